@@ -1,0 +1,217 @@
+"""End-to-end CPU training tests (reference analog:
+tests/python/test_basic.py, test_basic_models.py)."""
+import numpy as np
+import pytest
+
+import xgboost_amd as xgb
+from conftest import make_classification, make_regression
+
+
+def test_binary_classification_learns():
+    X, y = make_classification()
+    dtrain = xgb.DMatrix(X[:1500], label=y[:1500])
+    dvalid = xgb.DMatrix(X[1500:], label=y[1500:])
+    res = {}
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 4,
+                     "eta": 0.3, "eval_metric": ["logloss", "auc"]},
+                    dtrain, 30, evals=[(dvalid, "valid")],
+                    evals_result=res, verbose_eval=False)
+    assert res["valid"]["logloss"][-1] < 0.35
+    assert res["valid"]["auc"][-1] > 0.93
+    p = bst.predict(dvalid)
+    acc = ((p > 0.5) == y[1500:]).mean()
+    assert acc > 0.85
+    # monotone improvement in train loss
+    res2 = {}
+    xgb.train({"objective": "binary:logistic", "max_depth": 4, "eta": 0.3},
+              dtrain, 10, evals=[(dtrain, "train")], evals_result=res2,
+              verbose_eval=False)
+    ll = res2["train"]["logloss"]
+    assert ll[-1] < ll[0]
+
+
+def test_regression_squarederror():
+    X, y = make_regression()
+    dtrain = xgb.DMatrix(X[:1500], label=y[:1500])
+    dvalid = xgb.DMatrix(X[1500:], label=y[1500:])
+    res = {}
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 6,
+                     "eta": 0.3}, dtrain, 40, evals=[(dvalid, "valid")],
+                    evals_result=res, verbose_eval=False)
+    base_rmse = float(np.std(y[1500:]))
+    assert res["valid"]["rmse"][-1] < 0.5 * base_rmse
+    p = bst.predict(dvalid)
+    assert p.shape == (500,)
+
+
+def test_multiclass():
+    X, y = make_classification(n_class=4)
+    dtrain = xgb.DMatrix(X[:1500], label=y[:1500])
+    dvalid = xgb.DMatrix(X[1500:], label=y[1500:])
+    res = {}
+    bst = xgb.train({"objective": "multi:softprob", "num_class": 4,
+                     "max_depth": 4, "eta": 0.4}, dtrain, 15,
+                    evals=[(dvalid, "valid")], evals_result=res,
+                    verbose_eval=False)
+    p = bst.predict(dvalid)
+    assert p.shape == (500, 4)
+    assert np.allclose(p.sum(axis=1), 1.0, atol=1e-5)
+    acc = (p.argmax(axis=1) == y[1500:]).mean()
+    assert acc > 0.7
+    # softmax returns class ids
+    bst2 = xgb.train({"objective": "multi:softmax", "num_class": 4,
+                      "max_depth": 4, "eta": 0.4}, dtrain, 15,
+                     verbose_eval=False)
+    p2 = bst2.predict(dvalid)
+    assert p2.shape == (500,)
+    assert set(np.unique(p2)) <= {0.0, 1.0, 2.0, 3.0}
+
+
+def test_base_margin_and_base_score():
+    X, y = make_regression(500, 5)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "base_score": 3.0},
+                    d, 1, verbose_eval=False)
+    # first tree fits residuals vs 3.0
+    m = bst.predict(d, output_margin=True)
+    assert abs(np.mean(m) - (3.0 + np.mean(y - 3.0) * 0.3)) < 0.5
+
+
+def test_eta_zero_keeps_base():
+    X, y = make_regression(200, 3)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "eta": 0.0}, d, 2,
+                    verbose_eval=False)
+    p = bst.predict(d)
+    assert np.allclose(p, p[0])
+
+
+def test_max_depth_respected():
+    X, y = make_classification(1000, 8)
+    d = xgb.DMatrix(X, label=y)
+    for depth in (1, 3):
+        bst = xgb.train({"objective": "binary:logistic", "max_depth": depth},
+                        d, 3, verbose_eval=False)
+        for t in bst.trees:
+            assert t.max_depth() <= depth
+
+
+def test_max_leaves_lossguide():
+    X, y = make_classification(1000, 8)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "grow_policy": "lossguide",
+                     "max_leaves": 8, "max_depth": 0}, d, 3, verbose_eval=False)
+    for t in bst.trees:
+        n_leaves = sum(1 for nid in range(t.n_nodes) if t.is_leaf(nid))
+        assert n_leaves <= 8
+
+
+def test_custom_objective_and_metric():
+    X, y = make_regression(500, 5)
+    d = xgb.DMatrix(X, label=y)
+
+    def sq_obj(preds, dtrain):
+        g = preds - dtrain.get_label()
+        h = np.ones_like(g)
+        return g, h
+
+    def mae_metric(preds, dtrain):
+        return "my-mae", float(np.abs(preds - dtrain.get_label()).mean())
+
+    res = {}
+    xgb.train({"max_depth": 3, "eta": 0.3, "disable_default_eval_metric": 1},
+              d, 10, obj=sq_obj, custom_metric=mae_metric,
+              evals=[(d, "train")], evals_result=res, verbose_eval=False)
+    assert res["train"]["my-mae"][-1] < res["train"]["my-mae"][0]
+
+
+def test_early_stopping():
+    X, y = make_classification()
+    dtrain = xgb.DMatrix(X[:1500], label=y[:1500])
+    dvalid = xgb.DMatrix(X[1500:], label=y[1500:])
+    bst = xgb.train({"objective": "binary:logistic", "eta": 0.5,
+                     "max_depth": 6}, dtrain, 500,
+                    evals=[(dvalid, "valid")],
+                    early_stopping_rounds=5, verbose_eval=False)
+    assert bst.num_boosted_rounds() < 500
+    assert bst.best_iteration is not None
+
+
+def test_missing_values():
+    X, y = make_classification(1000, 5)
+    Xm = X.copy()
+    mask = np.random.RandomState(0).rand(*X.shape) < 0.2
+    Xm[mask] = np.nan
+    d = xgb.DMatrix(Xm, label=y)
+    res = {}
+    xgb.train({"objective": "binary:logistic", "max_depth": 4}, d, 10,
+              evals=[(d, "train")], evals_result=res, verbose_eval=False)
+    assert res["train"]["logloss"][-1] < 0.6
+
+
+def test_weights():
+    X, y = make_classification(1000, 5)
+    w = np.where(y == 1, 10.0, 1.0).astype(np.float32)
+    d = xgb.DMatrix(X, label=y, weight=w)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 3}, d, 5,
+                    verbose_eval=False)
+    p = bst.predict(d)
+    # heavily weighting positives shifts predictions up
+    d0 = xgb.DMatrix(X, label=y)
+    bst0 = xgb.train({"objective": "binary:logistic", "max_depth": 3}, d0, 5,
+                     verbose_eval=False)
+    assert p.mean() > bst0.predict(d0).mean()
+
+
+def test_subsample_colsample():
+    X, y = make_classification(2000, 10)
+    d = xgb.DMatrix(X, label=y)
+    res = {}
+    xgb.train({"objective": "binary:logistic", "max_depth": 4,
+               "subsample": 0.5, "colsample_bytree": 0.5,
+               "colsample_bylevel": 0.7, "colsample_bynode": 0.8,
+               "seed": 7}, d, 10, evals=[(d, "train")], evals_result=res,
+              verbose_eval=False)
+    assert res["train"]["logloss"][-1] < 0.5
+
+
+def test_determinism_same_seed():
+    X, y = make_classification(1000, 6)
+    d = xgb.DMatrix(X, label=y)
+    params = {"objective": "binary:logistic", "max_depth": 4,
+              "subsample": 0.8, "colsample_bytree": 0.8, "seed": 3}
+    p1 = xgb.train(params, d, 5, verbose_eval=False).predict(d)
+    d2 = xgb.DMatrix(X, label=y)
+    p2 = xgb.train(params, d2, 5, verbose_eval=False).predict(d2)
+    assert np.array_equal(p1, p2)
+
+
+def test_num_parallel_tree():
+    X, y = make_regression(800, 6)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "num_parallel_tree": 4,
+                     "max_depth": 3}, d, 3, verbose_eval=False)
+    assert len(bst.trees) == 12
+    assert bst.num_boosted_rounds() == 3
+
+
+def test_iteration_range_predict():
+    X, y = make_regression(500, 5)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror"}, d, 10, verbose_eval=False)
+    p5 = bst.predict(d, iteration_range=(0, 5))
+    sliced = bst[:5]
+    assert np.allclose(sliced.predict(d), p5, atol=1e-6)
+
+
+def test_pred_leaf():
+    X, y = make_classification(300, 4)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 3}, d, 4,
+                    verbose_eval=False)
+    leaves = bst.predict(d, pred_leaf=True)
+    assert leaves.shape == (300, 4)
+    for t in range(4):
+        tree = bst.trees[t]
+        for leaf_id in np.unique(leaves[:, t]).astype(int):
+            assert tree.is_leaf(leaf_id)

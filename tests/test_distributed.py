@@ -1,0 +1,175 @@
+"""Distributed data-parallel training over torch.distributed gloo
+(reference analog: tests/cpp/collective/test_worker.h in-process workers;
+our workers are real processes, world_size=2, loopback rendezvous)."""
+import os
+import pickle
+import subprocess
+import sys
+import tempfile
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import os, pickle, sys
+import numpy as np
+import torch
+import torch.distributed as dist
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+collective.init("gloo")
+
+rng = np.random.RandomState(0)
+n, f = 2000, 8
+X = rng.randn(n, f).astype(np.float32)
+w = rng.randn(f)
+y = (X @ w + 0.3 * rng.randn(n) > 0).astype(np.float32)
+
+# row shard
+shard = slice(rank * n // world, (rank + 1) * n // world)
+dtrain = xgb.DMatrix(X[shard], label=y[shard])
+params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+          "seed": 7, "debug_synchronize": True}
+bst = xgb.train(params, dtrain, 5, verbose_eval=False)
+
+# models must be identical across workers
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "model differs across workers"
+
+# predictions on the full data from the distributed model
+dfull = xgb.DMatrix(X, label=y)
+pred = bst.predict(dfull)
+out = os.environ["XGB_AMD_OUT"]
+if rank == 0:
+    with open(out, "wb") as fh:
+        pickle.dump({"pred": pred, "raw": raw}, fh)
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def _run_workers(world_size: int, script: str, env_extra=None) -> None:
+    procs = []
+    port = 29517
+    with tempfile.TemporaryDirectory() as td:
+        spath = os.path.join(td, "worker.py")
+        with open(spath, "w") as fh:
+            fh.write(script)
+        out = os.path.join(td, "out.pkl")
+        for r in range(world_size):
+            env = dict(os.environ)
+            env.update({
+                "RANK": str(r), "WORLD_SIZE": str(world_size),
+                "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                "XGB_AMD_REPO": REPO, "XGB_AMD_OUT": out,
+            })
+            if env_extra:
+                env.update(env_extra)
+            procs.append(subprocess.Popen(
+                [sys.executable, spath], env=env,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        outputs = []
+        ok = True
+        for p in procs:
+            stdout, _ = p.communicate(timeout=300)
+            outputs.append(stdout.decode())
+            ok = ok and p.returncode == 0
+        assert ok, "worker failed:\n" + "\n---\n".join(outputs)
+        with open(out, "rb") as fh:
+            return pickle.load(fh)
+
+
+def test_two_worker_training_sync():
+    res = _run_workers(2, WORKER)
+    # distributed model should learn the signal
+    rng = np.random.RandomState(0)
+    n, f = 2000, 8
+    X = rng.randn(n, f).astype(np.float32)
+    w = rng.randn(f)
+    y = (X @ w + 0.3 * rng.randn(n) > 0).astype(np.float32)
+    pred = res["pred"]
+    acc = ((pred > 0.5) == y).mean()
+    assert acc > 0.8, acc
+
+
+HIST_WORKER = r"""
+import os, sys
+import numpy as np
+import torch
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+from xgboost_amd import collective
+from xgboost_amd.data import DMatrix
+from xgboost_amd.quantile import make_cuts, HistogramCuts
+from xgboost_amd.data import quantize_dense
+from xgboost_amd.backend.cpu import CpuOps, GradQuantizer
+
+collective.init("gloo")
+rank = collective.get_rank()
+world = collective.get_world_size()
+
+rng = np.random.RandomState(1)
+n, f = 1000, 5
+X = rng.randn(n, f).astype(np.float32)
+gpair_full = torch.tensor(rng.randn(n, 2).astype(np.float32).clip(-1, 1))
+gpair_full[:, 1] = gpair_full[:, 1].abs() + 0.1
+
+cuts = make_cuts(X, 32)  # same cuts on all ranks
+
+shard = slice(rank * n // world, (rank + 1) * n // world)
+qm = quantize_dense(X[shard], cuts)
+ops = CpuOps(qm)
+gp = gpair_full[shard]
+quant = GradQuantizer(gp)   # max-abs is allreduced -> same scale everywhere
+qg = quant.quantize(gp)
+m = qg.shape[0]
+hist = ops.build_hist(qg, ops.make_ridx(m), [(0, m)])
+ops.allreduce_hist(hist)
+
+# oracle: single-process full-data histogram with the same (global) scale
+qm_full = quantize_dense(X, cuts)
+ops_full = CpuOps(qm_full)
+qg_full = quant.quantize(gpair_full)
+hist_full = ops_full.build_hist(qg_full, ops_full.make_ridx(n), [(0, n)])
+
+assert torch.equal(hist, hist_full), "allreduced hist != full-data hist"
+import torch.distributed as dist
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_hist_allreduce_exact():
+    _run_workers_simple(2, HIST_WORKER)
+
+
+def _run_workers_simple(world_size: int, script: str) -> None:
+    procs = []
+    port = 29531
+    with tempfile.TemporaryDirectory() as td:
+        spath = os.path.join(td, "worker.py")
+        with open(spath, "w") as fh:
+            fh.write(script)
+        for r in range(world_size):
+            env = dict(os.environ)
+            env.update({
+                "RANK": str(r), "WORLD_SIZE": str(world_size),
+                "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                "XGB_AMD_REPO": REPO,
+            })
+            procs.append(subprocess.Popen(
+                [sys.executable, spath], env=env,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        outputs = []
+        ok = True
+        for p in procs:
+            stdout, _ = p.communicate(timeout=300)
+            outputs.append(stdout.decode())
+            ok = ok and p.returncode == 0
+        assert ok, "worker failed:\n" + "\n---\n".join(outputs)

@@ -1,0 +1,137 @@
+"""Metric value spot checks (reference analog: tests/cpp/metric/*)."""
+import math
+
+import numpy as np
+import pytest
+
+from xgboost_amd.data import MetaInfo
+from xgboost_amd.metrics import create_metric
+
+
+def _info(y, weights=None, group=None, **kw):
+    y = np.asarray(y, np.float32)
+    info = MetaInfo(num_row=len(y), num_col=1, labels=y)
+    if weights is not None:
+        info.weights = np.asarray(weights, np.float32)
+    if group is not None:
+        info.group_ptr = np.concatenate([[0], np.cumsum(group)]).astype(np.int64)
+    for k, v in kw.items():
+        setattr(info, k, v)
+    return info
+
+
+def test_rmse():
+    m = create_metric("rmse")
+    assert m(np.array([0.0, 2.0]), _info([1.0, 1.0])) == pytest.approx(1.0)
+
+
+def test_mae():
+    m = create_metric("mae")
+    assert m(np.array([0.0, 3.0]), _info([1.0, 1.0])) == pytest.approx(1.5)
+
+
+def test_logloss():
+    m = create_metric("logloss")
+    v = m(np.array([0.9, 0.1]), _info([1.0, 0.0]))
+    assert v == pytest.approx(-math.log(0.9), rel=1e-6)
+
+
+def test_error_threshold():
+    m = create_metric("error")
+    preds = np.array([0.3, 0.7])
+    assert m(preds, _info([0.0, 0.0])) == pytest.approx(0.5)
+    m2 = create_metric("error@0.8")
+    assert m2(preds, _info([0.0, 0.0])) == pytest.approx(0.0)
+
+
+def test_auc_perfect_and_random():
+    m = create_metric("auc")
+    y = np.array([0, 0, 1, 1], np.float32)
+    assert m(np.array([0.1, 0.2, 0.8, 0.9]), _info(y)) == pytest.approx(1.0)
+    assert m(np.array([0.9, 0.8, 0.2, 0.1]), _info(y)) == pytest.approx(0.0)
+    # ties give 0.5
+    assert m(np.array([0.5, 0.5, 0.5, 0.5]), _info(y)) == pytest.approx(0.5)
+
+
+def test_auc_weighted():
+    m = create_metric("auc")
+    y = np.array([0, 1], np.float32)
+    v = m(np.array([0.4, 0.6]), _info(y, weights=[2.0, 3.0]))
+    assert v == pytest.approx(1.0)
+
+
+def test_aucpr_range():
+    m = create_metric("aucpr")
+    y = np.array([0, 0, 1, 1], np.float32)
+    v = m(np.array([0.1, 0.2, 0.8, 0.9]), _info(y))
+    assert 0.99 <= v <= 1.0
+
+
+def test_merror_mlogloss():
+    y = np.array([0, 1, 2], np.float32)
+    p = np.array([[0.8, 0.1, 0.1], [0.1, 0.8, 0.1], [0.8, 0.1, 0.1]])
+    assert create_metric("merror")(p, _info(y)) == pytest.approx(1 / 3)
+    assert create_metric("mlogloss")(p, _info(y)) == pytest.approx(
+        -(math.log(0.8) * 2 + math.log(0.1)) / 3, rel=1e-6)
+
+
+def test_ndcg():
+    m = create_metric("ndcg")
+    y = np.array([3, 2, 1, 0], np.float32)
+    perfect = m(np.array([4.0, 3.0, 2.0, 1.0]), _info(y, group=[4]))
+    assert perfect == pytest.approx(1.0)
+    worse = m(np.array([1.0, 2.0, 3.0, 4.0]), _info(y, group=[4]))
+    assert worse < 1.0
+
+
+def test_ndcg_topn():
+    m = create_metric("ndcg@2")
+    y = np.array([0, 0, 1, 1], np.float32)
+    v = m(np.array([0.9, 0.8, 0.2, 0.1]), _info(y, group=[4]))
+    assert v == pytest.approx(0.0)
+
+
+def test_map():
+    m = create_metric("map")
+    y = np.array([1, 0, 1, 0], np.float32)
+    v = m(np.array([0.9, 0.8, 0.7, 0.1]), _info(y, group=[4]))
+    # AP = (1/1 + 2/3) / 2
+    assert v == pytest.approx((1.0 + 2 / 3) / 2, rel=1e-6)
+
+
+def test_rmsle():
+    m = create_metric("rmsle")
+    v = m(np.array([math.e - 1]), _info([0.0]))
+    assert v == pytest.approx(1.0, rel=1e-6)
+
+
+def test_mape():
+    m = create_metric("mape")
+    assert m(np.array([2.0]), _info([1.0])) == pytest.approx(1.0)
+
+
+def test_quantile_metric():
+    m = create_metric("quantile@0.9")
+    # under-prediction penalized by alpha
+    assert m(np.array([0.0]), _info([1.0])) == pytest.approx(0.9)
+    assert m(np.array([1.0]), _info([0.0])) == pytest.approx(0.1)
+
+
+def test_poisson_nloglik():
+    m = create_metric("poisson-nloglik")
+    v = m(np.array([2.0]), _info([2.0]))
+    expected = 2.0 - 2.0 * math.log(2.0) + math.lgamma(3.0)
+    assert v == pytest.approx(expected, rel=1e-6)
+
+
+def test_interval_accuracy():
+    m = create_metric("interval-regression-accuracy")
+    info = _info([1.0, 1.0],
+                 label_lower_bound=np.array([0.5, 2.0], np.float32),
+                 label_upper_bound=np.array([1.5, 3.0], np.float32))
+    assert m(np.array([1.0, 1.0]), info) == pytest.approx(0.5)
+
+
+def test_unknown_metric_raises():
+    with pytest.raises(ValueError):
+        create_metric("bogus")

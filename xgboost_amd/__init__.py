@@ -1,0 +1,36 @@
+"""xgboost_amd — an MI355X-native gradient-boosted tree framework.
+
+A from-scratch re-implementation of the capabilities of dmlc/xgboost,
+designed for AMD Instinct MI355X (gfx950/CDNA4): hand-written HIP
+kernels for the training hot path (quantize, histogram build, split
+evaluation, partition, predict), torch tensors for memory management,
+and torch.distributed (RCCL over xGMI) for multi-GPU data parallelism.
+
+Public API mirrors the xgboost Python package: DMatrix, QuantileDMatrix,
+Booster, train, cv, callbacks, sklearn wrappers.
+"""
+from .collective import init as collective_init  # noqa: F401
+from .core import Booster  # noqa: F401
+from .data import DMatrix, QuantileDMatrix  # noqa: F401
+from .training import cv, train  # noqa: F401
+from . import callback  # noqa: F401
+from . import collective  # noqa: F401
+
+__version__ = "0.1.0"
+
+__all__ = [
+    "Booster", "DMatrix", "QuantileDMatrix", "train", "cv", "callback",
+    "collective",
+]
+
+
+def _lazy(name):
+    if name in ("XGBRegressor", "XGBClassifier", "XGBRanker", "XGBRFRegressor",
+                "XGBRFClassifier", "XGBModel"):
+        from . import sklearn as _sk
+        return getattr(_sk, name)
+    raise AttributeError(name)
+
+
+def __getattr__(name):
+    return _lazy(name)

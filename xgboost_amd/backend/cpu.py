@@ -1,0 +1,153 @@
+"""CPU backend ops — torch/numpy implementations of the tree-building
+primitives.
+
+This is both the production CPU training path (reference analog:
+grow_quantile_histmaker, src/tree/updater_quantile_hist.cc:664) and the
+fp64 numerics oracle the HIP kernels are tested against (SURVEY.md §4:
+CPU<->GPU cross-check is the primary kernel oracle).
+
+Determinism: gradients are quantized to int32 fixed point and histograms
+accumulated in int64, the same scheme the HIP histogram kernel uses
+(reference: src/tree/gpu_hist/quantiser.cuh:52) — so CPU and GPU
+histograms are bit-identical regardless of accumulation order.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .. import collective
+from ..data import QuantizedMatrix
+from ..params import TrainParam
+from ..splits import SplitEntry, evaluate_splits_np
+
+QSHIFT = 30  # fixed-point fraction bits; sums of 2^30-scaled int32 fit int64
+
+
+class GradQuantizer:
+    """Fixed-point gradient quantizer (reference quantiser.cuh:52).
+
+    scale = 2^30 / global_max_abs (allreduced), per g and h component.
+    """
+
+    def __init__(self, gpair: torch.Tensor):
+        g = gpair[..., 0]
+        h = gpair[..., 1]
+        max_g = float(g.abs().max()) if g.numel() else 0.0
+        max_h = float(h.abs().max()) if h.numel() else 0.0
+        max_g, max_h = collective.allreduce_max_scalars([max_g, max_h])
+        self.g_scale = (1 << QSHIFT) / max_g if max_g > 0 else 1.0
+        self.h_scale = (1 << QSHIFT) / max_h if max_h > 0 else 1.0
+
+    def quantize(self, gpair: torch.Tensor) -> torch.Tensor:
+        """float32 [n, 2] -> int32 [n, 2] (round half away from zero)."""
+        scale = torch.tensor([self.g_scale, self.h_scale],
+                             dtype=torch.float64, device=gpair.device)
+        return torch.round(gpair.double() * scale).to(torch.int32)
+
+    def dequantize_pair(self, qg: int, qh: int) -> Tuple[float, float]:
+        return qg / self.g_scale, qh / self.h_scale
+
+    def dequantize_hist(self, hist: torch.Tensor) -> np.ndarray:
+        """int64 [..., 2] -> float64 numpy."""
+        h = hist.to("cpu").numpy().astype(np.float64)
+        h[..., 0] /= self.g_scale
+        h[..., 1] /= self.h_scale
+        return h
+
+
+class CpuOps:
+    """Tree-building primitive ops on CPU tensors."""
+
+    device = torch.device("cpu")
+
+    def __init__(self, qm: QuantizedMatrix):
+        self.qm = qm
+        self.gidx_global = qm.global_gidx()  # int64 [n, f], -1 missing
+        self.n_bins = qm.cuts.total_bins
+
+    def make_ridx(self, n_rows: int) -> torch.Tensor:
+        return torch.arange(n_rows, dtype=torch.int64)
+
+    def root_sum(self, qgpair: torch.Tensor) -> Tuple[int, int]:
+        s = qgpair.to(torch.int64).sum(dim=0)
+        t = s.clone()
+        collective.allreduce_sum_(t)
+        return int(t[0]), int(t[1])
+
+    def build_hist(self, qgpair: torch.Tensor, ridx: torch.Tensor,
+                   segments: Sequence[Tuple[int, int]]) -> torch.Tensor:
+        """-> int64 [len(segments), n_bins, 2]."""
+        k = len(segments)
+        out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64)
+        for i, (s, e) in enumerate(segments):
+            rows = ridx[s:e]
+            g = self.gidx_global[rows]            # [m, f]
+            valid = g >= 0
+            flat = g[valid]
+            qg = qgpair[rows, 0].to(torch.int64).unsqueeze(1).expand_as(g)[valid]
+            qh = qgpair[rows, 1].to(torch.int64).unsqueeze(1).expand_as(g)[valid]
+            out[i, :, 0].index_add_(0, flat, qg)
+            out[i, :, 1].index_add_(0, flat, qh)
+        return out
+
+    def allreduce_hist(self, hist: torch.Tensor) -> torch.Tensor:
+        collective.allreduce_sum_(hist)
+        return hist
+
+    def evaluate_splits(self, hist: torch.Tensor, quantizer: GradQuantizer,
+                        parent_sums: Sequence[Tuple[int, int]],
+                        nids: Sequence[int], param: TrainParam,
+                        feature_sets=None, monotone=None, cat_mask=None,
+                        node_bounds=None) -> List[SplitEntry]:
+        h = quantizer.dequantize_hist(hist)
+        # parent_sums arrive already dequantized (float g, h)
+        pg = np.array([s[0] for s in parent_sums], dtype=np.float64)
+        ph = np.array([s[1] for s in parent_sums], dtype=np.float64)
+        return evaluate_splits_np(h, pg, ph, nids, self.qm.cuts.ptrs, param,
+                                  feature_sets=feature_sets, monotone=monotone,
+                                  cat_mask=cat_mask, node_bounds=node_bounds)
+
+    def partition(self, ridx: torch.Tensor,
+                  segments: Sequence[Tuple[int, int]],
+                  splits: Sequence[SplitEntry]
+                  ) -> List[Tuple[Tuple[int, int], Tuple[int, int]]]:
+        """Reorder ridx within each segment into [left | right].
+
+        Returns [(left_seg, right_seg), ...].  Stable on CPU (the GPU
+        kernel is unstable; histogram sums don't depend on order).
+        """
+        out = []
+        for (s, e), sp in zip(segments, splits):
+            rows = ridx[s:e]
+            bins = self.gidx_global[rows, sp.feature]
+            missing = bins < 0
+            if sp.is_cat:
+                fstart = int(self.qm.cuts.ptrs[sp.feature])
+                local = bins - fstart
+                go_right = torch.zeros_like(missing)
+                cats = torch.as_tensor(np.asarray(sp.cat_bits, np.int64))
+                go_right = torch.isin(local, cats)
+                go_left = ~go_right
+            else:
+                go_left = bins <= sp.split_bin
+            go_left = torch.where(missing,
+                                  torch.tensor(bool(sp.default_left)), go_left)
+            left_rows = rows[go_left]          # advanced indexing: copies
+            right_rows = rows[~go_left]        # copy BEFORE writing into ridx
+            nl = int(left_rows.numel())
+            ridx[s:s + nl] = left_rows
+            ridx[s + nl:e] = right_rows
+            out.append(((s, s + nl), (s + nl, e)))
+        return out
+
+    def leaf_partition(self, ridx: torch.Tensor,
+                       leaf_segments: Sequence[Tuple[int, int, int]],
+                       n_rows: int) -> torch.Tensor:
+        """-> int32 [n_rows] leaf node id per row (for prediction cache)."""
+        pos = torch.zeros(n_rows, dtype=torch.int32)
+        for nid, s, e in leaf_segments:
+            pos[ridx[s:e]] = nid
+        return pos

@@ -1,0 +1,651 @@
+"""Booster — training state machine, prediction, model IO.
+
+Reference behavior: src/learner.cc (LearnerImpl), src/gbm/gbtree.cc
+(GBTree::DoBoost/PredictBatch, prediction cache gbtree.h:192), JSON
+model schema learner.cc:868-899.
+
+The public API mirrors xgboost's Python Booster so reference users can
+switch directly: update/boost/eval_set/predict/save_model/load_model/
+dump_model/get_score/attributes/copy/slice.
+"""
+from __future__ import annotations
+
+import json
+import math
+import os
+from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
+
+import numpy as np
+import torch
+
+from . import collective
+from .backend.cpu import CpuOps, GradQuantizer
+from .data import DMatrix, QuantizedMatrix
+from .grower import TreeGrower
+from .metrics import create_metric
+from .objectives import Objective, create_objective
+from .params import (LEARNER_PARAMS, TrainParam, canonicalize, check_unknown,
+                     make_train_param)
+from .tree_model import RegTree
+
+VERSION = (3, 5, 0)
+
+
+def _resolve_device(params: Dict[str, Any]) -> torch.device:
+    dev = str(params.get("device", "cpu"))
+    if dev in ("cuda", "gpu"):
+        return torch.device("cuda", torch.cuda.current_device()
+                            if torch.cuda.is_available() else 0)
+    if dev.startswith(("cuda:", "gpu:")):
+        return torch.device("cuda", int(dev.split(":")[1]))
+    return torch.device("cpu")
+
+
+class Booster:
+    def __init__(self, params: Optional[Dict[str, Any]] = None,
+                 cache: Optional[Sequence[DMatrix]] = None,
+                 model_file: Optional[str] = None):
+        params = dict(params or {})
+        self.raw_params = canonicalize(params)
+        check_unknown(self.raw_params,
+                      bool(self.raw_params.get("validate_parameters", False)))
+        booster_kind = self.raw_params.get("booster", "gbtree")
+        if booster_kind == "dart":
+            booster_kind = "gbtree"  # DART deprecated -> gbtree (learner.cc:225)
+        if booster_kind == "gblinear":
+            raise NotImplementedError("gblinear is not implemented yet")
+        self.tparam: TrainParam = make_train_param(self.raw_params)
+        self.device = _resolve_device(self.raw_params)
+        obj_name = str(self.raw_params.get("objective", "reg:squarederror"))
+        self.objective: Objective = create_objective(obj_name, self.raw_params)
+        self.seed = int(self.raw_params.get("seed", 0))
+        self.seed_per_iteration = bool(self.raw_params.get("seed_per_iteration", False))
+
+        self.trees: List[RegTree] = []
+        self.tree_info: List[int] = []
+        self.iteration_indptr: List[int] = [0]
+        self.n_features: Optional[int] = None
+        self.n_targets = int(self.raw_params.get("num_target", 1))
+        self.base_score: Optional[float] = (
+            float(self.raw_params["base_score"])
+            if "base_score" in self.raw_params else None)
+        self._base_score_estimated = "base_score" in self.raw_params
+        self.attributes_: Dict[str, str] = {}
+        self.feature_names: Optional[List[str]] = None
+        self.feature_types: Optional[List[str]] = None
+        self.best_iteration: Optional[int] = None
+        self.best_score: Optional[float] = None
+
+        self._cache: Dict[int, Tuple[torch.Tensor, int]] = {}  # id(dmat) -> (margin, version)
+        self._ops_cache: Dict[int, Any] = {}
+        if model_file is not None:
+            self.load_model(model_file)
+        if cache:
+            for d in cache:
+                self._maybe_set_meta(d)
+
+    # ------------------------------------------------------------------
+    @property
+    def n_outputs(self) -> int:
+        return self.objective.n_outputs(self.n_targets)
+
+    def num_boosted_rounds(self) -> int:
+        return len(self.iteration_indptr) - 1
+
+    def num_features(self) -> int:
+        return self.n_features or 0
+
+    def _maybe_set_meta(self, dmat: DMatrix) -> None:
+        if self.n_features is None:
+            self.n_features = dmat.num_col()
+        elif self.n_features != dmat.num_col():
+            raise ValueError(
+                f"feature count mismatch: model has {self.n_features}, "
+                f"data has {dmat.num_col()}")
+        if self.feature_names is None:
+            self.feature_names = dmat.feature_names
+        if self.feature_types is None:
+            self.feature_types = dmat.feature_types
+        if self.n_targets == 1 and dmat.info.labels is not None \
+                and dmat.info.labels.ndim == 2 and dmat.info.labels.shape[1] > 1 \
+                and self.objective.task == "regression":
+            self.n_targets = dmat.info.labels.shape[1]
+
+    def _init_base_score(self, dtrain: DMatrix) -> None:
+        if self.base_score is not None and self._base_score_estimated:
+            return
+        if self.base_score is None:
+            if dtrain.info.labels is None:
+                self.base_score = 0.5
+            else:
+                self.base_score = float(
+                    self.objective.init_estimation(dtrain.info))
+                self.base_score = collective.broadcast_obj(self.base_score, 0)
+        self._base_score_estimated = True
+
+    def _base_margin_value(self) -> float:
+        bs = self.base_score if self.base_score is not None else 0.5
+        return float(self.objective.prob_to_margin(bs))
+
+    def _ops_for(self, dmat: DMatrix):
+        key = id(dmat)
+        ops = self._ops_cache.get(key)
+        if ops is None:
+            from .sketch import sketch_cuts
+            max_bin = self.tparam.max_bin
+            qm = dmat.quantized(max_bin, sketch_fn=sketch_cuts)
+            if self.device.type == "cuda":
+                from .backend.gpu import GpuOps
+                ops = GpuOps(qm.to(self.device))
+            else:
+                ops = CpuOps(qm)
+            self._ops_cache[key] = ops
+        return ops
+
+    # -- training ------------------------------------------------------
+    def update(self, dtrain: DMatrix, iteration: int,
+               fobj=None) -> None:
+        self._maybe_set_meta(dtrain)
+        self._init_base_score(dtrain)
+        margin = self._cached_margin(dtrain)
+        if fobj is not None:
+            preds = self.objective.pred_transform(margin).cpu().numpy()
+            grad, hess = fobj(np.squeeze(preds), dtrain)
+            grad = torch.as_tensor(np.asarray(grad, np.float32),
+                                   device=margin.device).view(margin.shape)
+            hess = torch.as_tensor(np.asarray(hess, np.float32),
+                                   device=margin.device).view(margin.shape)
+        else:
+            grad, hess = self.objective.get_gradient(margin, dtrain.info, iteration)
+        self.boost_gpair(dtrain, grad, hess, iteration)
+
+    def boost(self, dtrain: DMatrix, iteration: int = 0, grad=None, hess=None) -> None:
+        """Boost with user-supplied gradients (XGBoosterBoostOneIter)."""
+        self._maybe_set_meta(dtrain)
+        if self.base_score is None:
+            self.base_score = 0.5
+            self._base_score_estimated = True
+        dev = self.device
+        g = torch.as_tensor(np.asarray(grad, np.float32), device=dev)
+        h = torch.as_tensor(np.asarray(hess, np.float32), device=dev)
+        n = dtrain.num_row()
+        g = g.view(n, -1)
+        h = h.view(n, -1)
+        self.boost_gpair(dtrain, g, h, iteration)
+
+    def boost_gpair(self, dtrain: DMatrix, grad: torch.Tensor,
+                    hess: torch.Tensor, iteration: int) -> None:
+        ops = self._ops_for(dtrain)
+        n = dtrain.num_row()
+        n_out = grad.shape[1] if grad.dim() > 1 else 1
+        grad = grad.view(n, n_out)
+        hess = hess.view(n, n_out)
+        margin, version = self._cache[id(dtrain)]
+        seed = (self.seed + iteration if not self.seed_per_iteration
+                else self.seed + iteration * 2654435761)
+        new_trees = 0
+        eta_scale = 1.0 / max(1, self.tparam.num_parallel_tree)
+        for k in range(n_out):
+            for ptree in range(self.tparam.num_parallel_tree):
+                gpair = torch.stack([grad[:, k], hess[:, k]], dim=1).contiguous()
+                gpair = self._subsample(gpair, seed + 7919 * ptree + 104729 * k)
+                gpair = gpair.to(ops.device)
+                quantizer = GradQuantizer(gpair)
+                qg = quantizer.quantize(gpair)
+                tree = RegTree(self.n_features)
+                grower = TreeGrower(
+                    ops, self._scaled_param(eta_scale), quantizer, n,
+                    seed=seed + 31 * ptree + 17 * k,
+                    monotone=self._monotone_array(),
+                    interaction=self.tparam.interaction_constraints)
+                tree, positions = grower.grow(qg, tree)
+                self.trees.append(tree)
+                self.tree_info.append(k)
+                new_trees += 1
+                # update prediction cache from leaf positions
+                leaf_vals = torch.as_tensor(
+                    tree.split_cond[:tree.n_nodes].copy(), device=margin.device)
+                margin[:, k] += leaf_vals[positions.to(margin.device).long()]
+        self.iteration_indptr.append(self.iteration_indptr[-1] + new_trees)
+        self._cache[id(dtrain)] = (margin, len(self.trees))
+        if self.tparam.debug_synchronize:
+            collective.check_synchronized(
+                json.dumps(self.trees[-1].to_json()).encode(), "tree")
+
+    def _scaled_param(self, eta_scale: float) -> TrainParam:
+        if eta_scale == 1.0:
+            return self.tparam
+        import dataclasses as dc
+        p = dc.replace(self.tparam)
+        p.eta = self.tparam.eta * eta_scale
+        return p
+
+    def _monotone_array(self) -> Optional[np.ndarray]:
+        mc = self.tparam.monotone_constraints
+        if mc is None:
+            return None
+        if isinstance(mc, str):
+            mc = [int(x) for x in mc.strip("()[] ").split(",") if x.strip()]
+        arr = np.zeros(self.n_features, dtype=np.int64)
+        arr[:len(mc)] = np.asarray(list(mc), dtype=np.int64)[:self.n_features]
+        return arr
+
+    def _subsample(self, gpair: torch.Tensor, seed: int) -> torch.Tensor:
+        p = self.tparam.subsample
+        if p >= 1.0:
+            return gpair
+        gen = torch.Generator(device="cpu").manual_seed(seed & 0x7FFFFFFF)
+        if self.tparam.sampling_method == "gradient_based":
+            # MVS-style: keep prob proportional to sqrt(g^2 + lambda*h^2)
+            g = gpair[:, 0].cpu()
+            h = gpair[:, 1].cpu()
+            score = torch.sqrt(g * g + self.tparam.reg_lambda * h * h)
+            n = g.numel()
+            threshold = _sample_rate_threshold(score, p)
+            u = torch.rand(n, generator=gen)
+            keep_prob = torch.clamp(score / threshold, max=1.0)
+            keep = u < keep_prob
+            out = gpair.clone()
+            scale = (1.0 / keep_prob.clamp(min=1e-16)).to(gpair.device)
+            out[:, 0] = torch.where(keep.to(gpair.device),
+                                    gpair[:, 0] * scale, torch.zeros_like(gpair[:, 0]))
+            out[:, 1] = torch.where(keep.to(gpair.device),
+                                    gpair[:, 1] * scale, torch.zeros_like(gpair[:, 1]))
+            return out
+        mask = (torch.rand(gpair.shape[0], generator=gen) < p).to(gpair.device)
+        out = gpair.clone()
+        out[~mask] = 0.0
+        return out
+
+    # -- prediction ----------------------------------------------------
+    def _cached_margin(self, dmat: DMatrix) -> torch.Tensor:
+        key = id(dmat)
+        n = dmat.num_row()
+        entry = self._cache.get(key)
+        if entry is not None and entry[1] == len(self.trees):
+            return entry[0]
+        margin = self._predict_margin(dmat)
+        self._cache[key] = (margin, len(self.trees))
+        return margin
+
+    def _predict_margin(self, dmat: DMatrix,
+                        iteration_range: Tuple[int, int] = (0, 0)
+                        ) -> torch.Tensor:
+        n = dmat.num_row()
+        out = torch.full((n, self.n_outputs), self._base_margin_value(),
+                         dtype=torch.float32, device=self.device)
+        if dmat.info.base_margin is not None:
+            bm = torch.as_tensor(dmat.info.base_margin, dtype=torch.float32,
+                                 device=self.device)
+            out = bm.view(n, -1).expand(n, self.n_outputs).clone()
+        lo, hi = self._tree_range(iteration_range)
+        X = dmat.raw_data()
+        if self.device.type == "cuda" and (hi - lo) > 0:
+            from .backend.gpu import predict_margin_gpu
+            return predict_margin_gpu(self, dmat, out, lo, hi)
+        for t in range(lo, hi):
+            tree = self.trees[t]
+            pos = tree.predict_leaf_np(X, dmat.missing)
+            vals = tree.split_cond[:tree.n_nodes][pos]
+            out[:, self.tree_info[t]] += torch.as_tensor(vals, device=out.device)
+        return out
+
+    def _tree_range(self, iteration_range: Tuple[int, int]) -> Tuple[int, int]:
+        lo_it, hi_it = iteration_range
+        if hi_it == 0:
+            return (self.iteration_indptr[min(lo_it, len(self.iteration_indptr) - 1)],
+                    len(self.trees))
+        hi_it = min(hi_it, self.num_boosted_rounds())
+        return (self.iteration_indptr[lo_it], self.iteration_indptr[hi_it])
+
+    def predict(self, data: DMatrix, output_margin: bool = False,
+                pred_leaf: bool = False, pred_contribs: bool = False,
+                approx_contribs: bool = False, pred_interactions: bool = False,
+                validate_features: bool = True,
+                training: bool = False,
+                iteration_range: Tuple[int, int] = (0, 0),
+                strict_shape: bool = False) -> np.ndarray:
+        if not isinstance(data, DMatrix):
+            raise TypeError("predict expects a DMatrix; see inplace_predict")
+        if validate_features and self.n_features is not None \
+                and data.num_col() != self.n_features:
+            raise ValueError(
+                f"feature mismatch: {data.num_col()} vs {self.n_features}")
+        if pred_leaf:
+            lo, hi = self._tree_range(iteration_range)
+            X = data.raw_data()
+            out = np.stack([self.trees[t].predict_leaf_np(X, data.missing)
+                            for t in range(lo, hi)], axis=1)
+            return out.astype(np.float32)
+        if pred_interactions:
+            from .shap import shap_interactions
+            return shap_interactions(self, data, iteration_range)
+        if pred_contribs:
+            from .shap import shap_values
+            return shap_values(self, data, iteration_range, approx=approx_contribs)
+        margin = self._predict_margin(data, iteration_range)
+        if output_margin:
+            res = margin
+        else:
+            res = self.objective.pred_transform(margin)
+        arr = res.cpu().numpy()
+        if not strict_shape and arr.ndim == 2 and arr.shape[1] == 1:
+            arr = arr.reshape(-1)
+        return arr
+
+    def inplace_predict(self, data, iteration_range=(0, 0),
+                        predict_type: str = "value", missing: float = np.nan,
+                        validate_features: bool = True, base_margin=None,
+                        strict_shape: bool = False) -> np.ndarray:
+        d = DMatrix(data, missing=missing, base_margin=base_margin)
+        return self.predict(d, output_margin=(predict_type == "margin"),
+                            validate_features=validate_features,
+                            iteration_range=iteration_range,
+                            strict_shape=strict_shape)
+
+    # -- evaluation ----------------------------------------------------
+    def eval_set(self, evals: Sequence[Tuple[DMatrix, str]],
+                 iteration: int = 0, feval=None,
+                 output_margin: bool = True) -> str:
+        parts = [f"[{iteration}]"]
+        metric_names = self._metric_names()
+        for dmat, name in evals:
+            margin = self._cached_margin(dmat)
+            transformed = self.objective.pred_transform(margin)
+            tnp = transformed.cpu().numpy()
+            if tnp.shape[1] == 1:
+                tnp = tnp.reshape(-1)
+            for mname in metric_names:
+                m = create_metric(mname)
+                val = m(tnp, dmat.info)
+                parts.append(f"{name}-{mname}:{val:.5f}" if abs(val) >= 1e-5
+                             else f"{name}-{mname}:{val:g}")
+            if feval is not None:
+                res = feval(tnp, dmat)
+                if isinstance(res, list):
+                    for mn, v in res:
+                        parts.append(f"{name}-{mn}:{v:g}")
+                else:
+                    mn, v = res
+                    parts.append(f"{name}-{mn}:{v:g}")
+        return "\t".join(parts)
+
+    def _metric_names(self) -> List[str]:
+        em = self.raw_params.get("eval_metric")
+        if em is None:
+            if self.raw_params.get("disable_default_eval_metric"):
+                return []
+            dm = self.objective.default_metric
+            return [dm]
+        if isinstance(em, (list, tuple)):
+            return [str(m) for m in em]
+        return [str(em)]
+
+    def eval(self, data: DMatrix, name: str = "eval", iteration: int = 0) -> str:
+        return self.eval_set([(data, name)], iteration)
+
+    # -- attributes ----------------------------------------------------
+    def attr(self, key: str) -> Optional[str]:
+        return self.attributes_.get(key)
+
+    def set_attr(self, **kwargs) -> None:
+        for k, v in kwargs.items():
+            if v is None:
+                self.attributes_.pop(k, None)
+            else:
+                self.attributes_[k] = str(v)
+
+    def attributes(self) -> Dict[str, str]:
+        return dict(self.attributes_)
+
+    def set_param(self, params, value=None) -> None:
+        if isinstance(params, str):
+            params = {params: value}
+        elif isinstance(params, (list, tuple)):
+            params = dict(params)
+        self.raw_params.update(canonicalize(dict(params)))
+        self.tparam = make_train_param(self.raw_params)
+        self.device = _resolve_device(self.raw_params)
+
+    # -- model IO ------------------------------------------------------
+    def save_model(self, fname: str) -> None:
+        j = self._model_to_json()
+        if str(fname).endswith(".ubj"):
+            from .ubjson import dump_ubjson
+            with open(fname, "wb") as fh:
+                dump_ubjson(j, fh)
+        else:
+            with open(fname, "w") as fh:
+                json.dump(j, fh)
+
+    def save_raw(self, raw_format: str = "ubj") -> bytearray:
+        j = self._model_to_json()
+        if raw_format == "json":
+            return bytearray(json.dumps(j).encode())
+        from .ubjson import dumps_ubjson
+        return bytearray(dumps_ubjson(j))
+
+    def load_model(self, fname) -> None:
+        if isinstance(fname, (bytes, bytearray)):
+            data = bytes(fname)
+            j = _parse_model_bytes(data, force_ubj=False)
+        else:
+            with open(fname, "rb") as fh:
+                data = fh.read()
+            j = _parse_model_bytes(data,
+                                   force_ubj=str(fname).endswith(".ubj"))
+        self._model_from_json(j)
+
+    def _model_to_json(self) -> dict:
+        learner = {
+            "attributes": dict(self.attributes_),
+            "feature_names": self.feature_names or [],
+            "feature_types": self.feature_types or [],
+            "gradient_booster": {
+                "model": {
+                    "gbtree_model_param": {
+                        "num_trees": str(len(self.trees)),
+                        "num_parallel_tree": str(self.tparam.num_parallel_tree),
+                    },
+                    "iteration_indptr": list(self.iteration_indptr),
+                    "tree_info": list(self.tree_info),
+                    "trees": [t.to_json(i) for i, t in enumerate(self.trees)],
+                },
+                "name": "gbtree",
+            },
+            "learner_model_param": {
+                "base_score": f"{self.base_score if self.base_score is not None else 0.5:.9E}",
+                "boost_from_average": "1",
+                "num_class": str(getattr(self.objective, "num_class", 0)
+                                 if self.objective.task == "multiclass" else 0),
+                "num_feature": str(self.n_features or 0),
+                "num_target": str(self.n_targets),
+            },
+            "objective": self.objective.save_config(),
+        }
+        return {"learner": learner, "version": list(VERSION)}
+
+    def _model_from_json(self, j: dict) -> None:
+        learner = j["learner"]
+        lmp = learner["learner_model_param"]
+        self.n_features = int(lmp["num_feature"])
+        self.n_targets = max(1, int(lmp.get("num_target", "1")))
+        base_score_s = lmp.get("base_score", "5E-1")
+        self.base_score = float(base_score_s)
+        self._base_score_estimated = True
+        obj_cfg = learner["objective"]
+        obj_name = obj_cfg["name"]
+        obj_params = dict(self.raw_params)
+        for v in obj_cfg.values():
+            if isinstance(v, dict):
+                obj_params.update(v)
+        num_class = int(lmp.get("num_class", "0"))
+        if num_class > 1:
+            obj_params["num_class"] = num_class
+        self.objective = create_objective(obj_name, obj_params)
+        self.raw_params["objective"] = obj_name
+        gb = learner["gradient_booster"]
+        model = gb["model"]
+        self.trees = [RegTree.from_json(t) for t in model["trees"]]
+        self.tree_info = [int(x) for x in model["tree_info"]]
+        indptr = model.get("iteration_indptr")
+        if indptr:
+            self.iteration_indptr = [int(x) for x in indptr]
+        else:
+            n_group = max(1, num_class)
+            per_iter = n_group * self.tparam.num_parallel_tree
+            self.iteration_indptr = list(
+                range(0, len(self.trees) + 1, per_iter))
+        self.attributes_ = {k: str(v) for k, v in
+                            learner.get("attributes", {}).items()}
+        fn = learner.get("feature_names") or []
+        ft = learner.get("feature_types") or []
+        self.feature_names = list(fn) if fn else None
+        self.feature_types = list(ft) if ft else None
+        self._cache.clear()
+        self._ops_cache.clear()
+
+    def __getstate__(self):
+        state = {"raw": bytes(self.save_raw("json")),
+                 "params": self.raw_params}
+        return state
+
+    def __setstate__(self, state):
+        self.__init__(state["params"])
+        self.load_model(state["raw"])
+
+    def __copy__(self):
+        return self.copy()
+
+    def __deepcopy__(self, memo):
+        return self.copy()
+
+    def copy(self) -> "Booster":
+        b = Booster(self.raw_params)
+        b.load_model(self.save_raw("json"))
+        return b
+
+    def __getitem__(self, val) -> "Booster":
+        if isinstance(val, int):
+            val = slice(val, val + 1)
+        lo, hi, step = val.indices(self.num_boosted_rounds())
+        b = Booster(self.raw_params)
+        b.n_features = self.n_features
+        b.n_targets = self.n_targets
+        b.base_score = self.base_score
+        b._base_score_estimated = True
+        b.feature_names = self.feature_names
+        b.feature_types = self.feature_types
+        b.iteration_indptr = [0]
+        for it in range(lo, hi, step):
+            s, e = self.iteration_indptr[it], self.iteration_indptr[it + 1]
+            for t in range(s, e):
+                b.trees.append(self.trees[t])
+                b.tree_info.append(self.tree_info[t])
+            b.iteration_indptr.append(len(b.trees))
+        return b
+
+    def slice(self, begin: int, end: int = 0, step: int = 1) -> "Booster":
+        return self[slice(begin, end or self.num_boosted_rounds(), step)]
+
+    # -- introspection -------------------------------------------------
+    def get_dump(self, fmap: str = "", with_stats: bool = False,
+                 dump_format: str = "text") -> List[str]:
+        names = self.feature_names
+        return [t.dump(names, with_stats, dump_format) for t in self.trees]
+
+    def dump_model(self, fout, fmap: str = "", with_stats: bool = False,
+                   dump_format: str = "text") -> None:
+        dumps = self.get_dump(fmap, with_stats, dump_format)
+        with open(fout, "w") as fh:
+            if dump_format == "json":
+                fh.write("[\n" + ",\n".join(dumps) + "\n]")
+            else:
+                for i, d in enumerate(dumps):
+                    fh.write(f"booster[{i}]:\n{d}")
+
+    def get_score(self, fmap: str = "", importance_type: str = "weight"
+                  ) -> Dict[str, float]:
+        if importance_type not in ("weight", "gain", "cover",
+                                   "total_gain", "total_cover"):
+            raise ValueError(f"unknown importance_type {importance_type}")
+        counts: Dict[int, float] = {}
+        sums: Dict[int, float] = {}
+        for tree in self.trees:
+            for nid in range(tree.n_nodes):
+                if tree.is_leaf(nid):
+                    continue
+                f = int(tree.split_index[nid])
+                counts[f] = counts.get(f, 0.0) + 1.0
+                v = (float(tree.loss_chg[nid]) if "gain" in importance_type
+                     else float(tree.sum_hess[nid]))
+                sums[f] = sums.get(f, 0.0) + v
+        names = self.feature_names
+
+        def fname(f):
+            return names[f] if names and f < len(names) else f"f{f}"
+
+        if importance_type == "weight":
+            return {fname(f): c for f, c in counts.items()}
+        if importance_type.startswith("total"):
+            return {fname(f): sums[f] for f in sums}
+        return {fname(f): sums[f] / counts[f] for f in sums}
+
+    def get_fscore(self, fmap: str = "") -> Dict[str, float]:
+        return self.get_score(fmap, "weight")
+
+    def trees_to_dataframe(self, fmap: str = ""):
+        import pandas as pd
+        rows = []
+        for ti, tree in enumerate(self.trees):
+            for nid in range(tree.n_nodes):
+                leaf = tree.is_leaf(nid)
+                rows.append({
+                    "Tree": ti, "Node": nid, "ID": f"{ti}-{nid}",
+                    "Feature": "Leaf" if leaf else
+                    (self.feature_names[tree.split_index[nid]]
+                     if self.feature_names else f"f{tree.split_index[nid]}"),
+                    "Split": None if leaf else float(tree.split_cond[nid]),
+                    "Yes": None if leaf else f"{ti}-{tree.left[nid]}",
+                    "No": None if leaf else f"{ti}-{tree.right[nid]}",
+                    "Missing": None if leaf else
+                    f"{ti}-{tree.left[nid] if tree.default_left[nid] else tree.right[nid]}",
+                    "Gain": float(tree.split_cond[nid]) if leaf
+                    else float(tree.loss_chg[nid]),
+                    "Cover": float(tree.sum_hess[nid]),
+                })
+        return pd.DataFrame(rows)
+
+    def reset(self) -> "Booster":
+        self._cache.clear()
+        self._ops_cache.clear()
+        return self
+
+
+def _parse_model_bytes(data: bytes, force_ubj: bool) -> dict:
+    """Both JSON and UBJSON models may begin with b'{' — try JSON text
+    first (cheap), fall back to UBJSON."""
+    if not force_ubj:
+        try:
+            return json.loads(data)
+        except (UnicodeDecodeError, json.JSONDecodeError):
+            pass
+    from .ubjson import loads_ubjson
+    return loads_ubjson(data)
+
+
+def _sample_rate_threshold(score: torch.Tensor, p: float) -> float:
+    """Threshold u such that sum(min(score/u, 1)) == p*n (reference
+    sampler.cu:18 SampleRateDelta), via binary search."""
+    n = score.numel()
+    target = p * n
+    lo = float(score.min()) * 1e-6 + 1e-12
+    hi = float(score.max()) / max(p, 1e-6) + 1e-6
+    for _ in range(64):
+        mid = 0.5 * (lo + hi)
+        s = float(torch.clamp(score / mid, max=1.0).sum())
+        if s > target:
+            lo = mid
+        else:
+            hi = mid
+    return 0.5 * (lo + hi)

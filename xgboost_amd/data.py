@@ -1,0 +1,350 @@
+"""DMatrix family and MetaInfo.
+
+Reference behavior: include/xgboost/data.h:65 (MetaInfo), data.h:549
+(DMatrix), src/data/simple_dmatrix.h, src/data/iterative_dmatrix.h.
+
+Design (MI355X-native): the in-core representation is a dense torch
+float32 tensor plus a quantized bin matrix ("Ellpack") built lazily per
+max_bin.  With 288 GB HBM3E per GPU the quantized matrix for even very
+large datasets is device-resident; the raw float matrix is only needed
+for sketching and can stay on host.  Quantized bins are stored row-major
+as u8/u16 local (per-feature-relative) bin ids — the reference's
+"dense compressed" Ellpack mode (src/data/ellpack_page.cuh:26) — so the
+histogram kernel reads 1-2 bytes per (row, feature).
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import Any, Dict, List, Optional, Sequence, Union
+
+import numpy as np
+import torch
+
+from .quantile import HistogramCuts, make_cuts, search_bins
+
+
+@dataclasses.dataclass
+class MetaInfo:
+    """Per-dataset metadata (reference: include/xgboost/data.h:65)."""
+
+    num_row: int = 0
+    num_col: int = 0
+    labels: Optional[np.ndarray] = None          # [n] or [n, n_targets]
+    weights: Optional[np.ndarray] = None         # [n]
+    base_margin: Optional[np.ndarray] = None     # [n] or [n, n_out]
+    group_ptr: Optional[np.ndarray] = None       # [n_groups+1] for ranking
+    label_lower_bound: Optional[np.ndarray] = None  # survival
+    label_upper_bound: Optional[np.ndarray] = None
+    feature_names: Optional[List[str]] = None
+    feature_types: Optional[List[str]] = None
+
+    def validate(self) -> None:
+        if self.labels is not None and self.labels.shape[0] != self.num_row:
+            raise ValueError(
+                f"label length {self.labels.shape[0]} != num_row {self.num_row}")
+        if self.weights is not None:
+            if self.weights.shape[0] != self.num_row and (
+                    self.group_ptr is None
+                    or self.weights.shape[0] != len(self.group_ptr) - 1):
+                raise ValueError("weight length mismatch")
+            if np.any(self.weights < 0):
+                raise ValueError("weights must be non-negative")
+
+
+@dataclasses.dataclass
+class QuantizedMatrix:
+    """Dense quantized bin matrix (the Ellpack analog).
+
+    gidx: [n_rows, n_features] u8/u16/i32 of local bin ids; rows with a
+    missing value carry the per-feature sentinel `n_bins(f)` (only when
+    has_missing).
+    """
+
+    gidx: torch.Tensor
+    cuts: HistogramCuts
+    has_missing: bool
+
+    @property
+    def n_rows(self) -> int:
+        return self.gidx.shape[0]
+
+    @property
+    def n_features(self) -> int:
+        return self.gidx.shape[1]
+
+    @property
+    def device(self) -> torch.device:
+        return self.gidx.device
+
+    def to(self, device) -> "QuantizedMatrix":
+        if torch.device(device) == self.gidx.device:
+            return self
+        return QuantizedMatrix(self.gidx.to(device), self.cuts, self.has_missing)
+
+    def global_gidx(self) -> torch.Tensor:
+        """int64 [n, f] global bin ids, -1 for missing (CPU-oracle form)."""
+        ptrs = torch.from_numpy(self.cuts.ptrs).to(self.gidx.device)
+        n_bins = (ptrs[1:] - ptrs[:-1])
+        g = self.gidx.long()
+        missing = g >= n_bins.unsqueeze(0)
+        g = g + ptrs[:-1].unsqueeze(0)
+        g[missing] = -1
+        return g
+
+
+def _pick_bin_dtype(max_local: int) -> torch.dtype:
+    if max_local <= 255:
+        return torch.uint8
+    if max_local <= 65535:
+        return torch.int16  # stored as i16; kernels reinterpret as u16
+    return torch.int32
+
+
+def quantize_dense(X: np.ndarray, cuts: HistogramCuts,
+                   missing: float = np.nan) -> QuantizedMatrix:
+    """Host-side quantization: numeric bin = #cuts<=v, categorical bin = v."""
+    gidx_global = search_bins(X, cuts, missing)  # int32, -1 missing
+    has_missing = bool((gidx_global < 0).any())
+    offsets = cuts.ptrs[:-1].astype(np.int32)
+    n_bins = np.diff(cuts.ptrs).astype(np.int32)
+    local = gidx_global - offsets[None, :]
+    if has_missing:
+        local = np.where(gidx_global < 0, n_bins[None, :], local)
+        max_local = int(n_bins.max())
+    else:
+        max_local = int(n_bins.max()) - 1
+    dtype = _pick_bin_dtype(max_local)
+    t = torch.from_numpy(local.astype(_np_dtype(dtype), copy=False))
+    return QuantizedMatrix(gidx=t, cuts=cuts, has_missing=has_missing)
+
+
+def _np_dtype(td: torch.dtype):
+    return {torch.uint8: np.uint8, torch.int16: np.int16, torch.int32: np.int32}[td]
+
+
+class DMatrix:
+    """User-facing data holder (reference: python Booster/DMatrix API).
+
+    Accepts numpy 2-D arrays, torch tensors, scipy CSR/CSC, pandas
+    DataFrames and (nested) lists.  Internally dense float32.
+    """
+
+    def __init__(self, data: Any, label: Any = None, *, weight: Any = None,
+                 base_margin: Any = None, missing: float = np.nan,
+                 feature_names: Optional[Sequence[str]] = None,
+                 feature_types: Optional[Sequence[str]] = None,
+                 group: Any = None, qid: Any = None,
+                 label_lower_bound: Any = None, label_upper_bound: Any = None,
+                 nthread: Optional[int] = None, enable_categorical: bool = False,
+                 silent: bool = False):
+        self.missing = float("nan") if missing is None else float(missing)
+        X, inferred_names, inferred_types = _ingest(data, enable_categorical)
+        self._data = X  # np.float32 [n, f]
+        self.info = MetaInfo(num_row=X.shape[0], num_col=X.shape[1])
+        if label is not None:
+            self.info.labels = _as_float_array(label)
+        if weight is not None:
+            self.info.weights = _as_float_array(weight).reshape(-1)
+        if base_margin is not None:
+            self.info.base_margin = _as_float_array(base_margin)
+        if label_lower_bound is not None:
+            self.info.label_lower_bound = _as_float_array(label_lower_bound).reshape(-1)
+        if label_upper_bound is not None:
+            self.info.label_upper_bound = _as_float_array(label_upper_bound).reshape(-1)
+        if group is not None:
+            g = np.asarray(group, dtype=np.int64).reshape(-1)
+            self.info.group_ptr = np.concatenate([[0], np.cumsum(g)]).astype(np.int64)
+        elif qid is not None:
+            q = np.asarray(qid).reshape(-1)
+            if np.any(q[1:] < q[:-1]):
+                raise ValueError("qid must be sorted in non-decreasing order")
+            boundaries = np.nonzero(np.diff(q))[0] + 1
+            self.info.group_ptr = np.concatenate(
+                [[0], boundaries, [q.size]]).astype(np.int64)
+        self.info.feature_names = (list(feature_names) if feature_names
+                                   else inferred_names)
+        self.info.feature_types = (list(feature_types) if feature_types
+                                   else inferred_types)
+        if self.info.feature_types is not None and len(self.info.feature_types) != X.shape[1]:
+            raise ValueError("feature_types length mismatch")
+        self.info.validate()
+        self._quantized: Dict[int, QuantizedMatrix] = {}
+        self._ref_cuts: Optional[HistogramCuts] = None
+
+    # -- reference API surface ------------------------------------------------
+    def num_row(self) -> int:
+        return self.info.num_row
+
+    def num_col(self) -> int:
+        return self.info.num_col
+
+    def get_label(self) -> np.ndarray:
+        return (self.info.labels if self.info.labels is not None
+                else np.zeros(self.num_row(), dtype=np.float32))
+
+    def get_weight(self) -> np.ndarray:
+        return (self.info.weights if self.info.weights is not None
+                else np.ones(self.num_row(), dtype=np.float32))
+
+    def get_base_margin(self) -> Optional[np.ndarray]:
+        return self.info.base_margin
+
+    def set_info(self, *, label=None, weight=None, base_margin=None,
+                 group=None, qid=None, feature_names=None, feature_types=None,
+                 label_lower_bound=None, label_upper_bound=None) -> None:
+        if label is not None:
+            self.info.labels = _as_float_array(label)
+        if weight is not None:
+            self.info.weights = _as_float_array(weight).reshape(-1)
+        if base_margin is not None:
+            self.info.base_margin = _as_float_array(base_margin)
+        if label_lower_bound is not None:
+            self.info.label_lower_bound = _as_float_array(label_lower_bound).reshape(-1)
+        if label_upper_bound is not None:
+            self.info.label_upper_bound = _as_float_array(label_upper_bound).reshape(-1)
+        if group is not None:
+            g = np.asarray(group, dtype=np.int64).reshape(-1)
+            self.info.group_ptr = np.concatenate([[0], np.cumsum(g)]).astype(np.int64)
+        if qid is not None:
+            q = np.asarray(qid).reshape(-1)
+            boundaries = np.nonzero(np.diff(q))[0] + 1
+            self.info.group_ptr = np.concatenate([[0], boundaries, [q.size]]).astype(np.int64)
+        if feature_names is not None:
+            self.info.feature_names = list(feature_names)
+        if feature_types is not None:
+            self.info.feature_types = list(feature_types)
+        self.info.validate()
+
+    set_label = lambda self, label: self.set_info(label=label)
+    set_weight = lambda self, weight: self.set_info(weight=weight)
+    set_base_margin = lambda self, m: self.set_info(base_margin=m)
+    set_group = lambda self, group: self.set_info(group=group)
+
+    @property
+    def feature_names(self):
+        return self.info.feature_names
+
+    @property
+    def feature_types(self):
+        return self.info.feature_types
+
+    def slice(self, rindex: Sequence[int]) -> "DMatrix":
+        idx = np.asarray(rindex, dtype=np.int64)
+        out = DMatrix(self._data[idx], missing=self.missing,
+                      feature_names=self.info.feature_names,
+                      feature_types=self.info.feature_types)
+        if self.info.labels is not None:
+            out.info.labels = self.info.labels[idx]
+        if self.info.weights is not None:
+            out.info.weights = self.info.weights[idx]
+        if self.info.base_margin is not None:
+            out.info.base_margin = self.info.base_margin[idx]
+        return out
+
+    # -- internal --------------------------------------------------------------
+    def raw_data(self) -> np.ndarray:
+        return self._data
+
+    def set_ref_cuts(self, cuts: HistogramCuts) -> None:
+        """Bin this matrix with cut points from a training DMatrix
+        (reference: GetCutsFromRef, src/data/quantile_dmatrix.cc:19)."""
+        self._ref_cuts = cuts
+        self._quantized.clear()
+
+    def quantized(self, max_bin: int, sketch_fn=None) -> QuantizedMatrix:
+        """Lazily build (and cache) the quantized matrix for max_bin."""
+        qm = self._quantized.get(max_bin)
+        if qm is None:
+            if self._ref_cuts is not None:
+                cuts = self._ref_cuts
+            elif sketch_fn is not None:
+                cuts = sketch_fn(self, max_bin)
+            else:
+                cuts = make_cuts(self._data, max_bin,
+                                 weights=None,
+                                 feature_types=self.info.feature_types,
+                                 missing=self.missing)
+            qm = quantize_dense(self._data, cuts, self.missing)
+            self._quantized[max_bin] = qm
+        return qm
+
+    def cached_cuts(self) -> Optional[HistogramCuts]:
+        for qm in self._quantized.values():
+            return qm.cuts
+        return self._ref_cuts
+
+
+class QuantileDMatrix(DMatrix):
+    """Quantized-only DMatrix (reference: src/data/iterative_dmatrix.h:34).
+
+    Builds the bin matrix at construction; `ref` shares cut points with a
+    training matrix so validation data is binned identically.
+    """
+
+    def __init__(self, data: Any, label: Any = None, *, max_bin: int = 256,
+                 ref: Optional[DMatrix] = None, **kwargs):
+        super().__init__(data, label, **kwargs)
+        self.max_bin = max_bin
+        if ref is not None:
+            cuts = ref.cached_cuts()
+            if cuts is None:
+                cuts = ref.quantized(max_bin).cuts
+            self.set_ref_cuts(cuts)
+        self.quantized(max_bin)
+
+
+def _as_float_array(v: Any) -> np.ndarray:
+    if isinstance(v, torch.Tensor):
+        return v.detach().cpu().numpy().astype(np.float32, copy=False)
+    return np.asarray(v, dtype=np.float32)
+
+
+def _ingest(data: Any, enable_categorical: bool):
+    """Normalize input to dense np.float32 [n, f]; returns (X, names, types)."""
+    names = None
+    types = None
+    if isinstance(data, DMatrix):
+        raise TypeError("cannot construct a DMatrix from a DMatrix")
+    if isinstance(data, torch.Tensor):
+        X = data.detach().cpu().numpy()
+    elif hasattr(data, "toarray") and hasattr(data, "tocsr"):  # scipy sparse
+        X = data.toarray()
+    elif _is_pandas(data):
+        X, names, types = _from_pandas(data, enable_categorical)
+    else:
+        X = np.asarray(data)
+    if X.ndim == 1:
+        X = X.reshape(-1, 1)
+    if X.ndim != 2:
+        raise ValueError(f"expected 2-D data, got shape {X.shape}")
+    X = np.ascontiguousarray(X, dtype=np.float32)
+    return X, names, types
+
+
+def _is_pandas(data: Any) -> bool:
+    return type(data).__module__.startswith("pandas") and hasattr(data, "dtypes")
+
+
+def _from_pandas(df, enable_categorical: bool):
+    import pandas as pd
+    names = [str(c) for c in df.columns]
+    types: List[str] = []
+    cols = []
+    for c in df.columns:
+        s = df[c]
+        if isinstance(s.dtype, pd.CategoricalDtype):
+            if not enable_categorical:
+                raise ValueError(
+                    f"categorical column {c!r} needs enable_categorical=True")
+            cols.append(s.cat.codes.to_numpy(np.float32))
+            types.append("c")
+        else:
+            v = s.to_numpy(np.float32, na_value=np.nan)
+            cols.append(v)
+            types.append("int" if np.issubdtype(s.dtype, np.integer) else "float")
+    X = np.stack(cols, axis=1)
+    if "c" not in types:
+        types_out = None
+    else:
+        types_out = ["c" if t == "c" else "q" for t in types]
+    return X, names, types_out

@@ -1,0 +1,314 @@
+"""Histogram-based tree grower — the training hot loop.
+
+Reference behavior: src/tree/updater_gpu_hist.cu:588 (UpdateTree loop),
+driver.h (depthwise/lossguide priority queue), updater_gpu_hist.cuh:62
+(AssignNodes build-smaller/subtract-larger), hist/evaluate_splits.h,
+split_evaluator.h (monotone bounds), common/random.h:74 (ColumnSampler).
+
+Backend-agnostic: all data-touching primitives go through an `ops`
+object (CpuOps or GpuOps) so the same driver code runs the torch-CPU
+oracle path and the HIP kernel path.
+"""
+from __future__ import annotations
+
+import dataclasses
+import heapq
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from . import collective
+from .backend.cpu import GradQuantizer
+from .params import TrainParam
+from .splits import SplitEntry, calc_weight
+from .tree_model import RegTree
+
+
+class ColumnSampler:
+    """Nested per-tree/level/node feature sampling (reference
+    src/common/random.h:74).  Seeded identically on every worker so
+    feature sets agree without communication."""
+
+    def __init__(self, n_features: int, param: TrainParam, seed: int):
+        self.rng = np.random.RandomState(seed & 0x7FFFFFFF)
+        self.n_features = n_features
+        self.param = param
+        n = n_features
+        feats = np.arange(n_features)
+        if param.colsample_bytree < 1.0:
+            k = max(1, int(round(param.colsample_bytree * n)))
+            feats = np.sort(self.rng.choice(feats, size=k, replace=False))
+        self.tree_set = feats
+        self._level_cache: Dict[int, np.ndarray] = {}
+
+    def level_set(self, depth: int) -> np.ndarray:
+        if self.param.colsample_bylevel >= 1.0:
+            return self.tree_set
+        if depth not in self._level_cache:
+            k = max(1, int(round(self.param.colsample_bylevel * len(self.tree_set))))
+            self._level_cache[depth] = np.sort(
+                self.rng.choice(self.tree_set, size=k, replace=False))
+        return self._level_cache[depth]
+
+    def node_set(self, depth: int) -> Optional[np.ndarray]:
+        base = self.level_set(depth)
+        if self.param.colsample_bynode >= 1.0:
+            if (self.param.colsample_bytree >= 1.0
+                    and self.param.colsample_bylevel >= 1.0):
+                return None  # all features
+            return base
+        k = max(1, int(round(self.param.colsample_bynode * len(base))))
+        return np.sort(self.rng.choice(base, size=k, replace=False))
+
+
+class InteractionConstraints:
+    """Per-node allowed-feature sets (reference src/tree/constraints.cc:103)."""
+
+    def __init__(self, spec: Sequence[Sequence[int]], n_features: int):
+        self.sets = [frozenset(int(f) for f in s) for s in spec]
+        self.n_features = n_features
+        self.node_path: Dict[int, frozenset] = {0: frozenset()}
+
+    def split(self, nid: int, left: int, right: int, feature: int) -> None:
+        path = self.node_path.get(nid, frozenset()) | {int(feature)}
+        self.node_path[left] = path
+        self.node_path[right] = path
+
+    def allowed(self, nid: int) -> Optional[np.ndarray]:
+        path = self.node_path.get(nid, frozenset())
+        if not path:
+            return None
+        allowed = set(path)
+        for s in self.sets:
+            if path <= s:
+                allowed |= s
+        return np.array(sorted(allowed), dtype=np.int64)
+
+
+@dataclasses.dataclass(order=True)
+class _QueueEntry:
+    sort_key: float
+    seq: int
+    nid: int = dataclasses.field(compare=False)
+    depth: int = dataclasses.field(compare=False)
+    split: SplitEntry = dataclasses.field(compare=False)
+
+
+class TreeGrower:
+    """Grows one tree from quantized gradients."""
+
+    def __init__(self, ops, param: TrainParam, quantizer: GradQuantizer,
+                 n_rows: int, seed: int = 0,
+                 monotone: Optional[np.ndarray] = None,
+                 interaction: Optional[Sequence[Sequence[int]]] = None):
+        self.ops = ops
+        self.param = param
+        self.quantizer = quantizer
+        self.n_rows = n_rows
+        self.col_sampler = ColumnSampler(ops.qm.n_features, param, seed)
+        self.monotone = monotone
+        self.interaction = (InteractionConstraints(interaction, ops.qm.n_features)
+                            if interaction else None)
+        cuts = ops.qm.cuts
+        if cuts.feature_types is not None:
+            self.cat_mask = np.array([t == "c" for t in cuts.feature_types])
+            if not self.cat_mask.any():
+                self.cat_mask = None
+        else:
+            self.cat_mask = None
+
+    def grow(self, qgpair: torch.Tensor, tree: RegTree
+             ) -> Tuple[RegTree, torch.Tensor]:
+        """Returns (tree, leaf position per row int32)."""
+        param = self.param
+        ops = self.ops
+        ridx = ops.make_ridx(self.n_rows)
+        segments: Dict[int, Tuple[int, int]] = {0: (0, self.n_rows)}
+        node_sums: Dict[int, Tuple[float, float]] = {}
+        node_bounds: Dict[int, Tuple[float, float]] = {0: (-np.inf, np.inf)}
+        hists: Dict[int, torch.Tensor] = {}
+
+        # root
+        qg, qh = ops.root_sum(qgpair)
+        root_g = qg / self.quantizer.g_scale
+        root_h = qh / self.quantizer.h_scale
+        node_sums[0] = (root_g, root_h)
+        root_w = float(calc_weight(root_g, root_h, param))
+        tree.base_weight[0] = root_w
+        tree.sum_hess[0] = root_h
+        hist = ops.build_hist(qgpair, ridx, [segments[0]])
+        ops.allreduce_hist(hist)
+        hists[0] = hist[0]
+        root_entry = self._evaluate([0], node_sums, hists, node_bounds, depth=0)[0]
+
+        n_leaves = 1
+        seq = 0
+        heap: List[_QueueEntry] = []
+
+        def push(nid, depth, split):
+            nonlocal seq
+            if not self._expandable(split, depth, n_leaves):
+                return
+            key = (float(depth) if param.grow_policy == "depthwise"
+                   else -split.gain)
+            heapq.heappush(heap, _QueueEntry(key, seq, nid, depth, split))
+            seq += 1
+
+        push(0, 0, root_entry)
+
+        while heap:
+            # pop a batch: depthwise = whole level; lossguide = single best
+            batch = [heapq.heappop(heap)]
+            if param.grow_policy == "depthwise":
+                while heap and heap[0].sort_key == batch[0].sort_key:
+                    batch.append(heapq.heappop(heap))
+            # re-check leaf budget
+            batch = [b for b in batch
+                     if self._expandable(b.split, b.depth, n_leaves + 0)]
+            if param.max_leaves > 0:
+                keep = []
+                for b in batch:
+                    if n_leaves + len(keep) + 1 <= param.max_leaves:
+                        keep.append(b)
+                batch = keep
+            if not batch:
+                continue
+
+            # 1. apply splits to tree
+            children = []  # (entry, lnid, rnid)
+            for b in batch:
+                sp = b.split
+                cuts = ops.qm.cuts
+                if sp.is_cat:
+                    cond = np.nan
+                    cats_right = sp.cat_bits
+                else:
+                    cond = float(cuts.values[sp.split_bin])
+                    cats_right = None
+                wl = float(calc_weight(sp.left_g, sp.left_h, param))
+                wr = float(calc_weight(sp.right_g, sp.right_h, param))
+                lo, hi = node_bounds.get(b.nid, (-np.inf, np.inf))
+                wl, wr = self._clip_bounds(wl, wr, lo, hi)
+                l, r = tree.add_split(
+                    b.nid, sp.feature, cond, sp.default_left, sp.gain,
+                    float(tree.base_weight[b.nid]), wl, wr,
+                    sp.left_h + sp.right_h, sp.left_h, sp.right_h,
+                    categories_go_right=cats_right)
+                node_sums[l] = (sp.left_g, sp.left_h)
+                node_sums[r] = (sp.right_g, sp.right_h)
+                self._propagate_bounds(node_bounds, b.nid, l, r, sp, wl, wr)
+                if self.interaction is not None:
+                    self.interaction.split(b.nid, l, r, sp.feature)
+                children.append((b, l, r))
+                n_leaves += 1
+
+            # 2. partition rows
+            segs = [segments[b.nid] for b, _, _ in children]
+            splits = [b.split for b, _, _ in children]
+            new_segs = ops.partition(ridx, segs, splits)
+            for (b, l, r), (ls, rs) in zip(children, new_segs):
+                segments[l] = ls
+                segments[r] = rs
+
+            # 3. decide build vs subtract (smaller child built,
+            #    reference AssignNodes updater_gpu_hist.cuh:62)
+            build_nodes, subtract_nodes = [], []
+            for (b, l, r), (ls, rs) in zip(children, new_segs):
+                expand_more = (b.depth + 1 < (param.max_depth or 10 ** 9)
+                               or param.grow_policy == "lossguide")
+                if not expand_more and param.max_depth > 0:
+                    continue
+                if (ls[1] - ls[0]) <= (rs[1] - rs[0]):
+                    build_nodes.append((l, b.nid, r))
+                else:
+                    build_nodes.append((r, b.nid, l))
+            # 4. build + allreduce + subtract
+            if build_nodes:
+                bsegs = [segments[n] for n, _, _ in build_nodes]
+                bh = ops.build_hist(qgpair, ridx, bsegs)
+                ops.allreduce_hist(bh)
+                for i, (n, parent, sib) in enumerate(build_nodes):
+                    hists[n] = bh[i]
+                    hists[sib] = hists[parent] - bh[i]
+                    del hists[parent]
+
+            # 5. evaluate children
+            eval_nids = [n for (b, l, r) in children for n in (l, r)
+                         if n in hists]
+            if eval_nids:
+                depth = children[0][0].depth + 1
+                entries = self._evaluate(eval_nids, node_sums, hists,
+                                         node_bounds, depth)
+                for nid, e in zip(eval_nids, entries):
+                    push(nid, depth, e)
+
+        # finalize leaves
+        leaf_segments = []
+        for nid in range(tree.n_nodes):
+            if tree.is_leaf(nid):
+                w = float(tree.base_weight[nid])
+                tree.set_leaf(nid, w * param.eta)
+                if nid in segments:
+                    s, e = segments[nid]
+                    leaf_segments.append((nid, s, e))
+        positions = ops.leaf_partition(ridx, leaf_segments, self.n_rows)
+        return tree, positions
+
+    # ------------------------------------------------------------------
+    def _expandable(self, split: SplitEntry, depth: int, n_leaves: int) -> bool:
+        param = self.param
+        if not split.is_valid or split.gain <= param.gamma:
+            return False
+        if param.max_depth > 0 and depth >= param.max_depth:
+            return False
+        if param.max_leaves > 0 and n_leaves >= param.max_leaves:
+            return False
+        return True
+
+    def _evaluate(self, nids, node_sums, hists, node_bounds, depth
+                  ) -> List[SplitEntry]:
+        param = self.param
+        hist = torch.stack([hists[n] for n in nids])
+        parent_sums = [node_sums[n] for n in nids]
+        feature_sets = None
+        node_feats = self.col_sampler.node_set(depth)
+        inter_sets = ([self.interaction.allowed(n) for n in nids]
+                      if self.interaction is not None else None)
+        if node_feats is not None or inter_sets is not None:
+            feature_sets = []
+            for i, n in enumerate(nids):
+                fs = node_feats
+                if self.param.colsample_bynode < 1.0:
+                    fs = self.col_sampler.node_set(depth)  # fresh draw per node
+                if inter_sets is not None and inter_sets[i] is not None:
+                    fs = (np.intersect1d(fs, inter_sets[i]) if fs is not None
+                          else inter_sets[i])
+                feature_sets.append(fs)
+        bounds = (np.array([node_bounds.get(n, (-np.inf, np.inf)) for n in nids])
+                  if self.monotone is not None else None)
+        return self.ops.evaluate_splits(
+            hist, self.quantizer,
+            [(g, h) for g, h in parent_sums], nids, param,
+            feature_sets=feature_sets, monotone=self.monotone,
+            cat_mask=self.cat_mask, node_bounds=bounds)
+
+    def _clip_bounds(self, wl, wr, lo, hi):
+        return float(np.clip(wl, lo, hi)), float(np.clip(wr, lo, hi))
+
+    def _propagate_bounds(self, node_bounds, nid, l, r, sp, wl, wr) -> None:
+        lo, hi = node_bounds.get(nid, (-np.inf, np.inf))
+        node_bounds[l] = (lo, hi)
+        node_bounds[r] = (lo, hi)
+        if self.monotone is None:
+            return
+        c = int(self.monotone[sp.feature]) if sp.feature < len(self.monotone) else 0
+        if c == 0:
+            return
+        mid = (wl + wr) / 2.0
+        if c > 0:
+            node_bounds[l] = (lo, min(hi, mid))
+            node_bounds[r] = (max(lo, mid), hi)
+        else:
+            node_bounds[l] = (max(lo, mid), hi)
+            node_bounds[r] = (lo, min(hi, mid))
