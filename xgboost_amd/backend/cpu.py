@@ -102,11 +102,10 @@ class CpuOps:
                         nids: Sequence[int], param: TrainParam,
                         feature_sets=None, monotone=None, cat_mask=None,
                         node_bounds=None) -> List[SplitEntry]:
-        h = quantizer.dequantize_hist(hist)
-        # parent_sums arrive already dequantized (float g, h)
-        pg = np.array([s[0] for s in parent_sums], dtype=np.float64)
-        ph = np.array([s[1] for s in parent_sums], dtype=np.float64)
-        return evaluate_splits_np(h, pg, ph, nids, self.qm.cuts.ptrs, param,
+        # parent_sums are exact int64 (gq, hq) pairs
+        return evaluate_splits_np(hist.cpu().numpy(), parent_sums,
+                                  quantizer.g_scale, quantizer.h_scale,
+                                  nids, self.qm.cuts.ptrs, param,
                                   feature_sets=feature_sets, monotone=monotone,
                                   cat_mask=cat_mask, node_bounds=node_bounds)
 

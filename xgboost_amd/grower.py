@@ -125,7 +125,7 @@ class TreeGrower:
         ops = self.ops
         ridx = ops.make_ridx(self.n_rows)
         segments: Dict[int, Tuple[int, int]] = {0: (0, self.n_rows)}
-        node_sums: Dict[int, Tuple[float, float]] = {}
+        node_sums: Dict[int, Tuple[int, int]] = {}  # exact int64 (gq, hq)
         node_bounds: Dict[int, Tuple[float, float]] = {0: (-np.inf, np.inf)}
         hists: Dict[int, torch.Tensor] = {}
 
@@ -133,7 +133,7 @@ class TreeGrower:
         qg, qh = ops.root_sum(qgpair)
         root_g = qg / self.quantizer.g_scale
         root_h = qh / self.quantizer.h_scale
-        node_sums[0] = (root_g, root_h)
+        node_sums[0] = (qg, qh)
         root_w = float(calc_weight(root_g, root_h, param))
         tree.base_weight[0] = root_w
         tree.sum_hess[0] = root_h
@@ -195,8 +195,8 @@ class TreeGrower:
                     float(tree.base_weight[b.nid]), wl, wr,
                     sp.left_h + sp.right_h, sp.left_h, sp.right_h,
                     categories_go_right=cats_right)
-                node_sums[l] = (sp.left_g, sp.left_h)
-                node_sums[r] = (sp.right_g, sp.right_h)
+                node_sums[l] = (sp.left_gq, sp.left_hq)
+                node_sums[r] = (sp.right_gq, sp.right_hq)
                 self._propagate_bounds(node_bounds, b.nid, l, r, sp, wl, wr)
                 if self.interaction is not None:
                     self.interaction.split(b.nid, l, r, sp.feature)

@@ -4,11 +4,12 @@ Reference behavior: src/tree/split_evaluator.h (TreeEvaluator), param.h
 CalcGain/CalcWeight, src/tree/hist/evaluate_splits.h:30 and
 src/tree/gpu_hist/evaluate_splits.cu (EvaluateSplitAgent).
 
-This module holds (a) the scalar gain/weight formulas shared by the
-numpy oracle, the HIP kernel's verification tests, and leaf-value
-computation, and (b) a vectorized numpy evaluator used by the CPU
-backend.  Histogram inputs are deterministic int64 fixed-point sums;
-gain math runs in float64 after dequantization.
+Determinism: histograms and all left/right sums are int64 fixed-point
+(quantized gradients, see backend/cpu.py GradQuantizer) — scans and
+sibling subtraction are exact integer math, so the numpy oracle and the
+HIP kernel produce identical child sums; only the gain value itself is
+computed in float64 after dequantization, with a fixed operation order
+shared by both implementations.
 """
 from __future__ import annotations
 
@@ -20,14 +21,15 @@ import numpy as np
 from .params import TrainParam
 
 
-def threshold_l1(g: np.ndarray, alpha: float) -> np.ndarray:
+def threshold_l1(g, alpha: float):
     return np.sign(g) * np.maximum(np.abs(g) - alpha, 0.0)
 
 
 def calc_weight(g, h, param: TrainParam):
     """-ThresholdL1(G)/(H+lambda), optionally clipped by max_delta_step."""
-    w = -threshold_l1(np.asarray(g, dtype=np.float64), param.reg_alpha) / (
-        np.asarray(h, dtype=np.float64) + param.reg_lambda)
+    g = np.asarray(g, dtype=np.float64)
+    h = np.asarray(h, dtype=np.float64)
+    w = -threshold_l1(g, param.reg_alpha) / (h + param.reg_lambda)
     if param.max_delta_step > 0:
         w = np.clip(w, -param.max_delta_step, param.max_delta_step)
     return w
@@ -52,19 +54,39 @@ class SplitEntry:
     feature: int = -1
     split_bin: int = -1           # global bin index
     default_left: bool = False
-    left_g: float = 0.0           # dequantized sums
-    left_h: float = 0.0
-    right_g: float = 0.0
-    right_h: float = 0.0
+    left_gq: int = 0              # exact int64 fixed-point sums
+    left_hq: int = 0
+    right_gq: int = 0
+    right_hq: int = 0
+    g_scale: float = 1.0
+    h_scale: float = 1.0
     is_cat: bool = False
     cat_bits: Optional[np.ndarray] = None  # local bin ids going RIGHT
+
+    @property
+    def left_g(self) -> float:
+        return self.left_gq / self.g_scale
+
+    @property
+    def left_h(self) -> float:
+        return self.left_hq / self.h_scale
+
+    @property
+    def right_g(self) -> float:
+        return self.right_gq / self.g_scale
+
+    @property
+    def right_h(self) -> float:
+        return self.right_hq / self.h_scale
 
     @property
     def is_valid(self) -> bool:
         return self.feature >= 0 and np.isfinite(self.gain) and self.gain > 0
 
 
-def evaluate_splits_np(hist: np.ndarray, parent_g: np.ndarray, parent_h: np.ndarray,
+def evaluate_splits_np(hist_q: np.ndarray,
+                       parent_q: Sequence,
+                       g_scale: float, h_scale: float,
                        nids: Sequence[int], cut_ptrs: np.ndarray,
                        param: TrainParam,
                        feature_sets: Optional[List[np.ndarray]] = None,
@@ -72,50 +94,52 @@ def evaluate_splits_np(hist: np.ndarray, parent_g: np.ndarray, parent_h: np.ndar
                        cat_mask: Optional[np.ndarray] = None,
                        node_bounds: Optional[np.ndarray] = None,
                        ) -> List[SplitEntry]:
-    """Vectorized split evaluation over [n_nodes, n_bins, 2] float64 hists.
+    """Vectorized split evaluation over int64 [n_nodes, n_bins, 2] hists.
 
-    hist: dequantized float64 (G, H per global bin)
-    parent_g/h: [n_nodes] float64 node totals
-    feature_sets: per-node allowed features (colsample / interaction)
-    monotone: [n_features] in {-1, 0, +1}
-    cat_mask: [n_features] bool, True = categorical (one-hot eval)
-    node_bounds: [n_nodes, 2] (lower, upper) weight bounds from monotone
-      constraint propagation; leaf weights are clipped into these.
+    parent_q: [(gq, hq)] exact node totals (int64).
     """
-    n_nodes, n_bins, _ = hist.shape
+    n_nodes, n_bins, _ = hist_q.shape
     n_features = len(cut_ptrs) - 1
     widths = np.diff(cut_ptrs)
     feat_of_bin = np.repeat(np.arange(n_features), widths)
     seg_start = np.repeat(cut_ptrs[:-1], widths)
-
-    G = hist[:, :, 0]
-    H = hist[:, :, 1]
-    lam = param.reg_lambda
-
-    cumG = np.cumsum(G, axis=1)
-    cumH = np.cumsum(H, axis=1)
-    baseG = np.where(seg_start > 0, cumG[:, np.maximum(seg_start - 1, 0)], 0.0)
-    baseH = np.where(seg_start > 0, cumH[:, np.maximum(seg_start - 1, 0)], 0.0)
-    GL = cumG - baseG   # left sums including bin b (missing right)
-    HL = cumH - baseH
-
     seg_end = np.repeat(cut_ptrs[1:] - 1, widths)
-    featG = cumG[:, seg_end] - baseG   # per-bin: total of its feature
-    featH = cumH[:, seg_end] - baseH
-    missG = parent_g[:, None] - featG
-    missH = parent_h[:, None] - featH
 
-    parent_gain = calc_gain(parent_g, parent_h, param)  # [n_nodes]
+    pgq = np.array([p[0] for p in parent_q], dtype=np.int64)
+    phq = np.array([p[1] for p in parent_q], dtype=np.int64)
 
-    best = [SplitEntry(nid=int(nid)) for nid in nids]
+    Gq = hist_q[:, :, 0]
+    Hq = hist_q[:, :, 1]
+    cumG = np.cumsum(Gq, axis=1)
+    cumH = np.cumsum(Hq, axis=1)
+    baseG = np.where(seg_start > 0, cumG[:, np.maximum(seg_start - 1, 0)], 0)
+    baseH = np.where(seg_start > 0, cumH[:, np.maximum(seg_start - 1, 0)], 0)
+    GLq = cumG - baseG   # int64 exact: left sums including bin b
+    HLq = cumH - baseH
+    featGq = cumG[:, seg_end] - baseG
+    featHq = cumH[:, seg_end] - baseH
+    missGq = pgq[:, None] - featGq
+    missHq = phq[:, None] - featHq
 
-    is_last_bin = np.arange(n_bins) == seg_end  # splitting at last bin: right empty
+    inv_g = 1.0 / g_scale
+    inv_h = 1.0 / h_scale
+    parent_gain = calc_gain(pgq * inv_g, phq * inv_h, param)  # [n_nodes]
+
+    best = [SplitEntry(nid=int(nid), g_scale=g_scale, h_scale=h_scale)
+            for nid in nids]
+    is_last_bin = np.arange(n_bins) == seg_end
+
+    mono_bins = monotone[feat_of_bin][None, :] if monotone is not None else None
 
     for missing_left in (False, True):
-        gl = GL + (missG if missing_left else 0.0)
-        hl = HL + (missH if missing_left else 0.0)
-        gr = parent_g[:, None] - gl
-        hr = parent_h[:, None] - hl
+        glq = GLq + (missGq if missing_left else 0)
+        hlq = HLq + (missHq if missing_left else 0)
+        grq = pgq[:, None] - glq
+        hrq = phq[:, None] - hlq
+        gl = glq * inv_g
+        hl = hlq * inv_h
+        gr = grq * inv_g
+        hr = hrq * inv_h
         wl = calc_weight(gl, hl, param)
         wr = calc_weight(gr, hr, param)
         if node_bounds is not None:
@@ -128,15 +152,17 @@ def evaluate_splits_np(hist: np.ndarray, parent_g: np.ndarray, parent_h: np.ndar
                 - parent_gain[:, None])
         ok = (hl >= param.min_child_weight) & (hr >= param.min_child_weight)
         ok &= ~is_last_bin[None, :]
-        if monotone is not None:
-            c = monotone[feat_of_bin][None, :]
-            ok &= (c == 0) | ((c > 0) & (wl <= wr)) | ((c < 0) & (wl >= wr))
+        if mono_bins is not None:
+            ok &= ((mono_bins == 0) | ((mono_bins > 0) & (wl <= wr))
+                   | ((mono_bins < 0) & (wl >= wr)))
         gain = np.where(ok, gain, -np.inf)
         if cat_mask is not None and cat_mask.any():
-            gain = _onehot_cat_gains(gain, G, H, parent_g, parent_h, missG,
-                                     missH, feat_of_bin, cat_mask, param,
-                                     missing_left, parent_gain, node_bounds,
-                                     monotone)
+            gain, cat_lq = _onehot_cat_gains(
+                gain, Gq, Hq, pgq, phq, missGq, missHq, feat_of_bin, cat_mask,
+                param, missing_left, parent_gain, node_bounds, mono_bins,
+                inv_g, inv_h)
+        else:
+            cat_lq = None
         for i in range(n_nodes):
             row = gain[i]
             if feature_sets is not None and feature_sets[i] is not None:
@@ -151,40 +177,37 @@ def evaluate_splits_np(hist: np.ndarray, parent_g: np.ndarray, parent_h: np.ndar
                 f = int(feat_of_bin[b])
                 is_cat = bool(cat_mask is not None and cat_mask[f])
                 if is_cat:
-                    # one-hot: chosen category (stored set) goes RIGHT
-                    glv = float(parent_g[i]) - float(G[i, b]) - (
-                        0.0 if missing_left else float(missG[i, b]))
-                    hlv = float(parent_h[i]) - float(H[i, b]) - (
-                        0.0 if missing_left else float(missH[i, b]))
+                    lgq = int(cat_lq[0][i, b])
+                    lhq = int(cat_lq[1][i, b])
                 else:
-                    glv = float(GL[i, b]) + (float(missG[i, b]) if missing_left else 0.0)
-                    hlv = float(HL[i, b]) + (float(missH[i, b]) if missing_left else 0.0)
+                    lgq = int(glq[i, b])
+                    lhq = int(hlq[i, b])
                 e.gain = gval
                 e.feature = f
                 e.split_bin = b
                 e.default_left = missing_left
-                e.left_g = glv
-                e.left_h = hlv
-                e.right_g = float(parent_g[i]) - glv
-                e.right_h = float(parent_h[i]) - hlv
+                e.left_gq = lgq
+                e.left_hq = lhq
+                e.right_gq = int(pgq[i]) - lgq
+                e.right_hq = int(phq[i]) - lhq
                 e.is_cat = is_cat
                 if is_cat:
-                    # one-hot: category == this bin goes LEFT; others right
                     e.cat_bits = np.array([b - int(cut_ptrs[f])], dtype=np.int32)
     return best
 
 
-def _onehot_cat_gains(gain, G, H, parent_g, parent_h, missG, missH,
+def _onehot_cat_gains(gain, Gq, Hq, pgq, phq, missGq, missHq,
                       feat_of_bin, cat_mask, param, missing_left,
-                      parent_gain, node_bounds, monotone):
-    """For categorical features evaluate one-vs-rest per bin instead of the
-    cumulative scan (reference: one-hot split when n_cats is small)."""
+                      parent_gain, node_bounds, mono_bins, inv_g, inv_h):
+    """Categorical one-vs-rest: the chosen category (one bin) goes RIGHT,
+    everything else left (reference evaluate_splits.cu OneHot)."""
     cat_bins = cat_mask[feat_of_bin]
-    # chosen category goes RIGHT: left = parent - cat - (miss unless missing_left)
-    gl = parent_g[:, None] - G - (0.0 if missing_left else missG)
-    hl = parent_h[:, None] - H - (0.0 if missing_left else missH)
-    gr = parent_g[:, None] - gl
-    hr = parent_h[:, None] - hl
+    glq = pgq[:, None] - Gq - (0 if missing_left else missGq)
+    hlq = phq[:, None] - Hq - (0 if missing_left else missHq)
+    gl = glq * inv_g
+    hl = hlq * inv_h
+    gr = (pgq[:, None] - glq) * inv_g
+    hr = (phq[:, None] - hlq) * inv_h
     wl = calc_weight(gl, hl, param)
     wr = calc_weight(gr, hr, param)
     if node_bounds is not None:
@@ -196,8 +219,8 @@ def _onehot_cat_gains(gain, G, H, parent_g, parent_h, missG, missH,
           + calc_gain_given_weight(gr, hr, wr, param)
           - parent_gain[:, None])
     ok = (hl >= param.min_child_weight) & (hr >= param.min_child_weight)
-    if monotone is not None:
-        c = monotone[feat_of_bin][None, :]
-        ok &= (c == 0) | ((c > 0) & (wl <= wr)) | ((c < 0) & (wl >= wr))
+    if mono_bins is not None:
+        ok &= ((mono_bins == 0) | ((mono_bins > 0) & (wl <= wr))
+               | ((mono_bins < 0) & (wl >= wr)))
     g2 = np.where(ok, g2, -np.inf)
-    return np.where(cat_bins[None, :], g2, gain)
+    return np.where(cat_bins[None, :], g2, gain), (glq, hlq)
