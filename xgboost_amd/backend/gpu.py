@@ -1,0 +1,368 @@
+"""GPU backend ops — drives the HIP/CDNA4 kernels in ops/cpp.
+
+Same interface as backend/cpu.CpuOps; the grower is backend-agnostic.
+All launches go on torch's current HIP stream; the only host syncs per
+level are the partition-counter readback and the split-candidate
+readback (mirroring the reference's per-level pinned D2H sync,
+updater_gpu_hist.cu:613).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .. import collective
+from ..data import QuantizedMatrix
+from ..params import TrainParam
+from ..splits import SplitEntry
+from .cpu import GradQuantizer
+
+# LDS budget for the histogram kernel: 8192 bins * 16 B = 128 KiB
+# (160 KiB/CU on MI355X; leave headroom for occupancy)
+LDS_MAX_GROUP_BINS = 8192
+TARGET_HIST_TASKS = 2048  # fill 256 CUs x 8 blocks
+MIN_ROWS_PER_TASK = 1024
+
+
+def _chunk_tasks(segments: Sequence[Tuple[int, int]],
+                 slots: Optional[Sequence[int]] = None,
+                 target_tasks: int = TARGET_HIST_TASKS) -> np.ndarray:
+    """Build BlockTask array [(slot, row_begin, row_end, 0)] chunked so the
+    grid fills the chip (>=2048 blocks when rows allow)."""
+    total = sum(e - s for s, e in segments)
+    rows_per_task = max(MIN_ROWS_PER_TASK,
+                        (total + target_tasks - 1) // max(target_tasks, 1))
+    tasks = []
+    for i, (s, e) in enumerate(segments):
+        slot = slots[i] if slots is not None else i
+        b = s
+        while b < e:
+            tasks.append((slot, b, min(b + rows_per_task, e), 0))
+            b += rows_per_task
+        if s == e:  # empty segment still needs no task
+            pass
+    if not tasks:
+        tasks.append((0, 0, 0, 0))
+    return np.asarray(tasks, dtype=np.int32)
+
+
+class GpuOps:
+    def __init__(self, qm: QuantizedMatrix):
+        from .. import ops as hip_ops
+        if not torch.cuda.is_available():
+            raise RuntimeError("GpuOps requires a GPU")
+        self.lib = hip_ops.load()
+        self.hip = hip_ops
+        self.qm = qm
+        assert qm.gidx.is_cuda, "QuantizedMatrix must be on the GPU"
+        self.device = qm.gidx.device
+        cuts = qm.cuts
+        self.n_bins = cuts.total_bins
+        dev = self.device
+        self.cut_ptrs = torch.from_numpy(
+            cuts.ptrs.astype(np.int32)).to(dev)
+        self.n_bins_feat = torch.from_numpy(
+            np.diff(cuts.ptrs).astype(np.int32)).to(dev)
+        self.cut_values = torch.from_numpy(cuts.values).to(dev)
+        # feature groups for LDS privatization
+        widths = np.diff(cuts.ptrs)
+        groups_f = [0]
+        groups_b = [0]
+        acc = 0
+        for f, w in enumerate(widths):
+            if acc + w > LDS_MAX_GROUP_BINS and acc > 0:
+                groups_f.append(f)
+                groups_b.append(int(cuts.ptrs[f]))
+                acc = 0
+            acc += int(w)
+        groups_f.append(len(widths))
+        groups_b.append(int(cuts.ptrs[-1]))
+        self.feat_group_start = torch.tensor(groups_f, dtype=torch.int32,
+                                             device=dev)
+        self.bin_group_start = torch.tensor(groups_b, dtype=torch.int32,
+                                            device=dev)
+        self.n_groups = len(groups_f) - 1
+        self.max_group_bins = int(np.max(np.diff(groups_b)))
+        self.use_shared = 1 if self.max_group_bins <= LDS_MAX_GROUP_BINS else 0
+        self._ridx_out: Optional[torch.Tensor] = None
+        if cuts.feature_types is not None:
+            self.cat_feature = torch.tensor(
+                [1 if t == "c" else 0 for t in cuts.feature_types],
+                dtype=torch.uint8, device=dev)
+        else:
+            self.cat_feature = None
+        g8 = qm.gidx.dtype == torch.uint8
+        self._gidx8 = qm.gidx if g8 else None
+        self._gidx16 = None if g8 else qm.gidx
+
+    # ------------------------------------------------------------------
+    def _gidx_ptrs(self):
+        return self.hip.ptr(self._gidx8), self.hip.ptr(self._gidx16)
+
+    def make_ridx(self, n_rows: int) -> torch.Tensor:
+        self._ridx_out = torch.empty(n_rows, dtype=torch.int32,
+                                     device=self.device)
+        return torch.arange(n_rows, dtype=torch.int32, device=self.device)
+
+    def root_sum(self, qgpair: torch.Tensor) -> Tuple[int, int]:
+        s = qgpair.to(torch.int64).sum(dim=0)
+        collective.allreduce_sum_(s)
+        host = s.cpu()
+        return int(host[0]), int(host[1])
+
+    def build_hist(self, qgpair: torch.Tensor, ridx: torch.Tensor,
+                   segments: Sequence[Tuple[int, int]]) -> torch.Tensor:
+        k = len(segments)
+        out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64,
+                          device=self.device)
+        tasks_np = _chunk_tasks(segments)
+        tasks = torch.from_numpy(tasks_np).to(self.device, non_blocking=True)
+        p8, p16 = self._gidx_ptrs()
+        self.lib.gbt_hist(
+            p8, p16, self.qm.n_features, self.hip.ptr(qgpair),
+            self.hip.ptr(ridx), self.hip.ptr(tasks), len(tasks_np),
+            self.hip.ptr(out), self.n_bins,
+            self.hip.ptr(self.feat_group_start),
+            self.hip.ptr(self.bin_group_start), self.n_groups,
+            self.max_group_bins, self.hip.ptr(self.cut_ptrs),
+            self.use_shared, self.hip.stream())
+        return out
+
+    def allreduce_hist(self, hist: torch.Tensor) -> torch.Tensor:
+        collective.allreduce_sum_(hist)
+        return hist
+
+    def evaluate_splits(self, hist: torch.Tensor, quantizer: GradQuantizer,
+                        parent_sums: Sequence[Tuple[int, int]],
+                        nids: Sequence[int], param: TrainParam,
+                        feature_sets=None, monotone=None, cat_mask=None,
+                        node_bounds=None) -> List[SplitEntry]:
+        k = len(nids)
+        f = self.qm.n_features
+        dev = self.device
+        parents = torch.tensor(parent_sums, dtype=torch.int64,
+                               device=dev).view(k, 2)
+        gain = torch.empty((k, f), dtype=torch.float64, device=dev)
+        bins = torch.empty((k, f), dtype=torch.int32, device=dev)
+        dirs = torch.empty((k, f), dtype=torch.uint8, device=dev)
+        lsum = torch.empty((k, f, 2), dtype=torch.int64, device=dev)
+        mono_t = None
+        if monotone is not None:
+            mono_t = torch.from_numpy(
+                np.asarray(monotone, np.int8)).to(dev)
+        bounds_t = None
+        if node_bounds is not None:
+            bounds_t = torch.from_numpy(
+                np.ascontiguousarray(node_bounds, np.float64)).to(dev)
+        mask_t = None
+        if feature_sets is not None and any(fs is not None for fs in feature_sets):
+            m = np.zeros((k, f), dtype=np.uint8)
+            for i, fs in enumerate(feature_sets):
+                if fs is None:
+                    m[i] = 1
+                else:
+                    m[i, np.asarray(fs, np.int64)] = 1
+            mask_t = torch.from_numpy(m).to(dev)
+        self.lib.gbt_evaluate(
+            self.hip.ptr(hist), k, self.n_bins, f,
+            self.hip.ptr(self.cut_ptrs), self.hip.ptr(parents),
+            quantizer.g_scale, quantizer.h_scale,
+            param.reg_lambda, param.reg_alpha, param.max_delta_step,
+            param.min_child_weight, self.hip.ptr(mono_t),
+            self.hip.ptr(bounds_t), self.hip.ptr(mask_t),
+            self.hip.ptr(self.cat_feature),
+            self.hip.ptr(gain), self.hip.ptr(bins), self.hip.ptr(dirs),
+            self.hip.ptr(lsum), self.hip.stream())
+        best_f = torch.argmax(gain, dim=1)           # [k]
+        ar = torch.arange(k, device=dev)
+        sel_gain = gain[ar, best_f].cpu().numpy()
+        sel_bin = bins[ar, best_f].cpu().numpy()
+        sel_dir = dirs[ar, best_f].cpu().numpy()
+        sel_lsum = lsum[ar, best_f].cpu().numpy()
+        best_f = best_f.cpu().numpy()
+        cuts = self.qm.cuts
+        out = []
+        for i, nid in enumerate(nids):
+            e = SplitEntry(nid=int(nid), g_scale=quantizer.g_scale,
+                           h_scale=quantizer.h_scale)
+            if np.isfinite(sel_gain[i]) and sel_bin[i] >= 0:
+                fidx = int(best_f[i])
+                e.gain = float(sel_gain[i])
+                e.feature = fidx
+                e.split_bin = int(sel_bin[i])
+                e.default_left = bool(sel_dir[i])
+                e.left_gq = int(sel_lsum[i, 0])
+                e.left_hq = int(sel_lsum[i, 1])
+                e.right_gq = int(parent_sums[i][0]) - e.left_gq
+                e.right_hq = int(parent_sums[i][1]) - e.left_hq
+                e.is_cat = bool(cat_mask is not None and cat_mask[fidx])
+                if e.is_cat:
+                    e.cat_bits = np.array(
+                        [e.split_bin - int(cuts.ptrs[fidx])], dtype=np.int32)
+            out.append(e)
+        return out
+
+    def partition(self, ridx: torch.Tensor,
+                  segments: Sequence[Tuple[int, int]],
+                  splits: Sequence[SplitEntry]
+                  ) -> List[Tuple[Tuple[int, int], Tuple[int, int]]]:
+        k = len(segments)
+        dev = self.device
+        cuts = self.qm.cuts
+        feat = np.empty(k, np.int32)
+        sbin = np.empty(k, np.int32)
+        dleft = np.empty(k, np.uint8)
+        counters = np.empty((k, 2), np.int32)
+        cat_words: List[np.ndarray] = []
+        cat_offsets = np.zeros(k + 1, np.int32)
+        any_cat = False
+        for i, ((s, e), sp) in enumerate(zip(segments, splits)):
+            feat[i] = sp.feature
+            dleft[i] = 1 if sp.default_left else 0
+            counters[i] = (s, e)
+            if sp.is_cat:
+                any_cat = True
+                sbin[i] = -1
+                nw = (int(self.qm.cuts.ptrs[sp.feature + 1]
+                          - self.qm.cuts.ptrs[sp.feature]) + 31) // 32
+                w = np.zeros(nw, np.uint32)
+                for c in sp.cat_bits:
+                    w[c >> 5] |= np.uint32(1 << (c & 31))
+                cat_words.append(w)
+                cat_offsets[i + 1] = cat_offsets[i] + nw
+            else:
+                sbin[i] = sp.split_bin - int(cuts.ptrs[sp.feature])
+                cat_words.append(np.zeros(0, np.uint32))
+                cat_offsets[i + 1] = cat_offsets[i]
+        tasks_np = _chunk_tasks(segments)
+        tasks = torch.from_numpy(tasks_np).to(dev, non_blocking=True)
+        feat_t = torch.from_numpy(feat).to(dev, non_blocking=True)
+        sbin_t = torch.from_numpy(sbin).to(dev, non_blocking=True)
+        dleft_t = torch.from_numpy(dleft).to(dev, non_blocking=True)
+        cnt_t = torch.from_numpy(counters).to(dev, non_blocking=True)
+        if any_cat:
+            cat_bits_t = torch.from_numpy(
+                np.concatenate(cat_words).view(np.int32)).to(dev)
+            cat_off_t = torch.from_numpy(cat_offsets).to(dev)
+        else:
+            cat_bits_t = None
+            cat_off_t = None
+        p8, p16 = self._gidx_ptrs()
+        self.lib.gbt_partition(
+            p8, p16, self.qm.n_features, self.hip.ptr(ridx),
+            self.hip.ptr(self._ridx_out), self.hip.ptr(tasks), len(tasks_np),
+            self.hip.ptr(feat_t), self.hip.ptr(sbin_t), self.hip.ptr(dleft_t),
+            self.hip.ptr(cat_bits_t), self.hip.ptr(cat_off_t),
+            self.hip.ptr(self.n_bins_feat), self.hip.ptr(cnt_t),
+            self.hip.stream())
+        final = cnt_t.cpu().numpy()  # sync
+        out = []
+        for i, (s, e) in enumerate(segments):
+            mid = int(final[i, 0])
+            assert mid == int(final[i, 1]), (
+                f"partition counters disagree: {final[i]}")
+            # copy partitioned range back into the primary buffer
+            ridx[s:e] = self._ridx_out[s:e]
+            out.append(((s, mid), (mid, e)))
+        return out
+
+    def leaf_partition(self, ridx: torch.Tensor,
+                       leaf_segments: Sequence[Tuple[int, int, int]],
+                       n_rows: int) -> torch.Tensor:
+        pos = torch.zeros(n_rows, dtype=torch.int32, device=self.device)
+        if not leaf_segments:
+            return pos
+        segs = [(s, e) for _, s, e in leaf_segments]
+        leaf_ids = torch.tensor([nid for nid, _, _ in leaf_segments],
+                                dtype=torch.int32, device=self.device)
+        tasks_np = _chunk_tasks(segs)
+        tasks = torch.from_numpy(tasks_np).to(self.device, non_blocking=True)
+        self.lib.gbt_leaf_partition(
+            self.hip.ptr(ridx), self.hip.ptr(tasks), len(tasks_np),
+            self.hip.ptr(leaf_ids), self.hip.ptr(pos), self.hip.stream())
+        return pos
+
+
+# ---------------------------------------------------------------------------
+# prediction
+
+
+class _ForestArrays:
+    """Device SoA for a tree range [lo, hi) of a booster."""
+
+    def __init__(self, booster, lo: int, hi: int, device):
+        trees = booster.trees[lo:hi]
+        offs = np.zeros(len(trees) + 1, np.int32)
+        for i, t in enumerate(trees):
+            offs[i + 1] = offs[i] + t.n_nodes
+        total = int(offs[-1])
+        left = np.empty(total, np.int32)
+        right = np.empty(total, np.int32)
+        sidx = np.empty(total, np.int32)
+        cond = np.empty(total, np.float32)
+        dft = np.empty(total, np.uint8)
+        stype = np.empty(total, np.uint8)
+        cat_off = np.zeros(total + 1, np.int32)
+        cat_bits: List[np.ndarray] = []
+        pos = 0
+        for i, t in enumerate(trees):
+            n = t.n_nodes
+            o = offs[i]
+            left[o:o + n] = t.left[:n]
+            right[o:o + n] = t.right[:n]
+            sidx[o:o + n] = t.split_index[:n]
+            cond[o:o + n] = t.split_cond[:n]
+            dft[o:o + n] = t.default_left[:n]
+            stype[o:o + n] = t.split_type[:n]
+            for nid in range(n):
+                if t.split_type[nid] == 1 and nid in t.cat_segments:
+                    cats = t.cat_segments[nid]
+                    nw = (int(cats.max()) >> 5) + 1 if len(cats) else 0
+                    w = np.zeros(nw, np.uint32)
+                    for c in cats:
+                        w[c >> 5] |= np.uint32(1 << (c & 31))
+                    cat_bits.append(w)
+                    pos += nw
+                cat_off[o + nid + 1] = pos
+        self.tree_offsets = torch.from_numpy(offs).to(device)
+        self.left = torch.from_numpy(left).to(device)
+        self.right = torch.from_numpy(right).to(device)
+        self.split_index = torch.from_numpy(sidx).to(device)
+        self.split_cond = torch.from_numpy(cond).to(device)
+        self.default_left = torch.from_numpy(dft).to(device)
+        self.split_type = torch.from_numpy(stype).to(device)
+        self.cat_offsets = torch.from_numpy(cat_off).to(device)
+        if cat_bits:
+            self.cat_bits = torch.from_numpy(
+                np.concatenate(cat_bits).view(np.int32)).to(device)
+        else:
+            self.cat_bits = torch.zeros(1, dtype=torch.int32, device=device)
+        self.tree_group = torch.tensor(
+            [booster.tree_info[t] for t in range(lo, hi)],
+            dtype=torch.int32, device=device)
+        self.n_trees = len(trees)
+
+
+def predict_margin_gpu(booster, dmat, out_margin: torch.Tensor,
+                       lo: int, hi: int,
+                       out_leaf: Optional[torch.Tensor] = None) -> torch.Tensor:
+    from .. import ops as hip_ops
+    lib = hip_ops.load()
+    device = out_margin.device
+    fa = _ForestArrays(booster, lo, hi, device)
+    X = torch.from_numpy(dmat.raw_data()).to(device)
+    n = dmat.num_row()
+    missing = dmat.missing
+    missing_is_nan = 1 if np.isnan(missing) else 0
+    lib.gbt_predict(
+        hip_ops.ptr(X), n, dmat.num_col(), float(0.0 if missing_is_nan else missing),
+        missing_is_nan, hip_ops.ptr(fa.tree_offsets), hip_ops.ptr(fa.left),
+        hip_ops.ptr(fa.right), hip_ops.ptr(fa.split_index),
+        hip_ops.ptr(fa.split_cond), hip_ops.ptr(fa.default_left),
+        hip_ops.ptr(fa.split_type), hip_ops.ptr(fa.cat_offsets),
+        hip_ops.ptr(fa.cat_bits), hip_ops.ptr(fa.tree_group), fa.n_trees,
+        out_margin.shape[1], hip_ops.ptr(out_margin), hip_ops.ptr(out_leaf),
+        hip_ops.stream())
+    return out_margin

@@ -1,0 +1,84 @@
+"""ctypes loader for the HIP kernel library (libgbt_hip.so).
+
+The kernels are plain HIP compiled with hipcc for gfx950 — no
+compatibility layers.  Tensors are passed as raw device pointers plus
+torch's current HIP stream, so kernel launches land on the same stream
+as surrounding torch ops (no extra synchronization).
+
+On a GPU machine a missing/unbuildable library is a HARD error — the
+HIP path must never silently fall back to eager torch.
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Optional
+
+import torch
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_HERE, "libgbt_hip.so")
+_lib: Optional[ctypes.CDLL] = None
+
+_c = ctypes
+_p = _c.c_void_p
+_i = _c.c_int
+_i64 = _c.c_int64
+_f = _c.c_float
+_d = _c.c_double
+
+_SIGS = {
+    "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p],
+    "gbt_partition": [_p, _p, _i, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p, _p, _p],
+    "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _d, _d, _d, _d, _d, _d,
+                     _p, _p, _p, _p, _p, _p, _p, _p, _p],
+    "gbt_compress": [_p, _i64, _i, _p, _p, _p, _f, _i, _p, _p, _p],
+    "gbt_predict": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
+                    _p, _p, _i, _i, _p, _p, _p],
+    "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
+    "gbt_shap": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p, _p,
+                 _p, _p, _i, _i, _i, _p, _p],
+}
+
+
+def load() -> ctypes.CDLL:
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB_PATH):
+        try:
+            from .build_ext import build
+            build()
+        except Exception as e:  # noqa: BLE001
+            raise RuntimeError(
+                f"HIP kernel library missing and build failed: {e}. "
+                f"Run `python {os.path.join(_HERE, 'build_ext.py')}`."
+            ) from e
+    _lib = ctypes.CDLL(_LIB_PATH)
+    for name, argtypes in _SIGS.items():
+        try:
+            fn = getattr(_lib, name)
+        except AttributeError:
+            continue  # optional kernels (shap) may not be built yet
+        fn.argtypes = argtypes
+        fn.restype = None
+    return _lib
+
+
+def available() -> bool:
+    try:
+        load()
+        return True
+    except RuntimeError:
+        return False
+
+
+def ptr(t: Optional[torch.Tensor]):
+    if t is None:
+        return None
+    assert t.is_contiguous(), "kernel arg tensors must be contiguous"
+    return _c.c_void_p(t.data_ptr())
+
+
+def stream() -> _c.c_void_p:
+    return _c.c_void_p(torch.cuda.current_stream().cuda_stream)

@@ -1,0 +1,215 @@
+// Split evaluation for CDNA4: one wave64 per (feature, node).
+//
+// Reference behavior: src/tree/gpu_hist/evaluate_splits.cu
+// EvaluateSplitsKernel — which hard-codes 32-thread warps; this is a
+// wave64 re-design: 64 lanes scan a feature's bins in 64-wide chunks
+// with an int64 shuffle-based inclusive scan (exact: quantized sums),
+// then compute gains in float64 with the exact operation order of the
+// numpy oracle (xgboost_amd/splits.py) so results match bit-for-bit.
+#include "gbt_kernels.h"
+
+namespace {
+
+struct Best {
+  double gain;
+  int bin;       // global bin id
+  int dir;       // 1 = missing-left
+  long long lg;  // left sums (quantized)
+  long long lh;
+};
+
+__device__ __forceinline__ bool Better(const Best& a, const Best& b) {
+  // match numpy: strictly-greater gain wins; ties keep (dir asc, bin asc)
+  if (a.gain != b.gain) return a.gain > b.gain;
+  if (a.dir != b.dir) return a.dir < b.dir;
+  return a.bin < b.bin;
+}
+
+__device__ __forceinline__ double ThresholdL1(double g, double alpha) {
+  if (alpha == 0.0) return g;
+  double s = (g > 0.0) ? 1.0 : ((g < 0.0) ? -1.0 : 0.0);
+  double m = fabs(g) - alpha;
+  if (m < 0.0) m = 0.0;
+  return s * m;
+}
+
+struct Params {
+  double lam, alpha, mds, mcw;
+  double lo, hi;  // node weight bounds
+};
+
+__device__ __forceinline__ double CalcWeight(double g, double h,
+                                             const Params& p) {
+  double w = -ThresholdL1(g, p.alpha) / (h + p.lam);
+  if (p.mds > 0.0) {
+    w = fmin(fmax(w, -p.mds), p.mds);
+  }
+  return w;
+}
+
+__device__ __forceinline__ double GainGivenWeight(double g, double h, double w,
+                                                  const Params& p) {
+  return -(2.0 * g * w + (h + p.lam) * (w * w));
+}
+
+__device__ __forceinline__ long long ShflUpLL(long long v, int delta) {
+  return __shfl_up(v, delta, 64);
+}
+
+}  // namespace
+
+__global__ __launch_bounds__(64) void EvaluateKernel(
+    const int64_t* __restrict__ hist, int n_nodes, int n_bins, int n_features,
+    const int32_t* __restrict__ cut_ptrs,
+    const int64_t* __restrict__ parent_sums, double g_scale, double h_scale,
+    double reg_lambda, double reg_alpha, double max_delta_step,
+    double min_child_weight, const int8_t* __restrict__ monotone,
+    const double* __restrict__ node_bounds,
+    const uint8_t* __restrict__ feature_mask,
+    const uint8_t* __restrict__ cat_feature, double* __restrict__ out_gain,
+    int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dir,
+    int64_t* __restrict__ out_lsum) {
+  const int f = blockIdx.x;
+  const int node = blockIdx.y;
+  const int lane = threadIdx.x;
+  const size_t out_idx = (size_t)node * n_features + f;
+
+  if (feature_mask != nullptr && feature_mask[out_idx] == 0) {
+    if (lane == 0) {
+      out_gain[out_idx] = -INFINITY;
+      out_bin[out_idx] = -1;
+    }
+    return;
+  }
+
+  Params p;
+  p.lam = reg_lambda;
+  p.alpha = reg_alpha;
+  p.mds = max_delta_step;
+  p.mcw = min_child_weight;
+  p.lo = node_bounds ? node_bounds[2 * node] : -INFINITY;
+  p.hi = node_bounds ? node_bounds[2 * node + 1] : INFINITY;
+  const int mono = monotone ? (int)monotone[f] : 0;
+  const bool is_cat = cat_feature && cat_feature[f];
+
+  const int fb0 = cut_ptrs[f];
+  const int fb1 = cut_ptrs[f + 1];
+  const long long pg = parent_sums[2 * node];
+  const long long ph = parent_sums[2 * node + 1];
+  const double inv_g = 1.0 / g_scale;
+  const double inv_h = 1.0 / h_scale;
+
+  const int64_t* nh = hist + (size_t)node * n_bins * 2;
+
+  // pass 1: feature totals (wave reduce)
+  long long fg = 0, fh = 0;
+  for (int b = fb0 + lane; b < fb1; b += 64) {
+    fg += nh[2 * b];
+    fh += nh[2 * b + 1];
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    fg += __shfl_down(fg, off, 64);
+    fh += __shfl_down(fh, off, 64);
+  }
+  fg = __shfl(fg, 0, 64);
+  fh = __shfl(fh, 0, 64);
+  const long long miss_g = pg - fg;
+  const long long miss_h = ph - fh;
+
+  const double pw = CalcWeight(pg * inv_g, ph * inv_h, p);
+  const double parent_gain = GainGivenWeight(pg * inv_g, ph * inv_h, pw, p);
+
+  Best best{-INFINITY, -1, 0, 0, 0};
+
+  for (int dir = 0; dir < 2; ++dir) {
+    const long long add_g = dir ? miss_g : 0;
+    const long long add_h = dir ? miss_h : 0;
+    long long run_g = 0, run_h = 0;
+    for (int b0 = fb0; b0 < fb1; b0 += 64) {
+      const int b = b0 + lane;
+      const bool valid = b < fb1;
+      long long sg = valid ? (long long)nh[2 * b] : 0;
+      long long sh = valid ? (long long)nh[2 * b + 1] : 0;
+      long long glq, hlq;
+      if (is_cat) {
+        // one-vs-rest: category bin goes RIGHT
+        glq = pg - sg - (dir ? 0 : miss_g);
+        hlq = ph - sh - (dir ? 0 : miss_h);
+      } else {
+        // inclusive wave scan (int64: exact)
+        for (int off = 1; off < 64; off <<= 1) {
+          const long long tg = ShflUpLL(sg, off);
+          const long long th = ShflUpLL(sh, off);
+          if (lane >= off) {
+            sg += tg;
+            sh += th;
+          }
+        }
+        sg += run_g;
+        sh += run_h;
+        glq = sg + add_g;
+        hlq = sh + add_h;
+      }
+      if (valid) {
+        const bool last_bin = (b == fb1 - 1) && !is_cat;
+        const long long grq = pg - glq;
+        const long long hrq = ph - hlq;
+        const double gl = glq * inv_g;
+        const double hl = hlq * inv_h;
+        const double gr = grq * inv_g;
+        const double hr = hrq * inv_h;
+        double wl = CalcWeight(gl, hl, p);
+        double wr = CalcWeight(gr, hr, p);
+        wl = fmin(fmax(wl, p.lo), p.hi);
+        wr = fmin(fmax(wr, p.lo), p.hi);
+        bool ok = (hl >= p.mcw) && (hr >= p.mcw) && !last_bin;
+        if (mono > 0) ok = ok && (wl <= wr);
+        if (mono < 0) ok = ok && (wl >= wr);
+        if (ok) {
+          const double gain = GainGivenWeight(gl, hl, wl, p)
+                              + GainGivenWeight(gr, hr, wr, p) - parent_gain;
+          Best cand{gain, b, dir, glq, hlq};
+          if (isfinite(gain) && Better(cand, best)) best = cand;
+        }
+      }
+      if (!is_cat) {
+        run_g = __shfl(sg, 63, 64);
+        run_h = __shfl(sh, 63, 64);
+      }
+    }
+  }
+
+  // wave argmax reduce with full tie key
+  for (int off = 32; off > 0; off >>= 1) {
+    Best other;
+    other.gain = __shfl_down(best.gain, off, 64);
+    other.bin = __shfl_down(best.bin, off, 64);
+    other.dir = __shfl_down(best.dir, off, 64);
+    other.lg = __shfl_down(best.lg, off, 64);
+    other.lh = __shfl_down(best.lh, off, 64);
+    if (other.bin >= 0 && (best.bin < 0 || Better(other, best))) best = other;
+  }
+  if (lane == 0) {
+    out_gain[out_idx] = best.gain;
+    out_bin[out_idx] = best.bin;
+    out_dir[out_idx] = (uint8_t)best.dir;
+    out_lsum[2 * out_idx] = best.lg;
+    out_lsum[2 * out_idx + 1] = best.lh;
+  }
+}
+
+extern "C" void gbt_evaluate(
+    const int64_t* hist, int n_nodes, int n_bins, int n_features,
+    const int32_t* cut_ptrs, const int64_t* parent_sums, double g_scale,
+    double h_scale, double reg_lambda, double reg_alpha,
+    double max_delta_step, double min_child_weight, const int8_t* monotone,
+    const double* node_bounds, const uint8_t* feature_mask,
+    const uint8_t* cat_feature, double* out_gain, int32_t* out_bin,
+    uint8_t* out_dir, int64_t* out_lsum, hipStream_t stream) {
+  dim3 grid(n_features, n_nodes);
+  hipLaunchKernelGGL(EvaluateKernel, grid, dim3(64), 0, stream, hist, n_nodes,
+                     n_bins, n_features, cut_ptrs, parent_sums, g_scale,
+                     h_scale, reg_lambda, reg_alpha, max_delta_step,
+                     min_child_weight, monotone, node_bounds, feature_mask,
+                     cat_feature, out_gain, out_bin, out_dir, out_lsum);
+}

@@ -1,0 +1,122 @@
+// Gradient histogram build for CDNA4 (gfx950).
+//
+// Reference behavior: src/tree/gpu_hist/histogram.cu StHistKernel — but
+// re-designed for MI355X rather than translated:
+//  - LDS-privatized per-workgroup histogram covering a whole feature
+//    group (160 KiB LDS/CU lets one group hold up to ~9K bins at
+//    16 B/bin; Higgs 28x256=7168 bins fits in a single group, so the
+//    entire node histogram lives in LDS and is flushed once).
+//  - 64-wide wavefronts; block = 256 threads = 4 waves.
+//  - int32 quantized gradients accumulated into int64 bins via
+//    atomicAdd(u64) — ds_add_u64 on LDS, global_atomic_add_u64 on HBM:
+//    deterministic regardless of ordering (two's complement wraparound
+//    is exact for signed fixed-point).
+//  - blockIdx.x enumerates (node, row-chunk) tasks; blockIdx.y the
+//    feature group (grid-filling: tasks are sized so n_tasks >= ~2048
+//    when rows allow, covering 256 CUs across 8 XCDs).
+#include "gbt_kernels.h"
+
+#ifndef GBT_HIST_BLOCK
+#define GBT_HIST_BLOCK 256
+#endif
+
+template <typename BinT, bool kUseShared>
+__global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
+    const BinT* __restrict__ gidx, int n_features,
+    const int32_t* __restrict__ qgpair, const int32_t* __restrict__ ridx,
+    const BlockTask* __restrict__ tasks,
+    int64_t* __restrict__ out_hist, int n_bins,
+    const int32_t* __restrict__ feat_group_start,
+    const int32_t* __restrict__ bin_group_start,
+    const int32_t* __restrict__ cut_ptrs) {
+  const BlockTask task = tasks[blockIdx.x];
+  const int group = blockIdx.y;
+  const int f_begin = feat_group_start[group];
+  const int f_end = feat_group_start[group + 1];
+  const int bin_begin = bin_group_start[group];
+  const int bin_end = bin_group_start[group + 1];
+  const int group_bins = bin_end - bin_begin;
+
+  extern __shared__ unsigned long long smem[];  // [group_bins][2]
+  if (kUseShared) {
+    for (int i = threadIdx.x; i < group_bins * 2; i += blockDim.x) {
+      smem[i] = 0ULL;
+    }
+    __syncthreads();
+  }
+
+  unsigned long long* hist_s = smem;
+  int64_t* hist_g = out_hist + (size_t)task.out_slot * n_bins * 2;
+
+  // each thread walks rows with stride blockDim; per row it reads the
+  // gradient pair once and then the group's bin bytes.
+  for (int i = task.row_begin + threadIdx.x; i < task.row_end;
+       i += blockDim.x) {
+    const int row = ridx[i];
+    const long long g = qgpair[2 * (size_t)row];
+    const long long h = qgpair[2 * (size_t)row + 1];
+    const BinT* rowbins = gidx + (size_t)row * n_features;
+    for (int f = f_begin; f < f_end; ++f) {
+      const int local = (int)rowbins[f];
+      const int fbins = cut_ptrs[f + 1] - cut_ptrs[f];
+      if (local >= fbins) continue;  // missing sentinel
+      const int gbin = cut_ptrs[f] + local;
+      if (kUseShared) {
+        const int sbin = gbin - bin_begin;
+        atomicAdd(&hist_s[2 * sbin], (unsigned long long)g);
+        atomicAdd(&hist_s[2 * sbin + 1], (unsigned long long)h);
+      } else {
+        atomicAdd((unsigned long long*)&hist_g[2 * gbin],
+                  (unsigned long long)g);
+        atomicAdd((unsigned long long*)&hist_g[2 * gbin + 1],
+                  (unsigned long long)h);
+      }
+    }
+  }
+
+  if (kUseShared) {
+    __syncthreads();
+    for (int i = threadIdx.x; i < group_bins * 2; i += blockDim.x) {
+      const unsigned long long v = hist_s[i];
+      if (v != 0ULL) {
+        atomicAdd((unsigned long long*)&hist_g[2 * bin_begin + i], v);
+      }
+    }
+  }
+}
+
+extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
+                         int n_features, const int32_t* qgpair,
+                         const int32_t* ridx, const BlockTask* tasks,
+                         int n_tasks, int64_t* out_hist, int n_bins,
+                         const int32_t* feat_group_start,
+                         const int32_t* bin_group_start, int n_groups,
+                         int max_group_bins, const int32_t* cut_ptrs,
+                         int use_shared, hipStream_t stream) {
+  dim3 grid(n_tasks, n_groups);
+  dim3 block(GBT_HIST_BLOCK);
+  size_t shmem = use_shared ? (size_t)max_group_bins * 2 * sizeof(int64_t) : 0;
+  if (gidx8 != nullptr) {
+    if (use_shared) {
+      hipLaunchKernelGGL((HistKernel<uint8_t, true>), grid, block, shmem,
+                         stream, gidx8, n_features, qgpair, ridx, tasks,
+                         out_hist, n_bins, feat_group_start, bin_group_start,
+                         cut_ptrs);
+    } else {
+      hipLaunchKernelGGL((HistKernel<uint8_t, false>), grid, block, 0, stream,
+                         gidx8, n_features, qgpair, ridx, tasks, out_hist,
+                         n_bins, feat_group_start, bin_group_start, cut_ptrs);
+    }
+  } else {
+    if (use_shared) {
+      hipLaunchKernelGGL((HistKernel<uint16_t, true>), grid, block, shmem,
+                         stream, gidx16, n_features, qgpair, ridx, tasks,
+                         out_hist, n_bins, feat_group_start, bin_group_start,
+                         cut_ptrs);
+    } else {
+      hipLaunchKernelGGL((HistKernel<uint16_t, false>), grid, block, 0, stream,
+                         gidx16, n_features, qgpair, ridx, tasks, out_hist,
+                         n_bins, feat_group_start, bin_group_start, cut_ptrs);
+    }
+  }
+}
