@@ -40,7 +40,13 @@ def _unwind(path: List[List[float]], idx: int) -> None:
             n = t - path[j][3] * zero * (length - j) / (length + 1)
         else:
             path[j][3] = path[j][3] * (length + 1) / (zero * (length - j))
-    del path[idx]
+    # shift feature/zero/one down; pweights stay in place (the canonical
+    # UNWIND keeps the recomputed pweights at indices 0..length-1)
+    for j in range(idx, length):
+        path[j][0] = path[j + 1][0]
+        path[j][1] = path[j + 1][1]
+        path[j][2] = path[j + 1][2]
+    del path[length]
 
 
 def _unwound_sum(path: List[List[float]], idx: int) -> float:
