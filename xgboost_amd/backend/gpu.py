@@ -48,6 +48,64 @@ def _chunk_tasks(segments: Sequence[Tuple[int, int]],
     return np.asarray(tasks, dtype=np.int32)
 
 
+class _PinnedStager:
+    """Reusable pinned-host staging buffer: pack many small numpy arrays
+    into one H2D copy per level (profiling: dozens of tiny pageable
+    uploads per level cost more than the kernels)."""
+
+    ALIGN = 8
+    NSLOTS = 4  # ring: a slot is never reused before >=2 stream syncs
+
+    def __init__(self, device):
+        self.device = device
+        self.slots = [[torch.empty(1 << 16, dtype=torch.uint8,
+                                   pin_memory=True),
+                       torch.empty(1 << 16, dtype=torch.uint8, device=device)]
+                      for _ in range(self.NSLOTS)]
+        self.cur = 0
+
+    def upload(self, arrays):
+        """arrays: list of (np.ndarray | None); returns device tensors."""
+        slot = self.slots[self.cur]
+        self.cur = (self.cur + 1) % self.NSLOTS
+        offs = []
+        total = 0
+        for a in arrays:
+            if a is None:
+                offs.append((None, 0, 0))
+                continue
+            a = np.ascontiguousarray(a)
+            nbytes = a.nbytes
+            total = (total + self.ALIGN - 1) & ~(self.ALIGN - 1)
+            offs.append((a, total, nbytes))
+            total += nbytes
+        if total > slot[0].numel():
+            cap = 1 << max(int(total).bit_length(), 16)
+            slot[0] = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
+            slot[1] = torch.empty(cap, dtype=torch.uint8, device=self.device)
+        host = slot[0].numpy()
+        for a, off, nbytes in offs:
+            if a is not None:
+                host[off:off + nbytes] = a.view(np.uint8).reshape(-1)
+        slot[1][:total].copy_(slot[0][:total], non_blocking=True)
+        out = []
+        for a, off, nbytes in offs:
+            if a is None:
+                out.append(None)
+                continue
+            t = slot[1][off:off + nbytes]
+            td = t.view(_np_to_torch(a.dtype)).view(*a.shape)
+            out.append(td)
+        return out
+
+
+def _np_to_torch(dt: np.dtype) -> torch.dtype:
+    return {np.dtype(np.int32): torch.int32, np.dtype(np.uint8): torch.uint8,
+            np.dtype(np.int8): torch.int8, np.dtype(np.int64): torch.int64,
+            np.dtype(np.float64): torch.float64,
+            np.dtype(np.float32): torch.float32}[dt]
+
+
 class GpuOps:
     def __init__(self, qm: QuantizedMatrix):
         from .. import ops as hip_ops
@@ -96,6 +154,7 @@ class GpuOps:
         g8 = qm.gidx.dtype == torch.uint8
         self._gidx8 = qm.gidx if g8 else None
         self._gidx16 = None if g8 else qm.gidx
+        self.stager = _PinnedStager(dev)
 
     # ------------------------------------------------------------------
     def _gidx_ptrs(self):
@@ -118,7 +177,7 @@ class GpuOps:
         out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64,
                           device=self.device)
         tasks_np = _chunk_tasks(segments)
-        tasks = torch.from_numpy(tasks_np).to(self.device, non_blocking=True)
+        (tasks,) = self.stager.upload([tasks_np])
         p8, p16 = self._gidx_ptrs()
         self.lib.gbt_hist(
             p8, p16, self.qm.n_features, self.hip.ptr(qgpair),
@@ -142,29 +201,25 @@ class GpuOps:
         k = len(nids)
         f = self.qm.n_features
         dev = self.device
-        parents = torch.tensor(parent_sums, dtype=torch.int64,
-                               device=dev).view(k, 2)
         gain = torch.empty((k, f), dtype=torch.float64, device=dev)
         bins = torch.empty((k, f), dtype=torch.int32, device=dev)
         dirs = torch.empty((k, f), dtype=torch.uint8, device=dev)
         lsum = torch.empty((k, f, 2), dtype=torch.int64, device=dev)
-        mono_t = None
-        if monotone is not None:
-            mono_t = torch.from_numpy(
-                np.asarray(monotone, np.int8)).to(dev)
-        bounds_t = None
-        if node_bounds is not None:
-            bounds_t = torch.from_numpy(
-                np.ascontiguousarray(node_bounds, np.float64)).to(dev)
-        mask_t = None
+        parents_np = np.asarray(parent_sums, dtype=np.int64).reshape(k, 2)
+        mono_np = (np.asarray(monotone, np.int8)
+                   if monotone is not None else None)
+        bounds_np = (np.ascontiguousarray(node_bounds, np.float64)
+                     if node_bounds is not None else None)
+        mask_np = None
         if feature_sets is not None and any(fs is not None for fs in feature_sets):
-            m = np.zeros((k, f), dtype=np.uint8)
+            mask_np = np.zeros((k, f), dtype=np.uint8)
             for i, fs in enumerate(feature_sets):
                 if fs is None:
-                    m[i] = 1
+                    mask_np[i] = 1
                 else:
-                    m[i, np.asarray(fs, np.int64)] = 1
-            mask_t = torch.from_numpy(m).to(dev)
+                    mask_np[i, np.asarray(fs, np.int64)] = 1
+        parents, mono_t, bounds_t, mask_t = self.stager.upload(
+            [parents_np, mono_np, bounds_np, mask_np])
         self.lib.gbt_evaluate(
             self.hip.ptr(hist), k, self.n_bins, f,
             self.hip.ptr(self.cut_ptrs), self.hip.ptr(parents),
@@ -177,11 +232,21 @@ class GpuOps:
             self.hip.ptr(lsum), self.hip.stream())
         best_f = torch.argmax(gain, dim=1)           # [k]
         ar = torch.arange(k, device=dev)
-        sel_gain = gain[ar, best_f].cpu().numpy()
-        sel_bin = bins[ar, best_f].cpu().numpy()
-        sel_dir = dirs[ar, best_f].cpu().numpy()
-        sel_lsum = lsum[ar, best_f].cpu().numpy()
-        best_f = best_f.cpu().numpy()
+        # pack the per-node winners into ONE device tensor -> one D2H sync
+        packed = torch.stack([
+            gain[ar, best_f].view(torch.int64),
+            bins[ar, best_f].to(torch.int64),
+            dirs[ar, best_f].to(torch.int64),
+            lsum[ar, best_f, 0],
+            lsum[ar, best_f, 1],
+            best_f.to(torch.int64),
+        ], dim=1)
+        host = packed.cpu().numpy()
+        sel_gain = host[:, 0].view(np.float64)
+        sel_bin = host[:, 1]
+        sel_dir = host[:, 2]
+        sel_lsum = host[:, 3:5]
+        best_f = host[:, 5]
         cuts = self.qm.cuts
         out = []
         for i, nid in enumerate(nids):
@@ -237,18 +302,12 @@ class GpuOps:
                 cat_words.append(np.zeros(0, np.uint32))
                 cat_offsets[i + 1] = cat_offsets[i]
         tasks_np = _chunk_tasks(segments)
-        tasks = torch.from_numpy(tasks_np).to(dev, non_blocking=True)
-        feat_t = torch.from_numpy(feat).to(dev, non_blocking=True)
-        sbin_t = torch.from_numpy(sbin).to(dev, non_blocking=True)
-        dleft_t = torch.from_numpy(dleft).to(dev, non_blocking=True)
-        cnt_t = torch.from_numpy(counters).to(dev, non_blocking=True)
-        if any_cat:
-            cat_bits_t = torch.from_numpy(
-                np.concatenate(cat_words).view(np.int32)).to(dev)
-            cat_off_t = torch.from_numpy(cat_offsets).to(dev)
-        else:
-            cat_bits_t = None
-            cat_off_t = None
+        cat_bits_np = (np.concatenate(cat_words).view(np.int32)
+                       if any_cat else None)
+        cat_off_np = cat_offsets if any_cat else None
+        tasks, feat_t, sbin_t, dleft_t, cnt_t, cat_bits_t, cat_off_t = \
+            self.stager.upload([tasks_np, feat, sbin, dleft, counters,
+                                cat_bits_np, cat_off_np])
         p8, p16 = self._gidx_ptrs()
         self.lib.gbt_partition(
             p8, p16, self.qm.n_features, self.hip.ptr(ridx),
@@ -257,14 +316,16 @@ class GpuOps:
             self.hip.ptr(cat_bits_t), self.hip.ptr(cat_off_t),
             self.hip.ptr(self.n_bins_feat), self.hip.ptr(cnt_t),
             self.hip.stream())
+        # copy partitioned ranges back into the primary buffer (1 launch)
+        self.lib.gbt_copy_ranges(
+            self.hip.ptr(self._ridx_out), self.hip.ptr(ridx),
+            self.hip.ptr(tasks), len(tasks_np), self.hip.stream())
         final = cnt_t.cpu().numpy()  # sync
         out = []
         for i, (s, e) in enumerate(segments):
             mid = int(final[i, 0])
             assert mid == int(final[i, 1]), (
                 f"partition counters disagree: {final[i]}")
-            # copy partitioned range back into the primary buffer
-            ridx[s:e] = self._ridx_out[s:e]
             out.append(((s, mid), (mid, e)))
         return out
 
@@ -275,10 +336,9 @@ class GpuOps:
         if not leaf_segments:
             return pos
         segs = [(s, e) for _, s, e in leaf_segments]
-        leaf_ids = torch.tensor([nid for nid, _, _ in leaf_segments],
-                                dtype=torch.int32, device=self.device)
         tasks_np = _chunk_tasks(segs)
-        tasks = torch.from_numpy(tasks_np).to(self.device, non_blocking=True)
+        leaf_np = np.asarray([nid for nid, _, _ in leaf_segments], np.int32)
+        tasks, leaf_ids = self.stager.upload([tasks_np, leaf_np])
         self.lib.gbt_leaf_partition(
             self.hip.ptr(ridx), self.hip.ptr(tasks), len(tasks_np),
             self.hip.ptr(leaf_ids), self.hip.ptr(pos), self.hip.stream())

@@ -228,9 +228,13 @@ class TreeGrower:
                 bsegs = [segments[n] for n, _, _ in build_nodes]
                 bh = ops.build_hist(qgpair, ridx, bsegs)
                 ops.allreduce_hist(bh)
+                # batched sibling subtraction: one kernel for the level
+                parent_stack = torch.stack(
+                    [hists[p] for _, p, _ in build_nodes])
+                sib_stack = parent_stack - bh
                 for i, (n, parent, sib) in enumerate(build_nodes):
                     hists[n] = bh[i]
-                    hists[sib] = hists[parent] - bh[i]
+                    hists[sib] = sib_stack[i]
                     del hists[parent]
 
             # 5. evaluate children

@@ -36,6 +36,7 @@ _SIGS = {
     "gbt_predict": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
                     _p, _p, _i, _i, _p, _p, _p],
     "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
+    "gbt_copy_ranges": [_p, _p, _p, _i, _p],
     "gbt_shap": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p, _p,
                  _p, _p, _i, _i, _i, _p, _p],
 }
