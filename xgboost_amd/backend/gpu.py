@@ -230,18 +230,13 @@ class GpuOps:
             self.hip.ptr(self.cat_feature),
             self.hip.ptr(gain), self.hip.ptr(bins), self.hip.ptr(dirs),
             self.hip.ptr(lsum), self.hip.stream())
-        best_f = torch.argmax(gain, dim=1)           # [k]
-        ar = torch.arange(k, device=dev)
-        # pack the per-node winners into ONE device tensor -> one D2H sync
-        packed = torch.stack([
-            gain[ar, best_f].view(torch.int64),
-            bins[ar, best_f].to(torch.int64),
-            dirs[ar, best_f].to(torch.int64),
-            lsum[ar, best_f, 0],
-            lsum[ar, best_f, 1],
-            best_f.to(torch.int64),
-        ], dim=1)
-        host = packed.cpu().numpy()
+        # device-side per-node argmax + packing -> ONE D2H sync
+        out_best = torch.empty((k, 6), dtype=torch.int64, device=dev)
+        self.lib.gbt_select_best(
+            self.hip.ptr(gain), self.hip.ptr(bins), self.hip.ptr(dirs),
+            self.hip.ptr(lsum), k, f, self.hip.ptr(out_best),
+            self.hip.stream())
+        host = out_best.cpu().numpy()
         sel_gain = host[:, 0].view(np.float64)
         sel_bin = host[:, 1]
         sel_dir = host[:, 2]

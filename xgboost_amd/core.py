@@ -203,8 +203,11 @@ class Booster:
                 self.tree_info.append(k)
                 new_trees += 1
                 # update prediction cache from leaf positions
-                leaf_vals = torch.as_tensor(
-                    tree.split_cond[:tree.n_nodes].copy(), device=margin.device)
+                leaf_np = tree.split_cond[:tree.n_nodes].copy()
+                if hasattr(ops, "stager"):
+                    (leaf_vals,) = ops.stager.upload([leaf_np])
+                else:
+                    leaf_vals = torch.as_tensor(leaf_np, device=margin.device)
                 margin[:, k] += leaf_vals[positions.to(margin.device).long()]
         self.iteration_indptr.append(self.iteration_indptr[-1] + new_trees)
         self._cache[id(dtrain)] = (margin, len(self.trees))

@@ -236,14 +236,15 @@ class TreeGrower:
                     hists[n] = bh[i]
                     hists[sib] = sib_stack[i]
                     del hists[parent]
-
-            # 5. evaluate children
-            eval_nids = [n for (b, l, r) in children for n in (l, r)
-                         if n in hists]
-            if eval_nids:
+                # 5. evaluate children: [built..., subtracted...] order so
+                # the histogram input is one cat, not k small stacks
+                eval_nids = ([n for n, _, _ in build_nodes]
+                             + [s for _, _, s in build_nodes])
+                hist_stack = torch.cat([bh, sib_stack], dim=0)
                 depth = children[0][0].depth + 1
                 entries = self._evaluate(eval_nids, node_sums, hists,
-                                         node_bounds, depth)
+                                         node_bounds, depth,
+                                         hist_stack=hist_stack)
                 for nid, e in zip(eval_nids, entries):
                     push(nid, depth, e)
 
@@ -270,10 +271,12 @@ class TreeGrower:
             return False
         return True
 
-    def _evaluate(self, nids, node_sums, hists, node_bounds, depth
+    def _evaluate(self, nids, node_sums, hists, node_bounds, depth,
+                  hist_stack: Optional[torch.Tensor] = None
                   ) -> List[SplitEntry]:
         param = self.param
-        hist = torch.stack([hists[n] for n in nids])
+        hist = (hist_stack if hist_stack is not None
+                else torch.stack([hists[n] for n in nids]))
         parent_sums = [node_sums[n] for n in nids]
         feature_sets = None
         node_feats = self.col_sampler.node_set(depth)

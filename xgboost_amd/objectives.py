@@ -572,3 +572,157 @@ def _weighted_quantile(v: np.ndarray, w: Optional[np.ndarray], q: float) -> floa
     t = q * cw[-1]
     i = int(np.searchsorted(cw, t))
     return float(v[min(i, len(v) - 1)])
+
+
+# ---------------------------------------------------------------------------
+# learning to rank (reference: src/objective/lambdarank_obj.{cc,cu},
+# pair generation lambdarank_obj.cuh:76, NDCG deltas ranking_utils.h)
+
+
+class _LambdaRankBase(Objective):
+    task = "ranking"
+
+    def __init__(self, params=None):
+        super().__init__(params)
+        self.num_pair = int(self.params.get("lambdarank_num_pair_per_sample",
+                                            0) or 0)
+        self.pair_method = str(self.params.get("lambdarank_pair_method",
+                                               "topk"))
+        self.normalize = bool(self.params.get("lambdarank_normalization",
+                                              True))
+
+    def _groups(self, info):
+        if info.group_ptr is None:
+            return np.array([0, info.num_row], dtype=np.int64)
+        return np.asarray(info.group_ptr, dtype=np.int64)
+
+    def _delta(self, y_sorted, ranks_i, ranks_j, i_idx, j_idx, inv_idcg):
+        """|delta metric| for swapping documents at ranks_i/ranks_j."""
+        raise NotImplementedError
+
+    def get_gradient(self, preds, info, it):
+        p = preds.detach().cpu().numpy().reshape(-1).astype(np.float64)
+        y = np.asarray(info.labels, np.float64).reshape(-1)
+        gp = self._groups(info)
+        g = np.zeros_like(p)
+        h = np.zeros_like(p)
+        rng = np.random.RandomState(1234 + it)
+        for gi in range(len(gp) - 1):
+            s, e = int(gp[gi]), int(gp[gi + 1])
+            if e - s < 2:
+                continue
+            self._group_gradient(p[s:e], y[s:e], g[s:e], h[s:e], rng)
+        gt = torch.as_tensor(g.astype(np.float32),
+                             device=preds.device).view(preds.shape)
+        ht = torch.as_tensor(np.maximum(h, 1e-16).astype(np.float32),
+                             device=preds.device).view(preds.shape)
+        w = _weights(info, preds)
+        if w is not None and info.weights.shape[0] == len(gp) - 1:
+            # per-group weights
+            wr = np.repeat(np.asarray(info.weights, np.float32),
+                           np.diff(gp))
+            wt = torch.as_tensor(wr, device=preds.device).view(preds.shape)
+            gt, ht = gt * wt, ht * wt
+        elif w is not None:
+            gt, ht = gt * w.view(preds.shape), ht * w.view(preds.shape)
+        return gt, ht
+
+    def _make_pairs(self, n, y, order, rng):
+        """Yield (i, j) index pairs (into the group) with y[i] > y[j]."""
+        if self.pair_method == "mean" and self.num_pair > 0:
+            k = self.num_pair
+            pairs = []
+            for i in range(n):
+                js = rng.randint(0, n, size=k)
+                for j in js:
+                    if y[i] > y[j]:
+                        pairs.append((i, j))
+                    elif y[j] > y[i]:
+                        pairs.append((j, i))
+            return pairs
+        # topk/full: all label-discordant pairs (n<=512 full, else truncate)
+        pairs = []
+        cap = 512
+        idx = order[:cap]
+        for a in range(len(idx)):
+            for b in range(a + 1, len(idx)):
+                i, j = idx[a], idx[b]
+                if y[i] > y[j]:
+                    pairs.append((i, j))
+                elif y[j] > y[i]:
+                    pairs.append((j, i))
+        return pairs
+
+    def _group_gradient(self, p, y, g, h, rng):
+        n = len(p)
+        order = np.argsort(-p, kind="stable")
+        ranks = np.empty(n, dtype=np.int64)
+        ranks[order] = np.arange(n)  # 0-based rank by prediction
+        inv_idcg = self._inv_idcg(y)
+        pairs = self._make_pairs(n, y, order, rng)
+        if not pairs:
+            return
+        total_lambda = 0.0
+        for i, j in pairs:
+            delta = self._delta(y, ranks[i], ranks[j], i, j, inv_idcg)
+            sij = p[i] - p[j]
+            rho = 1.0 / (1.0 + np.exp(sij))  # d/ds of log(1+e^-s)
+            lam = -rho * delta
+            hess = max(rho * (1.0 - rho) * delta, 1e-16)
+            g[i] += lam
+            g[j] -= lam
+            h[i] += hess
+            h[j] += hess
+            total_lambda += abs(lam)
+        if self.normalize and total_lambda > 0:
+            norm = np.log2(1.0 + total_lambda) / total_lambda
+            g *= norm
+            h *= norm
+
+    def _inv_idcg(self, y):
+        gains = np.sort(2.0 ** y - 1.0)[::-1]
+        disc = 1.0 / np.log2(np.arange(2, len(y) + 2))
+        idcg = float((gains * disc).sum())
+        return 1.0 / idcg if idcg > 0 else 0.0
+
+    def init_estimation(self, info) -> float:
+        return 0.5
+
+    def prob_to_margin(self, base_score):
+        return base_score
+
+
+@register("rank:ndcg")
+class LambdaRankNDCG(_LambdaRankBase):
+    @property
+    def default_metric(self):
+        return "ndcg"
+
+    def _delta(self, y, rank_i, rank_j, i, j, inv_idcg):
+        gain_i = 2.0 ** y[i] - 1.0
+        gain_j = 2.0 ** y[j] - 1.0
+        disc_i = 1.0 / np.log2(rank_i + 2.0)
+        disc_j = 1.0 / np.log2(rank_j + 2.0)
+        return abs((gain_i - gain_j) * (disc_i - disc_j)) * inv_idcg
+
+
+@register("rank:map")
+class LambdaRankMAP(_LambdaRankBase):
+    @property
+    def default_metric(self):
+        return "map"
+
+    def _delta(self, y, rank_i, rank_j, i, j, inv_idcg):
+        # MAP delta approximated by reciprocal-rank difference on binary rel
+        ri, rj = min(rank_i, rank_j), max(rank_i, rank_j)
+        return abs(1.0 / (ri + 1.0) - 1.0 / (rj + 1.0))
+
+
+@register("rank:pairwise")
+class LambdaRankPairwise(_LambdaRankBase):
+    @property
+    def default_metric(self):
+        return "map"
+
+    def _delta(self, y, rank_i, rank_j, i, j, inv_idcg):
+        return 1.0

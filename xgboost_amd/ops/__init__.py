@@ -37,6 +37,7 @@ _SIGS = {
                     _p, _p, _i, _i, _p, _p, _p],
     "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
     "gbt_copy_ranges": [_p, _p, _p, _i, _p],
+    "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p],
     "gbt_shap": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p, _p,
                  _p, _p, _i, _i, _i, _p, _p],
 }

@@ -198,6 +198,64 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
   }
 }
 
+// Second stage: per-node argmax over features -> packed [n_nodes, 6]
+// (gain bits, bin, dir, left_gq, left_hq, feature).  One wave per node;
+// tie rule matches numpy flat argmax: higher gain wins, ties -> lower
+// feature index (the per-feature stage already resolved bin/dir ties).
+__global__ __launch_bounds__(64) void SelectBestKernel(
+    const double* __restrict__ gain, const int32_t* __restrict__ bins,
+    const uint8_t* __restrict__ dirs, const int64_t* __restrict__ lsum,
+    int n_features, int64_t* __restrict__ out_best) {
+  const int node = blockIdx.x;
+  const int lane = threadIdx.x;
+  const size_t base = (size_t)node * n_features;
+  double best_gain = -INFINITY;
+  int best_f = -1;
+  for (int f = lane; f < n_features; f += 64) {
+    const double gv = gain[base + f];
+    if (gv > best_gain) {
+      best_gain = gv;
+      best_f = f;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    const double og = __shfl_down(best_gain, off, 64);
+    const int of = __shfl_down(best_f, off, 64);
+    if (of >= 0 && (best_f < 0 || og > best_gain ||
+                    (og == best_gain && of < best_f))) {
+      best_gain = og;
+      best_f = of;
+    }
+  }
+  if (lane == 0) {
+    int64_t* out = out_best + (size_t)node * 6;
+    if (best_f < 0 || !isfinite(best_gain)) {
+      out[0] = 0;
+      out[1] = -1;
+      out[2] = 0;
+      out[3] = 0;
+      out[4] = 0;
+      out[5] = -1;
+    } else {
+      const size_t idx = base + best_f;
+      out[0] = __double_as_longlong(best_gain);
+      out[1] = bins[idx];
+      out[2] = dirs[idx];
+      out[3] = lsum[2 * idx];
+      out[4] = lsum[2 * idx + 1];
+      out[5] = best_f;
+    }
+  }
+}
+
+extern "C" void gbt_select_best(const double* gain, const int32_t* bins,
+                                const uint8_t* dirs, const int64_t* lsum,
+                                int n_nodes, int n_features,
+                                int64_t* out_best, hipStream_t stream) {
+  hipLaunchKernelGGL(SelectBestKernel, dim3(n_nodes), dim3(64), 0, stream,
+                     gain, bins, dirs, lsum, n_features, out_best);
+}
+
 extern "C" void gbt_evaluate(
     const int64_t* hist, int n_nodes, int n_bins, int n_features,
     const int32_t* cut_ptrs, const int64_t* parent_sums, double g_scale,
