@@ -1,0 +1,91 @@
+"""Tree methods & updaters: exact/approx, prune/refresh,
+process_type=update (reference analog: tests/python/test_updaters.py)."""
+import numpy as np
+import pytest
+
+import xgboost_amd as xgb
+from conftest import make_classification, make_regression
+
+
+def test_exact_learns():
+    X, y = make_regression(2000, 6)
+    d = xgb.DMatrix(X, label=y)
+    res = {}
+    xgb.train({"objective": "reg:squarederror", "tree_method": "exact",
+               "max_depth": 5, "eta": 0.3}, d, 15, evals=[(d, "t")],
+              evals_result=res, verbose_eval=False)
+    assert res["t"]["rmse"][-1] < 0.4 * np.std(y)
+
+
+def test_exact_matches_hist_quality():
+    X, y = make_classification(2000, 6)
+    d = xgb.DMatrix(X, label=y)
+    out = {}
+    for m in ("exact", "hist", "approx"):
+        res = {}
+        xgb.train({"objective": "binary:logistic", "tree_method": m,
+                   "max_depth": 4}, d, 10, evals=[(d, "t")],
+                  evals_result=res, verbose_eval=False)
+        out[m] = res["t"]["logloss"][-1]
+    assert all(v < 0.4 for v in out.values()), out
+    assert abs(out["exact"] - out["hist"]) < 0.1
+
+
+def test_exact_with_missing():
+    X, y = make_classification(1000, 5)
+    X[np.random.RandomState(0).rand(*X.shape) < 0.2] = np.nan
+    d = xgb.DMatrix(X, label=y)
+    res = {}
+    xgb.train({"objective": "binary:logistic", "tree_method": "exact",
+               "max_depth": 4}, d, 10, evals=[(d, "t")], evals_result=res,
+              verbose_eval=False)
+    assert res["t"]["logloss"][-1] < 0.55
+
+
+def test_process_type_update_refresh():
+    X, y = make_regression(1000, 5)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 4},
+                    d, 5, verbose_eval=False)
+    p_before = bst.predict(d)
+    # refresh on the SAME data should keep predictions roughly unchanged
+    X2, y2 = make_regression(1000, 5, seed=99)
+    d2 = xgb.DMatrix(X2, label=y2)
+    bst2 = xgb.train({"objective": "reg:squarederror",
+                      "process_type": "update", "updater": "refresh"},
+                     d2, 5, xgb_model=bst, verbose_eval=False)
+    assert len(bst2.trees) == len(bst.trees)
+    # structures identical, leaf values refreshed on new data
+    for t1, t2 in zip(bst.trees, bst2.trees):
+        assert t1.n_nodes == t2.n_nodes
+        assert np.array_equal(t1.split_index[:t1.n_nodes],
+                              t2.split_index[:t2.n_nodes])
+    p_after = bst2.predict(d2)
+    rmse_new = np.sqrt(np.mean((p_after - y2) ** 2))
+    assert rmse_new < np.std(y2)  # refreshed leaves fit the new data
+
+
+def test_prune_updater():
+    from xgboost_amd.updaters import prune_tree
+    from xgboost_amd.params import make_train_param
+    X, y = make_classification(1000, 6)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 6},
+                    d, 2, verbose_eval=False)
+    big = prune_tree(bst.trees[0], make_train_param({"gamma": 0.0}))
+    small = prune_tree(bst.trees[0], make_train_param({"gamma": 1e9}))
+    assert small.n_nodes == 1  # everything pruned to root leaf
+    assert big.n_nodes == bst.trees[0].n_nodes
+
+
+def test_refresh_leaf_false_keeps_leaves():
+    X, y = make_regression(500, 4)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 3},
+                    d, 3, verbose_eval=False)
+    p1 = bst.predict(d)
+    bst2 = xgb.train({"objective": "reg:squarederror",
+                      "process_type": "update", "updater": "refresh",
+                      "refresh_leaf": False}, d, 3, xgb_model=bst,
+                     verbose_eval=False)
+    assert np.allclose(bst2.predict(d), p1, atol=1e-5)

@@ -10,8 +10,10 @@ Public API mirrors the xgboost Python package: DMatrix, QuantileDMatrix,
 Booster, train, cv, callbacks, sklearn wrappers.
 """
 from .collective import init as collective_init  # noqa: F401
+from .config import config_context, get_config, set_config  # noqa: F401
 from .core import Booster  # noqa: F401
 from .data import DMatrix, QuantileDMatrix  # noqa: F401
+from .plotting import plot_importance, plot_tree, to_graphviz  # noqa: F401
 from .training import cv, train  # noqa: F401
 from . import callback  # noqa: F401
 from . import collective  # noqa: F401
@@ -20,7 +22,8 @@ __version__ = "0.1.0"
 
 __all__ = [
     "Booster", "DMatrix", "QuantileDMatrix", "train", "cv", "callback",
-    "collective",
+    "collective", "config_context", "set_config", "get_config",
+    "plot_importance", "plot_tree", "to_graphviz",
 ]
 
 

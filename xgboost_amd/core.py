@@ -127,7 +127,21 @@ class Booster:
         bs = self.base_score if self.base_score is not None else 0.5
         return float(self.objective.prob_to_margin(bs))
 
-    def _ops_for(self, dmat: DMatrix):
+    def _ops_for(self, dmat: DMatrix, hess: Optional[np.ndarray] = None):
+        if hess is not None:
+            # approx: hessian-weighted cuts regenerated per tree
+            # (reference: grow_histmaker/grow_gpu_approx, updater_approx.cc:46)
+            from .data import quantize_dense
+            from .quantile import make_cuts
+            cuts = make_cuts(dmat.raw_data(), self.tparam.max_bin,
+                             weights=np.abs(hess) + 1e-16,
+                             feature_types=dmat.info.feature_types,
+                             missing=dmat.missing)
+            qm = quantize_dense(dmat.raw_data(), cuts, dmat.missing)
+            if self.device.type == "cuda":
+                from .backend.gpu import GpuOps
+                return GpuOps(qm.to(self.device))
+            return CpuOps(qm)
         key = id(dmat)
         ops = self._ops_cache.get(key)
         if ops is None:
@@ -147,6 +161,9 @@ class Booster:
                fobj=None) -> None:
         self._maybe_set_meta(dtrain)
         self._init_base_score(dtrain)
+        if self.tparam.process_type == "update":
+            self._update_existing(dtrain, iteration)
+            return
         margin = self._cached_margin(dtrain)
         if fobj is not None:
             preds = self.objective.pred_transform(margin).cpu().numpy()
@@ -175,6 +192,9 @@ class Booster:
 
     def boost_gpair(self, dtrain: DMatrix, grad: torch.Tensor,
                     hess: torch.Tensor, iteration: int) -> None:
+        if self.tparam.tree_method == "exact":
+            self._boost_exact(dtrain, grad, hess, iteration)
+            return
         ops = self._ops_for(dtrain)
         n = dtrain.num_row()
         n_out = grad.shape[1] if grad.dim() > 1 else 1
@@ -185,7 +205,11 @@ class Booster:
                 else self.seed + iteration * 2654435761)
         new_trees = 0
         eta_scale = 1.0 / max(1, self.tparam.num_parallel_tree)
+        is_approx = self.tparam.tree_method == "approx"
         for k in range(n_out):
+            if is_approx:
+                ops = self._ops_for(
+                    dtrain, hess=hess[:, k].detach().cpu().numpy())
             for ptree in range(self.tparam.num_parallel_tree):
                 gpair = torch.stack([grad[:, k], hess[:, k]], dim=1).contiguous()
                 gpair = self._subsample(gpair, seed + 7919 * ptree + 104729 * k)
@@ -214,6 +238,74 @@ class Booster:
         if self.tparam.debug_synchronize:
             collective.check_synchronized(
                 json.dumps(self.trees[-1].to_json()).encode(), "tree")
+
+    def _update_existing(self, dtrain: DMatrix, iteration: int) -> None:
+        """process_type=update: run prune/refresh updaters over the trees
+        of `iteration` instead of growing new ones (reference gbtree.cc
+        process_type handling)."""
+        from .updaters import prune_tree, refresh_tree
+        if iteration >= self.num_boosted_rounds():
+            raise ValueError(
+                "process_type=update requires an existing model with at "
+                f"least {iteration + 1} boosted rounds")
+        updater_seq = (self.tparam.updater or "refresh").split(",")
+        margin = self._cached_margin(dtrain)
+        grad, hess = self.objective.get_gradient(margin, dtrain.info,
+                                                 iteration)
+        lo, hi = (self.iteration_indptr[iteration],
+                  self.iteration_indptr[iteration + 1])
+        X = dtrain.raw_data()
+        for t in range(lo, hi):
+            k = self.tree_info[t]
+            gpair = torch.stack([grad[:, k], hess[:, k]],
+                                dim=1).cpu().numpy()
+            for upd in updater_seq:
+                upd = upd.strip()
+                if upd == "prune":
+                    self.trees[t] = prune_tree(self.trees[t], self.tparam)
+                elif upd == "refresh":
+                    refresh_tree(self.trees[t], X, gpair, self.tparam,
+                                 dtrain.missing,
+                                 refresh_leaf=self.tparam.refresh_leaf)
+                else:
+                    raise ValueError(
+                        f"unsupported updater for process_type=update: {upd}")
+        self._cache.clear()  # leaf values changed; rebuild margins lazily
+
+    def _boost_exact(self, dtrain: DMatrix, grad: torch.Tensor,
+                     hess: torch.Tensor, iteration: int) -> None:
+        """tree_method=exact (reference ColMaker; CPU only like the
+        reference — src/tree/updater_colmaker.cc)."""
+        from .exact import grow_exact
+        if self.device.type == "cuda":
+            raise ValueError("tree_method=exact is CPU-only; use hist on GPU")
+        n = dtrain.num_row()
+        n_out = grad.shape[1]
+        margin, _ = self._cache[id(dtrain)]
+        seed = self.seed + iteration
+        cache = self.__dict__.setdefault("_exact_cache", {}).setdefault(
+            id(dtrain), {})
+        new_trees = 0
+        for k in range(n_out):
+            for ptree in range(self.tparam.num_parallel_tree):
+                gpair = torch.stack([grad[:, k], hess[:, k]], dim=1)
+                gpair = self._subsample(gpair, seed + 7919 * ptree + 11 * k)
+                tree = RegTree(self.n_features)
+                positions = grow_exact(dtrain.raw_data(),
+                                       gpair.cpu().numpy(),
+                                       self._scaled_param(
+                                           1.0 / max(1, self.tparam.num_parallel_tree)),
+                                       tree, dtrain.missing, cache)
+                self.trees.append(tree)
+                self.tree_info.append(k)
+                new_trees += 1
+                leaf_vals = torch.as_tensor(
+                    tree.split_cond[:tree.n_nodes].copy(), device=margin.device)
+                pos_t = torch.as_tensor(positions.astype(np.int64),
+                                        device=margin.device)
+                margin[:, k] += leaf_vals[pos_t]
+        self.iteration_indptr.append(self.iteration_indptr[-1] + new_trees)
+        self._cache[id(dtrain)] = (margin, len(self.trees))
 
     def _scaled_param(self, eta_scale: float) -> TrainParam:
         if eta_scale == 1.0:

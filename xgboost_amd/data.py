@@ -138,6 +138,16 @@ class DMatrix:
                  nthread: Optional[int] = None, enable_categorical: bool = False,
                  silent: bool = False):
         self.missing = float("nan") if missing is None else float(missing)
+        if isinstance(data, str) or hasattr(data, "__fspath__"):
+            # text file (libsvm format; deprecated upstream, data.cc:930)
+            import os as _os
+            from .libsvm import load_svmlight
+            path = _os.fspath(data).split("?", 1)[0]
+            data, file_labels, file_qid = load_svmlight(path)
+            if label is None:
+                label = file_labels
+            if qid is None and file_qid is not None:
+                qid = file_qid
         X, inferred_names, inferred_types = _ingest(data, enable_categorical)
         self._data = X  # np.float32 [n, f]
         self.info = MetaInfo(num_row=X.shape[0], num_col=X.shape[1])

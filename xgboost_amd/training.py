@@ -48,7 +48,11 @@ def train(params: Dict[str, Any], dtrain: DMatrix,
 
     cb = CallbackContainer(callbacks, metric=metric_fn)
     bst = cb.before_training(bst)
-    start = bst.num_boosted_rounds()
+    if params.get("process_type") == "update":
+        start = 0  # updaters revisit existing iterations
+        num_boost_round = min(num_boost_round, bst.num_boosted_rounds())
+    else:
+        start = bst.num_boosted_rounds()
     for i in range(start, start + num_boost_round):
         if cb.before_iteration(bst, i, dtrain, evals):
             break
