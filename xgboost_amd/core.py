@@ -52,8 +52,8 @@ class Booster:
         booster_kind = self.raw_params.get("booster", "gbtree")
         if booster_kind == "dart":
             booster_kind = "gbtree"  # DART deprecated -> gbtree (learner.cc:225)
-        if booster_kind == "gblinear":
-            raise NotImplementedError("gblinear is not implemented yet")
+        self.booster_kind = booster_kind
+        self._linear = None  # GBLinearModel when booster_kind == "gblinear"
         self.tparam: TrainParam = make_train_param(self.raw_params)
         self.device = _resolve_device(self.raw_params)
         obj_name = str(self.raw_params.get("objective", "reg:squarederror"))
@@ -192,6 +192,9 @@ class Booster:
 
     def boost_gpair(self, dtrain: DMatrix, grad: torch.Tensor,
                     hess: torch.Tensor, iteration: int) -> None:
+        if self.booster_kind == "gblinear":
+            self._boost_linear(dtrain, grad, hess, iteration)
+            return
         if self.tparam.tree_method == "exact":
             self._boost_exact(dtrain, grad, hess, iteration)
             return
@@ -238,6 +241,19 @@ class Booster:
         if self.tparam.debug_synchronize:
             collective.check_synchronized(
                 json.dumps(self.trees[-1].to_json()).encode(), "tree")
+
+    def _boost_linear(self, dtrain: DMatrix, grad: torch.Tensor,
+                      hess: torch.Tensor, iteration: int) -> None:
+        from .linear import GBLinearModel
+        if self._linear is None:
+            self._linear = GBLinearModel(self.n_features, self.n_outputs,
+                                         self.raw_params, self.device)
+        X = torch.as_tensor(dtrain.raw_data(), device=self.device)
+        siw = float(np.sum(dtrain.get_weight()))
+        self._linear.update(X, grad, hess, iteration, siw)
+        margin = self._linear.predict_margin(X) + self._base_margin_value()
+        self.iteration_indptr.append(self.iteration_indptr[-1])
+        self._cache[id(dtrain)] = (margin, len(self.trees))
 
     def _update_existing(self, dtrain: DMatrix, iteration: int) -> None:
         """process_type=update: run prune/refresh updaters over the trees
@@ -373,6 +389,11 @@ class Booster:
             bm = torch.as_tensor(dmat.info.base_margin, dtype=torch.float32,
                                  device=self.device)
             out = bm.view(n, -1).expand(n, self.n_outputs).clone()
+        if self.booster_kind == "gblinear":
+            if self._linear is not None:
+                X = torch.as_tensor(dmat.raw_data(), device=self.device)
+                out = out + self._linear.predict_margin(X)
+            return out
         lo, hi = self._tree_range(iteration_range)
         X = dmat.raw_data()
         if self.device.type == "cuda" and (hi - lo) > 0:
@@ -532,11 +553,10 @@ class Booster:
         self._model_from_json(j)
 
     def _model_to_json(self) -> dict:
-        learner = {
-            "attributes": dict(self.attributes_),
-            "feature_names": self.feature_names or [],
-            "feature_types": self.feature_types or [],
-            "gradient_booster": {
+        if self.booster_kind == "gblinear" and self._linear is not None:
+            gb = self._linear.to_json()
+        else:
+            gb = {
                 "model": {
                     "gbtree_model_param": {
                         "num_trees": str(len(self.trees)),
@@ -547,7 +567,12 @@ class Booster:
                     "trees": [t.to_json(i) for i, t in enumerate(self.trees)],
                 },
                 "name": "gbtree",
-            },
+            }
+        learner = {
+            "attributes": dict(self.attributes_),
+            "feature_names": self.feature_names or [],
+            "feature_types": self.feature_types or [],
+            "gradient_booster": gb,
             "learner_model_param": {
                 "base_score": f"{self.base_score if self.base_score is not None else 0.5:.9E}",
                 "boost_from_average": "1",
@@ -580,6 +605,20 @@ class Booster:
         self.objective = create_objective(obj_name, obj_params)
         self.raw_params["objective"] = obj_name
         gb = learner["gradient_booster"]
+        if gb.get("name") == "gblinear":
+            from .linear import GBLinearModel
+            self.booster_kind = "gblinear"
+            self._linear = GBLinearModel.from_json(
+                gb, self.n_features, max(1, num_class, self.n_targets),
+                self.raw_params, self.device)
+            self.trees = []
+            self.tree_info = []
+            self.iteration_indptr = [0]
+            self.attributes_ = {k: str(v) for k, v in
+                                learner.get("attributes", {}).items()}
+            self._cache.clear()
+            self._ops_cache.clear()
+            return
         model = gb["model"]
         self.trees = [RegTree.from_json(t) for t in model["trees"]]
         self.tree_info = [int(x) for x in model["tree_info"]]
