@@ -58,7 +58,42 @@ class GradQuantizer:
         return h
 
 
-class CpuOps:
+class SegmentedOpsMixin:
+    """Stateful node-segment bookkeeping shared by the in-core backends.
+
+    The grower talks to this interface only (reset / build_hist_nodes /
+    partition_nodes / leaf_positions / node_size); the external-memory
+    backend re-implements it with per-page state."""
+
+    def reset(self, n_rows: int) -> None:
+        self.ridx = self.make_ridx(n_rows)
+        self.segments = {0: (0, n_rows)}
+        self._n_rows = n_rows
+
+    def node_size(self, nid: int) -> int:
+        s, e = self.segments[nid]
+        return e - s
+
+    def build_hist_nodes(self, qgpair: torch.Tensor, nids) -> torch.Tensor:
+        return self.build_hist(qgpair, self.ridx,
+                               [self.segments[n] for n in nids])
+
+    def partition_nodes(self, parents, splits, children) -> None:
+        """parents: [nid], splits: [SplitEntry], children: [(l, r)].
+        Updates internal segments for the child nodes."""
+        segs = [self.segments[p] for p in parents]
+        new_segs = self.partition(self.ridx, segs, splits)
+        for (l, r), (ls, rs) in zip(children, new_segs):
+            self.segments[l] = ls
+            self.segments[r] = rs
+
+    def leaf_positions(self, leaf_nids) -> torch.Tensor:
+        segs = [(nid, *self.segments[nid]) for nid in leaf_nids
+                if nid in self.segments]
+        return self.leaf_partition(self.ridx, segs, self._n_rows)
+
+
+class CpuOps(SegmentedOpsMixin):
     """Tree-building primitive ops on CPU tensors."""
 
     device = torch.device("cpu")

@@ -106,7 +106,10 @@ def _np_to_torch(dt: np.dtype) -> torch.dtype:
             np.dtype(np.float32): torch.float32}[dt]
 
 
-class GpuOps:
+from .cpu import SegmentedOpsMixin
+
+
+class GpuOps(SegmentedOpsMixin):
     def __init__(self, qm: QuantizedMatrix):
         from .. import ops as hip_ops
         if not torch.cuda.is_available():
@@ -157,6 +160,14 @@ class GpuOps:
         self.stager = _PinnedStager(dev)
 
     # ------------------------------------------------------------------
+    def swap_gidx(self, gidx: torch.Tensor) -> None:
+        """Point the kernels at a different (streamed-in) quantized page
+        with the same cuts (external-memory path)."""
+        g8 = gidx.dtype == torch.uint8
+        self._gidx8 = gidx if g8 else None
+        self._gidx16 = None if g8 else gidx
+        self.qm = QuantizedMatrix(gidx, self.qm.cuts, self.qm.has_missing)
+
     def _gidx_ptrs(self):
         return self.hip.ptr(self._gidx8), self.hip.ptr(self._gidx16)
 

@@ -145,6 +145,11 @@ class Booster:
         key = id(dmat)
         ops = self._ops_cache.get(key)
         if ops is None:
+            from .extmem import ExtMemOps, ExtMemQuantileDMatrix
+            if isinstance(dmat, ExtMemQuantileDMatrix):
+                ops = ExtMemOps(dmat, self.device)
+                self._ops_cache[key] = ops
+                return ops
             from .sketch import sketch_cuts
             max_bin = self.tparam.max_bin
             qm = dmat.quantized(max_bin, sketch_fn=sketch_cuts)
@@ -395,6 +400,9 @@ class Booster:
                 out = out + self._linear.predict_margin(X)
             return out
         lo, hi = self._tree_range(iteration_range)
+        from .extmem import ExtMemQuantileDMatrix
+        if isinstance(dmat, ExtMemQuantileDMatrix):
+            return self._predict_margin_extmem(dmat, out, lo, hi)
         X = dmat.raw_data()
         if self.device.type == "cuda" and (hi - lo) > 0:
             from .backend.gpu import predict_margin_gpu
@@ -404,6 +412,20 @@ class Booster:
             pos = tree.predict_leaf_np(X, dmat.missing)
             vals = tree.split_cond[:tree.n_nodes][pos]
             out[:, self.tree_info[t]] += torch.as_tensor(vals, device=out.device)
+        return out
+
+    def _predict_margin_extmem(self, dmat, out: torch.Tensor,
+                               lo: int, hi: int) -> torch.Tensor:
+        """Bin-based traversal per quantized page (no raw values)."""
+        for pi, qm in enumerate(dmat.pages):
+            s, e = dmat.page_offsets[pi], dmat.page_offsets[pi + 1]
+            gg = qm.global_gidx().cpu().numpy()
+            for t in range(lo, hi):
+                tree = self.trees[t]
+                pos = tree.predict_leaf_bins(gg, dmat.cuts)
+                vals = tree.split_cond[:tree.n_nodes][pos]
+                out[s:e, self.tree_info[t]] += torch.as_tensor(
+                    vals, device=out.device)
         return out
 
     def _tree_range(self, iteration_range: Tuple[int, int]) -> Tuple[int, int]:

@@ -293,6 +293,23 @@ class QuantileDMatrix(DMatrix):
 
     def __init__(self, data: Any, label: Any = None, *, max_bin: int = 256,
                  ref: Optional[DMatrix] = None, **kwargs):
+        from .extmem import DataIter, _drive
+        if isinstance(data, DataIter):
+            # in-core QuantileDMatrix from an iterator: concatenate
+            # batches (reference IterativeDMatrix two-pass; with 288 GB
+            # HBM in-core concat is the right default)
+            Xs, ys, ws = [], [], []
+            for batch in _drive(data):
+                Xs.append(np.ascontiguousarray(batch["data"], np.float32))
+                if batch.get("label") is not None:
+                    ys.append(np.asarray(batch["label"], np.float32))
+                if batch.get("weight") is not None:
+                    ws.append(np.asarray(batch["weight"], np.float32))
+            data = np.concatenate(Xs)
+            if label is None and ys:
+                label = np.concatenate(ys)
+            if ws and "weight" not in kwargs:
+                kwargs["weight"] = np.concatenate(ws)
         super().__init__(data, label, **kwargs)
         self.max_bin = max_bin
         if ref is not None:

@@ -1,0 +1,289 @@
+"""External-memory training: DataIter, ExtMemQuantileDMatrix, paged ops.
+
+Reference behavior: src/data/extmem_quantile_dmatrix.{h,cu} (two passes
+over the user iterator: sketch then per-batch quantized pages cached in
+pinned host memory), src/data/sparse_page_source.h (prefetch ring),
+updater_gpu_hist.cu:371 (partition+hist per page pass).
+
+MI355X design: quantized pages live in pinned host memory; the GPU
+keeps a device-resident page cache sized to a HBM budget (288 GB/GPU
+makes most datasets fully cacheable — the reference's
+cache_host_ratio=0 case); pages beyond the budget are streamed with
+hipMemcpyAsync on torch's stream each sweep.
+"""
+from __future__ import annotations
+
+from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .data import DMatrix, MetaInfo, QuantizedMatrix, quantize_dense
+from .quantile import HistogramCuts
+from .sketch import cuts_from_summaries, sketch_cuts_batches, summarize_batch
+
+
+class DataIter:
+    """Base class for user batch iterators (reference:
+    python-package/xgboost/core.py DataIter)."""
+
+    def __init__(self, cache_prefix: Optional[str] = None,
+                 release_data: bool = True):
+        self.cache_prefix = cache_prefix
+        self._data_batches: List[Tuple] = []
+
+    def reset(self) -> None:
+        raise NotImplementedError
+
+    def next(self, input_data: Callable) -> bool:
+        """Call input_data(data=..., label=..., weight=...) and return
+        True while batches remain (xgboost>=2 convention)."""
+        raise NotImplementedError
+
+
+def _drive(it: DataIter):
+    """Iterate the DataIter once, yielding captured batch dicts."""
+    it.reset()
+    while True:
+        captured: Dict[str, Any] = {}
+
+        def input_data(**kwargs):
+            captured.update(kwargs)
+            return True
+
+        has_more = it.next(input_data)
+        if not has_more and not captured:
+            break
+        if captured:
+            yield captured
+        if not has_more:
+            break
+
+
+class ExtMemQuantileDMatrix(DMatrix):
+    """Quantized pages kept out-of-core (host memory), streamed during
+    training (reference: ExtMemQuantileDMatrix, extmem_quantile_dmatrix.h:29).
+    """
+
+    def __init__(self, data: DataIter, *, max_bin: int = 256,
+                 ref: Optional[DMatrix] = None, missing: float = np.nan,
+                 enable_categorical: bool = False,
+                 cache_host_ratio: Optional[float] = None,
+                 max_quantile_batches: Optional[int] = None,
+                 min_cache_page_bytes: Optional[int] = None,
+                 on_host: bool = True, nthread: Optional[int] = None):
+        # NOTE: deliberately does NOT call super().__init__ — no dense copy
+        self.missing = float("nan") if missing is None else float(missing)
+        self.max_bin = max_bin
+        self.cache_host_ratio = cache_host_ratio
+        self._quantized: Dict[int, QuantizedMatrix] = {}
+        self._ref_cuts: Optional[HistogramCuts] = None
+        self._data = None
+
+        labels, weights, margins, ftypes = [], [], [], None
+        # pass 1: sketch
+        if ref is not None:
+            cuts = ref.cached_cuts() or ref.quantized(max_bin).cuts
+        else:
+            summaries = []
+            n_features = None
+            for batch in _drive(data):
+                X = np.ascontiguousarray(batch["data"], dtype=np.float32)
+                n_features = X.shape[1]
+                ftypes = batch.get("feature_types", ftypes)
+                summaries.append(summarize_batch(
+                    X, self.missing, ftypes, max_bin,
+                    weights=batch.get("weight")))
+                if max_quantile_batches and \
+                        len(summaries) >= max_quantile_batches:
+                    pass  # summaries stay bounded per batch anyway
+            if n_features is None:
+                raise ValueError("DataIter yielded no batches")
+            cuts = sketch_cuts_batches(summaries, max_bin, n_features,
+                                       ftypes)
+        self.cuts = cuts
+        # pass 2: quantize pages
+        self.pages: List[QuantizedMatrix] = []
+        self.page_offsets = [0]
+        n_rows = 0
+        for batch in _drive(data):
+            X = np.ascontiguousarray(batch["data"], dtype=np.float32)
+            qm = quantize_dense(X, cuts, self.missing)
+            qm.gidx = qm.gidx.pin_memory() if torch.cuda.is_available() \
+                else qm.gidx
+            self.pages.append(qm)
+            n_rows += X.shape[0]
+            self.page_offsets.append(n_rows)
+            if batch.get("label") is not None:
+                labels.append(np.asarray(batch["label"], np.float32))
+            if batch.get("weight") is not None:
+                weights.append(np.asarray(batch["weight"], np.float32))
+            if batch.get("base_margin") is not None:
+                margins.append(np.asarray(batch["base_margin"], np.float32))
+        self.info = MetaInfo(num_row=n_rows,
+                             num_col=self.pages[0].n_features)
+        if labels:
+            self.info.labels = np.concatenate(labels)
+        if weights:
+            self.info.weights = np.concatenate(weights)
+        if margins:
+            self.info.base_margin = np.concatenate(margins)
+        self.info.feature_types = list(ftypes) if ftypes else None
+        self.info.feature_names = None
+        self.info.validate()
+
+    def num_row(self) -> int:
+        return self.info.num_row
+
+    def num_col(self) -> int:
+        return self.info.num_col
+
+    def raw_data(self):
+        raise RuntimeError(
+            "ExtMemQuantileDMatrix holds no raw feature values; "
+            "predict with the quantized pages (predict uses cut values)")
+
+    def cached_cuts(self) -> Optional[HistogramCuts]:
+        return self.cuts
+
+    def quantized(self, max_bin: int, sketch_fn=None):
+        raise RuntimeError("use make_extmem_ops for external-memory data")
+
+
+class ExtMemOps:
+    """Paged implementation of the grower ops interface: per-page row
+    index + segments; histograms accumulated across page sweeps."""
+
+    def __init__(self, dmat: ExtMemQuantileDMatrix, device: torch.device,
+                 device_cache_bytes: int = 200 << 30):
+        self.dmat = dmat
+        self.device = device
+        self.cuts = dmat.cuts
+        self.n_bins = self.cuts.total_bins
+        self.page_ops: List[Any] = []
+        budget = device_cache_bytes
+        for qm in dmat.pages:
+            if device.type == "cuda":
+                from .backend.gpu import GpuOps
+                size = qm.gidx.numel() * qm.gidx.element_size()
+                if size <= budget:
+                    budget -= size
+                    self.page_ops.append(GpuOps(qm.to(device)))
+                else:
+                    self.page_ops.append(_StreamedPage(qm, device))
+            else:
+                from .backend.cpu import CpuOps
+                self.page_ops.append(CpuOps(qm))
+        self.qm = dmat.pages[0]  # for cuts/n_features introspection
+
+    # -- stateful interface -------------------------------------------------
+    def reset(self, n_rows: int) -> None:
+        assert n_rows == self.dmat.num_row()
+        for ops, (s, e) in zip(self.page_ops, zip(
+                self.dmat.page_offsets[:-1], self.dmat.page_offsets[1:])):
+            ops.reset(e - s)
+        self._n_rows = n_rows
+
+    def node_size(self, nid: int) -> int:
+        return sum(ops.node_size(nid) for ops in self.page_ops)
+
+    def root_sum(self, qgpair: torch.Tensor) -> Tuple[int, int]:
+        from . import collective
+        s = qgpair.to(torch.int64).sum(dim=0)
+        collective.allreduce_sum_(s)
+        host = s.cpu()
+        return int(host[0]), int(host[1])
+
+    def _page_gpair(self, qgpair: torch.Tensor, i: int) -> torch.Tensor:
+        s, e = self.dmat.page_offsets[i], self.dmat.page_offsets[i + 1]
+        return qgpair[s:e]
+
+    def build_hist_nodes(self, qgpair: torch.Tensor, nids) -> torch.Tensor:
+        total = None
+        for i, ops in enumerate(self.page_ops):
+            h = ops.build_hist_nodes(self._page_gpair(qgpair, i), nids)
+            total = h if total is None else total + h
+        return total
+
+    def allreduce_hist(self, hist: torch.Tensor) -> torch.Tensor:
+        from . import collective
+        collective.allreduce_sum_(hist)
+        return hist
+
+    def evaluate_splits(self, *args, **kwargs):
+        return self.page_ops[0].evaluate_splits(*args, **kwargs)
+
+    def partition_nodes(self, parents, splits, children) -> None:
+        for ops in self.page_ops:
+            ops.partition_nodes(parents, splits, children)
+
+    def leaf_positions(self, leaf_nids) -> torch.Tensor:
+        out = torch.zeros(self._n_rows, dtype=torch.int32,
+                          device=self.device)
+        for i, ops in enumerate(self.page_ops):
+            s, e = self.dmat.page_offsets[i], self.dmat.page_offsets[i + 1]
+            out[s:e] = ops.leaf_positions(leaf_nids).to(self.device)
+        return out
+
+
+class _StreamedPage:
+    """A page whose quantized matrix stays in pinned host memory and is
+    copied to the device only for the op sweeps that read it
+    (beyond-HBM datasets).  The small per-page state (ridx, segments)
+    stays device-resident; only the big bin matrix streams."""
+
+    def __init__(self, qm: QuantizedMatrix, device: torch.device):
+        self.host_qm = qm
+        self.device = device
+        self._gpu_ops = None
+
+    def _ops_with_page(self):
+        from .backend.gpu import GpuOps
+        gidx = self.host_qm.gidx.to(self.device, non_blocking=True)
+        if self._gpu_ops is None:
+            self._gpu_ops = GpuOps(QuantizedMatrix(
+                gidx, self.host_qm.cuts, self.host_qm.has_missing))
+        else:
+            self._gpu_ops.swap_gidx(gidx)
+        return self._gpu_ops
+
+    def _drop_page(self) -> None:
+        if self._gpu_ops is not None:
+            self._gpu_ops.swap_gidx(
+                torch.zeros(0, dtype=self.host_qm.gidx.dtype,
+                            device=self.device).view(0, 1))
+
+    # ops that touch the bin matrix: stream the page in
+    def build_hist_nodes(self, qgpair, nids):
+        ops = self._ops_with_page()
+        out = ops.build_hist_nodes(qgpair, nids)
+        self._drop_page()
+        return out
+
+    def partition_nodes(self, parents, splits, children):
+        ops = self._ops_with_page()
+        out = ops.partition_nodes(parents, splits, children)
+        self._drop_page()
+        return out
+
+    # ops on small state only
+    def reset(self, n_rows):
+        return self._ops_with_page().reset(n_rows) if self._gpu_ops is None \
+            else self._gpu_ops.reset(n_rows)
+
+    def node_size(self, nid):
+        return self._gpu_ops.node_size(nid)
+
+    def leaf_positions(self, leaf_nids):
+        return self._gpu_ops.leaf_positions(leaf_nids)
+
+    def evaluate_splits(self, *a, **k):
+        return self._gpu_ops.evaluate_splits(*a, **k)
+
+    @property
+    def qm(self):
+        return self.host_qm
+
+
+def make_extmem_ops(dmat: ExtMemQuantileDMatrix, device) -> ExtMemOps:
+    return ExtMemOps(dmat, torch.device(device))

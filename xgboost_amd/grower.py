@@ -123,8 +123,7 @@ class TreeGrower:
         """Returns (tree, leaf position per row int32)."""
         param = self.param
         ops = self.ops
-        ridx = ops.make_ridx(self.n_rows)
-        segments: Dict[int, Tuple[int, int]] = {0: (0, self.n_rows)}
+        ops.reset(self.n_rows)
         node_sums: Dict[int, Tuple[int, int]] = {}  # exact int64 (gq, hq)
         node_bounds: Dict[int, Tuple[float, float]] = {0: (-np.inf, np.inf)}
         hists: Dict[int, torch.Tensor] = {}
@@ -137,7 +136,7 @@ class TreeGrower:
         root_w = float(calc_weight(root_g, root_h, param))
         tree.base_weight[0] = root_w
         tree.sum_hess[0] = root_h
-        hist = ops.build_hist(qgpair, ridx, [segments[0]])
+        hist = ops.build_hist_nodes(qgpair, [0])
         ops.allreduce_hist(hist)
         hists[0] = hist[0]
         root_entry = self._evaluate([0], node_sums, hists, node_bounds, depth=0)[0]
@@ -204,29 +203,27 @@ class TreeGrower:
                 n_leaves += 1
 
             # 2. partition rows
-            segs = [segments[b.nid] for b, _, _ in children]
+            parents = [b.nid for b, _, _ in children]
             splits = [b.split for b, _, _ in children]
-            new_segs = ops.partition(ridx, segs, splits)
-            for (b, l, r), (ls, rs) in zip(children, new_segs):
-                segments[l] = ls
-                segments[r] = rs
+            child_pairs = [(l, r) for _, l, r in children]
+            ops.partition_nodes(parents, splits, child_pairs)
 
             # 3. decide build vs subtract (smaller child built,
             #    reference AssignNodes updater_gpu_hist.cuh:62)
-            build_nodes, subtract_nodes = [], []
-            for (b, l, r), (ls, rs) in zip(children, new_segs):
+            build_nodes = []
+            for (b, l, r) in children:
                 expand_more = (b.depth + 1 < (param.max_depth or 10 ** 9)
                                or param.grow_policy == "lossguide")
                 if not expand_more and param.max_depth > 0:
                     continue
-                if (ls[1] - ls[0]) <= (rs[1] - rs[0]):
+                if ops.node_size(l) <= ops.node_size(r):
                     build_nodes.append((l, b.nid, r))
                 else:
                     build_nodes.append((r, b.nid, l))
             # 4. build + allreduce + subtract
             if build_nodes:
-                bsegs = [segments[n] for n, _, _ in build_nodes]
-                bh = ops.build_hist(qgpair, ridx, bsegs)
+                bh = ops.build_hist_nodes(qgpair,
+                                          [n for n, _, _ in build_nodes])
                 ops.allreduce_hist(bh)
                 # batched sibling subtraction: one kernel for the level
                 parent_stack = torch.stack(
@@ -249,15 +246,13 @@ class TreeGrower:
                     push(nid, depth, e)
 
         # finalize leaves
-        leaf_segments = []
+        leaf_nids = []
         for nid in range(tree.n_nodes):
             if tree.is_leaf(nid):
                 w = float(tree.base_weight[nid])
                 tree.set_leaf(nid, w * param.eta)
-                if nid in segments:
-                    s, e = segments[nid]
-                    leaf_segments.append((nid, s, e))
-        positions = ops.leaf_partition(ridx, leaf_segments, self.n_rows)
+                leaf_nids.append(nid)
+        positions = ops.leaf_positions(leaf_nids)
         return tree, positions
 
     # ------------------------------------------------------------------

@@ -1,70 +1,90 @@
-"""Distributed-aware sketch -> HistogramCuts.
+"""Sketch -> HistogramCuts: batch-incremental and distributed-aware.
 
 Reference behavior: src/common/quantile.cu:594 (SketchContainer::AllReduce
-merges per-worker GK summaries before MakeCuts).  Our distributed merge:
-each rank summarizes every feature into K weighted quantile points
-(K = 8*max_bin candidates, weight = local finite-count/K), the summaries
-are allgathered (small: n_features * K floats per rank) and the final
-cuts answer weighted rank queries on the pooled summary.  Single-process
-falls through to the exact sort-based cuts in quantile.py.
+merges per-worker GK summaries before MakeCuts); the same merge handles
+external-memory batches (reference: sketching per SparsePage chunk,
+hist_util.cu DeviceSketchWithHessian).
+
+Each data chunk (a batch from a DataIter, and/or a rank's shard) is
+summarized per feature into K weighted quantile points; summaries are
+pooled (allgathered across ranks when distributed) and the final cuts
+answer weighted rank queries on the pooled summary.  Single-process
+single-chunk input falls through to the exact sort-based cuts in
+quantile.py.
 """
 from __future__ import annotations
 
-from typing import Optional
+from typing import List, Optional, Sequence, Tuple
 
 import numpy as np
 
 from . import collective
 from .quantile import HistogramCuts, make_cuts, _categorical_cuts
 
-
-def sketch_cuts(dmat, max_bin: int) -> HistogramCuts:
-    if collective.get_world_size() <= 1:
-        return make_cuts(dmat.raw_data(), max_bin,
-                         feature_types=dmat.info.feature_types,
-                         missing=dmat.missing)
-    return _distributed_cuts(dmat, max_bin)
+# a per-feature summary: ("q", points, total_weight) | ("c", cats, count)
+Summary = Tuple[str, np.ndarray, float]
 
 
-def _distributed_cuts(dmat, max_bin: int) -> HistogramCuts:
-    X = dmat.raw_data()
-    missing = dmat.missing
-    ftypes = dmat.info.feature_types
-    n_features = X.shape[1]
+def summarize_batch(X: np.ndarray, missing: float,
+                    feature_types: Optional[List[str]], max_bin: int,
+                    weights: Optional[np.ndarray] = None
+                    ) -> List[Summary]:
     K = max(64, 8 * max_bin)
     qs = (np.arange(K) + 0.5) / K
-    summaries = []
-    for f in range(n_features):
+    out: List[Summary] = []
+    for f in range(X.shape[1]):
         col = X[:, f]
         if np.isnan(missing):
-            vals = col[~np.isnan(col)]
+            mask = ~np.isnan(col)
         else:
-            vals = col[(col != missing) & ~np.isnan(col)]
-        if ftypes is not None and ftypes[f] == "c":
-            cats = np.unique(vals).astype(np.float32)
-            summaries.append(("c", cats, float(vals.size)))
+            mask = (col != missing) & ~np.isnan(col)
+        vals = col[mask]
+        if feature_types is not None and feature_types[f] == "c":
+            out.append(("c", np.unique(vals).astype(np.float32),
+                        float(vals.size)))
         elif vals.size == 0:
-            summaries.append(("q", np.zeros(0, np.float32), 0.0))
+            out.append(("q", np.zeros(0, np.float32), 0.0))
         else:
-            pts = np.quantile(vals, qs).astype(np.float32)
+            if weights is not None:
+                w = weights[mask].astype(np.float64)
+                order = np.argsort(vals, kind="stable")
+                sv, sw = vals[order], w[order]
+                cw = np.cumsum(sw)
+                total = cw[-1]
+                targets = qs * total
+                pos = np.clip(np.searchsorted(cw, targets), 0, sv.size - 1)
+                pts = sv[pos].astype(np.float32)
+                wsum = float(total)
+            else:
+                pts = np.quantile(vals, qs).astype(np.float32)
+                wsum = float(vals.size)
             mn, mx = np.float32(vals.min()), np.float32(vals.max())
-            summaries.append(("q", np.concatenate([[mn], pts, [mx]]),
-                              float(vals.size)))
-    gathered = collective.allgather_obj(summaries)
+            out.append(("q", np.concatenate([[mn], pts, [mx]]), wsum))
+    return out
 
-    all_values, ptrs, min_vals = [], [0], np.zeros(n_features, np.float32)
+
+def cuts_from_summaries(chunks: Sequence[List[Summary]], max_bin: int,
+                        n_features: int,
+                        feature_types: Optional[List[str]]) -> HistogramCuts:
+    all_values, ptrs = [], [0]
+    min_vals = np.zeros(n_features, np.float32)
     for f in range(n_features):
-        kind = summaries[f][0]
+        kind = None
+        for ch in chunks:
+            if ch[f][2] > 0:
+                kind = ch[f][0]
+                break
         if kind == "c":
-            cats = np.unique(np.concatenate(
-                [g[f][1] for g in gathered if g[f][1].size]))
+            cats_list = [ch[f][1] for ch in chunks if ch[f][1].size]
+            cats = (np.unique(np.concatenate(cats_list))
+                    if cats_list else np.zeros(0, np.float32))
             cuts = (_categorical_cuts(cats) if cats.size
                     else np.array([0.0], np.float32))
             min_vals[f] = float(cats.min()) if cats.size else 0.0
         else:
             vals_list, w_list = [], []
-            for g in gathered:
-                pts, cnt = g[f][1], g[f][2]
+            for ch in chunks:
+                pts, cnt = ch[f][1], ch[f][2]
                 if cnt > 0 and pts.size:
                     vals_list.append(pts)
                     w_list.append(np.full(pts.size, cnt / pts.size))
@@ -87,7 +107,8 @@ def _distributed_cuts(dmat, max_bin: int) -> HistogramCuts:
                 else:
                     queries = np.arange(1, max_bin) * (total / max_bin)
                     first_idx = np.nonzero(distinct_mask)[0]
-                    last_idx = np.concatenate([first_idx[1:] - 1, [v.size - 1]])
+                    last_idx = np.concatenate([first_idx[1:] - 1,
+                                               [v.size - 1]])
                     rmax = cw[last_idx]
                     pos = np.clip(np.searchsorted(rmax, queries, "left"),
                                   0, distinct.size - 1)
@@ -104,4 +125,29 @@ def _distributed_cuts(dmat, max_bin: int) -> HistogramCuts:
         values=(np.concatenate(all_values) if all_values
                 else np.zeros(0, np.float32)),
         ptrs=np.asarray(ptrs, np.int64), min_vals=min_vals,
-        feature_types=list(ftypes) if ftypes else None)
+        feature_types=list(feature_types) if feature_types else None)
+
+
+def sketch_cuts(dmat, max_bin: int) -> HistogramCuts:
+    """Cuts for an in-core DMatrix; merges across ranks if distributed."""
+    if collective.get_world_size() <= 1:
+        return make_cuts(dmat.raw_data(), max_bin,
+                         feature_types=dmat.info.feature_types,
+                         missing=dmat.missing)
+    local = summarize_batch(dmat.raw_data(), dmat.missing,
+                            dmat.info.feature_types, max_bin)
+    gathered = collective.allgather_obj(local)
+    return cuts_from_summaries(gathered, max_bin, dmat.num_col(),
+                               dmat.info.feature_types)
+
+
+def sketch_cuts_batches(summaries: Sequence[List[Summary]], max_bin: int,
+                        n_features: int,
+                        feature_types: Optional[List[str]]) -> HistogramCuts:
+    """Cuts from per-batch summaries (external memory / DataIter);
+    distributed: each rank pools its batches, then ranks allgather."""
+    chunks = list(summaries)
+    if collective.get_world_size() > 1:
+        gathered = collective.allgather_obj(chunks)
+        chunks = [c for rank_chunks in gathered for c in rank_chunks]
+    return cuts_from_summaries(chunks, max_bin, n_features, feature_types)

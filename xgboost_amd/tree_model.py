@@ -156,6 +156,40 @@ class RegTree:
             active[idx] = self.left[pos[idx]] != -1
         return pos
 
+    def predict_leaf_bins(self, gidx_global: "np.ndarray",
+                          cuts) -> np.ndarray:
+        """Traversal over GLOBAL bin ids (external-memory predict where
+        raw values are gone): left iff bin <= split_bin; missing = -1."""
+        split_bin = np.zeros(self.n_nodes, dtype=np.int64)
+        for nid in range(self.n_nodes):
+            if not self.is_leaf(nid):
+                f = int(self.split_index[nid])
+                fc = cuts.feature_cuts(f)
+                b = int(np.searchsorted(fc, self.split_cond[nid], side="left"))
+                split_bin[nid] = cuts.ptrs[f] + b
+        n = gidx_global.shape[0]
+        pos = np.zeros(n, dtype=np.int32)
+        active = self.left[pos] != -1
+        while active.any():
+            idx = np.nonzero(active)[0]
+            nid = pos[idx]
+            feat = self.split_index[nid]
+            b = gidx_global[idx, feat]
+            missing = b < 0
+            go_left = np.where(missing, self.default_left[nid].astype(bool),
+                               b <= split_bin[nid])
+            cat_nodes = self.split_type[nid] == 1
+            if cat_nodes.any():
+                for k in np.nonzero(cat_nodes)[0]:
+                    if missing[k]:
+                        continue
+                    local = int(b[k] - cuts.ptrs[int(self.split_index[nid[k]])])
+                    cats = self.cat_segments[int(nid[k])]
+                    go_left[k] = local not in cats
+            pos[idx] = np.where(go_left, self.left[nid], self.right[nid])
+            active[idx] = self.left[pos[idx]] != -1
+        return pos
+
     # -- JSON schema ----------------------------------------------------------
     def to_json(self, tree_id: int = 0) -> dict:
         n = self.n_nodes
