@@ -252,3 +252,32 @@ def test_gpu_e2e_quality():
                      "eval_metric": "auc"}, d, 20, evals=[(d, "train")],
                     evals_result=res, verbose_eval=False)
     assert res["train"]["auc"][-1] > 0.9
+
+
+def test_gpu_shap_matches_cpu():
+    from xgboost_amd.shap import shap_values
+    X, y = _data(500, 8)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 5},
+                    d, 5, verbose_eval=False)
+    cpu_contribs = bst.predict(d, pred_contribs=True)
+    bst.set_param("device", "cuda")
+    bst.reset()
+    gpu_contribs = bst.predict(d, pred_contribs=True)
+    assert gpu_contribs.shape == cpu_contribs.shape
+    assert np.allclose(cpu_contribs, gpu_contribs, atol=1e-4), \
+        np.abs(cpu_contribs - gpu_contribs).max()
+    # efficiency: contribs sum to margin
+    margin = bst.predict(d, output_margin=True)
+    assert np.allclose(gpu_contribs.sum(axis=1), margin, atol=1e-3)
+
+
+def test_gpu_shap_with_missing():
+    from xgboost_amd.shap import shap_values
+    X, y = _data(300, 6, missing_frac=0.2)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 4,
+                     "device": "cuda"}, d, 4, verbose_eval=False)
+    contribs = bst.predict(d, pred_contribs=True)
+    margin = bst.predict(d, output_margin=True)
+    assert np.allclose(contribs.sum(axis=1), margin, atol=1e-3)

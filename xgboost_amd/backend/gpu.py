@@ -370,6 +370,7 @@ class _ForestArrays:
         cond = np.empty(total, np.float32)
         dft = np.empty(total, np.uint8)
         stype = np.empty(total, np.uint8)
+        hess = np.empty(total, np.float32)
         cat_off = np.zeros(total + 1, np.int32)
         cat_bits: List[np.ndarray] = []
         pos = 0
@@ -382,6 +383,7 @@ class _ForestArrays:
             cond[o:o + n] = t.split_cond[:n]
             dft[o:o + n] = t.default_left[:n]
             stype[o:o + n] = t.split_type[:n]
+            hess[o:o + n] = t.sum_hess[:n]
             for nid in range(n):
                 if t.split_type[nid] == 1 and nid in t.cat_segments:
                     cats = t.cat_segments[nid]
@@ -399,6 +401,7 @@ class _ForestArrays:
         self.split_cond = torch.from_numpy(cond).to(device)
         self.default_left = torch.from_numpy(dft).to(device)
         self.split_type = torch.from_numpy(stype).to(device)
+        self.sum_hess = torch.from_numpy(hess).to(device)
         self.cat_offsets = torch.from_numpy(cat_off).to(device)
         if cat_bits:
             self.cat_bits = torch.from_numpy(
@@ -409,6 +412,45 @@ class _ForestArrays:
             [booster.tree_info[t] for t in range(lo, hi)],
             dtype=torch.int32, device=device)
         self.n_trees = len(trees)
+
+
+def shap_gpu(booster, dmat, lo: int, hi: int, phi: np.ndarray) -> np.ndarray:
+    """GPU pred_contribs (path-dependent TreeSHAP kernel, shap.hip).
+    phi arrives [n, n_groups, f+1] float64 pre-filled with the base
+    margin in the bias column."""
+    from .. import ops as hip_ops
+    from ..shap import _expected_value
+    lib = hip_ops.load()
+    if not hasattr(lib, "gbt_shap"):
+        raise ImportError("gbt_shap kernel not built")
+    max_depth = max((t.max_depth() for t in booster.trees[lo:hi]), default=0)
+    n, n_groups, n_cols = phi.shape
+    if max_depth > 16 or n_cols > 129:
+        raise ImportError("GPU SHAP limits exceeded; falling back to CPU")
+    device = booster.device
+    fa = _ForestArrays(booster, lo, hi, device)
+    X = torch.from_numpy(dmat.raw_data()).to(device)
+    expected = torch.tensor(
+        [_expected_value(booster.trees[t]) for t in range(lo, hi)],
+        dtype=torch.float64, device=device)
+    out = torch.from_numpy(
+        np.ascontiguousarray(phi, np.float32)).to(device)
+    missing = dmat.missing
+    missing_is_nan = 1 if np.isnan(missing) else 0
+    lib.gbt_shap(
+        hip_ops.ptr(X), n, dmat.num_col(),
+        float(0.0 if missing_is_nan else missing), missing_is_nan,
+        hip_ops.ptr(fa.tree_offsets), hip_ops.ptr(fa.left),
+        hip_ops.ptr(fa.right), hip_ops.ptr(fa.split_index),
+        hip_ops.ptr(fa.split_cond), hip_ops.ptr(fa.default_left),
+        hip_ops.ptr(fa.split_type), hip_ops.ptr(fa.cat_offsets),
+        hip_ops.ptr(fa.cat_bits), hip_ops.ptr(fa.sum_hess),
+        hip_ops.ptr(fa.tree_group), fa.n_trees, n_groups, n_cols,
+        hip_ops.ptr(expected), hip_ops.ptr(out), hip_ops.stream())
+    res = out.cpu().numpy()
+    if n_groups == 1:
+        return res[:, 0, :]
+    return res
 
 
 def predict_margin_gpu(booster, dmat, out_margin: torch.Tensor,

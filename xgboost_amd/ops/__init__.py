@@ -39,7 +39,7 @@ _SIGS = {
     "gbt_copy_ranges": [_p, _p, _p, _i, _p],
     "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p],
     "gbt_shap": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p, _p,
-                 _p, _p, _i, _i, _i, _p, _p],
+                 _p, _p, _i, _i, _i, _p, _p, _p],
 }
 
 
