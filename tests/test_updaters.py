@@ -89,3 +89,39 @@ def test_refresh_leaf_false_keeps_leaves():
                       "refresh_leaf": False}, d, 3, xgb_model=bst,
                      verbose_eval=False)
     assert np.allclose(bst2.predict(d), p1, atol=1e-5)
+
+
+def test_multi_output_tree():
+    rng = np.random.RandomState(0)
+    X = rng.randn(1500, 6).astype(np.float32)
+    W = rng.randn(6, 3)
+    Y = (X @ W + 0.1 * rng.randn(1500, 3)).astype(np.float32)
+    d = xgb.DMatrix(X, label=Y)
+    res = {}
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 5,
+                     "multi_strategy": "multi_output_tree", "eta": 0.3},
+                    d, 15, evals=[(d, "t")], evals_result=res,
+                    verbose_eval=False)
+    assert len(bst.trees) == 15  # one vector-leaf tree per round
+    assert bst.trees[0].leaf_values is not None
+    p = bst.predict(d)
+    assert p.shape == (1500, 3)
+    assert res["t"]["rmse"][-1] < 0.6
+    # JSON round-trip with size_leaf_vector
+    raw = bst.save_raw("json")
+    b2 = xgb.Booster()
+    b2.load_model(bytes(raw))
+    assert np.allclose(b2.predict(d), p, atol=1e-6)
+
+
+def test_multi_target_one_output_per_tree():
+    rng = np.random.RandomState(0)
+    X = rng.randn(1000, 5).astype(np.float32)
+    W = rng.randn(5, 2)
+    Y = (X @ W).astype(np.float32)
+    d = xgb.DMatrix(X, label=Y)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 4},
+                    d, 10, verbose_eval=False)
+    assert len(bst.trees) == 20  # one tree per target per round
+    p = bst.predict(d)
+    assert p.shape == (1000, 2)

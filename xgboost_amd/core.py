@@ -213,6 +213,11 @@ class Booster:
                 else self.seed + iteration * 2654435761)
         new_trees = 0
         eta_scale = 1.0 / max(1, self.tparam.num_parallel_tree)
+        multi_strategy = str(self.raw_params.get("multi_strategy",
+                                                 "one_output_per_tree"))
+        if multi_strategy == "multi_output_tree" and n_out > 1:
+            self._boost_multi_target(dtrain, ops, grad, hess, margin, seed)
+            return
         is_approx = self.tparam.tree_method == "approx"
         for k in range(n_out):
             if is_approx:
@@ -258,6 +263,30 @@ class Booster:
         self._linear.update(X, grad, hess, iteration, siw)
         margin = self._linear.predict_margin(X) + self._base_margin_value()
         self.iteration_indptr.append(self.iteration_indptr[-1])
+        self._cache[id(dtrain)] = (margin, len(self.trees))
+
+    def _boost_multi_target(self, dtrain: DMatrix, ops, grad, hess,
+                            margin, seed: int) -> None:
+        """multi_strategy=multi_output_tree: one vector-leaf tree per
+        iteration (reference MultiTargetHistMaker)."""
+        from .grower import MultiTargetGrower
+        n, n_out = grad.shape
+        qgpairs, quantizers = [], []
+        for k in range(n_out):
+            gp = torch.stack([grad[:, k], hess[:, k]], dim=1).contiguous()
+            gp = gp.to(ops.device)
+            q = GradQuantizer(gp)
+            quantizers.append(q)
+            qgpairs.append(q.quantize(gp))
+        tree = RegTree(self.n_features, n_out)
+        grower = MultiTargetGrower(ops, self.tparam, quantizers, n, seed)
+        tree, positions = grower.grow(qgpairs, tree)
+        self.trees.append(tree)
+        self.tree_info.append(0)
+        self.iteration_indptr.append(self.iteration_indptr[-1] + 1)
+        leaf_vals = torch.as_tensor(
+            tree.leaf_values[:tree.n_nodes].copy(), device=margin.device)
+        margin += leaf_vals[positions.to(margin.device).long()]
         self._cache[id(dtrain)] = (margin, len(self.trees))
 
     def _update_existing(self, dtrain: DMatrix, iteration: int) -> None:
@@ -404,14 +433,20 @@ class Booster:
         if isinstance(dmat, ExtMemQuantileDMatrix):
             return self._predict_margin_extmem(dmat, out, lo, hi)
         X = dmat.raw_data()
-        if self.device.type == "cuda" and (hi - lo) > 0:
+        has_mt = any(t.leaf_values is not None for t in self.trees[lo:hi])
+        if self.device.type == "cuda" and (hi - lo) > 0 and not has_mt:
             from .backend.gpu import predict_margin_gpu
             return predict_margin_gpu(self, dmat, out, lo, hi)
         for t in range(lo, hi):
             tree = self.trees[t]
             pos = tree.predict_leaf_np(X, dmat.missing)
-            vals = tree.split_cond[:tree.n_nodes][pos]
-            out[:, self.tree_info[t]] += torch.as_tensor(vals, device=out.device)
+            if tree.leaf_values is not None:
+                out += torch.as_tensor(tree.leaf_values[:tree.n_nodes][pos],
+                                       device=out.device)
+            else:
+                vals = tree.split_cond[:tree.n_nodes][pos]
+                out[:, self.tree_info[t]] += torch.as_tensor(
+                    vals, device=out.device)
         return out
 
     def _predict_margin_extmem(self, dmat, out: torch.Tensor,

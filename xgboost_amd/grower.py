@@ -314,3 +314,146 @@ class TreeGrower:
         else:
             node_bounds[l] = (max(lo, mid), hi)
             node_bounds[r] = (lo, min(hi, mid))
+
+
+class MultiTargetGrower:
+    """Vector-leaf tree grower: ONE tree structure, per-target leaf
+    values (reference MultiTargetHistMaker, updater_gpu_hist.cuh:109;
+    target-major histograms histogram.cu:256; leaf sums leaf_sum.cuh).
+
+    Histograms are built per target over the same row partition; split
+    gain is the sum of per-target gains (multi_evaluate_splits.cu)."""
+
+    def __init__(self, ops, param: TrainParam, quantizers, n_rows: int,
+                 seed: int = 0):
+        self.ops = ops
+        self.param = param
+        self.quantizers = quantizers  # one GradQuantizer per target
+        self.n_rows = n_rows
+        self.col_sampler = ColumnSampler(ops.qm.n_features, param, seed)
+
+    def grow(self, qgpairs, tree: RegTree):
+        from .splits import evaluate_splits_multi_np
+        param = self.param
+        ops = self.ops
+        T = len(qgpairs)
+        ops.reset(self.n_rows)
+        g_scales = np.array([q.g_scale for q in self.quantizers])
+        h_scales = np.array([q.h_scale for q in self.quantizers])
+
+        node_sums: Dict[int, np.ndarray] = {}  # nid -> int64 [T, 2]
+        hists: Dict[int, torch.Tensor] = {}    # nid -> [T, bins, 2]
+
+        root = np.stack([np.asarray(ops.root_sum(qgpairs[t]), np.int64)
+                         for t in range(T)])
+        node_sums[0] = root
+        w0 = calc_weight(root[:, 0] / g_scales, root[:, 1] / h_scales, param)
+        tree.base_weight[0] = float(np.mean(w0))
+        tree.sum_hess[0] = float((root[:, 1] / h_scales).sum())
+
+        hist0 = torch.stack([ops.build_hist_nodes(qgpairs[t], [0])[0]
+                             for t in range(T)])
+        ops.allreduce_hist(hist0)
+        hists[0] = hist0
+
+        def evaluate(nids, depth):
+            h = torch.stack([hists[n] for n in nids], dim=1)  # [T,k,b,2]
+            parents = np.stack([node_sums[n] for n in nids])  # [k,T,2]
+            fs = None
+            node_feats = self.col_sampler.node_set(depth)
+            if node_feats is not None:
+                fs = [node_feats for _ in nids]
+            return evaluate_splits_multi_np(
+                h.cpu().numpy(), parents, g_scales, h_scales, nids,
+                ops.qm.cuts.ptrs, param, feature_sets=fs)
+
+        root_entry = evaluate([0], 0)[0]
+        n_leaves = 1
+        seq = 0
+        heap: List[_QueueEntry] = []
+
+        def push(nid, depth, split):
+            nonlocal seq
+            if not split.is_valid or split.gain <= param.gamma:
+                return
+            if param.max_depth > 0 and depth >= param.max_depth:
+                return
+            if param.max_leaves > 0 and n_leaves >= param.max_leaves:
+                return
+            key = (float(depth) if param.grow_policy == "depthwise"
+                   else -split.gain)
+            heapq.heappush(heap, _QueueEntry(key, seq, nid, depth, split))
+            seq += 1
+
+        push(0, 0, root_entry)
+        while heap:
+            batch = [heapq.heappop(heap)]
+            if param.grow_policy == "depthwise":
+                while heap and heap[0].sort_key == batch[0].sort_key:
+                    batch.append(heapq.heappop(heap))
+            if param.max_leaves > 0:
+                batch = batch[:max(0, param.max_leaves - n_leaves)]
+            if not batch:
+                continue
+            children = []
+            for b in batch:
+                sp = b.split
+                cond = float(ops.qm.cuts.values[sp.split_bin])
+                wl = calc_weight(sp.left_q[:, 0] / g_scales,
+                                 sp.left_q[:, 1] / h_scales, param)
+                wr = calc_weight(sp.right_q[:, 0] / g_scales,
+                                 sp.right_q[:, 1] / h_scales, param)
+                lh = float((sp.left_q[:, 1] / h_scales).sum())
+                rh = float((sp.right_q[:, 1] / h_scales).sum())
+                l, r = tree.add_split(
+                    b.nid, sp.feature, cond, sp.default_left, sp.gain,
+                    float(tree.base_weight[b.nid]),
+                    float(np.mean(wl)), float(np.mean(wr)),
+                    lh + rh, lh, rh)
+                tree.leaf_values[l] = wl.astype(np.float32)
+                tree.leaf_values[r] = wr.astype(np.float32)
+                node_sums[l] = sp.left_q
+                node_sums[r] = sp.right_q
+                children.append((b, l, r))
+                n_leaves += 1
+            parents = [b.nid for b, _, _ in children]
+            from .splits import SplitEntry
+            scalar_splits = [SplitEntry(
+                nid=b.nid, feature=b.split.feature,
+                split_bin=b.split.split_bin,
+                default_left=b.split.default_left) for b, _, _ in children]
+            ops.partition_nodes(parents, scalar_splits,
+                                [(l, r) for _, l, r in children])
+            build_nodes = []
+            for (b, l, r) in children:
+                if param.max_depth > 0 and b.depth + 1 >= param.max_depth \
+                        and param.grow_policy == "depthwise":
+                    continue
+                if ops.node_size(l) <= ops.node_size(r):
+                    build_nodes.append((l, b.nid, r))
+                else:
+                    build_nodes.append((r, b.nid, l))
+            if build_nodes:
+                bh = torch.stack([
+                    ops.build_hist_nodes(qgpairs[t],
+                                         [n for n, _, _ in build_nodes])
+                    for t in range(len(qgpairs))])  # [T, k, bins, 2]
+                ops.allreduce_hist(bh)
+                for i, (n, parent, sib) in enumerate(build_nodes):
+                    hists[n] = bh[:, i]
+                    hists[sib] = hists[parent] - bh[:, i]
+                    del hists[parent]
+                eval_nids = ([n for n, _, _ in build_nodes]
+                             + [s for _, _, s in build_nodes])
+                depth = children[0][0].depth + 1
+                for nid, e in zip(eval_nids, evaluate(eval_nids, depth)):
+                    push(nid, depth, e)
+
+        leaf_nids = []
+        for nid in range(tree.n_nodes):
+            if tree.is_leaf(nid):
+                tree.leaf_values[nid] = tree.leaf_values[nid] * param.eta
+                tree.set_leaf(nid, tree.leaf_values[nid])
+                leaf_nids.append(nid)
+        positions = ops.leaf_positions(leaf_nids)
+        return tree, positions

@@ -77,13 +77,18 @@ class Objective:
         return base_score
 
     def init_estimation(self, info) -> float:
-        """One Newton step at margin 0 (reference FitStump + PredTransform)."""
+        """One Newton step at margin 0 (reference FitStump + PredTransform).
+        Multi-target labels produce the mean of per-target estimates
+        (scalar base_score, like the reference's ParamArray mean)."""
         n = info.num_row
-        preds = torch.zeros((n, 1), dtype=torch.float32)
+        n_out = 1
+        if info.labels is not None and info.labels.ndim == 2:
+            n_out = info.labels.shape[1]
+        preds = torch.zeros((n, n_out), dtype=torch.float32)
         g, h = self.get_gradient(preds, info, 0)
-        gs = float(g.double().sum())
-        hs = float(h.double().sum())
-        margin = -gs / max(hs, 1e-16)
+        gs = g.double().sum(dim=0)
+        hs = h.double().sum(dim=0).clamp(min=1e-16)
+        margin = float((-gs / hs).mean())
         out = self.pred_transform(torch.tensor([margin])).item()
         return out
 

@@ -224,3 +224,104 @@ def _onehot_cat_gains(gain, Gq, Hq, pgq, phq, missGq, missHq,
                | ((mono_bins < 0) & (wl >= wr)))
     g2 = np.where(ok, g2, -np.inf)
     return np.where(cat_bins[None, :], g2, gain), (glq, hlq)
+
+
+def evaluate_splits_multi_np(hists: np.ndarray,
+                             parents_q: np.ndarray,
+                             g_scales: np.ndarray, h_scales: np.ndarray,
+                             nids: Sequence[int], cut_ptrs: np.ndarray,
+                             param: TrainParam,
+                             feature_sets: Optional[List[np.ndarray]] = None,
+                             ) -> List["MultiSplitEntry"]:
+    """Vector-leaf split evaluation (reference MultiHistEvaluator,
+    src/tree/gpu_hist/multi_evaluate_splits.cuh): gain per bin = sum of
+    per-target gains; one shared structure, per-target child sums.
+
+    hists: int64 [T, n_nodes, n_bins, 2]; parents_q: int64 [n_nodes, T, 2].
+    """
+    T, n_nodes, n_bins, _ = hists.shape
+    n_features = len(cut_ptrs) - 1
+    widths = np.diff(cut_ptrs)
+    feat_of_bin = np.repeat(np.arange(n_features), widths)
+    seg_start = np.repeat(cut_ptrs[:-1], widths)
+    seg_end = np.repeat(cut_ptrs[1:] - 1, widths)
+    is_last_bin = np.arange(n_bins) == seg_end
+
+    inv_g = (1.0 / g_scales).reshape(T, 1, 1)
+    inv_h = (1.0 / h_scales).reshape(T, 1, 1)
+
+    Gq = hists[:, :, :, 0]
+    Hq = hists[:, :, :, 1]
+    cumG = np.cumsum(Gq, axis=2)
+    cumH = np.cumsum(Hq, axis=2)
+    baseG = np.where(seg_start > 0, cumG[:, :, np.maximum(seg_start - 1, 0)], 0)
+    baseH = np.where(seg_start > 0, cumH[:, :, np.maximum(seg_start - 1, 0)], 0)
+    GLq = cumG - baseG
+    HLq = cumH - baseH
+    featGq = cumG[:, :, seg_end] - baseG
+    featHq = cumH[:, :, seg_end] - baseH
+    pq = parents_q.transpose(1, 0, 2)       # [T, n_nodes, 2]
+    pgq = pq[:, :, 0][:, :, None]
+    phq = pq[:, :, 1][:, :, None]
+    missGq = pgq - featGq
+    missHq = phq - featHq
+
+    pg = pgq * inv_g
+    ph = phq * inv_h
+    parent_gain = calc_gain(pg, ph, param).sum(axis=0)  # [n_nodes, 1]
+
+    best = [MultiSplitEntry(nid=int(nid), n_targets=T) for nid in nids]
+    for missing_left in (False, True):
+        glq = GLq + (missGq if missing_left else 0)
+        hlq = HLq + (missHq if missing_left else 0)
+        grq = pgq - glq
+        hrq = phq - hlq
+        gl = glq * inv_g
+        hl = hlq * inv_h
+        gr = grq * inv_g
+        hr = hrq * inv_h
+        wl = calc_weight(gl, hl, param)
+        wr = calc_weight(gr, hr, param)
+        gain = (calc_gain_given_weight(gl, hl, wl, param)
+                + calc_gain_given_weight(gr, hr, wr, param)).sum(axis=0) \
+            - parent_gain
+        ok = ((hl.sum(axis=0) >= param.min_child_weight)
+              & (hr.sum(axis=0) >= param.min_child_weight)
+              & ~is_last_bin[None, :])
+        gain = np.where(ok, gain, -np.inf)
+        for i in range(n_nodes):
+            row = gain[i]
+            if feature_sets is not None and feature_sets[i] is not None:
+                mask = np.zeros(n_bins, dtype=bool)
+                for f in feature_sets[i]:
+                    mask[cut_ptrs[f]:cut_ptrs[f + 1]] = True
+                row = np.where(mask, row, -np.inf)
+            b = int(np.argmax(row))
+            gval = float(row[b])
+            e = best[i]
+            if gval > e.gain and np.isfinite(gval):
+                e.gain = gval
+                e.feature = int(feat_of_bin[b])
+                e.split_bin = b
+                e.default_left = missing_left
+                e.left_q = np.stack([glq[:, i, b], hlq[:, i, b]], axis=1)
+                e.right_q = parents_q[i] - e.left_q
+    return best
+
+
+@dataclasses.dataclass
+class MultiSplitEntry:
+    nid: int
+    n_targets: int = 1
+    gain: float = -np.inf
+    feature: int = -1
+    split_bin: int = -1
+    default_left: bool = False
+    left_q: Optional[np.ndarray] = None    # int64 [T, 2]
+    right_q: Optional[np.ndarray] = None
+    is_cat: bool = False
+    cat_bits: Optional[np.ndarray] = None
+
+    @property
+    def is_valid(self) -> bool:
+        return self.feature >= 0 and np.isfinite(self.gain) and self.gain > 0
