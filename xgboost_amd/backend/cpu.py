@@ -33,11 +33,14 @@ class GradQuantizer:
     """
 
     def __init__(self, gpair: torch.Tensor):
-        g = gpair[..., 0]
-        h = gpair[..., 1]
-        max_g = float(g.abs().max()) if g.numel() else 0.0
-        max_h = float(h.abs().max()) if h.numel() else 0.0
-        max_g, max_h = collective.allreduce_max_scalars([max_g, max_h])
+        if gpair.numel():
+            m = gpair.abs().amax(dim=0)  # one reduce, one D2H sync
+            if collective.is_distributed():
+                collective.allreduce_max_(m)
+            mh = m.cpu()
+            max_g, max_h = float(mh[0]), float(mh[1])
+        else:
+            max_g, max_h = collective.allreduce_max_scalars([0.0, 0.0])
         self.g_scale = (1 << QSHIFT) / max_g if max_g > 0 else 1.0
         self.h_scale = (1 << QSHIFT) / max_h if max_h > 0 else 1.0
 

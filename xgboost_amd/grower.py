@@ -118,9 +118,27 @@ class TreeGrower:
         else:
             self.cat_mask = None
 
+    _monitor = None
+
+    @classmethod
+    def monitor(cls):
+        if cls._monitor is None:
+            from .monitor import Monitor
+            cls._monitor = Monitor("TreeGrower")
+        return cls._monitor
+
     def grow(self, qgpair: torch.Tensor, tree: RegTree
              ) -> Tuple[RegTree, torch.Tensor]:
         """Returns (tree, leaf position per row int32)."""
+        mon = self.monitor()
+        mon.start("grow")
+        try:
+            return self._grow(qgpair, tree)
+        finally:
+            mon.stop("grow")
+
+    def _grow(self, qgpair: torch.Tensor, tree: RegTree
+              ) -> Tuple[RegTree, torch.Tensor]:
         param = self.param
         ops = self.ops
         ops.reset(self.n_rows)
