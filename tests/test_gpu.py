@@ -281,3 +281,26 @@ def test_gpu_shap_with_missing():
     contribs = bst.predict(d, pred_contribs=True)
     margin = bst.predict(d, output_margin=True)
     assert np.allclose(contribs.sum(axis=1), margin, atol=1e-3)
+
+
+def test_device_data_ingestion():
+    X, y = _data(20000, 10)
+    Xd = torch.from_numpy(X).cuda()
+    d_dev = xgb.DMatrix(Xd, label=y)
+    assert d_dev.num_row() == 20000
+    res = {}
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 5,
+                     "device": "cuda", "eval_metric": "auc"}, d_dev, 10,
+                    evals=[(d_dev, "t")], evals_result=res,
+                    verbose_eval=False)
+    assert res["t"]["auc"][-1] > 0.9
+    # compare with host ingestion (same data): quality must match closely
+    d_host = xgb.DMatrix(X, label=y)
+    res2 = {}
+    xgb.train({"objective": "binary:logistic", "max_depth": 5,
+               "device": "cuda", "eval_metric": "auc"}, d_host, 10,
+              evals=[(d_host, "t")], evals_result=res2, verbose_eval=False)
+    assert abs(res["t"]["auc"][-1] - res2["t"]["auc"][-1]) < 0.02
+    # inplace predict from a device tensor
+    p = bst.inplace_predict(Xd)
+    assert p.shape == (20000,)

@@ -429,7 +429,8 @@ def shap_gpu(booster, dmat, lo: int, hi: int, phi: np.ndarray) -> np.ndarray:
         raise ImportError("GPU SHAP limits exceeded; falling back to CPU")
     device = booster.device
     fa = _ForestArrays(booster, lo, hi, device)
-    X = torch.from_numpy(dmat.raw_data()).to(device)
+    dd = dmat.device_data() if hasattr(dmat, 'device_data') else None
+    X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
     expected = torch.tensor(
         [_expected_value(booster.trees[t]) for t in range(lo, hi)],
         dtype=torch.float64, device=device)
@@ -460,7 +461,8 @@ def predict_margin_gpu(booster, dmat, out_margin: torch.Tensor,
     lib = hip_ops.load()
     device = out_margin.device
     fa = _ForestArrays(booster, lo, hi, device)
-    X = torch.from_numpy(dmat.raw_data()).to(device)
+    dd = dmat.device_data() if hasattr(dmat, "device_data") else None
+    X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
     n = dmat.num_row()
     missing = dmat.missing
     missing_is_nan = 1 if np.isnan(missing) else 0

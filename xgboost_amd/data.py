@@ -122,6 +122,85 @@ def _np_dtype(td: torch.dtype):
     return {torch.uint8: np.uint8, torch.int16: np.int16, torch.int32: np.int32}[td]
 
 
+def make_cuts_device(Xd: torch.Tensor, max_bin: int, missing: float,
+                     feature_types=None) -> HistogramCuts:
+    """Cuts from a device-resident matrix: nanquantile on the GPU, final
+    cut selection on host from the small [K, f] summary."""
+    from .sketch import cuts_from_summaries
+    n, f = Xd.shape
+    if np.isnan(missing):
+        Xm = Xd
+    else:
+        Xm = torch.where(Xd == missing, torch.full_like(Xd, float("nan")), Xd)
+    finite = torch.isfinite(Xm)
+    cnt = finite.sum(dim=0)
+    K = max(64, 8 * max_bin)
+    qs = torch.from_numpy(((np.arange(K) + 0.5) / K).astype(np.float32)
+                          ).to(Xd.device)
+    # nanquantile with 'lower' keeps actual data values (distinct-value
+    # cuts must be exact data points)
+    pts = torch.nanquantile(Xm.double(), qs.double(), dim=0,
+                            interpolation="lower").float()  # [K, f]
+    mins = torch.where(cnt > 0,
+                       torch.nan_to_num(Xm, nan=float("inf")).amin(dim=0),
+                       torch.zeros(f, device=Xd.device))
+    maxs = torch.where(cnt > 0,
+                       torch.nan_to_num(Xm, nan=float("-inf")).amax(dim=0),
+                       torch.zeros(f, device=Xd.device))
+    pts_h = pts.cpu().numpy()
+    cnt_h = cnt.cpu().numpy()
+    mins_h = mins.cpu().numpy()
+    maxs_h = maxs.cpu().numpy()
+    summaries = []
+    for j in range(f):
+        if feature_types is not None and feature_types[j] == "c":
+            col = Xm[:, j]
+            cats = torch.unique(col[finite[:, j]]).cpu().numpy().astype(
+                np.float32)
+            summaries.append(("c", cats, float(cnt_h[j])))
+        elif cnt_h[j] == 0:
+            summaries.append(("q", np.zeros(0, np.float32), 0.0))
+        else:
+            arr = np.concatenate([[mins_h[j]], pts_h[:, j], [maxs_h[j]]])
+            summaries.append(("q", arr.astype(np.float32), float(cnt_h[j])))
+    # distributed: allgather the per-rank summaries before cut selection
+    from .sketch import sketch_cuts_batches
+    return sketch_cuts_batches([summaries], max_bin, f,
+                               list(feature_types) if feature_types else None)
+
+
+def quantize_dense_device(Xd: torch.Tensor, cuts: HistogramCuts,
+                          missing: float) -> QuantizedMatrix:
+    """GPU compression via the HIP kernel (gbt_compress)."""
+    from . import ops as hip_ops
+    lib = hip_ops.load()
+    n, f = Xd.shape
+    dev = Xd.device
+    has_missing = bool(torch.isnan(Xd).any().item()) if np.isnan(missing) \
+        else bool(((Xd == missing) | torch.isnan(Xd)).any().item())
+    n_bins = np.diff(cuts.ptrs).astype(np.int64)
+    max_local = int(n_bins.max()) - (0 if has_missing else 1)
+    dtype = _pick_bin_dtype(max_local)
+    cut_vals = torch.from_numpy(cuts.values).to(dev)
+    cut_ptrs = torch.from_numpy(cuts.ptrs.astype(np.int32)).to(dev)
+    cat_t = None
+    if cuts.feature_types is not None:
+        cat_t = torch.tensor([1 if t == "c" else 0
+                              for t in cuts.feature_types],
+                             dtype=torch.uint8, device=dev)
+    out = torch.empty((n, f), dtype=dtype, device=dev)
+    missing_is_nan = 1 if np.isnan(missing) else 0
+    Xc = Xd.contiguous()
+    lib.gbt_compress(
+        hip_ops.ptr(Xc), n, f, hip_ops.ptr(cut_vals), hip_ops.ptr(cut_ptrs),
+        hip_ops.ptr(cat_t), float(0.0 if missing_is_nan else missing),
+        missing_is_nan,
+        hip_ops.ptr(out) if dtype == torch.uint8 else None,
+        hip_ops.ptr(out) if dtype != torch.uint8 else None,
+        hip_ops.stream())
+    return QuantizedMatrix(gidx=out, cuts=cuts, has_missing=has_missing)
+
+
 class DMatrix:
     """User-facing data holder (reference: python Booster/DMatrix API).
 
@@ -138,6 +217,17 @@ class DMatrix:
                  nthread: Optional[int] = None, enable_categorical: bool = False,
                  silent: bool = False):
         self.missing = float("nan") if missing is None else float(missing)
+        self._device_data: Optional[torch.Tensor] = None
+        if isinstance(data, torch.Tensor) and data.is_cuda:
+            # zero-copy device ingestion (reference CupyAdapter,
+            # src/data/device_adapter.cuh); sketch+compress run on GPU
+            self._device_data = data.to(torch.float32).contiguous()
+            data = None
+        elif hasattr(data, "__cuda_array_interface__"):
+            self._device_data = torch.as_tensor(data).to(
+                torch.float32).contiguous()
+            data = None
+        inferred_names = inferred_types = None
         if isinstance(data, str) or hasattr(data, "__fspath__"):
             # text file (libsvm format; deprecated upstream, data.cc:930)
             import os as _os
@@ -148,9 +238,14 @@ class DMatrix:
                 label = file_labels
             if qid is None and file_qid is not None:
                 qid = file_qid
-        X, inferred_names, inferred_types = _ingest(data, enable_categorical)
-        self._data = X  # np.float32 [n, f]
-        self.info = MetaInfo(num_row=X.shape[0], num_col=X.shape[1])
+        if self._device_data is None:
+            X, inferred_names, inferred_types = _ingest(data, enable_categorical)
+            self._data = X  # np.float32 [n, f]
+            n_row, n_col = X.shape
+        else:
+            self._data = None
+            n_row, n_col = self._device_data.shape
+        self.info = MetaInfo(num_row=n_row, num_col=n_col)
         if label is not None:
             self.info.labels = _as_float_array(label)
         if weight is not None:
@@ -253,7 +348,13 @@ class DMatrix:
 
     # -- internal --------------------------------------------------------------
     def raw_data(self) -> np.ndarray:
+        if self._data is None and self._device_data is not None:
+            self._data = np.ascontiguousarray(
+                self._device_data.cpu().numpy(), dtype=np.float32)
         return self._data
+
+    def device_data(self) -> Optional[torch.Tensor]:
+        return self._device_data
 
     def set_ref_cuts(self, cuts: HistogramCuts) -> None:
         """Bin this matrix with cut points from a training DMatrix
@@ -265,18 +366,29 @@ class DMatrix:
         """Lazily build (and cache) the quantized matrix for max_bin."""
         qm = self._quantized.get(max_bin)
         if qm is None:
-            if self._ref_cuts is not None:
-                cuts = self._ref_cuts
-            elif sketch_fn is not None:
-                cuts = sketch_fn(self, max_bin)
+            if self._device_data is not None:
+                qm = self._quantize_on_device(max_bin)
             else:
-                cuts = make_cuts(self._data, max_bin,
-                                 weights=None,
-                                 feature_types=self.info.feature_types,
-                                 missing=self.missing)
-            qm = quantize_dense(self._data, cuts, self.missing)
+                if self._ref_cuts is not None:
+                    cuts = self._ref_cuts
+                elif sketch_fn is not None:
+                    cuts = sketch_fn(self, max_bin)
+                else:
+                    cuts = make_cuts(self._data, max_bin,
+                                     weights=None,
+                                     feature_types=self.info.feature_types,
+                                     missing=self.missing)
+                qm = quantize_dense(self._data, cuts, self.missing)
             self._quantized[max_bin] = qm
         return qm
+
+    def _quantize_on_device(self, max_bin: int) -> QuantizedMatrix:
+        """GPU sketch + compress for device-resident input (reference
+        AdapterDeviceSketch hist_util.cuh:323 + CompressBinEllpackKernel)."""
+        cuts = self._ref_cuts or make_cuts_device(
+            self._device_data, max_bin, self.missing,
+            self.info.feature_types)
+        return quantize_dense_device(self._device_data, cuts, self.missing)
 
     def cached_cuts(self) -> Optional[HistogramCuts]:
         for qm in self._quantized.values():
