@@ -335,6 +335,126 @@ class GpuOps(SegmentedOpsMixin):
             out.append(((s, mid), (mid, e)))
         return out
 
+    # -- native C++ level-loop driver (driver.hip) ----------------------
+    def grow_tree_native(self, qgpair: torch.Tensor, tree, param,
+                         quantizer, monotone: Optional[np.ndarray],
+                         root_sums: Tuple[int, int]):
+        """Run the whole per-tree loop in C++ (gbt_grow_tree).  Returns
+        (tree, positions) or None when the config is unsupported (the
+        Python driver handles those)."""
+        import ctypes
+        if not hasattr(self.lib, "gbt_grow_tree"):
+            return None
+        if param.grow_policy != "depthwise" or param.max_leaves > 0:
+            return None
+        if param.max_depth <= 0 or param.max_depth > 14:
+            return None
+        n_rows = qgpair.shape[0]
+        max_build = max(1, 1 << max(param.max_depth - 2, 0))
+        pool_rows = 2 * max_build
+        pool_bytes = pool_rows * self.n_bins * 16
+        if pool_bytes > (32 << 30):  # fall back rather than blow memory
+            return None
+        dev = self.device
+        ws = getattr(self, "_native_ws", None)
+        if ws is None or ws["pool_rows"] < pool_rows or ws["n"] < n_rows:
+            ws = {
+                "pool_rows": pool_rows, "n": n_rows,
+                "ridx": torch.empty(n_rows, dtype=torch.int32, device=dev),
+                "ridx_out": torch.empty(n_rows, dtype=torch.int32,
+                                        device=dev),
+                "pool_a": torch.empty((pool_rows, self.n_bins, 2),
+                                      dtype=torch.int64, device=dev),
+                "pool_b": torch.empty((pool_rows, self.n_bins, 2),
+                                      dtype=torch.int64, device=dev),
+                "eval_gain": torch.empty((pool_rows, self.qm.n_features),
+                                         dtype=torch.float64, device=dev),
+                "eval_bin": torch.empty((pool_rows, self.qm.n_features),
+                                        dtype=torch.int32, device=dev),
+                "eval_dir": torch.empty((pool_rows, self.qm.n_features),
+                                        dtype=torch.uint8, device=dev),
+                "eval_lsum": torch.empty((pool_rows, self.qm.n_features, 2),
+                                         dtype=torch.int64, device=dev),
+                "eval_best": torch.empty((pool_rows, 6), dtype=torch.int64,
+                                         device=dev),
+                "pos": torch.zeros(n_rows, dtype=torch.int32, device=dev),
+                "driver": self.lib.gbt_driver_create(),
+            }
+            self._native_ws = ws
+        cap = 1 << (param.max_depth + 1)
+        host = {name: np.zeros(cap, dt) for name, dt in [
+            ("left", np.int32), ("right", np.int32), ("parent", np.int32),
+            ("split_index", np.int32), ("split_cond", np.float32),
+            ("default_left", np.uint8), ("loss_chg", np.float32),
+            ("sum_hess", np.float32), ("base_weight", np.float32)]}
+        mono_dev = mono_host = None
+        if monotone is not None:
+            m8 = np.ascontiguousarray(monotone, np.int8)
+            self._mono_dev_t = torch.from_numpy(m8).to(dev)
+            self._mono_host = m8
+            mono_dev = self.hip.ptr(self._mono_dev_t)
+            mono_host = m8.ctypes.data_as(ctypes.c_void_p)
+        cuts = self.qm.cuts
+        cut_ptrs_host = np.ascontiguousarray(cuts.ptrs, np.int32)
+        cut_values_host = np.ascontiguousarray(cuts.values, np.float32)
+        cb = None
+        if collective.is_distributed():
+            pool_a, pool_b = ws["pool_a"], ws["pool_b"]
+            base_a, base_b = pool_a.data_ptr(), pool_b.data_ptr()
+
+            def _allreduce(ptr, n_elems):
+                addr = ctypes.addressof(ptr.contents)
+                if base_a <= addr < base_a + pool_a.numel() * 8:
+                    base, pool = base_a, pool_a
+                else:
+                    base, pool = base_b, pool_b
+                off = (addr - base) // 8
+                view = pool.view(-1)[off:off + n_elems]
+                collective.allreduce_sum_(view)
+
+            cb = self.hip.ALLREDUCE_FN(_allreduce)
+        p8, p16 = self._gidx_ptrs()
+        rc = self.lib.gbt_grow_tree(
+            ws["driver"], p8, p16, self.qm.n_features, n_rows,
+            self.hip.ptr(qgpair),
+            self.hip.ptr(self.cut_ptrs),
+            cut_values_host.ctypes.data_as(ctypes.c_void_p),
+            cut_ptrs_host.ctypes.data_as(ctypes.c_void_p),
+            self.hip.ptr(self.n_bins_feat),
+            self.hip.ptr(self.feat_group_start),
+            self.hip.ptr(self.bin_group_start), self.n_groups,
+            self.max_group_bins, self.use_shared, self.n_bins,
+            self.hip.ptr(ws["ridx"]), self.hip.ptr(ws["ridx_out"]),
+            self.hip.ptr(ws["pool_a"]), self.hip.ptr(ws["pool_b"]),
+            self.hip.ptr(ws["eval_gain"]), self.hip.ptr(ws["eval_bin"]),
+            self.hip.ptr(ws["eval_dir"]), self.hip.ptr(ws["eval_lsum"]),
+            self.hip.ptr(ws["eval_best"]), self.hip.ptr(ws["pos"]),
+            max_build,
+            quantizer.g_scale, quantizer.h_scale,
+            root_sums[0], root_sums[1],
+            param.reg_lambda, param.reg_alpha, param.max_delta_step,
+            param.min_child_weight, param.gamma, param.eta, param.max_depth,
+            mono_dev, mono_host, cb,
+            *[host[k].ctypes.data_as(ctypes.c_void_p) for k in (
+                "left", "right", "parent", "split_index", "split_cond",
+                "default_left", "loss_chg", "sum_hess", "base_weight")],
+            self.hip.stream())
+        if rc <= 0:
+            raise RuntimeError(f"gbt_grow_tree failed: rc={rc}")
+        n = rc
+        tree._ensure(n)
+        tree.n_nodes = n
+        tree.left[:n] = host["left"][:n]
+        tree.right[:n] = host["right"][:n]
+        tree.parent[:n] = host["parent"][:n]
+        tree.split_index[:n] = host["split_index"][:n]
+        tree.split_cond[:n] = host["split_cond"][:n]
+        tree.default_left[:n] = host["default_left"][:n]
+        tree.loss_chg[:n] = host["loss_chg"][:n]
+        tree.sum_hess[:n] = host["sum_hess"][:n]
+        tree.base_weight[:n] = host["base_weight"][:n]
+        return tree, ws["pos"]
+
     def leaf_partition(self, ridx: torch.Tensor,
                        leaf_segments: Sequence[Tuple[int, int, int]],
                        n_rows: int) -> torch.Tensor:

@@ -133,9 +133,28 @@ class TreeGrower:
         mon = self.monitor()
         mon.start("grow")
         try:
+            native = self._try_native(qgpair, tree)
+            if native is not None:
+                return native
             return self._grow(qgpair, tree)
         finally:
             mon.stop("grow")
+
+    def _try_native(self, qgpair, tree):
+        """C++ level-loop fast path (ops/cpp/driver.hip) for the common
+        depthwise numeric config; feature-rich paths stay in Python."""
+        p = self.param
+        ops = self.ops
+        if not hasattr(ops, "grow_tree_native"):
+            return None
+        if (self.cat_mask is not None or self.interaction is not None
+                or p.colsample_bytree < 1.0 or p.colsample_bylevel < 1.0
+                or p.colsample_bynode < 1.0):
+            return None
+        root_sums = ops.root_sum(qgpair)
+        out = ops.grow_tree_native(qgpair, tree, p, self.quantizer,
+                                   self.monotone, root_sums)
+        return out
 
     def _grow(self, qgpair: torch.Tensor, tree: RegTree
               ) -> Tuple[RegTree, torch.Tensor]:

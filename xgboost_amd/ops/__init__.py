@@ -38,6 +38,25 @@ _SIGS = {
     "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
     "gbt_copy_ranges": [_p, _p, _p, _i, _p],
     "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p],
+    # native level-loop driver
+    "gbt_driver_create": [],
+    "gbt_driver_destroy": [_p],
+    "gbt_grow_tree": [
+        _p,                      # ctx
+        _p, _p, _i, _i64, _p,    # gidx8/16, n_features, n_rows, qgpair
+        _p, _p, _p, _p,          # cut_ptrs_dev, cut_values_host, cut_ptrs_host, n_bins_feat_dev
+        _p, _p, _i, _i, _i, _i,  # groups, n_groups, max_group_bins, use_shared, n_bins
+        _p, _p, _p, _p,          # ridx, ridx_out, pool_a, pool_b
+        _p, _p, _p, _p, _p, _p,  # eval_gain/bin/dir/lsum/best, pos_out
+        _i,                      # max_nodes_level
+        _d, _d, _i64, _i64,      # scales, root sums
+        _d, _d, _d, _d, _d, _d,  # lambda, alpha, mds, mcw, gamma, eta
+        _i,                      # max_depth
+        _p, _p,                  # monotone dev/host
+        _p,                      # allreduce callback
+        _p, _p, _p, _p, _p, _p, _p, _p, _p,  # tree out arrays
+        _p,                      # stream
+    ],
     "gbt_shap": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p, _p,
                  _p, _p, _i, _i, _i, _p, _p, _p],
 }
@@ -64,7 +83,13 @@ def load() -> ctypes.CDLL:
             continue  # optional kernels (shap) may not be built yet
         fn.argtypes = argtypes
         fn.restype = None
+    if hasattr(_lib, "gbt_driver_create"):
+        _lib.gbt_driver_create.restype = _p
+        _lib.gbt_grow_tree.restype = _c.c_int
     return _lib
+
+
+ALLREDUCE_FN = _c.CFUNCTYPE(None, _c.POINTER(_c.c_longlong), _c.c_longlong)
 
 
 def available() -> bool:

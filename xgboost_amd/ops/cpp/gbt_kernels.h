@@ -85,4 +85,12 @@ void gbt_leaf_partition(const int32_t* ridx, const BlockTask* tasks,
                         int n_tasks, const int32_t* leaf_ids,
                         int32_t* out_pos, hipStream_t stream);
 
+void gbt_copy_ranges(const int32_t* src, int32_t* dst,
+                     const BlockTask* tasks, int n_tasks,
+                     hipStream_t stream);
+
+void gbt_select_best(const double* gain, const int32_t* bins,
+                     const uint8_t* dirs, const int64_t* lsum, int n_nodes,
+                     int n_features, int64_t* out_best, hipStream_t stream);
+
 }  // extern "C"
