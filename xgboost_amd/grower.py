@@ -279,8 +279,14 @@ class TreeGrower:
                 entries = self._evaluate(eval_nids, node_sums, hists,
                                          node_bounds, depth,
                                          hist_stack=hist_stack)
-                for nid, e in zip(eval_nids, entries):
-                    push(nid, depth, e)
+                entry_of = dict(zip(eval_nids, entries))
+                # push in (left, right) pair order so node numbering at
+                # the next level matches the native C++ driver
+                for (b, l, r) in children:
+                    if l in entry_of:
+                        push(l, depth, entry_of[l])
+                    if r in entry_of:
+                        push(r, depth, entry_of[r])
 
         # finalize leaves
         leaf_nids = []
@@ -483,8 +489,12 @@ class MultiTargetGrower:
                 eval_nids = ([n for n, _, _ in build_nodes]
                              + [s for _, _, s in build_nodes])
                 depth = children[0][0].depth + 1
-                for nid, e in zip(eval_nids, evaluate(eval_nids, depth)):
-                    push(nid, depth, e)
+                entry_of = dict(zip(eval_nids, evaluate(eval_nids, depth)))
+                for (b, l, r) in children:
+                    if l in entry_of:
+                        push(l, depth, entry_of[l])
+                    if r in entry_of:
+                        push(r, depth, entry_of[r])
 
         leaf_nids = []
         for nid in range(tree.n_nodes):
