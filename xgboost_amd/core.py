@@ -150,6 +150,13 @@ class Booster:
                 ops = ExtMemOps(dmat, self.device)
                 self._ops_cache[key] = ops
                 return ops
+            if getattr(dmat, "_sparse_data", None) is not None:
+                from .sparse import CsrCpuOps, CsrGpuOps
+                sqm = dmat.sparse_quantized(self.tparam.max_bin)
+                ops = (CsrGpuOps(sqm, self.device)
+                       if self.device.type == "cuda" else CsrCpuOps(sqm))
+                self._ops_cache[key] = ops
+                return ops
             from .sketch import sketch_cuts
             max_bin = self.tparam.max_bin
             qm = dmat.quantized(max_bin, sketch_fn=sketch_cuts)
@@ -432,6 +439,8 @@ class Booster:
         from .extmem import ExtMemQuantileDMatrix
         if isinstance(dmat, ExtMemQuantileDMatrix):
             return self._predict_margin_extmem(dmat, out, lo, hi)
+        if getattr(dmat, "_sparse_data", None) is not None:
+            return self._predict_margin_sparse(dmat, out, lo, hi)
         X = dmat.raw_data()
         has_mt = any(t.leaf_values is not None for t in self.trees[lo:hi])
         if self.device.type == "cuda" and (hi - lo) > 0 and not has_mt:
@@ -447,6 +456,40 @@ class Booster:
                 vals = tree.split_cond[:tree.n_nodes][pos]
                 out[:, self.tree_info[t]] += torch.as_tensor(
                     vals, device=out.device)
+        return out
+
+    def _predict_margin_sparse(self, dmat, out: torch.Tensor,
+                               lo: int, hi: int) -> torch.Tensor:
+        """Sparse predict: densify ONLY the features used by the trees,
+        absent entries become NaN (missing -> default direction)."""
+        csr = dmat.sparse_data()
+        n = csr.shape[0]
+        used = sorted(set(
+            int(f) for t in range(lo, hi)
+            for nid in range(self.trees[t].n_nodes)
+            if not self.trees[t].is_leaf(nid)
+            for f in [self.trees[t].split_index[nid]]))
+        if not used:
+            return out
+        col_of = {f: i for i, f in enumerate(used)}
+        sub = csr[:, used].tocoo()
+        X_sub = np.full((n, len(used)), np.nan, dtype=np.float32)
+        X_sub[sub.row, sub.col] = sub.data
+        for t in range(lo, hi):
+            tree = self.trees[t]
+            remap = tree.split_index[:tree.n_nodes].copy()
+            saved = remap.copy()
+            for nid in range(tree.n_nodes):
+                if not tree.is_leaf(nid):
+                    remap[nid] = col_of[int(saved[nid])]
+            tree.split_index[:tree.n_nodes] = remap
+            try:
+                pos = tree.predict_leaf_np(X_sub, np.nan)
+            finally:
+                tree.split_index[:tree.n_nodes] = saved
+            vals = tree.split_cond[:tree.n_nodes][pos]
+            out[:, self.tree_info[t]] += torch.as_tensor(
+                vals, device=out.device)
         return out
 
     def _predict_margin_extmem(self, dmat, out: torch.Tensor,

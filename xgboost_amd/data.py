@@ -238,7 +238,16 @@ class DMatrix:
                 label = file_labels
             if qid is None and file_qid is not None:
                 qid = file_qid
-        if self._device_data is None:
+        self._sparse_data = None
+        if (hasattr(data, "tocsr") and hasattr(data, "nnz")
+                and self._device_data is None):
+            # scipy sparse: keep CSR — absent entries are MISSING values
+            # (reference SparsePage semantics), never densified
+            self._sparse_data = data.tocsr().astype(np.float32)
+            self._data = None
+            n_row, n_col = self._sparse_data.shape
+            inferred_names = inferred_types = None
+        elif self._device_data is None:
             X, inferred_names, inferred_types = _ingest(data, enable_categorical)
             self._data = X  # np.float32 [n, f]
             n_row, n_col = X.shape
@@ -355,6 +364,20 @@ class DMatrix:
 
     def device_data(self) -> Optional[torch.Tensor]:
         return self._device_data
+
+    def sparse_data(self):
+        return self._sparse_data
+
+    def sparse_quantized(self, max_bin: int):
+        """Quantized CSR (sparse path; see sparse.py)."""
+        key = ("sparse", max_bin)
+        sqm = self._quantized.get(key)
+        if sqm is None:
+            from .sparse import quantize_csr, sketch_csr
+            cuts = self._ref_cuts or sketch_csr(self._sparse_data, max_bin)
+            sqm = quantize_csr(self._sparse_data, cuts)
+            self._quantized[key] = sqm
+        return sqm
 
     def set_ref_cuts(self, cuts: HistogramCuts) -> None:
         """Bin this matrix with cut points from a training DMatrix
