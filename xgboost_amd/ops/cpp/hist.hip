@@ -17,8 +17,9 @@
 #include "gbt_kernels.h"
 
 #ifndef GBT_HIST_BLOCK
-#define GBT_HIST_BLOCK 256
+#define GBT_HIST_BLOCK 1024
 #endif
+#define GBT_HIST_MAX_F 1024  // per-block staged feature metadata cap
 
 template <typename BinT, bool kUseShared>
 __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
@@ -36,40 +37,69 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
   const int bin_begin = bin_group_start[group];
   const int bin_end = bin_group_start[group + 1];
   const int group_bins = bin_end - bin_begin;
+  const int gf = f_end - f_begin;
+
+  // stage per-feature (start bin - bin_begin, n_bins) in LDS so the hot
+  // loop does no global cut_ptrs loads
+  __shared__ int s_start[GBT_HIST_MAX_F];
+  __shared__ int s_width[GBT_HIST_MAX_F];
+  const bool stage_meta = gf <= GBT_HIST_MAX_F;
+  if (stage_meta) {
+    for (int f = threadIdx.x; f < gf; f += blockDim.x) {
+      const int c0 = cut_ptrs[f_begin + f];
+      s_start[f] = c0 - (kUseShared ? bin_begin : 0);
+      s_width[f] = cut_ptrs[f_begin + f + 1] - c0;
+    }
+  }
 
   extern __shared__ unsigned long long smem[];  // [group_bins][2]
   if (kUseShared) {
     for (int i = threadIdx.x; i < group_bins * 2; i += blockDim.x) {
       smem[i] = 0ULL;
     }
-    __syncthreads();
   }
+  __syncthreads();
 
   unsigned long long* hist_s = smem;
   int64_t* hist_g = out_hist + (size_t)task.out_slot * n_bins * 2;
 
-  // each thread walks rows with stride blockDim; per row it reads the
-  // gradient pair once and then the group's bin bytes.
-  for (int i = task.row_begin + threadIdx.x; i < task.row_end;
+  for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
        i += blockDim.x) {
     const int row = ridx[i];
     const long long g = qgpair[2 * (size_t)row];
     const long long h = qgpair[2 * (size_t)row + 1];
-    const BinT* rowbins = gidx + (size_t)row * n_features;
-    for (int f = f_begin; f < f_end; ++f) {
-      const int local = (int)rowbins[f];
-      const int fbins = cut_ptrs[f + 1] - cut_ptrs[f];
-      if (local >= fbins) continue;  // missing sentinel
-      const int gbin = cut_ptrs[f] + local;
-      if (kUseShared) {
-        const int sbin = gbin - bin_begin;
-        atomicAdd(&hist_s[2 * sbin], (unsigned long long)g);
-        atomicAdd(&hist_s[2 * sbin + 1], (unsigned long long)h);
-      } else {
-        atomicAdd((unsigned long long*)&hist_g[2 * gbin],
-                  (unsigned long long)g);
-        atomicAdd((unsigned long long*)&hist_g[2 * gbin + 1],
-                  (unsigned long long)h);
+    const BinT* rowbins = gidx + (size_t)row * n_features + f_begin;
+    if (stage_meta) {
+      for (int f = 0; f < gf; ++f) {
+        const int local = (int)rowbins[f];
+        if (local >= s_width[f]) continue;  // missing sentinel
+        const int sbin = s_start[f] + local;
+        if (kUseShared) {
+          atomicAdd(&hist_s[2 * sbin], (unsigned long long)g);
+          atomicAdd(&hist_s[2 * sbin + 1], (unsigned long long)h);
+        } else {
+          atomicAdd((unsigned long long*)&hist_g[2 * sbin],
+                    (unsigned long long)g);
+          atomicAdd((unsigned long long*)&hist_g[2 * sbin + 1],
+                    (unsigned long long)h);
+        }
+      }
+    } else {
+      for (int f = f_begin; f < f_end; ++f) {
+        const int local = (int)rowbins[f - f_begin];
+        const int c0 = cut_ptrs[f];
+        if (local >= cut_ptrs[f + 1] - c0) continue;
+        const int gbin = c0 + local;
+        if (kUseShared) {
+          const int sbin = gbin - bin_begin;
+          atomicAdd(&hist_s[2 * sbin], (unsigned long long)g);
+          atomicAdd(&hist_s[2 * sbin + 1], (unsigned long long)h);
+        } else {
+          atomicAdd((unsigned long long*)&hist_g[2 * gbin],
+                    (unsigned long long)g);
+          atomicAdd((unsigned long long*)&hist_g[2 * gbin + 1],
+                    (unsigned long long)h);
+        }
       }
     }
   }

@@ -66,12 +66,32 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
   const int my_begin = task.row_begin + (int)threadIdx.x * chunk;
   const int my_end = min(my_begin + chunk, task.row_end);
 
+  // decision cache: avoid the second gather pass when the task fits
+  // the LDS bitmask (64K rows = 8 KiB)
+  constexpr int kMaxCacheRows = 65536;
+  __shared__ uint32_t decide_bits[kMaxCacheRows / 32];
+  const bool use_cache = n_rows <= kMaxCacheRows;
+  if (use_cache) {
+    const int n_words = (n_rows + 31) / 32;
+    for (int i = threadIdx.x; i < n_words; i += blockDim.x) {
+      decide_bits[i] = 0u;
+    }
+    __syncthreads();
+  }
+
   // phase A: count left in my contiguous sub-range
   int my_left = 0;
   for (int i = my_begin; i < my_end; ++i) {
     const int row = ridx_in[i];
     const int local = (int)gidx[(size_t)row * n_features + feature];
-    my_left += DecideLeft(local, fbins, sbin, dleft, cats, cat_words) ? 1 : 0;
+    const bool left = DecideLeft(local, fbins, sbin, dleft, cats, cat_words);
+    my_left += left ? 1 : 0;
+    if (use_cache) {
+      const int k = i - task.row_begin;
+      // each thread owns a contiguous bit range; no races within a word
+      // except at range boundaries -> use atomicOr for safety
+      if (left) atomicOr(&decide_bits[k >> 5], 1u << (k & 31));
+    }
   }
   const int my_rows = max(my_end - my_begin, 0);
   const int my_right = my_rows - my_left;
@@ -111,13 +131,20 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
   }
   __syncthreads();
 
-  // phase C: re-read and scatter
+  // phase C: scatter (cached decisions when they fit, else re-gather)
   int dl = base_l + excl_l;
   int dr = base_r + excl_r;
   for (int i = my_begin; i < my_end; ++i) {
     const int row = ridx_in[i];
-    const int local = (int)gidx[(size_t)row * n_features + feature];
-    if (DecideLeft(local, fbins, sbin, dleft, cats, cat_words)) {
+    bool left;
+    if (use_cache) {
+      const int k = i - task.row_begin;
+      left = (decide_bits[k >> 5] >> (k & 31)) & 1u;
+    } else {
+      const int local = (int)gidx[(size_t)row * n_features + feature];
+      left = DecideLeft(local, fbins, sbin, dleft, cats, cat_words);
+    }
+    if (left) {
       ridx_out[dl++] = row;
     } else {
       ridx_out[dr++] = row;
