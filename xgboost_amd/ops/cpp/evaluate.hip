@@ -69,9 +69,9 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
     const uint8_t* __restrict__ cat_feature, double* __restrict__ out_gain,
     int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dir,
     int64_t* __restrict__ out_lsum) {
-  const int f = blockIdx.x;
   const int node = blockIdx.y;
   const int lane = threadIdx.x;
+  for (int f = blockIdx.x; f < n_features; f += gridDim.x) {
   const size_t out_idx = (size_t)node * n_features + f;
 
   if (feature_mask != nullptr && feature_mask[out_idx] == 0) {
@@ -79,7 +79,7 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
       out_gain[out_idx] = -INFINITY;
       out_bin[out_idx] = -1;
     }
-    return;
+    continue;
   }
 
   Params p;
@@ -151,7 +151,6 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
         hlq = sh + add_h;
       }
       if (valid) {
-        const bool last_bin = (b == fb1 - 1) && !is_cat;
         const long long grq = pg - glq;
         const long long hrq = ph - hlq;
         const double gl = glq * inv_g;
@@ -162,7 +161,9 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
         double wr = CalcWeight(gr, hr, p);
         wl = fmin(fmax(wl, p.lo), p.hi);
         wr = fmin(fmax(wr, p.lo), p.hi);
-        bool ok = (hl >= p.mcw) && (hr >= p.mcw) && !last_bin;
+        // empty-side splits rejected via exact hessian counts; the
+        // last-bin + missing-right split (present vs absent) is valid
+        bool ok = (hl >= p.mcw) && (hr >= p.mcw) && hlq > 0 && hrq > 0;
         if (mono > 0) ok = ok && (wl <= wr);
         if (mono < 0) ok = ok && (wl >= wr);
         if (ok) {
@@ -196,6 +197,7 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
     out_lsum[2 * out_idx] = best.lg;
     out_lsum[2 * out_idx + 1] = best.lh;
   }
+  }  // f stride loop
 }
 
 // Second stage: per-node argmax over features -> packed [n_nodes, 6]
@@ -264,7 +266,7 @@ extern "C" void gbt_evaluate(
     const double* node_bounds, const uint8_t* feature_mask,
     const uint8_t* cat_feature, double* out_gain, int32_t* out_bin,
     uint8_t* out_dir, int64_t* out_lsum, hipStream_t stream) {
-  dim3 grid(n_features, n_nodes);
+  dim3 grid(n_features > 65535 ? 65535 : n_features, n_nodes);
   hipLaunchKernelGGL(EvaluateKernel, grid, dim3(64), 0, stream, hist, n_nodes,
                      n_bins, n_features, cut_ptrs, parent_sums, g_scale,
                      h_scale, reg_lambda, reg_alpha, max_delta_step,

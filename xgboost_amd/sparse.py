@@ -55,6 +55,16 @@ def sketch_csr(X_csr, max_bin: int) -> HistogramCuts:
     """Cuts from the nonzero values of each column (absent = missing)."""
     csc = X_csr.tocsc()
     n_features = csc.shape[1]
+    # fast path: single-valued data (one-hot / binary indicator matrices
+    # at Criteo scale have 1e6 columns — the per-column loop would
+    # dominate).  Every column's cuts = [v + |v| + 1e-5] (one bin).
+    if csc.nnz and csc.data.size and float(csc.data.min()) == float(csc.data.max()):
+        v = float(csc.data.flat[0])
+        sentinel = np.float32(v + (abs(v) + 1e-5))
+        values = np.full(n_features, sentinel, dtype=np.float32)
+        ptrs = np.arange(n_features + 1, dtype=np.int64)
+        min_vals = np.full(n_features, np.float32(v), dtype=np.float32)
+        return HistogramCuts(values=values, ptrs=ptrs, min_vals=min_vals)
     all_cuts: List[np.ndarray] = []
     min_vals = np.zeros(n_features, dtype=np.float32)
     from .quantile import _cuts_for_column
@@ -71,8 +81,16 @@ def sketch_csr(X_csr, max_bin: int) -> HistogramCuts:
 
 
 def quantize_csr(X_csr, cuts: HistogramCuts) -> SparseQuantizedMatrix:
-    csr = X_csr.tocsr().sorted_indices()
+    csr = X_csr.tocsr()
+    csr.sort_indices()
     n, f = csr.shape
+    # fast path: one bin per feature -> global bin == column index
+    if cuts.total_bins == f and np.array_equal(
+            cuts.ptrs, np.arange(f + 1, dtype=cuts.ptrs.dtype)):
+        return SparseQuantizedMatrix(
+            row_ptr=torch.from_numpy(csr.indptr.astype(np.int64)),
+            bin_idx=torch.from_numpy(csr.indices.astype(np.int32)),
+            cuts=cuts, n_features=f)
     bins = np.empty(csr.nnz, dtype=np.int32)
     # per column quantization via CSC, then map back by position
     csc = csr.tocsc()

@@ -151,7 +151,10 @@ def evaluate_splits_np(hist_q: np.ndarray,
                 + calc_gain_given_weight(gr, hr, wr, param)
                 - parent_gain[:, None])
         ok = (hl >= param.min_child_weight) & (hr >= param.min_child_weight)
-        ok &= ~is_last_bin[None, :]
+        # degenerate last-bin split (empty right) is rejected by the
+        # hessian checks / zero gain; last-bin + missing-right is a VALID
+        # "present vs absent" split (essential for one-hot sparse data)
+        ok &= (hrq > 0) & (hlq > 0)
         if mono_bins is not None:
             ok &= ((mono_bins == 0) | ((mono_bins > 0) & (wl <= wr))
                    | ((mono_bins < 0) & (wl >= wr)))
@@ -287,7 +290,7 @@ def evaluate_splits_multi_np(hists: np.ndarray,
             - parent_gain
         ok = ((hl.sum(axis=0) >= param.min_child_weight)
               & (hr.sum(axis=0) >= param.min_child_weight)
-              & ~is_last_bin[None, :])
+              & (hlq.sum(axis=0) > 0) & (hrq.sum(axis=0) > 0))
         gain = np.where(ok, gain, -np.inf)
         for i in range(n_nodes):
             row = gain[i]
