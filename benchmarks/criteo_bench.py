@@ -26,12 +26,13 @@ def synth_csr(n, f, nnz_row, seed):
     y = (rng.rand(n) > 0.5)
     # one indicator nonzero per row: cols [0,100) for positives,
     # [100,200) for negatives (with 20% label noise)
-    first = rows % nnz_row == 0
+    first = np.zeros(len(rows), dtype=bool)
+    first[::nnz_row] = True  # one indicator slot per row
     noisy = rng.rand(n) < 0.2
     pos_like = np.where(noisy, ~y, y)
     cols[first] = np.where(pos_like[rows[first]],
-                           rng.randint(0, 100, first.sum()),
-                           100 + rng.randint(0, 100, first.sum()))
+                           rng.randint(0, 100, n),
+                           100 + rng.randint(0, 100, n))
     vals = np.ones(n * nnz_row, np.float32)
     X = sp.csr_matrix((vals, (rows, cols)), shape=(n, f))
     X.sum_duplicates()
