@@ -163,3 +163,30 @@ def test_categorical_onehot():
     bst2 = xgb.Booster()
     bst2.load_model(bytes(raw))
     assert np.allclose(bst2.predict(d), p, atol=1e-6)
+
+
+def test_categorical_partition_split():
+    """Wide categorical features use sorted-partition subsets
+    (reference: max_cat_to_onehot threshold + SortHistogram path)."""
+    rng = np.random.RandomState(0)
+    n = 4000
+    cat = rng.randint(0, 30, n).astype(np.float32)
+    good = [1, 5, 7, 12, 19, 22, 28]
+    y = (np.isin(cat, good).astype(np.float32) * 2.0
+         + 0.1 * rng.randn(n)).astype(np.float32)
+    X = np.stack([cat, rng.randn(n).astype(np.float32)], axis=1)
+    d = xgb.DMatrix(X, label=y, feature_types=["c", "q"])
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 3,
+                     "eta": 0.5}, d, 8, verbose_eval=False)
+    p = bst.predict(d)
+    assert np.sqrt(np.mean((p - y) ** 2)) < 0.3
+    t = bst.trees[0]
+    assert t.split_type[0] == 1
+    cats_right = set(int(c) for c in t.cat_segments[0])
+    # the good categories must be cleanly separated at the root
+    assert cats_right == set(range(30)) - set(good) or \
+        cats_right == set(good)
+    raw = bst.save_raw("json")
+    b2 = xgb.Booster()
+    b2.load_model(bytes(raw))
+    assert np.allclose(b2.predict(d), p, atol=1e-6)

@@ -84,6 +84,64 @@ class SplitEntry:
         return self.feature >= 0 and np.isfinite(self.gain) and self.gain > 0
 
 
+def _sorted_cat_split(e: SplitEntry, node_hist: np.ndarray, pgq: int, phq: int,
+                      f: int, cut_ptrs: np.ndarray, param: TrainParam,
+                      inv_g: float, inv_h: float, parent_gain: float) -> None:
+    """Partition-based categorical split for wide categorical features
+    (reference gpu_hist/evaluate_splits.cuh SortHistogram + Partition
+    agent): categories sorted by gradient ratio, scanned like a numeric
+    feature; the best prefix goes LEFT, the complement is the stored
+    go-RIGHT set.  Updates `e` in place if a better split is found."""
+    b0, b1 = int(cut_ptrs[f]), int(cut_ptrs[f + 1])
+    Gc = node_hist[b0:b1, 0].astype(np.int64)
+    Hc = node_hist[b0:b1, 1].astype(np.int64)
+    present = Hc != 0
+    if present.sum() < 2:
+        return
+    ratio = np.where(present,
+                     (Gc * inv_g) / (Hc * inv_h + param.reg_lambda), np.inf)
+    order = np.argsort(ratio, kind="stable")
+    order = order[present[order]]          # present categories, sorted
+    order = order[:param.max_cat_threshold + 1]
+    cg = np.cumsum(Gc[order])
+    ch = np.cumsum(Hc[order])
+    featG, featH = int(Gc.sum()), int(Hc.sum())
+    missG, missH = pgq - featG, phq - featH
+    for missing_left in (False, True):
+        gl = cg + (missG if missing_left else 0)
+        hl = ch + (missH if missing_left else 0)
+        gr = pgq - gl
+        hr = phq - hl
+        glf, hlf = gl * inv_g, hl * inv_h
+        grf, hrf = gr * inv_g, hr * inv_h
+        wl = calc_weight(glf, hlf, param)
+        wr = calc_weight(grf, hrf, param)
+        gains = (calc_gain_given_weight(glf, hlf, wl, param)
+                 + calc_gain_given_weight(grf, hrf, wr, param) - parent_gain)
+        ok = ((hlf >= param.min_child_weight) & (hrf >= param.min_child_weight)
+              & (hl > 0) & (hr > 0))
+        gains = np.where(ok, gains, -np.inf)
+        if not np.isfinite(gains).any():
+            continue
+        i = int(np.argmax(gains))
+        gv = float(gains[i])
+        if gv > e.gain:
+            e.gain = gv
+            e.feature = f
+            e.split_bin = b0  # informational; cat_bits carries the split
+            e.default_left = missing_left
+            e.left_gq = int(gl[i])
+            e.left_hq = int(hl[i])
+            e.right_gq = pgq - e.left_gq
+            e.right_hq = phq - e.left_hq
+            e.is_cat = True
+            # everything NOT in the chosen prefix goes RIGHT
+            left_set = set(int(c) for c in order[:i + 1])
+            e.cat_bits = np.array(
+                sorted(c for c in range(b1 - b0) if c not in left_set),
+                dtype=np.int32)
+
+
 def evaluate_splits_np(hist_q: np.ndarray,
                        parent_q: Sequence,
                        g_scale: float, h_scale: float,
@@ -160,10 +218,15 @@ def evaluate_splits_np(hist_q: np.ndarray,
                    | ((mono_bins < 0) & (wl >= wr)))
         gain = np.where(ok, gain, -np.inf)
         if cat_mask is not None and cat_mask.any():
+            # one-hot only below max_cat_to_onehot; wide categorical
+            # features use the sorted-partition path below
+            narrow_cat = cat_mask & (widths <= param.max_cat_to_onehot)
+            wide_bins = (cat_mask & ~narrow_cat)[feat_of_bin]
             gain, cat_lq = _onehot_cat_gains(
-                gain, Gq, Hq, pgq, phq, missGq, missHq, feat_of_bin, cat_mask,
-                param, missing_left, parent_gain, node_bounds, mono_bins,
-                inv_g, inv_h)
+                gain, Gq, Hq, pgq, phq, missGq, missHq, feat_of_bin,
+                narrow_cat, param, missing_left, parent_gain, node_bounds,
+                mono_bins, inv_g, inv_h)
+            gain = np.where(wide_bins[None, :], -np.inf, gain)
         else:
             cat_lq = None
         for i in range(n_nodes):
@@ -196,6 +259,22 @@ def evaluate_splits_np(hist_q: np.ndarray,
                 e.is_cat = is_cat
                 if is_cat:
                     e.cat_bits = np.array([b - int(cut_ptrs[f])], dtype=np.int32)
+    # sorted-partition splits for wide categorical features
+    if cat_mask is not None and cat_mask.any():
+        wide_feats = np.nonzero(cat_mask
+                                & (widths > param.max_cat_to_onehot))[0]
+        if len(wide_feats):
+            for i in range(n_nodes):
+                allowed = (set(int(x) for x in feature_sets[i])
+                           if feature_sets is not None
+                           and feature_sets[i] is not None else None)
+                pgain = parent_gain[i]
+                for f in wide_feats:
+                    if allowed is not None and int(f) not in allowed:
+                        continue
+                    _sorted_cat_split(best[i], hist_q[i], int(pgq[i]),
+                                      int(phq[i]), int(f), cut_ptrs, param,
+                                      inv_g, inv_h, float(pgain))
     return best
 
 
