@@ -825,6 +825,32 @@ class Booster:
             return {fname(f): sums[f] for f in sums}
         return {fname(f): sums[f] / counts[f] for f in sums}
 
+    def get_split_value_histogram(self, feature: str, fmap: str = "",
+                                  bins=None, as_pandas: bool = True):
+        """Histogram of split values for a feature across all trees
+        (reference core.py get_split_value_histogram)."""
+        names = self.feature_names
+        values = []
+        for t in self.trees:
+            for nid in range(t.n_nodes):
+                if t.is_leaf(nid):
+                    continue
+                f = int(t.split_index[nid])
+                fn = names[f] if names and f < len(names) else f"f{f}"
+                if fn == feature or f"f{f}" == feature:
+                    values.append(float(t.split_cond[nid]))
+        values = np.asarray(values, dtype=np.float64)
+        nbins = max(min(int(bins) if bins else 10, values.size), 1)
+        hist, edges = np.histogram(values, bins=nbins)
+        out = np.column_stack([edges[1:], hist])
+        if as_pandas:
+            try:
+                import pandas as pd
+                return pd.DataFrame(out, columns=["SplitValue", "Count"])
+            except ImportError:
+                pass
+        return out
+
     def get_fscore(self, fmap: str = "") -> Dict[str, float]:
         return self.get_score(fmap, "weight")
 

@@ -510,3 +510,7 @@ def _from_pandas(df, enable_categorical: bool):
     else:
         types_out = ["c" if t == "c" else "q" for t in types]
     return X, names, types_out
+
+
+# deprecated alias kept for API compatibility (reference: core.py)
+DeviceQuantileDMatrix = QuantileDMatrix
