@@ -16,6 +16,8 @@
 //    when rows allow, covering 256 CUs across 8 XCDs).
 #include "gbt_kernels.h"
 
+#include <cstdlib>
+
 #ifndef GBT_HIST_BLOCK
 #define GBT_HIST_BLOCK 1024
 #endif
@@ -123,8 +125,13 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
                          const int32_t* bin_group_start, int n_groups,
                          int max_group_bins, const int32_t* cut_ptrs,
                          int use_shared, hipStream_t stream) {
+  static int block_size = [] {
+    const char* e = getenv("GBT_HIST_BLOCK_SIZE");
+    int v = e ? atoi(e) : GBT_HIST_BLOCK;
+    return (v >= 64 && v <= 1024) ? (v & ~63) : GBT_HIST_BLOCK;
+  }();
   dim3 grid(n_tasks, n_groups);
-  dim3 block(GBT_HIST_BLOCK);
+  dim3 block(block_size);
   size_t shmem = use_shared ? (size_t)max_group_bins * 2 * sizeof(int64_t) : 0;
   if (gidx8 != nullptr) {
     if (use_shared) {
