@@ -1,0 +1,133 @@
+"""Callbacks, config context, UBJSON typed arrays, monitor, cv
+(reference analog: tests/python/test_callback.py, test_config.py)."""
+import io
+
+import numpy as np
+import pytest
+
+import xgboost_amd as xgb
+from xgboost_amd.callback import (EarlyStopping, EvaluationMonitor,
+                                  LearningRateScheduler, TrainingCheckPoint)
+from conftest import make_classification, make_regression
+
+
+def test_config_context():
+    from xgboost_amd.config import config_context, get_config, set_config
+    assert get_config()["verbosity"] == 1
+    with config_context(verbosity=3):
+        assert get_config()["verbosity"] == 3
+    assert get_config()["verbosity"] == 1
+    with pytest.raises(ValueError):
+        set_config(bogus=1)
+
+
+def test_learning_rate_scheduler():
+    X, y = make_regression(500, 4)
+    d = xgb.DMatrix(X, label=y)
+    rates = [0.5, 0.4, 0.3, 0.2, 0.1]
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 2}, d, 5,
+                    callbacks=[LearningRateScheduler(rates)],
+                    verbose_eval=False)
+    assert bst.num_boosted_rounds() == 5
+
+
+def test_checkpoint_callback(tmp_path):
+    X, y = make_regression(300, 4)
+    d = xgb.DMatrix(X, label=y)
+    xgb.train({"objective": "reg:squarederror"}, d, 6,
+              callbacks=[TrainingCheckPoint(directory=str(tmp_path),
+                                            name="ckpt", interval=2)],
+              verbose_eval=False)
+    files = sorted(p.name for p in tmp_path.iterdir())
+    assert len(files) >= 2
+    bst = xgb.Booster(model_file=str(tmp_path / files[0]))
+    assert bst.num_features() == 4
+
+
+def test_evaluation_monitor_output(capsys):
+    X, y = make_classification(400, 4)
+    d = xgb.DMatrix(X, label=y)
+    xgb.train({"objective": "binary:logistic"}, d, 3,
+              evals=[(d, "train")], verbose_eval=True)
+    out = capsys.readouterr().out
+    assert "train-logloss" in out
+    assert "[0]" in out
+
+
+def test_early_stopping_save_best():
+    X, y = make_classification(1500, 6)
+    dtrain = xgb.DMatrix(X[:1000], label=y[:1000])
+    dvalid = xgb.DMatrix(X[1000:], label=y[1000:])
+    es = EarlyStopping(rounds=3, save_best=True)
+    bst = xgb.train({"objective": "binary:logistic", "eta": 0.5,
+                     "max_depth": 6}, dtrain, 100,
+                    evals=[(dvalid, "valid")], callbacks=[es],
+                    verbose_eval=False)
+    assert bst.num_boosted_rounds() == bst.best_iteration + 1
+
+
+def test_cv_runs():
+    X, y = make_classification(600, 5)
+    d = xgb.DMatrix(X, label=y)
+    res = xgb.cv({"objective": "binary:logistic", "max_depth": 3}, d,
+                 num_boost_round=5, nfold=3, seed=1)
+    cols = list(res.columns) if hasattr(res, "columns") else list(res)
+    assert any("test-logloss-mean" in c for c in cols)
+    n = len(res)
+    assert n == 5
+
+
+def test_cv_early_stopping():
+    X, y = make_classification(600, 5)
+    d = xgb.DMatrix(X, label=y)
+    res = xgb.cv({"objective": "binary:logistic", "max_depth": 6,
+                  "eta": 0.8}, d, num_boost_round=50, nfold=3,
+                 early_stopping_rounds=3, seed=1)
+    assert len(res) < 50
+
+
+def test_ubjson_typed_array_roundtrip():
+    """Reference UBJSON writers emit optimized typed arrays
+    ([$<type>#<count>); our reader must parse them."""
+    from xgboost_amd.ubjson import dumps_ubjson, loads_ubjson
+    obj = {"a": np.array([1.5, 2.5], np.float32),
+           "b": np.array([1, 2, 3], np.int64),
+           "c": "text", "d": [1, True, None], "e": {"n": 7}}
+    blob = dumps_ubjson(obj)
+    back = loads_ubjson(blob)
+    assert back["a"] == [1.5, 2.5]
+    assert back["b"] == [1, 2, 3]
+    assert back["c"] == "text"
+    assert back["d"] == [1, True, None]
+    assert back["e"] == {"n": 7}
+
+
+def test_monitor_accumulates():
+    from xgboost_amd.monitor import Monitor
+    m = Monitor("test")
+    m.start("phase")
+    m.stop("phase")
+    assert m.counts["phase"] == 1
+    assert "phase" in m.report()
+
+
+def test_validate_parameters():
+    X, y = make_regression(100, 3)
+    d = xgb.DMatrix(X, label=y)
+    with pytest.raises(ValueError):
+        xgb.train({"objective": "reg:squarederror", "typo_param": 1,
+                   "validate_parameters": True}, d, 1, verbose_eval=False)
+    # without validation: accepted silently (reference warns)
+    xgb.train({"objective": "reg:squarederror", "typo_param": 1}, d, 1,
+              verbose_eval=False)
+
+
+def test_seed_per_iteration():
+    X, y = make_classification(800, 5)
+    d = xgb.DMatrix(X, label=y)
+    p = {"objective": "binary:logistic", "subsample": 0.7,
+         "seed_per_iteration": True, "seed": 3}
+    b1 = xgb.train(p, d, 3, verbose_eval=False)
+    d2 = xgb.DMatrix(X, label=y)
+    b2 = xgb.train(p, d2, 3, verbose_eval=False)
+    assert np.array_equal(b1.predict(d), b2.predict(d2))
