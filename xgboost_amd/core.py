@@ -177,6 +177,8 @@ class Booster:
             self._update_existing(dtrain, iteration)
             return
         margin = self._cached_margin(dtrain)
+        if fobj is None and self._boost_fused(dtrain, margin, iteration):
+            return
         if fobj is not None:
             preds = self.objective.pred_transform(margin).cpu().numpy()
             grad, hess = fobj(np.squeeze(preds), dtrain)
@@ -258,6 +260,91 @@ class Booster:
         if self.tparam.debug_synchronize:
             collective.check_synchronized(
                 json.dumps(self.trees[-1].to_json()).encode(), "tree")
+
+    def _boost_fused(self, dtrain: DMatrix, margin: torch.Tensor,
+                     iteration: int) -> bool:
+        """GPU fast path: fused gradient+quantize HIP kernel for the hot
+        objectives (ops/cpp/gpair.hip); returns False when inapplicable
+        (the generic torch path handles those)."""
+        if self.device.type != "cuda":
+            return False
+        obj_id = {"binary:logistic": 0, "reg:squarederror": 1}.get(
+            self.objective.name)
+        if obj_id is None or self.n_outputs != 1:
+            return False
+        tp = self.tparam
+        if (tp.subsample < 1.0 or tp.num_parallel_tree != 1
+                or tp.tree_method not in ("auto", "hist")
+                or self.booster_kind == "gblinear"
+                or tp.process_type == "update"
+                or str(self.raw_params.get("multi_strategy",
+                                           "one_output_per_tree"))
+                != "one_output_per_tree"):
+            return False
+        from . import ops as hip_ops
+        try:
+            lib = hip_ops.load()
+        except RuntimeError:
+            return False
+        if not hasattr(lib, "gbt_gpair_fused"):
+            return False
+        ops = self._ops_for(dtrain)
+        n = dtrain.num_row()
+        key = id(dtrain)
+        fc = self.__dict__.setdefault("_fused_cache", {})
+        ent = fc.get(key)
+        if ent is None:
+            dev = self.device
+            labels = torch.as_tensor(
+                np.ascontiguousarray(dtrain.get_label(), np.float32),
+                device=dev)
+            w = dtrain.info.weights
+            weights = (torch.as_tensor(np.ascontiguousarray(w, np.float32),
+                                       device=dev) if w is not None else None)
+            ent = {"labels": labels, "weights": weights,
+                   "gh": torch.empty((n, 2), dtype=torch.float32, device=dev),
+                   "qg": torch.empty((n, 2), dtype=torch.int32, device=dev),
+                   "maxabs": torch.zeros(2, dtype=torch.float32, device=dev)}
+            fc[key] = ent
+        ent["maxabs"].zero_()
+        spw = float(self.raw_params.get("scale_pos_weight", 1.0))
+        lib.gbt_gpair_fused(
+            obj_id, hip_ops.ptr(margin.contiguous().view(-1)),
+            hip_ops.ptr(ent["labels"]), hip_ops.ptr(ent["weights"]),
+            spw, n, hip_ops.ptr(ent["gh"]), hip_ops.ptr(ent["maxabs"]),
+            hip_ops.stream())
+        m = ent["maxabs"]
+        if collective.is_distributed():
+            collective.allreduce_max_(m)
+        mh = m.cpu()
+        max_g, max_h = float(mh[0]), float(mh[1])
+        quantizer = GradQuantizer.__new__(GradQuantizer)
+        quantizer.g_scale = (1 << 30) / max_g if max_g > 0 else 1.0
+        quantizer.h_scale = (1 << 30) / max_h if max_h > 0 else 1.0
+        lib.gbt_quantize(hip_ops.ptr(ent["gh"]), n, quantizer.g_scale,
+                         quantizer.h_scale, hip_ops.ptr(ent["qg"]),
+                         hip_ops.stream())
+        seed = (self.seed + iteration if not self.seed_per_iteration
+                else self.seed + iteration * 2654435761)
+        tree = RegTree(self.n_features)
+        grower = TreeGrower(ops, self.tparam, quantizer, n, seed=seed,
+                            monotone=self._monotone_array(),
+                            interaction=tp.interaction_constraints)
+        tree, positions = grower.grow(ent["qg"], tree)
+        self.trees.append(tree)
+        self.tree_info.append(0)
+        self.iteration_indptr.append(self.iteration_indptr[-1] + 1)
+        leaf_np = tree.split_cond[:tree.n_nodes].copy()
+        if hasattr(ops, "stager"):
+            (leaf_vals,) = ops.stager.upload([leaf_np])
+        else:
+            leaf_vals = torch.as_tensor(leaf_np, device=margin.device)
+        margin[:, 0] += leaf_vals[positions.to(margin.device).long()]
+        self._cache[key] = (margin, len(self.trees))
+        if self.tparam.debug_synchronize:
+            collective.check_synchronized(
+                json.dumps(self.trees[-1].to_json()).encode(), "tree")
+        return True
 
     def _boost_linear(self, dtrain: DMatrix, grad: torch.Tensor,
                       hess: torch.Tensor, iteration: int) -> None:
