@@ -131,3 +131,31 @@ def test_seed_per_iteration():
     d2 = xgb.DMatrix(X, label=y)
     b2 = xgb.train(p, d2, 3, verbose_eval=False)
     assert np.array_equal(b1.predict(d), b2.predict(d2))
+
+
+def test_category_recode_between_frames():
+    """Predict-frame category dictionaries are re-coded to the training
+    dictionary (reference src/encoder/ordinal.h Recode)."""
+    import pandas as pd
+    rng = np.random.RandomState(0)
+    n = 1000
+    colors = rng.choice(["red", "green", "blue"], n)
+    df = pd.DataFrame({"color": pd.Categorical(colors),
+                       "x": rng.randn(n).astype(np.float32)})
+    y = ((colors == "red") * 2.0).astype(np.float32)
+    d = xgb.DMatrix(df, label=y, enable_categorical=True)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 3}, d, 5,
+                    verbose_eval=False)
+    p_train = bst.predict(d)
+    # same data, shuffled category order
+    df2 = df.copy()
+    df2["color"] = pd.Categorical(colors, categories=["blue", "red", "green"])
+    d2 = xgb.DMatrix(df2, label=y, enable_categorical=True)
+    assert np.allclose(bst.predict(d2), p_train, atol=1e-6)
+    # unseen category behaves as missing (no crash, finite output)
+    df3 = pd.DataFrame({
+        "color": pd.Categorical(["purple"] * 4,
+                                categories=["purple", "red"]),
+        "x": np.zeros(4, np.float32)})
+    d3 = xgb.DMatrix(df3, label=np.zeros(4), enable_categorical=True)
+    assert np.isfinite(bst.predict(d3)).all()

@@ -110,6 +110,10 @@ class Booster:
                 and dmat.info.labels.ndim == 2 and dmat.info.labels.shape[1] > 1 \
                 and self.objective.task == "regression":
             self.n_targets = dmat.info.labels.shape[1]
+        if not getattr(self, "cat_categories_", None) and \
+                getattr(dmat, "categories_", None):
+            self.cat_categories_ = {int(k): list(v)
+                                    for k, v in dmat.categories_.items()}
 
     def _init_base_score(self, dtrain: DMatrix) -> None:
         if self.base_score is not None and self._base_score_estimated:
@@ -529,8 +533,18 @@ class Booster:
         if getattr(dmat, "_sparse_data", None) is not None:
             return self._predict_margin_sparse(dmat, out, lo, hi)
         X = dmat.raw_data()
+        train_cats = getattr(self, "cat_categories_", None)
+        pred_cats = getattr(dmat, "categories_", None)
+        aligned = False
+        if train_cats and pred_cats and pred_cats != train_cats:
+            # predict-frame categories re-coded to the training dictionary
+            # (reference encoder/ordinal.h Recode)
+            from .data import align_categories
+            X = align_categories(X, pred_cats, train_cats)
+            aligned = True
         has_mt = any(t.leaf_values is not None for t in self.trees[lo:hi])
-        if self.device.type == "cuda" and (hi - lo) > 0 and not has_mt:
+        if self.device.type == "cuda" and (hi - lo) > 0 and not has_mt \
+                and not aligned:
             from .backend.gpu import predict_margin_gpu
             return predict_margin_gpu(self, dmat, out, lo, hi)
         for t in range(lo, hi):

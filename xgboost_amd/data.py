@@ -248,7 +248,8 @@ class DMatrix:
             n_row, n_col = self._sparse_data.shape
             inferred_names = inferred_types = None
         elif self._device_data is None:
-            X, inferred_names, inferred_types = _ingest(data, enable_categorical)
+            (X, inferred_names, inferred_types,
+             self.categories_) = _ingest(data, enable_categorical)
             self._data = X  # np.float32 [n, f]
             n_row, n_col = X.shape
         else:
@@ -472,7 +473,10 @@ def _ingest(data: Any, enable_categorical: bool):
     elif hasattr(data, "toarray") and hasattr(data, "tocsr"):  # scipy sparse
         X = data.toarray()
     elif _is_pandas(data):
-        X, names, types = _from_pandas(data, enable_categorical)
+        X, names, types, cats = _from_pandas(data, enable_categorical)
+        if X.ndim == 1:
+            X = X.reshape(-1, 1)
+        return np.ascontiguousarray(X, dtype=np.float32), names, types, cats
     else:
         X = np.asarray(data)
     if X.ndim == 1:
@@ -480,7 +484,7 @@ def _ingest(data: Any, enable_categorical: bool):
     if X.ndim != 2:
         raise ValueError(f"expected 2-D data, got shape {X.shape}")
     X = np.ascontiguousarray(X, dtype=np.float32)
-    return X, names, types
+    return X, names, types, {}
 
 
 def _is_pandas(data: Any) -> bool:
@@ -492,13 +496,17 @@ def _from_pandas(df, enable_categorical: bool):
     names = [str(c) for c in df.columns]
     types: List[str] = []
     cols = []
-    for c in df.columns:
+    categories: Dict[int, list] = {}
+    for j, c in enumerate(df.columns):
         s = df[c]
         if isinstance(s.dtype, pd.CategoricalDtype):
             if not enable_categorical:
                 raise ValueError(
                     f"categorical column {c!r} needs enable_categorical=True")
-            cols.append(s.cat.codes.to_numpy(np.float32))
+            codes = s.cat.codes.to_numpy(np.float32)
+            codes = np.where(codes < 0, np.nan, codes)  # NaN category
+            cols.append(codes)
+            categories[j] = list(s.cat.categories)
             types.append("c")
         else:
             v = s.to_numpy(np.float32, na_value=np.nan)
@@ -509,7 +517,31 @@ def _from_pandas(df, enable_categorical: bool):
         types_out = None
     else:
         types_out = ["c" if t == "c" else "q" for t in types]
-    return X, names, types_out
+    return X, names, types_out, categories
+
+
+def align_categories(X: np.ndarray, pred_cats: Dict[int, list],
+                     train_cats: Dict[int, list]) -> np.ndarray:
+    """Re-code categorical codes from a prediction frame's dictionary to
+    the training dictionary (reference: src/encoder/ordinal.h:349 Recode,
+    data/cat_container.h).  Unseen categories become missing (NaN)."""
+    X = X.copy()
+    for j, cats in pred_cats.items():
+        tcats = train_cats.get(j)
+        if tcats is None or list(cats) == list(tcats):
+            continue
+        lut = {c: i for i, c in enumerate(tcats)}
+        remap = np.full(len(cats), np.nan, dtype=np.float32)
+        for i, c in enumerate(cats):
+            if c in lut:
+                remap[i] = lut[c]
+        col = X[:, j]
+        valid = ~np.isnan(col)
+        idx = col[valid].astype(np.int64)
+        idx = np.clip(idx, 0, len(cats) - 1)
+        col[valid] = remap[idx]
+        X[:, j] = col
+    return X
 
 
 # deprecated alias kept for API compatibility (reference: core.py)
