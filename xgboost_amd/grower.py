@@ -406,6 +406,11 @@ class MultiTargetGrower:
             node_feats = self.col_sampler.node_set(depth)
             if node_feats is not None:
                 fs = [node_feats for _ in nids]
+            if h.is_cuda:
+                out = self._evaluate_gpu(h, parents, g_scales, h_scales,
+                                         nids, fs)
+                if out is not None:
+                    return out
             return evaluate_splits_multi_np(
                 h.cpu().numpy(), parents, g_scales, h_scales, nids,
                 ops.qm.cuts.ptrs, param, feature_sets=fs)
@@ -504,3 +509,74 @@ class MultiTargetGrower:
                 leaf_nids.append(nid)
         positions = ops.leaf_positions(leaf_nids)
         return tree, positions
+
+    def _evaluate_gpu(self, h, parents_np, g_scales, h_scales, nids, fs):
+        """Device MT split evaluation (ops/cpp/mt_evaluate.hip); left
+        sums for the winning bin are derived with torch cumsum on device
+        (reference ScanHistogramKernel + MT EvaluateSplitsKernel)."""
+        from .splits import MultiSplitEntry
+        ops = self.ops
+        if not hasattr(ops, "lib") or not hasattr(ops.lib, "gbt_mt_evaluate"):
+            return None
+        hip = ops.hip
+        dev = h.device
+        T, k, n_bins, _ = h.shape
+        f = ops.qm.n_features
+        cuts = ops.qm.cuts
+        hc = h.contiguous()
+        parents_t = torch.from_numpy(
+            np.ascontiguousarray(parents_np, np.int64)).to(dev)
+        gs_t = torch.from_numpy(np.ascontiguousarray(
+            g_scales, np.float64)).to(dev)
+        hs_t = torch.from_numpy(np.ascontiguousarray(
+            h_scales, np.float64)).to(dev)
+        mask_t = None
+        if fs is not None:
+            m = np.zeros((k, f), np.uint8)
+            for i, s in enumerate(fs):
+                if s is None:
+                    m[i] = 1
+                else:
+                    m[i, np.asarray(s, np.int64)] = 1
+            mask_t = torch.from_numpy(m).to(dev)
+        gain = torch.empty((k, f), dtype=torch.float64, device=dev)
+        bins = torch.empty((k, f), dtype=torch.int32, device=dev)
+        dirs = torch.empty((k, f), dtype=torch.uint8, device=dev)
+        p = self.param
+        ops.lib.gbt_mt_evaluate(
+            hip.ptr(hc), T, k, n_bins, f, hip.ptr(ops.cut_ptrs),
+            hip.ptr(parents_t), hip.ptr(gs_t), hip.ptr(hs_t),
+            p.reg_lambda, p.reg_alpha, p.max_delta_step, p.min_child_weight,
+            hip.ptr(mask_t), hip.ptr(gain), hip.ptr(bins), hip.ptr(dirs),
+            hip.stream())
+        best_f = torch.argmax(gain, dim=1)
+        ar = torch.arange(k, device=dev)
+        packed = torch.stack([gain[ar, best_f].view(torch.int64),
+                              bins[ar, best_f].to(torch.int64),
+                              dirs[ar, best_f].to(torch.int64),
+                              best_f], dim=1).cpu().numpy()
+        ptrs = cuts.ptrs
+        out = []
+        for i, nid in enumerate(nids):
+            e = MultiSplitEntry(nid=int(nid), n_targets=T)
+            gv = packed[i, 0:1].view(np.float64)[0]
+            b = int(packed[i, 1])
+            if b >= 0 and np.isfinite(gv):
+                fi = int(packed[i, 3])
+                e.gain = float(gv)
+                e.feature = fi
+                e.split_bin = b
+                e.default_left = bool(packed[i, 2])
+                fb0, fb1 = int(ptrs[fi]), int(ptrs[fi + 1])
+                seg = hc[:, i, fb0:fb1, :]           # [T, w, 2]
+                cum = seg.cumsum(dim=1)
+                left = cum[:, b - fb0, :]            # [T, 2]
+                if e.default_left:
+                    parent_i = parents_t[i]          # [T, 2]
+                    feat_tot = cum[:, -1, :]
+                    left = left + (parent_i - feat_tot)
+                lq = left.cpu().numpy().astype(np.int64)
+                e.left_q = lq
+                e.right_q = parents_np[i] - lq
+            out.append(e)
+        return out

@@ -38,6 +38,8 @@ _SIGS = {
     "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
     "gbt_copy_ranges": [_p, _p, _p, _i, _p],
     "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p],
+    "gbt_mt_evaluate": [_p, _i, _i, _i, _i, _p, _p, _p, _p,
+                        _d, _d, _d, _d, _p, _p, _p, _p, _p],
     "gbt_gpair_fused": [_i, _p, _p, _p, _f, _i64, _p, _p, _p],
     "gbt_quantize": [_p, _i64, _d, _d, _p, _p],
     "gbt_hist_csr": [_p, _p, _p, _p, _p, _i, _p, _i, _p],

@@ -9,7 +9,7 @@ SRC = os.path.join(HERE, "cpp")
 OUT = os.path.join(HERE, "libgbt_hip.so")
 
 SOURCES = ["hist.hip", "partition.hip", "evaluate.hip", "compress.hip",
-           "predict.hip", "shap.hip", "driver.hip", "csr.hip", "gpair.hip"]
+           "predict.hip", "shap.hip", "driver.hip", "csr.hip", "gpair.hip", "mt_evaluate.hip"]
 
 
 def build(force: bool = False) -> str:
