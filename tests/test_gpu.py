@@ -389,3 +389,28 @@ def test_fused_gpair_path_matches_torch():
     xgb.train(params, d3, 5, evals=[(d3, "t")], evals_result=res3,
               verbose_eval=False)
     assert res3["t"]["auc"][-1] > 0.9
+
+
+def test_mt_gpu_eval_matches_cpu():
+    """Vector-leaf trees: the MT evaluation kernel must agree with the
+    numpy multi-target evaluator."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(20000, 8).astype(np.float32)
+    W = rng.randn(8, 3)
+    Y = (X @ W + 0.1 * rng.randn(20000, 3)).astype(np.float32)
+    params = {"objective": "reg:squarederror", "max_depth": 5,
+              "multi_strategy": "multi_output_tree", "eta": 0.3,
+              "max_bin": 128}
+    b_cpu = xgb.train(dict(params, device="cpu"),
+                      xgb.DMatrix(X, label=Y), 5, verbose_eval=False)
+    b_gpu = xgb.train(dict(params, device="cuda"),
+                      xgb.DMatrix(X, label=Y), 5, verbose_eval=False)
+    for tc, tg in zip(b_cpu.trees, b_gpu.trees):
+        assert tc.n_nodes == tg.n_nodes
+        assert np.array_equal(tc.split_index[:tc.n_nodes],
+                              tg.split_index[:tg.n_nodes])
+        assert np.allclose(tc.leaf_values[:tc.n_nodes],
+                           tg.leaf_values[:tg.n_nodes], rtol=1e-4, atol=1e-5)
+    p1 = b_cpu.predict(xgb.DMatrix(X))
+    p2 = b_gpu.predict(xgb.DMatrix(X))
+    assert np.allclose(p1, p2, atol=1e-4)
