@@ -20,6 +20,19 @@ from . import collective  # noqa: F401
 
 __version__ = "0.1.0"
 
+
+def build_info():
+    """Build/runtime info (reference: xgboost.build_info)."""
+    import torch
+    from . import ops as _ops
+    return {
+        "version": __version__,
+        "rocm_torch": torch.version.hip or "",
+        "gfx_arch": "gfx950",
+        "native_kernels": _ops.available(),
+        "USE_RCCL": True,
+    }
+
 __all__ = [
     "Booster", "DMatrix", "QuantileDMatrix", "train", "cv", "callback",
     "collective", "config_context", "set_config", "get_config",

@@ -724,6 +724,28 @@ class Booster:
         self.tparam = make_train_param(self.raw_params)
         self.device = _resolve_device(self.raw_params)
 
+    def save_config(self) -> str:
+        """JSON dump of the effective configuration (reference
+        Learner::SaveConfig, learner.cc:630)."""
+        import dataclasses as dc
+        cfg = {
+            "learner": {
+                "generic_param": {"device": str(self.device),
+                                  "seed": str(self.seed)},
+                "gradient_booster": {"name": self.booster_kind,
+                                     "tree_train_param": {
+                                         k: str(v) for k, v in
+                                         dc.asdict(self.tparam).items()
+                                         if v is not None}},
+                "learner_train_param": {
+                    "booster": self.booster_kind,
+                    "objective": self.objective.name},
+                "objective": self.objective.save_config(),
+            },
+            "version": list(VERSION),
+        }
+        return json.dumps(cfg)
+
     # -- model IO ------------------------------------------------------
     def save_model(self, fname: str) -> None:
         j = self._model_to_json()

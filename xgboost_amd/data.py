@@ -343,6 +343,38 @@ class DMatrix:
     def feature_types(self):
         return self.info.feature_types
 
+    def save_binary(self, fname: str, silent: bool = True) -> None:
+        """Binary DMatrix cache (reference XGDMatrixSaveBinary); ours is
+        an npz with the dense matrix + metadata."""
+        payload = {"X": self.raw_data(), "missing": self.missing}
+        for k in ("labels", "weights", "base_margin", "group_ptr"):
+            v = getattr(self.info, k)
+            if v is not None:
+                payload[k] = v
+        if self.info.feature_names:
+            payload["feature_names"] = np.asarray(self.info.feature_names)
+        if self.info.feature_types:
+            payload["feature_types"] = np.asarray(self.info.feature_types)
+        np.savez_compressed(fname, **payload)
+
+    @classmethod
+    def load_binary(cls, fname: str) -> "DMatrix":
+        z = np.load(fname, allow_pickle=False)
+        d = cls(z["X"], missing=float(z["missing"]))
+        if "labels" in z:
+            d.info.labels = z["labels"]
+        if "weights" in z:
+            d.info.weights = z["weights"]
+        if "base_margin" in z:
+            d.info.base_margin = z["base_margin"]
+        if "group_ptr" in z:
+            d.info.group_ptr = z["group_ptr"]
+        if "feature_names" in z:
+            d.info.feature_names = [str(x) for x in z["feature_names"]]
+        if "feature_types" in z:
+            d.info.feature_types = [str(x) for x in z["feature_types"]]
+        return d
+
     def slice(self, rindex: Sequence[int]) -> "DMatrix":
         idx = np.asarray(rindex, dtype=np.int64)
         out = DMatrix(self._data[idx], missing=self.missing,
