@@ -339,11 +339,18 @@ class Booster:
         self.tree_info.append(0)
         self.iteration_indptr.append(self.iteration_indptr[-1] + 1)
         leaf_np = tree.split_cond[:tree.n_nodes].copy()
-        if hasattr(ops, "stager"):
+        if hasattr(ops, "stager") and hasattr(lib, "gbt_margin_add") \
+                and positions.device == margin.device:
             (leaf_vals,) = ops.stager.upload([leaf_np])
+            mc = margin.contiguous()
+            lib.gbt_margin_add(hip_ops.ptr(mc), hip_ops.ptr(positions),
+                               hip_ops.ptr(leaf_vals), n,
+                               mc.stride(0), 0, hip_ops.stream())
+            if mc.data_ptr() != margin.data_ptr():
+                margin.copy_(mc)
         else:
             leaf_vals = torch.as_tensor(leaf_np, device=margin.device)
-        margin[:, 0] += leaf_vals[positions.to(margin.device).long()]
+            margin[:, 0] += leaf_vals[positions.to(margin.device).long()]
         self._cache[key] = (margin, len(self.trees))
         if self.tparam.debug_synchronize:
             collective.check_synchronized(

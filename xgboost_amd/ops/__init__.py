@@ -42,6 +42,7 @@ _SIGS = {
                         _d, _d, _d, _d, _p, _p, _p, _p, _p],
     "gbt_gpair_fused": [_i, _p, _p, _p, _f, _i64, _p, _p, _p],
     "gbt_quantize": [_p, _i64, _d, _d, _p, _p],
+    "gbt_margin_add": [_p, _p, _p, _i64, _i, _i, _p],
     "gbt_hist_csr": [_p, _p, _p, _p, _p, _i, _p, _i, _p],
     "gbt_partition_csr": [_p, _p, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p],
     # native level-loop driver

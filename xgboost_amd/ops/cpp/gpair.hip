@@ -105,3 +105,24 @@ void gbt_quantize(const float* gh, long long n, double g_scale,
 }
 
 }  // extern "C"
+
+namespace {
+
+__global__ __launch_bounds__(256) void MarginAddKernel(
+    float* __restrict__ margin, const int32_t* __restrict__ pos,
+    const float* __restrict__ leaf_vals, long long n, int stride, int col) {
+  const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (long long i = i0; i < n; i += (long long)gridDim.x * blockDim.x) {
+    margin[i * stride + col] += leaf_vals[pos[i]];
+  }
+}
+
+}  // namespace
+
+extern "C" void gbt_margin_add(float* margin, const int32_t* pos,
+                               const float* leaf_vals, long long n,
+                               int stride, int col, hipStream_t stream) {
+  const int blocks = (int)std::min<long long>((n + 255) / 256, 4096);
+  hipLaunchKernelGGL(MarginAddKernel, dim3(blocks), dim3(256), 0, stream,
+                     margin, pos, leaf_vals, n, stride, col);
+}

@@ -6,7 +6,8 @@
 //    group (160 KiB LDS/CU lets one group hold up to ~9K bins at
 //    16 B/bin; Higgs 28x256=7168 bins fits in a single group, so the
 //    entire node histogram lives in LDS and is flushed once).
-//  - 64-wide wavefronts; block = 256 threads = 4 waves.
+//  - 64-wide wavefronts; 1024-thread blocks (16 waves/CU at the
+//    1-block LDS occupancy) hide ds_add latency.
 //  - int32 quantized gradients accumulated into int64 bins via
 //    atomicAdd(u64) — ds_add_u64 on LDS, global_atomic_add_u64 on HBM:
 //    deterministic regardless of ordering (two's complement wraparound

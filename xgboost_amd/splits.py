@@ -185,7 +185,6 @@ def evaluate_splits_np(hist_q: np.ndarray,
 
     best = [SplitEntry(nid=int(nid), g_scale=g_scale, h_scale=h_scale)
             for nid in nids]
-    is_last_bin = np.arange(n_bins) == seg_end
 
     mono_bins = monotone[feat_of_bin][None, :] if monotone is not None else None
 
@@ -327,7 +326,6 @@ def evaluate_splits_multi_np(hists: np.ndarray,
     feat_of_bin = np.repeat(np.arange(n_features), widths)
     seg_start = np.repeat(cut_ptrs[:-1], widths)
     seg_end = np.repeat(cut_ptrs[1:] - 1, widths)
-    is_last_bin = np.arange(n_bins) == seg_end
 
     inv_g = (1.0 / g_scales).reshape(T, 1, 1)
     inv_h = (1.0 / h_scales).reshape(T, 1, 1)
