@@ -28,11 +28,13 @@ MIN_ROWS_PER_TASK = 1024
 
 def _chunk_tasks(segments: Sequence[Tuple[int, int]],
                  slots: Optional[Sequence[int]] = None,
-                 target_tasks: int = TARGET_HIST_TASKS) -> np.ndarray:
+                 target_tasks: int = TARGET_HIST_TASKS,
+                 min_rows: int = MIN_ROWS_PER_TASK) -> np.ndarray:
     """Build BlockTask array [(slot, row_begin, row_end, 0)] chunked so the
-    grid fills the chip (>=2048 blocks when rows allow)."""
+    grid fills the chip; hist callers pass fewer/bigger tasks so the LDS
+    zero+flush cost amortizes."""
     total = sum(e - s for s, e in segments)
-    rows_per_task = max(MIN_ROWS_PER_TASK,
+    rows_per_task = max(min_rows,
                         (total + target_tasks - 1) // max(target_tasks, 1))
     tasks = []
     for i, (s, e) in enumerate(segments):
@@ -187,7 +189,7 @@ class GpuOps(SegmentedOpsMixin):
         k = len(segments)
         out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64,
                           device=self.device)
-        tasks_np = _chunk_tasks(segments)
+        tasks_np = _chunk_tasks(segments, target_tasks=512)
         (tasks,) = self.stager.upload([tasks_np])
         p8, p16 = self._gidx_ptrs()
         self.lib.gbt_hist(

@@ -15,6 +15,7 @@
 
 #include <algorithm>
 #include <cmath>
+#include <cstdlib>
 #include <cstring>
 #include <vector>
 
@@ -103,11 +104,13 @@ double CalcWeight(double g, double h, const HostParams& p) {
   return w;
 }
 
-void ChunkTasks(const std::vector<Node*>& nodes, std::vector<BlockTask>* out) {
+void ChunkTasks(const std::vector<Node*>& nodes, std::vector<BlockTask>* out,
+                long long min_rows = 1024, long long target_tasks = 2048) {
   out->clear();
   long long total = 0;
   for (auto* n : nodes) total += n->seg_end - n->seg_begin;
-  long long rows_per_task = std::max<long long>(1024, (total + 2047) / 2048);
+  long long rows_per_task =
+      std::max<long long>(min_rows, (total + target_tasks - 1) / target_tasks);
   for (size_t i = 0; i < nodes.size(); ++i) {
     int b = nodes[i]->seg_begin;
     while (b < nodes[i]->seg_end) {
@@ -207,9 +210,17 @@ int gbt_grow_tree(
   out_base_weight[0] = (float)CalcWeight(root_gq * inv_g, root_hq * inv_h, p);
   out_sum_hess[0] = (float)(root_hq * inv_h);
 
+  // hist wants FEWER, BIGGER tasks than partition: every block pays a
+  // full LDS zero+flush of the group histogram, so rows/task must
+  // amortize ~2*group_bins atomics (env GBT_HIST_TASKS to tune)
+  static long long hist_tasks = [] {
+    const char* e = getenv("GBT_HIST_TASKS");
+    long long v = e ? atoll(e) : 512;
+    return v >= 64 && v <= 16384 ? v : 512;
+  }();
   auto build_hists = [&](std::vector<Node*>& nodes, int64_t* pool) -> int {
     std::vector<BlockTask> tasks;
-    ChunkTasks(nodes, &tasks);
+    ChunkTasks(nodes, &tasks, 2048, hist_tasks);
     const int slot = ctx->ring.next();
     size_t bytes = tasks.size() * sizeof(BlockTask);
     if (int e = ctx->ring.ensure(slot, bytes)) return e;
