@@ -13,7 +13,7 @@ histograms are bit-identical regardless of accumulation order.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Sequence, Tuple
 
 import numpy as np
 import torch

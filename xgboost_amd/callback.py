@@ -8,8 +8,6 @@ from __future__ import annotations
 import os
 from typing import Dict, List, Optional, Sequence
 
-import numpy as np
-
 from . import collective
 
 

@@ -15,7 +15,6 @@ from __future__ import annotations
 import os
 from typing import List, Optional, Sequence
 
-import numpy as np
 import torch
 import torch.distributed as dist
 

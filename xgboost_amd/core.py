@@ -11,20 +11,19 @@ dump_model/get_score/attributes/copy/slice.
 from __future__ import annotations
 
 import json
-import math
 import os
-from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch
 
 from . import collective
 from .backend.cpu import CpuOps, GradQuantizer
-from .data import DMatrix, QuantizedMatrix
+from .data import DMatrix
 from .grower import TreeGrower
 from .metrics import create_metric
 from .objectives import Objective, create_objective
-from .params import (LEARNER_PARAMS, TrainParam, canonicalize, check_unknown,
+from .params import (TrainParam, canonicalize, check_unknown,
                      make_train_param)
 from .tree_model import RegTree
 

@@ -15,7 +15,7 @@ histogram kernel reads 1-2 bytes per (row, feature).
 from __future__ import annotations
 
 import dataclasses
-from typing import Any, Dict, List, Optional, Sequence, Union
+from typing import Any, Dict, List, Optional, Sequence
 
 import numpy as np
 import torch

@@ -13,14 +13,14 @@ hipMemcpyAsync on torch's stream each sweep.
 """
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Callable, Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
 
 from .data import DMatrix, MetaInfo, QuantizedMatrix, quantize_dense
 from .quantile import HistogramCuts
-from .sketch import cuts_from_summaries, sketch_cuts_batches, summarize_batch
+from .sketch import sketch_cuts_batches, summarize_batch
 
 
 class DataIter:

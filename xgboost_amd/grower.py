@@ -18,7 +18,6 @@ from typing import Dict, List, Optional, Sequence, Tuple
 import numpy as np
 import torch
 
-from . import collective
 from .backend.cpu import GradQuantizer
 from .params import TrainParam
 from .splits import SplitEntry, calc_weight

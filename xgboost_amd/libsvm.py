@@ -4,7 +4,7 @@ still accepted — data.cc:930).  Supports 'label idx:val ...' lines and
 optional 'qid:' tokens."""
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import numpy as np
 

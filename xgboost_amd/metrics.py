@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import math
 import re
-from typing import Callable, Dict, Optional, Tuple
+from typing import Callable, Dict, Tuple
 
 import numpy as np
 

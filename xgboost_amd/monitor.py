@@ -9,7 +9,7 @@ import atexit
 import ctypes
 import time
 from collections import defaultdict
-from typing import Dict, Optional
+from typing import Dict
 
 from .config import verbosity
 

@@ -8,7 +8,7 @@ reference's documented semantics.
 from __future__ import annotations
 
 import dataclasses
-from typing import Any, Dict, List, Optional, Sequence, Union
+from typing import Any, Dict, List, Optional, Sequence
 
 ALIASES = {
     "reg_lambda": "lambda",

@@ -13,7 +13,7 @@ last column the bias (expected value incl. base_score); interactions
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 

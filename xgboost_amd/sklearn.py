@@ -5,13 +5,13 @@ XGBClassifier :1728, XGBRegressor :2023, XGBRanker :2165, RF variants).
 """
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, Optional, Sequence, Tuple, Union
 
 import numpy as np
 
-from .callback import EarlyStopping, TrainingCallback
+from .callback import TrainingCallback
 from .core import Booster
-from .data import DMatrix, QuantileDMatrix
+from .data import DMatrix
 from .training import train as _train
 
 _PARAM_NAMES = [

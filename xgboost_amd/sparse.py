@@ -16,13 +16,12 @@ is shared unchanged.
 from __future__ import annotations
 
 import dataclasses
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Tuple
 
 import numpy as np
 import torch
 
-from .quantile import HistogramCuts, make_cuts
-from .params import TrainParam
+from .quantile import HistogramCuts
 from .splits import SplitEntry, evaluate_splits_np
 
 

@@ -10,7 +10,6 @@ arrays — that is the layout the HIP predictor kernel consumes directly.
 """
 from __future__ import annotations
 
-import dataclasses
 from typing import Dict, List, Optional
 
 import numpy as np

@@ -7,8 +7,6 @@ Used by process_type=update (reference gbtree.cc updater config).
 """
 from __future__ import annotations
 
-from typing import Tuple
-
 import numpy as np
 
 from .params import TrainParam
