@@ -912,9 +912,25 @@ class Booster:
         return self[slice(begin, end or self.num_boosted_rounds(), step)]
 
     # -- introspection -------------------------------------------------
+    def _fmap_names(self, fmap: str) -> Optional[List[str]]:
+        """Parse a featmap.txt file (reference format:
+        '<index>\t<name>\t<type>' per line)."""
+        if not fmap or not os.path.exists(fmap):
+            return self.feature_names
+        names: Dict[int, str] = {}
+        with open(fmap) as fh:
+            for line in fh:
+                parts = line.strip().split("\t")
+                if len(parts) >= 2:
+                    names[int(parts[0])] = parts[1]
+        if not names:
+            return self.feature_names
+        n = max(names) + 1
+        return [names.get(i, f"f{i}") for i in range(n)]
+
     def get_dump(self, fmap: str = "", with_stats: bool = False,
                  dump_format: str = "text") -> List[str]:
-        names = self.feature_names
+        names = self._fmap_names(fmap)
         return [t.dump(names, with_stats, dump_format) for t in self.trees]
 
     def dump_model(self, fout, fmap: str = "", with_stats: bool = False,
@@ -943,7 +959,7 @@ class Booster:
                 v = (float(tree.loss_chg[nid]) if "gain" in importance_type
                      else float(tree.sum_hess[nid]))
                 sums[f] = sums.get(f, 0.0) + v
-        names = self.feature_names
+        names = self._fmap_names(fmap)
 
         def fname(f):
             return names[f] if names and f < len(names) else f"f{f}"
