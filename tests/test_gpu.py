@@ -414,3 +414,49 @@ def test_mt_gpu_eval_matches_cpu():
     p1 = b_cpu.predict(xgb.DMatrix(X))
     p2 = b_gpu.predict(xgb.DMatrix(X))
     assert np.allclose(p1, p2, atol=1e-4)
+
+
+def test_extmem_streamed_pages():
+    """Force the beyond-HBM streaming path (_StreamedPage) and verify
+    it trains identically to the device-cached path."""
+    from xgboost_amd.extmem import DataIter, ExtMemQuantileDMatrix, ExtMemOps
+    from xgboost_amd.core import Booster
+
+    class It(DataIter):
+        def __init__(self):
+            super().__init__()
+            self.i = 0
+
+        def reset(self):
+            self.i = 0
+
+        def next(self, input_data):
+            if self.i >= 3:
+                return False
+            rng = np.random.RandomState(self.i)
+            Xb = rng.randn(5000, 8).astype(np.float32)
+            yb = (Xb[:, 0] > 0).astype(np.float32)
+            input_data(data=Xb, label=yb)
+            self.i += 1
+            return True
+
+    d1 = ExtMemQuantileDMatrix(It(), max_bin=64)
+    b1 = Booster({"objective": "binary:logistic", "max_depth": 4,
+                  "device": "cuda", "max_bin": 64}, cache=[d1])
+    # force every page through the streamed path (budget 0)
+    b1._ops_cache[id(d1)] = ExtMemOps(d1, torch.device("cuda"),
+                                      device_cache_bytes=0)
+    from xgboost_amd.extmem import _StreamedPage
+    assert all(isinstance(p, _StreamedPage)
+               for p in b1._ops_cache[id(d1)].page_ops)
+    for i in range(5):
+        b1.update(d1, i)
+    d2 = ExtMemQuantileDMatrix(It(), max_bin=64)
+    b2 = Booster({"objective": "binary:logistic", "max_depth": 4,
+                  "device": "cuda", "max_bin": 64}, cache=[d2])
+    for i in range(5):
+        b2.update(d2, i)
+    for t1, t2 in zip(b1.trees, b2.trees):
+        assert t1.n_nodes == t2.n_nodes
+        assert np.array_equal(t1.split_index[:t1.n_nodes],
+                              t2.split_index[:t2.n_nodes])

@@ -405,8 +405,8 @@ def aft_nloglik(preds, info, param=None):
     logp = torch.log(torch.clamp(p, min=eps))
     z_lo = (torch.log(torch.clamp(lo, min=eps)) - logp) / s
     z_hi = (torch.log(torch.clamp(hi, min=eps)) - logp) / s
-    pdf_l, cdf_l, _ = obj._pdf_cdf(z_lo)
-    pdf_u, cdf_u, _ = obj._pdf_cdf(z_hi)
+    pdf_l, cdf_l, _, _ = obj._pdf_cdf(z_lo)
+    pdf_u, cdf_u, _, _ = obj._pdf_cdf(z_hi)
     uncensored = torch.isfinite(hi) & (lo == hi)
     cdf_u = torch.where(torch.isfinite(hi), cdf_u, torch.ones_like(cdf_u))
     cdf_l = torch.where(lo > 0, cdf_l, torch.zeros_like(cdf_l))

@@ -509,21 +509,26 @@ class AFT(Objective):
         return lo_t, hi_t
 
     def _pdf_cdf(self, z):
+        """Returns (pdf, cdf, dlogpdf/dz, -d2logpdf/dz2) — closed forms
+        per distribution (reference probability_distribution.h:35)."""
         if self.dist == "normal":
             pdf = torch.exp(-0.5 * z * z) / math.sqrt(2 * math.pi)
             cdf = 0.5 * (1 + torch.erf(z / math.sqrt(2)))
-            grad_pdf = -z  # d log pdf/dz
+            grad_pdf = -z
+            curv = torch.ones_like(z)
         elif self.dist == "logistic":
             ez = torch.exp(z)
             pdf = ez / (1 + ez) ** 2
             cdf = ez / (1 + ez)
             grad_pdf = (1 - ez) / (1 + ez)
+            curv = 2 * ez / (1 + ez) ** 2
         else:  # extreme (Gumbel)
             ez = torch.exp(z)
             pdf = ez * torch.exp(-ez)
             cdf = 1 - torch.exp(-ez)
             grad_pdf = 1 - ez
-        return pdf, cdf, grad_pdf
+            curv = ez
+        return pdf, cdf, grad_pdf, curv
 
     def get_gradient(self, preds, info, it):
         lo, hi = self._bounds(info, preds.device)
@@ -533,15 +538,15 @@ class AFT(Objective):
         z_lo = (torch.log(torch.clamp(lo, min=eps)) - p) / s
         z_hi = (torch.log(torch.clamp(hi, min=eps)) - p) / s
         uncensored = torch.isfinite(hi) & (lo == hi)
-        pdf_l, cdf_l, glp_l = self._pdf_cdf(z_lo)
-        pdf_u, cdf_u, _ = self._pdf_cdf(z_hi)
+        pdf_l, cdf_l, glp_l, curv_l = self._pdf_cdf(z_lo)
+        pdf_u, cdf_u, _, _ = self._pdf_cdf(z_hi)
         cdf_u = torch.where(torch.isfinite(hi), cdf_u, torch.ones_like(cdf_u))
         pdf_u = torch.where(torch.isfinite(hi), pdf_u, torch.zeros_like(pdf_u))
         cdf_l = torch.where(lo > 0, cdf_l, torch.zeros_like(cdf_l))
         pdf_l = torch.where(lo > 0, pdf_l, torch.zeros_like(pdf_l))
         # uncensored: -log pdf(z)/ (s t); censored: -log(cdf_u - cdf_l)
         g_unc = glp_l / s  # d/dp of -log pdf(z_lo): -(dlogpdf/dz)(dz/dp)= glp/s
-        h_unc = torch.ones_like(p) / (s * s)  # upper-bounded curvature
+        h_unc = torch.clamp(curv_l, min=1e-6) / (s * s)
         denom = torch.clamp(cdf_u - cdf_l, min=eps)
         g_cen = (pdf_u - pdf_l) / (s * denom)
         h_cen = torch.clamp(g_cen * g_cen, min=1e-16) + 1e-6
