@@ -190,3 +190,38 @@ def test_categorical_partition_split():
     b2 = xgb.Booster()
     b2.load_model(bytes(raw))
     assert np.allclose(b2.predict(d), p, atol=1e-6)
+
+
+def test_native_cpu_hist_matches_torch_oracle():
+    """The C (OpenMP) CPU kernels must match the pure-torch oracle
+    bit-for-bit (same int64 fixed-point scheme)."""
+    rng = np.random.RandomState(3)
+    X = rng.randn(5000, 7).astype(np.float32)
+    X[rng.rand(5000, 7) < 0.1] = np.nan
+    d = DMatrix(X)
+    qm = d.quantized(64)
+    native = CpuOps(qm, use_native=True)
+    oracle = CpuOps(qm, use_native=False)
+    if native.lib is None:
+        pytest.skip("native CPU kernels unavailable")
+    gpair = torch.tensor(np.stack([rng.randn(5000),
+                                   rng.rand(5000) + 0.1],
+                                  axis=1).astype(np.float32))
+    quant = GradQuantizer(gpair)
+    qg = quant.quantize(gpair)
+    segs = [(0, 2000), (2000, 5000)]
+    h1 = native.build_hist(qg, native.make_ridx(5000), segs)
+    h2 = oracle.build_hist(qg, oracle.make_ridx(5000), segs)
+    assert torch.equal(h1, h2)
+    # partition equality (stable on both)
+    param = make_train_param({"max_depth": 4})
+    s = qg.to(torch.int64).sum(0)
+    splits = oracle.evaluate_splits(h2[:1] * 0 + oracle.build_hist(
+        qg, oracle.make_ridx(5000), [(0, 5000)]), quant,
+        [(int(s[0]), int(s[1]))], [0], param)
+    r1 = native.make_ridx(5000)
+    r2 = oracle.make_ridx(5000)
+    seg1 = native.partition(r1, [(0, 5000)], splits)
+    seg2 = oracle.partition(r2, [(0, 5000)], splits)
+    assert seg1 == seg2
+    assert torch.equal(r1, r2)  # both stable -> identical order

@@ -97,14 +97,42 @@ class SegmentedOpsMixin:
 
 
 class CpuOps(SegmentedOpsMixin):
-    """Tree-building primitive ops on CPU tensors."""
+    """Tree-building primitive ops on CPU tensors.
+
+    Two implementations behind one interface:
+    - native C (OpenMP) kernels in libgbt_hip.so (gbt_hist_cpu /
+      gbt_partition_cpu) — the production CPU path, same int64
+      fixed-point scheme => bit-identical results;
+    - pure torch/numpy (`use_native=False`) — the independent oracle the
+      HIP and C kernels are tested against.
+    """
 
     device = torch.device("cpu")
 
-    def __init__(self, qm: QuantizedMatrix):
+    def __init__(self, qm: QuantizedMatrix, use_native: bool = True):
         self.qm = qm
         self.gidx_global = qm.global_gidx()  # int64 [n, f], -1 missing
         self.n_bins = qm.cuts.total_bins
+        self.lib = None
+        if use_native:
+            try:
+                from .. import ops as hip_ops
+                lib = hip_ops.load()
+                if hasattr(lib, "gbt_hist_cpu"):
+                    self.lib = lib
+                    self.hip = hip_ops
+            except RuntimeError:
+                pass
+        self._gidx_np = np.ascontiguousarray(qm.gidx.numpy())
+        self._cut_ptrs_np = np.ascontiguousarray(qm.cuts.ptrs, np.int32)
+        self._scratch = None
+
+    def _gidx_c_ptrs(self):
+        import ctypes
+        p = self._gidx_np.ctypes.data_as(ctypes.c_void_p)
+        if self._gidx_np.dtype == np.uint8:
+            return p, None
+        return None, p
 
     def make_ridx(self, n_rows: int) -> torch.Tensor:
         return torch.arange(n_rows, dtype=torch.int64)
@@ -120,6 +148,23 @@ class CpuOps(SegmentedOpsMixin):
         """-> int64 [len(segments), n_bins, 2]."""
         k = len(segments)
         out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64)
+        if self.lib is not None:
+            import ctypes
+            q_np = np.ascontiguousarray(qgpair.cpu().numpy(), np.int32)
+            r_np = np.ascontiguousarray(ridx.numpy(), np.int64)
+            sb = np.ascontiguousarray([s for s, _ in segments], np.int64)
+            se = np.ascontiguousarray([e for _, e in segments], np.int64)
+            o_np = out.numpy()
+            p8, p16 = self._gidx_c_ptrs()
+            self.lib.gbt_hist_cpu(
+                p8, p16, self.qm.n_features,
+                q_np.ctypes.data_as(ctypes.c_void_p),
+                r_np.ctypes.data_as(ctypes.c_void_p),
+                sb.ctypes.data_as(ctypes.c_void_p),
+                se.ctypes.data_as(ctypes.c_void_p), k,
+                self._cut_ptrs_np.ctypes.data_as(ctypes.c_void_p),
+                o_np.ctypes.data_as(ctypes.c_void_p), self.n_bins)
+            return out
         for i, (s, e) in enumerate(segments):
             rows = ridx[s:e]
             g = self.gidx_global[rows]            # [m, f]
@@ -156,6 +201,8 @@ class CpuOps(SegmentedOpsMixin):
         Returns [(left_seg, right_seg), ...].  Stable on CPU (the GPU
         kernel is unstable; histogram sums don't depend on order).
         """
+        if self.lib is not None:
+            return self._partition_native(ridx, segments, splits)
         out = []
         for (s, e), sp in zip(segments, splits):
             rows = ridx[s:e]
@@ -178,6 +225,40 @@ class CpuOps(SegmentedOpsMixin):
             ridx[s:s + nl] = left_rows
             ridx[s + nl:e] = right_rows
             out.append(((s, s + nl), (s + nl, e)))
+        return out
+
+    def _partition_native(self, ridx, segments, splits):
+        import ctypes
+        r_np = np.ascontiguousarray(ridx.numpy(), np.int64)
+        assert r_np.base is not None or r_np.ctypes.data == \
+            ridx.data_ptr(), "ridx must share memory"
+        max_seg = max((e - s for s, e in segments), default=0)
+        if self._scratch is None or self._scratch.size < max_seg:
+            self._scratch = np.empty(max(max_seg, 1024), np.int64)
+        cuts = self.qm.cuts
+        out = []
+        p8, p16 = self._gidx_c_ptrs()
+        for (s, e), sp in zip(segments, splits):
+            fbins = int(cuts.ptrs[sp.feature + 1] - cuts.ptrs[sp.feature])
+            if sp.is_cat:
+                nw = (fbins + 31) // 32
+                w = np.zeros(nw, np.uint32)
+                for c in sp.cat_bits:
+                    w[c >> 5] |= np.uint32(1 << (c & 31))
+                cat_ptr = w.ctypes.data_as(ctypes.c_void_p)
+                cat_words = nw
+                sbin = -1
+            else:
+                cat_ptr = None
+                cat_words = 0
+                sbin = sp.split_bin - int(cuts.ptrs[sp.feature])
+            nl = self.lib.gbt_partition_cpu(
+                p8, p16, self.qm.n_features,
+                r_np.ctypes.data_as(ctypes.c_void_p), s, e,
+                sp.feature, sbin, 1 if sp.default_left else 0,
+                cat_ptr, cat_words, fbins,
+                self._scratch.ctypes.data_as(ctypes.c_void_p))
+            out.append(((s, s + int(nl)), (s + int(nl), e)))
         return out
 
     def leaf_partition(self, ridx: torch.Tensor,

@@ -43,6 +43,9 @@ _SIGS = {
     "gbt_gpair_fused": [_i, _p, _p, _p, _f, _i64, _p, _p, _p],
     "gbt_quantize": [_p, _i64, _d, _d, _p, _p],
     "gbt_margin_add": [_p, _p, _p, _i64, _i, _i, _p],
+    "gbt_hist_cpu": [_p, _p, _i, _p, _p, _p, _p, _i, _p, _p, _i],
+    "gbt_partition_cpu": [_p, _p, _i, _p, _i64, _i64, _i, _i, _i, _p, _i,
+                          _i, _p],
     "gbt_hist_csr": [_p, _p, _p, _p, _p, _i, _p, _i, _p],
     "gbt_partition_csr": [_p, _p, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p],
     # native level-loop driver
@@ -93,6 +96,8 @@ def load() -> ctypes.CDLL:
     if hasattr(_lib, "gbt_driver_create"):
         _lib.gbt_driver_create.restype = _p
         _lib.gbt_grow_tree.restype = _c.c_int
+    if hasattr(_lib, "gbt_partition_cpu"):
+        _lib.gbt_partition_cpu.restype = _c.c_longlong
     return _lib
 
 

@@ -9,7 +9,7 @@ SRC = os.path.join(HERE, "cpp")
 OUT = os.path.join(HERE, "libgbt_hip.so")
 
 SOURCES = ["hist.hip", "partition.hip", "evaluate.hip", "compress.hip",
-           "predict.hip", "shap.hip", "driver.hip", "csr.hip", "gpair.hip", "mt_evaluate.hip"]
+           "predict.hip", "shap.hip", "driver.hip", "csr.hip", "gpair.hip", "mt_evaluate.hip", "cpu_hist.cpp"]
 
 
 def build(force: bool = False) -> str:
@@ -23,6 +23,7 @@ def build(force: bool = False) -> str:
     hipcc = os.environ.get("HIPCC", "hipcc")
     cmd = [hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17",
            "-ffp-contract=off",  # bit-match the numpy fp64 oracle
+           "-fopenmp",           # native CPU hist/partition (cpu_hist.cpp)
            "-shared", "-fPIC", "-o", OUT] + sources
     print("+", " ".join(cmd), flush=True)
     subprocess.check_call(cmd)
