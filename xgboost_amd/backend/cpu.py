@@ -185,12 +185,67 @@ class CpuOps(SegmentedOpsMixin):
                         nids: Sequence[int], param: TrainParam,
                         feature_sets=None, monotone=None, cat_mask=None,
                         node_bounds=None) -> List[SplitEntry]:
+        if (self.lib is not None and cat_mask is None
+                and hasattr(self.lib, "gbt_evaluate_cpu")):
+            return self._evaluate_native(hist, quantizer, parent_sums, nids,
+                                         param, feature_sets, monotone,
+                                         node_bounds)
         # parent_sums are exact int64 (gq, hq) pairs
         return evaluate_splits_np(hist.cpu().numpy(), parent_sums,
                                   quantizer.g_scale, quantizer.h_scale,
                                   nids, self.qm.cuts.ptrs, param,
                                   feature_sets=feature_sets, monotone=monotone,
                                   cat_mask=cat_mask, node_bounds=node_bounds)
+
+    def _evaluate_native(self, hist, quantizer, parent_sums, nids, param,
+                         feature_sets, monotone, node_bounds):
+        """C evaluator (gbt_evaluate_cpu): same semantics/tie rules as
+        numpy and the HIP kernel."""
+        import ctypes
+
+        def cp(a):
+            return a.ctypes.data_as(ctypes.c_void_p) if a is not None else None
+
+        k = len(nids)
+        f = self.qm.n_features
+        h_np = np.ascontiguousarray(hist.numpy())
+        parents = np.ascontiguousarray(parent_sums, np.int64).reshape(k, 2)
+        mono = (np.ascontiguousarray(monotone, np.int8)
+                if monotone is not None else None)
+        bounds = (np.ascontiguousarray(node_bounds, np.float64)
+                  if node_bounds is not None else None)
+        mask = None
+        if feature_sets is not None and any(
+                fs is not None for fs in feature_sets):
+            mask = np.zeros((k, f), np.uint8)
+            for i, fs in enumerate(feature_sets):
+                if fs is None:
+                    mask[i] = 1
+                else:
+                    mask[i, np.asarray(fs, np.int64)] = 1
+        out = np.zeros((k, 6), np.int64)
+        self.lib.gbt_evaluate_cpu(
+            cp(h_np), k, self.n_bins, f, cp(self._cut_ptrs_np), cp(parents),
+            quantizer.g_scale, quantizer.h_scale, param.reg_lambda,
+            param.reg_alpha, param.max_delta_step, param.min_child_weight,
+            cp(mono), cp(bounds), cp(mask), cp(out))
+        entries = []
+        for i, nid in enumerate(nids):
+            e = SplitEntry(nid=int(nid), g_scale=quantizer.g_scale,
+                           h_scale=quantizer.h_scale)
+            if out[i, 1] >= 0 and out[i, 5] >= 0:
+                gain = out[i, 0:1].view(np.float64)[0]
+                if np.isfinite(gain):
+                    e.gain = float(gain)
+                    e.split_bin = int(out[i, 1])
+                    e.default_left = bool(out[i, 2])
+                    e.left_gq = int(out[i, 3])
+                    e.left_hq = int(out[i, 4])
+                    e.feature = int(out[i, 5])
+                    e.right_gq = int(parents[i, 0]) - e.left_gq
+                    e.right_hq = int(parents[i, 1]) - e.left_hq
+            entries.append(e)
+        return entries
 
     def partition(self, ridx: torch.Tensor,
                   segments: Sequence[Tuple[int, int]],

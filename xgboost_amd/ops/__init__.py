@@ -46,6 +46,8 @@ _SIGS = {
     "gbt_hist_cpu": [_p, _p, _i, _p, _p, _p, _p, _i, _p, _p, _i],
     "gbt_partition_cpu": [_p, _p, _i, _p, _i64, _i64, _i, _i, _i, _p, _i,
                           _i, _p],
+    "gbt_evaluate_cpu": [_p, _i, _i, _i, _p, _p, _d, _d, _d, _d, _d, _d,
+                         _p, _p, _p, _p],
     "gbt_hist_csr": [_p, _p, _p, _p, _p, _i, _p, _i, _p],
     "gbt_partition_csr": [_p, _p, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p],
     # native level-loop driver
