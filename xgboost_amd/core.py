@@ -674,9 +674,9 @@ class Booster:
         metric_names = self._metric_names()
         for dmat, name in evals:
             margin = self._cached_margin(dmat)
-            transformed = self.objective.pred_transform(margin)
+            transformed = self.objective.eval_transform(margin)
             tnp = transformed.cpu().numpy()
-            if tnp.shape[1] == 1:
+            if tnp.ndim == 2 and tnp.shape[1] == 1:
                 tnp = tnp.reshape(-1)
             for mname in metric_names:
                 m = create_metric(mname)

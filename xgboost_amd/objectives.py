@@ -54,6 +54,21 @@ def _labels(info, t: torch.Tensor) -> torch.Tensor:
     return torch.as_tensor(info.labels, dtype=torch.float32, device=t.device)
 
 
+
+
+def _parse_alphas(alpha):
+    """Accept float, list, numpy array, or the stringified forms the
+    config writer emits ("0.5" / "[0.1, 0.5]")."""
+    if isinstance(alpha, str):
+        alpha = alpha.strip()
+        if alpha.startswith("["):
+            return [float(x) for x in alpha.strip("[] ").split(",") if x.strip()]
+        return [float(alpha)]
+    if isinstance(alpha, (list, tuple, np.ndarray)):
+        return [float(a) for a in alpha]
+    return [float(alpha)]
+
+
 class Objective:
     name = "base"
     n_class = 1
@@ -72,6 +87,13 @@ class Objective:
 
     def pred_transform(self, margin: torch.Tensor) -> torch.Tensor:
         return margin
+
+    def eval_transform(self, margin: torch.Tensor) -> torch.Tensor:
+        """Transform used for metric evaluation (reference
+        ObjFunction::EvalTransform — differs from PredTransform for
+        multi:softmax, which argmaxes for prediction but evaluates
+        mlogloss on probabilities)."""
+        return self.pred_transform(margin)
 
     def prob_to_margin(self, base_score: float) -> float:
         return base_score
@@ -196,10 +218,7 @@ class QuantileError(Objective):
     def __init__(self, params=None):
         super().__init__(params)
         alpha = self.params.get("quantile_alpha", 0.5)
-        if isinstance(alpha, (list, tuple, np.ndarray)):
-            self.alphas = [float(a) for a in alpha]
-        else:
-            self.alphas = [float(alpha)]
+        self.alphas = _parse_alphas(alpha)
         for a in self.alphas:
             if not 0.0 < a < 1.0:
                 raise ValueError("quantile_alpha must be in (0, 1)")
@@ -236,10 +255,7 @@ class ExpectileError(Objective):
     def __init__(self, params=None):
         super().__init__(params)
         alpha = self.params.get("expectile_alpha", 0.5)
-        if isinstance(alpha, (list, tuple, np.ndarray)):
-            self.alphas = [float(a) for a in alpha]
-        else:
-            self.alphas = [float(alpha)]
+        self.alphas = _parse_alphas(alpha)
 
     def n_outputs(self, n_targets: int = 1) -> int:
         return len(self.alphas)
@@ -424,6 +440,9 @@ class SoftmaxMulti(Objective):
         if self.output_prob:
             return torch.softmax(margin, dim=1)
         return torch.argmax(margin, dim=1).to(torch.float32)
+
+    def eval_transform(self, margin):
+        return torch.softmax(margin, dim=1)
 
     def init_estimation(self, info) -> float:
         return 0.5
