@@ -170,7 +170,10 @@ class ExtMemOps:
                     budget -= size
                     self.page_ops.append(GpuOps(qm.to(device)))
                 else:
-                    self.page_ops.append(_StreamedPage(qm, device))
+                    if not hasattr(self, "_copy_stream"):
+                        self._copy_stream = torch.cuda.Stream()
+                    self.page_ops.append(
+                        _StreamedPage(qm, device, self._copy_stream))
             else:
                 from .backend.cpu import CpuOps
                 self.page_ops.append(CpuOps(qm))
@@ -198,9 +201,15 @@ class ExtMemOps:
         s, e = self.dmat.page_offsets[i], self.dmat.page_offsets[i + 1]
         return qgpair[s:e]
 
+    def _prefetch(self, i: int) -> None:
+        if i < len(self.page_ops) and hasattr(self.page_ops[i], "prefetch"):
+            self.page_ops[i].prefetch()
+
     def build_hist_nodes(self, qgpair: torch.Tensor, nids) -> torch.Tensor:
         total = None
+        self._prefetch(0)
         for i, ops in enumerate(self.page_ops):
+            self._prefetch(i + 1)  # overlap next page's H2D with compute
             h = ops.build_hist_nodes(self._page_gpair(qgpair, i), nids)
             total = h if total is None else total + h
         return total
@@ -214,7 +223,9 @@ class ExtMemOps:
         return self.page_ops[0].evaluate_splits(*args, **kwargs)
 
     def partition_nodes(self, parents, splits, children) -> None:
-        for ops in self.page_ops:
+        self._prefetch(0)
+        for i, ops in enumerate(self.page_ops):
+            self._prefetch(i + 1)
             ops.partition_nodes(parents, splits, children)
 
     def leaf_positions(self, leaf_nids) -> torch.Tensor:
@@ -230,16 +241,40 @@ class _StreamedPage:
     """A page whose quantized matrix stays in pinned host memory and is
     copied to the device only for the op sweeps that read it
     (beyond-HBM datasets).  The small per-page state (ridx, segments)
-    stays device-resident; only the big bin matrix streams."""
+    stays device-resident; only the big bin matrix streams.  H2D runs
+    on a dedicated copy stream so the NEXT page's upload overlaps the
+    current page's kernels (reference: sparse_page_source.h prefetch
+    ring + ext-mem copy stream)."""
 
-    def __init__(self, qm: QuantizedMatrix, device: torch.device):
+    def __init__(self, qm: QuantizedMatrix, device: torch.device,
+                 copy_stream: Optional["torch.cuda.Stream"] = None):
         self.host_qm = qm
         self.device = device
         self._gpu_ops = None
+        self._copy_stream = copy_stream
+        self._pending = None
+        self._event = None
+
+    def prefetch(self) -> None:
+        """Start this page's H2D on the copy stream (non-blocking)."""
+        if self._pending is not None or self._copy_stream is None:
+            return
+        with torch.cuda.stream(self._copy_stream):
+            self._pending = self.host_qm.gidx.to(self.device,
+                                                 non_blocking=True)
+            self._event = torch.cuda.Event()
+            self._event.record(self._copy_stream)
+
+    def _take_page(self):
+        if self._pending is not None:
+            torch.cuda.current_stream().wait_event(self._event)
+            gidx, self._pending, self._event = self._pending, None, None
+            return gidx
+        return self.host_qm.gidx.to(self.device, non_blocking=True)
 
     def _ops_with_page(self):
         from .backend.gpu import GpuOps
-        gidx = self.host_qm.gidx.to(self.device, non_blocking=True)
+        gidx = self._take_page()
         if self._gpu_ops is None:
             self._gpu_ops = GpuOps(QuantizedMatrix(
                 gidx, self.host_qm.cuts, self.host_qm.has_missing))
