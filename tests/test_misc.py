@@ -159,3 +159,26 @@ def test_category_recode_between_frames():
         "x": np.zeros(4, np.float32)})
     d3 = xgb.DMatrix(df3, label=np.zeros(4), enable_categorical=True)
     assert np.isfinite(bst.predict(d3)).all()
+
+
+def test_cv_fpreproc_and_dict_result():
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(200, 4).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    d = xgb.DMatrix(X, label=y)
+    seen = []
+
+    def fpreproc(dtr, dte, params):
+        # reference-style hook: rescale a param per fold
+        seen.append(dtr.num_row())
+        params["eta"] = 0.1
+        return dtr, dte, params
+
+    res = xgb.cv({"objective": "binary:logistic", "max_depth": 2},
+                 d, num_boost_round=3, nfold=4, fpreproc=fpreproc,
+                 as_pandas=False, seed=1)
+    assert len(seen) == 4
+    assert isinstance(res, dict)
+    assert len(res["test-logloss-mean"]) == 3

@@ -80,7 +80,8 @@ class CVPack:
 
 
 def mknfold(dall: DMatrix, nfold: int, params, seed: int,
-            stratified=False, folds=None, shuffle=True) -> List[CVPack]:
+            stratified=False, folds=None, shuffle=True,
+            fpreproc=None) -> List[CVPack]:
     rng = np.random.RandomState(seed)
     n = dall.num_row()
     if folds is not None:
@@ -98,7 +99,13 @@ def mknfold(dall: DMatrix, nfold: int, params, seed: int,
                        chunks[i]) for i in range(nfold)]
     packs = []
     for tr_idx, te_idx in splits:
-        packs.append(CVPack(dall.slice(tr_idx), dall.slice(te_idx), params))
+        dtr, dte = dall.slice(tr_idx), dall.slice(te_idx)
+        fparams = params
+        if fpreproc is not None:
+            # reference semantics (python-package training.py mknfold):
+            # per-fold hook returns possibly-new matrices and params
+            dtr, dte, fparams = fpreproc(dtr, dte, dict(params))
+        packs.append(CVPack(dtr, dte, fparams))
     return packs
 
 
@@ -124,7 +131,8 @@ def cv(params, dtrain, num_boost_round=10, nfold=3, stratified=False,
     if metrics:
         params["eval_metric"] = list(metrics) if len(list(metrics)) > 1 \
             else list(metrics)[0]
-    packs = mknfold(dtrain, nfold, params, seed, stratified, folds, shuffle)
+    packs = mknfold(dtrain, nfold, params, seed, stratified, folds, shuffle,
+                    fpreproc)
     results: Dict[str, List[float]] = {}
     metric_fn = custom_metric or feval
     best_iter = None
