@@ -182,3 +182,30 @@ def test_cv_fpreproc_and_dict_result():
     assert len(seen) == 4
     assert isinstance(res, dict)
     assert len(res["test-logloss-mean"]) == 3
+
+
+def test_feature_weights_bias_column_sampling():
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(3)
+    X = rng.randn(500, 6).astype(np.float32)
+    y = (X[:, 5] + 0.05 * rng.randn(500)).astype(np.float32)
+    # weight 0 on the informative feature -> it can never be sampled
+    fw = np.ones(6, dtype=np.float32)
+    fw[5] = 0.0
+    d = xgb.DMatrix(X, label=y, feature_weights=fw)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 3,
+                     "colsample_bynode": 0.5, "seed": 7}, d, 10)
+    score = bst.get_score(importance_type="weight")
+    assert "f5" not in score and len(score) > 0
+    # weight heavily toward f5 -> it is used
+    fw2 = np.full(6, 1e-6, dtype=np.float32)
+    fw2[5] = 1.0
+    d2 = xgb.DMatrix(X, label=y, feature_weights=fw2)
+    bst2 = xgb.train({"objective": "reg:squarederror", "max_depth": 3,
+                      "colsample_bynode": 0.5, "seed": 7}, d2, 10)
+    assert "f5" in bst2.get_score(importance_type="weight")
+    # validation
+    import pytest
+    with pytest.raises(ValueError):
+        xgb.DMatrix(X, label=y, feature_weights=np.ones(3))

@@ -246,7 +246,8 @@ class Booster:
                     ops, self._scaled_param(eta_scale), quantizer, n,
                     seed=seed + 31 * ptree + 17 * k,
                     monotone=self._monotone_array(),
-                    interaction=self.tparam.interaction_constraints)
+                    interaction=self.tparam.interaction_constraints,
+                    feature_weights=dtrain.info.feature_weights)
                 tree, positions = grower.grow(qg, tree)
                 self.trees.append(tree)
                 self.tree_info.append(k)
@@ -332,7 +333,8 @@ class Booster:
         tree = RegTree(self.n_features)
         grower = TreeGrower(ops, self.tparam, quantizer, n, seed=seed,
                             monotone=self._monotone_array(),
-                            interaction=tp.interaction_constraints)
+                            interaction=tp.interaction_constraints,
+                            feature_weights=dtrain.info.feature_weights)
         tree, positions = grower.grow(ent["qg"], tree)
         self.trees.append(tree)
         self.tree_info.append(0)
@@ -383,7 +385,9 @@ class Booster:
             quantizers.append(q)
             qgpairs.append(q.quantize(gp))
         tree = RegTree(self.n_features, n_out)
-        grower = MultiTargetGrower(ops, self.tparam, quantizers, n, seed)
+        grower = MultiTargetGrower(
+            ops, self.tparam, quantizers, n, seed,
+            feature_weights=dtrain.info.feature_weights)
         tree, positions = grower.grow(qgpairs, tree)
         self.trees.append(tree)
         self.tree_info.append(0)

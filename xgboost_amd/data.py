@@ -37,6 +37,7 @@ class MetaInfo:
     label_upper_bound: Optional[np.ndarray] = None
     feature_names: Optional[List[str]] = None
     feature_types: Optional[List[str]] = None
+    feature_weights: Optional[np.ndarray] = None  # [n_col] colsample weights
 
     def validate(self) -> None:
         if self.labels is not None and self.labels.shape[0] != self.num_row:
@@ -215,7 +216,7 @@ class DMatrix:
                  group: Any = None, qid: Any = None,
                  label_lower_bound: Any = None, label_upper_bound: Any = None,
                  nthread: Optional[int] = None, enable_categorical: bool = False,
-                 silent: bool = False):
+                 feature_weights: Any = None, silent: bool = False):
         self.missing = float("nan") if missing is None else float(missing)
         self._device_data: Optional[torch.Tensor] = None
         if isinstance(data, torch.Tensor) and data.is_cuda:
@@ -282,6 +283,8 @@ class DMatrix:
                                    else inferred_types)
         if self.info.feature_types is not None and len(self.info.feature_types) != X.shape[1]:
             raise ValueError("feature_types length mismatch")
+        if feature_weights is not None:
+            self.set_info(feature_weights=feature_weights)
         self.info.validate()
         self._quantized: Dict[int, QuantizedMatrix] = {}
         self._ref_cuts: Optional[HistogramCuts] = None
@@ -306,7 +309,8 @@ class DMatrix:
 
     def set_info(self, *, label=None, weight=None, base_margin=None,
                  group=None, qid=None, feature_names=None, feature_types=None,
-                 label_lower_bound=None, label_upper_bound=None) -> None:
+                 label_lower_bound=None, label_upper_bound=None,
+                 feature_weights=None) -> None:
         if label is not None:
             self.info.labels = _as_float_array(label)
         if weight is not None:
@@ -328,6 +332,15 @@ class DMatrix:
             self.info.feature_names = list(feature_names)
         if feature_types is not None:
             self.info.feature_types = list(feature_types)
+        if feature_weights is not None:
+            fw = _as_float_array(feature_weights).reshape(-1)
+            if fw.size != self.info.num_col:
+                raise ValueError(
+                    f"feature_weights length {fw.size} != num_col "
+                    f"{self.info.num_col}")
+            if (fw < 0).any():
+                raise ValueError("feature_weights must be non-negative")
+            self.info.feature_weights = fw
         self.info.validate()
 
     set_label = lambda self, label: self.set_info(label=label)

@@ -269,6 +269,11 @@ class _StreamedPage:
         if self._pending is not None:
             torch.cuda.current_stream().wait_event(self._event)
             gidx, self._pending, self._event = self._pending, None, None
+            # the tensor was allocated on the copy stream but is consumed
+            # by kernels on the current stream: without this, freeing it
+            # (next swap_gidx) lets the allocator recycle the memory for
+            # the NEXT prefetch while kernels still read it
+            gidx.record_stream(torch.cuda.current_stream())
             return gidx
         return self.host_qm.gidx.to(self.device, non_blocking=True)
 

@@ -27,27 +27,43 @@ from .tree_model import RegTree
 class ColumnSampler:
     """Nested per-tree/level/node feature sampling (reference
     src/common/random.h:74).  Seeded identically on every worker so
-    feature sets agree without communication."""
+    feature sets agree without communication.  With `feature_weights`
+    set, sampling is weighted without replacement via the
+    Efraimidis-Spirakis key trick (rand^(1/w), top-k — reference
+    WeightedSamplingWithoutReplacement, random.h:104)."""
 
-    def __init__(self, n_features: int, param: TrainParam, seed: int):
+    def __init__(self, n_features: int, param: TrainParam, seed: int,
+                 feature_weights: Optional[np.ndarray] = None):
         self.rng = np.random.RandomState(seed & 0x7FFFFFFF)
         self.n_features = n_features
         self.param = param
-        n = n_features
+        self.weights = (np.asarray(feature_weights, dtype=np.float64)
+                        if feature_weights is not None else None)
         feats = np.arange(n_features)
         if param.colsample_bytree < 1.0:
-            k = max(1, int(round(param.colsample_bytree * n)))
-            feats = np.sort(self.rng.choice(feats, size=k, replace=False))
+            k = max(1, int(round(param.colsample_bytree * n_features)))
+            feats = self._sample(feats, k)
         self.tree_set = feats
         self._level_cache: Dict[int, np.ndarray] = {}
+
+    def _sample(self, feats: np.ndarray, k: int) -> np.ndarray:
+        if self.weights is None:
+            return np.sort(self.rng.choice(feats, size=k, replace=False))
+        w = self.weights[feats]
+        pos = w > 0
+        feats, w = feats[pos], w[pos]
+        if feats.size == 0:
+            raise ValueError("all sampled feature_weights are zero")
+        k = min(k, feats.size)
+        keys = self.rng.random_sample(feats.size) ** (1.0 / w)
+        return np.sort(feats[np.argsort(-keys)[:k]])
 
     def level_set(self, depth: int) -> np.ndarray:
         if self.param.colsample_bylevel >= 1.0:
             return self.tree_set
         if depth not in self._level_cache:
             k = max(1, int(round(self.param.colsample_bylevel * len(self.tree_set))))
-            self._level_cache[depth] = np.sort(
-                self.rng.choice(self.tree_set, size=k, replace=False))
+            self._level_cache[depth] = self._sample(self.tree_set, k)
         return self._level_cache[depth]
 
     def node_set(self, depth: int) -> Optional[np.ndarray]:
@@ -58,7 +74,7 @@ class ColumnSampler:
                 return None  # all features
             return base
         k = max(1, int(round(self.param.colsample_bynode * len(base))))
-        return np.sort(self.rng.choice(base, size=k, replace=False))
+        return self._sample(base, k)
 
 
 class InteractionConstraints:
@@ -100,12 +116,14 @@ class TreeGrower:
     def __init__(self, ops, param: TrainParam, quantizer: GradQuantizer,
                  n_rows: int, seed: int = 0,
                  monotone: Optional[np.ndarray] = None,
-                 interaction: Optional[Sequence[Sequence[int]]] = None):
+                 interaction: Optional[Sequence[Sequence[int]]] = None,
+                 feature_weights: Optional[np.ndarray] = None):
         self.ops = ops
         self.param = param
         self.quantizer = quantizer
         self.n_rows = n_rows
-        self.col_sampler = ColumnSampler(ops.qm.n_features, param, seed)
+        self.col_sampler = ColumnSampler(ops.qm.n_features, param, seed,
+                                         feature_weights)
         self.monotone = monotone
         self.interaction = (InteractionConstraints(interaction, ops.qm.n_features)
                             if interaction else None)
@@ -367,12 +385,14 @@ class MultiTargetGrower:
     gain is the sum of per-target gains (multi_evaluate_splits.cu)."""
 
     def __init__(self, ops, param: TrainParam, quantizers, n_rows: int,
-                 seed: int = 0):
+                 seed: int = 0,
+                 feature_weights: Optional[np.ndarray] = None):
         self.ops = ops
         self.param = param
         self.quantizers = quantizers  # one GradQuantizer per target
         self.n_rows = n_rows
-        self.col_sampler = ColumnSampler(ops.qm.n_features, param, seed)
+        self.col_sampler = ColumnSampler(ops.qm.n_features, param, seed,
+                                         feature_weights)
 
     def grow(self, qgpairs, tree: RegTree):
         from .splits import evaluate_splits_multi_np

@@ -141,6 +141,8 @@ class XGBModel:
             base_margin_eval_set=None, verbose: Union[bool, int] = True,
             xgb_model=None, feature_weights=None) -> "XGBModel":
         dtrain = self._make_dmatrix(X, y, sample_weight, base_margin)
+        if feature_weights is not None:
+            dtrain.set_info(feature_weights=feature_weights)
         evals = []
         if eval_set:
             for i, (ex, ey) in enumerate(eval_set):
