@@ -288,10 +288,14 @@ class _StreamedPage:
         return self._gpu_ops
 
     def _drop_page(self) -> None:
+        # keep the column count: evaluate_splits (and anything else that
+        # runs between sweeps) reads qm.n_features from the swapped-in
+        # tensor's shape
         if self._gpu_ops is not None:
             self._gpu_ops.swap_gidx(
-                torch.zeros(0, dtype=self.host_qm.gidx.dtype,
-                            device=self.device).view(0, 1))
+                torch.zeros((0, self.host_qm.gidx.shape[1]),
+                            dtype=self.host_qm.gidx.dtype,
+                            device=self.device))
 
     # ops that touch the bin matrix: stream the page in
     def build_hist_nodes(self, qgpair, nids):
