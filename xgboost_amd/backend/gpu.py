@@ -380,6 +380,16 @@ class GpuOps(SegmentedOpsMixin):
                 "eval_best": torch.empty((pool_rows, 6), dtype=torch.int64,
                                          device=dev),
                 "pos": torch.zeros(n_rows, dtype=torch.int32, device=dev),
+                # 1-sync driver: persistent partition counters (expand
+                # nodes per level <= 2^(max_depth-1)), device-generated
+                # hist task buffer + task-gen scratch
+                "counters": torch.empty(max(4, 1 << param.max_depth),
+                                        dtype=torch.int32, device=dev),
+                "hist_tasks_cap": 16384 + max_build + 8,
+                "hist_tasks": torch.empty((16384 + max_build + 8, 4),
+                                          dtype=torch.int32, device=dev),
+                "tg_scratch": torch.empty(3 * max_build + 8,
+                                          dtype=torch.int32, device=dev),
                 "driver": self.lib.gbt_driver_create(),
             }
             self._native_ws = ws
@@ -432,6 +442,8 @@ class GpuOps(SegmentedOpsMixin):
             self.hip.ptr(ws["eval_dir"]), self.hip.ptr(ws["eval_lsum"]),
             self.hip.ptr(ws["eval_best"]), self.hip.ptr(ws["pos"]),
             max_build,
+            self.hip.ptr(ws["counters"]), self.hip.ptr(ws["hist_tasks"]),
+            ws["hist_tasks_cap"], self.hip.ptr(ws["tg_scratch"]),
             quantizer.g_scale, quantizer.h_scale,
             root_sums[0], root_sums[1],
             param.reg_lambda, param.reg_alpha, param.max_delta_step,

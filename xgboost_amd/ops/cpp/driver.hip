@@ -9,6 +9,17 @@
 // Depthwise growth, numeric splits, optional monotone constraints.
 // Produces bit-identical trees to the Python GPU driver: same kernels,
 // same double-precision host math, compiled with -ffp-contract=off.
+//
+// ONE host sync per level: the whole phase chain
+//   partition -> child-hist task generation (on device, from the
+//   partition counters) -> hist build -> sibling subtraction ->
+//   split evaluation
+// is enqueued in a single burst, then ONE hipStreamSynchronize reads
+// back the per-node best splits AND the partition counters together.
+// The build-vs-subtract sibling choice uses the hessian sums known
+// from the parent's evaluation (exact row counts are not yet on the
+// host at enqueue time); histogram subtraction is exact int64, so the
+// choice affects only performance, never the resulting tree.
 #include "gbt_kernels.h"
 
 #include <hip/hip_runtime.h>
@@ -143,6 +154,79 @@ __global__ void SubtractHistKernel(const int64_t* __restrict__ parents,
   }
 }
 
+// Generate the child-hist BlockTask array on device: the child
+// segments come from the partition counters (still in flight on the
+// stream when the host enqueues this), so the host never has to wait
+// for the partition before launching the hist build.
+//   desc[j] = {parent_begin, parent_end, counter_slot, is_left}
+// Child j's segment: left  -> [parent_begin, counters[2*slot])
+//                    right -> [counters[2*slot], parent_end)
+// The task array is padded with empty tasks up to max_tasks (the
+// hist/partition kernels early-return on them); scratch holds
+// {begin, end, task_prefix} per child.
+__global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
+                                  const int32_t* __restrict__ desc, int k,
+                                  long long min_rows, long long target_tasks,
+                                  int max_tasks,
+                                  int32_t* __restrict__ scratch,
+                                  BlockTask* __restrict__ out_tasks) {
+  __shared__ long long s_rpt;
+  __shared__ int s_total_tasks;
+  if (threadIdx.x == 0) {
+    long long total = 0;
+    for (int j = 0; j < k; ++j) {
+      const int pb = desc[4 * j], pe = desc[4 * j + 1];
+      const int split = counters[2 * desc[4 * j + 2]];
+      const int b = desc[4 * j + 3] ? pb : split;
+      const int e = desc[4 * j + 3] ? split : pe;
+      scratch[3 * j] = b;
+      scratch[3 * j + 1] = e;
+      total += e - b;
+    }
+    long long rpt =
+        std::max(min_rows, (total + target_tasks - 1) / target_tasks);
+    // defensive: never overflow the task buffer (host bound should
+    // already guarantee this)
+    for (;;) {
+      long long need = 0;
+      for (int j = 0; j < k; ++j) {
+        const long long sz = scratch[3 * j + 1] - scratch[3 * j];
+        need += (sz + rpt - 1) / rpt;
+      }
+      if (need <= max_tasks) break;
+      rpt <<= 1;
+    }
+    int pref = 0;
+    for (int j = 0; j < k; ++j) {
+      scratch[3 * j + 2] = pref;
+      const long long sz = scratch[3 * j + 1] - scratch[3 * j];
+      pref += (int)((sz + rpt - 1) / rpt);
+    }
+    s_rpt = rpt;
+    s_total_tasks = pref;
+  }
+  __syncthreads();
+  const long long rpt = s_rpt;
+  const int total_tasks = s_total_tasks;
+  for (int t = threadIdx.x; t < max_tasks; t += blockDim.x) {
+    if (t >= total_tasks) {
+      out_tasks[t] = BlockTask{0, 0, 0, 0};
+      continue;
+    }
+    // binary search: largest j with prefix[j] <= t
+    int lo = 0, hi = k - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (scratch[3 * mid + 2] <= t) lo = mid; else hi = mid - 1;
+    }
+    const int j = lo;
+    const long long idx = t - scratch[3 * j + 2];
+    const int b = scratch[3 * j] + (int)(idx * rpt);
+    const int e = (int)std::min<long long>(b + rpt, scratch[3 * j + 1]);
+    out_tasks[t] = BlockTask{j, b, e, 0};
+  }
+}
+
 struct LeafSeg {
   int nid, begin, end;
 };
@@ -169,6 +253,10 @@ int gbt_grow_tree(
     int64_t* hist_pool_b, double* eval_gain, int32_t* eval_bin,
     uint8_t* eval_dir, int64_t* eval_lsum, int64_t* eval_best,
     int32_t* pos_out, int max_nodes_level,
+    int32_t* part_counters,      // [>= 2 * max expand nodes]
+    BlockTask* hist_tasks_dev,   // [hist_tasks_cap]
+    int hist_tasks_cap,
+    int32_t* tg_scratch,         // [3 * max_nodes_level + 4]
     // scalars
     double g_scale, double h_scale, long long root_gq, long long root_hq,
     double reg_lambda, double reg_alpha, double max_delta_step,
@@ -218,29 +306,32 @@ int gbt_grow_tree(
     long long v = e ? atoll(e) : 512;
     return v >= 64 && v <= 16384 ? v : 512;
   }();
-  auto build_hists = [&](std::vector<Node*>& nodes, int64_t* pool) -> int {
+  const long long hist_min_rows = 2048;
+
+  // ---- root histogram: tasks host-generated (root segment is known)
+  {
+    std::vector<Node*> nodes{&root};
     std::vector<BlockTask> tasks;
-    ChunkTasks(nodes, &tasks, 2048, hist_tasks);
+    ChunkTasks(nodes, &tasks, hist_min_rows, hist_tasks);
     const int slot = ctx->ring.next();
     size_t bytes = tasks.size() * sizeof(BlockTask);
     if (int e = ctx->ring.ensure(slot, bytes)) return e;
     memcpy(ctx->ring.host[slot], tasks.data(), bytes);
     HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot], bytes,
                              hipMemcpyHostToDevice, stream));
-    HIP_CHECK(hipMemsetAsync(pool, 0,
-                             nodes.size() * hist_row * sizeof(int64_t),
+    HIP_CHECK(hipMemsetAsync(hist_pool_a, 0, hist_row * sizeof(int64_t),
                              stream));
     gbt_hist(gidx8, gidx16, n_features, qgpair, ridx,
-             (const BlockTask*)ctx->ring.dev[slot], (int)tasks.size(), pool,
-             n_bins, feat_group_start_dev, bin_group_start_dev, n_groups,
-             max_group_bins, cut_ptrs_dev, use_shared, stream);
-    if (allreduce) {
-      allreduce((long long*)pool, (long long)nodes.size() * hist_row);
-    }
-    return 0;
-  };
+             (const BlockTask*)ctx->ring.dev[slot], (int)tasks.size(),
+             hist_pool_a, n_bins, feat_group_start_dev, bin_group_start_dev,
+             n_groups, max_group_bins, cut_ptrs_dev, use_shared, stream);
+    if (allreduce) allreduce((long long*)hist_pool_a, hist_row);
+  }
 
-  auto evaluate = [&](std::vector<Node*>& nodes, const int64_t* hists) -> int {
+  // enqueue split evaluation for `nodes` against `hists`; the best-split
+  // D2H is enqueued but NOT synced — the caller syncs
+  auto evaluate_enqueue = [&](std::vector<Node*>& nodes,
+                              const int64_t* hists) -> int {
     const int k = (int)nodes.size();
     const int slot = ctx->ring.next();
     size_t off_ps = 0;
@@ -270,13 +361,11 @@ int gbt_grow_tree(
                  nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k, n_features,
                     eval_best, stream);
-    size_t rb = (size_t)k * 6 * sizeof(int64_t);
-    if (int e = ctx->ensure_readback(rb)) return e;
-    HIP_CHECK(hipMemcpyAsync(ctx->readback_host, eval_best, rb,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipStreamSynchronize(stream));
-    const int64_t* best = (const int64_t*)ctx->readback_host;
-    for (int i = 0; i < k; ++i) {
+    return 0;
+  };
+
+  auto parse_best = [&](std::vector<Node*>& nodes, const int64_t* best) {
+    for (size_t i = 0; i < nodes.size(); ++i) {
       Node* nd = nodes[i];
       double gain;
       memcpy(&gain, &best[6 * i], sizeof(double));
@@ -287,11 +376,12 @@ int gbt_grow_tree(
       nd->feature = (int)best[6 * i + 5];
       nd->gain = (nd->bin >= 0 && std::isfinite(gain)) ? gain : -INFINITY;
     }
-    return 0;
   };
 
-  auto partition = [&](std::vector<Node*>& nodes,
-                       std::vector<int>* left_counts) -> int {
+  // enqueue partition of `nodes` (segments known on host); counters are
+  // left in part_counters for the device-side hist task generation and
+  // the piggybacked readback at the level sync
+  auto partition_enqueue = [&](std::vector<Node*>& nodes) -> int {
     const int k = (int)nodes.size();
     std::vector<BlockTask> tasks;
     ChunkTasks(nodes, &tasks);
@@ -319,32 +409,58 @@ int gbt_grow_tree(
     }
     HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
                              hipMemcpyHostToDevice, stream));
+    // counter init goes to the persistent buffer (read by the task-gen
+    // kernel and by the level-sync readback)
+    HIP_CHECK(hipMemcpyAsync(part_counters, h + off_cnt,
+                             (size_t)k * 2 * sizeof(int32_t),
+                             hipMemcpyHostToDevice, stream));
     char* d = (char*)ctx->ring.dev[slot];
     gbt_partition(gidx8, gidx16, n_features, ridx, ridx_out,
                   (const BlockTask*)(d + off_tasks), (int)tasks.size(),
                   (const int32_t*)(d + off_feat),
                   (const int32_t*)(d + off_sbin), (const uint8_t*)(d + off_dl),
-                  nullptr, nullptr, n_bins_feat_dev, (int32_t*)(d + off_cnt),
-                  stream);
+                  nullptr, nullptr, n_bins_feat_dev, part_counters, stream);
     gbt_copy_ranges(ridx_out, ridx, (const BlockTask*)(d + off_tasks),
                     (int)tasks.size(), stream);
-    size_t rb = (size_t)k * 2 * sizeof(int32_t);
-    if (int e = ctx->ensure_readback(rb)) return e;
-    HIP_CHECK(hipMemcpyAsync(ctx->readback_host, d + off_cnt, rb,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipStreamSynchronize(stream));
-    const int32_t* fin = (const int32_t*)ctx->readback_host;
-    left_counts->resize(k);
-    for (int i = 0; i < k; ++i) {
-      (*left_counts)[i] = fin[2 * i] - nodes[i]->seg_begin;
-    }
     return 0;
   };
 
-  // ---- root ----
-  std::vector<Node*> frontier{&root};
-  if (int e = build_hists(frontier, hist_pool_a)) return e;
-  if (int e = evaluate(frontier, hist_pool_a)) return e;
+  // one D2H burst + ONE sync per level: best splits for `eval_nodes`
+  // (may be empty on the final level) + partition counters for
+  // `n_expand` nodes
+  auto level_sync = [&](int n_eval, int n_expand,
+                        const int64_t** best_out,
+                        const int32_t** cnt_out) -> int {
+    size_t off_best = 0;
+    size_t off_cnt = ((size_t)n_eval * 6 * sizeof(int64_t) + 63) & ~63ULL;
+    size_t bytes = off_cnt + (size_t)n_expand * 2 * sizeof(int32_t);
+    if (int e = ctx->ensure_readback(bytes)) return e;
+    char* h = (char*)ctx->readback_host;
+    if (n_eval > 0) {
+      HIP_CHECK(hipMemcpyAsync(h + off_best, eval_best,
+                               (size_t)n_eval * 6 * sizeof(int64_t),
+                               hipMemcpyDeviceToHost, stream));
+    }
+    if (n_expand > 0) {
+      HIP_CHECK(hipMemcpyAsync(h + off_cnt, part_counters,
+                               (size_t)n_expand * 2 * sizeof(int32_t),
+                               hipMemcpyDeviceToHost, stream));
+    }
+    HIP_CHECK(hipStreamSynchronize(stream));
+    *best_out = (const int64_t*)(h + off_best);
+    *cnt_out = (const int32_t*)(h + off_cnt);
+    return 0;
+  };
+
+  // ---- root evaluation (root-only sync) ----
+  {
+    std::vector<Node*> frontier{&root};
+    if (int e = evaluate_enqueue(frontier, hist_pool_a)) return e;
+    const int64_t* best;
+    const int32_t* cnt;
+    if (int e = level_sync(1, 0, &best, &cnt)) return e;
+    parse_best(frontier, best);
+  }
 
   std::vector<Node> level_nodes{root};
   std::vector<Node> next_level;
@@ -365,7 +481,8 @@ int gbt_grow_tree(
       level_nodes.clear();
       break;
     }
-    // apply splits on host
+    // apply splits on host (child sums/bounds are known from the
+    // parents' evaluation; child SEGMENTS arrive at the level sync)
     next_level.clear();
     next_level.reserve(2 * expand.size());
     for (Node* nd : expand) {
@@ -398,6 +515,8 @@ int gbt_grow_tree(
       ln.hq = nd->lhq;
       rn.gq = rgq;
       rn.hq = rhq;
+      ln.seg_begin = ln.seg_end = -1;  // set at the level sync
+      rn.seg_begin = rn.seg_end = -1;
       ln.lo = rn.lo = nd->lo;
       ln.hi = rn.hi = nd->hi;
       ln.gain = rn.gain = -INFINITY;
@@ -418,62 +537,111 @@ int gbt_grow_tree(
       next_level.push_back(ln);
       next_level.push_back(rn);
     }
-    std::vector<int> left_counts;
-    if (int e = partition(expand, &left_counts)) return e;
-    for (size_t i = 0; i < expand.size(); ++i) {
-      Node& ln = next_level[2 * i];
-      Node& rn = next_level[2 * i + 1];
-      ln.seg_begin = expand[i]->seg_begin;
-      ln.seg_end = expand[i]->seg_begin + left_counts[i];
-      rn.seg_begin = ln.seg_end;
-      rn.seg_end = expand[i]->seg_end;
-    }
+    const int n_expand = (int)expand.size();
+    if (int e = partition_enqueue(expand)) return e;
+
+    auto set_child_segs = [&](const int32_t* fin) {
+      for (int i = 0; i < n_expand; ++i) {
+        Node& ln = next_level[2 * i];
+        Node& rn = next_level[2 * i + 1];
+        ln.seg_begin = expand[i]->seg_begin;
+        ln.seg_end = fin[2 * i];  // begin + n_left
+        rn.seg_begin = ln.seg_end;
+        rn.seg_end = expand[i]->seg_end;
+      }
+    };
+
     if (depth + 1 >= max_depth) {
+      const int64_t* best;
+      const int32_t* cnt;
+      if (int e = level_sync(0, n_expand, &best, &cnt)) return e;
+      set_child_segs(cnt);
       for (auto& nd : next_level) {
         leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end});
       }
       level_nodes.clear();
       break;
     }
-    // build smaller sibling, subtract larger
+    // choose the sibling to build by hessian sum (a proxy for row
+    // count — exact counts are still on the device); subtraction is
+    // exact int64, so this is a performance choice only
     std::vector<Node*> build;
     std::vector<int32_t> parent_slots;
     std::vector<Node*> subtracted;
-    for (size_t i = 0; i < expand.size(); ++i) {
+    std::vector<int32_t> desc;  // [k][4] for HistTaskGenKernel
+    build.reserve(n_expand);
+    desc.reserve(4 * n_expand);
+    for (int i = 0; i < n_expand; ++i) {
       Node& ln = next_level[2 * i];
       Node& rn = next_level[2 * i + 1];
-      Node* small = (ln.seg_end - ln.seg_begin <= rn.seg_end - rn.seg_begin)
-                        ? &ln : &rn;
+      Node* small = (ln.hq <= rn.hq) ? &ln : &rn;
       Node* big = (small == &ln) ? &rn : &ln;
       small->hist_slot = (int)build.size();
       build.push_back(small);
       parent_slots.push_back(expand[i]->hist_slot);
       subtracted.push_back(big);
+      desc.push_back(expand[i]->seg_begin);
+      desc.push_back(expand[i]->seg_end);
+      desc.push_back(i);                       // counter slot
+      desc.push_back(small == &ln ? 1 : 0);    // is_left
     }
     if (2 * (int)build.size() > 2 * max_nodes_level) return -9999;
-    if (int e = build_hists(build, next_pool)) return e;
+    const int kb = (int)build.size();
+    // host bound on the device-generated task count
+    long long bound_total = 0;
+    for (Node* nd : expand) bound_total += nd->seg_end - nd->seg_begin;
+    int max_tasks = (int)std::min<long long>(
+        std::min<long long>(bound_total / hist_min_rows, hist_tasks) + kb + 1,
+        hist_tasks_cap);
     {
-      const int k = (int)build.size();
       const int slot = ctx->ring.next();
-      size_t bytes = (size_t)k * sizeof(int32_t);
+      size_t bytes = desc.size() * sizeof(int32_t);
+      if (int e = ctx->ring.ensure(slot, bytes)) return e;
+      memcpy(ctx->ring.host[slot], desc.data(), bytes);
+      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
+                               bytes, hipMemcpyHostToDevice, stream));
+      hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
+                         part_counters, (const int32_t*)ctx->ring.dev[slot],
+                         kb, hist_min_rows, hist_tasks, max_tasks, tg_scratch,
+                         hist_tasks_dev);
+    }
+    HIP_CHECK(hipMemsetAsync(next_pool, 0,
+                             (size_t)kb * hist_row * sizeof(int64_t), stream));
+    gbt_hist(gidx8, gidx16, n_features, qgpair, ridx, hist_tasks_dev,
+             max_tasks, next_pool, n_bins, feat_group_start_dev,
+             bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
+             use_shared, stream);
+    if (allreduce) {
+      allreduce((long long*)next_pool, (long long)kb * hist_row);
+    }
+    {
+      const int slot = ctx->ring.next();
+      size_t bytes = (size_t)kb * sizeof(int32_t);
       if (int e = ctx->ring.ensure(slot, bytes)) return e;
       memcpy(ctx->ring.host[slot], parent_slots.data(), bytes);
       HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
                                bytes, hipMemcpyHostToDevice, stream));
-      int64_t* sub_out = next_pool + (long long)k * hist_row;
-      const long long total = (long long)k * hist_row;
+      int64_t* sub_out = next_pool + (long long)kb * hist_row;
+      const long long total = (long long)kb * hist_row;
       int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
       hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
                          stream, cur_pool, next_pool, sub_out,
                          (const int32_t*)ctx->ring.dev[slot], (int)hist_row,
-                         k);
-      for (int i = 0; i < k; ++i) subtracted[i]->hist_slot = k + i;
+                         kb);
+      for (int i = 0; i < kb; ++i) subtracted[i]->hist_slot = kb + i;
     }
     std::vector<Node*> eval_nodes;
-    eval_nodes.reserve(2 * build.size());
+    eval_nodes.reserve(2 * kb);
     for (Node* b : build) eval_nodes.push_back(b);
     for (Node* s : subtracted) eval_nodes.push_back(s);
-    if (int e = evaluate(eval_nodes, next_pool)) return e;
+    if (int e = evaluate_enqueue(eval_nodes, next_pool)) return e;
+    // ---- the ONE sync for this level ----
+    const int64_t* best;
+    const int32_t* cnt;
+    if (int e = level_sync((int)eval_nodes.size(), n_expand, &best, &cnt))
+      return e;
+    parse_best(eval_nodes, best);
+    set_child_segs(cnt);
     level_nodes.swap(next_level);
     std::swap(cur_pool, next_pool);
   }

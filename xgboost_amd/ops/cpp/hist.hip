@@ -34,6 +34,10 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
     const int32_t* __restrict__ bin_group_start,
     const int32_t* __restrict__ cut_ptrs) {
   const BlockTask task = tasks[blockIdx.x];
+  // device-generated task arrays are padded with empty tasks up to the
+  // launched grid (task count is not host-known in the 1-sync driver);
+  // return before any LDS work — uniform across the block
+  if (task.row_begin >= task.row_end) return;
   const int group = blockIdx.y;
   const int f_begin = feat_group_start[group];
   const int f_end = feat_group_start[group + 1];

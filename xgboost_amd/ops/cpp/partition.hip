@@ -48,6 +48,7 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
     const int32_t* __restrict__ n_bins_feat,
     int32_t* __restrict__ counters) {
   const BlockTask task = tasks[blockIdx.x];
+  if (task.row_begin >= task.row_end) return;  // padded empty task
   const int slot = task.out_slot;
   const int feature = split_feature[slot];
   const int sbin = split_bin_local[slot];
