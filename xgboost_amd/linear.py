@@ -80,6 +80,30 @@ class GBLinearModel:
                 dbias = self.eta * (-sg / sh)
                 self.weights[-1, k] += dbias
                 g += h * dbias
+            if self.updater == "shotgun":
+                # Parallel coordinate step (reference updater_shotgun.cc):
+                # all deltas computed against the same residual in two
+                # matvecs, then applied together.  The reference races
+                # Hogwild-style across threads; computing the whole batch
+                # from one residual snapshot is the deterministic
+                # equivalent and maps to GEMV instead of n_features
+                # host-synced dot products.
+                sum_grad = X.t() @ g + lam * self.weights[:-1, k]
+                sum_hess = Xsq.t() @ h
+                dw = -_threshold_l1(sum_grad, alp) / (sum_hess + lam)
+                dw = self.eta * torch.where(
+                    sum_hess > 1e-16, dw, torch.zeros_like(dw))
+                if self.top_k > 0:
+                    keep = torch.zeros_like(dw)
+                    sel = torch.as_tensor(
+                        np.ascontiguousarray(
+                            self._feature_order(iteration, None)),
+                        device=dw.device, dtype=torch.long)
+                    keep[sel] = dw[sel]
+                    dw = keep
+                self.weights[:-1, k] += dw
+                g += h * (X @ dw)
+                continue
             if self.feature_selector in ("greedy", "thrifty"):
                 gf = (X * g.view(-1, 1)).sum(dim=0).abs().cpu().numpy()
             else:
