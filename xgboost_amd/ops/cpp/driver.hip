@@ -177,8 +177,12 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     for (int j = 0; j < k; ++j) {
       const int pb = desc[4 * j], pe = desc[4 * j + 1];
       const int split = counters[2 * desc[4 * j + 2]];
-      const int b = desc[4 * j + 3] ? pb : split;
-      const int e = desc[4 * j + 3] ? split : pe;
+      // mode: 1 = build left, 0 = build right, 2 = build the smaller
+      // child (single-GPU only: local row counts are rank-dependent)
+      int mode = desc[4 * j + 3];
+      if (mode == 2) mode = (split - pb) <= (pe - split) ? 1 : 0;
+      const int b = mode ? pb : split;
+      const int e = mode ? split : pe;
       scratch[3 * j] = b;
       scratch[3 * j + 1] = e;
       total += e - b;
@@ -225,6 +229,58 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     const int e = (int)std::min<long long>(b + rpt, scratch[3 * j + 1]);
     out_tasks[t] = BlockTask{j, b, e, 0};
   }
+}
+
+// Sum quantized gradient pairs over the BUILT children (reusing the
+// device-generated hist task array): needed because with the
+// device-chosen sibling the host does not know which child each hist
+// slot holds, so the evaluator's per-slot parent sums must also be
+// produced on device.
+__global__ __launch_bounds__(256) void SumPairsKernel(
+    const int32_t* __restrict__ qgpair, const int32_t* __restrict__ ridx,
+    const BlockTask* __restrict__ tasks, int64_t* __restrict__ ps) {
+  const BlockTask task = tasks[blockIdx.x];
+  if (task.row_begin >= task.row_end) return;
+  long long g = 0, h = 0;
+  for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
+       i += blockDim.x) {
+    const int row = ridx[i];
+    g += qgpair[2 * (size_t)row];
+    h += qgpair[2 * (size_t)row + 1];
+  }
+  __shared__ long long sg[256 / 64], sh[256 / 64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  for (int off = 32; off > 0; off >>= 1) {
+    g += __shfl_down(g, off, 64);
+    h += __shfl_down(h, off, 64);
+  }
+  if (lane == 0) {
+    sg[wave] = g;
+    sh[wave] = h;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    long long tg = 0, th = 0;
+    for (int w = 0; w < (int)blockDim.x / 64; ++w) {
+      tg += sg[w];
+      th += sh[w];
+    }
+    if (tg) atomicAdd((unsigned long long*)&ps[2 * task.out_slot],
+                      (unsigned long long)tg);
+    if (th) atomicAdd((unsigned long long*)&ps[2 * task.out_slot + 1],
+                      (unsigned long long)th);
+  }
+}
+
+// subtracted sibling's sums = parent - built (slots kb..2kb-1)
+__global__ void DeriveSiblingSumsKernel(int64_t* __restrict__ ps,
+                                        const int64_t* __restrict__ parent_ps,
+                                        int kb) {
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= kb) return;
+  ps[2 * (kb + j)] = parent_ps[2 * j] - ps[2 * j];
+  ps[2 * (kb + j) + 1] = parent_ps[2 * j + 1] - ps[2 * j + 1];
 }
 
 struct LeafSeg {
@@ -328,37 +384,52 @@ int gbt_grow_tree(
     if (allreduce) allreduce((long long*)hist_pool_a, hist_row);
   }
 
-  // enqueue split evaluation for `nodes` against `hists`; the best-split
-  // D2H is enqueued but NOT synced — the caller syncs
-  auto evaluate_enqueue = [&](std::vector<Node*>& nodes,
-                              const int64_t* hists) -> int {
-    const int k = (int)nodes.size();
-    const int slot = ctx->ring.next();
-    size_t off_ps = 0;
-    size_t off_bd = ((size_t)k * 2 * sizeof(int64_t) + 7) & ~7ULL;
-    size_t bytes = off_bd + (has_mono ? (size_t)k * 2 * sizeof(double) : 0);
-    if (int e = ctx->ring.ensure(slot, bytes)) return e;
-    char* h = (char*)ctx->ring.host[slot];
-    int64_t* ps = (int64_t*)(h + off_ps);
-    for (int i = 0; i < k; ++i) {
-      ps[2 * i] = nodes[i]->gq;
-      ps[2 * i + 1] = nodes[i]->hq;
-    }
-    if (has_mono) {
-      double* bd = (double*)(h + off_bd);
+  // enqueue split evaluation against `hists`; the best-split D2H is
+  // enqueued but NOT synced — the caller syncs.  Node sums are staged
+  // from `nodes` unless a device-resident sum buffer is given
+  // (device-chosen siblings compute their sums on the GPU).
+  auto evaluate_enqueue = [&](int k, std::vector<Node*>* nodes,
+                              const int64_t* hists,
+                              const int64_t* ps_dev) -> int {
+    const int64_t* ps_arg = ps_dev;
+    if (ps_dev == nullptr) {
+      const int slot = ctx->ring.next();
+      size_t off_ps = 0;
+      size_t off_bd = ((size_t)k * 2 * sizeof(int64_t) + 7) & ~7ULL;
+      size_t bytes = off_bd + (has_mono ? (size_t)k * 2 * sizeof(double) : 0);
+      if (int e = ctx->ring.ensure(slot, bytes)) return e;
+      char* h = (char*)ctx->ring.host[slot];
+      int64_t* ps = (int64_t*)(h + off_ps);
       for (int i = 0; i < k; ++i) {
-        bd[2 * i] = nodes[i]->lo;
-        bd[2 * i + 1] = nodes[i]->hi;
+        ps[2 * i] = (*nodes)[i]->gq;
+        ps[2 * i + 1] = (*nodes)[i]->hq;
+      }
+      if (has_mono) {
+        double* bd = (double*)(h + off_bd);
+        for (int i = 0; i < k; ++i) {
+          bd[2 * i] = (*nodes)[i]->lo;
+          bd[2 * i + 1] = (*nodes)[i]->hi;
+        }
+      }
+      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
+                               hipMemcpyHostToDevice, stream));
+      char* d = (char*)ctx->ring.dev[slot];
+      ps_arg = (const int64_t*)(d + off_ps);
+      if (has_mono) {
+        gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
+                     g_scale, h_scale, reg_lambda, reg_alpha, max_delta_step,
+                     min_child_weight, monotone_dev,
+                     (const double*)(d + off_bd), nullptr, nullptr, eval_gain,
+                     eval_bin, eval_dir, eval_lsum, stream);
+        gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k,
+                        n_features, eval_best, stream);
+        return 0;
       }
     }
-    HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
-                             hipMemcpyHostToDevice, stream));
-    char* d = (char*)ctx->ring.dev[slot];
-    gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev,
-                 (const int64_t*)(d + off_ps), g_scale, h_scale, reg_lambda,
-                 reg_alpha, max_delta_step, min_child_weight, monotone_dev,
-                 has_mono ? (const double*)(d + off_bd) : nullptr, nullptr,
-                 nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, stream);
+    gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg, g_scale,
+                 h_scale, reg_lambda, reg_alpha, max_delta_step,
+                 min_child_weight, monotone_dev, nullptr, nullptr, nullptr,
+                 eval_gain, eval_bin, eval_dir, eval_lsum, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k, n_features,
                     eval_best, stream);
     return 0;
@@ -455,7 +526,8 @@ int gbt_grow_tree(
   // ---- root evaluation (root-only sync) ----
   {
     std::vector<Node*> frontier{&root};
-    if (int e = evaluate_enqueue(frontier, hist_pool_a)) return e;
+    if (int e = evaluate_enqueue(1, &frontier, hist_pool_a, nullptr))
+      return e;
     const int64_t* best;
     const int32_t* cnt;
     if (int e = level_sync(1, 0, &best, &cnt)) return e;
@@ -562,31 +634,42 @@ int gbt_grow_tree(
       level_nodes.clear();
       break;
     }
-    // choose the sibling to build by hessian sum (a proxy for row
-    // count — exact counts are still on the device); subtraction is
-    // exact int64, so this is a performance choice only
+    // Sibling to build vs subtract (exact int64 subtraction makes this
+    // a performance choice only, never a correctness one):
+    //  - single GPU, no monotone bounds: the DEVICE picks the
+    //    smaller-row child inside HistTaskGenKernel (exact local
+    //    counts); its per-slot sums for evaluation are computed on
+    //    device too (SumPairsKernel), and the host learns the choice
+    //    from the counters at the level sync.
+    //  - distributed or monotone: the HOST picks by global hessian sum
+    //    (rank-identical, so every rank builds/allreduces the same
+    //    slot layout) and stages sums/bounds as usual.
+    const bool dev_choice = (allreduce == nullptr) && !has_mono;
     std::vector<Node*> build;
     std::vector<int32_t> parent_slots;
     std::vector<Node*> subtracted;
     std::vector<int32_t> desc;  // [k][4] for HistTaskGenKernel
-    build.reserve(n_expand);
     desc.reserve(4 * n_expand);
     for (int i = 0; i < n_expand; ++i) {
       Node& ln = next_level[2 * i];
       Node& rn = next_level[2 * i + 1];
-      Node* small = (ln.hq <= rn.hq) ? &ln : &rn;
-      Node* big = (small == &ln) ? &rn : &ln;
-      small->hist_slot = (int)build.size();
-      build.push_back(small);
       parent_slots.push_back(expand[i]->hist_slot);
-      subtracted.push_back(big);
       desc.push_back(expand[i]->seg_begin);
       desc.push_back(expand[i]->seg_end);
-      desc.push_back(i);                       // counter slot
-      desc.push_back(small == &ln ? 1 : 0);    // is_left
+      desc.push_back(i);  // counter slot
+      if (dev_choice) {
+        desc.push_back(2);  // device picks the smaller child
+      } else {
+        Node* small = (ln.hq <= rn.hq) ? &ln : &rn;
+        Node* big = (small == &ln) ? &rn : &ln;
+        small->hist_slot = (int)build.size();
+        build.push_back(small);
+        subtracted.push_back(big);
+        desc.push_back(small == &ln ? 1 : 0);  // is_left
+      }
     }
-    if (2 * (int)build.size() > 2 * max_nodes_level) return -9999;
-    const int kb = (int)build.size();
+    if (2 * n_expand > 2 * max_nodes_level) return -9999;
+    const int kb = n_expand;
     // host bound on the device-generated task count
     long long bound_total = 0;
     for (Node* nd : expand) bound_total += nd->seg_end - nd->seg_begin;
@@ -614,6 +697,32 @@ int gbt_grow_tree(
     if (allreduce) {
       allreduce((long long*)next_pool, (long long)kb * hist_row);
     }
+    const int64_t* eval_ps_dev = nullptr;
+    if (dev_choice) {
+      // built-child sums on device: [parent sums (H2D) | ps (memset)]
+      const int slot = ctx->ring.next();
+      size_t off_ps = ((size_t)kb * 2 * sizeof(int64_t) + 63) & ~63ULL;
+      size_t bytes = off_ps + (size_t)kb * 4 * sizeof(int64_t);
+      if (int e = ctx->ring.ensure(slot, bytes)) return e;
+      int64_t* pp = (int64_t*)ctx->ring.host[slot];
+      for (int i = 0; i < kb; ++i) {
+        pp[2 * i] = expand[i]->gq;
+        pp[2 * i + 1] = expand[i]->hq;
+      }
+      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
+                               (size_t)kb * 2 * sizeof(int64_t),
+                               hipMemcpyHostToDevice, stream));
+      char* d = (char*)ctx->ring.dev[slot];
+      int64_t* ps = (int64_t*)(d + off_ps);
+      HIP_CHECK(hipMemsetAsync(ps, 0, (size_t)kb * 4 * sizeof(int64_t),
+                               stream));
+      hipLaunchKernelGGL(SumPairsKernel, dim3(max_tasks), dim3(256), 0,
+                         stream, qgpair, ridx, hist_tasks_dev, ps);
+      hipLaunchKernelGGL(DeriveSiblingSumsKernel,
+                         dim3((kb + 255) / 256), dim3(256), 0, stream, ps,
+                         (const int64_t*)d, kb);
+      eval_ps_dev = ps;
+    }
     {
       const int slot = ctx->ring.next();
       size_t bytes = (size_t)kb * sizeof(int32_t);
@@ -628,20 +737,41 @@ int gbt_grow_tree(
                          stream, cur_pool, next_pool, sub_out,
                          (const int32_t*)ctx->ring.dev[slot], (int)hist_row,
                          kb);
-      for (int i = 0; i < kb; ++i) subtracted[i]->hist_slot = kb + i;
+      for (int i = 0; i < (int)subtracted.size(); ++i) {
+        subtracted[i]->hist_slot = kb + i;
+      }
     }
     std::vector<Node*> eval_nodes;
-    eval_nodes.reserve(2 * kb);
-    for (Node* b : build) eval_nodes.push_back(b);
-    for (Node* s : subtracted) eval_nodes.push_back(s);
-    if (int e = evaluate_enqueue(eval_nodes, next_pool)) return e;
+    if (!dev_choice) {
+      eval_nodes.reserve(2 * kb);
+      for (Node* b : build) eval_nodes.push_back(b);
+      for (Node* s : subtracted) eval_nodes.push_back(s);
+    }
+    if (int e = evaluate_enqueue(2 * kb, dev_choice ? nullptr : &eval_nodes,
+                                 next_pool, eval_ps_dev)) return e;
     // ---- the ONE sync for this level ----
     const int64_t* best;
     const int32_t* cnt;
-    if (int e = level_sync((int)eval_nodes.size(), n_expand, &best, &cnt))
-      return e;
-    parse_best(eval_nodes, best);
+    if (int e = level_sync(2 * kb, n_expand, &best, &cnt)) return e;
     set_child_segs(cnt);
+    if (dev_choice) {
+      // recover the device's smaller-child choice from the counters
+      // (same rule as HistTaskGenKernel) and map eval slots to nodes
+      eval_nodes.resize(2 * kb);
+      for (int i = 0; i < kb; ++i) {
+        Node& ln = next_level[2 * i];
+        Node& rn = next_level[2 * i + 1];
+        const int nl = ln.seg_end - ln.seg_begin;
+        const int nr = rn.seg_end - rn.seg_begin;
+        Node* small = (nl <= nr) ? &ln : &rn;
+        Node* big = (small == &ln) ? &rn : &ln;
+        small->hist_slot = i;
+        big->hist_slot = kb + i;
+        eval_nodes[i] = small;
+        eval_nodes[kb + i] = big;
+      }
+    }
+    parse_best(eval_nodes, best);
     level_nodes.swap(next_level);
     std::swap(cur_pool, next_pool);
   }
