@@ -199,7 +199,7 @@ class GpuOps(SegmentedOpsMixin):
             self.hip.ptr(self.feat_group_start),
             self.hip.ptr(self.bin_group_start), self.n_groups,
             self.max_group_bins, self.hip.ptr(self.cut_ptrs),
-            self.use_shared, self.hip.stream())
+            self.use_shared, None, self.hip.stream())
         return out
 
     def allreduce_hist(self, hist: torch.Tensor) -> torch.Tensor:

@@ -28,7 +28,7 @@ _f = _c.c_float
 _d = _c.c_double
 
 _SIGS = {
-    "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p],
+    "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p, _p],
     "gbt_partition": [_p, _p, _i, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p, _p, _p],
     "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _d, _d, _d, _d, _d, _d,
                      _p, _p, _p, _p, _p, _p, _p, _p, _p],

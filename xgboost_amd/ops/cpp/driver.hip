@@ -142,8 +142,15 @@ __global__ void SubtractHistKernel(const int64_t* __restrict__ parents,
                                    const int64_t* __restrict__ built,
                                    int64_t* __restrict__ out,
                                    const int32_t* __restrict__ parent_slot,
-                                   int n_bins2, int k) {
+                                   int n_bins2, int k,
+                                   int64_t* __restrict__ ps,
+                                   const int64_t* __restrict__ parent_ps) {
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  // piggybacked: subtracted-sibling sums = parent - built (slots k..2k-1)
+  if (ps != nullptr && idx < k) {
+    ps[2 * (k + idx)] = parent_ps[2 * idx] - ps[2 * idx];
+    ps[2 * (k + idx) + 1] = parent_ps[2 * idx + 1] - ps[2 * idx + 1];
+  }
   const long long total = (long long)k * n_bins2;
   for (long long i = idx; i < total; i += (long long)gridDim.x * blockDim.x) {
     const int row = (int)(i / n_bins2);
@@ -169,7 +176,12 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
                                   long long min_rows, long long target_tasks,
                                   int max_tasks,
                                   int32_t* __restrict__ scratch,
-                                  BlockTask* __restrict__ out_tasks) {
+                                  BlockTask* __restrict__ out_tasks,
+                                  int64_t* __restrict__ ps /* [2k][2] to
+                                      zero, or null */) {
+  if (ps != nullptr) {
+    for (int i = threadIdx.x; i < 4 * k; i += blockDim.x) ps[i] = 0;
+  }
   __shared__ long long s_rpt;
   __shared__ int s_total_tasks;
   if (threadIdx.x == 0) {
@@ -187,24 +199,30 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
       scratch[3 * j + 1] = e;
       total += e - b;
     }
-    long long rpt =
+    // round rows/task UP to a power of two: all the per-node
+    // ceil-divides below become shifts (serial thread-0 64-bit
+    // divisions measured ~2x the whole kernel's budget), and the
+    // task layout has no effect on results (atomics are
+    // order-independent)
+    long long rpt0 =
         std::max(min_rows, (total + target_tasks - 1) / target_tasks);
-    // defensive: never overflow the task buffer (host bound should
-    // already guarantee this)
-    for (;;) {
+    int shift = 0;
+    while ((1LL << shift) < rpt0) ++shift;
+    for (;;) {  // defensive: never overflow the task buffer
       long long need = 0;
       for (int j = 0; j < k; ++j) {
         const long long sz = scratch[3 * j + 1] - scratch[3 * j];
-        need += (sz + rpt - 1) / rpt;
+        need += (sz + (1LL << shift) - 1) >> shift;
       }
       if (need <= max_tasks) break;
-      rpt <<= 1;
+      ++shift;
     }
+    const long long rpt = 1LL << shift;
     int pref = 0;
     for (int j = 0; j < k; ++j) {
       scratch[3 * j + 2] = pref;
       const long long sz = scratch[3 * j + 1] - scratch[3 * j];
-      pref += (int)((sz + rpt - 1) / rpt);
+      pref += (int)((sz + rpt - 1) >> shift);
     }
     s_rpt = rpt;
     s_total_tasks = pref;
@@ -229,58 +247,6 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     const int e = (int)std::min<long long>(b + rpt, scratch[3 * j + 1]);
     out_tasks[t] = BlockTask{j, b, e, 0};
   }
-}
-
-// Sum quantized gradient pairs over the BUILT children (reusing the
-// device-generated hist task array): needed because with the
-// device-chosen sibling the host does not know which child each hist
-// slot holds, so the evaluator's per-slot parent sums must also be
-// produced on device.
-__global__ __launch_bounds__(256) void SumPairsKernel(
-    const int32_t* __restrict__ qgpair, const int32_t* __restrict__ ridx,
-    const BlockTask* __restrict__ tasks, int64_t* __restrict__ ps) {
-  const BlockTask task = tasks[blockIdx.x];
-  if (task.row_begin >= task.row_end) return;
-  long long g = 0, h = 0;
-  for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
-       i += blockDim.x) {
-    const int row = ridx[i];
-    g += qgpair[2 * (size_t)row];
-    h += qgpair[2 * (size_t)row + 1];
-  }
-  __shared__ long long sg[256 / 64], sh[256 / 64];
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  for (int off = 32; off > 0; off >>= 1) {
-    g += __shfl_down(g, off, 64);
-    h += __shfl_down(h, off, 64);
-  }
-  if (lane == 0) {
-    sg[wave] = g;
-    sh[wave] = h;
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    long long tg = 0, th = 0;
-    for (int w = 0; w < (int)blockDim.x / 64; ++w) {
-      tg += sg[w];
-      th += sh[w];
-    }
-    if (tg) atomicAdd((unsigned long long*)&ps[2 * task.out_slot],
-                      (unsigned long long)tg);
-    if (th) atomicAdd((unsigned long long*)&ps[2 * task.out_slot + 1],
-                      (unsigned long long)th);
-  }
-}
-
-// subtracted sibling's sums = parent - built (slots kb..2kb-1)
-__global__ void DeriveSiblingSumsKernel(int64_t* __restrict__ ps,
-                                        const int64_t* __restrict__ parent_ps,
-                                        int kb) {
-  const int j = blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= kb) return;
-  ps[2 * (kb + j)] = parent_ps[2 * j] - ps[2 * j];
-  ps[2 * (kb + j) + 1] = parent_ps[2 * j + 1] - ps[2 * j + 1];
 }
 
 struct LeafSeg {
@@ -380,7 +346,8 @@ int gbt_grow_tree(
     gbt_hist(gidx8, gidx16, n_features, qgpair, ridx,
              (const BlockTask*)ctx->ring.dev[slot], (int)tasks.size(),
              hist_pool_a, n_bins, feat_group_start_dev, bin_group_start_dev,
-             n_groups, max_group_bins, cut_ptrs_dev, use_shared, stream);
+             n_groups, max_group_bins, cut_ptrs_dev, use_shared, nullptr,
+             stream);
     if (allreduce) allreduce((long long*)hist_pool_a, hist_row);
   }
 
@@ -676,30 +643,14 @@ int gbt_grow_tree(
     int max_tasks = (int)std::min<long long>(
         std::min<long long>(bound_total / hist_min_rows, hist_tasks) + kb + 1,
         hist_tasks_cap);
-    {
-      const int slot = ctx->ring.next();
-      size_t bytes = desc.size() * sizeof(int32_t);
-      if (int e = ctx->ring.ensure(slot, bytes)) return e;
-      memcpy(ctx->ring.host[slot], desc.data(), bytes);
-      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
-                               bytes, hipMemcpyHostToDevice, stream));
-      hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
-                         part_counters, (const int32_t*)ctx->ring.dev[slot],
-                         kb, hist_min_rows, hist_tasks, max_tasks, tg_scratch,
-                         hist_tasks_dev);
-    }
-    HIP_CHECK(hipMemsetAsync(next_pool, 0,
-                             (size_t)kb * hist_row * sizeof(int64_t), stream));
-    gbt_hist(gidx8, gidx16, n_features, qgpair, ridx, hist_tasks_dev,
-             max_tasks, next_pool, n_bins, feat_group_start_dev,
-             bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
-             use_shared, stream);
-    if (allreduce) {
-      allreduce((long long*)next_pool, (long long)kb * hist_row);
-    }
-    const int64_t* eval_ps_dev = nullptr;
+    // device-choice mode: stage parent pair-sums and give the hist
+    // kernel a device buffer to accumulate the built children's own
+    // sums into (zeroed by the task-gen kernel, completed by the
+    // subtraction kernel) — the evaluator then reads per-slot sums
+    // without the host ever knowing which sibling was built
+    int64_t* eval_ps = nullptr;
+    const int64_t* parent_ps_dev = nullptr;
     if (dev_choice) {
-      // built-child sums on device: [parent sums (H2D) | ps (memset)]
       const int slot = ctx->ring.next();
       size_t off_ps = ((size_t)kb * 2 * sizeof(int64_t) + 63) & ~63ULL;
       size_t bytes = off_ps + (size_t)kb * 4 * sizeof(int64_t);
@@ -713,15 +664,29 @@ int gbt_grow_tree(
                                (size_t)kb * 2 * sizeof(int64_t),
                                hipMemcpyHostToDevice, stream));
       char* d = (char*)ctx->ring.dev[slot];
-      int64_t* ps = (int64_t*)(d + off_ps);
-      HIP_CHECK(hipMemsetAsync(ps, 0, (size_t)kb * 4 * sizeof(int64_t),
-                               stream));
-      hipLaunchKernelGGL(SumPairsKernel, dim3(max_tasks), dim3(256), 0,
-                         stream, qgpair, ridx, hist_tasks_dev, ps);
-      hipLaunchKernelGGL(DeriveSiblingSumsKernel,
-                         dim3((kb + 255) / 256), dim3(256), 0, stream, ps,
-                         (const int64_t*)d, kb);
-      eval_ps_dev = ps;
+      parent_ps_dev = (const int64_t*)d;
+      eval_ps = (int64_t*)(d + off_ps);
+    }
+    {
+      const int slot = ctx->ring.next();
+      size_t bytes = desc.size() * sizeof(int32_t);
+      if (int e = ctx->ring.ensure(slot, bytes)) return e;
+      memcpy(ctx->ring.host[slot], desc.data(), bytes);
+      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
+                               bytes, hipMemcpyHostToDevice, stream));
+      hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
+                         part_counters, (const int32_t*)ctx->ring.dev[slot],
+                         kb, hist_min_rows, hist_tasks, max_tasks, tg_scratch,
+                         hist_tasks_dev, eval_ps);
+    }
+    HIP_CHECK(hipMemsetAsync(next_pool, 0,
+                             (size_t)kb * hist_row * sizeof(int64_t), stream));
+    gbt_hist(gidx8, gidx16, n_features, qgpair, ridx, hist_tasks_dev,
+             max_tasks, next_pool, n_bins, feat_group_start_dev,
+             bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
+             use_shared, eval_ps, stream);
+    if (allreduce) {
+      allreduce((long long*)next_pool, (long long)kb * hist_row);
     }
     {
       const int slot = ctx->ring.next();
@@ -736,7 +701,7 @@ int gbt_grow_tree(
       hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
                          stream, cur_pool, next_pool, sub_out,
                          (const int32_t*)ctx->ring.dev[slot], (int)hist_row,
-                         kb);
+                         kb, eval_ps, parent_ps_dev);
       for (int i = 0; i < (int)subtracted.size(); ++i) {
         subtracted[i]->hist_slot = kb + i;
       }
@@ -748,7 +713,7 @@ int gbt_grow_tree(
       for (Node* s : subtracted) eval_nodes.push_back(s);
     }
     if (int e = evaluate_enqueue(2 * kb, dev_choice ? nullptr : &eval_nodes,
-                                 next_pool, eval_ps_dev)) return e;
+                                 next_pool, eval_ps)) return e;
     // ---- the ONE sync for this level ----
     const int64_t* best;
     const int32_t* cnt;

@@ -30,7 +30,9 @@ void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
               const int32_t* bin_group_start,   // [n_groups+1] global bin idx
               int n_groups, int max_group_bins,
               const int32_t* cut_ptrs,          // [n_features+1]
-              int use_shared, hipStream_t stream);
+              int use_shared,
+              int64_t* node_sums,  // [n_slots,2] or null: per-slot pair sums
+              hipStream_t stream);
 
 void gbt_partition(const uint8_t* gidx8, const uint16_t* gidx16,
                    int n_features, const int32_t* ridx_in, int32_t* ridx_out,

@@ -32,7 +32,10 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
     int64_t* __restrict__ out_hist, int n_bins,
     const int32_t* __restrict__ feat_group_start,
     const int32_t* __restrict__ bin_group_start,
-    const int32_t* __restrict__ cut_ptrs) {
+    const int32_t* __restrict__ cut_ptrs,
+    int64_t* __restrict__ node_sums /* [k,2] or null: per-slot
+        gradient sums, accumulated by group-0 blocks (used by the
+        1-sync native driver for device-chosen siblings) */) {
   const BlockTask task = tasks[blockIdx.x];
   // device-generated task arrays are padded with empty tasks up to the
   // launched grid (task count is not host-known in the 1-sync driver);
@@ -70,11 +73,17 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
   unsigned long long* hist_s = smem;
   int64_t* hist_g = out_hist + (size_t)task.out_slot * n_bins * 2;
 
+  const bool do_sums = node_sums != nullptr && group == 0;
+  long long sum_g = 0, sum_h = 0;
   for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
        i += blockDim.x) {
     const int row = ridx[i];
     const long long g = qgpair[2 * (size_t)row];
     const long long h = qgpair[2 * (size_t)row + 1];
+    if (do_sums) {
+      sum_g += g;
+      sum_h += h;
+    }
     const BinT* rowbins = gidx + (size_t)row * n_features + f_begin;
     if (stage_meta) {
       for (int f = 0; f < gf; ++f) {
@@ -120,6 +129,33 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
       }
     }
   }
+  if (do_sums) {
+    for (int off = 32; off > 0; off >>= 1) {
+      sum_g += __shfl_down(sum_g, off, 64);
+      sum_h += __shfl_down(sum_h, off, 64);
+    }
+    __shared__ long long wg[GBT_HIST_BLOCK / 64];
+    __shared__ long long wh[GBT_HIST_BLOCK / 64];
+    const int lane = (int)threadIdx.x & 63;
+    const int wave = (int)threadIdx.x >> 6;
+    if (lane == 0) {
+      wg[wave] = sum_g;
+      wh[wave] = sum_h;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      long long tg = 0, th = 0;
+      for (int w = 0; w < (int)blockDim.x / 64; ++w) {
+        tg += wg[w];
+        th += wh[w];
+      }
+      if (tg) atomicAdd((unsigned long long*)&node_sums[2 * task.out_slot],
+                        (unsigned long long)tg);
+      if (th) atomicAdd(
+          (unsigned long long*)&node_sums[2 * task.out_slot + 1],
+          (unsigned long long)th);
+    }
+  }
 }
 
 extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
@@ -129,7 +165,8 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
                          const int32_t* feat_group_start,
                          const int32_t* bin_group_start, int n_groups,
                          int max_group_bins, const int32_t* cut_ptrs,
-                         int use_shared, hipStream_t stream) {
+                         int use_shared, int64_t* node_sums,
+                         hipStream_t stream) {
   static int block_size = [] {
     const char* e = getenv("GBT_HIST_BLOCK_SIZE");
     int v = e ? atoi(e) : GBT_HIST_BLOCK;
@@ -143,22 +180,24 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
       hipLaunchKernelGGL((HistKernel<uint8_t, true>), grid, block, shmem,
                          stream, gidx8, n_features, qgpair, ridx, tasks,
                          out_hist, n_bins, feat_group_start, bin_group_start,
-                         cut_ptrs);
+                         cut_ptrs, node_sums);
     } else {
       hipLaunchKernelGGL((HistKernel<uint8_t, false>), grid, block, 0, stream,
                          gidx8, n_features, qgpair, ridx, tasks, out_hist,
-                         n_bins, feat_group_start, bin_group_start, cut_ptrs);
+                         n_bins, feat_group_start, bin_group_start, cut_ptrs,
+                         node_sums);
     }
   } else {
     if (use_shared) {
       hipLaunchKernelGGL((HistKernel<uint16_t, true>), grid, block, shmem,
                          stream, gidx16, n_features, qgpair, ridx, tasks,
                          out_hist, n_bins, feat_group_start, bin_group_start,
-                         cut_ptrs);
+                         cut_ptrs, node_sums);
     } else {
       hipLaunchKernelGGL((HistKernel<uint16_t, false>), grid, block, 0, stream,
                          gidx16, n_features, qgpair, ridx, tasks, out_hist,
-                         n_bins, feat_group_start, bin_group_start, cut_ptrs);
+                         n_bins, feat_group_start, bin_group_start, cut_ptrs,
+                         node_sums);
     }
   }
 }
