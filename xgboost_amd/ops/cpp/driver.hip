@@ -419,54 +419,11 @@ int gbt_grow_tree(
   // enqueue partition of `nodes` (segments known on host); counters are
   // left in part_counters for the device-side hist task generation and
   // the piggybacked readback at the level sync
-  auto partition_enqueue = [&](std::vector<Node*>& nodes) -> int {
-    const int k = (int)nodes.size();
-    std::vector<BlockTask> tasks;
-    ChunkTasks(nodes, &tasks);
-    const int slot = ctx->ring.next();
-    size_t off_tasks = 0;
-    size_t off_feat = (tasks.size() * sizeof(BlockTask) + 7) & ~7ULL;
-    size_t off_sbin = (off_feat + (size_t)k * 4 + 7) & ~7ULL;
-    size_t off_dl = (off_sbin + (size_t)k * 4 + 7) & ~7ULL;
-    size_t off_cnt = (off_dl + (size_t)k + 7) & ~7ULL;
-    size_t bytes = off_cnt + (size_t)k * 2 * sizeof(int32_t);
-    if (int e = ctx->ring.ensure(slot, bytes)) return e;
-    char* h = (char*)ctx->ring.host[slot];
-    memcpy(h + off_tasks, tasks.data(), tasks.size() * sizeof(BlockTask));
-    int32_t* feat = (int32_t*)(h + off_feat);
-    int32_t* sbin = (int32_t*)(h + off_sbin);
-    uint8_t* dl = (uint8_t*)(h + off_dl);
-    int32_t* cnt = (int32_t*)(h + off_cnt);
-    for (int i = 0; i < k; ++i) {
-      Node* nd = nodes[i];
-      feat[i] = nd->feature;
-      sbin[i] = nd->bin - cut_ptrs_host[nd->feature];
-      dl[i] = (uint8_t)nd->dir;
-      cnt[2 * i] = nd->seg_begin;
-      cnt[2 * i + 1] = nd->seg_end;
-    }
-    HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
-                             hipMemcpyHostToDevice, stream));
-    // counter init goes to the persistent buffer (read by the task-gen
-    // kernel and by the level-sync readback)
-    HIP_CHECK(hipMemcpyAsync(part_counters, h + off_cnt,
-                             (size_t)k * 2 * sizeof(int32_t),
-                             hipMemcpyHostToDevice, stream));
-    char* d = (char*)ctx->ring.dev[slot];
-    gbt_partition(gidx8, gidx16, n_features, ridx, ridx_out,
-                  (const BlockTask*)(d + off_tasks), (int)tasks.size(),
-                  (const int32_t*)(d + off_feat),
-                  (const int32_t*)(d + off_sbin), (const uint8_t*)(d + off_dl),
-                  nullptr, nullptr, n_bins_feat_dev, part_counters, stream);
-    gbt_copy_ranges(ridx_out, ridx, (const BlockTask*)(d + off_tasks),
-                    (int)tasks.size(), stream);
-    return 0;
-  };
-
   // one D2H burst + ONE sync per level: best splits for `eval_nodes`
   // (may be empty on the final level) + partition counters for
   // `n_expand` nodes
   auto level_sync = [&](int n_eval, int n_expand,
+                        const int32_t* cnt_dev,
                         const int64_t** best_out,
                         const int32_t** cnt_out) -> int {
     size_t off_best = 0;
@@ -480,7 +437,7 @@ int gbt_grow_tree(
                                hipMemcpyDeviceToHost, stream));
     }
     if (n_expand > 0) {
-      HIP_CHECK(hipMemcpyAsync(h + off_cnt, part_counters,
+      HIP_CHECK(hipMemcpyAsync(h + off_cnt, cnt_dev,
                                (size_t)n_expand * 2 * sizeof(int32_t),
                                hipMemcpyDeviceToHost, stream));
     }
@@ -497,7 +454,7 @@ int gbt_grow_tree(
       return e;
     const int64_t* best;
     const int32_t* cnt;
-    if (int e = level_sync(1, 0, &best, &cnt)) return e;
+    if (int e = level_sync(1, 0, nullptr, &best, &cnt)) return e;
     parse_best(frontier, best);
   }
 
@@ -577,7 +534,45 @@ int gbt_grow_tree(
       next_level.push_back(rn);
     }
     const int n_expand = (int)expand.size();
-    if (int e = partition_enqueue(expand)) return e;
+    // Sibling to build vs subtract (exact int64 subtraction makes this
+    // a performance choice only, never a correctness one):
+    //  - single GPU, no monotone bounds: the DEVICE picks the
+    //    smaller-row child inside HistTaskGenKernel (exact local
+    //    counts) and the hist kernel accumulates its sums; the host
+    //    learns the choice from the counters at the level sync.
+    //  - distributed or monotone: the HOST picks by global hessian sum
+    //    (rank-identical, so every rank builds/allreduces the same
+    //    slot layout) and stages sums/bounds as usual.
+    const bool last = depth + 1 >= max_depth;
+    const bool dev_choice = (allreduce == nullptr) && !has_mono;
+    std::vector<Node*> build;
+    std::vector<int32_t> parent_slots;
+    std::vector<Node*> subtracted;
+    std::vector<int32_t> desc;  // [k][4] for HistTaskGenKernel
+    if (!last) {
+      desc.reserve(4 * n_expand);
+      for (int i = 0; i < n_expand; ++i) {
+        Node& ln = next_level[2 * i];
+        Node& rn = next_level[2 * i + 1];
+        parent_slots.push_back(expand[i]->hist_slot);
+        desc.push_back(expand[i]->seg_begin);
+        desc.push_back(expand[i]->seg_end);
+        desc.push_back(i);  // counter slot
+        if (dev_choice) {
+          desc.push_back(2);  // device picks the smaller child
+        } else {
+          Node* small = (ln.hq <= rn.hq) ? &ln : &rn;
+          Node* big = (small == &ln) ? &rn : &ln;
+          small->hist_slot = (int)build.size();
+          build.push_back(small);
+          subtracted.push_back(big);
+          desc.push_back(small == &ln ? 1 : 0);  // is_left
+        }
+      }
+    }
+    if (2 * n_expand > 2 * max_nodes_level) return -9999;
+    const int kb = n_expand;
+    const bool use_ps = !last && dev_choice;
 
     auto set_child_segs = [&](const int32_t* fin) {
       for (int i = 0; i < n_expand; ++i) {
@@ -590,10 +585,81 @@ int gbt_grow_tree(
       }
     };
 
-    if (depth + 1 >= max_depth) {
+    // ---- ONE staging upload for the whole level ----
+    // [partition tasks | feat | sbin | default_left | counters | desc |
+    //  parent_slots | parent pair-sums]  + (device-only) ps region.
+    // Counters live HERE (ring slot): partition atomics update them,
+    // the task-gen kernel reads them, and the level sync reads them
+    // back — no separate init copy, no persistent buffer.
+    std::vector<BlockTask> ptasks;
+    ChunkTasks(expand, &ptasks);
+    const int k = n_expand;
+    const int slot = ctx->ring.next();
+    size_t off_feat = (ptasks.size() * sizeof(BlockTask) + 7) & ~7ULL;
+    size_t off_sbin = (off_feat + (size_t)k * 4 + 7) & ~7ULL;
+    size_t off_dl = (off_sbin + (size_t)k * 4 + 7) & ~7ULL;
+    size_t off_cnt = (off_dl + (size_t)k + 7) & ~7ULL;
+    size_t off_desc = (off_cnt + (size_t)k * 8 + 7) & ~7ULL;
+    size_t off_pslots = (off_desc + desc.size() * 4 + 7) & ~7ULL;
+    size_t off_pps =
+        (off_pslots + parent_slots.size() * 4 + 63) & ~63ULL;
+    size_t upload_bytes =
+        off_pps + (use_ps ? (size_t)kb * 2 * sizeof(int64_t) : 0);
+    size_t off_ps = (upload_bytes + 63) & ~63ULL;
+    size_t total_bytes =
+        off_ps + (use_ps ? (size_t)kb * 4 * sizeof(int64_t) : 0);
+    if (int e = ctx->ring.ensure(slot, total_bytes)) return e;
+    char* h = (char*)ctx->ring.host[slot];
+    memcpy(h, ptasks.data(), ptasks.size() * sizeof(BlockTask));
+    {
+      int32_t* feat = (int32_t*)(h + off_feat);
+      int32_t* sbin = (int32_t*)(h + off_sbin);
+      uint8_t* dl = (uint8_t*)(h + off_dl);
+      int32_t* cnt = (int32_t*)(h + off_cnt);
+      for (int i = 0; i < k; ++i) {
+        Node* nd = expand[i];
+        feat[i] = nd->feature;
+        sbin[i] = nd->bin - cut_ptrs_host[nd->feature];
+        dl[i] = (uint8_t)nd->dir;
+        cnt[2 * i] = nd->seg_begin;
+        cnt[2 * i + 1] = nd->seg_end;
+      }
+      if (!desc.empty()) {
+        memcpy(h + off_desc, desc.data(), desc.size() * 4);
+      }
+      if (!parent_slots.empty()) {
+        memcpy(h + off_pslots, parent_slots.data(),
+               parent_slots.size() * 4);
+      }
+      if (use_ps) {
+        int64_t* pp = (int64_t*)(h + off_pps);
+        for (int i = 0; i < kb; ++i) {
+          pp[2 * i] = expand[i]->gq;
+          pp[2 * i + 1] = expand[i]->hq;
+        }
+      }
+    }
+    HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, upload_bytes,
+                             hipMemcpyHostToDevice, stream));
+    char* d = (char*)ctx->ring.dev[slot];
+    int32_t* cnt_dev = (int32_t*)(d + off_cnt);
+    int64_t* eval_ps = use_ps ? (int64_t*)(d + off_ps) : nullptr;
+    const int64_t* parent_ps_dev =
+        use_ps ? (const int64_t*)(d + off_pps) : nullptr;
+
+    gbt_partition(gidx8, gidx16, n_features, ridx, ridx_out,
+                  (const BlockTask*)d, (int)ptasks.size(),
+                  (const int32_t*)(d + off_feat),
+                  (const int32_t*)(d + off_sbin),
+                  (const uint8_t*)(d + off_dl), nullptr, nullptr,
+                  n_bins_feat_dev, cnt_dev, stream);
+    gbt_copy_ranges(ridx_out, ridx, (const BlockTask*)d,
+                    (int)ptasks.size(), stream);
+
+    if (last) {
       const int64_t* best;
       const int32_t* cnt;
-      if (int e = level_sync(0, n_expand, &best, &cnt)) return e;
+      if (int e = level_sync(0, n_expand, cnt_dev, &best, &cnt)) return e;
       set_child_segs(cnt);
       for (auto& nd : next_level) {
         leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end});
@@ -601,84 +667,16 @@ int gbt_grow_tree(
       level_nodes.clear();
       break;
     }
-    // Sibling to build vs subtract (exact int64 subtraction makes this
-    // a performance choice only, never a correctness one):
-    //  - single GPU, no monotone bounds: the DEVICE picks the
-    //    smaller-row child inside HistTaskGenKernel (exact local
-    //    counts); its per-slot sums for evaluation are computed on
-    //    device too (SumPairsKernel), and the host learns the choice
-    //    from the counters at the level sync.
-    //  - distributed or monotone: the HOST picks by global hessian sum
-    //    (rank-identical, so every rank builds/allreduces the same
-    //    slot layout) and stages sums/bounds as usual.
-    const bool dev_choice = (allreduce == nullptr) && !has_mono;
-    std::vector<Node*> build;
-    std::vector<int32_t> parent_slots;
-    std::vector<Node*> subtracted;
-    std::vector<int32_t> desc;  // [k][4] for HistTaskGenKernel
-    desc.reserve(4 * n_expand);
-    for (int i = 0; i < n_expand; ++i) {
-      Node& ln = next_level[2 * i];
-      Node& rn = next_level[2 * i + 1];
-      parent_slots.push_back(expand[i]->hist_slot);
-      desc.push_back(expand[i]->seg_begin);
-      desc.push_back(expand[i]->seg_end);
-      desc.push_back(i);  // counter slot
-      if (dev_choice) {
-        desc.push_back(2);  // device picks the smaller child
-      } else {
-        Node* small = (ln.hq <= rn.hq) ? &ln : &rn;
-        Node* big = (small == &ln) ? &rn : &ln;
-        small->hist_slot = (int)build.size();
-        build.push_back(small);
-        subtracted.push_back(big);
-        desc.push_back(small == &ln ? 1 : 0);  // is_left
-      }
-    }
-    if (2 * n_expand > 2 * max_nodes_level) return -9999;
-    const int kb = n_expand;
     // host bound on the device-generated task count
     long long bound_total = 0;
     for (Node* nd : expand) bound_total += nd->seg_end - nd->seg_begin;
     int max_tasks = (int)std::min<long long>(
         std::min<long long>(bound_total / hist_min_rows, hist_tasks) + kb + 1,
         hist_tasks_cap);
-    // device-choice mode: stage parent pair-sums and give the hist
-    // kernel a device buffer to accumulate the built children's own
-    // sums into (zeroed by the task-gen kernel, completed by the
-    // subtraction kernel) — the evaluator then reads per-slot sums
-    // without the host ever knowing which sibling was built
-    int64_t* eval_ps = nullptr;
-    const int64_t* parent_ps_dev = nullptr;
-    if (dev_choice) {
-      const int slot = ctx->ring.next();
-      size_t off_ps = ((size_t)kb * 2 * sizeof(int64_t) + 63) & ~63ULL;
-      size_t bytes = off_ps + (size_t)kb * 4 * sizeof(int64_t);
-      if (int e = ctx->ring.ensure(slot, bytes)) return e;
-      int64_t* pp = (int64_t*)ctx->ring.host[slot];
-      for (int i = 0; i < kb; ++i) {
-        pp[2 * i] = expand[i]->gq;
-        pp[2 * i + 1] = expand[i]->hq;
-      }
-      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
-                               (size_t)kb * 2 * sizeof(int64_t),
-                               hipMemcpyHostToDevice, stream));
-      char* d = (char*)ctx->ring.dev[slot];
-      parent_ps_dev = (const int64_t*)d;
-      eval_ps = (int64_t*)(d + off_ps);
-    }
-    {
-      const int slot = ctx->ring.next();
-      size_t bytes = desc.size() * sizeof(int32_t);
-      if (int e = ctx->ring.ensure(slot, bytes)) return e;
-      memcpy(ctx->ring.host[slot], desc.data(), bytes);
-      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
-                               bytes, hipMemcpyHostToDevice, stream));
-      hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
-                         part_counters, (const int32_t*)ctx->ring.dev[slot],
-                         kb, hist_min_rows, hist_tasks, max_tasks, tg_scratch,
-                         hist_tasks_dev, eval_ps);
-    }
+    hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
+                       cnt_dev, (const int32_t*)(d + off_desc), kb,
+                       hist_min_rows, hist_tasks, max_tasks, tg_scratch,
+                       hist_tasks_dev, eval_ps);
     HIP_CHECK(hipMemsetAsync(next_pool, 0,
                              (size_t)kb * hist_row * sizeof(int64_t), stream));
     gbt_hist(gidx8, gidx16, n_features, qgpair, ridx, hist_tasks_dev,
@@ -689,18 +687,12 @@ int gbt_grow_tree(
       allreduce((long long*)next_pool, (long long)kb * hist_row);
     }
     {
-      const int slot = ctx->ring.next();
-      size_t bytes = (size_t)kb * sizeof(int32_t);
-      if (int e = ctx->ring.ensure(slot, bytes)) return e;
-      memcpy(ctx->ring.host[slot], parent_slots.data(), bytes);
-      HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], ctx->ring.host[slot],
-                               bytes, hipMemcpyHostToDevice, stream));
       int64_t* sub_out = next_pool + (long long)kb * hist_row;
       const long long total = (long long)kb * hist_row;
       int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
       hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
                          stream, cur_pool, next_pool, sub_out,
-                         (const int32_t*)ctx->ring.dev[slot], (int)hist_row,
+                         (const int32_t*)(d + off_pslots), (int)hist_row,
                          kb, eval_ps, parent_ps_dev);
       for (int i = 0; i < (int)subtracted.size(); ++i) {
         subtracted[i]->hist_slot = kb + i;
@@ -717,7 +709,8 @@ int gbt_grow_tree(
     // ---- the ONE sync for this level ----
     const int64_t* best;
     const int32_t* cnt;
-    if (int e = level_sync(2 * kb, n_expand, &best, &cnt)) return e;
+    if (int e = level_sync(2 * kb, n_expand, cnt_dev, &best, &cnt))
+      return e;
     set_child_segs(cnt);
     if (dev_choice) {
       // recover the device's smaller-child choice from the counters
