@@ -453,6 +453,8 @@ class GpuOps(SegmentedOpsMixin):
                 "left", "right", "parent", "split_index", "split_cond",
                 "default_left", "loss_chg", "sum_hess", "base_weight")],
             self.hip.stream())
+        if rc in (-9998, -9999):
+            return None  # capacity guard tripped: python driver handles it
         if rc <= 0:
             raise RuntimeError(f"gbt_grow_tree failed: rc={rc}")
         n = rc

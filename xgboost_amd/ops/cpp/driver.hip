@@ -182,22 +182,29 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
   if (ps != nullptr) {
     for (int i = threadIdx.x; i < 4 * k; i += blockDim.x) ps[i] = 0;
   }
+  // parallel load phase: thread-0 doing k dependent global loads costs
+  // more than the whole rest of the kernel; gather child segments into
+  // LDS cooperatively first (k is capped by the driver's level guard)
+  __shared__ int s_begin[2048], s_end[2048];
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    const int pb = desc[4 * j], pe = desc[4 * j + 1];
+    const int split = counters[2 * desc[4 * j + 2]];
+    // mode: 1 = build left, 0 = build right, 2 = build the smaller
+    // child (single-GPU only: local row counts are rank-dependent)
+    int mode = desc[4 * j + 3];
+    if (mode == 2) mode = (split - pb) <= (pe - split) ? 1 : 0;
+    s_begin[j] = mode ? pb : split;
+    s_end[j] = mode ? split : pe;
+  }
   __shared__ long long s_rpt;
   __shared__ int s_total_tasks;
+  __syncthreads();
   if (threadIdx.x == 0) {
     long long total = 0;
     for (int j = 0; j < k; ++j) {
-      const int pb = desc[4 * j], pe = desc[4 * j + 1];
-      const int split = counters[2 * desc[4 * j + 2]];
-      // mode: 1 = build left, 0 = build right, 2 = build the smaller
-      // child (single-GPU only: local row counts are rank-dependent)
-      int mode = desc[4 * j + 3];
-      if (mode == 2) mode = (split - pb) <= (pe - split) ? 1 : 0;
-      const int b = mode ? pb : split;
-      const int e = mode ? split : pe;
-      scratch[3 * j] = b;
-      scratch[3 * j + 1] = e;
-      total += e - b;
+      scratch[3 * j] = s_begin[j];
+      scratch[3 * j + 1] = s_end[j];
+      total += s_end[j] - s_begin[j];
     }
     // round rows/task UP to a power of two: all the per-node
     // ceil-divides below become shifts (serial thread-0 64-bit
@@ -211,7 +218,7 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     for (;;) {  // defensive: never overflow the task buffer
       long long need = 0;
       for (int j = 0; j < k; ++j) {
-        const long long sz = scratch[3 * j + 1] - scratch[3 * j];
+        const long long sz = s_end[j] - s_begin[j];
         need += (sz + (1LL << shift) - 1) >> shift;
       }
       if (need <= max_tasks) break;
@@ -221,7 +228,7 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     int pref = 0;
     for (int j = 0; j < k; ++j) {
       scratch[3 * j + 2] = pref;
-      const long long sz = scratch[3 * j + 1] - scratch[3 * j];
+      const long long sz = s_end[j] - s_begin[j];
       pref += (int)((sz + rpt - 1) >> shift);
     }
     s_rpt = rpt;
@@ -243,8 +250,9 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     }
     const int j = lo;
     const long long idx = t - scratch[3 * j + 2];
-    const int b = scratch[3 * j] + (int)(idx * rpt);
-    const int e = (int)std::min<long long>(b + rpt, scratch[3 * j + 1]);
+    const int b = s_begin[j] + (int)(idx * rpt);
+    const int e = (int)std::min<long long>((long long)b + rpt,
+                                           (long long)s_end[j]);
     out_tasks[t] = BlockTask{j, b, e, 0};
   }
 }
@@ -570,7 +578,11 @@ int gbt_grow_tree(
         }
       }
     }
-    if (2 * n_expand > 2 * max_nodes_level) return -9999;
+    // capacity guards bound the HIST pool slots and the task-gen
+    // LDS arrays — the final level only partitions, so up to
+    // 2*max_nodes_level nodes may expand there without either
+    if (!last && 2 * n_expand > 2 * max_nodes_level) return -9999;
+    if (!last && n_expand > 2048) return -9998;  // task-gen LDS cap
     const int kb = n_expand;
     const bool use_ps = !last && dev_choice;
 
