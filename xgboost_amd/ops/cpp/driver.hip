@@ -612,7 +612,8 @@ int gbt_grow_tree(
     size_t off_dl = (off_sbin + (size_t)k * 4 + 7) & ~7ULL;
     size_t off_cnt = (off_dl + (size_t)k + 7) & ~7ULL;
     size_t off_desc = (off_cnt + (size_t)k * 8 + 7) & ~7ULL;
-    size_t off_pslots = (off_desc + desc.size() * 4 + 7) & ~7ULL;
+    size_t desc_bytes = desc.empty() ? (size_t)k * 8 : desc.size() * 4;
+    size_t off_pslots = (off_desc + desc_bytes + 7) & ~7ULL;
     size_t off_pps =
         (off_pslots + parent_slots.size() * 4 + 63) & ~63ULL;
     size_t upload_bytes =
@@ -638,6 +639,14 @@ int gbt_grow_tree(
       }
       if (!desc.empty()) {
         memcpy(h + off_desc, desc.data(), desc.size() * 4);
+      } else if (last) {
+        // final level: (left nid, right nid) per expand node for the
+        // direct leaf-position write
+        int32_t* kids = (int32_t*)(h + off_desc);
+        for (int i = 0; i < k; ++i) {
+          kids[2 * i] = next_level[2 * i].nid;
+          kids[2 * i + 1] = next_level[2 * i + 1].nid;
+        }
       }
       if (!parent_slots.empty()) {
         memcpy(h + off_pslots, parent_slots.data(),
@@ -659,6 +668,19 @@ int gbt_grow_tree(
     const int64_t* parent_ps_dev =
         use_ps ? (const int64_t*)(d + off_pps) : nullptr;
 
+    if (last) {
+      // final level: the children are all leaves — write their
+      // positions directly (one decide pass) instead of
+      // partition + copy + counter sync + a later leaf sweep
+      gbt_leaf_decide(gidx8, gidx16, n_features, ridx, (const BlockTask*)d,
+                      (int)ptasks.size(), (const int32_t*)(d + off_feat),
+                      (const int32_t*)(d + off_sbin),
+                      (const uint8_t*)(d + off_dl),
+                      (const int32_t*)(d + off_desc), n_bins_feat_dev,
+                      pos_out, stream);
+      level_nodes.clear();
+      break;
+    }
     gbt_partition(gidx8, gidx16, n_features, ridx, ridx_out,
                   (const BlockTask*)d, (int)ptasks.size(),
                   (const int32_t*)(d + off_feat),
@@ -667,18 +689,6 @@ int gbt_grow_tree(
                   n_bins_feat_dev, cnt_dev, stream);
     gbt_copy_ranges(ridx_out, ridx, (const BlockTask*)d,
                     (int)ptasks.size(), stream);
-
-    if (last) {
-      const int64_t* best;
-      const int32_t* cnt;
-      if (int e = level_sync(0, n_expand, cnt_dev, &best, &cnt)) return e;
-      set_child_segs(cnt);
-      for (auto& nd : next_level) {
-        leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end});
-      }
-      level_nodes.clear();
-      break;
-    }
     // host bound on the device-generated task count
     long long bound_total = 0;
     for (Node* nd : expand) bound_total += nd->seg_end - nd->seg_begin;

@@ -83,6 +83,15 @@ void gbt_predict(const float* X, int64_t n_rows, int n_features,
                  int32_t* out_leaf,            // [n_rows, n_trees] or null
                  hipStream_t stream);
 
+void gbt_leaf_decide(const uint8_t* gidx8, const uint16_t* gidx16,
+                     int n_features, const int32_t* ridx,
+                     const BlockTask* tasks, int n_tasks,
+                     const int32_t* split_feature,
+                     const int32_t* split_bin_local,
+                     const uint8_t* default_left, const int32_t* kids,
+                     const int32_t* n_bins_feat, int32_t* out_pos,
+                     hipStream_t stream);
+
 void gbt_leaf_partition(const int32_t* ridx, const BlockTask* tasks,
                         int n_tasks, const int32_t* leaf_ids,
                         int32_t* out_pos, hipStream_t stream);

@@ -205,6 +205,54 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void LeafPartitionKernel(
   }
 }
 
+template <typename BinT>
+__global__ __launch_bounds__(GBT_PART_BLOCK) void LeafDecideKernel(
+    const BinT* __restrict__ gidx, int n_features,
+    const int32_t* __restrict__ ridx, const BlockTask* __restrict__ tasks,
+    const int32_t* __restrict__ split_feature,
+    const int32_t* __restrict__ split_bin_local,
+    const uint8_t* __restrict__ default_left,
+    const int32_t* __restrict__ kids /* [k][2] = (left nid, right nid) */,
+    const int32_t* __restrict__ n_bins_feat, int32_t* __restrict__ out_pos) {
+  const BlockTask task = tasks[blockIdx.x];
+  if (task.row_begin >= task.row_end) return;
+  const int slot = task.out_slot;
+  const int feature = split_feature[slot];
+  const int sbin = split_bin_local[slot];
+  const bool dleft = default_left[slot] != 0;
+  const int fbins = n_bins_feat[feature];
+  const int lnid = kids[2 * slot], rnid = kids[2 * slot + 1];
+  for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
+       i += blockDim.x) {
+    const int row = ridx[i];
+    const int local = (int)gidx[(size_t)row * n_features + feature];
+    const bool left = DecideLeft(local, fbins, sbin, dleft, nullptr, 0);
+    out_pos[row] = left ? lnid : rnid;
+  }
+}
+
+extern "C" void gbt_leaf_decide(const uint8_t* gidx8, const uint16_t* gidx16,
+                                int n_features, const int32_t* ridx,
+                                const BlockTask* tasks, int n_tasks,
+                                const int32_t* split_feature,
+                                const int32_t* split_bin_local,
+                                const uint8_t* default_left,
+                                const int32_t* kids,
+                                const int32_t* n_bins_feat, int32_t* out_pos,
+                                hipStream_t stream) {
+  if (gidx8 != nullptr) {
+    hipLaunchKernelGGL((LeafDecideKernel<uint8_t>), dim3(n_tasks),
+                       dim3(GBT_PART_BLOCK), 0, stream, gidx8, n_features,
+                       ridx, tasks, split_feature, split_bin_local,
+                       default_left, kids, n_bins_feat, out_pos);
+  } else {
+    hipLaunchKernelGGL((LeafDecideKernel<uint16_t>), dim3(n_tasks),
+                       dim3(GBT_PART_BLOCK), 0, stream, gidx16, n_features,
+                       ridx, tasks, split_feature, split_bin_local,
+                       default_left, kids, n_bins_feat, out_pos);
+  }
+}
+
 extern "C" void gbt_leaf_partition(const int32_t* ridx, const BlockTask* tasks,
                                    int n_tasks, const int32_t* leaf_ids,
                                    int32_t* out_pos, hipStream_t stream) {
