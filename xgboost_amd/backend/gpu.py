@@ -340,7 +340,7 @@ class GpuOps(SegmentedOpsMixin):
     # -- native C++ level-loop driver (driver.hip) ----------------------
     def grow_tree_native(self, qgpair: torch.Tensor, tree, param,
                          quantizer, monotone: Optional[np.ndarray],
-                         root_sums: Tuple[int, int]):
+                         root_sums: torch.Tensor):
         """Run the whole per-tree loop in C++ (gbt_grow_tree).  Returns
         (tree, positions) or None when the config is unsupported (the
         Python driver handles those)."""
@@ -444,8 +444,8 @@ class GpuOps(SegmentedOpsMixin):
             max_build,
             self.hip.ptr(ws["counters"]), self.hip.ptr(ws["hist_tasks"]),
             ws["hist_tasks_cap"], self.hip.ptr(ws["tg_scratch"]),
+            self.hip.ptr(root_sums),
             quantizer.g_scale, quantizer.h_scale,
-            root_sums[0], root_sums[1],
             param.reg_lambda, param.reg_alpha, param.max_delta_step,
             param.min_child_weight, param.gamma, param.eta, param.max_depth,
             mono_dev, mono_host, cb,

@@ -168,9 +168,14 @@ class TreeGrower:
                 or p.colsample_bytree < 1.0 or p.colsample_bylevel < 1.0
                 or p.colsample_bynode < 1.0):
             return None
-        root_sums = ops.root_sum(qgpair)
+        # root sums stay on device: their readback piggybacks on the
+        # driver's root-eval sync (one fewer host round-trip per tree)
+        from . import collective
+        rs = qgpair.to(torch.int64).sum(dim=0)
+        if collective.is_distributed():
+            collective.allreduce_sum_(rs)
         out = ops.grow_tree_native(qgpair, tree, p, self.quantizer,
-                                   self.monotone, root_sums)
+                                   self.monotone, rs.contiguous())
         return out
 
     def _grow(self, qgpair: torch.Tensor, tree: RegTree

@@ -62,7 +62,8 @@ _SIGS = {
         _p, _p, _p, _p, _p, _p,  # eval_gain/bin/dir/lsum/best, pos_out
         _i,                      # max_nodes_level
         _p, _p, _i, _p,          # part_counters, hist_tasks_dev, cap, tg_scratch
-        _d, _d, _i64, _i64,      # scales, root sums
+        _p,                      # root_sums_dev [2] int64
+        _d, _d,                  # scales
         _d, _d, _d, _d, _d, _d,  # lambda, alpha, mds, mcw, gamma, eta
         _i,                      # max_depth
         _p, _p,                  # monotone dev/host

@@ -287,8 +287,9 @@ int gbt_grow_tree(
     BlockTask* hist_tasks_dev,   // [hist_tasks_cap]
     int hist_tasks_cap,
     int32_t* tg_scratch,         // [3 * max_nodes_level + 4]
+    const int64_t* root_sums_dev,  // [2] exact quantized (g, h) totals
     // scalars
-    double g_scale, double h_scale, long long root_gq, long long root_hq,
+    double g_scale, double h_scale,
     double reg_lambda, double reg_alpha, double max_delta_step,
     double min_child_weight, double gamma, double eta, int max_depth,
     const int8_t* monotone_dev, const int8_t* monotone_host,
@@ -320,13 +321,11 @@ int gbt_grow_tree(
   root.nid = 0;
   root.seg_begin = 0;
   root.seg_end = (int)n_rows;
-  root.gq = root_gq;
-  root.hq = root_hq;
   root.hist_slot = 0;
   root.lo = -INFINITY;
   root.hi = INFINITY;
-  out_base_weight[0] = (float)CalcWeight(root_gq * inv_g, root_hq * inv_h, p);
-  out_sum_hess[0] = (float)(root_hq * inv_h);
+  // root.gq/hq arrive with the root-eval readback (root_sums_dev);
+  // the evaluator reads the device copy directly
 
   // hist wants FEWER, BIGGER tasks than partition: every block pays a
   // full LDS zero+flush of the group histogram, so rows/task must
@@ -455,14 +454,25 @@ int gbt_grow_tree(
     return 0;
   };
 
-  // ---- root evaluation (root-only sync) ----
+  // ---- root evaluation (root-only sync; root sums ride along) ----
   {
     std::vector<Node*> frontier{&root};
-    if (int e = evaluate_enqueue(1, &frontier, hist_pool_a, nullptr))
+    if (int e = evaluate_enqueue(1, nullptr, hist_pool_a, root_sums_dev))
       return e;
+    const int rslot = ctx->ring.next();
+    if (int e = ctx->ring.ensure(rslot, 2 * sizeof(int64_t))) return e;
+    HIP_CHECK(hipMemcpyAsync(ctx->ring.host[rslot], root_sums_dev,
+                             2 * sizeof(int64_t), hipMemcpyDeviceToHost,
+                             stream));
     const int64_t* best;
     const int32_t* cnt;
     if (int e = level_sync(1, 0, nullptr, &best, &cnt)) return e;
+    const int64_t* rs = (const int64_t*)ctx->ring.host[rslot];
+    root.gq = rs[0];
+    root.hq = rs[1];
+    out_base_weight[0] =
+        (float)CalcWeight(root.gq * inv_g, root.hq * inv_h, p);
+    out_sum_hess[0] = (float)(root.hq * inv_h);
     parse_best(frontier, best);
   }
 
