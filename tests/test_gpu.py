@@ -324,7 +324,7 @@ def test_native_driver_matches_python_driver():
     gops = GpuOps(qm.to("cuda"))
     grower = TreeGrower(gops, param, quant, 30000)
     t_native = RegTree(10)
-    rs = gops.root_sum(qg)
+    rs = qg.to(torch.int64).sum(dim=0).contiguous()  # device-resident
     res = gops.grow_tree_native(qg, t_native, param, quant, None, rs)
     assert res is not None, "native driver refused a supported config"
     t_native, pos_native = res
