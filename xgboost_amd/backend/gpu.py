@@ -235,7 +235,7 @@ class GpuOps(SegmentedOpsMixin):
             [parents_np, mono_np, bounds_np, mask_np])
         self.lib.gbt_evaluate(
             self.hip.ptr(hist), k, self.n_bins, f,
-            self.hip.ptr(self.cut_ptrs), self.hip.ptr(parents),
+            self.hip.ptr(self.cut_ptrs), self.hip.ptr(parents), None,
             quantizer.g_scale, quantizer.h_scale,
             param.reg_lambda, param.reg_alpha, param.max_delta_step,
             param.min_child_weight, self.hip.ptr(mono_t),
@@ -426,6 +426,10 @@ class GpuOps(SegmentedOpsMixin):
 
             cb = self.hip.ALLREDUCE_FN(_allreduce)
         p8, p16 = self._gidx_ptrs()
+        ma = getattr(quantizer, "maxabs_dev", None)
+        out_scales = np.zeros(2, dtype=np.float64)
+        gsc = quantizer.g_scale if ma is None else 0.0
+        hsc = quantizer.h_scale if ma is None else 0.0
         rc = self.lib.gbt_grow_tree(
             ws["driver"], p8, p16, self.qm.n_features, n_rows,
             self.hip.ptr(qgpair),
@@ -445,7 +449,9 @@ class GpuOps(SegmentedOpsMixin):
             self.hip.ptr(ws["counters"]), self.hip.ptr(ws["hist_tasks"]),
             ws["hist_tasks_cap"], self.hip.ptr(ws["tg_scratch"]),
             self.hip.ptr(root_sums),
-            quantizer.g_scale, quantizer.h_scale,
+            self.hip.ptr(ma),
+            out_scales.ctypes.data_as(ctypes.c_void_p),
+            gsc, hsc,
             param.reg_lambda, param.reg_alpha, param.max_delta_step,
             param.min_child_weight, param.gamma, param.eta, param.max_depth,
             mono_dev, mono_host, cb,
@@ -457,6 +463,11 @@ class GpuOps(SegmentedOpsMixin):
             return None  # capacity guard tripped: python driver handles it
         if rc <= 0:
             raise RuntimeError(f"gbt_grow_tree failed: rc={rc}")
+        if ma is not None:
+            # host copies of the device-derived scales (bit-identical
+            # formula), for any later host-side use of this quantizer
+            quantizer.g_scale = float(out_scales[0])
+            quantizer.h_scale = float(out_scales[1])
         n = rc
         tree._ensure(n)
         tree.n_nodes = n

@@ -320,13 +320,15 @@ class Booster:
         m = ent["maxabs"]
         if collective.is_distributed():
             collective.allreduce_max_(m)
-        mh = m.cpu()
-        max_g, max_h = float(mh[0]), float(mh[1])
+        # scales are derived ON DEVICE (QuantizeKernel / root eval) and
+        # come back to the host with the tree driver's root sync — this
+        # path has no per-round max-abs readback sync at all
         quantizer = GradQuantizer.__new__(GradQuantizer)
-        quantizer.g_scale = (1 << 30) / max_g if max_g > 0 else 1.0
-        quantizer.h_scale = (1 << 30) / max_h if max_h > 0 else 1.0
-        lib.gbt_quantize(hip_ops.ptr(ent["gh"]), n, quantizer.g_scale,
-                         quantizer.h_scale, hip_ops.ptr(ent["qg"]),
+        quantizer.g_scale = None
+        quantizer.h_scale = None
+        quantizer.maxabs_dev = m
+        lib.gbt_quantize(hip_ops.ptr(ent["gh"]), n, 0.0, 0.0,
+                         hip_ops.ptr(m), hip_ops.ptr(ent["qg"]),
                          hip_ops.stream())
         seed = (self.seed + iteration if not self.seed_per_iteration
                 else self.seed + iteration * 2654435761)

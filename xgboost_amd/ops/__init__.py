@@ -30,7 +30,7 @@ _d = _c.c_double
 _SIGS = {
     "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p, _p],
     "gbt_partition": [_p, _p, _i, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p, _p, _p],
-    "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _d, _d, _d, _d, _d, _d,
+    "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _p, _d, _d, _d, _d, _d, _d,
                      _p, _p, _p, _p, _p, _p, _p, _p, _p],
     "gbt_compress": [_p, _i64, _i, _p, _p, _p, _f, _i, _p, _p, _p],
     "gbt_predict": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
@@ -41,7 +41,7 @@ _SIGS = {
     "gbt_mt_evaluate": [_p, _i, _i, _i, _i, _p, _p, _p, _p,
                         _d, _d, _d, _d, _p, _p, _p, _p, _p],
     "gbt_gpair_fused": [_i, _p, _p, _p, _f, _i64, _p, _p, _p],
-    "gbt_quantize": [_p, _i64, _d, _d, _p, _p],
+    "gbt_quantize": [_p, _i64, _d, _d, _p, _p, _p],
     "gbt_margin_add": [_p, _p, _p, _i64, _i, _i, _p],
     "gbt_hist_cpu": [_p, _p, _i, _p, _p, _p, _p, _i, _p, _p, _i],
     "gbt_partition_cpu": [_p, _p, _i, _p, _i64, _i64, _i, _i, _i, _p, _i,
@@ -63,6 +63,7 @@ _SIGS = {
         _i,                      # max_nodes_level
         _p, _p, _i, _p,          # part_counters, hist_tasks_dev, cap, tg_scratch
         _p,                      # root_sums_dev [2] int64
+        _p, _p,                  # maxabs_dev [2] f32 | out_scales [2] f64
         _d, _d,                  # scales
         _d, _d, _d, _d, _d, _d,  # lambda, alpha, mds, mcw, gamma, eta
         _i,                      # max_depth

@@ -288,6 +288,11 @@ int gbt_grow_tree(
     int hist_tasks_cap,
     int32_t* tg_scratch,         // [3 * max_nodes_level + 4]
     const int64_t* root_sums_dev,  // [2] exact quantized (g, h) totals
+    const float* maxabs_dev,  // null, or [2] gradient max-abs: scales are
+                              // derived on device for the root phase and
+                              // on host after the root sync (no separate
+                              // max-abs readback sync per round)
+    double* out_scales,       // [2] out: the derived (g_scale, h_scale)
     // scalars
     double g_scale, double h_scale,
     double reg_lambda, double reg_alpha, double max_delta_step,
@@ -303,7 +308,7 @@ int gbt_grow_tree(
   hipStream_t stream = (hipStream_t)stream_v;
   HostParams p{reg_lambda, reg_alpha, max_delta_step};
   const bool has_mono = monotone_host != nullptr;
-  const double inv_g = 1.0 / g_scale, inv_h = 1.0 / h_scale;
+  double inv_g = 1.0 / g_scale, inv_h = 1.0 / h_scale;
   const long long hist_row = (long long)n_bins * 2;
 
   {
@@ -363,8 +368,8 @@ int gbt_grow_tree(
   // from `nodes` unless a device-resident sum buffer is given
   // (device-chosen siblings compute their sums on the GPU).
   auto evaluate_enqueue = [&](int k, std::vector<Node*>* nodes,
-                              const int64_t* hists,
-                              const int64_t* ps_dev) -> int {
+                              const int64_t* hists, const int64_t* ps_dev,
+                              const float* maxabs_eval = nullptr) -> int {
     const int64_t* ps_arg = ps_dev;
     if (ps_dev == nullptr) {
       const int slot = ctx->ring.next();
@@ -391,6 +396,7 @@ int gbt_grow_tree(
       ps_arg = (const int64_t*)(d + off_ps);
       if (has_mono) {
         gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
+                     maxabs_eval,
                      g_scale, h_scale, reg_lambda, reg_alpha, max_delta_step,
                      min_child_weight, monotone_dev,
                      (const double*)(d + off_bd), nullptr, nullptr, eval_gain,
@@ -400,7 +406,8 @@ int gbt_grow_tree(
         return 0;
       }
     }
-    gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg, g_scale,
+    gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
+                 maxabs_eval, g_scale,
                  h_scale, reg_lambda, reg_alpha, max_delta_step,
                  min_child_weight, monotone_dev, nullptr, nullptr, nullptr,
                  eval_gain, eval_bin, eval_dir, eval_lsum, stream);
@@ -454,22 +461,40 @@ int gbt_grow_tree(
     return 0;
   };
 
-  // ---- root evaluation (root-only sync; root sums ride along) ----
+  // ---- root evaluation (root-only sync; root sums + max-abs ride
+  // along, so neither needs its own host round-trip) ----
   {
     std::vector<Node*> frontier{&root};
-    if (int e = evaluate_enqueue(1, nullptr, hist_pool_a, root_sums_dev))
+    if (int e = evaluate_enqueue(1, nullptr, hist_pool_a, root_sums_dev,
+                                 maxabs_dev))
       return e;
     const int rslot = ctx->ring.next();
-    if (int e = ctx->ring.ensure(rslot, 2 * sizeof(int64_t))) return e;
-    HIP_CHECK(hipMemcpyAsync(ctx->ring.host[rslot], root_sums_dev,
-                             2 * sizeof(int64_t), hipMemcpyDeviceToHost,
-                             stream));
+    if (int e = ctx->ring.ensure(rslot, 4 * sizeof(int64_t))) return e;
+    char* rh = (char*)ctx->ring.host[rslot];
+    HIP_CHECK(hipMemcpyAsync(rh, root_sums_dev, 2 * sizeof(int64_t),
+                             hipMemcpyDeviceToHost, stream));
+    if (maxabs_dev != nullptr) {
+      HIP_CHECK(hipMemcpyAsync(rh + 2 * sizeof(int64_t), maxabs_dev,
+                               2 * sizeof(float), hipMemcpyDeviceToHost,
+                               stream));
+    }
     const int64_t* best;
     const int32_t* cnt;
     if (int e = level_sync(1, 0, nullptr, &best, &cnt)) return e;
-    const int64_t* rs = (const int64_t*)ctx->ring.host[rslot];
+    const int64_t* rs = (const int64_t*)rh;
     root.gq = rs[0];
     root.hq = rs[1];
+    if (maxabs_dev != nullptr) {
+      const float* ma = (const float*)(rh + 2 * sizeof(int64_t));
+      g_scale = ma[0] > 0.f ? 1073741824.0 / (double)ma[0] : 1.0;
+      h_scale = ma[1] > 0.f ? 1073741824.0 / (double)ma[1] : 1.0;
+      inv_g = 1.0 / g_scale;
+      inv_h = 1.0 / h_scale;
+      if (out_scales != nullptr) {
+        out_scales[0] = g_scale;
+        out_scales[1] = h_scale;
+      }
+    }
     out_base_weight[0] =
         (float)CalcWeight(root.gq * inv_g, root.hq * inv_h, p);
     out_sum_hess[0] = (float)(root.hq * inv_h);

@@ -61,7 +61,10 @@ __device__ __forceinline__ long long ShflUpLL(long long v, int delta) {
 __global__ __launch_bounds__(64) void EvaluateKernel(
     const int64_t* __restrict__ hist, int n_nodes, int n_bins, int n_features,
     const int32_t* __restrict__ cut_ptrs,
-    const int64_t* __restrict__ parent_sums, double g_scale, double h_scale,
+    const int64_t* __restrict__ parent_sums,
+    const float* __restrict__ maxabs /* null, or [2]: derive scales on
+        device so no host sync is needed before the root evaluation */,
+    double g_scale, double h_scale,
     double reg_lambda, double reg_alpha, double max_delta_step,
     double min_child_weight, const int8_t* __restrict__ monotone,
     const double* __restrict__ node_bounds,
@@ -96,6 +99,11 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
   const int fb1 = cut_ptrs[f + 1];
   const long long pg = parent_sums[2 * node];
   const long long ph = parent_sums[2 * node + 1];
+  if (maxabs != nullptr) {
+    // identical derivation to the host/QuantizeKernel formula
+    g_scale = maxabs[0] > 0.f ? 1073741824.0 / (double)maxabs[0] : 1.0;
+    h_scale = maxabs[1] > 0.f ? 1073741824.0 / (double)maxabs[1] : 1.0;
+  }
   const double inv_g = 1.0 / g_scale;
   const double inv_h = 1.0 / h_scale;
 
@@ -260,7 +268,8 @@ extern "C" void gbt_select_best(const double* gain, const int32_t* bins,
 
 extern "C" void gbt_evaluate(
     const int64_t* hist, int n_nodes, int n_bins, int n_features,
-    const int32_t* cut_ptrs, const int64_t* parent_sums, double g_scale,
+    const int32_t* cut_ptrs, const int64_t* parent_sums,
+    const float* maxabs, double g_scale,
     double h_scale, double reg_lambda, double reg_alpha,
     double max_delta_step, double min_child_weight, const int8_t* monotone,
     const double* node_bounds, const uint8_t* feature_mask,
@@ -268,7 +277,8 @@ extern "C" void gbt_evaluate(
     uint8_t* out_dir, int64_t* out_lsum, hipStream_t stream) {
   dim3 grid(n_features > 65535 ? 65535 : n_features, n_nodes);
   hipLaunchKernelGGL(EvaluateKernel, grid, dim3(64), 0, stream, hist, n_nodes,
-                     n_bins, n_features, cut_ptrs, parent_sums, g_scale,
+                     n_bins, n_features, cut_ptrs, parent_sums, maxabs,
+                     g_scale,
                      h_scale, reg_lambda, reg_alpha, max_delta_step,
                      min_child_weight, monotone, node_bounds, feature_mask,
                      cat_feature, out_gain, out_bin, out_dir, out_lsum);

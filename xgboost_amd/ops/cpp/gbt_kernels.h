@@ -49,6 +49,8 @@ void gbt_partition(const uint8_t* gidx8, const uint16_t* gidx16,
 void gbt_evaluate(const int64_t* hist /* [n_nodes, n_bins, 2] */,
                   int n_nodes, int n_bins, int n_features,
                   const int32_t* cut_ptrs, const int64_t* parent_sums,
+                  const float* maxabs,  // null, or [2] device max-abs:
+                                        // scales derived in-kernel
                   double g_scale, double h_scale,
                   double reg_lambda, double reg_alpha, double max_delta_step,
                   double min_child_weight,

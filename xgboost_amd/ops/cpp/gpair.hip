@@ -77,7 +77,13 @@ __global__ __launch_bounds__(256) void GpairKernel(
 
 __global__ __launch_bounds__(256) void QuantizeKernel(
     const float* __restrict__ gh, long long n, double g_scale, double h_scale,
+    const float* __restrict__ maxabs /* null, or [2]: derive the scales
+        here so the host never has to read max-abs back */,
     int32_t* __restrict__ out) {
+  if (maxabs != nullptr) {
+    g_scale = maxabs[0] > 0.f ? 1073741824.0 / (double)maxabs[0] : 1.0;
+    h_scale = maxabs[1] > 0.f ? 1073741824.0 / (double)maxabs[1] : 1.0;
+  }
   const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   for (long long i = i0; i < n; i += (long long)gridDim.x * blockDim.x) {
     // round-half-to-even (llrint, FE_TONEAREST) matches torch.round
@@ -110,10 +116,11 @@ void gbt_gpair_fused(int objective, const float* margin, const float* label,
 }
 
 void gbt_quantize(const float* gh, long long n, double g_scale,
-                  double h_scale, int32_t* out, hipStream_t stream) {
+                  double h_scale, const float* maxabs, int32_t* out,
+                  hipStream_t stream) {
   const int blocks = (int)std::min<long long>((n + 255) / 256, 4096);
   hipLaunchKernelGGL(QuantizeKernel, dim3(blocks), dim3(256), 0, stream, gh,
-                     n, g_scale, h_scale, out);
+                     n, g_scale, h_scale, maxabs, out);
 }
 
 }  // extern "C"
