@@ -176,19 +176,19 @@ class TreeGrower:
             collective.allreduce_sum_(rs)
         out = ops.grow_tree_native(qgpair, tree, p, self.quantizer,
                                    self.monotone, rs.contiguous())
-        if out is None and getattr(self.quantizer, "g_scale", 1.0) is None:
-            # native path refused AND the quantizer was device-lazy:
-            # materialize the scales for the python driver (rare)
-            ma = self.quantizer.maxabs_dev.cpu()
-            mg, mh = float(ma[0]), float(ma[1])
-            self.quantizer.g_scale = (1 << 30) / mg if mg > 0 else 1.0
-            self.quantizer.h_scale = (1 << 30) / mh if mh > 0 else 1.0
         return out
 
     def _grow(self, qgpair: torch.Tensor, tree: RegTree
               ) -> Tuple[RegTree, torch.Tensor]:
         param = self.param
         ops = self.ops
+        if getattr(self.quantizer, "g_scale", 1.0) is None:
+            # the fused GPU path leaves scales on the device; the python
+            # driver needs host copies (same formula, one sync)
+            ma = self.quantizer.maxabs_dev.cpu()
+            mg, mh = float(ma[0]), float(ma[1])
+            self.quantizer.g_scale = (1 << 30) / mg if mg > 0 else 1.0
+            self.quantizer.h_scale = (1 << 30) / mh if mh > 0 else 1.0
         ops.reset(self.n_rows)
         node_sums: Dict[int, Tuple[int, int]] = {}  # exact int64 (gq, hq)
         node_bounds: Dict[int, Tuple[float, float]] = {0: (-np.inf, np.inf)}

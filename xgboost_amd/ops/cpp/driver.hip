@@ -259,6 +259,7 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
 
 struct LeafSeg {
   int nid, begin, end;
+  int parity;  // which ping-pong ridx buffer holds these rows
 };
 
 }  // namespace
@@ -316,6 +317,13 @@ int gbt_grow_tree(
     hipLaunchKernelGGL(IotaKernel, dim3(blocks), dim3(256), 0, stream, ridx,
                        n_rows);
   }
+  // ping-pong row-index buffers: the partition scatters cur -> alt and
+  // the buffers swap, so no copy-back pass is needed.  Rows of a node
+  // that stops expanding stay put in whichever buffer was current at
+  // that level (later partitions only write other, disjoint ranges),
+  // so each leaf records its buffer parity for the final position pass.
+  int32_t* cur_ridx = ridx;
+  int32_t* alt_ridx = ridx_out;
 
   int n_tree_nodes = 1;
   out_left[0] = -1;
@@ -508,12 +516,13 @@ int gbt_grow_tree(
   int64_t* next_pool = hist_pool_b;
 
   for (int depth = 0; depth < max_depth && !level_nodes.empty(); ++depth) {
+    const int parity = (cur_ridx == ridx) ? 0 : 1;
     std::vector<Node*> expand;
     for (auto& nd : level_nodes) {
       if (nd.gain > gamma && std::isfinite(nd.gain)) {
         expand.push_back(&nd);
       } else {
-        leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end});
+        leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end, parity});
       }
     }
     if (expand.empty()) {
@@ -707,7 +716,8 @@ int gbt_grow_tree(
       // final level: the children are all leaves — write their
       // positions directly (one decide pass) instead of
       // partition + copy + counter sync + a later leaf sweep
-      gbt_leaf_decide(gidx8, gidx16, n_features, ridx, (const BlockTask*)d,
+      gbt_leaf_decide(gidx8, gidx16, n_features, cur_ridx,
+                      (const BlockTask*)d,
                       (int)ptasks.size(), (const int32_t*)(d + off_feat),
                       (const int32_t*)(d + off_sbin),
                       (const uint8_t*)(d + off_dl),
@@ -716,14 +726,13 @@ int gbt_grow_tree(
       level_nodes.clear();
       break;
     }
-    gbt_partition(gidx8, gidx16, n_features, ridx, ridx_out,
+    gbt_partition(gidx8, gidx16, n_features, cur_ridx, alt_ridx,
                   (const BlockTask*)d, (int)ptasks.size(),
                   (const int32_t*)(d + off_feat),
                   (const int32_t*)(d + off_sbin),
                   (const uint8_t*)(d + off_dl), nullptr, nullptr,
                   n_bins_feat_dev, cnt_dev, stream);
-    gbt_copy_ranges(ridx_out, ridx, (const BlockTask*)d,
-                    (int)ptasks.size(), stream);
+    std::swap(cur_ridx, alt_ridx);  // children now live in cur_ridx
     // host bound on the device-generated task count
     long long bound_total = 0;
     for (Node* nd : expand) bound_total += nd->seg_end - nd->seg_begin;
@@ -736,7 +745,7 @@ int gbt_grow_tree(
                        hist_tasks_dev, eval_ps);
     HIP_CHECK(hipMemsetAsync(next_pool, 0,
                              (size_t)kb * hist_row * sizeof(int64_t), stream));
-    gbt_hist(gidx8, gidx16, n_features, qgpair, ridx, hist_tasks_dev,
+    gbt_hist(gidx8, gidx16, n_features, qgpair, cur_ridx, hist_tasks_dev,
              max_tasks, next_pool, n_bins, feat_group_start_dev,
              bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
              use_shared, eval_ps, stream);
@@ -790,8 +799,11 @@ int gbt_grow_tree(
     level_nodes.swap(next_level);
     std::swap(cur_pool, next_pool);
   }
-  for (auto& nd : level_nodes) {
-    leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end});
+  {
+    const int parity = (cur_ridx == ridx) ? 0 : 1;
+    for (auto& nd : level_nodes) {
+      leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end, parity});
+    }
   }
 
   // leaf values
@@ -800,30 +812,37 @@ int gbt_grow_tree(
       out_split_cond[nid] = (float)(out_base_weight[nid] * eta);
     }
   }
-  // leaf positions
-  {
+  // leaf positions: one sweep per ping-pong parity (a leaf's rows sit
+  // in whichever buffer was current when it stopped expanding)
+  for (int par = 0; par < 2; ++par) {
+    std::vector<const LeafSeg*> group;
+    for (const auto& lf : leaves) {
+      if (lf.parity == par) group.push_back(&lf);
+    }
+    if (group.empty()) continue;
     std::vector<BlockTask> tasks;
-    std::vector<Node> lnodes(leaves.size());
-    std::vector<Node*> lptrs(leaves.size());
-    for (size_t i = 0; i < leaves.size(); ++i) {
-      lnodes[i].seg_begin = leaves[i].begin;
-      lnodes[i].seg_end = leaves[i].end;
+    std::vector<Node> lnodes(group.size());
+    std::vector<Node*> lptrs(group.size());
+    for (size_t i = 0; i < group.size(); ++i) {
+      lnodes[i].seg_begin = group[i]->begin;
+      lnodes[i].seg_end = group[i]->end;
       lptrs[i] = &lnodes[i];
     }
     ChunkTasks(lptrs, &tasks);
     const int slot = ctx->ring.next();
     size_t off_tasks = 0;
     size_t off_ids = (tasks.size() * sizeof(BlockTask) + 7) & ~7ULL;
-    size_t bytes = off_ids + leaves.size() * sizeof(int32_t);
+    size_t bytes = off_ids + group.size() * sizeof(int32_t);
     if (int e = ctx->ring.ensure(slot, bytes)) return e;
     char* h = (char*)ctx->ring.host[slot];
     memcpy(h + off_tasks, tasks.data(), tasks.size() * sizeof(BlockTask));
     int32_t* ids = (int32_t*)(h + off_ids);
-    for (size_t i = 0; i < leaves.size(); ++i) ids[i] = leaves[i].nid;
+    for (size_t i = 0; i < group.size(); ++i) ids[i] = group[i]->nid;
     HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
                              hipMemcpyHostToDevice, stream));
     char* d = (char*)ctx->ring.dev[slot];
-    gbt_leaf_partition(ridx, (const BlockTask*)(d + off_tasks),
+    gbt_leaf_partition(par == 0 ? ridx : ridx_out,
+                       (const BlockTask*)(d + off_tasks),
                        (int)tasks.size(), (const int32_t*)(d + off_ids),
                        pos_out, stream);
   }
