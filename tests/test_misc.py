@@ -209,3 +209,44 @@ def test_feature_weights_bias_column_sampling():
     import pytest
     with pytest.raises(ValueError):
         xgb.DMatrix(X, label=y, feature_weights=np.ones(3))
+
+
+def test_public_api_surface_matches_reference():
+    """Every name in the reference package's __all__ resolves here
+    (reference python-package/xgboost/__init__.py)."""
+    import xgboost_amd as m
+    ref_all = ["Booster", "DMatrix", "DataIter", "ExtMemQuantileDMatrix",
+               "QuantileDMatrix", "RabitTracker", "XGBClassifier",
+               "XGBModel", "XGBRFClassifier", "XGBRFRegressor",
+               "XGBRanker", "XGBRegressor", "build_info", "collective",
+               "config_context", "cv", "get_config", "interpret",
+               "plot_importance", "plot_tree", "set_config", "to_graphviz",
+               "train"]
+    for name in ref_all:
+        assert getattr(m, name, None) is not None, name
+
+
+def test_interpret_shap_values_split():
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(300, 5).astype(np.float32)
+    y = (X[:, 0] + 0.5 * X[:, 1]).astype(np.float32)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 3}, d, 8)
+    values, bias = xgb.interpret.shap_values(bst, X)
+    assert values.shape == (300, 5)
+    margin = bst.predict(d, output_margin=True)
+    np.testing.assert_allclose(values.sum(axis=1) + bias, margin, atol=1e-4)
+
+
+def test_rabit_tracker_rendezvous():
+    import xgboost_amd as xgb
+    t = xgb.RabitTracker(n_workers=2)
+    t.start()
+    args = t.worker_args()
+    assert args["MASTER_ADDR"] == "127.0.0.1"
+    assert args["WORLD_SIZE"] == 2
+    assert args["MASTER_PORT"] == args["DMLC_TRACKER_PORT"]
+    t.wait_for()
+    t.free()

@@ -34,9 +34,12 @@ def build_info():
     }
 
 __all__ = [
-    "Booster", "DMatrix", "QuantileDMatrix", "train", "cv", "callback",
+    "Booster", "DMatrix", "QuantileDMatrix", "DataIter",
+    "ExtMemQuantileDMatrix", "train", "cv", "callback",
     "collective", "config_context", "set_config", "get_config",
-    "plot_importance", "plot_tree", "to_graphviz",
+    "plot_importance", "plot_tree", "to_graphviz", "interpret",
+    "RabitTracker", "XGBModel", "XGBRegressor", "XGBClassifier",
+    "XGBRanker", "XGBRFRegressor", "XGBRFClassifier", "build_info",
 ]
 
 
@@ -45,6 +48,15 @@ def _lazy(name):
                 "XGBRFClassifier", "XGBModel"):
         from . import sklearn as _sk
         return getattr(_sk, name)
+    if name in ("DataIter", "ExtMemQuantileDMatrix"):
+        from . import extmem as _em
+        return getattr(_em, name)
+    if name == "RabitTracker":
+        from .tracker import RabitTracker
+        return RabitTracker
+    if name == "interpret":
+        import importlib
+        return importlib.import_module(".interpret", __name__)
     raise AttributeError(name)
 
 
