@@ -173,3 +173,17 @@ def _run_workers_simple(world_size: int, script: str) -> None:
             outputs.append(stdout.decode())
             ok = ok and p.returncode == 0
         assert ok, "worker failed:\n" + "\n---\n".join(outputs)
+
+
+def test_collective_compat_surface_single_process():
+    """Reference collective API names exist and behave sanely without a
+    process group (reference python-package/xgboost/collective.py)."""
+    import numpy as np
+    from xgboost_amd import collective as c
+    a = c.allreduce(np.arange(3.0), c.Op.SUM)
+    assert np.array_equal(a, np.arange(3.0))
+    assert c.broadcast({"x": 1}, 0) == {"x": 1}
+    assert isinstance(c.get_processor_name(), str)
+    cfg = c.Config(tracker_host_ip="127.0.0.1")
+    assert cfg.tracker_host_ip == "127.0.0.1"
+    assert int(c.Op.MAX) == 0 and int(c.Op.SUM) == 2

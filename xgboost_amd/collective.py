@@ -32,11 +32,17 @@ def get_world_size() -> int:
 
 
 def init(backend: Optional[str] = None, **kwargs) -> None:
-    """Initialize from torchrun-style env vars (RANK/WORLD_SIZE/MASTER_*)."""
+    """Initialize from torchrun-style env vars (RANK/WORLD_SIZE/MASTER_*).
+
+    Reference-style dmlc_* / xgboost_* kwargs (tracker host, task id,
+    communicator choice) are accepted and ignored — rendezvous comes
+    from the env, and the communicator IS torch.distributed."""
     if is_distributed():
         return
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    kwargs = {k: v for k, v in kwargs.items()
+              if not k.lower().startswith(("dmlc_", "xgboost_"))}
     dist.init_process_group(backend=backend, **kwargs)
 
 
@@ -132,3 +138,101 @@ def check_synchronized(payload: bytes, what: str = "model") -> None:
     if ref != payload:
         raise RuntimeError(f"{what} differs across workers "
                            f"(rank {get_rank()})")
+
+
+# ---------------------------------------------------------------------------
+# Reference-API compatibility surface (python-package/xgboost/collective.py):
+# the names user code calls directly. All map onto torch.distributed.
+
+import dataclasses
+from enum import IntEnum
+from typing import Any, Optional as _Optional
+
+
+@dataclasses.dataclass
+class Config:
+    """Rendezvous configuration (reference collective.py:25). With
+    torch.distributed the retry/tracker fields are handled by the
+    store; kept for call-site compatibility."""
+
+    retry: _Optional[int] = None
+    timeout: _Optional[int] = None
+    tracker_host_ip: _Optional[str] = None
+    tracker_port: _Optional[int] = None
+    tracker_task_id: _Optional[str] = None
+
+
+class Op(IntEnum):
+    """Reduce operation for :func:`allreduce` (reference collective.py:264)."""
+
+    MAX = 0
+    MIN = 1
+    SUM = 2
+    BITWISE_AND = 3
+    BITWISE_OR = 4
+    BITWISE_XOR = 5
+
+
+_TORCH_OPS = None
+
+
+def _torch_op(op: "Op"):
+    global _TORCH_OPS
+    if _TORCH_OPS is None:
+        _TORCH_OPS = {
+            Op.MAX: dist.ReduceOp.MAX, Op.MIN: dist.ReduceOp.MIN,
+            Op.SUM: dist.ReduceOp.SUM, Op.BITWISE_AND: dist.ReduceOp.BAND,
+            Op.BITWISE_OR: dist.ReduceOp.BOR,
+            Op.BITWISE_XOR: dist.ReduceOp.BXOR,
+        }
+    return _TORCH_OPS[op]
+
+
+def allreduce(data, op: "Op"):
+    """In-place allreduce of a numpy array; returns it (reference
+    collective.py:275)."""
+    import numpy as _np
+    data = _np.asarray(data)
+    if not is_distributed():
+        return data
+    t = torch.from_numpy(data)
+    dist.all_reduce(t, op=_torch_op(op))
+    return data
+
+
+def broadcast(data, root: int):
+    """Broadcast any picklable object from `root` (reference
+    collective.py:190)."""
+    return broadcast_obj(data, root)
+
+
+def communicator_print(msg: Any) -> None:
+    """Print with rank prefix (reference collective.py:156)."""
+    print(f"[{get_rank()}] {msg}", flush=True)
+
+
+def get_processor_name() -> str:
+    import socket
+    return socket.gethostname()
+
+
+def signal_error() -> None:
+    """Abort the process group after an unrecoverable worker error."""
+    if is_distributed():
+        dist.destroy_process_group()
+    raise RuntimeError("collective worker signalled an error")
+
+
+class CommunicatorContext:
+    """`with CommunicatorContext(**args):` init/finalize wrapper
+    (reference collective.py:350)."""
+
+    def __init__(self, **args: Any) -> None:
+        self.args = args
+
+    def __enter__(self) -> dict:
+        init(**self.args)
+        return self.args
+
+    def __exit__(self, *exc) -> None:
+        finalize()
