@@ -250,3 +250,50 @@ def test_rabit_tracker_rendezvous():
     assert args["MASTER_PORT"] == args["DMLC_TRACKER_PORT"]
     t.wait_for()
     t.free()
+
+
+def test_dmatrix_info_accessors():
+    import numpy as np
+    from scipy import sparse as sp
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(50, 4).astype(np.float32)
+    X[2, 1] = np.nan
+    y = rng.rand(50).astype(np.float32)
+    w = rng.rand(50).astype(np.float32)
+    d = xgb.DMatrix(X, label=y, weight=w)
+    assert np.allclose(d.get_float_info("label"), y)
+    assert np.allclose(d.get_float_info("weight"), w)
+    d.set_float_info("base_margin", np.zeros(50, np.float32))
+    assert d.get_float_info("base_margin").shape == (50,)
+    assert d.num_nonmissing() == 50 * 4 - 1
+    csr = d.get_data()
+    assert sp.issparse(csr) and csr.shape == (50, 4)
+    assert csr.nnz == 50 * 4 - 1
+    # quantile cuts
+    d.quantized(32)
+    indptr, values = d.get_quantile_cut()
+    assert indptr.shape == (5,) and values.shape[0] == int(indptr[-1])
+    # group accessors
+    dq = xgb.DMatrix(X, label=y, qid=np.repeat(np.arange(5), 10))
+    assert np.array_equal(dq.get_group(), np.full(5, 10))
+    assert dq.get_uint_info("group_ptr").shape == (6,)
+
+
+def test_booster_load_config_roundtrip():
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(1)
+    X = rng.randn(100, 3).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 4,
+                     "eta": 0.21}, d, 2)
+    cfg = bst.save_config()
+    bst2 = xgb.Booster({"objective": "reg:squarederror"}, cache=[d])
+    bst2.load_config(cfg)
+    import json
+    c2 = json.loads(bst2.save_config())
+    tp = c2["learner"]["gradient_booster"]["tree_train_param"]
+    assert float(tp["eta"]) == 0.21
+    assert int(tp["max_depth"]) == 4

@@ -736,6 +736,40 @@ class Booster:
         self.tparam = make_train_param(self.raw_params)
         self.device = _resolve_device(self.raw_params)
 
+    def load_config(self, config: str) -> None:
+        """Apply a configuration produced by save_config (reference
+        Learner::LoadConfig, learner.cc:560): parameters only — model
+        state is untouched."""
+        cfg = json.loads(config)
+        learner = cfg.get("learner", {})
+        params = {}
+        params.update(learner.get("learner_train_param", {}))
+        gb = learner.get("gradient_booster", {})
+        params.update(gb.get("tree_train_param", {}))
+        obj = learner.get("objective", {})
+        if obj.get("name"):
+            params["objective"] = obj["name"]
+        gp = learner.get("generic_param", {})
+        if gp.get("device"):
+            params["device"] = gp["device"]
+        if gp.get("seed"):
+            params["seed"] = gp["seed"]
+        params.pop("booster", None)
+        self.set_param({k: v for k, v in params.items() if v != "None"})
+
+    def get_categories(self, export_to_arrow: bool = False):
+        """Per-feature category values recorded at train time
+        (reference Booster.get_categories); None when the model was
+        trained without categorical features."""
+        if export_to_arrow:
+            raise NotImplementedError("arrow export is not supported")
+        cats = getattr(self, "cat_categories_", None)
+        if not cats:
+            return None
+        names = self.feature_names
+        return {(names[f] if names else f"f{f}"): list(v)
+                for f, v in cats.items()}
+
     def save_config(self) -> str:
         """JSON dump of the effective configuration (reference
         Learner::SaveConfig, learner.cc:630)."""

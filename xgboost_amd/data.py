@@ -343,6 +343,88 @@ class DMatrix:
             self.info.feature_weights = fw
         self.info.validate()
 
+    # generic info accessors (reference core.py get_float_info/...)
+    _FLOAT_FIELDS = {"label": "labels", "weight": "weights",
+                     "base_margin": "base_margin",
+                     "label_lower_bound": "label_lower_bound",
+                     "label_upper_bound": "label_upper_bound"}
+
+    def get_float_info(self, field: str) -> np.ndarray:
+        if field in self._FLOAT_FIELDS:
+            v = getattr(self.info, self._FLOAT_FIELDS[field])
+            return (np.array([], dtype=np.float32) if v is None
+                    else np.asarray(v, dtype=np.float32).reshape(-1))
+        raise ValueError(f"unknown float field: {field}")
+
+    def set_float_info(self, field: str, data) -> None:
+        if field in self._FLOAT_FIELDS:
+            self.set_info(**{field: data})
+            return
+        raise ValueError(f"unknown float field: {field}")
+
+    set_float_info_npy2d = set_float_info
+
+    def get_uint_info(self, field: str) -> np.ndarray:
+        if field in ("group_ptr", "group"):
+            g = self.info.group_ptr
+            return (np.array([], dtype=np.uint32) if g is None
+                    else np.asarray(g, dtype=np.uint32))
+        raise ValueError(f"unknown uint field: {field}")
+
+    def set_uint_info(self, field: str, data) -> None:
+        if field == "group":
+            self.set_info(group=data)
+            return
+        raise ValueError(f"unknown uint field: {field}")
+
+    def get_group(self) -> np.ndarray:
+        g = self.info.group_ptr
+        return (np.array([], dtype=np.int64) if g is None
+                else np.diff(np.asarray(g, dtype=np.int64)))
+
+    def get_data(self):
+        """The feature matrix as scipy CSR (reference DMatrix.get_data);
+        missing entries are absent from the CSR."""
+        from scipy import sparse as sp
+        if self._sparse_data is not None:
+            return self._sparse_data.copy()
+        if self._data is None:
+            raise RuntimeError("no host feature data available")
+        X = self._data
+        mask = ~np.isnan(X) if np.isnan(self.missing) else X != self.missing
+        return _dense_to_csr(X, mask)
+
+    def num_nonmissing(self) -> int:
+        if self._sparse_data is not None:
+            return int(self._sparse_data.nnz)
+        if self._data is None:
+            return self.num_row() * self.num_col()
+        X = self._data
+        mask = ~np.isnan(X) if np.isnan(self.missing) else X != self.missing
+        return int(mask.sum())
+
+    def get_quantile_cut(self):
+        """(indptr, values) of the quantile cuts (reference
+        DMatrix.get_quantile_cut); requires a quantized matrix."""
+        cuts = self.cached_cuts()
+        if cuts is None:
+            if not self._quantized:
+                raise RuntimeError(
+                    "no quantile cuts yet: train or quantize first")
+            cuts = next(iter(self._quantized.values())).cuts
+        return (np.asarray(cuts.ptrs, dtype=np.uint64),
+                np.asarray(cuts.values, dtype=np.float32))
+
+    def get_categories(self, export_to_arrow: bool = False):
+        if export_to_arrow:
+            raise NotImplementedError("arrow export is not supported")
+        cats = getattr(self, "categories_", None)
+        if not cats:
+            return None
+        names = self.info.feature_names
+        return {(names[f] if names else f"f{f}"): list(v)
+                for f, v in cats.items()}
+
     set_label = lambda self, label: self.set_info(label=label)
     set_weight = lambda self, weight: self.set_info(weight=weight)
     set_base_margin = lambda self, m: self.set_info(base_margin=m)
@@ -591,3 +673,11 @@ def align_categories(X: np.ndarray, pred_cats: Dict[int, list],
 
 # deprecated alias kept for API compatibility (reference: core.py)
 DeviceQuantileDMatrix = QuantileDMatrix
+
+
+def _dense_to_csr(X: np.ndarray, mask: np.ndarray):
+    from scipy import sparse as sp
+    indptr = np.zeros(X.shape[0] + 1, dtype=np.int64)
+    np.cumsum(mask.sum(axis=1), out=indptr[1:])
+    rows, cols = np.nonzero(mask)
+    return sp.csr_matrix((X[rows, cols], cols, indptr), shape=X.shape)
