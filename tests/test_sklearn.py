@@ -135,3 +135,30 @@ def test_unfitted_raises():
     m = XGBRegressor()
     with pytest.raises(ValueError):
         m.get_booster()
+
+
+def test_sklearn_extended_surface():
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(200, 4).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    m = xgb.XGBRegressor(n_estimators=5, booster="gblinear")
+    m.fit(X, y, verbose=False)
+    assert m.coef_.shape == (4,)
+    assert m.get_num_boosting_rounds() == 5
+    # best_iteration raises without early stopping
+    import pytest
+    with pytest.raises(AttributeError):
+        _ = m.best_iteration
+    m2 = xgb.XGBRegressor(n_estimators=50, early_stopping_rounds=3)
+    m2.fit(X, y, eval_set=[(X, y)], verbose=False)
+    assert isinstance(m2.best_iteration, int)
+    assert np.isfinite(m2.best_score)
+    # ranker score
+    q = np.repeat(np.arange(10), 20)
+    yr = (rng.rand(200) * 3).astype(np.float32)
+    r = xgb.XGBRanker(n_estimators=4)
+    r.fit(X, yr, qid=q, verbose=False)
+    s = r.score(X, yr, qid=q)
+    assert 0.0 <= s <= 1.0

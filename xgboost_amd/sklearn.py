@@ -170,13 +170,9 @@ class XGBModel:
         return "reg:squarederror"
 
     def _set_fitted_attrs(self, dtrain: DMatrix) -> None:
-        b = self.get_booster()
+        # best_iteration/best_score/feature_names_in_ are properties
+        # derived from the booster
         self.n_features_in_ = dtrain.num_col()
-        if dtrain.feature_names:
-            self.feature_names_in_ = np.asarray(dtrain.feature_names)
-        if b.best_iteration is not None:
-            self.best_iteration = b.best_iteration
-            self.best_score = b.best_score
 
     def _iteration_range(self) -> Tuple[int, int]:
         b = self.get_booster()
@@ -218,6 +214,46 @@ class XGBModel:
     @property
     def intercept_(self) -> np.ndarray:
         return np.array([self.get_booster().base_score], dtype=np.float32)
+
+    @property
+    def coef_(self) -> np.ndarray:
+        """Linear coefficients (gblinear boosters only, like upstream)."""
+        if getattr(self, "booster", None) not in ("gblinear",):
+            raise AttributeError(
+                "coef_ is only defined for booster=gblinear")
+        b = self.get_booster()
+        j = b._model_to_json()
+        w = np.array(j["learner"]["gradient_booster"]["model"]["weights"],
+                     dtype=np.float32)
+        n = self.n_features_in_
+        w = w.reshape(n + 1, -1)[:n]
+        return w[:, 0] if w.shape[1] == 1 else w.T
+
+    @property
+    def best_iteration(self) -> int:
+        bi = self.get_booster().best_iteration
+        if bi is None:
+            raise AttributeError(
+                "best_iteration is only defined when early stopping is used")
+        return bi
+
+    @property
+    def best_score(self) -> float:
+        bs = self.get_booster().best_score
+        if bs is None:
+            raise AttributeError(
+                "best_score is only defined when early stopping is used")
+        return bs
+
+    @property
+    def feature_names_in_(self) -> np.ndarray:
+        names = self.get_booster().feature_names
+        if names is None:
+            raise AttributeError("feature names are not available")
+        return np.array(names, dtype=object)
+
+    def get_num_boosting_rounds(self) -> int:
+        return self.n_estimators_
 
     def save_model(self, fname: str) -> None:
         self.get_booster().save_model(fname)
@@ -324,6 +360,15 @@ class XGBRanker(XGBModel):
             xgb_model=xgb_model)
         self._set_fitted_attrs(dtrain)
         return self
+
+    def score(self, X, y, qid=None):
+        """Mean NDCG over query groups (reference sklearn.py
+        XGBRanker.score semantics)."""
+        from .metrics import create_metric
+        d = self._make_dmatrix(X, y, qid=qid) if qid is not None \
+            else self._make_dmatrix(X, y)
+        margin = self.get_booster().predict(d, output_margin=True)
+        return float(create_metric("ndcg")(margin, d.info))
 
 
 class XGBRFRegressor(XGBRegressor):
