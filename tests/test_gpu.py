@@ -494,3 +494,38 @@ def test_gpu_colsample_matches_cpu():
         assert tg.n_nodes == tc.n_nodes
         assert np.array_equal(tg.split_index[:tg.n_nodes],
                               tc.split_index[:tc.n_nodes])
+
+
+def test_native_driver_distributed_branch_rccl_world1():
+    """Initialize a 1-rank RCCL process group so the native driver takes
+    its DISTRIBUTED branch (global-hessian sibling choice, host-staged
+    eval sums, allreduce callback into torch.distributed) — the exact
+    code the multi-GPU scaling run exercises — and check the trees
+    equal the non-distributed CPU oracle."""
+    import os
+    import torch.distributed as dist
+    from xgboost_amd import collective
+
+    X, y = _data(20000, 8, seed=31)
+    pd = {"objective": "binary:logistic", "max_depth": 6, "max_bin": 64,
+          "seed": 2}
+    dc = xgb.DMatrix(X, label=y)
+    bc = xgb.train(pd, dc, 6)  # CPU reference first (no process group)
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29771")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        assert collective.is_distributed()
+        dg = xgb.DMatrix(X, label=y)
+        bg = xgb.train({**pd, "device": "cuda"}, dg, 6)
+        for tg, tc in zip(bg.trees, bc.trees):
+            assert tg.n_nodes == tc.n_nodes
+            assert np.array_equal(tg.split_index[:tg.n_nodes],
+                                  tc.split_index[:tc.n_nodes])
+            assert np.array_equal(tg.left[:tg.n_nodes],
+                                  tc.left[:tc.n_nodes])
+    finally:
+        dist.destroy_process_group()
