@@ -242,12 +242,12 @@ class GpuOps(SegmentedOpsMixin):
             self.hip.ptr(bounds_t), self.hip.ptr(mask_t),
             self.hip.ptr(self.cat_feature),
             self.hip.ptr(gain), self.hip.ptr(bins), self.hip.ptr(dirs),
-            self.hip.ptr(lsum), self.hip.stream())
+            self.hip.ptr(lsum), None, self.hip.stream())
         # device-side per-node argmax + packing -> ONE D2H sync
         out_best = torch.empty((k, 6), dtype=torch.int64, device=dev)
         self.lib.gbt_select_best(
             self.hip.ptr(gain), self.hip.ptr(bins), self.hip.ptr(dirs),
-            self.hip.ptr(lsum), k, f, self.hip.ptr(out_best),
+            self.hip.ptr(lsum), k, f, self.hip.ptr(out_best), None,
             self.hip.stream())
         host = out_best.cpu().numpy()
         sel_gain = host[:, 0].view(np.float64)
@@ -392,6 +392,20 @@ class GpuOps(SegmentedOpsMixin):
                                           dtype=torch.int32, device=dev),
                 "driver": self.lib.gbt_driver_create(),
             }
+            # whole-tree mode arena (single sync per tree): per-level
+            # best/segment records + device-side partition/task args
+            pool = 2 * max_build
+            if pool <= 2048:
+                rec = 1 + param.max_depth * pool
+                mp = n_rows // 1024 + pool + 2
+                wt_bytes = (48 * rec + 8 * rec + 8 * (param.max_depth + 2)
+                            + 96 * pool + 16 * mp + 32 * 1024)
+                ws["wt_max_ptasks"] = mp
+                ws["wt_ws"] = torch.empty(wt_bytes, dtype=torch.uint8,
+                                          device=dev)
+            else:
+                ws["wt_max_ptasks"] = 0
+                ws["wt_ws"] = None
             self._native_ws = ws
         cap = 1 << (param.max_depth + 1)
         host = {name: np.zeros(cap, dt) for name, dt in [
@@ -448,6 +462,9 @@ class GpuOps(SegmentedOpsMixin):
             max_build,
             self.hip.ptr(ws["counters"]), self.hip.ptr(ws["hist_tasks"]),
             ws["hist_tasks_cap"], self.hip.ptr(ws["tg_scratch"]),
+            self.hip.ptr(ws["wt_ws"]),
+            0 if ws["wt_ws"] is None else ws["wt_ws"].numel(),
+            ws["wt_max_ptasks"],
             self.hip.ptr(root_sums),
             self.hip.ptr(ma),
             out_scales.ctypes.data_as(ctypes.c_void_p),

@@ -31,13 +31,13 @@ _SIGS = {
     "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p, _p],
     "gbt_partition": [_p, _p, _i, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p, _p, _p],
     "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _p, _d, _d, _d, _d, _d, _d,
-                     _p, _p, _p, _p, _p, _p, _p, _p, _p],
+                     _p, _p, _p, _p, _p, _p, _p, _p, _p, _p],
     "gbt_compress": [_p, _i64, _i, _p, _p, _p, _f, _i, _p, _p, _p],
     "gbt_predict": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
                     _p, _p, _i, _i, _p, _p, _p],
     "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
     "gbt_copy_ranges": [_p, _p, _p, _i, _p],
-    "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p],
+    "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p, _p],
     "gbt_mt_evaluate": [_p, _i, _i, _i, _i, _p, _p, _p, _p,
                         _d, _d, _d, _d, _p, _p, _p, _p, _p],
     "gbt_gpair_fused": [_i, _p, _p, _p, _f, _i64, _p, _p, _p],
@@ -62,6 +62,7 @@ _SIGS = {
         _p, _p, _p, _p, _p, _p,  # eval_gain/bin/dir/lsum/best, pos_out
         _i,                      # max_nodes_level
         _p, _p, _i, _p,          # part_counters, hist_tasks_dev, cap, tg_scratch
+        _p, _i64, _i,            # wt_ws, wt_ws_bytes, wt_max_ptasks
         _p,                      # root_sums_dev [2] int64
         _p, _p,                  # maxabs_dev [2] f32 | out_scales [2] f64
         _d, _d,                  # scales

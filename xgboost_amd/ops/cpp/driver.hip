@@ -144,7 +144,12 @@ __global__ void SubtractHistKernel(const int64_t* __restrict__ parents,
                                    const int32_t* __restrict__ parent_slot,
                                    int n_bins2, int k,
                                    int64_t* __restrict__ ps,
-                                   const int64_t* __restrict__ parent_ps) {
+                                   const int64_t* __restrict__ parent_ps,
+                                   const int32_t* __restrict__ kp_dev) {
+  if (kp_dev != nullptr) {
+    k = *kp_dev;
+    out += (long long)k * n_bins2;  // subtracted slots follow the built
+  }
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   // piggybacked: subtracted-sibling sums = parent - built (slots k..2k-1)
   if (ps != nullptr && idx < k) {
@@ -178,7 +183,13 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
                                   int32_t* __restrict__ scratch,
                                   BlockTask* __restrict__ out_tasks,
                                   int64_t* __restrict__ ps /* [2k][2] to
-                                      zero, or null */) {
+                                      zero, or null */,
+                                  const int32_t* __restrict__ kp_dev,
+                                  int32_t* __restrict__ seg_out
+                                  /* null, or [2k][2] child segments in
+                                     SLOT order: built j, subtracted
+                                     k+j (whole-tree mode) */) {
+  if (kp_dev != nullptr) k = *kp_dev;
   if (ps != nullptr) {
     for (int i = threadIdx.x; i < 4 * k; i += blockDim.x) ps[i] = 0;
   }
@@ -195,6 +206,13 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     if (mode == 2) mode = (split - pb) <= (pe - split) ? 1 : 0;
     s_begin[j] = mode ? pb : split;
     s_end[j] = mode ? split : pe;
+    if (seg_out != nullptr) {
+      seg_out[2 * j] = s_begin[j];
+      seg_out[2 * j + 1] = s_end[j];
+      // the subtracted sibling is the complement range
+      seg_out[2 * (k + j)] = mode ? split : pb;
+      seg_out[2 * (k + j) + 1] = mode ? pe : split;
+    }
   }
   __shared__ long long s_rpt;
   __shared__ int s_total_tasks;
@@ -257,6 +275,134 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
   }
 }
 
+// Whole-tree mode: apply one level's split decisions entirely on
+// device.  Reads the level's best-split records and segments, compacts
+// the expanding nodes IN PAIR ORDER (left0, right0, left1, ...) so the
+// host replay and the python driver number children identically, and
+// emits everything the next phase chain needs: partition args + padded
+// task list, counters, task-gen descriptors, and the gathered parent
+// pair-sums for the sibling-subtraction derivation.
+__global__ __launch_bounds__(256) void ApplyKernel(
+    const int64_t* __restrict__ best,    // [kn][6] this level's nodes
+    const int32_t* __restrict__ segs,    // [kn][2] slot order
+    const int32_t* __restrict__ kn_dev,  // node count of this level
+    const int32_t* __restrict__ kp_prev_dev,  // pair count (0 at root)
+    const int64_t* __restrict__ ps_prev,      // [kn][2] node pair sums
+    double gamma_, const int32_t* __restrict__ cut_ptrs,
+    long long min_rows, long long target_tasks, int max_ptasks,
+    int32_t* __restrict__ kn_out, int32_t* __restrict__ kp_out,
+    int32_t* __restrict__ feat, int32_t* __restrict__ sbin,
+    uint8_t* __restrict__ dl, int32_t* __restrict__ cnt,
+    BlockTask* __restrict__ out_tasks, int32_t* __restrict__ desc,
+    int64_t* __restrict__ parent_ps, int32_t* __restrict__ parent_slot) {
+  __shared__ int s_order[2048];   // slots in pair order
+  __shared__ uint8_t s_flag[2048];
+  __shared__ int s_exp[2048];     // expanding slots (pair order)
+  __shared__ int s_b[2048], s_e[2048];
+  __shared__ int s_kb;
+  const int kn = *kn_dev;
+  const int kp_prev = *kp_prev_dev;
+  // pair-order slot sequence: pair p = slots (p, kp_prev + p); left is
+  // the one whose end == the other's begin
+  if (kp_prev == 0) {
+    if (threadIdx.x == 0) s_order[0] = 0;  // root
+  } else {
+    for (int p = threadIdx.x; p < kp_prev; p += blockDim.x) {
+      const int a = p, b = kp_prev + p;
+      const bool a_left = segs[2 * a + 1] == segs[2 * b];
+      s_order[2 * p] = a_left ? a : b;
+      s_order[2 * p + 1] = a_left ? b : a;
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < kn; i += blockDim.x) {
+    const int slot = s_order[i];
+    const double gv = __longlong_as_double(best[6 * slot]);
+    const long long bin = best[6 * slot + 1];
+    s_flag[i] = (bin >= 0 && isfinite(gv) && gv > gamma_) ? 1 : 0;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int kb = 0;
+    for (int i = 0; i < kn; ++i) {
+      if (s_flag[i]) s_exp[kb++] = s_order[i];
+    }
+    s_kb = kb;
+    *kp_out = kb;
+    *kn_out = 2 * kb;
+  }
+  __syncthreads();
+  const int kb = s_kb;
+  for (int j = threadIdx.x; j < kb; j += blockDim.x) {
+    const int e = s_exp[j];
+    const int f = (int)best[6 * e + 5];
+    const int sb = segs[2 * e], se = segs[2 * e + 1];
+    feat[j] = f;
+    sbin[j] = (int)best[6 * e + 1] - cut_ptrs[f];
+    dl[j] = (uint8_t)best[6 * e + 2];
+    cnt[2 * j] = sb;
+    cnt[2 * j + 1] = se;
+    desc[4 * j] = sb;
+    desc[4 * j + 1] = se;
+    desc[4 * j + 2] = j;
+    desc[4 * j + 3] = 2;  // device picks the smaller child
+    parent_ps[2 * j] = ps_prev[2 * e];
+    parent_ps[2 * j + 1] = ps_prev[2 * e + 1];
+    parent_slot[j] = e;
+    s_b[j] = sb;
+    s_e[j] = se;
+  }
+  // partition tasks over the expand segments, padded to max_ptasks
+  __shared__ long long s_rpt;
+  __shared__ int s_tt;
+  __shared__ int s_pref[2048];
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    long long total = 0;
+    for (int j = 0; j < kb; ++j) total += s_e[j] - s_b[j];
+    long long rpt0 =
+        std::max(min_rows, (total + target_tasks - 1) / target_tasks);
+    int shift = 0;
+    while ((1LL << shift) < rpt0) ++shift;
+    for (;;) {
+      long long need = 0;
+      for (int j = 0; j < kb; ++j) {
+        need += ((long long)(s_e[j] - s_b[j]) + (1LL << shift) - 1) >> shift;
+      }
+      if (need <= max_ptasks) break;
+      ++shift;
+    }
+    int pref = 0;
+    for (int j = 0; j < kb; ++j) {
+      s_pref[j] = pref;
+      pref += (int)(((long long)(s_e[j] - s_b[j]) + (1LL << shift) - 1)
+                    >> shift);
+    }
+    s_rpt = 1LL << shift;
+    s_tt = pref;
+  }
+  __syncthreads();
+  const long long rpt = s_rpt;
+  const int tt = s_tt;
+  for (int t = threadIdx.x; t < max_ptasks; t += blockDim.x) {
+    if (t >= tt || kb == 0) {
+      out_tasks[t] = BlockTask{0, 0, 0, 0};
+      continue;
+    }
+    int lo = 0, hi = kb - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (s_pref[mid] <= t) lo = mid; else hi = mid - 1;
+    }
+    const int j = lo;
+    const long long idx = t - s_pref[j];
+    const int b = s_b[j] + (int)(idx * rpt);
+    const int e = (int)std::min<long long>((long long)b + rpt,
+                                           (long long)s_e[j]);
+    out_tasks[t] = BlockTask{j, b, e, 0};
+  }
+}
+
 struct LeafSeg {
   int nid, begin, end;
   int parity;  // which ping-pong ridx buffer holds these rows
@@ -288,6 +434,8 @@ int gbt_grow_tree(
     BlockTask* hist_tasks_dev,   // [hist_tasks_cap]
     int hist_tasks_cap,
     int32_t* tg_scratch,         // [3 * max_nodes_level + 4]
+    void* wt_ws,                 // whole-tree record/arg arena, or null
+    long long wt_ws_bytes, int wt_max_ptasks,
     const int64_t* root_sums_dev,  // [2] exact quantized (g, h) totals
     const float* maxabs_dev,  // null, or [2] gradient max-abs: scales are
                               // derived on device for the root phase and
@@ -408,9 +556,9 @@ int gbt_grow_tree(
                      g_scale, h_scale, reg_lambda, reg_alpha, max_delta_step,
                      min_child_weight, monotone_dev,
                      (const double*)(d + off_bd), nullptr, nullptr, eval_gain,
-                     eval_bin, eval_dir, eval_lsum, stream);
+                     eval_bin, eval_dir, eval_lsum, nullptr, stream);
         gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k,
-                        n_features, eval_best, stream);
+                        n_features, eval_best, nullptr, stream);
         return 0;
       }
     }
@@ -418,9 +566,9 @@ int gbt_grow_tree(
                  maxabs_eval, g_scale,
                  h_scale, reg_lambda, reg_alpha, max_delta_step,
                  min_child_weight, monotone_dev, nullptr, nullptr, nullptr,
-                 eval_gain, eval_bin, eval_dir, eval_lsum, stream);
+                 eval_gain, eval_bin, eval_dir, eval_lsum, nullptr, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k, n_features,
-                    eval_best, stream);
+                    eval_best, nullptr, stream);
     return 0;
   };
 
@@ -469,7 +617,309 @@ int gbt_grow_tree(
     return 0;
   };
 
-  // ---- root evaluation (root-only sync; root sums + max-abs ride
+  std::vector<Node> level_nodes;
+  std::vector<Node> next_level;
+  std::vector<LeafSeg> leaves;
+  int64_t* cur_pool = hist_pool_a;
+  int64_t* next_pool = hist_pool_b;
+
+  // ================= WHOLE-TREE MODE =================
+  // Single GPU, no monotone bounds: every level's phase chain —
+  // apply-splits, partition, task generation, hist, subtraction,
+  // evaluation — is enqueued with NO host sync (the expansion decision
+  // and all task lists are computed on device); ONE readback at the
+  // end returns the per-level best-split and segment records, and the
+  // host replays them to build the tree arrays (bit-identical: the
+  // replay applies the same comparisons to the same doubles).
+  const bool whole_tree =
+      wt_ws != nullptr && allreduce == nullptr && !has_mono &&
+      maxabs_dev != nullptr && max_nodes_level <= 1024 && max_depth >= 2 &&
+      wt_max_ptasks > 0;
+  if (whole_tree) {
+    const int pool = 2 * max_nodes_level;
+    const int rec_slots = 1 + max_depth * pool;
+    size_t off = 0;
+    auto carve = [&](size_t bytes) {
+      size_t o = off;
+      off = (off + bytes + 255) & ~255ULL;
+      return o;
+    };
+    const size_t o_best = carve((size_t)rec_slots * 6 * 8);
+    const size_t o_seg = carve((size_t)rec_slots * 2 * 4);
+    const size_t o_kn = carve((size_t)(max_depth + 2) * 4);
+    const size_t o_kp = carve((size_t)(max_depth + 2) * 4);
+    const size_t o_descw = carve((size_t)pool * 16);
+    const size_t o_featw = carve((size_t)pool * 4);
+    const size_t o_sbinw = carve((size_t)pool * 4);
+    const size_t o_dlw = carve((size_t)pool);
+    const size_t o_cntw = carve((size_t)pool * 8);
+    const size_t o_pt = carve((size_t)wt_max_ptasks * sizeof(BlockTask));
+    const size_t o_pps = carve((size_t)pool * 16);
+    const size_t o_pslot = carve((size_t)pool * 4);
+    const size_t o_psa = carve((size_t)pool * 16);
+    const size_t o_psb = carve((size_t)pool * 16);
+    if ((long long)off > wt_ws_bytes) return -9997;
+    char* w = (char*)wt_ws;
+    int64_t* best_rec = (int64_t*)(w + o_best);
+    int32_t* seg_rec = (int32_t*)(w + o_seg);
+    int32_t* kn_arr = (int32_t*)(w + o_kn);
+    int32_t* kp_arr = (int32_t*)(w + o_kp);
+    int32_t* d_desc = (int32_t*)(w + o_descw);
+    int32_t* d_feat = (int32_t*)(w + o_featw);
+    int32_t* d_sbin = (int32_t*)(w + o_sbinw);
+    uint8_t* d_dl = (uint8_t*)(w + o_dlw);
+    int32_t* d_cnt = (int32_t*)(w + o_cntw);
+    BlockTask* d_pt = (BlockTask*)(w + o_pt);
+    int64_t* d_pps = (int64_t*)(w + o_pps);
+    int32_t* d_pslot = (int32_t*)(w + o_pslot);
+    int64_t* ps_bufs[2] = {(int64_t*)(w + o_psa), (int64_t*)(w + o_psb)};
+
+    {  // knodes[0]=1, kpairs[0]=0, seg_rec[0]={0, n_rows}
+      const int slot = ctx->ring.next();
+      if (int e = ctx->ring.ensure(slot, 16)) return e;
+      int32_t* hh = (int32_t*)ctx->ring.host[slot];
+      hh[0] = 1;
+      hh[1] = 0;
+      hh[2] = 0;
+      hh[3] = (int)n_rows;
+      HIP_CHECK(hipMemcpyAsync(kn_arr, &hh[0], 4, hipMemcpyHostToDevice,
+                               stream));
+      HIP_CHECK(hipMemcpyAsync(kp_arr, &hh[1], 4, hipMemcpyHostToDevice,
+                               stream));
+      HIP_CHECK(hipMemcpyAsync(seg_rec, &hh[2], 8, hipMemcpyHostToDevice,
+                               stream));
+    }
+    // root evaluation straight into best_rec[0]
+    gbt_evaluate(hist_pool_a, 1, n_bins, n_features, cut_ptrs_dev,
+                 root_sums_dev, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
+                 max_delta_step, min_child_weight, nullptr, nullptr, nullptr,
+                 nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, nullptr,
+                 stream);
+    gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, 1, n_features,
+                    best_rec, nullptr, stream);
+    const int wt_max_htasks = (int)std::min<long long>(
+        std::min<long long>(n_rows / hist_min_rows, hist_tasks) +
+            max_nodes_level + 1,
+        (long long)hist_tasks_cap);
+    const int64_t* ps_prev = root_sums_dev;
+    for (int L = 0; L + 1 < max_depth; ++L) {
+      const int64_t* bl =
+          best_rec + (L == 0 ? 0 : (1 + (size_t)(L - 1) * pool) * 6);
+      const int32_t* sl =
+          seg_rec + (L == 0 ? 0 : (1 + (size_t)(L - 1) * pool) * 2);
+      int64_t* bo = best_rec + (1 + (size_t)L * pool) * 6;
+      int32_t* so = seg_rec + (1 + (size_t)L * pool) * 2;
+      int64_t* ps_next = ps_bufs[L & 1];
+      hipLaunchKernelGGL(ApplyKernel, dim3(1), dim3(256), 0, stream, bl, sl,
+                         kn_arr + L, kp_arr + L, ps_prev, gamma, cut_ptrs_dev,
+                         (long long)1024, (long long)2048, wt_max_ptasks,
+                         kn_arr + L + 1, kp_arr + L + 1, d_feat, d_sbin,
+                         d_dl, d_cnt, d_pt, d_desc, d_pps, d_pslot);
+      gbt_partition(gidx8, gidx16, n_features, cur_ridx, alt_ridx, d_pt,
+                    wt_max_ptasks, d_feat, d_sbin, d_dl, nullptr, nullptr,
+                    n_bins_feat_dev, d_cnt, stream);
+      std::swap(cur_ridx, alt_ridx);
+      hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
+                         d_cnt, d_desc, 0, hist_min_rows, hist_tasks,
+                         wt_max_htasks, tg_scratch, hist_tasks_dev, ps_next,
+                         kp_arr + L + 1, so);
+      HIP_CHECK(hipMemsetAsync(next_pool, 0,
+                               (size_t)pool * hist_row * sizeof(int64_t),
+                               stream));
+      gbt_hist(gidx8, gidx16, n_features, qgpair, cur_ridx, hist_tasks_dev,
+               wt_max_htasks, next_pool, n_bins, feat_group_start_dev,
+               bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
+               use_shared, ps_next, stream);
+      {
+        const long long total = (long long)max_nodes_level * hist_row;
+        int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
+        hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
+                           stream, cur_pool, next_pool, next_pool, d_pslot,
+                           (int)hist_row, 0, ps_next, d_pps, kp_arr + L + 1);
+      }
+      gbt_evaluate(next_pool, pool, n_bins, n_features, cut_ptrs_dev,
+                   ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
+                   max_delta_step, min_child_weight, nullptr, nullptr,
+                   nullptr, nullptr, eval_gain, eval_bin, eval_dir,
+                   eval_lsum, kn_arr + L + 1, stream);
+      gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, pool,
+                      n_features, bo, kn_arr + L + 1, stream);
+      ps_prev = ps_next;
+      std::swap(cur_pool, next_pool);
+    }
+    // ---- the ONE sync: all per-level records + root sums + scales ----
+    size_t r_best = 0;
+    size_t r_seg = ((size_t)rec_slots * 48 + 63) & ~63ULL;
+    size_t r_kp = (r_seg + (size_t)rec_slots * 8 + 63) & ~63ULL;
+    size_t r_rs = (r_kp + (size_t)(max_depth + 2) * 4 + 63) & ~63ULL;
+    size_t r_ma = r_rs + 16;
+    size_t r_total = r_ma + 8;
+    if (int e = ctx->ensure_readback(r_total)) return e;
+    char* rb = (char*)ctx->readback_host;
+    HIP_CHECK(hipMemcpyAsync(rb + r_best, best_rec, (size_t)rec_slots * 48,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipMemcpyAsync(rb + r_seg, seg_rec, (size_t)rec_slots * 8,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipMemcpyAsync(rb + r_kp, kp_arr,
+                             (size_t)(max_depth + 2) * 4,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipMemcpyAsync(rb + r_rs, root_sums_dev, 16,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipMemcpyAsync(rb + r_ma, maxabs_dev, 8,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    const int64_t* h_best = (const int64_t*)(rb + r_best);
+    const int32_t* h_seg = (const int32_t*)(rb + r_seg);
+    const int32_t* h_kp = (const int32_t*)(rb + r_kp);
+    const int64_t* h_rs = (const int64_t*)(rb + r_rs);
+    const float* h_ma = (const float*)(rb + r_ma);
+    g_scale = h_ma[0] > 0.f ? 1073741824.0 / (double)h_ma[0] : 1.0;
+    h_scale = h_ma[1] > 0.f ? 1073741824.0 / (double)h_ma[1] : 1.0;
+    inv_g = 1.0 / g_scale;
+    inv_h = 1.0 / h_scale;
+    if (out_scales != nullptr) {
+      out_scales[0] = g_scale;
+      out_scales[1] = h_scale;
+    }
+    root.gq = h_rs[0];
+    root.hq = h_rs[1];
+    out_base_weight[0] =
+        (float)CalcWeight(root.gq * inv_g, root.hq * inv_h, p);
+    out_sum_hess[0] = (float)(root.hq * inv_h);
+    {
+      double gn;
+      memcpy(&gn, &h_best[0], 8);
+      root.bin = (int)h_best[1];
+      root.dir = (int)h_best[2];
+      root.lgq = h_best[3];
+      root.lhq = h_best[4];
+      root.feature = (int)h_best[5];
+      root.gain = (root.bin >= 0 && std::isfinite(gn)) ? gn : -INFINITY;
+    }
+    // ---- host replay of the records ----
+    level_nodes.push_back(root);
+    for (int depth = 0; depth < max_depth && !level_nodes.empty(); ++depth) {
+      const int parity = depth & 1;
+      std::vector<Node*> expand;
+      for (auto& nd : level_nodes) {
+        if (nd.gain > gamma && std::isfinite(nd.gain)) {
+          expand.push_back(&nd);
+        } else {
+          leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end, parity});
+        }
+      }
+      if (expand.empty()) {
+        level_nodes.clear();
+        break;
+      }
+      next_level.clear();
+      next_level.reserve(2 * expand.size());
+      for (Node* nd : expand) {
+        const int l = n_tree_nodes, r = n_tree_nodes + 1;
+        n_tree_nodes += 2;
+        out_left[nd->nid] = l;
+        out_right[nd->nid] = r;
+        out_left[l] = out_right[l] = -1;
+        out_left[r] = out_right[r] = -1;
+        out_parent[l] = nd->nid;
+        out_parent[r] = nd->nid;
+        out_split_index[nd->nid] = nd->feature;
+        out_split_cond[nd->nid] = cut_values_host[nd->bin];
+        out_default_left[nd->nid] = (uint8_t)nd->dir;
+        out_loss_chg[nd->nid] = (float)nd->gain;
+        const long long rgq = nd->gq - nd->lgq, rhq = nd->hq - nd->lhq;
+        const double wl = CalcWeight(nd->lgq * inv_g, nd->lhq * inv_h, p);
+        const double wr = CalcWeight(rgq * inv_g, rhq * inv_h, p);
+        out_sum_hess[nd->nid] = (float)((nd->lhq + rhq) * inv_h);
+        out_base_weight[l] = (float)wl;
+        out_base_weight[r] = (float)wr;
+        out_sum_hess[l] = (float)(nd->lhq * inv_h);
+        out_sum_hess[r] = (float)(rhq * inv_h);
+        Node ln{}, rn{};
+        ln.nid = l;
+        rn.nid = r;
+        ln.gq = nd->lgq;
+        ln.hq = nd->lhq;
+        rn.gq = rgq;
+        rn.hq = rhq;
+        ln.lo = rn.lo = -INFINITY;
+        ln.hi = rn.hi = INFINITY;
+        ln.gain = rn.gain = -INFINITY;
+        ln.bin = rn.bin = -1;
+        next_level.push_back(ln);
+        next_level.push_back(rn);
+      }
+      const int kbl = (int)expand.size();
+      if (depth + 1 >= max_depth) {
+        // final level: direct leaf-position decide (host-staged args)
+        std::vector<BlockTask> tasks;
+        ChunkTasks(expand, &tasks);
+        const int slot = ctx->ring.next();
+        size_t offf = (tasks.size() * sizeof(BlockTask) + 7) & ~7ULL;
+        size_t offs = (offf + (size_t)kbl * 4 + 7) & ~7ULL;
+        size_t offd = (offs + (size_t)kbl * 4 + 7) & ~7ULL;
+        size_t offk = (offd + (size_t)kbl + 7) & ~7ULL;
+        size_t bytes = offk + (size_t)kbl * 8;
+        if (int e = ctx->ring.ensure(slot, bytes)) return e;
+        char* h = (char*)ctx->ring.host[slot];
+        memcpy(h, tasks.data(), tasks.size() * sizeof(BlockTask));
+        int32_t* hf = (int32_t*)(h + offf);
+        int32_t* hs = (int32_t*)(h + offs);
+        uint8_t* hd = (uint8_t*)(h + offd);
+        int32_t* hk = (int32_t*)(h + offk);
+        for (int j = 0; j < kbl; ++j) {
+          Node* nd = expand[j];
+          hf[j] = nd->feature;
+          hs[j] = nd->bin - cut_ptrs_host[nd->feature];
+          hd[j] = (uint8_t)nd->dir;
+          hk[2 * j] = next_level[2 * j].nid;
+          hk[2 * j + 1] = next_level[2 * j + 1].nid;
+        }
+        HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
+                                 hipMemcpyHostToDevice, stream));
+        char* d = (char*)ctx->ring.dev[slot];
+        gbt_leaf_decide(gidx8, gidx16, n_features, cur_ridx,
+                        (const BlockTask*)d, (int)tasks.size(),
+                        (const int32_t*)(d + offf),
+                        (const int32_t*)(d + offs),
+                        (const uint8_t*)(d + offd),
+                        (const int32_t*)(d + offk), n_bins_feat_dev,
+                        pos_out, stream);
+        level_nodes.clear();
+        break;
+      }
+      if (h_kp[depth + 1] != kbl) return -9991;  // replay divergence
+      const int64_t* bo = h_best + (1 + (size_t)depth * pool) * 6;
+      const int32_t* so = h_seg + (1 + (size_t)depth * pool) * 2;
+      for (int j = 0; j < kbl; ++j) {
+        Node& ln = next_level[2 * j];
+        Node& rn = next_level[2 * j + 1];
+        const int sA = j, sB = kbl + j;
+        const int ab = so[2 * sA], ae = so[2 * sA + 1];
+        const int bb2 = so[2 * sB], be = so[2 * sB + 1];
+        const bool a_left = (ae == bb2);
+        ln.seg_begin = a_left ? ab : bb2;
+        ln.seg_end = a_left ? ae : be;
+        rn.seg_begin = a_left ? bb2 : ab;
+        rn.seg_end = a_left ? be : ae;
+        auto parse1 = [&](Node& nd, int slot2) {
+          double gn;
+          memcpy(&gn, &bo[6 * slot2], 8);
+          nd.bin = (int)bo[6 * slot2 + 1];
+          nd.dir = (int)bo[6 * slot2 + 2];
+          nd.lgq = bo[6 * slot2 + 3];
+          nd.lhq = bo[6 * slot2 + 4];
+          nd.feature = (int)bo[6 * slot2 + 5];
+          nd.gain = (nd.bin >= 0 && std::isfinite(gn)) ? gn : -INFINITY;
+        };
+        parse1(a_left ? ln : rn, sA);
+        parse1(a_left ? rn : ln, sB);
+      }
+      level_nodes.swap(next_level);
+    }
+  } else {
+  // ================= PER-LEVEL MODE =================
+// ---- root evaluation (root-only sync; root sums + max-abs ride
   // along, so neither needs its own host round-trip) ----
   {
     std::vector<Node*> frontier{&root};
@@ -509,11 +959,7 @@ int gbt_grow_tree(
     parse_best(frontier, best);
   }
 
-  std::vector<Node> level_nodes{root};
-  std::vector<Node> next_level;
-  std::vector<LeafSeg> leaves;
-  int64_t* cur_pool = hist_pool_a;
-  int64_t* next_pool = hist_pool_b;
+  level_nodes.push_back(root);
 
   for (int depth = 0; depth < max_depth && !level_nodes.empty(); ++depth) {
     const int parity = (cur_ridx == ridx) ? 0 : 1;
@@ -742,7 +1188,7 @@ int gbt_grow_tree(
     hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
                        cnt_dev, (const int32_t*)(d + off_desc), kb,
                        hist_min_rows, hist_tasks, max_tasks, tg_scratch,
-                       hist_tasks_dev, eval_ps);
+                       hist_tasks_dev, eval_ps, nullptr, nullptr);
     HIP_CHECK(hipMemsetAsync(next_pool, 0,
                              (size_t)kb * hist_row * sizeof(int64_t), stream));
     gbt_hist(gidx8, gidx16, n_features, qgpair, cur_ridx, hist_tasks_dev,
@@ -759,7 +1205,7 @@ int gbt_grow_tree(
       hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
                          stream, cur_pool, next_pool, sub_out,
                          (const int32_t*)(d + off_pslots), (int)hist_row,
-                         kb, eval_ps, parent_ps_dev);
+                         kb, eval_ps, parent_ps_dev, nullptr);
       for (int i = 0; i < (int)subtracted.size(); ++i) {
         subtracted[i]->hist_slot = kb + i;
       }
@@ -805,6 +1251,7 @@ int gbt_grow_tree(
       leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end, parity});
     }
   }
+  }  // end per-level mode
 
   // leaf values
   for (int nid = 0; nid < n_tree_nodes; ++nid) {

@@ -64,6 +64,8 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
     const int64_t* __restrict__ parent_sums,
     const float* __restrict__ maxabs /* null, or [2]: derive scales on
         device so no host sync is needed before the root evaluation */,
+    const int32_t* __restrict__ k_dev /* null, or the live node count:
+        whole-tree mode launches a worst-case grid */,
     double g_scale, double h_scale,
     double reg_lambda, double reg_alpha, double max_delta_step,
     double min_child_weight, const int8_t* __restrict__ monotone,
@@ -73,6 +75,7 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
     int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dir,
     int64_t* __restrict__ out_lsum) {
   const int node = blockIdx.y;
+  if (k_dev != nullptr && node >= *k_dev) return;
   const int lane = threadIdx.x;
   for (int f = blockIdx.x; f < n_features; f += gridDim.x) {
   const size_t out_idx = (size_t)node * n_features + f;
@@ -215,8 +218,10 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
 __global__ __launch_bounds__(64) void SelectBestKernel(
     const double* __restrict__ gain, const int32_t* __restrict__ bins,
     const uint8_t* __restrict__ dirs, const int64_t* __restrict__ lsum,
-    int n_features, int64_t* __restrict__ out_best) {
+    int n_features, int64_t* __restrict__ out_best,
+    const int32_t* __restrict__ k_dev) {
   const int node = blockIdx.x;
+  if (k_dev != nullptr && node >= *k_dev) return;
   const int lane = threadIdx.x;
   const size_t base = (size_t)node * n_features;
   double best_gain = -INFINITY;
@@ -261,9 +266,10 @@ __global__ __launch_bounds__(64) void SelectBestKernel(
 extern "C" void gbt_select_best(const double* gain, const int32_t* bins,
                                 const uint8_t* dirs, const int64_t* lsum,
                                 int n_nodes, int n_features,
-                                int64_t* out_best, hipStream_t stream) {
+                                int64_t* out_best, const int32_t* k_dev,
+                                 hipStream_t stream) {
   hipLaunchKernelGGL(SelectBestKernel, dim3(n_nodes), dim3(64), 0, stream,
-                     gain, bins, dirs, lsum, n_features, out_best);
+                     gain, bins, dirs, lsum, n_features, out_best, k_dev);
 }
 
 extern "C" void gbt_evaluate(
@@ -274,10 +280,11 @@ extern "C" void gbt_evaluate(
     double max_delta_step, double min_child_weight, const int8_t* monotone,
     const double* node_bounds, const uint8_t* feature_mask,
     const uint8_t* cat_feature, double* out_gain, int32_t* out_bin,
-    uint8_t* out_dir, int64_t* out_lsum, hipStream_t stream) {
+    uint8_t* out_dir, int64_t* out_lsum, const int32_t* k_dev,
+    hipStream_t stream) {
   dim3 grid(n_features > 65535 ? 65535 : n_features, n_nodes);
   hipLaunchKernelGGL(EvaluateKernel, grid, dim3(64), 0, stream, hist, n_nodes,
-                     n_bins, n_features, cut_ptrs, parent_sums, maxabs,
+                     n_bins, n_features, cut_ptrs, parent_sums, maxabs, k_dev,
                      g_scale,
                      h_scale, reg_lambda, reg_alpha, max_delta_step,
                      min_child_weight, monotone, node_bounds, feature_mask,

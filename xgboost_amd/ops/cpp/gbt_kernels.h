@@ -62,6 +62,7 @@ void gbt_evaluate(const int64_t* hist /* [n_nodes, n_bins, 2] */,
                   int32_t* out_bin,              // [n_nodes, n_features]
                   uint8_t* out_dir,              // [n_nodes, n_features]
                   int64_t* out_lsum,             // [n_nodes, n_features, 2]
+                  const int32_t* k_dev,          // null, or live node count
                   hipStream_t stream);
 
 void gbt_compress(const float* X, int64_t n_rows, int n_features,
@@ -104,6 +105,7 @@ void gbt_copy_ranges(const int32_t* src, int32_t* dst,
 
 void gbt_select_best(const double* gain, const int32_t* bins,
                      const uint8_t* dirs, const int64_t* lsum, int n_nodes,
-                     int n_features, int64_t* out_best, hipStream_t stream);
+                     int n_features, int64_t* out_best,
+                     const int32_t* k_dev, hipStream_t stream);
 
 }  // extern "C"
