@@ -430,7 +430,9 @@ int gbt_grow_tree(
     int64_t* hist_pool_b, double* eval_gain, int32_t* eval_bin,
     uint8_t* eval_dir, int64_t* eval_lsum, int64_t* eval_best,
     int32_t* pos_out, int max_nodes_level,
-    int32_t* part_counters,      // [>= 2 * max expand nodes]
+    int32_t* part_counters,      // unused (counters live in the ring
+                                 // slot / whole-tree arena); kept in the
+                                 // ABI for wrapper stability
     BlockTask* hist_tasks_dev,   // [hist_tasks_cap]
     int hist_tasks_cap,
     int32_t* tg_scratch,         // [3 * max_nodes_level + 4]
@@ -586,9 +588,6 @@ int gbt_grow_tree(
     }
   };
 
-  // enqueue partition of `nodes` (segments known on host); counters are
-  // left in part_counters for the device-side hist task generation and
-  // the piggybacked readback at the level sync
   // one D2H burst + ONE sync per level: best splits for `eval_nodes`
   // (may be empty on the final level) + partition counters for
   // `n_expand` nodes
