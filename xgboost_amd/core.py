@@ -327,9 +327,17 @@ class Booster:
         quantizer.g_scale = None
         quantizer.h_scale = None
         quantizer.maxabs_dev = m
+        if "rootsum" not in ent:
+            ent["rootsum"] = torch.zeros(2, dtype=torch.int64,
+                                         device=margin.device)
+        else:
+            ent["rootsum"].zero_()
         lib.gbt_quantize(hip_ops.ptr(ent["gh"]), n, 0.0, 0.0,
                          hip_ops.ptr(m), hip_ops.ptr(ent["qg"]),
-                         hip_ops.stream())
+                         hip_ops.ptr(ent["rootsum"]), hip_ops.stream())
+        if collective.is_distributed():
+            collective.allreduce_sum_(ent["rootsum"])
+        quantizer.root_sums_dev = ent["rootsum"]
         seed = (self.seed + iteration if not self.seed_per_iteration
                 else self.seed + iteration * 2654435761)
         tree = RegTree(self.n_features)

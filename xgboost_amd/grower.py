@@ -169,13 +169,16 @@ class TreeGrower:
                 or p.colsample_bynode < 1.0):
             return None
         # root sums stay on device: their readback piggybacks on the
-        # driver's root-eval sync (one fewer host round-trip per tree)
-        from . import collective
-        rs = qgpair.to(torch.int64).sum(dim=0)
-        if collective.is_distributed():
-            collective.allreduce_sum_(rs)
+        # driver's root-eval sync (one fewer host round-trip per tree);
+        # the fused gpair path accumulates them inside QuantizeKernel
+        rs = getattr(self.quantizer, "root_sums_dev", None)
+        if rs is None:
+            from . import collective
+            rs = qgpair.to(torch.int64).sum(dim=0).contiguous()
+            if collective.is_distributed():
+                collective.allreduce_sum_(rs)
         out = ops.grow_tree_native(qgpair, tree, p, self.quantizer,
-                                   self.monotone, rs.contiguous())
+                                   self.monotone, rs)
         return out
 
     def _grow(self, qgpair: torch.Tensor, tree: RegTree

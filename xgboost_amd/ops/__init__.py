@@ -41,7 +41,7 @@ _SIGS = {
     "gbt_mt_evaluate": [_p, _i, _i, _i, _i, _p, _p, _p, _p,
                         _d, _d, _d, _d, _p, _p, _p, _p, _p],
     "gbt_gpair_fused": [_i, _p, _p, _p, _f, _i64, _p, _p, _p],
-    "gbt_quantize": [_p, _i64, _d, _d, _p, _p, _p],
+    "gbt_quantize": [_p, _i64, _d, _d, _p, _p, _p, _p],
     "gbt_margin_add": [_p, _p, _p, _i64, _i, _i, _p],
     "gbt_hist_cpu": [_p, _p, _i, _p, _p, _p, _p, _i, _p, _p, _i],
     "gbt_partition_cpu": [_p, _p, _i, _p, _i64, _i64, _i, _i, _i, _p, _i,

@@ -79,16 +79,49 @@ __global__ __launch_bounds__(256) void QuantizeKernel(
     const float* __restrict__ gh, long long n, double g_scale, double h_scale,
     const float* __restrict__ maxabs /* null, or [2]: derive the scales
         here so the host never has to read max-abs back */,
-    int32_t* __restrict__ out) {
+    int32_t* __restrict__ out,
+    int64_t* __restrict__ out_sums /* null, or [2] zero-initialized:
+        exact totals of the quantized pairs (the tree driver's root
+        sums), accumulated here instead of a separate 16 MB int64
+        materialization + reduction */) {
   if (maxabs != nullptr) {
     g_scale = maxabs[0] > 0.f ? 1073741824.0 / (double)maxabs[0] : 1.0;
     h_scale = maxabs[1] > 0.f ? 1073741824.0 / (double)maxabs[1] : 1.0;
   }
+  long long sg = 0, sh = 0;
   const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   for (long long i = i0; i < n; i += (long long)gridDim.x * blockDim.x) {
     // round-half-to-even (llrint, FE_TONEAREST) matches torch.round
-    out[2 * i] = (int32_t)llrint((double)gh[2 * i] * g_scale);
-    out[2 * i + 1] = (int32_t)llrint((double)gh[2 * i + 1] * h_scale);
+    const int32_t qg = (int32_t)llrint((double)gh[2 * i] * g_scale);
+    const int32_t qh = (int32_t)llrint((double)gh[2 * i + 1] * h_scale);
+    out[2 * i] = qg;
+    out[2 * i + 1] = qh;
+    sg += qg;
+    sh += qh;
+  }
+  if (out_sums != nullptr) {
+    for (int off = 32; off > 0; off >>= 1) {
+      sg += __shfl_down(sg, off, 64);
+      sh += __shfl_down(sh, off, 64);
+    }
+    __shared__ long long wg[256 / 64], wh[256 / 64];
+    const int lane = (int)threadIdx.x & 63, wave = (int)threadIdx.x >> 6;
+    if (lane == 0) {
+      wg[wave] = sg;
+      wh[wave] = sh;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      long long tg = 0, th = 0;
+      for (int w = 0; w < (int)blockDim.x / 64; ++w) {
+        tg += wg[w];
+        th += wh[w];
+      }
+      if (tg) atomicAdd((unsigned long long*)&out_sums[0],
+                        (unsigned long long)tg);
+      if (th) atomicAdd((unsigned long long*)&out_sums[1],
+                        (unsigned long long)th);
+    }
   }
 }
 
@@ -117,10 +150,11 @@ void gbt_gpair_fused(int objective, const float* margin, const float* label,
 
 void gbt_quantize(const float* gh, long long n, double g_scale,
                   double h_scale, const float* maxabs, int32_t* out,
-                  hipStream_t stream) {
-  const int blocks = (int)std::min<long long>((n + 255) / 256, 4096);
+                  int64_t* out_sums, hipStream_t stream) {
+  // same-line atomic lesson as the gpair kernel: few blocks
+  const int blocks = (int)std::min<long long>((n + 255) / 256, 512);
   hipLaunchKernelGGL(QuantizeKernel, dim3(blocks), dim3(256), 0, stream, gh,
-                     n, g_scale, h_scale, maxabs, out);
+                     n, g_scale, h_scale, maxabs, out, out_sums);
 }
 
 }  // extern "C"
