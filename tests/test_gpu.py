@@ -543,23 +543,26 @@ def test_gpu_categorical_matches_cpu():
     logits = (np.isin(Xc[:, 0], [1, 4, 7]) * 2.0 - 1.0) + 0.5 * Xn[:, 0]
     y = (logits + 0.3 * rng.randn(n) > 0).astype(np.float32)
     ft = ["c", "c", "q", "q", "q"]
-    params = {"objective": "binary:logistic", "max_depth": 5,
-              "max_bin": 64, "max_cat_to_onehot": 4, "seed": 0}
-    dg = xgb.DMatrix(X, label=y, feature_types=ft)
-    dc = xgb.DMatrix(X, label=y, feature_types=ft)
-    bg = xgb.train({**params, "device": "cuda"}, dg, 6)
-    bc = xgb.train(params, dc, 6)
-    used_cat = False
-    for tg, tc in zip(bg.trees, bc.trees):
-        assert tg.n_nodes == tc.n_nodes
-        assert np.array_equal(tg.split_index[:tg.n_nodes],
-                              tc.split_index[:tc.n_nodes])
-        assert np.array_equal(tg.left[:tg.n_nodes], tc.left[:tc.n_nodes])
-        used_cat = used_cat or any(
-            tg.split_index[i] < 2 and tg.left[i] >= 0
-            for i in range(tg.n_nodes))
-    assert used_cat, "categorical feature never used - test is vacuous"
-    # prediction parity on the categorical model
-    pg = bg.predict(dg)
-    pc = bc.predict(dc)
-    np.testing.assert_allclose(pg, pc, atol=2e-6)
+    # onehot=4 -> sorted-partition subsets; onehot=32 -> in-kernel one-hot
+    for onehot in (4, 32):
+        params = {"objective": "binary:logistic", "max_depth": 5,
+                  "max_bin": 64, "max_cat_to_onehot": onehot, "seed": 0}
+        dg = xgb.DMatrix(X, label=y, feature_types=ft)
+        dc = xgb.DMatrix(X, label=y, feature_types=ft)
+        bg = xgb.train({**params, "device": "cuda"}, dg, 6)
+        bc = xgb.train(params, dc, 6)
+        used_cat = False
+        for tg, tc in zip(bg.trees, bc.trees):
+            assert tg.n_nodes == tc.n_nodes, f"onehot={onehot}"
+            assert np.array_equal(tg.split_index[:tg.n_nodes],
+                                  tc.split_index[:tc.n_nodes])
+            assert np.array_equal(tg.left[:tg.n_nodes],
+                                  tc.left[:tc.n_nodes])
+            used_cat = used_cat or any(
+                tg.split_index[i] < 2 and tg.left[i] >= 0
+                for i in range(tg.n_nodes))
+        assert used_cat, "categorical feature never used - test is vacuous"
+        # prediction parity on the categorical model
+        pg = bg.predict(dg)
+        pc = bc.predict(dc)
+        np.testing.assert_allclose(pg, pc, atol=2e-6)

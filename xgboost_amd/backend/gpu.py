@@ -231,6 +231,21 @@ class GpuOps(SegmentedOpsMixin):
                     mask_np[i] = 1
                 else:
                     mask_np[i, np.asarray(fs, np.int64)] = 1
+        # wide categorical features (cardinality > max_cat_to_onehot) use
+        # sorted-partition splits, evaluated HOST-side with the SAME
+        # routine as the CPU oracle (splits._sorted_cat_split) so GPU and
+        # CPU trees stay identical; mask them out of the kernel's
+        # one-hot/numeric scan
+        wide_feats = None
+        if cat_mask is not None:
+            widths = np.diff(self.qm.cuts.ptrs)
+            wm = np.asarray(cat_mask, bool) & (
+                widths > param.max_cat_to_onehot)
+            if wm.any():
+                wide_feats = np.nonzero(wm)[0]
+                if mask_np is None:
+                    mask_np = np.ones((k, f), dtype=np.uint8)
+                mask_np[:, wide_feats] = 0
         parents, mono_t, bounds_t, mask_t = self.stager.upload(
             [parents_np, mono_np, bounds_np, mask_np])
         self.lib.gbt_evaluate(
@@ -275,7 +290,41 @@ class GpuOps(SegmentedOpsMixin):
                     e.cat_bits = np.array(
                         [e.split_bin - int(cuts.ptrs[fidx])], dtype=np.int32)
             out.append(e)
+        if wide_feats is not None:
+            self._host_sorted_cat(out, hist, parents_np, quantizer, param,
+                                  wide_feats, feature_sets)
         return out
+
+    def _host_sorted_cat(self, out, hist, parents_np, quantizer, param,
+                         wide_feats, feature_sets) -> None:
+        """Sorted-partition categorical splits on the pulled histogram
+        columns — identical math to the CPU oracle (splits.py)."""
+        from ..splits import (_sorted_cat_split, calc_gain_given_weight,
+                              calc_weight)
+        cuts = self.qm.cuts
+        k = len(out)
+        cols = np.concatenate([
+            np.arange(cuts.ptrs[fi], cuts.ptrs[fi + 1]) for fi in wide_feats])
+        cols_t = torch.as_tensor(cols, dtype=torch.long, device=self.device)
+        hist_cat = hist.index_select(1, cols_t).cpu().numpy()  # [k,|cols|,2]
+        hist_np = np.zeros((k, self.n_bins, 2), dtype=np.int64)
+        hist_np[:, cols, :] = hist_cat
+        inv_g = 1.0 / quantizer.g_scale
+        inv_h = 1.0 / quantizer.h_scale
+        for i, e in enumerate(out):
+            pgq, phq = int(parents_np[i, 0]), int(parents_np[i, 1])
+            pw = calc_weight(pgq * inv_g, phq * inv_h, param)
+            pgain = calc_gain_given_weight(pgq * inv_g, phq * inv_h, pw,
+                                           param)
+            allowed = None
+            if feature_sets is not None and feature_sets[i] is not None:
+                allowed = set(int(x) for x in feature_sets[i])
+            for fi in wide_feats:
+                if allowed is not None and int(fi) not in allowed:
+                    continue
+                _sorted_cat_split(e, hist_np[i], pgq, phq, int(fi),
+                                  cuts.ptrs, param, inv_g, inv_h,
+                                  float(pgain))
 
     def partition(self, ridx: torch.Tensor,
                   segments: Sequence[Tuple[int, int]],
