@@ -566,3 +566,19 @@ def test_gpu_categorical_matches_cpu():
         pg = bg.predict(dg)
         pc = bc.predict(dc)
         np.testing.assert_allclose(pg, pc, atol=2e-6)
+
+
+def test_gpu_approx_matches_cpu():
+    """tree_method=approx re-sketches with hessian weights each
+    iteration; cuts come from the same host sketch, so CPU and GPU
+    trees must still agree exactly."""
+    X, y = _data(8000, 6, seed=13)
+    params = {"objective": "binary:logistic", "tree_method": "approx",
+              "max_depth": 4, "max_bin": 64, "seed": 0}
+    bc = xgb.train(params, xgb.DMatrix(X, label=y), 5)
+    bg = xgb.train({**params, "device": "cuda"},
+                   xgb.DMatrix(X, label=y), 5)
+    for tc, tg in zip(bc.trees, bg.trees):
+        assert tc.n_nodes == tg.n_nodes
+        assert np.array_equal(tc.split_index[:tc.n_nodes],
+                              tg.split_index[:tg.n_nodes])
