@@ -1,4 +1,8 @@
 """BASELINE.json config 1: agaricus, tree_method=hist, CPU, 10 rounds
+
+Data: the classic UCI mushroom dataset in libsvm form (the exact
+files BASELINE.json's config 1 names, demo/data in the reference;
+public-domain UCI data, included here as test fixtures).
 binary:logistic — the reference's canonical plumbing test
 (demo/guide-python; tests/python/test_basic.py uses the same files)."""
 import os
