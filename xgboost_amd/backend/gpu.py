@@ -257,7 +257,10 @@ class GpuOps(SegmentedOpsMixin):
             self.hip.ptr(bounds_t), self.hip.ptr(mask_t),
             self.hip.ptr(self.cat_feature),
             self.hip.ptr(gain), self.hip.ptr(bins), self.hip.ptr(dirs),
-            self.hip.ptr(lsum), None, self.hip.stream())
+            self.hip.ptr(lsum), None,
+            8 if f > 4096 else 0,  # scalar eval for narrow features on
+                                   # one-hot-scale data (see evaluate.hip)
+            self.hip.stream())
         # device-side per-node argmax + packing -> ONE D2H sync
         out_best = torch.empty((k, 6), dtype=torch.int64, device=dev)
         self.lib.gbt_select_best(

@@ -31,7 +31,7 @@ _SIGS = {
     "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p, _p],
     "gbt_partition": [_p, _p, _i, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p, _p, _p],
     "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _p, _d, _d, _d, _d, _d, _d,
-                     _p, _p, _p, _p, _p, _p, _p, _p, _p, _p],
+                     _p, _p, _p, _p, _p, _p, _p, _p, _p, _i, _p],
     "gbt_compress": [_p, _i64, _i, _p, _p, _p, _f, _i, _p, _p, _p],
     "gbt_predict": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
                     _p, _p, _i, _i, _p, _p, _p],

@@ -558,7 +558,7 @@ int gbt_grow_tree(
                      g_scale, h_scale, reg_lambda, reg_alpha, max_delta_step,
                      min_child_weight, monotone_dev,
                      (const double*)(d + off_bd), nullptr, nullptr, eval_gain,
-                     eval_bin, eval_dir, eval_lsum, nullptr, stream);
+                     eval_bin, eval_dir, eval_lsum, nullptr, 0, stream);
         gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k,
                         n_features, eval_best, nullptr, stream);
         return 0;
@@ -568,7 +568,7 @@ int gbt_grow_tree(
                  maxabs_eval, g_scale,
                  h_scale, reg_lambda, reg_alpha, max_delta_step,
                  min_child_weight, monotone_dev, nullptr, nullptr, nullptr,
-                 eval_gain, eval_bin, eval_dir, eval_lsum, nullptr, stream);
+                 eval_gain, eval_bin, eval_dir, eval_lsum, nullptr, 0, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k, n_features,
                     eval_best, nullptr, stream);
     return 0;
@@ -693,7 +693,7 @@ int gbt_grow_tree(
                  root_sums_dev, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
                  max_delta_step, min_child_weight, nullptr, nullptr, nullptr,
                  nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, nullptr,
-                 stream);
+                 0, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, 1, n_features,
                     best_rec, nullptr, stream);
     const int wt_max_htasks = (int)std::min<long long>(
@@ -740,7 +740,7 @@ int gbt_grow_tree(
                    ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
                    max_delta_step, min_child_weight, nullptr, nullptr,
                    nullptr, nullptr, eval_gain, eval_bin, eval_dir,
-                   eval_lsum, kn_arr + L + 1, stream);
+                   eval_lsum, kn_arr + L + 1, 0, stream);
       gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, pool,
                       n_features, bo, kn_arr + L + 1, stream);
       ps_prev = ps_next;

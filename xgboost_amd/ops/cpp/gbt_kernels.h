@@ -63,6 +63,7 @@ void gbt_evaluate(const int64_t* hist /* [n_nodes, n_bins, 2] */,
                   uint8_t* out_dir,              // [n_nodes, n_features]
                   int64_t* out_lsum,             // [n_nodes, n_features, 2]
                   const int32_t* k_dev,          // null, or live node count
+                  int narrow_max,  // >0: scalar kernel for narrow features
                   hipStream_t stream);
 
 void gbt_compress(const float* X, int64_t n_rows, int n_features,
