@@ -280,6 +280,17 @@ __global__ __launch_bounds__(256) void EvaluateNarrowKernel(
       fg += nh[2 * b];
       fh += nh[2 * b + 1];
     }
+    if (fh == 0 && fg == 0) {
+      // feature absent from this node's rows: every candidate split is
+      // empty-vs-all and fails the exact hessian-count validity — the
+      // common case for one-hot features in deep nodes
+      out_gain[out_idx] = -INFINITY;
+      out_bin[out_idx] = -1;
+      out_dir[out_idx] = 0;
+      out_lsum[2 * out_idx] = 0;
+      out_lsum[2 * out_idx + 1] = 0;
+      continue;
+    }
     const long long miss_g = pg - fg;
     const long long miss_h = ph - fh;
     const double pw = CalcWeight(pg * inv_g, ph * inv_h, p);
