@@ -631,12 +631,18 @@ class _ForestArrays:
 
 
 def shap_gpu(booster, dmat, lo: int, hi: int, phi: np.ndarray) -> np.ndarray:
-    """GPU pred_contribs (path-dependent TreeSHAP kernel, shap.hip).
-    phi arrives [n, n_groups, f+1] float64 pre-filled with the base
-    margin in the bias column."""
+    """GPU pred_contribs.  Numeric forests use the path-table kernel
+    (shap_paths.hip: host-precomputed path decomposition, register
+    Extend/Unwind, ~2 orders of magnitude faster than the old scratch
+    DFS); categorical forests fall back to the DFS kernel."""
     from .. import ops as hip_ops
     from ..shap import _expected_value
     lib = hip_ops.load()
+    has_cat = any(
+        t.split_type[:t.n_nodes].any() for t in booster.trees[lo:hi]
+        if hasattr(t, "split_type"))
+    if not has_cat and hasattr(lib, "gbt_shap_paths"):
+        return _shap_gpu_paths(booster, dmat, lo, hi, phi)
     if not hasattr(lib, "gbt_shap"):
         raise ImportError("gbt_shap kernel not built")
     max_depth = max((t.max_depth() for t in booster.trees[lo:hi]), default=0)
@@ -664,6 +670,43 @@ def shap_gpu(booster, dmat, lo: int, hi: int, phi: np.ndarray) -> np.ndarray:
         hip_ops.ptr(fa.cat_bits), hip_ops.ptr(fa.sum_hess),
         hip_ops.ptr(fa.tree_group), fa.n_trees, n_groups, n_cols,
         hip_ops.ptr(expected), hip_ops.ptr(out), hip_ops.stream())
+    res = out.cpu().numpy()
+    if n_groups == 1:
+        return res[:, 0, :]
+    return res
+
+
+def _shap_gpu_paths(booster, dmat, lo: int, hi: int,
+                    phi: np.ndarray) -> np.ndarray:
+    from .. import ops as hip_ops
+    from ..shap_paths import build_path_table
+    lib = hip_ops.load()
+    n, n_groups, n_cols = phi.shape
+    device = booster.device
+    pp, pg, ef, elo, ehi, emiss, ez, pv, bias = build_path_table(
+        booster.trees[lo:hi], booster.tree_info[lo:hi])
+    max_elems = int(np.diff(pp).max()) if len(pg) else 0
+    if max_elems > 16:
+        raise ImportError("GPU SHAP path length > 16; CPU fallback")
+    dd = dmat.device_data() if hasattr(dmat, "device_data") else None
+    X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
+    out = torch.from_numpy(np.ascontiguousarray(phi, np.float64)).to(device)
+    for grp in range(n_groups):
+        out[:, grp, n_cols - 1] += float(bias[grp]) if grp < len(bias) else 0.0
+    t = {}
+    for name, arr in (("pp", pp), ("pg", pg), ("ef", ef), ("elo", elo),
+                      ("ehi", ehi), ("emiss", emiss), ("ez", ez),
+                      ("pv", pv)):
+        t[name] = torch.from_numpy(np.ascontiguousarray(arr)).to(device)
+    missing = dmat.missing
+    missing_is_nan = 1 if np.isnan(missing) else 0
+    fn = lib.gbt_shap_paths if max_elems <= 8 else lib.gbt_shap_paths16
+    fn(hip_ops.ptr(X), n, dmat.num_col(),
+       float(0.0 if missing_is_nan else missing), missing_is_nan,
+       hip_ops.ptr(t["pp"]), hip_ops.ptr(t["pg"]), hip_ops.ptr(t["ef"]),
+       hip_ops.ptr(t["elo"]), hip_ops.ptr(t["ehi"]),
+       hip_ops.ptr(t["emiss"]), hip_ops.ptr(t["ez"]), hip_ops.ptr(t["pv"]),
+       len(pg), n_groups, n_cols, hip_ops.ptr(out), hip_ops.stream())
     res = out.cpu().numpy()
     if n_groups == 1:
         return res[:, 0, :]
