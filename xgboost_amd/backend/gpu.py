@@ -693,10 +693,11 @@ def _shap_gpu_paths(booster, dmat, lo: int, hi: int,
     out = torch.from_numpy(np.ascontiguousarray(phi, np.float64)).to(device)
     for grp in range(n_groups):
         out[:, grp, n_cols - 1] += float(bias[grp]) if grp < len(bias) else 0.0
+    rz = np.where(ez > 0, 1.0 / np.maximum(ez, 1e-300), 0.0)
     t = {}
     for name, arr in (("pp", pp), ("pg", pg), ("ef", ef), ("elo", elo),
                       ("ehi", ehi), ("emiss", emiss), ("ez", ez),
-                      ("pv", pv)):
+                      ("rz", rz), ("pv", pv)):
         t[name] = torch.from_numpy(np.ascontiguousarray(arr)).to(device)
     missing = dmat.missing
     missing_is_nan = 1 if np.isnan(missing) else 0
@@ -705,7 +706,8 @@ def _shap_gpu_paths(booster, dmat, lo: int, hi: int,
        float(0.0 if missing_is_nan else missing), missing_is_nan,
        hip_ops.ptr(t["pp"]), hip_ops.ptr(t["pg"]), hip_ops.ptr(t["ef"]),
        hip_ops.ptr(t["elo"]), hip_ops.ptr(t["ehi"]),
-       hip_ops.ptr(t["emiss"]), hip_ops.ptr(t["ez"]), hip_ops.ptr(t["pv"]),
+       hip_ops.ptr(t["emiss"]), hip_ops.ptr(t["ez"]), hip_ops.ptr(t["rz"]),
+       hip_ops.ptr(t["pv"]),
        len(pg), n_groups, n_cols, hip_ops.ptr(out), hip_ops.stream())
     res = out.cpu().numpy()
     if n_groups == 1:

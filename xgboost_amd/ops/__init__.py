@@ -76,9 +76,9 @@ _SIGS = {
     "gbt_shap": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p, _p,
                  _p, _p, _i, _i, _i, _p, _p, _p],
     "gbt_shap_paths": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p,
-                       _p, _p, _i64, _i, _i, _p, _p],
+                       _p, _p, _p, _i64, _i, _i, _p, _p],
     "gbt_shap_paths16": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p,
-                         _p, _p, _i64, _i, _i, _p, _p],
+                         _p, _p, _p, _i64, _i, _i, _p, _p],
 }
 
 

@@ -15,6 +15,15 @@
 
 namespace {
 
+// reciprocals of the small integer divisors in Extend/Unwind: fp64
+// division has no hardware path (~10 instructions each) and the inner
+// loops execute millions of them per row; multiplying by these differs
+// from the oracle's division by <= 1 ulp per op (tolerance-tested)
+__constant__ double kRcp[20] = {
+    0.0,      1.0,      1.0 / 2,  1.0 / 3,  1.0 / 4,  1.0 / 5,  1.0 / 6,
+    1.0 / 7,  1.0 / 8,  1.0 / 9,  1.0 / 10, 1.0 / 11, 1.0 / 12, 1.0 / 13,
+    1.0 / 14, 1.0 / 15, 1.0 / 16, 1.0 / 17, 1.0 / 18, 1.0 / 19};
+
 // kD: compile-time element cap — all loops fully unroll so the pw
 // array and element state stay in registers (predicated dead
 // iterations cost ~2x for shallow trees; a dynamic bound would spill
@@ -26,7 +35,8 @@ __global__ __launch_bounds__(256) void ShapPathsKernel(
     const int64_t* __restrict__ path_ptr, const int32_t* __restrict__ pgrp,
     const int32_t* __restrict__ ef, const float* __restrict__ elo,
     const float* __restrict__ ehi, const uint8_t* __restrict__ emiss,
-    const double* __restrict__ ez, const double* __restrict__ pv,
+    const double* __restrict__ ez, const double* __restrict__ erz,
+    const double* __restrict__ pv,
     long long n_paths, int n_groups, int n_cols /* n_features + 1 */,
     double* __restrict__ phi /* [n_rows, n_groups, n_cols] */) {
   const long long row0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -41,6 +51,7 @@ __global__ __launch_bounds__(256) void ShapPathsKernel(
       const double v = pv[p];
       if (v == 0.0) continue;
       double zs[kD];
+      double rzs[kD];
       double ones[kD];
 #pragma unroll
       for (int j = 0; j < kD; ++j) {
@@ -53,6 +64,7 @@ __global__ __launch_bounds__(256) void ShapPathsKernel(
                              : (x >= elo[s + j] && x < ehi[s + j]);
         ones[j] = ok ? 1.0 : 0.0;
         zs[j] = ez[s + j];
+        rzs[j] = erz[s + j];
       }
       // extend: pw[0..M] with the implicit (1,1) base already applied
       double pw[kD + 1];
@@ -61,14 +73,15 @@ __global__ __launch_bounds__(256) void ShapPathsKernel(
       for (int j = 0; j < kD; ++j) {
         if (j >= M) break;
         const int mm = j + 1;  // path length before this extend
-        pw[mm] = ones[j] * pw[mm - 1] * mm / (mm + 1);
+        const double rcp = kRcp[mm + 1];
+        pw[mm] = ones[j] * pw[mm - 1] * mm * rcp;
 #pragma unroll
         for (int i = kD - 1; i >= 1; --i) {
           if (i > mm - 1) continue;
-          pw[i] = ones[j] * pw[i - 1] * i / (mm + 1)
-                  + zs[j] * pw[i] * (mm - i) / (mm + 1);
+          pw[i] = ones[j] * pw[i - 1] * i * rcp
+                  + zs[j] * pw[i] * (mm - i) * rcp;
         }
-        pw[0] = zs[j] * pw[0] * mm / (mm + 1);
+        pw[0] = zs[j] * pw[0] * mm * rcp;
       }
       const int d = M;
       const int grp = pgrp[p];
@@ -79,20 +92,22 @@ __global__ __launch_bounds__(256) void ShapPathsKernel(
         const double o = ones[i];
         const double z = zs[i];
         double total = 0.0;
-        if (o != 0.0) {
+        const double rd1 = kRcp[d + 1];
+        if (o != 0.0) {  // o == 1 in this formulation
           double nxt = pw[d];
 #pragma unroll
           for (int j = kD - 1; j >= 0; --j) {
             if (j > d - 1) continue;
-            const double tmp = nxt * (d + 1) / ((j + 1) * o);
+            const double tmp = nxt * (d + 1) * kRcp[j + 1];
             total += tmp;
-            nxt = pw[j] - tmp * z * (d - j) / (d + 1);
+            nxt = pw[j] - tmp * z * (d - j) * rd1;
           }
         } else {
+          const double rz = rzs[i];
 #pragma unroll
           for (int j = kD - 1; j >= 0; --j) {
             if (j > d - 1) continue;
-            total += pw[j] * (d + 1) / (z * (d - j));
+            total += pw[j] * (d + 1) * rz * kRcp[d - j];
           }
         }
         phig[ef[s + i]] += v * (o - z) * total;
@@ -107,14 +122,15 @@ extern "C" void gbt_shap_paths(
     const float* X, long long n_rows, int n_features, float missing_value,
     int missing_is_nan, const int64_t* path_ptr, const int32_t* pgrp,
     const int32_t* ef, const float* elo, const float* ehi,
-    const uint8_t* emiss, const double* ez, const double* pv,
+    const uint8_t* emiss, const double* ez, const double* erz,
+    const double* pv,
     long long n_paths, int n_groups, int n_cols, double* phi,
     hipStream_t stream) {
   const int blocks =
       (int)((n_rows + 255) / 256 < 16384 ? (n_rows + 255) / 256 : 16384);
   hipLaunchKernelGGL(ShapPathsKernel<8>, dim3(blocks), dim3(256), 0, stream,
                      X, n_rows, n_features, missing_value, missing_is_nan,
-                     path_ptr, pgrp, ef, elo, ehi, emiss, ez, pv, n_paths,
+                     path_ptr, pgrp, ef, elo, ehi, emiss, ez, erz, pv, n_paths,
                      n_groups, n_cols, phi);
 }
 
@@ -122,13 +138,14 @@ extern "C" void gbt_shap_paths16(
     const float* X, long long n_rows, int n_features, float missing_value,
     int missing_is_nan, const int64_t* path_ptr, const int32_t* pgrp,
     const int32_t* ef, const float* elo, const float* ehi,
-    const uint8_t* emiss, const double* ez, const double* pv,
+    const uint8_t* emiss, const double* ez, const double* erz,
+    const double* pv,
     long long n_paths, int n_groups, int n_cols, double* phi,
     hipStream_t stream) {
   const int blocks =
       (int)((n_rows + 255) / 256 < 16384 ? (n_rows + 255) / 256 : 16384);
   hipLaunchKernelGGL(ShapPathsKernel<16>, dim3(blocks), dim3(256), 0, stream,
                      X, n_rows, n_features, missing_value, missing_is_nan,
-                     path_ptr, pgrp, ef, elo, ehi, emiss, ez, pv, n_paths,
+                     path_ptr, pgrp, ef, elo, ehi, emiss, ez, erz, pv, n_paths,
                      n_groups, n_cols, phi);
 }
