@@ -12,7 +12,7 @@ def run(n_rows, n_paths, M, iters=3):
     dev = "cuda"
     F = 28
     rng = np.random.RandomState(0)
-    X = torch.from_numpy(rng.randn(n_rows, F).astype(np.float32)).to(dev)
+    X = torch.from_numpy(rng.randn(F, n_rows).astype(np.float32)).to(dev)
     pp = np.arange(n_paths + 1, dtype=np.int64) * M
     pg = np.zeros(n_paths, dtype=np.int32)
     ef = rng.randint(0, F, size=n_paths * M).astype(np.int32)
@@ -25,7 +25,7 @@ def run(n_rows, n_paths, M, iters=3):
     t = {k: torch.from_numpy(v).to(dev) for k, v in
          dict(pp=pp, pg=pg, ef=ef, elo=elo, ehi=ehi, em=em, ez=ez,
               rz=rz, pv=pv).items()}
-    phi = torch.zeros(n_rows, 1, F + 1, dtype=torch.float64, device=dev)
+    phi = torch.zeros(1, F + 1, n_rows, dtype=torch.float64, device=dev)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):

@@ -690,9 +690,13 @@ def _shap_gpu_paths(booster, dmat, lo: int, hi: int,
         raise ImportError("GPU SHAP path length > 16; CPU fallback")
     dd = dmat.device_data() if hasattr(dmat, "device_data") else None
     X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
-    out = torch.from_numpy(np.ascontiguousarray(phi, np.float64)).to(device)
+    X = X.t().contiguous()  # [F, n]: coalesced per-feature gathers
+    # phi transposed to [groups*cols, n] so the accumulation is coalesced
+    out = torch.from_numpy(
+        np.ascontiguousarray(phi, np.float64)).to(device)
     for grp in range(n_groups):
         out[:, grp, n_cols - 1] += float(bias[grp]) if grp < len(bias) else 0.0
+    out = out.permute(1, 2, 0).contiguous()  # [groups, cols, n]
     rz = np.where(ez > 0, 1.0 / np.maximum(ez, 1e-300), 0.0)
     t = {}
     for name, arr in (("pp", pp), ("pg", pg), ("ef", ef), ("elo", elo),
@@ -709,7 +713,7 @@ def _shap_gpu_paths(booster, dmat, lo: int, hi: int,
        hip_ops.ptr(t["emiss"]), hip_ops.ptr(t["ez"]), hip_ops.ptr(t["rz"]),
        hip_ops.ptr(t["pv"]),
        len(pg), n_groups, n_cols, hip_ops.ptr(out), hip_ops.stream())
-    res = out.cpu().numpy()
+    res = out.permute(2, 0, 1).contiguous().cpu().numpy()
     if n_groups == 1:
         return res[:, 0, :]
     return res
