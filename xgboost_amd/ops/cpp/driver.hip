@@ -709,6 +709,8 @@ int gbt_grow_tree(
       int64_t* bo = best_rec + (1 + (size_t)L * pool) * 6;
       int32_t* so = seg_rec + (1 + (size_t)L * pool) * 2;
       int64_t* ps_next = ps_bufs[L & 1];
+      // level-L capacity: at most 2^L nodes expand, 2^(L+1) children
+      const int cap_kids = (int)std::min<long long>(2LL << L, (long long)pool);
       hipLaunchKernelGGL(ApplyKernel, dim3(1), dim3(256), 0, stream, bl, sl,
                          kn_arr + L, kp_arr + L, ps_prev, gamma, cut_ptrs_dev,
                          (long long)1024, (long long)2048, wt_max_ptasks,
@@ -723,25 +725,25 @@ int gbt_grow_tree(
                          wt_max_htasks, tg_scratch, hist_tasks_dev, ps_next,
                          kp_arr + L + 1, so);
       HIP_CHECK(hipMemsetAsync(next_pool, 0,
-                               (size_t)pool * hist_row * sizeof(int64_t),
+                               (size_t)cap_kids * hist_row * sizeof(int64_t),
                                stream));
       gbt_hist(gidx8, gidx16, n_features, qgpair, cur_ridx, hist_tasks_dev,
                wt_max_htasks, next_pool, n_bins, feat_group_start_dev,
                bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
                use_shared, ps_next, stream);
       {
-        const long long total = (long long)max_nodes_level * hist_row;
+        const long long total = (long long)(cap_kids / 2 + 1) * hist_row;
         int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
         hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
                            stream, cur_pool, next_pool, next_pool, d_pslot,
                            (int)hist_row, 0, ps_next, d_pps, kp_arr + L + 1);
       }
-      gbt_evaluate(next_pool, pool, n_bins, n_features, cut_ptrs_dev,
+      gbt_evaluate(next_pool, cap_kids, n_bins, n_features, cut_ptrs_dev,
                    ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
                    max_delta_step, min_child_weight, nullptr, nullptr,
                    nullptr, nullptr, eval_gain, eval_bin, eval_dir,
                    eval_lsum, kn_arr + L + 1, 0, stream);
-      gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, pool,
+      gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, cap_kids,
                       n_features, bo, kn_arr + L + 1, stream);
       ps_prev = ps_next;
       std::swap(cur_pool, next_pool);
