@@ -582,3 +582,21 @@ def test_gpu_approx_matches_cpu():
         assert tc.n_nodes == tg.n_nodes
         assert np.array_equal(tc.split_index[:tc.n_nodes],
                               tg.split_index[:tg.n_nodes])
+
+
+def test_gpu_shap_multiclass_matches_cpu():
+    """Path-table SHAP with n_groups > 1: per-group bias and the
+    transposed phi indexing must match the CPU oracle."""
+    rng = np.random.RandomState(17)
+    X = rng.randn(3000, 6).astype(np.float32)
+    y = np.abs(X[:, :3]).argmax(axis=1).astype(np.float32)
+    params = {"objective": "multi:softprob", "num_class": 3,
+              "max_depth": 4, "max_bin": 64, "seed": 0}
+    dc = xgb.DMatrix(X, label=y)
+    bc = xgb.train(params, dc, 4)
+    ref = bc.predict(dc, pred_contribs=True)
+    dg = xgb.DMatrix(X, label=y)
+    bg = xgb.train({**params, "device": "cuda"}, dg, 4)
+    got = bg.predict(dg, pred_contribs=True)
+    assert got.shape == ref.shape == (3000, 3, 7)
+    np.testing.assert_allclose(got, ref, atol=5e-4, rtol=1e-3)
