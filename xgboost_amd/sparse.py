@@ -51,19 +51,50 @@ class SparseQuantizedMatrix:
 
 
 def sketch_csr(X_csr, max_bin: int) -> HistogramCuts:
-    """Cuts from the nonzero values of each column (absent = missing)."""
+    """Cuts from the nonzero values of each column (absent = missing).
+    Distributed: per-rank column summaries are allgathered and merged so
+    every rank quantizes with IDENTICAL cuts (same guarantee as the
+    dense sketch; rank-local cuts would break the histogram allreduce).
+    """
+    from . import collective
     csc = X_csr.tocsc()
     n_features = csc.shape[1]
+    distributed = collective.get_world_size() > 1
     # fast path: single-valued data (one-hot / binary indicator matrices
     # at Criteo scale have 1e6 columns — the per-column loop would
     # dominate).  Every column's cuts = [v + |v| + 1e-5] (one bin).
-    if csc.nnz and csc.data.size and float(csc.data.min()) == float(csc.data.max()):
+    # Distributed: the single-value test must be GLOBAL.
+    lo = float(csc.data.min()) if csc.nnz else np.inf
+    hi = float(csc.data.max()) if csc.nnz else -np.inf
+    if distributed:
+        lo = -collective.allreduce_max_scalars([-lo])[0]
+        (hi,) = collective.allreduce_max_scalars([hi])
+    if csc.nnz and csc.data.size and lo == hi:
         v = float(csc.data.flat[0])
         sentinel = np.float32(v + (abs(v) + 1e-5))
         values = np.full(n_features, sentinel, dtype=np.float32)
         ptrs = np.arange(n_features + 1, dtype=np.int64)
         min_vals = np.full(n_features, np.float32(v), dtype=np.float32)
         return HistogramCuts(values=values, ptrs=ptrs, min_vals=min_vals)
+    if distributed:
+        # per-column quantile summaries, merged like the dense path
+        from .sketch import cuts_from_summaries
+        K = max(64, 8 * max_bin)
+        qs = (np.arange(K) + 0.5) / K
+        local = []
+        for f in range(n_features):
+            vals = csc.data[csc.indptr[f]:csc.indptr[f + 1]].astype(
+                np.float32)
+            vals = vals[~np.isnan(vals)]
+            if vals.size == 0:
+                local.append(("q", np.zeros(0, np.float32), 0.0))
+            else:
+                pts = np.quantile(vals, qs).astype(np.float32)
+                local.append(("q", np.concatenate(
+                    [[np.float32(vals.min())], pts,
+                     [np.float32(vals.max())]]), float(vals.size)))
+        gathered = collective.allgather_obj(local)
+        return cuts_from_summaries(gathered, max_bin, n_features, None)
     all_cuts: List[np.ndarray] = []
     min_vals = np.zeros(n_features, dtype=np.float32)
     from .quantile import _cuts_for_column
