@@ -234,3 +234,57 @@ def test_two_process_sparse_training():
     caught rank-local cuts breaking the collective)."""
     blob = _run_workers(2, _SPARSE_SCRIPT)
     assert len(blob["raw"]) > 100
+
+
+_EXTMEM_SCRIPT = """
+import os, pickle, sys
+import numpy as np
+import torch.distributed as dist
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+rank = int(os.environ["RANK"])
+collective.init("gloo")
+
+
+class It(xgb.DataIter):
+    def __init__(self):
+        super().__init__()
+        self.i = 0
+
+    def reset(self):
+        self.i = 0
+
+    def next(self, input_data):
+        if self.i >= 3:
+            return False
+        rng = np.random.RandomState(100 * rank + self.i)
+        Xb = rng.randn(1500, 6).astype(np.float32)
+        yb = (Xb[:, 0] > 0).astype(np.float32)
+        input_data(data=Xb, label=yb)
+        self.i += 1
+        return True
+
+
+d = xgb.ExtMemQuantileDMatrix(It(), max_bin=64)
+params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+          "seed": 11, "debug_synchronize": True}
+bst = xgb.train(params, d, 4, verbose_eval=False)
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "extmem model differs across workers"
+out = os.environ["XGB_AMD_OUT"]
+if rank == 0:
+    with open(out, "wb") as fh:
+        pickle.dump({"raw": raw}, fh)
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_two_process_extmem_training():
+    """Distributed external-memory training: per-rank DataIter batches,
+    allgathered sketch summaries, identical models on every worker."""
+    blob = _run_workers(2, _EXTMEM_SCRIPT)
+    assert len(blob["raw"]) > 100
