@@ -288,3 +288,46 @@ def test_two_process_extmem_training():
     allgathered sketch summaries, identical models on every worker."""
     blob = _run_workers(2, _EXTMEM_SCRIPT)
     assert len(blob["raw"]) > 100
+
+
+_MT_SCRIPT = """
+import os, pickle, sys
+import numpy as np
+import torch.distributed as dist
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+collective.init("gloo")
+
+rng = np.random.RandomState(0)
+n, f = 1600, 6
+X = rng.randn(n, f).astype(np.float32)
+Y = np.stack([X[:, 0] + 0.1 * rng.randn(n),
+              X[:, 1] - X[:, 2] + 0.1 * rng.randn(n)], axis=1).astype(
+    np.float32)
+shard = slice(rank * n // world, (rank + 1) * n // world)
+d = xgb.DMatrix(X[shard], label=Y[shard])
+params = {"objective": "reg:squarederror", "max_depth": 4,
+          "multi_strategy": "multi_output_tree", "seed": 2,
+          "debug_synchronize": True}
+bst = xgb.train(params, d, 3, verbose_eval=False)
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "multi-target model differs across workers"
+out = os.environ["XGB_AMD_OUT"]
+if rank == 0:
+    with open(out, "wb") as fh:
+        pickle.dump({"raw": raw}, fh)
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_two_process_multi_target_training():
+    """Distributed vector-leaf trees: per-target hist allreduce +
+    summed-gain evaluation give identical models on every worker."""
+    blob = _run_workers(2, _MT_SCRIPT)
+    assert len(blob["raw"]) > 100

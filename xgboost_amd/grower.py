@@ -285,7 +285,11 @@ class TreeGrower:
                                or param.grow_policy == "lossguide")
                 if not expand_more and param.max_depth > 0:
                     continue
-                if ops.node_size(l) <= ops.node_size(r):
+                # build the smaller sibling by GLOBAL hessian sum:
+                # local row counts are rank-dependent, and ranks must
+                # build/allreduce the SAME node set (exact int64
+                # subtraction makes the choice performance-only)
+                if node_sums[l][1] <= node_sums[r][1]:
                     build_nodes.append((l, b.nid, r))
                 else:
                     build_nodes.append((r, b.nid, l))
@@ -511,7 +515,9 @@ class MultiTargetGrower:
                 if param.max_depth > 0 and b.depth + 1 >= param.max_depth \
                         and param.grow_policy == "depthwise":
                     continue
-                if ops.node_size(l) <= ops.node_size(r):
+                # global hessian-sum choice: rank-identical (see the
+                # single-target driver note above)
+                if node_sums[l][:, 1].sum() <= node_sums[r][:, 1].sum():
                     build_nodes.append((l, b.nid, r))
                 else:
                     build_nodes.append((r, b.nid, l))
