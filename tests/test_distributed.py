@@ -331,3 +331,44 @@ def test_two_process_multi_target_training():
     summed-gain evaluation give identical models on every worker."""
     blob = _run_workers(2, _MT_SCRIPT)
     assert len(blob["raw"]) > 100
+
+
+_APPROX_SCRIPT = """
+import os, pickle, sys
+import numpy as np
+import torch.distributed as dist
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+collective.init("gloo")
+
+rng = np.random.RandomState(0)
+n, f = 2000, 6
+X = rng.randn(n, f).astype(np.float32)
+y = (X[:, 0] + X[:, 1] ** 2 > 0.5).astype(np.float32)
+shard = slice(rank * n // world, (rank + 1) * n // world)
+d = xgb.DMatrix(X[shard], label=y[shard])
+params = {"objective": "binary:logistic", "tree_method": "approx",
+          "max_depth": 4, "seed": 5, "debug_synchronize": True}
+bst = xgb.train(params, d, 3, verbose_eval=False)
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "approx model differs across workers"
+out = os.environ["XGB_AMD_OUT"]
+if rank == 0:
+    with open(out, "wb") as fh:
+        pickle.dump({"raw": raw}, fh)
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_two_process_approx_training():
+    """tree_method=approx re-sketches with hessian weights every
+    iteration; the weighted summaries must merge across ranks so cuts
+    (and models) stay identical."""
+    blob = _run_workers(2, _APPROX_SCRIPT)
+    assert len(blob["raw"]) > 100

@@ -136,10 +136,23 @@ class Booster:
             # (reference: grow_histmaker/grow_gpu_approx, updater_approx.cc:46)
             from .data import quantize_dense
             from .quantile import make_cuts
-            cuts = make_cuts(dmat.raw_data(), self.tparam.max_bin,
-                             weights=np.abs(hess) + 1e-16,
-                             feature_types=dmat.info.feature_types,
-                             missing=dmat.missing)
+            if collective.is_distributed():
+                # ranks must agree on the re-sketched cuts: merge the
+                # weighted per-rank summaries like the in-core sketch
+                from .sketch import cuts_from_summaries, summarize_batch
+                local = summarize_batch(
+                    dmat.raw_data(), dmat.missing,
+                    dmat.info.feature_types, self.tparam.max_bin,
+                    weights=np.abs(hess) + 1e-16)
+                gathered = collective.allgather_obj(local)
+                cuts = cuts_from_summaries(
+                    gathered, self.tparam.max_bin, dmat.num_col(),
+                    dmat.info.feature_types)
+            else:
+                cuts = make_cuts(dmat.raw_data(), self.tparam.max_bin,
+                                 weights=np.abs(hess) + 1e-16,
+                                 feature_types=dmat.info.feature_types,
+                                 missing=dmat.missing)
             qm = quantize_dense(dmat.raw_data(), cuts, dmat.missing)
             if self.device.type == "cuda":
                 from .backend.gpu import GpuOps
