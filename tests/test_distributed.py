@@ -372,3 +372,47 @@ def test_two_process_approx_training():
     (and models) stay identical."""
     blob = _run_workers(2, _APPROX_SCRIPT)
     assert len(blob["raw"]) > 100
+
+
+_KITCHEN_SCRIPT = """
+import os, pickle, sys
+import numpy as np
+import torch.distributed as dist
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+collective.init("gloo")
+
+rng = np.random.RandomState(0)
+n, f = 2400, 8
+X = rng.randn(n, f).astype(np.float32)
+w = rng.randn(f)
+y = (X @ w + 0.3 * rng.randn(n)).astype(np.float32)
+shard = slice(rank * n // world, (rank + 1) * n // world)
+d = xgb.DMatrix(X[shard], label=y[shard])
+params = {"objective": "reg:squarederror", "grow_policy": "lossguide",
+          "max_leaves": 24, "max_depth": 0, "colsample_bynode": 0.7,
+          "subsample": 0.9, "monotone_constraints": "(1,0,0,0,0,0,0,0)",
+          "seed": 9, "debug_synchronize": True}
+bst = xgb.train(params, d, 4, verbose_eval=False)
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "kitchen-sink model differs across workers"
+out = os.environ["XGB_AMD_OUT"]
+if rank == 0:
+    with open(out, "wb") as fh:
+        pickle.dump({"raw": raw}, fh)
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_two_process_lossguide_colsample_monotone():
+    """The feature-rich python-driver paths (lossguide heap, column
+    sampling, subsample, monotone bounds) must also be bit-identical
+    across workers."""
+    blob = _run_workers(2, _KITCHEN_SCRIPT)
+    assert len(blob["raw"]) > 100
