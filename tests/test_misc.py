@@ -297,3 +297,27 @@ def test_booster_load_config_roundtrip():
     tp = c2["learner"]["gradient_booster"]["tree_train_param"]
     assert float(tp["eta"]) == 0.21
     assert int(tp["max_depth"]) == 4
+
+
+def test_bench_contract_cpu():
+    """bench.py must run standalone (no flags -> N=1 quick run) and
+    print one JSON line with the driver-contract fields."""
+    import json
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"),
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300, cwd=repo)
+    assert out.returncode == 0, out.stdout + out.stderr
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    blob = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in blob, key
+    assert blob["metric"] == "boosting_rounds_per_sec"
+    assert blob["scaling"] == "weak"
+    assert blob["data"] == "synthetic"
