@@ -495,8 +495,9 @@ int gbt_grow_tree(
   // amortize ~2*group_bins atomics (env GBT_HIST_TASKS to tune)
   static long long hist_tasks = [] {
     const char* e = getenv("GBT_HIST_TASKS");
-    long long v = e ? atoll(e) : 512;
-    return v >= 64 && v <= 16384 ? v : 512;
+    long long v = e ? atoll(e) : 256;  // swept: 256 beats 512 by ~4%
+                                       // under whole-tree mode
+    return v >= 64 && v <= 16384 ? v : 256;
   }();
   const long long hist_min_rows = 2048;
 
