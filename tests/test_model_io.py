@@ -133,3 +133,24 @@ def test_feature_names_mismatch_raises():
     bad = xgb.DMatrix(X[:, :4], label=y)
     with pytest.raises(ValueError):
         bst.predict(bad)
+
+
+def test_attributes_and_names_roundtrip(tmp_path):
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(120, 3).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    d = xgb.DMatrix(X, label=y, feature_names=["alpha", "beta", "gamma"],
+                    feature_types=["q", "q", "q"])
+    bst = xgb.train({"objective": "reg:squarederror", "max_depth": 3}, d, 3)
+    bst.set_attr(best_iteration="2", my_tag="hello")
+    path = str(tmp_path / "m.json")
+    bst.save_model(path)
+    bst2 = xgb.Booster(model_file=path)
+    assert bst2.attr("my_tag") == "hello"
+    assert bst2.attr("best_iteration") == "2"
+    assert bst2.feature_names == ["alpha", "beta", "gamma"]
+    assert bst2.feature_types == ["q", "q", "q"]
+    score = bst2.get_score(importance_type="gain")
+    assert all(k in ("alpha", "beta", "gamma") for k in score)
