@@ -321,3 +321,30 @@ def test_bench_contract_cpu():
     assert blob["metric"] == "boosting_rounds_per_sec"
     assert blob["scaling"] == "weak"
     assert blob["data"] == "synthetic"
+
+
+def test_base_score_estimation_boost_from_average():
+    """reference test_intercept.py: with no base_score given, the
+    intercept is fitted from the labels (one Newton step), and a
+    0-round model predicts it."""
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(500, 3).astype(np.float32)
+    y = (rng.rand(500) < 0.25).astype(np.float32)  # 25% positives
+    d = xgb.DMatrix(X, label=y)
+    # estimation happens at the first boost (reference: lazy configure
+    # + FitStump); the fitted intercept lands in base_score
+    bst = xgb.train({"objective": "binary:logistic", "eta": 0.0}, d, 1)
+    # ONE Newton step at margin 0 (reference FitStump): sigmoid(4(m-.5))
+    expect = 1.0 / (1.0 + np.exp(-4.0 * (y.mean() - 0.5)))
+    np.testing.assert_allclose(bst.base_score, expect, rtol=1e-3)
+    # explicit base_score wins
+    bst2 = xgb.train({"objective": "binary:logistic", "base_score": 0.5,
+                      "eta": 0.0}, xgb.DMatrix(X, label=y), 1)
+    np.testing.assert_allclose(bst2.base_score, 0.5, atol=1e-6)
+    # regression: mean label
+    yr = (X[:, 0] * 2 + 5).astype(np.float32)
+    bst3 = xgb.train({"objective": "reg:squarederror", "eta": 0.0},
+                     xgb.DMatrix(X, label=yr), 1)
+    np.testing.assert_allclose(bst3.base_score, yr.mean(), rtol=1e-3)
