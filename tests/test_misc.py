@@ -348,3 +348,27 @@ def test_base_score_estimation_boost_from_average():
     bst3 = xgb.train({"objective": "reg:squarederror", "eta": 0.0},
                      xgb.DMatrix(X, label=yr), 1)
     np.testing.assert_allclose(bst3.base_score, yr.mean(), rtol=1e-3)
+
+
+def test_early_stopping_last_dataset_last_metric():
+    """reference semantics: without explicit names, early stopping
+    watches the LAST metric on the LAST eval set."""
+    import numpy as np
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(600, 4).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    Xv = rng.randn(300, 4).astype(np.float32)
+    yv = (rng.rand(300) > 0.5).astype(np.float32)  # pure noise valid
+    res = {}
+    bst = xgb.train({"objective": "binary:logistic",
+                     "eval_metric": ["logloss", "auc"], "max_depth": 3},
+                    xgb.DMatrix(X, label=y), 60,
+                    evals=[(xgb.DMatrix(X, label=y), "train"),
+                           (xgb.DMatrix(Xv, label=yv), "valid")],
+                    early_stopping_rounds=5, evals_result=res,
+                    verbose_eval=False)
+    # stopped early on noise validation AUC (maximize inferred)
+    assert bst.best_iteration is not None
+    assert len(res["valid"]["auc"]) < 60
+    assert "logloss" in res["valid"] and "auc" in res["valid"]
