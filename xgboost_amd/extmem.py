@@ -8,8 +8,9 @@ updater_gpu_hist.cu:371 (partition+hist per page pass).
 MI355X design: quantized pages live in pinned host memory; the GPU
 keeps a device-resident page cache sized to a HBM budget (288 GB/GPU
 makes most datasets fully cacheable — the reference's
-cache_host_ratio=0 case); pages beyond the budget are streamed with
-hipMemcpyAsync on torch's stream each sweep.
+cache_host_ratio=0 case); pages beyond the budget are streamed on a
+dedicated copy stream with one-page lookahead, so the next page's H2D
+overlaps the current page's kernels.
 """
 from __future__ import annotations
 
