@@ -67,6 +67,25 @@ void HistOneSeg(const BinT* gidx, int n_features, const int32_t* qgpair,
   }
 }
 
+template <typename BinT>
+void HistSegSerial(const BinT* gidx, int n_features, const int32_t* qgpair,
+                   const int64_t* ridx, int64_t begin, int64_t end,
+                   const int32_t* cut_ptrs, int64_t* hist) {
+  for (int64_t i = begin; i < end; ++i) {
+    const int64_t row = ridx[i];
+    const int64_t g = qgpair[2 * row];
+    const int64_t h = qgpair[2 * row + 1];
+    const BinT* rb = gidx + (size_t)row * n_features;
+    for (int f = 0; f < n_features; ++f) {
+      const int local_bin = (int)rb[f];
+      if (local_bin >= cut_ptrs[f + 1] - cut_ptrs[f]) continue;
+      const int b = cut_ptrs[f] + local_bin;
+      hist[2 * b] += g;
+      hist[2 * b + 1] += h;
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -75,6 +94,28 @@ void gbt_hist_cpu(const uint8_t* gidx8, const uint16_t* gidx16,
                   int n_features, const int32_t* qgpair, const int64_t* ridx,
                   const int64_t* seg_begin, const int64_t* seg_end, int n_segs,
                   const int32_t* cut_ptrs, int64_t* out_hist, int n_bins) {
+#ifdef _OPENMP
+  const int n_threads = omp_get_max_threads();
+#else
+  const int n_threads = 1;
+#endif
+  if (n_segs >= n_threads) {
+    // deep levels: one thread per segment writing straight into its own
+    // output slot — the thread-local-copy scheme would spend more time
+    // zeroing and reducing 16B/bin scratch than visiting rows
+#pragma omp parallel for schedule(dynamic)
+    for (int s = 0; s < n_segs; ++s) {
+      int64_t* hist = out_hist + (size_t)s * n_bins * 2;
+      if (gidx8 != nullptr) {
+        HistSegSerial<uint8_t>(gidx8, n_features, qgpair, ridx,
+                               seg_begin[s], seg_end[s], cut_ptrs, hist);
+      } else {
+        HistSegSerial<uint16_t>(gidx16, n_features, qgpair, ridx,
+                                seg_begin[s], seg_end[s], cut_ptrs, hist);
+      }
+    }
+    return;
+  }
   for (int s = 0; s < n_segs; ++s) {
     int64_t* hist = out_hist + (size_t)s * n_bins * 2;
     if (gidx8 != nullptr) {
