@@ -77,9 +77,18 @@ class SegmentedOpsMixin:
         s, e = self.segments[nid]
         return e - s
 
-    def build_hist_nodes(self, qgpair: torch.Tensor, nids) -> torch.Tensor:
+    def build_hist_nodes(self, qgpair: torch.Tensor, nids,
+                         out: "torch.Tensor" = None) -> torch.Tensor:
         return self.build_hist(qgpair, self.ridx,
-                               [self.segments[n] for n in nids])
+                               [self.segments[n] for n in nids], out=out)
+
+    def alloc_hist(self, k: int) -> torch.Tensor:
+        """Uninitialized [k, n_bins, 2] on this backend's device — the
+        level driver builds/subtracts into slices of one buffer instead
+        of concatenating per-node histograms."""
+        dev = getattr(self, "device", None)
+        return torch.empty((k, self.n_bins, 2), dtype=torch.int64,
+                           device=dev if dev is not None else "cpu")
 
     def partition_nodes(self, parents, splits, children) -> None:
         """parents: [nid], splits: [SplitEntry], children: [(l, r)].
@@ -144,13 +153,25 @@ class CpuOps(SegmentedOpsMixin):
         return int(t[0]), int(t[1])
 
     def build_hist(self, qgpair: torch.Tensor, ridx: torch.Tensor,
-                   segments: Sequence[Tuple[int, int]]) -> torch.Tensor:
+                   segments: Sequence[Tuple[int, int]],
+                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
         """-> int64 [len(segments), n_bins, 2]."""
         k = len(segments)
-        out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64)
+        if out is None:
+            out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64)
+        else:
+            assert out.is_contiguous()
+            out.zero_()
         if self.lib is not None:
             import ctypes
-            q_np = np.ascontiguousarray(qgpair.cpu().numpy(), np.int32)
+            # the quantized gradient tensor is identical across every
+            # call within a tree: cache the int32 view by data pointer
+            key = (qgpair.data_ptr(), qgpair._version)
+            if getattr(self, "_qnp_key", None) != key:
+                self._qnp_key = key
+                self._qnp = np.ascontiguousarray(
+                    qgpair.cpu().numpy(), np.int32)
+            q_np = self._qnp
             r_np = np.ascontiguousarray(ridx.numpy(), np.int64)
             sb = np.ascontiguousarray([s for s, _ in segments], np.int64)
             se = np.ascontiguousarray([e for _, e in segments], np.int64)

@@ -185,10 +185,18 @@ class GpuOps(SegmentedOpsMixin):
         return int(host[0]), int(host[1])
 
     def build_hist(self, qgpair: torch.Tensor, ridx: torch.Tensor,
-                   segments: Sequence[Tuple[int, int]]) -> torch.Tensor:
+                   segments: Sequence[Tuple[int, int]],
+                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
         k = len(segments)
+        if out is not None:
+            assert out.is_contiguous()
+            out.zero_()
+            return self._build_hist_into(qgpair, ridx, segments, out)
         out = torch.zeros((k, self.n_bins, 2), dtype=torch.int64,
                           device=self.device)
+        return self._build_hist_into(qgpair, ridx, segments, out)
+
+    def _build_hist_into(self, qgpair, ridx, segments, out):
         tasks_np = _chunk_tasks(segments, target_tasks=512)
         (tasks,) = self.stager.upload([tasks_np])
         p8, p16 = self._gidx_ptrs()

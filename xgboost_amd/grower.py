@@ -293,24 +293,35 @@ class TreeGrower:
                     build_nodes.append((l, b.nid, r))
                 else:
                     build_nodes.append((r, b.nid, l))
-            # 4. build + allreduce + subtract
+            # 4. build + allreduce + subtract — all into ONE level
+            # buffer ([built... | subtracted...]) so evaluation reads a
+            # contiguous stack with no concatenation copies
             if build_nodes:
-                bh = ops.build_hist_nodes(qgpair,
-                                          [n for n, _, _ in build_nodes])
+                kb = len(build_nodes)
+                stack = (ops.alloc_hist(2 * kb)
+                         if hasattr(ops, "alloc_hist") else None)
+                bnids = [n for n, _, _ in build_nodes]
+                if stack is None:  # extmem / csr ops: no out= support
+                    bh = ops.build_hist_nodes(qgpair, bnids)
+                else:
+                    bh = ops.build_hist_nodes(qgpair, bnids,
+                                              out=stack[:kb])
                 ops.allreduce_hist(bh)
-                # batched sibling subtraction: one kernel for the level
                 parent_stack = torch.stack(
                     [hists[p] for _, p, _ in build_nodes])
-                sib_stack = parent_stack - bh
+                if stack is None:
+                    sib_stack = parent_stack - bh
+                    hist_stack = torch.cat([bh, sib_stack], dim=0)
+                else:
+                    torch.sub(parent_stack, bh, out=stack[kb:])
+                    sib_stack = stack[kb:]
+                    hist_stack = stack
                 for i, (n, parent, sib) in enumerate(build_nodes):
                     hists[n] = bh[i]
                     hists[sib] = sib_stack[i]
                     del hists[parent]
-                # 5. evaluate children: [built..., subtracted...] order so
-                # the histogram input is one cat, not k small stacks
                 eval_nids = ([n for n, _, _ in build_nodes]
                              + [s for _, _, s in build_nodes])
-                hist_stack = torch.cat([bh, sib_stack], dim=0)
                 depth = children[0][0].depth + 1
                 entries = self._evaluate(eval_nids, node_sums, hists,
                                          node_bounds, depth,
