@@ -340,8 +340,13 @@ class CpuOps(SegmentedOpsMixin):
     def leaf_partition(self, ridx: torch.Tensor,
                        leaf_segments: Sequence[Tuple[int, int, int]],
                        n_rows: int) -> torch.Tensor:
-        """-> int32 [n_rows] leaf node id per row (for prediction cache)."""
-        pos = torch.zeros(n_rows, dtype=torch.int32)
+        """-> int32 [n_rows] leaf node id per row (for prediction cache).
+
+        One contiguous fill per leaf + a single scatter: per-leaf fancy
+        indexing cost ~0.25 ms x n_leaves in torch."""
+        vals = torch.zeros(n_rows, dtype=torch.int32)
         for nid, s, e in leaf_segments:
-            pos[ridx[s:e]] = nid
+            vals[s:e] = nid  # positional (contiguous) fill
+        pos = torch.zeros(n_rows, dtype=torch.int32)
+        pos[ridx.long()] = vals
         return pos
