@@ -1,10 +1,11 @@
 """GPU backend ops — drives the HIP/CDNA4 kernels in ops/cpp.
 
 Same interface as backend/cpu.CpuOps; the grower is backend-agnostic.
-All launches go on torch's current HIP stream; the only host syncs per
-level are the partition-counter readback and the split-candidate
-readback (mirroring the reference's per-level pinned D2H sync,
-updater_gpu_hist.cu:613).
+All launches go on torch's current HIP stream.  These python-side ops
+back the feature-rich driver paths (categorical, sampling, lossguide,
+external memory); the hot path is the native C++ driver
+(ops/cpp/driver.hip) wrapped by grow_tree_native below, which in
+whole-tree mode runs with ONE host sync per tree.
 """
 from __future__ import annotations
 
