@@ -416,3 +416,52 @@ def test_two_process_lossguide_colsample_monotone():
     across workers."""
     blob = _run_workers(2, _KITCHEN_SCRIPT)
     assert len(blob["raw"]) > 100
+
+
+INTERCEPT_WORKER = r"""
+import os, pickle, sys
+import numpy as np
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+collective.init("gloo")
+rank = collective.get_rank()
+world = collective.get_world_size()
+
+rng = np.random.RandomState(5)
+n, f = 1200, 4
+X = rng.randn(n, f).astype(np.float32)
+# labels with a strongly rank-dependent mean: a local fit-stump would
+# produce a different intercept on each shard
+y = (np.arange(n) / n * 10.0 + rng.rand(n)).astype(np.float32)
+
+shard = slice(rank * n // world, (rank + 1) * n // world)
+dtrain = xgb.DMatrix(X[shard], label=y[shard])
+params = {"objective": "reg:squarederror", "max_depth": 2, "eta": 0.1}
+bst = xgb.train(params, dtrain, 1, verbose_eval=False)
+base = float(bst.base_score)
+
+if rank == 0:
+    with open(os.environ["XGB_AMD_OUT"], "wb") as fh:
+        pickle.dump({"base": base}, fh)
+import torch.distributed as dist
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_two_process_global_intercept():
+    """Distributed fit-stump must equal the single-process full-data
+    intercept (reference GlobalSum in src/tree/fit_stump.cu:46-49)."""
+    import xgboost_amd as xgb
+    res = _run_workers(2, INTERCEPT_WORKER)
+    rng = np.random.RandomState(5)
+    n, f = 1200, 4
+    X = rng.randn(n, f).astype(np.float32)
+    y = (np.arange(n) / n * 10.0 + rng.rand(n)).astype(np.float32)
+    solo = xgb.train({"objective": "reg:squarederror", "max_depth": 2,
+                      "eta": 0.1}, xgb.DMatrix(X, label=y), 1,
+                     verbose_eval=False)
+    assert abs(res["base"] - float(solo.base_score)) < 1e-6, (
+        res["base"], float(solo.base_score))

@@ -190,13 +190,28 @@ def _torch_op(op: "Op"):
 
 def allreduce(data, op: "Op"):
     """In-place allreduce of a numpy array; returns it (reference
-    collective.py:275)."""
+    collective.py:275).
+
+    Routed through the backend's communication device like
+    allreduce_sum_ (rccl wants GPU tensors); bitwise ops are not
+    provided by either nccl/rccl or gloo and raise loudly instead of
+    hanging."""
     import numpy as _np
     data = _np.asarray(data)
     if not is_distributed():
         return data
+    if op in (Op.BITWISE_AND, Op.BITWISE_OR, Op.BITWISE_XOR):
+        raise NotImplementedError(
+            f"bitwise allreduce ({op.name}) is not supported by the "
+            f"{dist.get_backend()} backend")
     t = torch.from_numpy(data)
-    dist.all_reduce(t, op=_torch_op(op))
+    dev = _comm_device()
+    if dev.type != "cpu":
+        tmp = t.to(dev)
+        dist.all_reduce(tmp, op=_torch_op(op))
+        t.copy_(tmp.to(t.device))
+    else:
+        dist.all_reduce(t, op=_torch_op(op))
     return data
 
 

@@ -77,6 +77,8 @@ class Booster:
 
         self._cache: Dict[int, Tuple[torch.Tensor, int]] = {}  # id(dmat) -> (margin, version)
         self._ops_cache: Dict[int, Any] = {}
+        # id(dmat) -> weakref with an eviction callback (see _pin)
+        self._cache_refs: Dict[int, Any] = {}
         if model_file is not None:
             self.load_model(model_file)
         if cache:
@@ -93,6 +95,32 @@ class Booster:
 
     def num_features(self) -> int:
         return self.n_features or 0
+
+    def _pin(self, dmat) -> int:
+        """Cache key for a DMatrix that cannot go stale: caches are keyed
+        by id(dmat), and CPython reuses ids after collection — so a new
+        DMatrix could silently alias a dead one's cached margin/quantized
+        matrix (reference avoids this with DMatrixCache weak_ptr eviction,
+        include/xgboost/cache.h:26).  A weakref eviction callback clears
+        every cache family for the id the moment the DMatrix dies."""
+        key = id(dmat)
+        refs = self.__dict__.setdefault("_cache_refs", {})
+        if key not in refs:
+            import weakref
+            self_ref = weakref.ref(self)
+
+            def _evict(_, key=key, self_ref=self_ref):
+                b = self_ref()
+                if b is None:
+                    return
+                b._cache.pop(key, None)
+                b._ops_cache.pop(key, None)
+                b.__dict__.get("_fused_cache", {}).pop(key, None)
+                b.__dict__.get("_exact_cache", {}).pop(key, None)
+                b.__dict__.get("_cache_refs", {}).pop(key, None)
+
+            refs[key] = weakref.ref(dmat, _evict)
+        return key
 
     def _maybe_set_meta(self, dmat: DMatrix) -> None:
         if self.n_features is None:
@@ -158,7 +186,7 @@ class Booster:
                 from .backend.gpu import GpuOps
                 return GpuOps(qm.to(self.device))
             return CpuOps(qm)
-        key = id(dmat)
+        key = self._pin(dmat)
         ops = self._ops_cache.get(key)
         if ops is None:
             from .extmem import ExtMemOps, ExtMemQuantileDMatrix
@@ -307,7 +335,7 @@ class Booster:
             return False
         ops = self._ops_for(dtrain)
         n = dtrain.num_row()
-        key = id(dtrain)
+        key = self._pin(dtrain)
         fc = self.__dict__.setdefault("_fused_cache", {})
         ent = fc.get(key)
         if ent is None:
@@ -535,7 +563,7 @@ class Booster:
 
     # -- prediction ----------------------------------------------------
     def _cached_margin(self, dmat: DMatrix) -> torch.Tensor:
-        key = id(dmat)
+        key = self._pin(dmat)
         n = dmat.num_row()
         entry = self._cache.get(key)
         if entry is not None and entry[1] == len(self.trees):

@@ -348,7 +348,10 @@ class TreeGrower:
     # ------------------------------------------------------------------
     def _expandable(self, split: SplitEntry, depth: int, n_leaves: int) -> bool:
         param = self.param
-        if not split.is_valid or split.gain <= param.gamma:
+        # reference prunes loss_chg < min_split_loss (driver.h:37): a
+        # split with gain exactly == gamma IS expanded; the separate
+        # gain > 0 validity lives in SplitEntry.is_valid
+        if not split.is_valid or split.gain < param.gamma:
             return False
         if param.max_depth > 0 and depth >= param.max_depth:
             return False
@@ -471,7 +474,7 @@ class MultiTargetGrower:
 
         def push(nid, depth, split):
             nonlocal seq
-            if not split.is_valid or split.gain <= param.gamma:
+            if not split.is_valid or split.gain < param.gamma:
                 return
             if param.max_depth > 0 and depth >= param.max_depth:
                 return

@@ -101,7 +101,12 @@ class Objective:
     def init_estimation(self, info) -> float:
         """One Newton step at margin 0 (reference FitStump + PredTransform).
         Multi-target labels produce the mean of per-target estimates
-        (scalar base_score, like the reference's ParamArray mean)."""
+        (scalar base_score, like the reference's ParamArray mean).
+
+        Distributed: the G/H sums are allreduced so the intercept is the
+        GLOBAL -sum(G)/sum(H), matching the reference's GlobalSum inside
+        FitStump (src/tree/fit_stump.cu:46-49) — not a local estimate."""
+        from . import collective
         n = info.num_row
         n_out = 1
         if info.labels is not None and info.labels.ndim == 2:
@@ -109,7 +114,14 @@ class Objective:
         preds = torch.zeros((n, n_out), dtype=torch.float32)
         g, h = self.get_gradient(preds, info, 0)
         gs = g.double().sum(dim=0)
-        hs = h.double().sum(dim=0).clamp(min=1e-16)
+        hs = h.double().sum(dim=0)
+        if collective.is_distributed():
+            k = gs.numel()
+            glob = collective.allreduce_sum_scalars(
+                gs.cpu().tolist() + hs.cpu().tolist())
+            gs = torch.tensor(glob[:k], dtype=torch.float64)
+            hs = torch.tensor(glob[k:], dtype=torch.float64)
+        hs = hs.clamp(min=1e-16)
         margin = float((-gs / hs).mean())
         out = self.pred_transform(torch.tensor([margin])).item()
         return out

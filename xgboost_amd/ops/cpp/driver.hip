@@ -319,7 +319,10 @@ __global__ __launch_bounds__(256) void ApplyKernel(
     const int slot = s_order[i];
     const double gv = __longlong_as_double(best[6 * slot]);
     const long long bin = best[6 * slot + 1];
-    s_flag[i] = (bin >= 0 && isfinite(gv) && gv > gamma_) ? 1 : 0;
+    // expand iff gain > 0 (validity) AND gain >= gamma (reference
+    // driver.h:37 prunes loss_chg < min_split_loss, boundary included)
+    s_flag[i] =
+        (bin >= 0 && isfinite(gv) && gv > 0.0 && gv >= gamma_) ? 1 : 0;
   }
   __syncthreads();
   if (threadIdx.x == 0) {
@@ -804,7 +807,7 @@ int gbt_grow_tree(
       const int parity = depth & 1;
       std::vector<Node*> expand;
       for (auto& nd : level_nodes) {
-        if (nd.gain > gamma && std::isfinite(nd.gain)) {
+        if (nd.gain > 0.0 && nd.gain >= gamma && std::isfinite(nd.gain)) {
           expand.push_back(&nd);
         } else {
           leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end, parity});
@@ -967,7 +970,7 @@ int gbt_grow_tree(
     const int parity = (cur_ridx == ridx) ? 0 : 1;
     std::vector<Node*> expand;
     for (auto& nd : level_nodes) {
-      if (nd.gain > gamma && std::isfinite(nd.gain)) {
+      if (nd.gain > 0.0 && nd.gain >= gamma && std::isfinite(nd.gain)) {
         expand.push_back(&nd);
       } else {
         leaves.push_back({nd.nid, nd.seg_begin, nd.seg_end, parity});
