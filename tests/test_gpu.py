@@ -507,10 +507,16 @@ def test_native_driver_distributed_branch_rccl_world1():
     from xgboost_amd import collective
 
     X, y = _data(20000, 8, seed=31)
-    pd = {"objective": "binary:logistic", "max_depth": 6, "max_bin": 64,
-          "seed": 2}
-    dc = xgb.DMatrix(X, label=y)
-    bc = xgb.train(pd, dc, 6)  # CPU reference first (no process group)
+    # two configs: full-expansion depth 6, and depth 8 with gamma so some
+    # levels only partially expand (exercises the padded fixed-count
+    # hist/pair-sum allreduces of the distributed WHOLE-TREE chain)
+    param_sets = [
+        {"objective": "binary:logistic", "max_depth": 6, "max_bin": 64,
+         "seed": 2},
+        {"objective": "binary:logistic", "max_depth": 8, "max_bin": 128,
+         "gamma": 0.5, "seed": 2},
+    ]
+    refs = [xgb.train(pd, xgb.DMatrix(X, label=y), 6) for pd in param_sets]
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29771")
@@ -519,14 +525,15 @@ def test_native_driver_distributed_branch_rccl_world1():
     dist.init_process_group("nccl", rank=0, world_size=1)
     try:
         assert collective.is_distributed()
-        dg = xgb.DMatrix(X, label=y)
-        bg = xgb.train({**pd, "device": "cuda"}, dg, 6)
-        for tg, tc in zip(bg.trees, bc.trees):
-            assert tg.n_nodes == tc.n_nodes
-            assert np.array_equal(tg.split_index[:tg.n_nodes],
-                                  tc.split_index[:tc.n_nodes])
-            assert np.array_equal(tg.left[:tg.n_nodes],
-                                  tc.left[:tc.n_nodes])
+        for pd, bc in zip(param_sets, refs):
+            dg = xgb.DMatrix(X, label=y)
+            bg = xgb.train({**pd, "device": "cuda"}, dg, 6)
+            for tg, tc in zip(bg.trees, bc.trees):
+                assert tg.n_nodes == tc.n_nodes
+                assert np.array_equal(tg.split_index[:tg.n_nodes],
+                                      tc.split_index[:tc.n_nodes])
+                assert np.array_equal(tg.left[:tg.n_nodes],
+                                      tc.left[:tc.n_nodes])
     finally:
         dist.destroy_process_group()
 
@@ -600,3 +607,23 @@ def test_gpu_shap_multiclass_matches_cpu():
     got = bg.predict(dg, pred_contribs=True)
     assert got.shape == ref.shape == (3000, 3, 7)
     np.testing.assert_allclose(got, ref, atol=5e-4, rtol=1e-3)
+
+
+def test_native_driver_monotone_whole_tree_matches_cpu_trees():
+    """Monotone configs used to force the per-level sync driver; the
+    whole-tree chain now propagates fp64 weight bounds inside
+    ApplyKernel (device) and the host replay — trees must still match
+    the CPU oracle exactly."""
+    X, y = _data(30000, 6, seed=13)
+    pd = {"objective": "reg:squarederror", "max_depth": 8, "max_bin": 128,
+          "monotone_constraints": [1, -1, 0, 0, 1, 0], "eta": 0.4}
+    bc = xgb.train(pd, xgb.DMatrix(X, label=y), 8)
+    bg = xgb.train({**pd, "device": "cuda"}, xgb.DMatrix(X, label=y), 8)
+    assert len(bg.trees) == len(bc.trees)
+    for tg, tc in zip(bg.trees, bc.trees):
+        assert tg.n_nodes == tc.n_nodes
+        assert np.array_equal(tg.split_index[:tg.n_nodes],
+                              tc.split_index[:tc.n_nodes])
+        assert np.array_equal(tg.left[:tg.n_nodes], tc.left[:tc.n_nodes])
+        assert np.allclose(tg.split_cond[:tg.n_nodes],
+                           tc.split_cond[:tc.n_nodes], rtol=1e-6)

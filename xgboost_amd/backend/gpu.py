@@ -459,8 +459,11 @@ class GpuOps(SegmentedOpsMixin):
             if pool <= 2048:
                 rec = 1 + param.max_depth * pool
                 mp = n_rows // 1024 + pool + 2
+                # + 2 fp64 bound ping-pong buffers (monotone whole-tree)
+                # + per-level built-is-left mode records
                 wt_bytes = (48 * rec + 8 * rec + 8 * (param.max_depth + 2)
-                            + 96 * pool + 16 * mp + 32 * 1024)
+                            + 96 * pool + 32 * pool + 16 * mp
+                            + param.max_depth * pool + 32 * 1024)
                 ws["wt_max_ptasks"] = mp
                 ws["wt_ws"] = torch.empty(wt_bytes, dtype=torch.uint8,
                                           device=dev)
@@ -486,18 +489,26 @@ class GpuOps(SegmentedOpsMixin):
         cut_values_host = np.ascontiguousarray(cuts.values, np.float32)
         cb = None
         if collective.is_distributed():
-            pool_a, pool_b = ws["pool_a"], ws["pool_b"]
-            base_a, base_b = pool_a.data_ptr(), pool_b.data_ptr()
+            # the driver hands back raw device pointers into the hist
+            # pools (per-level + whole-tree hist reduce) or the wt arena
+            # (whole-tree pair-sum reduce); resolve to the owning torch
+            # tensor so torch.distributed handles stream ordering
+            regions = [(t.data_ptr(), t.numel() * t.element_size(), t)
+                       for t in (ws["pool_a"], ws["pool_b"])]
+            if ws["wt_ws"] is not None:
+                w = ws["wt_ws"]
+                regions.append((w.data_ptr(), w.numel(), w))
 
             def _allreduce(ptr, n_elems):
                 addr = ctypes.addressof(ptr.contents)
-                if base_a <= addr < base_a + pool_a.numel() * 8:
-                    base, pool = base_a, pool_a
-                else:
-                    base, pool = base_b, pool_b
-                off = (addr - base) // 8
-                view = pool.view(-1)[off:off + n_elems]
-                collective.allreduce_sum_(view)
+                for base, nbytes, t in regions:
+                    if base <= addr < base + nbytes:
+                        off = addr - base
+                        i64 = t.view(torch.uint8).view(-1)[
+                            off:off + 8 * n_elems].view(torch.int64)
+                        collective.allreduce_sum_(i64)
+                        return
+                raise RuntimeError("allreduce pointer outside workspaces")
 
             cb = self.hip.ALLREDUCE_FN(_allreduce)
         p8, p16 = self._gidx_ptrs()

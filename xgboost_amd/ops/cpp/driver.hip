@@ -115,6 +115,23 @@ double CalcWeight(double g, double h, const HostParams& p) {
   return w;
 }
 
+// device mirror of the host weight math (same doubles, same op order,
+// -ffp-contract=off ⇒ the whole-tree replay reproduces device bounds
+// bit-exactly)
+__device__ double DevCalcWeight(double g, double h, double lam, double alpha,
+                                double mds) {
+  double t = g;
+  if (alpha != 0.0) {
+    double s = (g > 0.0) ? 1.0 : ((g < 0.0) ? -1.0 : 0.0);
+    double m = fabs(g) - alpha;
+    if (m < 0.0) m = 0.0;
+    t = s * m;
+  }
+  double w = -t / (h + lam);
+  if (mds > 0.0) w = fmin(fmax(w, -mds), mds);
+  return w;
+}
+
 void ChunkTasks(const std::vector<Node*>& nodes, std::vector<BlockTask>* out,
                 long long min_rows = 1024, long long target_tasks = 2048) {
   out->clear();
@@ -185,13 +202,25 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
                                   int64_t* __restrict__ ps /* [2k][2] to
                                       zero, or null */,
                                   const int32_t* __restrict__ kp_dev,
-                                  int32_t* __restrict__ seg_out
+                                  int32_t* __restrict__ seg_out,
                                   /* null, or [2k][2] child segments in
                                      SLOT order: built j, subtracted
-                                     k+j (whole-tree mode) */) {
+                                     k+j (whole-tree mode) */
+                                  int zero_n, /* distributed whole-tree:
+                                     zero ps up to this host-known
+                                     worst-case bound so the padded
+                                     fixed-count allreduce sums zeros,
+                                     never stale ping-pong garbage */
+                                  uint8_t* __restrict__ mode_out
+                                  /* null, or [k]: resolved built-is-left
+                                     flag per pair — the EXPLICIT record
+                                     of which sibling was built (segment
+                                     adjacency is ambiguous for nodes
+                                     whose local shard is empty) */) {
   if (kp_dev != nullptr) k = *kp_dev;
   if (ps != nullptr) {
-    for (int i = threadIdx.x; i < 4 * k; i += blockDim.x) ps[i] = 0;
+    const int zn = max(4 * k, zero_n);
+    for (int i = threadIdx.x; i < zn; i += blockDim.x) ps[i] = 0;
   }
   // parallel load phase: thread-0 doing k dependent global loads costs
   // more than the whole rest of the kernel; gather child segments into
@@ -206,6 +235,7 @@ __global__ void HistTaskGenKernel(const int32_t* __restrict__ counters,
     if (mode == 2) mode = (split - pb) <= (pe - split) ? 1 : 0;
     s_begin[j] = mode ? pb : split;
     s_end[j] = mode ? split : pe;
+    if (mode_out != nullptr) mode_out[j] = (uint8_t)mode;
     if (seg_out != nullptr) {
       seg_out[2 * j] = s_begin[j];
       seg_out[2 * j + 1] = s_end[j];
@@ -294,7 +324,21 @@ __global__ __launch_bounds__(256) void ApplyKernel(
     int32_t* __restrict__ feat, int32_t* __restrict__ sbin,
     uint8_t* __restrict__ dl, int32_t* __restrict__ cnt,
     BlockTask* __restrict__ out_tasks, int32_t* __restrict__ desc,
-    int64_t* __restrict__ parent_ps, int32_t* __restrict__ parent_slot) {
+    int64_t* __restrict__ parent_ps, int32_t* __restrict__ parent_slot,
+    // distributed / monotone extensions (all null / 0 otherwise):
+    int choice_global,  // 1: build the smaller-GLOBAL-hessian child so
+                        // every rank builds (and allreduces) the SAME
+                        // sibling slot; 0: HistTaskGenKernel picks the
+                        // smaller LOCAL-row child (single-GPU only)
+    const float* __restrict__ maxabs,     // [2] scale derivation (mono)
+    const int8_t* __restrict__ monotone,  // [F] or null
+    double lam, double alpha, double mds,
+    const double* __restrict__ bounds_prev,  // [kn][2] slot order | null
+    double* __restrict__ bounds_out,         // [2kb][2] slot order | null
+    const uint8_t* __restrict__ mode_prev) { // [kp_prev] built-is-left
+                                             // flags of the PREVIOUS
+                                             // level's pairs, or null
+                                             // (root / legacy adjacency)
   __shared__ int s_order[2048];   // slots in pair order
   __shared__ uint8_t s_flag[2048];
   __shared__ int s_exp[2048];     // expanding slots (pair order)
@@ -309,7 +353,12 @@ __global__ __launch_bounds__(256) void ApplyKernel(
   } else {
     for (int p = threadIdx.x; p < kp_prev; p += blockDim.x) {
       const int a = p, b = kp_prev + p;
-      const bool a_left = segs[2 * a + 1] == segs[2 * b];
+      // which slot is the LEFT child: from the explicit built-is-left
+      // record when available (segment adjacency is ambiguous when this
+      // rank's local shard of the pair is empty — distributed mode)
+      const bool a_left = mode_prev != nullptr
+                              ? (mode_prev[p] != 0)
+                              : (segs[2 * a + 1] == segs[2 * b]);
       s_order[2 * p] = a_left ? a : b;
       s_order[2 * p + 1] = a_left ? b : a;
     }
@@ -348,12 +397,53 @@ __global__ __launch_bounds__(256) void ApplyKernel(
     desc[4 * j] = sb;
     desc[4 * j + 1] = se;
     desc[4 * j + 2] = j;
-    desc[4 * j + 3] = 2;  // device picks the smaller child
-    parent_ps[2 * j] = ps_prev[2 * e];
-    parent_ps[2 * j + 1] = ps_prev[2 * e + 1];
+    const long long pgq = ps_prev[2 * e], phq = ps_prev[2 * e + 1];
+    const long long lgq = best[6 * e + 3], lhq = best[6 * e + 4];
+    int mode = 2;  // device picks the smaller LOCAL child
+    if (choice_global) {
+      // rank-identical: global hessians come from the (allreduced)
+      // histogram's best record, so every rank builds the same slot
+      mode = (lhq <= phq - lhq) ? 1 : 0;
+    }
+    desc[4 * j + 3] = mode;
+    parent_ps[2 * j] = pgq;
+    parent_ps[2 * j + 1] = phq;
     parent_slot[j] = e;
     s_b[j] = sb;
     s_e[j] = se;
+    if (bounds_out != nullptr) {
+      // monotone bound propagation (same math as the per-level host
+      // path): child bounds in SLOT order — built j, subtracted kb+j
+      double lo = bounds_prev ? bounds_prev[2 * e] : -INFINITY;
+      double hi = bounds_prev ? bounds_prev[2 * e + 1] : INFINITY;
+      double llo = lo, lhi = hi, rlo = lo, rhi = hi;
+      const int c = monotone ? (int)monotone[f] : 0;
+      if (c != 0) {
+        const double gsc =
+            maxabs[0] > 0.f ? 1073741824.0 / (double)maxabs[0] : 1.0;
+        const double hsc =
+            maxabs[1] > 0.f ? 1073741824.0 / (double)maxabs[1] : 1.0;
+        const double ig = 1.0 / gsc, ih = 1.0 / hsc;
+        double wl = DevCalcWeight(lgq * ig, lhq * ih, lam, alpha, mds);
+        double wr = DevCalcWeight((pgq - lgq) * ig, (phq - lhq) * ih, lam,
+                                  alpha, mds);
+        wl = fmin(fmax(wl, lo), hi);
+        wr = fmin(fmax(wr, lo), hi);
+        const double mid = (wl + wr) / 2.0;
+        if (c > 0) {
+          lhi = fmin(hi, mid);
+          rlo = fmax(lo, mid);
+        } else {
+          llo = fmax(lo, mid);
+          rhi = fmin(hi, mid);
+        }
+      }
+      const bool built_left = (mode == 1);
+      bounds_out[2 * j] = built_left ? llo : rlo;
+      bounds_out[2 * j + 1] = built_left ? lhi : rhi;
+      bounds_out[2 * (kb + j)] = built_left ? rlo : llo;
+      bounds_out[2 * (kb + j) + 1] = built_left ? rhi : lhi;
+    }
   }
   // partition tasks over the expand segments, padded to max_ptasks
   __shared__ long long s_rpt;
@@ -634,11 +724,18 @@ int gbt_grow_tree(
   // end returns the per-level best-split and segment records, and the
   // host replays them to build the tree arrays (bit-identical: the
   // replay applies the same comparisons to the same doubles).
+  // Distributed (allreduce != null): same single-enqueue chain — the
+  // per-level hist/pair-sum allreduces are enqueued with a host-known
+  // WORST-CASE padded count (zero slots reduce to zeros), and the
+  // sibling build/subtract choice comes from GLOBAL hessians so every
+  // rank reduces the same slot layout.  Monotone: fp64 bound
+  // propagation runs inside ApplyKernel (scales derived on device).
   const bool whole_tree =
-      wt_ws != nullptr && allreduce == nullptr && !has_mono &&
+      wt_ws != nullptr &&
       maxabs_dev != nullptr && max_nodes_level <= 1024 && max_depth >= 2 &&
       wt_max_ptasks > 0;
   if (whole_tree) {
+    const int choice_global = (allreduce != nullptr || has_mono) ? 1 : 0;
     const int pool = 2 * max_nodes_level;
     const int rec_slots = 1 + max_depth * pool;
     size_t off = 0;
@@ -661,6 +758,9 @@ int gbt_grow_tree(
     const size_t o_pslot = carve((size_t)pool * 4);
     const size_t o_psa = carve((size_t)pool * 16);
     const size_t o_psb = carve((size_t)pool * 16);
+    const size_t o_bnda = has_mono ? carve((size_t)pool * 16) : 0;
+    const size_t o_bndb = has_mono ? carve((size_t)pool * 16) : 0;
+    const size_t o_mode = carve((size_t)max_depth * pool);
     if ((long long)off > wt_ws_bytes) return -9997;
     char* w = (char*)wt_ws;
     int64_t* best_rec = (int64_t*)(w + o_best);
@@ -676,6 +776,10 @@ int gbt_grow_tree(
     int64_t* d_pps = (int64_t*)(w + o_pps);
     int32_t* d_pslot = (int32_t*)(w + o_pslot);
     int64_t* ps_bufs[2] = {(int64_t*)(w + o_psa), (int64_t*)(w + o_psb)};
+    double* bnd_bufs[2] = {
+        has_mono ? (double*)(w + o_bnda) : nullptr,
+        has_mono ? (double*)(w + o_bndb) : nullptr};
+    uint8_t* d_mode = (uint8_t*)(w + o_mode);  // [max_depth][pool]
 
     {  // knodes[0]=1, kpairs[0]=0, seg_rec[0]={0, n_rows}
       const int slot = ctx->ring.next();
@@ -695,7 +799,8 @@ int gbt_grow_tree(
     // root evaluation straight into best_rec[0]
     gbt_evaluate(hist_pool_a, 1, n_bins, n_features, cut_ptrs_dev,
                  root_sums_dev, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
-                 max_delta_step, min_child_weight, nullptr, nullptr, nullptr,
+                 max_delta_step, min_child_weight,
+                 has_mono ? monotone_dev : nullptr, nullptr, nullptr,
                  nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, nullptr,
                  0, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, 1, n_features,
@@ -719,7 +824,13 @@ int gbt_grow_tree(
                          kn_arr + L, kp_arr + L, ps_prev, gamma, cut_ptrs_dev,
                          (long long)1024, (long long)2048, wt_max_ptasks,
                          kn_arr + L + 1, kp_arr + L + 1, d_feat, d_sbin,
-                         d_dl, d_cnt, d_pt, d_desc, d_pps, d_pslot);
+                         d_dl, d_cnt, d_pt, d_desc, d_pps, d_pslot,
+                         choice_global, maxabs_dev,
+                         has_mono ? monotone_dev : nullptr,
+                         reg_lambda, reg_alpha, max_delta_step,
+                         (L == 0 || !has_mono) ? nullptr : bnd_bufs[(L - 1) & 1],
+                         has_mono ? bnd_bufs[L & 1] : nullptr,
+                         L == 0 ? nullptr : d_mode + (size_t)(L - 1) * pool);
       gbt_partition(gidx8, gidx16, n_features, cur_ridx, alt_ridx, d_pt,
                     wt_max_ptasks, d_feat, d_sbin, d_dl, nullptr, nullptr,
                     n_bins_feat_dev, d_cnt, stream);
@@ -727,7 +838,9 @@ int gbt_grow_tree(
       hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
                          d_cnt, d_desc, 0, hist_min_rows, hist_tasks,
                          wt_max_htasks, tg_scratch, hist_tasks_dev, ps_next,
-                         kp_arr + L + 1, so);
+                         kp_arr + L + 1, so,
+                         allreduce != nullptr ? 2 * cap_kids : 0,
+                         d_mode + (size_t)L * pool);
       HIP_CHECK(hipMemsetAsync(next_pool, 0,
                                (size_t)cap_kids * hist_row * sizeof(int64_t),
                                stream));
@@ -735,6 +848,16 @@ int gbt_grow_tree(
                wt_max_htasks, next_pool, n_bins, feat_group_start_dev,
                bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
                use_shared, ps_next, stream);
+      if (allreduce) {
+        // fixed worst-case counts (host-known, rank-identical): built
+        // hist slots 0..cap/2-1 (unused slots are zeros — the memset
+        // covers the pool, HistTaskGenKernel zeroed the ps padding) and
+        // the built-child pair sums.  Sibling subtraction AFTER the
+        // reduce then yields global histograms/sums for every child.
+        allreduce((long long*)next_pool,
+                  (long long)(cap_kids / 2) * hist_row);
+        allreduce((long long*)ps_next, (long long)cap_kids);
+      }
       {
         const long long total = (long long)(cap_kids / 2 + 1) * hist_row;
         int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
@@ -744,7 +867,9 @@ int gbt_grow_tree(
       }
       gbt_evaluate(next_pool, cap_kids, n_bins, n_features, cut_ptrs_dev,
                    ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
-                   max_delta_step, min_child_weight, nullptr, nullptr,
+                   max_delta_step, min_child_weight,
+                   has_mono ? monotone_dev : nullptr,
+                   has_mono ? bnd_bufs[L & 1] : nullptr,
                    nullptr, nullptr, eval_gain, eval_bin, eval_dir,
                    eval_lsum, kn_arr + L + 1, 0, stream);
       gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, cap_kids,
@@ -758,7 +883,8 @@ int gbt_grow_tree(
     size_t r_kp = (r_seg + (size_t)rec_slots * 8 + 63) & ~63ULL;
     size_t r_rs = (r_kp + (size_t)(max_depth + 2) * 4 + 63) & ~63ULL;
     size_t r_ma = r_rs + 16;
-    size_t r_total = r_ma + 8;
+    size_t r_mode = (r_ma + 8 + 63) & ~63ULL;
+    size_t r_total = r_mode + (size_t)max_depth * pool;
     if (int e = ctx->ensure_readback(r_total)) return e;
     char* rb = (char*)ctx->readback_host;
     HIP_CHECK(hipMemcpyAsync(rb + r_best, best_rec, (size_t)rec_slots * 48,
@@ -772,12 +898,15 @@ int gbt_grow_tree(
                              hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipMemcpyAsync(rb + r_ma, maxabs_dev, 8,
                              hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipMemcpyAsync(rb + r_mode, d_mode, (size_t)max_depth * pool,
+                             hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     const int64_t* h_best = (const int64_t*)(rb + r_best);
     const int32_t* h_seg = (const int32_t*)(rb + r_seg);
     const int32_t* h_kp = (const int32_t*)(rb + r_kp);
     const int64_t* h_rs = (const int64_t*)(rb + r_rs);
     const float* h_ma = (const float*)(rb + r_ma);
+    const uint8_t* h_mode = (const uint8_t*)(rb + r_mode);
     g_scale = h_ma[0] > 0.f ? 1073741824.0 / (double)h_ma[0] : 1.0;
     h_scale = h_ma[1] > 0.f ? 1073741824.0 / (double)h_ma[1] : 1.0;
     inv_g = 1.0 / g_scale;
@@ -833,8 +962,12 @@ int gbt_grow_tree(
         out_default_left[nd->nid] = (uint8_t)nd->dir;
         out_loss_chg[nd->nid] = (float)nd->gain;
         const long long rgq = nd->gq - nd->lgq, rhq = nd->hq - nd->lhq;
-        const double wl = CalcWeight(nd->lgq * inv_g, nd->lhq * inv_h, p);
-        const double wr = CalcWeight(rgq * inv_g, rhq * inv_h, p);
+        double wl = CalcWeight(nd->lgq * inv_g, nd->lhq * inv_h, p);
+        double wr = CalcWeight(rgq * inv_g, rhq * inv_h, p);
+        // monotone: clamp to this node's bounds and propagate to the
+        // children — bit-identical replay of ApplyKernel's fp64 math
+        wl = std::fmin(std::fmax(wl, nd->lo), nd->hi);
+        wr = std::fmin(std::fmax(wr, nd->lo), nd->hi);
         out_sum_hess[nd->nid] = (float)((nd->lhq + rhq) * inv_h);
         out_base_weight[l] = (float)wl;
         out_base_weight[r] = (float)wr;
@@ -847,8 +980,21 @@ int gbt_grow_tree(
         ln.hq = nd->lhq;
         rn.gq = rgq;
         rn.hq = rhq;
-        ln.lo = rn.lo = -INFINITY;
-        ln.hi = rn.hi = INFINITY;
+        ln.lo = rn.lo = nd->lo;
+        ln.hi = rn.hi = nd->hi;
+        if (has_mono && nd->feature < n_features) {
+          const int c = monotone_host[nd->feature];
+          if (c != 0) {
+            const double mid = (wl + wr) / 2.0;
+            if (c > 0) {
+              ln.hi = std::fmin(nd->hi, mid);
+              rn.lo = std::fmax(nd->lo, mid);
+            } else {
+              ln.lo = std::fmax(nd->lo, mid);
+              rn.hi = std::fmin(nd->hi, mid);
+            }
+          }
+        }
         ln.gain = rn.gain = -INFINITY;
         ln.bin = rn.bin = -1;
         next_level.push_back(ln);
@@ -896,13 +1042,14 @@ int gbt_grow_tree(
       if (h_kp[depth + 1] != kbl) return -9991;  // replay divergence
       const int64_t* bo = h_best + (1 + (size_t)depth * pool) * 6;
       const int32_t* so = h_seg + (1 + (size_t)depth * pool) * 2;
+      const uint8_t* mo = h_mode + (size_t)depth * pool;
       for (int j = 0; j < kbl; ++j) {
         Node& ln = next_level[2 * j];
         Node& rn = next_level[2 * j + 1];
         const int sA = j, sB = kbl + j;
         const int ab = so[2 * sA], ae = so[2 * sA + 1];
         const int bb2 = so[2 * sB], be = so[2 * sB + 1];
-        const bool a_left = (ae == bb2);
+        const bool a_left = mo[j] != 0;  // built-is-left record
         ln.seg_begin = a_left ? ab : bb2;
         ln.seg_end = a_left ? ae : be;
         rn.seg_begin = a_left ? bb2 : ab;
@@ -1193,7 +1340,8 @@ int gbt_grow_tree(
     hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
                        cnt_dev, (const int32_t*)(d + off_desc), kb,
                        hist_min_rows, hist_tasks, max_tasks, tg_scratch,
-                       hist_tasks_dev, eval_ps, nullptr, nullptr);
+                       hist_tasks_dev, eval_ps, nullptr, nullptr, 0,
+                       nullptr);
     HIP_CHECK(hipMemsetAsync(next_pool, 0,
                              (size_t)kb * hist_row * sizeof(int64_t), stream));
     gbt_hist(gidx8, gidx16, n_features, qgpair, cur_ridx, hist_tasks_dev,
