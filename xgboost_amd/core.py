@@ -162,8 +162,32 @@ class Booster:
         if hess is not None:
             # approx: hessian-weighted cuts regenerated per tree
             # (reference: grow_histmaker/grow_gpu_approx, updater_approx.cc:46)
+            if self.device.type == "cuda":
+                # fully on-device regen like the reference's device path
+                # (updater_gpu_hist.cu:809 ApproxBatch regen): DeviceSketch
+                # weighted cuts + gbt_compress, no host round-trip; the
+                # sketch handles the distributed merge internally
+                from .backend.gpu import GpuOps
+                from .data import quantize_dense_device
+                from .gpu_sketch import device_cuts
+                key = self._pin(dmat)
+                xc = self.__dict__.setdefault("_approx_x_cache", {})
+                Xd = xc.get(key)
+                if Xd is None:
+                    dd = (dmat.device_data()
+                          if hasattr(dmat, "device_data") else None)
+                    Xd = dd if dd is not None else torch.as_tensor(
+                        dmat.raw_data(), device=self.device)
+                    xc[key] = Xd
+                h = torch.as_tensor(hess, device=Xd.device).abs() + 1e-16
+                cuts = device_cuts(Xd, self.tparam.max_bin,
+                                   missing=dmat.missing, weights=h,
+                                   feature_types=dmat.info.feature_types)
+                qm = quantize_dense_device(Xd, cuts, dmat.missing)
+                return GpuOps(qm)
             from .data import quantize_dense
             from .quantile import make_cuts
+            hess = np.asarray(hess)
             if collective.is_distributed():
                 # ranks must agree on the re-sketched cuts: merge the
                 # weighted per-rank summaries like the in-core sketch
@@ -182,9 +206,6 @@ class Booster:
                                  feature_types=dmat.info.feature_types,
                                  missing=dmat.missing)
             qm = quantize_dense(dmat.raw_data(), cuts, dmat.missing)
-            if self.device.type == "cuda":
-                from .backend.gpu import GpuOps
-                return GpuOps(qm.to(self.device))
             return CpuOps(qm)
         key = self._pin(dmat)
         ops = self._ops_cache.get(key)
@@ -275,7 +296,7 @@ class Booster:
         for k in range(n_out):
             if is_approx:
                 ops = self._ops_for(
-                    dtrain, hess=hess[:, k].detach().cpu().numpy())
+                    dtrain, hess=hess[:, k].detach())
             for ptree in range(self.tparam.num_parallel_tree):
                 gpair = torch.stack([grad[:, k], hess[:, k]], dim=1).contiguous()
                 gpair = self._subsample(gpair, seed + 7919 * ptree + 104729 * k)

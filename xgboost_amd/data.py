@@ -124,50 +124,15 @@ def _np_dtype(td: torch.dtype):
 
 
 def make_cuts_device(Xd: torch.Tensor, max_bin: int, missing: float,
-                     feature_types=None) -> HistogramCuts:
-    """Cuts from a device-resident matrix: nanquantile on the GPU, final
-    cut selection on host from the small [K, f] summary."""
-    from .sketch import cuts_from_summaries
-    n, f = Xd.shape
-    if np.isnan(missing):
-        Xm = Xd
-    else:
-        Xm = torch.where(Xd == missing, torch.full_like(Xd, float("nan")), Xd)
-    finite = torch.isfinite(Xm)
-    cnt = finite.sum(dim=0)
-    K = max(64, 8 * max_bin)
-    qs = torch.from_numpy(((np.arange(K) + 0.5) / K).astype(np.float32)
-                          ).to(Xd.device)
-    # nanquantile with 'lower' keeps actual data values (distinct-value
-    # cuts must be exact data points)
-    pts = torch.nanquantile(Xm.double(), qs.double(), dim=0,
-                            interpolation="lower").float()  # [K, f]
-    mins = torch.where(cnt > 0,
-                       torch.nan_to_num(Xm, nan=float("inf")).amin(dim=0),
-                       torch.zeros(f, device=Xd.device))
-    maxs = torch.where(cnt > 0,
-                       torch.nan_to_num(Xm, nan=float("-inf")).amax(dim=0),
-                       torch.zeros(f, device=Xd.device))
-    pts_h = pts.cpu().numpy()
-    cnt_h = cnt.cpu().numpy()
-    mins_h = mins.cpu().numpy()
-    maxs_h = maxs.cpu().numpy()
-    summaries = []
-    for j in range(f):
-        if feature_types is not None and feature_types[j] == "c":
-            col = Xm[:, j]
-            cats = torch.unique(col[finite[:, j]]).cpu().numpy().astype(
-                np.float32)
-            summaries.append(("c", cats, float(cnt_h[j])))
-        elif cnt_h[j] == 0:
-            summaries.append(("q", np.zeros(0, np.float32), 0.0))
-        else:
-            arr = np.concatenate([[mins_h[j]], pts_h[:, j], [maxs_h[j]]])
-            summaries.append(("q", arr.astype(np.float32), float(cnt_h[j])))
-    # distributed: allgather the per-rank summaries before cut selection
-    from .sketch import sketch_cuts_batches
-    return sketch_cuts_batches([summaries], max_bin, f,
-                               list(feature_types) if feature_types else None)
+                     feature_types=None, weights=None) -> HistogramCuts:
+    """Cuts from a device-resident matrix via the weighted device
+    quantile sketch (gpu_sketch.DeviceSketch — the SketchContainer
+    equivalent, reference src/common/quantile.cuh:41): exact weighted
+    ranks + bounded-error prune, rank-merged when distributed."""
+    from .gpu_sketch import device_cuts
+    return device_cuts(Xd, max_bin, missing=missing, weights=weights,
+                       feature_types=(list(feature_types)
+                                      if feature_types else None))
 
 
 def quantize_dense_device(Xd: torch.Tensor, cuts: HistogramCuts,
