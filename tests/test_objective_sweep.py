@@ -19,7 +19,12 @@ def _data_for(name, n=400, seed=0):
         y = np.abs(X[:, :3]).argmax(axis=1).astype(np.float32)
         params["num_class"] = 3
     elif name.startswith("rank"):
-        y = np.clip((X[:, 0] * 2 + 2).astype(int), 0, 3).astype(np.float32)
+        if name == "rank:map":
+            # reference CheckPreLabels: MAP requires binary relevance
+            y = (X[:, 0] > 0).astype(np.float32)
+        else:
+            y = np.clip((X[:, 0] * 2 + 2).astype(int), 0,
+                        3).astype(np.float32)
         kwargs["qid"] = np.repeat(np.arange(20), n // 20)
     elif name == "survival:cox":
         t = np.exp(X[:, 0] * 0.5 + 2)

@@ -621,6 +621,23 @@ def _weighted_quantile(v: np.ndarray, w: Optional[np.ndarray], q: float) -> floa
 
 
 class _LambdaRankBase(Objective):
+    """LambdaMART gradients, fully vectorized torch ops (device-resident
+    when predictions live on the GPU).
+
+    Reference math reproduced exactly (src/objective/lambdarank_obj.h):
+    - LambdaGrad (:93): sigma = sigmoid(s_high - s_low),
+      lambda = (sigma - 1) * |delta|, hessian = max(sigma(1-sigma), eps)
+      * |delta| * 2; optional score normalization |delta| /= (|ds|+0.01)
+      when the group is not degenerate (lambdarank_score_normalization,
+      default true).
+    - MakePairs (:225): topk = truncation pairs (i<min(k,n), j>i over the
+      model-ranked list, k default 32); mean = per-doc sampled opponents
+      outside the label bucket (num_pair default 1).
+    - Normalization (lambdarank_obj.cc:230): topk
+      log2(1+sum_lambda)/sum_lambda per group; mean 1/num_pair; group
+      weights scaled by n_groups/sum(weights) (ranking_utils.cc:44).
+    """
+
     task = "ranking"
 
     def __init__(self, params=None):
@@ -631,106 +648,214 @@ class _LambdaRankBase(Objective):
                                                "topk"))
         self.normalize = bool(self.params.get("lambdarank_normalization",
                                               True))
+        self.score_norm = bool(self.params.get(
+            "lambdarank_score_normalization", True))
+        self.exp_gain = bool(self.params.get("ndcg_exp_gain", True))
+        self.unbiased = bool(self.params.get("lambdarank_unbiased", False))
+        if self.unbiased:
+            raise NotImplementedError(
+                "lambdarank_unbiased (position debiasing) is not implemented")
+
+    def _k(self) -> int:
+        """reference LambdaRankParam::NumPair (ranking_utils.h:102)."""
+        if self.num_pair > 0:
+            return self.num_pair
+        return 32 if self.pair_method == "topk" else 1
 
     def _groups(self, info):
         if info.group_ptr is None:
             return np.array([0, info.num_row], dtype=np.int64)
         return np.asarray(info.group_ptr, dtype=np.int64)
 
-    def _delta(self, y_sorted, ranks_i, ranks_j, i_idx, j_idx, inv_idcg):
-        """|delta metric| for swapping documents at ranks_i/ranks_j."""
+    # -- pair construction -------------------------------------------------
+    @staticmethod
+    def _topk_pairs(cnt: torch.Tensor, gp: torch.Tensor, k: int):
+        """All (i, j) rank-position pairs with i < min(k, cnt), i < j < cnt
+        (reference MakePairs truncation branch).  Returns flat rank-list
+        positions (a, b) and the pair's group id."""
+        dev = cnt.device
+        G = cnt.numel()
+        kk = torch.minimum(cnt, torch.tensor(k, device=dev))
+        m = kk * (cnt - 1) - kk * (kk - 1) // 2  # pairs per group
+        M = int(m.sum())
+        if M == 0:
+            z = torch.zeros(0, dtype=torch.long, device=dev)
+            return z, z, z
+        pg = torch.repeat_interleave(torch.arange(G, device=dev), m)
+        off = torch.zeros(G + 1, dtype=torch.long, device=dev)
+        torch.cumsum(m, 0, out=off[1:])
+        t = torch.arange(M, device=dev) - off[pg]
+        c = cnt[pg]
+        # i = largest i0 with prefix(i0) <= t, prefix(i) = i(c-1) - i(i-1)/2
+        b = (2 * c - 1).double()
+        i = ((b - (b * b - 8.0 * t.double()).clamp(min=0).sqrt()) / 2
+             ).floor().long().clamp(min=0)
+        for _ in range(2):  # fp fix-up
+            pref = i * (c - 1) - i * (i - 1) // 2
+            i = (i - (pref > t).long()).clamp(min=0)
+            nxt = (i + 1) * (c - 1) - (i + 1) * i // 2
+            i = i + ((nxt <= t) & (i + 1 < c)).long()
+        pref = i * (c - 1) - i * (i - 1) // 2
+        j = i + 1 + (t - pref)
+        return gp[pg] + i, gp[pg] + j, pg
+
+    def _mean_pairs(self, y_by_rank: torch.Tensor, cnt: torch.Tensor,
+                    gp: torch.Tensor, n_samples: int, seed: int):
+        """Sampled pairs: for each doc, n_samples opponents drawn from
+        outside its label bucket on the label-sorted list (reference
+        MakePairs sampling branch)."""
+        dev = y_by_rank.device
+        G = cnt.numel()
+        N = y_by_rank.numel()
+        gid = torch.repeat_interleave(torch.arange(G, device=dev), cnt)
+        # label-desc order over rank positions, grouped
+        o1 = torch.argsort(-y_by_rank, stable=True)
+        o2 = torch.argsort(gid[o1], stable=True)
+        ls = o1[o2]  # ls[q] = rank position of q-th label-sorted doc
+        ys = y_by_rank[ls]
+        # bucket boundaries within groups
+        start = torch.zeros(N, dtype=torch.bool, device=dev)
+        start[gp[:-1][cnt > 0]] = True
+        if N > 1:
+            start[1:] |= (ys[1:] != ys[:-1]) & (gid[1:] == gid[:-1])
+        bucket = torch.cumsum(start.long(), 0) - 1
+        b_first = torch.nonzero(start, as_tuple=True)[0]
+        b_last = torch.empty_like(b_first)
+        if b_first.numel() > 1:
+            b_last[:-1] = b_first[1:] - 1
+        b_last[-1] = N - 1
+        i_of = (b_first - gp[gid[b_first]])[bucket]   # bucket start (i)
+        j_of = (b_last - gp[gid[b_last]] + 1)[bucket]  # bucket end (j)
+        c = cnt[gid]
+        n_lefts = i_of
+        n_rights = c - j_of
+        tot = n_lefts + n_rights
+        valid = tot > 0
+        # deterministic sampling (CPU RNG, then moved to the device)
+        rng = np.random.RandomState(seed)
+        r = torch.as_tensor(
+            rng.randint(0, 2 ** 31 - 1, size=(n_samples, N)), device=dev)
+        out_a, out_b, out_g = [], [], []
+        for sidx in range(n_samples):
+            ridx = r[sidx] % tot.clamp(min=1)
+            ridx = torch.where(ridx >= n_lefts, ridx - i_of + j_of, ridx)
+            bpos = (gp[gid] + ridx).clamp(max=N - 1)
+            out_a.append(ls[valid])
+            out_b.append(ls[bpos][valid])
+            out_g.append(gid[valid])
+        return (torch.cat(out_a), torch.cat(out_b), torch.cat(out_g))
+
+    # -- delta hooks -------------------------------------------------------
+    def _prepare(self, y_by_rank, cnt, gp, pos):
+        """Per-iteration group statistics for the delta (subclasses)."""
+        return None
+
+    def _delta_vec(self, stats, y_high, y_low, rank_high, rank_low, g):
         raise NotImplementedError
 
+    # -- main --------------------------------------------------------------
     def get_gradient(self, preds, info, it):
-        p = preds.detach().cpu().numpy().reshape(-1).astype(np.float64)
-        y = np.asarray(info.labels, np.float64).reshape(-1)
-        gp = self._groups(info)
-        g = np.zeros_like(p)
-        h = np.zeros_like(p)
-        rng = np.random.RandomState(1234 + it)
-        for gi in range(len(gp) - 1):
-            s, e = int(gp[gi]), int(gp[gi + 1])
-            if e - s < 2:
-                continue
-            self._group_gradient(p[s:e], y[s:e], g[s:e], h[s:e], rng)
-        gt = torch.as_tensor(g.astype(np.float32),
-                             device=preds.device).view(preds.shape)
-        ht = torch.as_tensor(np.maximum(h, 1e-16).astype(np.float32),
-                             device=preds.device).view(preds.shape)
-        w = _weights(info, preds)
-        if w is not None and info.weights.shape[0] == len(gp) - 1:
-            # per-group weights
-            wr = np.repeat(np.asarray(info.weights, np.float32),
-                           np.diff(gp))
-            wt = torch.as_tensor(wr, device=preds.device).view(preds.shape)
-            gt, ht = gt * wt, ht * wt
-        elif w is not None:
-            gt, ht = gt * w.view(preds.shape), ht * w.view(preds.shape)
+        dev = preds.device
+        p = preds.detach().double().reshape(-1)
+        N = p.numel()
+        y = torch.as_tensor(np.asarray(info.labels, np.float64).reshape(-1),
+                            device=dev)
+        gp_np = self._groups(info)
+        gp = torch.as_tensor(gp_np, device=dev)
+        cnt = gp[1:] - gp[:-1]
+        G = cnt.numel()
+        gid = torch.repeat_interleave(torch.arange(G, device=dev), cnt)
+        # model-rank order (pred desc, stable) within each group
+        o1 = torch.argsort(-p, stable=True)
+        o2 = torch.argsort(gid[o1], stable=True)
+        order = o1[o2]           # order[gp[g]+r] = doc index at rank r
+        pos = torch.arange(N, device=dev) - gp[gid]
+        y_by_rank = y[order]
+        stats = self._prepare(y_by_rank, cnt, gp, pos)
+
+        if self.pair_method == "topk":
+            a_flat, b_flat, pg = self._topk_pairs(cnt, gp, self._k())
+        else:
+            a_flat, b_flat, pg = self._mean_pairs(
+                y_by_rank, cnt, gp, self._k(), seed=1234 + it)
+        g_out = torch.zeros(N, dtype=torch.float64, device=dev)
+        h_out = torch.zeros(N, dtype=torch.float64, device=dev)
+        if a_flat.numel():
+            da = order[a_flat]
+            db = order[b_flat]
+            ya, yb = y[da], y[db]
+            keep = ya != yb
+            da, db, ya, yb = da[keep], db[keep], ya[keep], yb[keep]
+            ra = (a_flat - gp[pg])[keep]
+            rb = (b_flat - gp[pg])[keep]
+            pgk = pg[keep]
+            swap = ya < yb
+            idx_high = torch.where(swap, db, da)
+            idx_low = torch.where(swap, da, db)
+            y_high = torch.where(swap, yb, ya)
+            y_low = torch.where(swap, ya, yb)
+            rank_high = torch.where(swap, rb, ra)
+            rank_low = torch.where(swap, ra, rb)
+            s_high = p[idx_high]
+            s_low = p[idx_low]
+            sig = torch.sigmoid(s_high - s_low)
+            delta = self._delta_vec(stats, y_high, y_low, rank_high,
+                                    rank_low, pgk).abs()
+            if self.score_norm:
+                # skip for degenerate groups (best score == worst score,
+                # reference LambdaGrad best/worst check)
+                best = p[order[gp[:-1].clamp(max=max(N - 1, 0))]]
+                worst = p[order[(gp[1:] - 1).clamp(min=0)]]
+                ok = (best != worst)[pgk]
+                delta = torch.where(
+                    ok, delta / ((s_high - s_low).abs() + 0.01), delta)
+            lam = (sig - 1.0) * delta
+            hess = torch.clamp(sig * (1.0 - sig), min=1e-16) * delta * 2.0
+            g_out.index_add_(0, idx_high, lam)
+            g_out.index_add_(0, idx_low, -lam)
+            h_out.index_add_(0, idx_high, hess)
+            h_out.index_add_(0, idx_low, hess)
+            if self.normalize:
+                if self.pair_method == "topk":
+                    s_g = torch.zeros(G, dtype=torch.float64, device=dev)
+                    s_g.index_add_(0, pgk, -2.0 * lam)
+                    norm_g = torch.where(
+                        s_g > 0,
+                        torch.log2(1.0 + s_g) / s_g.clamp(min=1e-300),
+                        torch.ones_like(s_g))
+                    g_out *= norm_g[gid]
+                    h_out *= norm_g[gid]
+                else:
+                    g_out /= self._k()
+                    h_out /= self._k()
+        # group weights * weight_norm (ranking_utils.cc:44)
+        if info.weights is not None:
+            w_np = np.asarray(info.weights, np.float64).reshape(-1)
+            if w_np.shape[0] == G:
+                wn = G / max(float(w_np.sum()), 1e-300)
+                w_doc = torch.as_tensor(w_np, device=dev)[gid] * wn
+            else:
+                w_doc = torch.as_tensor(w_np, device=dev)
+            g_out *= w_doc
+            h_out *= w_doc
+        gt = g_out.float().view(preds.shape)
+        ht = torch.clamp(h_out, min=1e-16).float().view(preds.shape)
         return gt, ht
-
-    def _make_pairs(self, n, y, order, rng):
-        """Yield (i, j) index pairs (into the group) with y[i] > y[j]."""
-        if self.pair_method == "mean" and self.num_pair > 0:
-            k = self.num_pair
-            pairs = []
-            for i in range(n):
-                js = rng.randint(0, n, size=k)
-                for j in js:
-                    if y[i] > y[j]:
-                        pairs.append((i, j))
-                    elif y[j] > y[i]:
-                        pairs.append((j, i))
-            return pairs
-        # topk/full: all label-discordant pairs (n<=512 full, else truncate)
-        pairs = []
-        cap = 512
-        idx = order[:cap]
-        for a in range(len(idx)):
-            for b in range(a + 1, len(idx)):
-                i, j = idx[a], idx[b]
-                if y[i] > y[j]:
-                    pairs.append((i, j))
-                elif y[j] > y[i]:
-                    pairs.append((j, i))
-        return pairs
-
-    def _group_gradient(self, p, y, g, h, rng):
-        n = len(p)
-        order = np.argsort(-p, kind="stable")
-        ranks = np.empty(n, dtype=np.int64)
-        ranks[order] = np.arange(n)  # 0-based rank by prediction
-        inv_idcg = self._inv_idcg(y)
-        pairs = self._make_pairs(n, y, order, rng)
-        if not pairs:
-            return
-        total_lambda = 0.0
-        for i, j in pairs:
-            delta = self._delta(y, ranks[i], ranks[j], i, j, inv_idcg)
-            sij = p[i] - p[j]
-            rho = 1.0 / (1.0 + np.exp(sij))  # d/ds of log(1+e^-s)
-            lam = -rho * delta
-            hess = max(rho * (1.0 - rho) * delta, 1e-16)
-            g[i] += lam
-            g[j] -= lam
-            h[i] += hess
-            h[j] += hess
-            total_lambda += abs(lam)
-        if self.normalize and total_lambda > 0:
-            norm = np.log2(1.0 + total_lambda) / total_lambda
-            g *= norm
-            h *= norm
-
-    def _inv_idcg(self, y):
-        gains = np.sort(2.0 ** y - 1.0)[::-1]
-        disc = 1.0 / np.log2(np.arange(2, len(y) + 2))
-        idcg = float((gains * disc).sum())
-        return 1.0 / idcg if idcg > 0 else 0.0
 
     def init_estimation(self, info) -> float:
         return 0.5
 
     def prob_to_margin(self, base_score):
         return base_score
+
+    def save_config(self):
+        return {"name": self.name, "lambdarank_param": {
+            "lambdarank_pair_method": self.pair_method,
+            "lambdarank_num_pair_per_sample": str(self._k()),
+            "lambdarank_normalization": str(int(self.normalize)),
+            "lambdarank_score_normalization": str(int(self.score_norm)),
+            "ndcg_exp_gain": str(int(self.exp_gain)),
+        }}
 
 
 @register("rank:ndcg")
@@ -739,12 +864,37 @@ class LambdaRankNDCG(_LambdaRankBase):
     def default_metric(self):
         return "ndcg"
 
-    def _delta(self, y, rank_i, rank_j, i, j, inv_idcg):
-        gain_i = 2.0 ** y[i] - 1.0
-        gain_j = 2.0 ** y[j] - 1.0
-        disc_i = 1.0 / np.log2(rank_i + 2.0)
-        disc_j = 1.0 / np.log2(rank_j + 2.0)
-        return abs((gain_i - gain_j) * (disc_i - disc_j)) * inv_idcg
+    def _prepare(self, y_by_rank, cnt, gp, pos):
+        """inv_IDCG per group, truncated at TopK like NDCGCache
+        (ranking_utils.cc:96-108)."""
+        dev = y_by_rank.device
+        G = cnt.numel()
+        N = y_by_rank.numel()
+        gid = torch.repeat_interleave(torch.arange(G, device=dev), cnt)
+        # label-desc within group
+        o1 = torch.argsort(-y_by_rank, stable=True)
+        o2 = torch.argsort(gid[o1], stable=True)
+        ysort = y_by_rank[o1[o2]]
+        gain = (2.0 ** ysort - 1.0) if self.exp_gain else ysort
+        disc = 1.0 / torch.log2(pos.double() + 2.0)
+        topk = self._k() if self.pair_method == "topk" else N + 1
+        mask = pos < topk
+        idcg = torch.zeros(G, dtype=torch.float64, device=dev)
+        idcg.index_add_(0, gid[mask], (gain * disc)[mask])
+        inv_idcg = torch.where(idcg > 0, 1.0 / idcg.clamp(min=1e-300),
+                               torch.zeros_like(idcg))
+        return {"inv_idcg": inv_idcg}
+
+    def _delta_vec(self, stats, y_high, y_low, rank_high, rank_low, g):
+        """DeltaNDCG (lambdarank_obj.h:42)."""
+        if self.exp_gain:
+            gain_h = 2.0 ** y_high - 1.0
+            gain_l = 2.0 ** y_low - 1.0
+        else:
+            gain_h, gain_l = y_high, y_low
+        disc_h = 1.0 / torch.log2(rank_high.double() + 2.0)
+        disc_l = 1.0 / torch.log2(rank_low.double() + 2.0)
+        return (gain_h - gain_l) * (disc_h - disc_l) * stats["inv_idcg"][g]
 
 
 @register("rank:map")
@@ -753,10 +903,50 @@ class LambdaRankMAP(_LambdaRankBase):
     def default_metric(self):
         return "map"
 
-    def _delta(self, y, rank_i, rank_j, i, j, inv_idcg):
-        # MAP delta approximated by reciprocal-rank difference on binary rel
-        ri, rj = min(rank_i, rank_j), max(rank_i, rank_j)
-        return abs(1.0 / (ri + 1.0) - 1.0 / (rj + 1.0))
+    def _prepare(self, y_by_rank, cnt, gp, pos):
+        """MAPStat (lambdarank_obj.cc): n_rel = running count of relevant
+        docs down the ranked list, acc = running sum of label/rank."""
+        if not bool(((y_by_rank == 0) | (y_by_rank == 1)).all()):
+            raise ValueError("rank:map requires binary labels")
+        dev = y_by_rank.device
+        G = cnt.numel()
+        gid = torch.repeat_interleave(torch.arange(G, device=dev), cnt)
+        nz = cnt > 0
+        # segmented cumsums via global cumsum minus per-group base
+        cw = torch.cumsum(y_by_rank, 0)
+        base = torch.zeros(G, dtype=torch.float64, device=dev)
+        base[nz] = cw[gp[:-1][nz]] - y_by_rank[gp[:-1][nz]]
+        n_rel = cw - base[gid]
+        a = y_by_rank / (pos.double() + 1.0)
+        ca = torch.cumsum(a, 0)
+        base_a = torch.zeros(G, dtype=torch.float64, device=dev)
+        base_a[nz] = ca[gp[:-1][nz]] - a[gp[:-1][nz]]
+        acc = ca - base_a[gid]
+        last = (gp[1:] - 1).clamp(min=0)
+        n_total = torch.where(nz, n_rel[last],
+                              torch.zeros(G, dtype=torch.float64, device=dev))
+        return {"n_rel": n_rel, "acc": acc, "n_total": n_total, "gp": gp}
+
+    def _delta_vec(self, stats, y_high, y_low, rank_high, rank_low, g):
+        """DeltaMAP (lambdarank_obj.h:62) with the rank min/max swap the
+        reference's caller applies (lambdarank_obj.cu:484-491): positions
+        are ordered while the labels keep their high/low roles."""
+        gp = stats["gp"]
+        rh = torch.minimum(rank_high, rank_low)
+        rl = torch.maximum(rank_high, rank_low)
+        base = gp[g]
+        n_rel = stats["n_rel"]
+        acc = stats["acc"]
+        n_tot = stats["n_total"][g].clamp(min=1e-300)
+        r_h = rh.double() + 1.0
+        r_l = rl.double() + 1.0
+        m = n_rel[base + rl]
+        n = n_rel[base + rh]
+        b = acc[base + rl - 1] - acc[base + rh]
+        swapped = y_high < y_low  # the relevant doc already ranks higher
+        a1 = m / r_l - (n + 1.0) / r_h   # y_high < y_low branch
+        a2 = n / r_h - m / r_l           # y_high > y_low branch
+        return torch.where(swapped, a1 - b, a2 + b) / n_tot
 
 
 @register("rank:pairwise")
@@ -765,5 +955,5 @@ class LambdaRankPairwise(_LambdaRankBase):
     def default_metric(self):
         return "map"
 
-    def _delta(self, y, rank_i, rank_j, i, j, inv_idcg):
-        return 1.0
+    def _delta_vec(self, stats, y_high, y_low, rank_high, rank_low, g):
+        return torch.ones_like(y_high)
