@@ -651,3 +651,38 @@ def test_gpu_lambdarank_device_gradients():
                       "device": "cuda"}, d2, 10, verbose_eval=False)
     res2 = bst2.eval_set([(d2, "train")], 9)
     assert float(res2.split(":")[-1]) > 0.8, res2
+
+
+def test_gpu_shap_interactions_matches_cpu():
+    """pred_interactions on GPU (shap_ix.hip) vs the exact CPU
+    conditional TreeSHAP."""
+    from xgboost_amd.shap import shap_interactions
+    X, y = _data(800, 6, seed=23, missing_frac=0.05)
+    d = xgb.DMatrix(X, label=y)
+    bst_c = xgb.train({"objective": "binary:logistic", "max_depth": 5,
+                       "eta": 0.3}, d, 8, verbose_eval=False)
+    ref = shap_interactions(bst_c, d)
+    dg = xgb.DMatrix(X, label=y)
+    bst_g = xgb.train({"objective": "binary:logistic", "max_depth": 5,
+                       "eta": 0.3, "device": "cuda"}, dg, 8,
+                      verbose_eval=False)
+    # same data/params -> identical trees (tested elsewhere); compare
+    # the GPU interactions of the GPU model vs CPU interactions of it
+    bst_g.device = bst_g.device  # noqa: B018
+    got = shap_interactions(bst_g, dg)
+    cpu_of_g = None
+    import torch as _t
+    dev = bst_g.device
+    try:
+        bst_g.device = _t.device("cpu")
+        cpu_of_g = shap_interactions(bst_g, dg)
+    finally:
+        bst_g.device = dev
+    assert got.shape == cpu_of_g.shape
+    assert np.allclose(got, cpu_of_g, atol=2e-4), np.abs(
+        got.astype(np.float64) - cpu_of_g.astype(np.float64)).max()
+    # row sums reproduce the margin (SHAP completeness)
+    margin = bst_g.predict(dg, output_margin=True)
+    total = got.sum(axis=(1, 2))
+    assert np.allclose(total, margin, atol=1e-3)
+    del ref

@@ -80,3 +80,101 @@ def test_multiclass_contribs():
     assert contribs.shape == (200, 3, 5)
     margin = bst.predict(d, output_margin=True)
     assert np.allclose(contribs.sum(axis=2), margin, atol=1e-4)
+
+
+def test_path_pair_decomposition_matches_exact_interactions():
+    """Python mirror of the shap_ix.hip algorithm (extend once, unwind
+    each element b, accumulate conditional pair terms) vs the exact
+    conditioned-TreeSHAP CPU implementation — validates the math the
+    GPU kernel runs."""
+    import xgboost_amd as xgb
+    from xgboost_amd.shap import shap_interactions, shap_values
+    from xgboost_amd.shap_paths import build_path_table
+
+    def extend_pw(zs, ones):
+        pw = [1.0]
+        for j, (z, o) in enumerate(zip(zs, ones)):
+            mm = j + 1
+            pw.append(o * pw[mm - 1] * mm / (mm + 1))
+            for i in range(mm - 1, 0, -1):
+                pw[i] = (o * pw[i - 1] * i / (mm + 1)
+                         + z * pw[i] * (mm - i) / (mm + 1))
+            pw[0] = z * pw[0] * mm / (mm + 1)
+        return pw
+
+    def unwind_pw(pw, z, o, d):
+        out = [0.0] * d
+        n_ = pw[d]
+        if o != 0:
+            for j in range(d - 1, -1, -1):
+                t = n_ * (d + 1) / ((j + 1) * o)
+                out[j] = t
+                n_ = pw[j] - t * z * (d - j) / (d + 1)
+        else:
+            for j in range(d - 1, -1, -1):
+                out[j] = pw[j] * (d + 1) / (z * (d - j))
+        return out
+
+    def unwound_sum_pw(pw, z, o, d):
+        total = 0.0
+        n_ = pw[d]
+        if o != 0:
+            for j in range(d - 1, -1, -1):
+                t = n_ * (d + 1) / ((j + 1) * o)
+                total += t
+                n_ = pw[j] - t * z * (d - j) / (d + 1)
+        else:
+            for j in range(d - 1, -1, -1):
+                total += pw[j] * (d + 1) / (z * (d - j))
+        return total
+
+    rng = np.random.RandomState(11)
+    n, f = 150, 5
+    X = rng.randn(n, f).astype(np.float32)
+    X[rng.rand(n, f) < 0.05] = np.nan
+    y = (np.nan_to_num(X[:, 0] * X[:, 1] + X[:, 2]) > 0).astype(np.float32)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 4, "eta": 0.5,
+                     "objective": "binary:logistic"}, d, 4,
+                    verbose_eval=False)
+    ref = shap_interactions(bst, d).astype(np.float64)
+
+    pp, pg, ef, elo, ehi, emiss, ez, pv, bias = build_path_table(
+        bst.trees, bst.tree_info)
+    base = shap_values(bst, d).astype(np.float64)
+    C = f + 1
+    out = np.zeros((n, C, C))
+    for p in range(len(pg)):
+        s, e = pp[p], pp[p + 1]
+        M = e - s
+        if M < 2 or pv[p] == 0.0:
+            continue
+        zs = list(ez[s:e])
+        fs = ef[s:e]
+        v = pv[p]
+        for i in range(n):
+            ones = []
+            for j in range(M):
+                x = X[i, fs[j]]
+                if np.isnan(x):
+                    ok = bool(emiss[s + j])
+                else:
+                    ok = elo[s + j] <= x < ehi[s + j]
+                ones.append(1.0 if ok else 0.0)
+            pw = extend_pw(zs, ones)
+            for b in range(M):
+                mult = 0.5 * v * (ones[b] - zs[b])
+                if mult == 0.0:
+                    continue
+                pwb = unwind_pw(pw, zs[b], ones[b], M)
+                for a in range(M):
+                    if a == b:
+                        continue
+                    U = unwound_sum_pw(pwb, zs[a], ones[a], M - 1)
+                    w = mult * (ones[a] - zs[a]) * U
+                    out[i, fs[a], fs[b]] += w
+                    out[i, fs[b], fs[a]] += w
+    for i_ in range(C):
+        out[:, i_, i_] = base[:, i_] - (out[:, i_, :].sum(axis=-1)
+                                        - out[:, i_, i_])
+    assert np.allclose(out, ref, atol=1e-4), np.abs(out - ref).max()

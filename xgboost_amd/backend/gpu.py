@@ -739,6 +739,79 @@ def _shap_gpu_paths(booster, dmat, lo: int, hi: int,
     return res
 
 
+def shap_interactions_gpu(booster, dmat, lo: int, hi: int,
+                          iteration_range=(0, 0)) -> np.ndarray:
+    """GPU pred_interactions via the path-table decomposition
+    (shap_ix.hip; reference QuadratureShapInteractionTaskKernel,
+    src/predictor/interpretability/shap.cu:1068).
+
+    Per (row, path): extend once, unwind each element b, and accumulate
+    0.5*v*(one_b-z_b)*(one_a-z_a)*UnwoundSum_a over the reduced path —
+    the exact conditional-TreeSHAP pair terms (validated against the
+    CPU implementation by tests/test_shap.py).  Diagonal/bias cells are
+    completed from the contribution vector like the CPU path."""
+    from .. import ops as hip_ops
+    from ..shap import shap_values
+    from ..shap_paths import build_path_table
+    lib = hip_ops.load()
+    if not hasattr(lib, "gbt_shap_ix"):
+        raise ImportError("gbt_shap_ix kernel not built")
+    has_cat = any(
+        t.split_type[:t.n_nodes].any() for t in booster.trees[lo:hi]
+        if hasattr(t, "split_type"))
+    if has_cat:
+        raise ImportError("GPU SHAP interactions: categorical fallback")
+    n = dmat.num_row()
+    f = dmat.num_col()
+    n_groups = booster.n_outputs
+    C = f + 1
+    device = booster.device
+    pp, pg, ef, elo, ehi, emiss, ez, pv, bias = build_path_table(
+        booster.trees[lo:hi], booster.tree_info[lo:hi])
+    m = np.diff(pp)
+    if len(m) and int(m.max()) > 16:
+        raise ImportError("GPU SHAP interactions: path length > 16")
+    out_bytes = n * n_groups * C * C * 8
+    free, _total = torch.cuda.mem_get_info()
+    if out_bytes > free - (2 << 30):
+        raise ImportError("GPU SHAP interactions: output exceeds HBM")
+    rz = np.where(ez > 0, 1.0 / np.maximum(ez, 1e-300), 0.0)
+    t = {}
+    for name, arr in (("pp", pp), ("pg", pg), ("ef", ef), ("elo", elo),
+                      ("ehi", ehi), ("emiss", emiss), ("ez", ez),
+                      ("rz", rz), ("pv", pv)):
+        t[name] = torch.from_numpy(np.ascontiguousarray(arr)).to(device)
+    dd = dmat.device_data() if hasattr(dmat, "device_data") else None
+    X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
+    X = X.t().contiguous()  # [F, n] coalesced gathers
+    out = torch.zeros((n_groups, C, C, n), dtype=torch.float64,
+                      device=device)
+    missing = dmat.missing
+    missing_is_nan = 1 if np.isnan(missing) else 0
+    lib.gbt_shap_ix(
+        hip_ops.ptr(X), n, f, float(0.0 if missing_is_nan else missing),
+        missing_is_nan, hip_ops.ptr(t["pp"]), hip_ops.ptr(t["pg"]),
+        hip_ops.ptr(t["ef"]), hip_ops.ptr(t["elo"]), hip_ops.ptr(t["ehi"]),
+        hip_ops.ptr(t["emiss"]), hip_ops.ptr(t["ez"]), hip_ops.ptr(t["rz"]),
+        hip_ops.ptr(t["pv"]), len(pg), n_groups, C, hip_ops.ptr(out),
+        hip_ops.stream())
+    # diagonal completion from the (GPU) contribution vector
+    base = shap_values(booster, dmat, iteration_range)
+    base_t = torch.as_tensor(np.ascontiguousarray(base, np.float64),
+                             device=device)
+    if n_groups == 1:
+        base_t = base_t[:, None, :]
+    ix = out.permute(3, 0, 1, 2)  # [n, g, C, C] view
+    off_sum = ix.sum(dim=-1) - torch.diagonal(ix, dim1=-2, dim2=-1)
+    diag = base_t - off_sum
+    res = ix.clone()
+    res.diagonal(dim1=-2, dim2=-1).copy_(diag)
+    res = res.cpu().numpy()
+    if n_groups == 1:
+        return res[:, 0].astype(np.float32)
+    return res.astype(np.float32)
+
+
 def predict_margin_gpu(booster, dmat, out_margin: torch.Tensor,
                        lo: int, hi: int,
                        out_leaf: Optional[torch.Tensor] = None) -> torch.Tensor:

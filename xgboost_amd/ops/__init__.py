@@ -79,6 +79,8 @@ _SIGS = {
                        _p, _p, _p, _i64, _i, _i, _p, _p],
     "gbt_shap_paths16": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p,
                          _p, _p, _p, _i64, _i, _i, _p, _p],
+    "gbt_shap_ix": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
+                    _p, _i64, _i, _i, _p, _p],
 }
 
 

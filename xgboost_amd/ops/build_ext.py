@@ -9,8 +9,9 @@ SRC = os.path.join(HERE, "cpp")
 OUT = os.path.join(HERE, "libgbt_hip.so")
 
 SOURCES = ["hist.hip", "partition.hip", "evaluate.hip", "compress.hip",
-           "predict.hip", "shap.hip",
-    "shap_paths.hip", "driver.hip", "csr.hip", "gpair.hip", "mt_evaluate.hip", "cpu_hist.cpp"]
+           "predict.hip", "shap.hip", "shap_paths.hip", "shap_ix.hip",
+           "driver.hip", "csr.hip", "gpair.hip", "mt_evaluate.hip",
+           "cpu_hist.cpp"]
 
 
 def build(force: bool = False) -> str:

@@ -193,6 +193,13 @@ def _saabas(tree: RegTree, X: np.ndarray, phi: np.ndarray,
 
 def shap_interactions(booster, dmat, iteration_range=(0, 0)) -> np.ndarray:
     lo, hi = booster._tree_range(iteration_range)
+    if booster.device.type == "cuda":
+        try:
+            from .backend.gpu import shap_interactions_gpu
+            return shap_interactions_gpu(booster, dmat, lo, hi,
+                                         iteration_range)
+        except (ImportError, AttributeError):
+            pass  # categorical / deep-path forests: exact CPU fallback
     X = dmat.raw_data()
     n, f = X.shape
     n_groups = booster.n_outputs
