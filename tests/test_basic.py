@@ -215,3 +215,40 @@ def test_pred_leaf():
         tree = bst.trees[t]
         for leaf_id in np.unique(leaves[:, t]).astype(int):
             assert tree.is_leaf(leaf_id)
+
+
+def test_inplace_predict_zero_copy_proxy():
+    """inplace_predict wraps numpy/torch input in a proxy with no
+    DMatrix materialization (reference GBTree::InplacePredict via
+    ProxyDMatrix) and matches the DMatrix prediction exactly."""
+    import torch
+    rng = np.random.RandomState(3)
+    X = rng.randn(500, 6).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 4},
+                    xgb.DMatrix(X, label=y), 5, verbose_eval=False)
+    ref = bst.predict(xgb.DMatrix(X))
+    got_np = bst.inplace_predict(X)
+    assert np.allclose(got_np, ref, atol=1e-7)
+    got_t = bst.inplace_predict(torch.from_numpy(X))
+    assert np.allclose(got_t, ref, atol=1e-7)
+    # margin type
+    m_ref = bst.predict(xgb.DMatrix(X), output_margin=True)
+    m = bst.inplace_predict(X, predict_type="margin")
+    assert np.allclose(m, m_ref, atol=1e-6)
+    # base_margin honored
+    bm = np.full(500, 0.7, np.float32)
+    r1 = bst.predict(xgb.DMatrix(X, base_margin=bm))
+    r2 = bst.inplace_predict(X, base_margin=bm)
+    assert np.allclose(r1, r2, atol=1e-6)
+    # feature mismatch raises
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        bst.inplace_predict(X[:, :4])
+    # custom missing value
+    Xm = X.copy()
+    Xm[rng.rand(500, 6) < 0.1] = -999.0
+    Xn = np.where(Xm == -999.0, np.nan, Xm)
+    r3 = bst.inplace_predict(Xm, missing=-999.0)
+    r4 = bst.predict(xgb.DMatrix(Xn))
+    assert np.allclose(r3, r4, atol=1e-6)

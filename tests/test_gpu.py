@@ -686,3 +686,23 @@ def test_gpu_shap_interactions_matches_cpu():
     total = got.sum(axis=(1, 2))
     assert np.allclose(total, margin, atol=1e-3)
     del ref
+
+
+def test_gpu_inplace_predict_device_resident():
+    """inplace_predict on a cuda tensor: zero-copy proxy straight into
+    the HIP predict kernel, forest SoA cached across calls."""
+    X, y = _data(20000, 8, seed=41)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 6,
+                     "device": "cuda"}, d, 10, verbose_eval=False)
+    ref = bst.predict(xgb.DMatrix(X))
+    Xd = torch.from_numpy(X).cuda()
+    got = bst.inplace_predict(Xd)
+    assert np.allclose(got, ref, atol=1e-6)
+    # repeated call reuses the cached device forest (same object)
+    fc = bst.__dict__.get("_forest_dev_cache", {})
+    assert len(fc) == 1
+    fa0 = next(iter(fc.values()))
+    got2 = bst.inplace_predict(Xd)
+    assert next(iter(fc.values())) is fa0
+    assert np.allclose(got2, ref, atol=1e-6)

@@ -812,13 +812,26 @@ def shap_interactions_gpu(booster, dmat, lo: int, hi: int,
     return res.astype(np.float32)
 
 
+def _cached_forest(booster, lo: int, hi: int, device) -> _ForestArrays:
+    """Device forest SoA cached across predict calls (serving path:
+    repeated inplace_predict uploads nothing but the rows)."""
+    key = (lo, hi, len(booster.trees), str(device))
+    fc = booster.__dict__.setdefault("_forest_dev_cache", {})
+    fa = fc.get(key)
+    if fa is None:
+        fc.clear()  # model changed or different range: drop stale SoA
+        fa = _ForestArrays(booster, lo, hi, device)
+        fc[key] = fa
+    return fa
+
+
 def predict_margin_gpu(booster, dmat, out_margin: torch.Tensor,
                        lo: int, hi: int,
                        out_leaf: Optional[torch.Tensor] = None) -> torch.Tensor:
     from .. import ops as hip_ops
     lib = hip_ops.load()
     device = out_margin.device
-    fa = _ForestArrays(booster, lo, hi, device)
+    fa = _cached_forest(booster, lo, hi, device)
     dd = dmat.device_data() if hasattr(dmat, "device_data") else None
     X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
     n = dmat.num_row()

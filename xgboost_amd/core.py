@@ -736,11 +736,34 @@ class Booster:
                         predict_type: str = "value", missing: float = np.nan,
                         validate_features: bool = True, base_margin=None,
                         strict_shape: bool = False) -> np.ndarray:
-        d = DMatrix(data, missing=missing, base_margin=base_margin)
-        return self.predict(d, output_margin=(predict_type == "margin"),
-                            validate_features=validate_features,
-                            iteration_range=iteration_range,
-                            strict_shape=strict_shape)
+        """Predict straight from a user array with NO DMatrix
+        materialization (reference GBTree::InplacePredict via
+        ProxyDMatrix, src/gbm/gbtree.cc / src/data/proxy_dmatrix.h):
+        numpy / torch (cpu or cuda) / __cuda_array_interface__ inputs
+        are wrapped in a zero-copy proxy and fed directly to the
+        predictor kernels; forest device arrays are cached across calls
+        so repeated serving calls upload nothing but the rows."""
+        from .data import _ProxyMatrix
+        proxy = _ProxyMatrix.wrap(data, missing=missing,
+                                  base_margin=base_margin,
+                                  device=self.device)
+        if proxy is None:  # unsupported input shape: full DMatrix path
+            d = DMatrix(data, missing=missing, base_margin=base_margin)
+            return self.predict(d, output_margin=(predict_type == "margin"),
+                                validate_features=validate_features,
+                                iteration_range=iteration_range,
+                                strict_shape=strict_shape)
+        if validate_features and self.n_features is not None \
+                and proxy.num_col() != self.n_features:
+            raise ValueError(
+                f"feature mismatch: {proxy.num_col()} vs {self.n_features}")
+        margin = self._predict_margin(proxy, iteration_range)
+        res = margin if predict_type == "margin" \
+            else self.objective.pred_transform(margin)
+        arr = res.cpu().numpy()
+        if not strict_shape and arr.ndim == 2 and arr.shape[1] == 1:
+            arr = arr.reshape(-1)
+        return arr
 
     # -- evaluation ----------------------------------------------------
     def eval_set(self, evals: Sequence[Tuple[DMatrix, str]],

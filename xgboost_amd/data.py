@@ -646,3 +646,65 @@ def _dense_to_csr(X: np.ndarray, mask: np.ndarray):
     np.cumsum(mask.sum(axis=1), out=indptr[1:])
     rows, cols = np.nonzero(mask)
     return sp.csr_matrix((X[rows, cols], cols, indptr), shape=X.shape)
+
+
+class _ProxyMatrix:
+    """Zero-copy inplace-predict proxy (reference ProxyDMatrix,
+    src/data/proxy_dmatrix.h): wraps a user array without building a
+    DMatrix.  Quacks like DMatrix for the predictor paths only
+    (num_row/num_col/raw_data/device_data/missing/info)."""
+
+    def __init__(self):
+        self.info = MetaInfo()
+        self.missing = float("nan")
+        self._device_data: Optional[torch.Tensor] = None
+        self._host: Optional[np.ndarray] = None
+        self.feature_names = None
+        self.feature_types = None
+
+    @staticmethod
+    def wrap(data, missing, base_margin, device) -> Optional["_ProxyMatrix"]:
+        p = _ProxyMatrix()
+        p.missing = float(missing)
+        if isinstance(data, torch.Tensor):
+            if data.dim() != 2:
+                return None
+            t = data.to(torch.float32)
+            if not t.is_contiguous():
+                t = t.contiguous()
+            if t.is_cuda:
+                p._device_data = t
+            else:
+                p._host = t.numpy()
+        elif hasattr(data, "__cuda_array_interface__"):
+            t = torch.as_tensor(data, device=device)
+            if t.dim() != 2:
+                return None
+            t = t.to(torch.float32)
+            p._device_data = t if t.is_contiguous() else t.contiguous()
+        elif isinstance(data, np.ndarray):
+            if data.ndim != 2:
+                return None
+            p._host = np.ascontiguousarray(data, np.float32)
+        else:
+            return None  # DataFrame / CSR etc: full DMatrix path
+        n, c = (p._device_data.shape if p._device_data is not None
+                else p._host.shape)
+        p.info.num_row, p.info.num_col = int(n), int(c)
+        if base_margin is not None:
+            p.info.base_margin = np.asarray(base_margin, np.float32)
+        return p
+
+    def num_row(self) -> int:
+        return self.info.num_row
+
+    def num_col(self) -> int:
+        return self.info.num_col
+
+    def raw_data(self) -> np.ndarray:
+        if self._host is None and self._device_data is not None:
+            self._host = self._device_data.cpu().numpy()
+        return self._host
+
+    def device_data(self) -> Optional[torch.Tensor]:
+        return self._device_data
