@@ -408,8 +408,9 @@ class GpuOps(SegmentedOpsMixin):
         import ctypes
         if not hasattr(self.lib, "gbt_grow_tree"):
             return None
-        if param.grow_policy != "depthwise" or param.max_leaves > 0:
-            return None
+        # lossguide / max_leaves configs run the SAME depthwise chain:
+        # the grower replays the priority-queue policy on the recorded
+        # candidates afterwards (TreeGrower._policy_replay)
         if param.max_depth <= 0 or param.max_depth > 14:
             return None
         n_rows = qgpair.shape[0]

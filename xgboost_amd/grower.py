@@ -168,6 +168,12 @@ class TreeGrower:
                 or p.colsample_bytree < 1.0 or p.colsample_bylevel < 1.0
                 or p.colsample_bynode < 1.0):
             return None
+        needs_replay = p.grow_policy == "lossguide" or p.max_leaves > 0
+        if needs_replay:
+            if p.max_depth <= 0 or p.max_depth > 14:
+                return None  # unbounded-depth lossguide: python driver
+            if p.max_leaves > 0 and (1 << p.max_depth) > 16 * p.max_leaves:
+                return None  # the full depthwise chain would be wasteful
         # root sums stay on device: their readback piggybacks on the
         # driver's root-eval sync (one fewer host round-trip per tree);
         # the fused gpair path accumulates them inside QuantizeKernel
@@ -179,7 +185,101 @@ class TreeGrower:
                 collective.allreduce_sum_(rs)
         out = ops.grow_tree_native(qgpair, tree, p, self.quantizer,
                                    self.monotone, rs)
-        return out
+        if out is None or not needs_replay:
+            return out
+        dtree, pos = out
+        return self._policy_replay(dtree, pos, p)
+
+    def _policy_replay(self, dt: RegTree, pos: torch.Tensor,
+                       param) -> Tuple[RegTree, torch.Tensor]:
+        """Re-drive the grow-policy priority queue over the native
+        driver's depthwise expansion records (reference Driver,
+        src/tree/driver.h:17-51).
+
+        The depthwise chain expands EVERY gain-valid node, a superset
+        of what lossguide / max_leaves would expand (a node's candidate
+        split depends only on its own histogram, never on which other
+        nodes expanded), so the policy can be replayed on the host: pop
+        candidates by (-gain, seq) — or by (depth, seq) for depthwise
+        with a leaf cap — renumber nodes in pop order exactly like the
+        Python driver, truncate at max_leaves, and remap row positions
+        from pruned subtrees onto their covering leaf."""
+        import heapq
+        n = dt.n_nodes
+        internal = dt.left[:n] >= 0
+        nt = RegTree(dt.n_features)
+        nt._ensure(n)
+        new_of = np.full(n, -1, np.int32)
+        new_of[0] = 0
+        nt.base_weight[0] = dt.base_weight[0]
+        nt.sum_hess[0] = dt.sum_hess[0]
+        nt.left[0] = nt.right[0] = nt.parent[0] = -1
+        state = {"n_new": 1, "n_leaves": 1, "seq": 0}
+        heap: List[Tuple[float, int, int, int]] = []
+
+        def push(dnid: int, depth: int) -> None:
+            if not internal[dnid]:
+                return
+            if param.max_depth > 0 and depth >= param.max_depth:
+                return
+            if param.max_leaves > 0 and state["n_leaves"] >= param.max_leaves:
+                return
+            key = (float(depth) if param.grow_policy == "depthwise"
+                   else -float(dt.loss_chg[dnid]))
+            heapq.heappush(heap, (key, state["seq"], dnid, depth))
+            state["seq"] += 1
+
+        push(0, 0)
+        while heap:
+            batch = [heapq.heappop(heap)]
+            if param.grow_policy == "depthwise":
+                while heap and heap[0][0] == batch[0][0]:
+                    batch.append(heapq.heappop(heap))
+            if param.max_leaves > 0:
+                batch = batch[:max(0, param.max_leaves - state["n_leaves"])]
+            if not batch:
+                continue
+            for _key, _sq, dnid, _depth in batch:
+                nn = int(new_of[dnid])
+                l, r = state["n_new"], state["n_new"] + 1
+                state["n_new"] += 2
+                dl, dr = int(dt.left[dnid]), int(dt.right[dnid])
+                new_of[dl], new_of[dr] = l, r
+                nt.left[nn], nt.right[nn] = l, r
+                nt.parent[l] = nt.parent[r] = nn
+                nt.split_index[nn] = dt.split_index[dnid]
+                nt.split_cond[nn] = dt.split_cond[dnid]
+                nt.default_left[nn] = dt.default_left[dnid]
+                nt.loss_chg[nn] = dt.loss_chg[dnid]
+                nt.sum_hess[nn] = dt.sum_hess[dnid]
+                for c_new, c_d in ((l, dl), (r, dr)):
+                    nt.base_weight[c_new] = dt.base_weight[c_d]
+                    nt.sum_hess[c_new] = dt.sum_hess[c_d]
+                    nt.left[c_new] = nt.right[c_new] = -1
+                state["n_leaves"] += 1
+            for _key, _sq, dnid, depth in batch:
+                push(int(dt.left[dnid]), depth + 1)
+                push(int(dt.right[dnid]), depth + 1)
+        nt.n_nodes = state["n_new"]
+        for i in range(nt.n_nodes):
+            if nt.left[i] == -1:
+                nt.split_cond[i] = np.float32(
+                    float(nt.base_weight[i]) * param.eta)
+        # remap positions: every depthwise leaf -> its covering new leaf
+        lut = np.zeros(n, np.int32)
+        stack = [(0, -1)]
+        while stack:
+            dnid, cover = stack.pop()
+            if cover < 0:
+                nn = int(new_of[dnid])
+                if nn >= 0 and nt.left[nn] == -1:
+                    cover = nn
+            lut[dnid] = cover if cover >= 0 else 0
+            if internal[dnid]:
+                stack.append((int(dt.left[dnid]), cover))
+                stack.append((int(dt.right[dnid]), cover))
+        lut_t = torch.from_numpy(lut).to(pos.device)
+        return nt, lut_t[pos.long()].to(torch.int32)
 
     def _grow(self, qgpair: torch.Tensor, tree: RegTree
               ) -> Tuple[RegTree, torch.Tensor]:
