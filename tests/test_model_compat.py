@@ -1,0 +1,180 @@
+"""Cross-implementation model-format evidence.
+
+The reference cannot be compiled in this image (its dmlc-core submodule
+is empty and there is no network), so instead of a binary round-trip we
+pin format compatibility two ways:
+
+1. A model JSON authored BY HAND to the reference WRITER's layout
+   (src/learner.cc:868-899 + tree_model.cc:1084-1208, including the
+   extra fields the reference emits that our own writer does not:
+   objective reg_loss_param, categorical segment arrays, scientific
+   base_score) must load and predict exactly what the trees encode.
+2. Our dump_model(dump_format="json") output must satisfy the
+   reference-authored JSON schema doc/dump.schema (required fields
+   nodeid/depth/yes/no/split/children for splits, nodeid/leaf for
+   leaves, <= 2 children).
+"""
+import json
+import math
+
+import numpy as np
+import pytest
+
+import xgboost_amd as xgb
+
+# A reference-layout model: binary:logistic, 2 features, two trees —
+# one depth-1 split on f0 at 0.5 (missing goes left), one leaf-only
+# stump.  Field order/types follow the reference writer, including
+# string-encoded scalars and the categorical segment arrays (empty).
+REFERENCE_STYLE_MODEL = {
+    "learner": {
+        "attributes": {"best_iteration": "1"},
+        "feature_names": ["f0", "f1"],
+        "feature_types": ["float", "float"],
+        "gradient_booster": {
+            "model": {
+                "gbtree_model_param": {
+                    "num_trees": "2",
+                    "num_parallel_tree": "1",
+                },
+                "iteration_indptr": [0, 1, 2],
+                "tree_info": [0, 0],
+                "trees": [
+                    {
+                        "tree_param": {
+                            "num_feature": "2",
+                            "num_nodes": "3",
+                            "size_leaf_vector": "1",
+                        },
+                        "id": 0,
+                        "loss_changes": [10.5, 0.0, 0.0],
+                        "sum_hessian": [8.0, 3.0, 5.0],
+                        "base_weights": [0.0, -0.4, 0.6],
+                        "split_indices": [0, 0, 0],
+                        "split_conditions": [0.5, -0.2, 0.3],
+                        "default_left": [1, 0, 0],
+                        "left_children": [1, -1, -1],
+                        "right_children": [2, -1, -1],
+                        "parents": [2147483647, 0, 0],
+                        "split_type": [0, 0, 0],
+                        "categories": [],
+                        "categories_nodes": [],
+                        "categories_segments": [],
+                        "categories_sizes": [],
+                    },
+                    {
+                        "tree_param": {
+                            "num_feature": "2",
+                            "num_nodes": "1",
+                            "size_leaf_vector": "1",
+                        },
+                        "id": 1,
+                        "loss_changes": [0.0],
+                        "sum_hessian": [8.0],
+                        "base_weights": [0.05],
+                        "split_indices": [0],
+                        "split_conditions": [0.05],
+                        "default_left": [0],
+                        "left_children": [-1],
+                        "right_children": [-1],
+                        "parents": [2147483647],
+                        "split_type": [0],
+                        "categories": [],
+                        "categories_nodes": [],
+                        "categories_segments": [],
+                        "categories_sizes": [],
+                    },
+                ],
+            },
+            "name": "gbtree",
+        },
+        "learner_model_param": {
+            "base_score": "5E-1",
+            "boost_from_average": "1",
+            "num_class": "0",
+            "num_feature": "2",
+            "num_target": "1",
+        },
+        "objective": {
+            "name": "binary:logistic",
+            "reg_loss_param": {"scale_pos_weight": "1"},
+        },
+    },
+    "version": [3, 5, 0],
+}
+
+
+def test_load_reference_authored_json(tmp_path):
+    path = str(tmp_path / "ref_model.json")
+    with open(path, "w") as fh:
+        json.dump(REFERENCE_STYLE_MODEL, fh)
+    bst = xgb.Booster(model_file=path)
+    assert bst.num_features() == 2
+    assert bst.num_boosted_rounds() == 2
+    X = np.array([[0.0, 9.9], [1.0, -1.0], [np.nan, 0.0]], np.float32)
+    pred = bst.predict(xgb.DMatrix(X))
+    base = math.log(0.5 / 0.5)  # logit of base_score 5E-1
+    # row 0: f0 < 0.5 -> left leaf -0.2; stump +0.05
+    # row 1: f0 >= 0.5 -> right leaf 0.3; stump +0.05
+    # row 2: missing, default_left=1 -> left leaf -0.2; stump +0.05
+    exp = [1 / (1 + math.exp(-(base + m + 0.05))) for m in (-0.2, 0.3, -0.2)]
+    assert np.allclose(pred, exp, atol=1e-6), (pred, exp)
+    # attributes survive
+    assert bst.attributes().get("best_iteration") == "1"
+    # re-save keeps the tree content (round-trip through our writer)
+    raw = bytes(bst.save_raw("json"))
+    bst2 = xgb.Booster()
+    bst2.load_model(bytearray(raw))
+    assert np.allclose(bst2.predict(xgb.DMatrix(X)), pred, atol=1e-7)
+
+
+def test_load_reference_authored_categorical_json(tmp_path):
+    """Categorical split encoded the reference way: categories_nodes /
+    segments / sizes arrays with the category list that goes RIGHT."""
+    model = json.loads(json.dumps(REFERENCE_STYLE_MODEL))
+    t0 = model["learner"]["gradient_booster"]["model"]["trees"][0]
+    t0["split_type"] = [1, 0, 0]
+    t0["categories"] = [1, 3]        # cats {1,3} go right
+    t0["categories_nodes"] = [0]
+    t0["categories_segments"] = [0]
+    t0["categories_sizes"] = [2]
+    t0["split_conditions"] = [0.0, -0.2, 0.3]
+    model["learner"]["feature_types"] = ["c", "float"]
+    path = str(tmp_path / "ref_cat.json")
+    with open(path, "w") as fh:
+        json.dump(model, fh)
+    bst = xgb.Booster(model_file=path)
+    X = np.array([[1.0, 0.0], [2.0, 0.0], [3.0, 0.0]], np.float32)
+    pred = bst.predict(xgb.DMatrix(X, feature_types=["c", "float"]))
+    base = 0.0
+    exp = [1 / (1 + math.exp(-(base + m + 0.05)))
+           for m in (0.3, -0.2, 0.3)]  # cat 1 -> right, 2 -> left, 3 -> right
+    assert np.allclose(pred, exp, atol=1e-6), (pred, exp)
+
+
+def _validate_dump_node(node, depth=0):
+    """Recursive check against the reference's doc/dump.schema: split
+    nodes require nodeid/depth/yes/no/split/children (<= 2 children),
+    leaves require nodeid/leaf."""
+    assert isinstance(node["nodeid"], int) and node["nodeid"] >= 0
+    if "leaf" in node:
+        assert isinstance(node["leaf"], float)
+        return
+    for key in ("depth", "yes", "no", "split", "children"):
+        assert key in node, f"missing {key} in split node"
+    assert isinstance(node["split"], str)
+    assert isinstance(node["children"], list) and len(node["children"]) <= 2
+    for ch in node["children"]:
+        _validate_dump_node(ch, depth + 1)
+
+
+def test_dump_model_matches_reference_schema(tmp_path):
+    rng = np.random.RandomState(0)
+    X = rng.randn(500, 5).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 0).astype(np.float32)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 4},
+                    xgb.DMatrix(X, label=y), 3, verbose_eval=False)
+    dumps = bst.get_dump(dump_format="json")
+    assert len(dumps) == 3
+    for d in dumps:
+        _validate_dump_node(json.loads(d))
