@@ -107,3 +107,44 @@ def test_extmem_gpu_training(batched_data):
                "device": "cuda", "max_bin": 64}, d_ext, 10,
               evals=[(d_ext, "t")], evals_result=res, verbose_eval=False)
     assert res["t"]["logloss"][-1] < 0.45
+
+
+class CachedBatchIter(NumpyBatchIter):
+    def __init__(self, Xs, ys, cache_prefix):
+        super().__init__(Xs, ys)
+        self.cache_prefix = cache_prefix
+
+
+def test_extmem_disk_spill_matches_incore(batched_data, tmp_path):
+    """Pages beyond the host budget spill to <cache_prefix>.pageN.bin
+    and stream back through the 2-deep read-ahead ring (reference
+    sparse_page_source.h disk cache): training must match the all-RAM
+    external-memory path exactly."""
+    X, y, Xs, ys = batched_data
+    prefix = str(tmp_path / "cache")
+    # budget fits only the first page -> pages 1..3 go to disk
+    page_bytes = Xs[0].shape[0] * Xs[0].shape[1]  # u8 bins
+    d_disk = ExtMemQuantileDMatrix(
+        CachedBatchIter(Xs, ys, prefix), max_bin=64,
+        max_host_cache_bytes=page_bytes + 1)
+    assert sum(d_disk.store.is_disk(i) for i in range(4)) == 3
+    import os
+    assert os.path.exists(prefix + ".page1.bin")
+    d_ram = ExtMemQuantileDMatrix(NumpyBatchIter(Xs, ys), max_bin=64)
+    params = {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3}
+    b1 = xgb.train(params, d_disk, 5, verbose_eval=False)
+    b2 = xgb.train(params, d_ram, 5, verbose_eval=False)
+    assert bytes(b1.save_raw("json")) == bytes(b2.save_raw("json"))
+    # predict traverses disk pages through page_qm
+    p1 = b1.predict(d_disk)
+    p2 = b2.predict(d_ram)
+    assert np.allclose(p1, p2, atol=1e-7)
+
+
+def test_extmem_disk_spill_requires_prefix(batched_data):
+    """Without a cache_prefix the budget is ignored (pages stay in
+    host RAM) — spilling must be an explicit opt-in."""
+    X, y, Xs, ys = batched_data
+    d = ExtMemQuantileDMatrix(NumpyBatchIter(Xs, ys), max_bin=64,
+                              max_host_cache_bytes=1)
+    assert not any(d.store.is_disk(i) for i in range(4))

@@ -136,6 +136,13 @@ class CpuOps(SegmentedOpsMixin):
         self._cut_ptrs_np = np.ascontiguousarray(qm.cuts.ptrs, np.int32)
         self._scratch = None
 
+    def swap_gidx(self, gidx: torch.Tensor) -> None:
+        """Point the kernels at a different (streamed-in) quantized page
+        with the same cuts (external-memory disk path)."""
+        self.qm = QuantizedMatrix(gidx, self.qm.cuts, self.qm.has_missing)
+        self.gidx_global = self.qm.global_gidx()
+        self._gidx_np = np.ascontiguousarray(gidx.numpy())
+
     def _gidx_c_ptrs(self):
         import ctypes
         p = self._gidx_np.ctypes.data_as(ctypes.c_void_p)

@@ -678,7 +678,8 @@ class Booster:
     def _predict_margin_extmem(self, dmat, out: torch.Tensor,
                                lo: int, hi: int) -> torch.Tensor:
         """Bin-based traversal per quantized page (no raw values)."""
-        for pi, qm in enumerate(dmat.pages):
+        for pi in range(len(dmat.pages)):
+            qm = dmat.page_qm(pi)  # loads disk-spilled pages on demand
             s, e = dmat.page_offsets[pi], dmat.page_offsets[pi + 1]
             gg = qm.global_gidx().cpu().numpy()
             for t in range(lo, hi):
