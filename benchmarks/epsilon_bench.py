@@ -22,7 +22,7 @@ from xgboost_amd import collective  # noqa: E402
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--rows", type=int, default=2_000_000)
+    ap.add_argument("--rows", type=int, default=6_250_000)  # 5e7 / 8 GPUs
     ap.add_argument("--features", type=int, default=2000)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
@@ -44,10 +44,18 @@ def main():
     gen = torch.Generator(device="cuda" if has_gpu else "cpu")
     gen.manual_seed(99 + rank)
     dev = "cuda" if has_gpu else "cpu"
-    X = torch.randn(n, f, generator=gen, device=dev)
+    # 6.25e6 x 2000 f32 = 50 GB generated and kept ON DEVICE (288 GB
+    # HBM); labels computed in chunks to bound the matmul workspace
+    X = torch.empty(n, f, device=dev)
     w = torch.randn(f, generator=gen, device=dev) / np.sqrt(f)
-    y = ((X @ w) + 0.3 * torch.randn(n, generator=gen, device=dev)
-         > 0).float().cpu().numpy()
+    y = torch.empty(n, device=dev)
+    chunk = 1_000_000
+    for s0 in range(0, n, chunk):
+        e0 = min(s0 + chunk, n)
+        X[s0:e0] = torch.randn(e0 - s0, f, generator=gen, device=dev)
+        y[s0:e0] = (X[s0:e0] @ w
+                    + 0.3 * torch.randn(e0 - s0, generator=gen, device=dev))
+    y = (y > 0).float().cpu().numpy()
     d = xgb.DMatrix(X if has_gpu else X.cpu().numpy(), label=y)
     bst = xgb.Booster({"objective": "binary:logistic",
                        "max_depth": args.max_depth, "max_bin": args.max_bin,

@@ -359,7 +359,20 @@ def device_cuts(X: torch.Tensor, max_bin: int,
                 missing: float = float("nan"),
                 weights: Optional[torch.Tensor] = None,
                 feature_types: Optional[List[str]] = None) -> HistogramCuts:
-    """One-shot device sketch (handles distributed merge internally)."""
-    sk = DeviceSketch(X.shape[1], max_bin, feature_types)
-    sk.push(X, weights=weights, missing=missing)
+    """One-shot device sketch (handles distributed merge internally).
+
+    Very large inputs are pushed in row batches so the flat segmented
+    index tensors stay bounded (~2 GB); the documented two-prune error
+    bound covers exactly this batched path."""
+    n, f = X.shape
+    sk = DeviceSketch(f, max_bin, feature_types)
+    max_elems = 1 << 28
+    batch_rows = max(1, max_elems // max(f, 1))
+    if n <= batch_rows:
+        sk.push(X, weights=weights, missing=missing)
+    else:
+        for s0 in range(0, n, batch_rows):
+            e0 = min(s0 + batch_rows, n)
+            w = weights[s0:e0] if weights is not None else None
+            sk.push(X[s0:e0], weights=w, missing=missing)
     return sk.make_cuts()
