@@ -476,7 +476,7 @@ class GpuOps(SegmentedOpsMixin):
         host = {name: np.zeros(cap, dt) for name, dt in [
             ("left", np.int32), ("right", np.int32), ("parent", np.int32),
             ("split_index", np.int32), ("split_cond", np.float32),
-            ("default_left", np.uint8), ("loss_chg", np.float32),
+            ("default_left", np.uint8), ("loss_chg", np.float64),
             ("sum_hess", np.float32), ("base_weight", np.float32)]}
         mono_dev = mono_host = None
         if monotone is not None:
@@ -568,6 +568,9 @@ class GpuOps(SegmentedOpsMixin):
         tree.split_cond[:n] = host["split_cond"][:n]
         tree.default_left[:n] = host["default_left"][:n]
         tree.loss_chg[:n] = host["loss_chg"][:n]
+        # exact fp64 gains: the grow-policy replay's heap keys must
+        # match the Python driver's fp64 ordering bit-for-bit
+        tree._gain64 = host["loss_chg"][:n].copy()
         tree.sum_hess[:n] = host["sum_hess"][:n]
         tree.base_weight[:n] = host["base_weight"][:n]
         return tree, ws["pos"]

@@ -645,6 +645,8 @@ class Booster:
                                lo: int, hi: int) -> torch.Tensor:
         """Sparse predict: densify ONLY the features used by the trees,
         absent entries become NaN (missing -> default direction)."""
+        if hi <= lo:
+            return out  # no trees yet (first margin of training)
         csr = dmat.sparse_data()
         n = csr.shape[0]
         used = sorted(set(

@@ -207,6 +207,9 @@ class TreeGrower:
         import heapq
         n = dt.n_nodes
         internal = dt.left[:n] >= 0
+        gains = getattr(dt, "_gain64", None)
+        if gains is None:
+            gains = dt.loss_chg[:n].astype(np.float64)
         nt = RegTree(dt.n_features)
         nt._ensure(n)
         new_of = np.full(n, -1, np.int32)
@@ -225,7 +228,7 @@ class TreeGrower:
             if param.max_leaves > 0 and state["n_leaves"] >= param.max_leaves:
                 return
             key = (float(depth) if param.grow_policy == "depthwise"
-                   else -float(dt.loss_chg[dnid]))
+                   else -float(gains[dnid]))
             heapq.heappush(heap, (key, state["seq"], dnid, depth))
             state["seq"] += 1
 

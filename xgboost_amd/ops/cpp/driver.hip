@@ -546,7 +546,10 @@ int gbt_grow_tree(
     // host tree outputs (caller-sized to 2^(max_depth+1))
     int32_t* out_left, int32_t* out_right, int32_t* out_parent,
     int32_t* out_split_index, float* out_split_cond,
-    uint8_t* out_default_left, float* out_loss_chg, float* out_sum_hess,
+    uint8_t* out_default_left,
+    double* out_loss_chg,  // fp64: the grow-policy replay orders its
+                           // priority queue by these exact gains
+    float* out_sum_hess,
     float* out_base_weight, void* stream_v) {
   DriverCtx* ctx = (DriverCtx*)vctx;
   hipStream_t stream = (hipStream_t)stream_v;
@@ -960,7 +963,7 @@ int gbt_grow_tree(
         out_split_index[nd->nid] = nd->feature;
         out_split_cond[nd->nid] = cut_values_host[nd->bin];
         out_default_left[nd->nid] = (uint8_t)nd->dir;
-        out_loss_chg[nd->nid] = (float)nd->gain;
+        out_loss_chg[nd->nid] = nd->gain;
         const long long rgq = nd->gq - nd->lgq, rhq = nd->hq - nd->lhq;
         double wl = CalcWeight(nd->lgq * inv_g, nd->lhq * inv_h, p);
         double wr = CalcWeight(rgq * inv_g, rhq * inv_h, p);
@@ -1143,7 +1146,7 @@ int gbt_grow_tree(
       out_split_index[nd->nid] = nd->feature;
       out_split_cond[nd->nid] = cut_values_host[nd->bin];
       out_default_left[nd->nid] = (uint8_t)nd->dir;
-      out_loss_chg[nd->nid] = (float)nd->gain;
+      out_loss_chg[nd->nid] = nd->gain;
       const long long rgq = nd->gq - nd->lgq, rhq = nd->hq - nd->lhq;
       double wl = CalcWeight(nd->lgq * inv_g, nd->lhq * inv_h, p);
       double wr = CalcWeight(rgq * inv_g, rhq * inv_h, p);
