@@ -708,6 +708,72 @@ def test_gpu_inplace_predict_device_resident():
     assert np.allclose(got2, ref, atol=1e-6)
 
 
+def test_gpu_lossguide_native_replay_matches_python_gpu_driver():
+    """lossguide / max_leaves on the native driver: depthwise chain +
+    host policy replay must match the PYTHON GPU lossguide driver
+    node-for-node.  (The GPU python driver is the right oracle: its
+    heap keys are the same fp64 gains the native records carry.  A
+    CPU-trained tree is NOT node-comparable for lossguide: CPU and GPU
+    gains agree only to ~1e-12 relative, and the lossguide POP ORDER
+    is sensitive to last-ulp differences — depthwise expansion is not.)
+    """
+    from xgboost_amd.grower import TreeGrower
+    X, y = _data(30000, 8, seed=19)
+    for params in (
+        {"grow_policy": "lossguide", "max_depth": 8, "max_leaves": 31},
+        {"grow_policy": "lossguide", "max_depth": 6, "max_leaves": 0},
+        {"grow_policy": "depthwise", "max_depth": 7, "max_leaves": 40},
+        {"grow_policy": "lossguide", "max_depth": 8, "max_leaves": 31,
+         "monotone_constraints": [1, 0, 0, 0, -1, 0, 0, 0]},
+    ):
+        pd = {"objective": "binary:logistic", "eta": 0.3, "max_bin": 128,
+              "device": "cuda", **params}
+        bg = xgb.train(pd, xgb.DMatrix(X, label=y), 5)  # native + replay
+        orig = TreeGrower._try_native
+        TreeGrower._try_native = lambda self, qg, t: None
+        try:
+            bp = xgb.train(pd, xgb.DMatrix(X, label=y), 5)  # python driver
+        finally:
+            TreeGrower._try_native = orig
+        for tg, tc in zip(bg.trees, bp.trees):
+            assert tg.n_nodes == tc.n_nodes, params
+            assert np.array_equal(tg.split_index[:tg.n_nodes],
+                                  tc.split_index[:tc.n_nodes]), params
+            assert np.array_equal(tg.left[:tg.n_nodes],
+                                  tc.left[:tc.n_nodes]), params
+            assert np.allclose(tg.split_cond[:tg.n_nodes],
+                               tc.split_cond[:tc.n_nodes],
+                               rtol=1e-6, atol=1e-7), params
+        # and the CPU oracle agrees on QUALITY (training accuracy)
+        bc = xgb.train({k: v for k, v in pd.items() if k != "device"},
+                       xgb.DMatrix(X, label=y), 5)
+        pg = bg.predict(xgb.DMatrix(X))
+        pc = bc.predict(xgb.DMatrix(X))
+        acc_g = (((pg > 0.5) == y).mean())
+        acc_c = (((pc > 0.5) == y).mean())
+        assert abs(acc_g - acc_c) < 0.02, (params, acc_g, acc_c)
+
+
+def test_gpu_inplace_predict_device_resident():
+    """inplace_predict on a cuda tensor: zero-copy proxy straight into
+    the HIP predict kernel, forest SoA cached across calls."""
+    X, y = _data(20000, 8, seed=41)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 6,
+                     "device": "cuda"}, d, 10, verbose_eval=False)
+    ref = bst.predict(xgb.DMatrix(X))
+    Xd = torch.from_numpy(X).cuda()
+    got = bst.inplace_predict(Xd)
+    assert np.allclose(got, ref, atol=1e-6)
+    # repeated call reuses the cached device forest (same object)
+    fc = bst.__dict__.get("_forest_dev_cache", {})
+    assert len(fc) == 1
+    fa0 = next(iter(fc.values()))
+    got2 = bst.inplace_predict(Xd)
+    assert next(iter(fc.values())) is fa0
+    assert np.allclose(got2, ref, atol=1e-6)
+
+
 def test_gpu_lossguide_native_replay_matches_cpu():
     """lossguide / max_leaves on the native driver: depthwise chain +
     host policy replay must match the Python lossguide driver's trees
