@@ -772,29 +772,3 @@ def test_gpu_inplace_predict_device_resident():
     got2 = bst.inplace_predict(Xd)
     assert next(iter(fc.values())) is fa0
     assert np.allclose(got2, ref, atol=1e-6)
-
-
-def test_gpu_lossguide_native_replay_matches_cpu():
-    """lossguide / max_leaves on the native driver: depthwise chain +
-    host policy replay must match the Python lossguide driver's trees
-    node-for-node."""
-    X, y = _data(30000, 8, seed=19)
-    for params in (
-        {"grow_policy": "lossguide", "max_depth": 8, "max_leaves": 31},
-        {"grow_policy": "lossguide", "max_depth": 6, "max_leaves": 0},
-        {"grow_policy": "depthwise", "max_depth": 7, "max_leaves": 40},
-        {"grow_policy": "lossguide", "max_depth": 8, "max_leaves": 31,
-         "monotone_constraints": [1, 0, 0, 0, -1, 0, 0, 0]},
-    ):
-        pd = {"objective": "binary:logistic", "eta": 0.3, "max_bin": 128,
-              **params}
-        bc = xgb.train(pd, xgb.DMatrix(X, label=y), 5)
-        bg = xgb.train({**pd, "device": "cuda"}, xgb.DMatrix(X, label=y), 5)
-        for tg, tc in zip(bg.trees, bc.trees):
-            assert tg.n_nodes == tc.n_nodes, params
-            assert np.array_equal(tg.split_index[:tg.n_nodes],
-                                  tc.split_index[:tc.n_nodes]), params
-            assert np.array_equal(tg.left[:tg.n_nodes],
-                                  tc.left[:tc.n_nodes]), params
-            assert np.allclose(tg.split_cond[:tg.n_nodes],
-                               tc.split_cond[:tc.n_nodes], rtol=1e-6), params
