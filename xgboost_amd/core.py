@@ -777,15 +777,20 @@ class Booster:
         for dmat, name in evals:
             margin = self._cached_margin(dmat)
             transformed = self.objective.eval_transform(margin)
-            tnp = transformed.cpu().numpy()
-            if tnp.ndim == 2 and tnp.shape[1] == 1:
-                tnp = tnp.reshape(-1)
+            # metrics stay device-resident when the margin is on GPU
+            # (reference device AUC / elementwise metrics, auc.cu:168);
+            # torch-unaware metrics fall back to host inside the wrapper
+            tv = transformed.detach()
+            if tv.dim() == 2 and tv.shape[1] == 1:
+                tv = tv.reshape(-1)
+            tnp = None
             for mname in metric_names:
                 m = create_metric(mname)
-                val = m(tnp, dmat.info)
+                val = m(tv, dmat.info)
                 parts.append(f"{name}-{mname}:{val:.5f}" if abs(val) >= 1e-5
                              else f"{name}-{mname}:{val:g}")
             if feval is not None:
+                tnp = tv.cpu().numpy()
                 res = feval(tnp, dmat)
                 if isinstance(res, list):
                     for mn, v in res:

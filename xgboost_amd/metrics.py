@@ -41,13 +41,42 @@ def create_metric(name: str):
     fn = _REGISTRY[base]
 
     def call(preds, info):
+        import torch as _torch
+        if isinstance(preds, _torch.Tensor) and \
+                not getattr(fn, "_torch_ok", False):
+            preds = preds.detach().cpu().numpy()
         return fn(preds, info, param)
 
     call.metric_name = name
     return call
 
 
-def _yw(preds: np.ndarray, info) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+def _torch_ok(fn):
+    """Marks a metric as device-resident capable: it accepts torch
+    tensors (CPU or HIP) and reduces on the tensor's device, syncing
+    one scalar pair to the host (reference device AUC / elementwise
+    metrics, src/metric/auc.cu:168, elementwise_metric.cu)."""
+    fn._torch_ok = True
+    return fn
+
+
+def _yw(preds, info):
+    """(labels, preds, weights) as matching fp64 arrays — torch tensors
+    on preds.device when preds is a tensor, else numpy."""
+    import torch as _torch
+    if isinstance(preds, _torch.Tensor):
+        dev = preds.device
+        p = preds.detach().double()
+        if p.dim() == 1:
+            p = p.view(-1, 1)
+        y = _torch.as_tensor(np.asarray(info.labels, np.float64),
+                             device=dev).view(p.shape[0], -1)
+        if y.shape[1] == 1 and p.shape[1] > 1:
+            y = y.expand(p.shape)
+        w = (_torch.as_tensor(np.asarray(info.weights, np.float64),
+                              device=dev) if info.weights is not None
+             else _torch.ones(p.shape[0], dtype=_torch.float64, device=dev))
+        return y, p, w
     y = np.asarray(info.labels, dtype=np.float64).reshape(preds.shape[0], -1)
     p = np.asarray(preds, dtype=np.float64)
     if p.ndim == 1:
@@ -64,14 +93,16 @@ def _ratio(num: float, den: float) -> float:
     return num / max(den, 1e-16)
 
 
-def _wmean(err: np.ndarray, w: np.ndarray) -> float:
-    # err is [n, k]; weight applies per row
+def _wmean(err, w) -> float:
+    # err is [n, k]; weight applies per row (torch or numpy — ONE host
+    # sync for the two scalars when err lives on the GPU)
     num = float((err * w[:, None]).sum())
-    den = float(w.sum() * err.shape[1])
+    den = float(w.sum()) * err.shape[1]
     return _ratio(num, den)
 
 
 @register("rmse")
+@_torch_ok
 def rmse(preds, info, param=None):
     y, p, w = _yw(preds, info)
     return math.sqrt(_wmean((p - y) ** 2, w))
@@ -84,9 +115,10 @@ def rmsle(preds, info, param=None):
 
 
 @register("mae")
+@_torch_ok
 def mae(preds, info, param=None):
     y, p, w = _yw(preds, info)
-    return _wmean(np.abs(p - y), w)
+    return _wmean(abs(p - y), w)
 
 
 @register("mape")
@@ -104,24 +136,48 @@ def mphe(preds, info, param=None):
 
 
 @register("logloss")
+@_torch_ok
 def logloss(preds, info, param=None):
+    import torch as _torch
     y, p, w = _yw(preds, info)
     eps = 1e-16
-    p = np.clip(p, eps, 1 - eps)
-    ll = -(y * np.log(p) + (1 - y) * np.log(1 - p))
+    if isinstance(p, _torch.Tensor):
+        p = p.clamp(eps, 1 - eps)
+        ll = -(y * _torch.log(p) + (1 - y) * _torch.log(1 - p))
+    else:
+        p = np.clip(p, eps, 1 - eps)
+        ll = -(y * np.log(p) + (1 - y) * np.log(1 - p))
     return _wmean(ll, w)
 
 
 @register("error")
+@_torch_ok
 def error(preds, info, param=None):
+    import torch as _torch
     t = float(param) if param else 0.5
     y, p, w = _yw(preds, info)
-    wrong = np.where(p > t, y != 1.0, y != 0.0).astype(np.float64)
+    if isinstance(p, _torch.Tensor):
+        wrong = _torch.where(p > t, y != 1.0, y != 0.0).double()
+    else:
+        wrong = np.where(p > t, y != 1.0, y != 0.0).astype(np.float64)
     return _wmean(wrong, w)
 
 
 @register("merror")
+@_torch_ok
 def merror(preds, info, param=None):
+    import torch as _torch
+    if isinstance(preds, _torch.Tensor):
+        dev = preds.device
+        y = _torch.as_tensor(np.asarray(info.labels, np.int64),
+                             device=dev).view(-1)
+        p = preds.detach().double()
+        cls = p.argmax(dim=1) if p.dim() == 2 and p.shape[1] > 1 \
+            else p.view(-1)
+        w = (_torch.as_tensor(np.asarray(info.weights, np.float64),
+                              device=dev) if info.weights is not None
+             else _torch.ones(y.shape[0], dtype=_torch.float64, device=dev))
+        return _ratio(float(((cls != y) * w).sum()), float(w.sum()))
     y = np.asarray(info.labels, dtype=np.int64).reshape(-1)
     p = np.asarray(preds, dtype=np.float64)
     cls = p.argmax(axis=1) if p.ndim == 2 and p.shape[1] > 1 else p.reshape(-1)
@@ -131,7 +187,19 @@ def merror(preds, info, param=None):
 
 
 @register("mlogloss")
+@_torch_ok
 def mlogloss(preds, info, param=None):
+    import torch as _torch
+    if isinstance(preds, _torch.Tensor):
+        dev = preds.device
+        y = _torch.as_tensor(np.asarray(info.labels, np.int64),
+                             device=dev).view(-1)
+        p = preds.detach().double().clamp(1e-16, 1 - 1e-16)
+        w = (_torch.as_tensor(np.asarray(info.weights, np.float64),
+                              device=dev) if info.weights is not None
+             else _torch.ones(y.shape[0], dtype=_torch.float64, device=dev))
+        ll = -_torch.log(p[_torch.arange(y.numel(), device=dev), y])
+        return _ratio(float((ll * w).sum()), float(w.sum()))
     y = np.asarray(info.labels, dtype=np.int64).reshape(-1)
     p = np.asarray(preds, dtype=np.float64)
     eps = 1e-16
@@ -203,7 +271,24 @@ def expectile_loss(preds, info, param=None):
 
 
 @register("auc")
+@_torch_ok
 def auc(preds, info, param=None):
+    import torch as _torch
+    if isinstance(preds, _torch.Tensor):
+        p = preds.detach().double()
+        if not (p.dim() == 2 and p.shape[1] > 1) and (
+                info.group_ptr is None or len(info.group_ptr) <= 2):
+            dev = p.device
+            y = _torch.as_tensor(np.asarray(info.labels, np.float64),
+                                 device=dev).view(-1)
+            w = (_torch.as_tensor(np.asarray(info.weights, np.float64),
+                                  device=dev) if info.weights is not None
+                 else _torch.ones(y.shape[0], dtype=_torch.float64,
+                                  device=dev))
+            a, valid = _binary_auc_t(p.view(-1), y, w)
+            s_, v_ = collective.allreduce_sum_scalars([a * valid, valid])
+            return s_ / max(v_, 1e-16)
+        preds = p.cpu().numpy()
     p = np.asarray(preds, dtype=np.float64)
     if p.ndim == 2 and p.shape[1] > 1:
         return _multi_auc(p, info)
@@ -216,6 +301,33 @@ def auc(preds, info, param=None):
     # distributed: weighted mean of per-worker AUC (reference auc.cc:125)
     s, v = collective.allreduce_sum_scalars([a * valid, valid])
     return s / max(v, 1e-16)
+
+
+def _binary_auc_t(p, y, w) -> Tuple[float, float]:
+    """Device-resident weighted ROC AUC: sort + cumsum + tie merge, one
+    host sync (reference BinaryROCAUC, src/metric/auc.cu:172)."""
+    import torch as _torch
+    order = _torch.argsort(-p, stable=True)
+    p, y, w = p[order], y[order], w[order]
+    pos = (w * y).sum()
+    neg = (w * (1 - y)).sum()
+    tp = _torch.cumsum(w * y, 0)
+    fp = _torch.cumsum(w * (1 - y), 0)
+    # keep the last index of each distinct prediction value
+    n = p.numel()
+    keep = _torch.ones(n, dtype=_torch.bool, device=p.device)
+    if n > 1:
+        keep[:-1] = p[:-1] != p[1:]
+    tp, fp = tp[keep], fp[keep]
+    z = _torch.zeros(1, dtype=tp.dtype, device=tp.device)
+    tp0 = _torch.cat([z, tp[:-1]])
+    fp0 = _torch.cat([z, fp[:-1]])
+    area = ((fp - fp0) * (tp + tp0) * 0.5).sum()
+    stats = _torch.stack([pos, neg, area]).cpu()  # ONE sync
+    pos_, neg_, area_ = (float(stats[0]), float(stats[1]), float(stats[2]))
+    if pos_ == 0 or neg_ == 0:
+        return 0.5, 0.0
+    return area_ / (pos_ * neg_), 1.0
 
 
 def _binary_auc(p, y, w) -> Tuple[float, float]:
