@@ -288,11 +288,16 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
                          int max_group_bins, const int32_t* cut_ptrs,
                          int use_shared, int64_t* node_sums,
                          hipStream_t stream) {
-  static int block_size = [] {
+  static int block_env = [] {
     const char* e = getenv("GBT_HIST_BLOCK_SIZE");
-    int v = e ? atoi(e) : GBT_HIST_BLOCK;
-    return (v >= 64 && v <= 1024) ? (v & ~63) : GBT_HIST_BLOCK;
+    int v = e ? atoi(e) : 0;
+    return (v >= 64 && v <= 1024) ? (v & ~63) : 0;
   }();
+  const int block_size = block_env ? block_env : GBT_HIST_BLOCK;
+  // round-2 sweep: the register-metadata kernel peaks at 512 threads
+  // (972 vs 918 rounds/s at 1024 — fewer waves contending the LDS
+  // atomic pipe, still enough to hide ds latency)
+  const int reg_block = block_env ? block_env : 512;
   dim3 grid(n_tasks, n_groups);
   dim3 block(block_size);
   size_t shmem = use_shared ? (size_t)max_group_bins * 2 * sizeof(int64_t) : 0;
@@ -306,13 +311,14 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
       n_features <= 32) {
     const bool vec = (n_features % 4 == 0) &&
                      (((uintptr_t)gidx8 & 3u) == 0);
+    dim3 rblock(reg_block);
     if (vec) {
-      hipLaunchKernelGGL((HistKernelReg<32, true>), dim3(n_tasks), block,
+      hipLaunchKernelGGL((HistKernelReg<32, true>), dim3(n_tasks), rblock,
                          shmem, stream, gidx8, n_features, qgpair, ridx,
                          tasks, out_hist, n_bins, n_features, cut_ptrs,
                          node_sums);
     } else {
-      hipLaunchKernelGGL((HistKernelReg<32, false>), dim3(n_tasks), block,
+      hipLaunchKernelGGL((HistKernelReg<32, false>), dim3(n_tasks), rblock,
                          shmem, stream, gidx8, n_features, qgpair, ridx,
                          tasks, out_hist, n_bins, n_features, cut_ptrs,
                          node_sums);
