@@ -465,3 +465,117 @@ def test_two_process_global_intercept():
                      verbose_eval=False)
     assert abs(res["base"] - float(solo.base_score)) < 1e-6, (
         res["base"], float(solo.base_score))
+
+
+TRACKER_WORKER = r"""
+import os, sys
+import numpy as np
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+# NO RANK in the env: the worker claims one from the tracker's store
+assert "RANK" not in os.environ
+collective.init("gloo", timeout=60)
+rank = collective.get_rank()
+world = collective.get_world_size()
+assert world == 2
+
+rng = np.random.RandomState(0)
+n, f = 1000, 6
+X = rng.randn(n, f).astype(np.float32)
+y = (X[:, 0] > 0).astype(np.float32)
+shard = slice(rank * n // world, (rank + 1) * n // world)
+bst = xgb.train({"objective": "binary:logistic", "max_depth": 3},
+                xgb.DMatrix(X[shard], label=y[shard]), 3,
+                verbose_eval=False)
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "model differs across tracker-assigned ranks"
+collective.finalize()
+"""
+
+
+def test_tracker_owned_rendezvous_and_rank_assignment():
+    """Builder-owned tracker: workers arrive WITHOUT ranks, claim them
+    from the tracker's store (reference tracker.cc rank assignment),
+    train in sync, and wait_for() observes completion."""
+    import xgboost_amd.tracker as tr
+    tracker = tr.RabitTracker(n_workers=2)
+    tracker.start()
+    args = tracker.worker_args()
+    procs = []
+    with tempfile.TemporaryDirectory() as td:
+        spath = os.path.join(td, "worker.py")
+        with open(spath, "w") as fh:
+            fh.write(TRACKER_WORKER)
+        for _ in range(2):
+            env = {k: v for k, v in os.environ.items() if k != "RANK"}
+            env.update({str(k): str(v) for k, v in args.items()})
+            env["XGB_AMD_REPO"] = REPO
+            procs.append(subprocess.Popen(
+                [sys.executable, spath], env=env,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        outs = [p.communicate(timeout=180)[0].decode() for p in procs]
+        assert all(p.returncode == 0 for p in procs), "\n---\n".join(outs)
+    tracker.wait_for(timeout=30)  # both workers posted completion
+
+
+FAILURE_WORKER = r"""
+import os, sys, time
+import numpy as np
+import torch
+import torch.distributed as dist
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+from xgboost_amd import collective
+
+collective.init("gloo", timeout=10)
+rank = collective.get_rank()
+t = torch.ones(4)
+dist.all_reduce(t)  # both alive: fine
+if rank == 1:
+    os._exit(17)  # die WITHOUT closing the group
+# rank 0: the next collective must fail within the watchdog budget,
+# not hang forever (reference watchdog + ncclCommAbort, coll.cu:157)
+t0 = time.monotonic()
+try:
+    for _ in range(100):
+        dist.all_reduce(t)
+        time.sleep(0.2)
+    print("SURVIVED-UNEXPECTEDLY", flush=True)
+    sys.exit(3)
+except Exception as e:  # noqa: BLE001
+    el = time.monotonic() - t0
+    print(f"DETECTED after {el:.1f}s: {type(e).__name__}", flush=True)
+    assert el < 60, el
+    try:
+        collective.signal_error(f"peer failure: {type(e).__name__}")
+    except RuntimeError:
+        pass
+    sys.exit(0)
+"""
+
+
+def test_collective_failure_detection():
+    """A dead worker must surface as an error on the survivors within
+    the watchdog budget instead of hanging the job."""
+    procs = []
+    port = 29611
+    with tempfile.TemporaryDirectory() as td:
+        spath = os.path.join(td, "worker.py")
+        with open(spath, "w") as fh:
+            fh.write(FAILURE_WORKER)
+        for r in range(2):
+            env = dict(os.environ)
+            env.update({"RANK": str(r), "WORLD_SIZE": "2",
+                        "MASTER_ADDR": "127.0.0.1",
+                        "MASTER_PORT": str(port),
+                        "XGB_AMD_REPO": REPO})
+            procs.append(subprocess.Popen(
+                [sys.executable, spath], env=env,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        out0, _ = procs[0].communicate(timeout=150)
+        procs[1].wait(timeout=30)
+        assert procs[1].returncode == 17
+        assert procs[0].returncode == 0, out0.decode()
+        assert b"DETECTED" in out0, out0.decode()

@@ -248,7 +248,11 @@ def test_rabit_tracker_rendezvous():
     assert args["MASTER_ADDR"] == "127.0.0.1"
     assert args["WORLD_SIZE"] == 2
     assert args["MASTER_PORT"] == args["DMLC_TRACKER_PORT"]
-    t.wait_for()
+    # wait_for now genuinely blocks on worker completion (reference
+    # XGTrackerWaitFor); with no workers it must time out, not no-op
+    import pytest as _pytest
+    with _pytest.raises(TimeoutError):
+        t.wait_for(timeout=1)
     t.free()
 
 

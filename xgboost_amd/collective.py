@@ -31,22 +31,68 @@ def get_world_size() -> int:
     return dist.get_world_size() if is_distributed() else 1
 
 
-def init(backend: Optional[str] = None, **kwargs) -> None:
-    """Initialize from torchrun-style env vars (RANK/WORLD_SIZE/MASTER_*).
+_TRACKER_STORE = None  # worker-side handle for done/error posting
 
-    Reference-style dmlc_* / xgboost_* kwargs (tracker host, task id,
-    communicator choice) are accepted and ignored — rendezvous comes
-    from the env, and the communicator IS torch.distributed."""
+
+def init(backend: Optional[str] = None, timeout: Optional[float] = None,
+         **kwargs) -> None:
+    """Initialize the communicator.
+
+    Rendezvous: torchrun-style env (RANK/WORLD_SIZE/MASTER_*), or a
+    builder-owned RabitTracker via DMLC_TRACKER_URI/PORT — in the
+    tracker case the worker connects the tracker's TCPStore, claims a
+    rank if it has none (reference tracker.cc rank assignment), and
+    builds the process group on that store.
+
+    Watchdog: `timeout` (seconds; env XGB_AMD_COLL_TIMEOUT; default
+    1800 like the reference's collective timeout) bounds every
+    collective.  For RCCL this arms torch's NCCL watchdog (async error
+    handling + comm abort on timeout — the reference's AsyncLaunch
+    watchdog + ncclCommAbort, src/collective/coll.cu:93-182); for gloo
+    the ops raise directly."""
+    global _TRACKER_STORE
     if is_distributed():
         return
+    import datetime
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
-    kwargs = {k: v for k, v in kwargs.items()
-              if not k.lower().startswith(("dmlc_", "xgboost_"))}
-    dist.init_process_group(backend=backend, **kwargs)
+    if timeout is None:
+        t = (kwargs.get("dmlc_timeout") or kwargs.get("DMLC_TIMEOUT")
+             or os.environ.get("XGB_AMD_COLL_TIMEOUT"))
+        timeout = float(t) if t else 1800.0
+    td = datetime.timedelta(seconds=float(timeout))
+    if backend == "nccl":
+        # abort the communicator (and surface the error) on watchdog
+        # timeout instead of hanging the job
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
+    tracker_uri = (kwargs.get("dmlc_tracker_uri")
+                   or os.environ.get("DMLC_TRACKER_URI"))
+    extra = {k: v for k, v in kwargs.items()
+             if not k.lower().startswith(("dmlc_", "xgboost_"))}
+    if tracker_uri and "store" not in extra:
+        from .tracker import connect_tracker
+        port = int(kwargs.get("dmlc_tracker_port")
+                   or os.environ.get("DMLC_TRACKER_PORT"))
+        world = int(kwargs.get("dmlc_num_worker")
+                    or os.environ.get("WORLD_SIZE", "0"))
+        rank_env = os.environ.get("RANK")
+        store, rank = connect_tracker(
+            tracker_uri, port, world,
+            int(rank_env) if rank_env is not None else None,
+            float(timeout))
+        _TRACKER_STORE = store
+        dist.init_process_group(backend=backend, store=store, rank=rank,
+                                world_size=world, timeout=td)
+        return
+    dist.init_process_group(backend=backend, timeout=td, **extra)
 
 
 def finalize() -> None:
+    global _TRACKER_STORE
+    if _TRACKER_STORE is not None:
+        from .tracker import post_done
+        post_done(_TRACKER_STORE)
+        _TRACKER_STORE = None
     if is_distributed():
         dist.destroy_process_group()
 
@@ -231,11 +277,15 @@ def get_processor_name() -> str:
     return socket.gethostname()
 
 
-def signal_error() -> None:
-    """Abort the process group after an unrecoverable worker error."""
+def signal_error(msg: str = "worker error") -> None:
+    """Abort the process group after an unrecoverable worker error and
+    post it to the tracker (reference SignalError, comm.h:44-101)."""
+    if _TRACKER_STORE is not None:
+        from .tracker import post_error
+        post_error(_TRACKER_STORE, msg)
     if is_distributed():
         dist.destroy_process_group()
-    raise RuntimeError("collective worker signalled an error")
+    raise RuntimeError(f"collective worker signalled an error: {msg}")
 
 
 class CommunicatorContext:
