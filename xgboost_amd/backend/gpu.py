@@ -309,34 +309,99 @@ class GpuOps(SegmentedOpsMixin):
 
     def _host_sorted_cat(self, out, hist, parents_np, quantizer, param,
                          wide_feats, feature_sets) -> None:
-        """Sorted-partition categorical splits on the pulled histogram
-        columns — identical math to the CPU oracle (splits.py)."""
-        from ..splits import (_sorted_cat_split, calc_gain_given_weight,
-                              calc_weight)
+        """Sorted-partition categorical splits (reference
+        gpu_hist/evaluate_splits.cuh SortHistogram): the per-category
+        ratio sort, present-compaction and prefix cumsums run ON DEVICE;
+        only the small (k x width) cumsum/order tables come to the host,
+        where the final gain math is the SAME numpy ops as the CPU
+        oracle (splits._sorted_cat_split), batched over nodes — so GPU
+        and CPU trees stay identical."""
+        from ..splits import calc_gain_given_weight, calc_weight
         cuts = self.qm.cuts
         k = len(out)
-        cols = np.concatenate([
-            np.arange(cuts.ptrs[fi], cuts.ptrs[fi + 1]) for fi in wide_feats])
-        cols_t = torch.as_tensor(cols, dtype=torch.long, device=self.device)
-        hist_cat = hist.index_select(1, cols_t).cpu().numpy()  # [k,|cols|,2]
-        hist_np = np.zeros((k, self.n_bins, 2), dtype=np.int64)
-        hist_np[:, cols, :] = hist_cat
         inv_g = 1.0 / quantizer.g_scale
         inv_h = 1.0 / quantizer.h_scale
-        for i, e in enumerate(out):
-            pgq, phq = int(parents_np[i, 0]), int(parents_np[i, 1])
-            pw = calc_weight(pgq * inv_g, phq * inv_h, param)
-            pgain = calc_gain_given_weight(pgq * inv_g, phq * inv_h, pw,
-                                           param)
-            allowed = None
-            if feature_sets is not None and feature_sets[i] is not None:
-                allowed = set(int(x) for x in feature_sets[i])
-            for fi in wide_feats:
-                if allowed is not None and int(fi) not in allowed:
-                    continue
-                _sorted_cat_split(e, hist_np[i], pgq, phq, int(fi),
-                                  cuts.ptrs, param, inv_g, inv_h,
-                                  float(pgain))
+        pgq_v = parents_np[:, 0].astype(np.int64)
+        phq_v = parents_np[:, 1].astype(np.int64)
+        pw_v = calc_weight(pgq_v * inv_g, phq_v * inv_h, param)
+        pgain_v = calc_gain_given_weight(pgq_v * inv_g, phq_v * inv_h,
+                                         pw_v, param)
+        allowed = [None] * k
+        if feature_sets is not None:
+            for i, fs in enumerate(feature_sets):
+                if fs is not None:
+                    allowed[i] = set(int(x) for x in fs)
+        mct = param.max_cat_threshold + 1
+        for fi in wide_feats:
+            b0, b1 = int(cuts.ptrs[fi]), int(cuts.ptrs[fi + 1])
+            w = b1 - b0
+            Gc = hist[:, b0:b1, 0]                      # int64 [k, w]
+            Hc = hist[:, b0:b1, 1]
+            present = Hc != 0
+            ratio = torch.where(
+                present,
+                (Gc.double() * inv_g) / (Hc.double() * inv_h
+                                         + param.reg_lambda),
+                torch.full_like(Gc, float("inf"), dtype=torch.float64))
+            order = torch.argsort(ratio, dim=1, stable=True)  # absent last
+            Gs = Gc.gather(1, order)
+            Hs = Hc.gather(1, order)
+            cg_t = torch.cumsum(Gs, 1)
+            ch_t = torch.cumsum(Hs, 1)
+            npres_t = present.sum(1)
+            # ONE compact download per feature: [k, w] cumsums + order
+            cg = cg_t.cpu().numpy()
+            ch = ch_t.cpu().numpy()
+            order_h = order.cpu().numpy()
+            npres = npres_t.cpu().numpy()
+            featG = cg[:, -1]
+            featH = ch[:, -1]
+            missG = pgq_v - featG
+            missH = phq_v - featH
+            # positions beyond the present prefix / threshold are invalid
+            pos = np.arange(w)[None, :]
+            limit = np.minimum(npres, mct)[:, None]
+            pos_ok = pos < limit
+            for missing_left in (False, True):
+                gl = cg + (missG[:, None] if missing_left else 0)
+                hl = ch + (missH[:, None] if missing_left else 0)
+                gr = pgq_v[:, None] - gl
+                hr = phq_v[:, None] - hl
+                glf, hlf = gl * inv_g, hl * inv_h
+                grf, hrf = gr * inv_g, hr * inv_h
+                wl = calc_weight(glf, hlf, param)
+                wr = calc_weight(grf, hrf, param)
+                gains = (calc_gain_given_weight(glf, hlf, wl, param)
+                         + calc_gain_given_weight(grf, hrf, wr, param)
+                         - pgain_v[:, None])
+                ok = (pos_ok & (hlf >= param.min_child_weight)
+                      & (hrf >= param.min_child_weight)
+                      & (hl > 0) & (hr > 0))
+                gains = np.where(ok, gains, -np.inf)
+                for i, e in enumerate(out):
+                    if npres[i] < 2:
+                        continue
+                    if allowed[i] is not None and int(fi) not in allowed[i]:
+                        continue
+                    row = gains[i]
+                    if not np.isfinite(row).any():
+                        continue
+                    j = int(np.argmax(row))
+                    gv = float(row[j])
+                    if gv > e.gain:
+                        e.gain = gv
+                        e.feature = int(fi)
+                        e.split_bin = b0
+                        e.default_left = missing_left
+                        e.left_gq = int(gl[i, j])
+                        e.left_hq = int(hl[i, j])
+                        e.right_gq = int(pgq_v[i]) - e.left_gq
+                        e.right_hq = int(phq_v[i]) - e.left_hq
+                        e.is_cat = True
+                        left_set = set(int(c) for c in order_h[i, :j + 1])
+                        e.cat_bits = np.array(
+                            sorted(c for c in range(w)
+                                   if c not in left_set), dtype=np.int32)
 
     def partition(self, ridx: torch.Tensor,
                   segments: Sequence[Tuple[int, int]],
