@@ -215,6 +215,62 @@ class GpuOps(SegmentedOpsMixin):
         collective.allreduce_sum_(hist)
         return hist
 
+    def _mt_grouping(self, T: int):
+        """Feature groups sized so T histograms fit the LDS budget
+        (reference MtHistKernel AllocateBlocks, histogram.cu:309)."""
+        cache = self.__dict__.setdefault("_mt_group_cache", {})
+        if T in cache:
+            return cache[T]
+        budget = LDS_MAX_GROUP_BINS // T
+        widths = np.diff(self.qm.cuts.ptrs)
+        if int(widths.max()) > budget:
+            cache[T] = None  # a single feature exceeds the MT budget
+            return None
+        groups_f, groups_b, acc = [0], [0], 0
+        for f, w in enumerate(widths):
+            if acc + w > budget and acc > 0:
+                groups_f.append(f)
+                groups_b.append(int(self.qm.cuts.ptrs[f]))
+                acc = 0
+            acc += int(w)
+        groups_f.append(len(widths))
+        groups_b.append(int(self.qm.cuts.ptrs[-1]))
+        dev = self.device
+        out = (torch.tensor(groups_f, dtype=torch.int32, device=dev),
+               torch.tensor(groups_b, dtype=torch.int32, device=dev),
+               len(groups_f) - 1,
+               int(np.max(np.diff(groups_b))))
+        cache[T] = out
+        return out
+
+    def build_hist_nodes_mt(self, qg_mt: torch.Tensor, nids):
+        """Fused multi-target histogram: ONE pass over the bin matrix
+        accumulates all targets (gbt_hist_mt; reference MtHistKernel).
+        qg_mt: [n, T, 2] int32 contiguous.  Returns [T, k, n_bins, 2]
+        or None when the shape is unsupported (caller falls back to the
+        per-target loop)."""
+        T = int(qg_mt.shape[1])
+        if T > 8 or not hasattr(self.lib, "gbt_hist_mt"):
+            return None
+        grouping = self._mt_grouping(T)
+        if grouping is None:
+            return None
+        fg, bg, n_groups, max_gb = grouping
+        segs = [self.segments[n] for n in nids]
+        tasks_np = _chunk_tasks(segs, target_tasks=512)
+        (tasks,) = self.stager.upload([tasks_np])
+        k = len(nids)
+        out = torch.zeros((T, k, self.n_bins, 2), dtype=torch.int64,
+                          device=self.device)
+        p8, p16 = self._gidx_ptrs()
+        self.lib.gbt_hist_mt(
+            p8, p16, self.qm.n_features, self.hip.ptr(qg_mt), T,
+            self.hip.ptr(self.ridx), self.hip.ptr(tasks), len(tasks_np),
+            self.hip.ptr(out), self.n_bins, k,
+            self.hip.ptr(fg), self.hip.ptr(bg), n_groups, max_gb,
+            self.hip.ptr(self.cut_ptrs), self.hip.stream())
+        return out
+
     def evaluate_splits(self, hist: torch.Tensor, quantizer: GradQuantizer,
                         parent_sums: Sequence[Tuple[int, int]],
                         nids: Sequence[int], param: TrainParam,

@@ -549,8 +549,21 @@ class MultiTargetGrower:
         tree.base_weight[0] = float(np.mean(w0))
         tree.sum_hess[0] = float((root[:, 1] / h_scales).sum())
 
-        hist0 = torch.stack([ops.build_hist_nodes(qgpairs[t], [0])[0]
-                             for t in range(T)])
+        qg_mt = None
+        if hasattr(ops, "build_hist_nodes_mt") and T <= 8:
+            qg_mt = torch.stack(qgpairs, dim=1).contiguous()
+
+        def mt_hist(nids):
+            """[T, k, bins, 2] — fused single-pass kernel when
+            available (gbt_hist_mt), else per-target launches."""
+            if qg_mt is not None:
+                bh = ops.build_hist_nodes_mt(qg_mt, nids)
+                if bh is not None:
+                    return bh
+            return torch.stack([ops.build_hist_nodes(qgpairs[t], nids)
+                                for t in range(T)])
+
+        hist0 = mt_hist([0])[:, 0]
         ops.allreduce_hist(hist0)
         hists[0] = hist0
 
@@ -639,10 +652,8 @@ class MultiTargetGrower:
                 else:
                     build_nodes.append((r, b.nid, l))
             if build_nodes:
-                bh = torch.stack([
-                    ops.build_hist_nodes(qgpairs[t],
-                                         [n for n, _, _ in build_nodes])
-                    for t in range(len(qgpairs))])  # [T, k, bins, 2]
+                bh = mt_hist([n for n, _, _ in build_nodes])
+                bh = bh.contiguous()
                 ops.allreduce_hist(bh)
                 for i, (n, parent, sib) in enumerate(build_nodes):
                     hists[n] = bh[:, i]

@@ -81,6 +81,8 @@ _SIGS = {
                          _p, _p, _p, _i64, _i, _i, _p, _p],
     "gbt_shap_ix": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
                     _p, _i64, _i, _i, _p, _p],
+    "gbt_hist_mt": [_p, _p, _i, _p, _i, _p, _p, _i, _p, _i, _i,
+                    _p, _p, _i, _i, _p, _p],
 }
 
 

@@ -351,3 +351,103 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
     }
   }
 }
+
+// Fused multi-target histogram (reference MtHistKernel,
+// src/tree/gpu_hist/histogram.cu:256): ONE pass over the rows
+// accumulates every target's (g, h) — the python fallback launches the
+// single-target kernel T times and re-reads the bin matrix per target.
+// LDS layout [group_bins][T][2] u64; the caller regroups features so
+// T * group_bins * 16 B fits the LDS budget.  Output is target-major:
+// out[slot][T][n_bins][2] (the MT grower's stacked-tensor layout).
+#define GBT_HIST_MT_MAX_T 8
+
+template <typename BinT>
+__global__ __launch_bounds__(512) void HistMtKernel(
+    const BinT* __restrict__ gidx, int n_features,
+    const int32_t* __restrict__ qg /* [n][T][2] */, int T,
+    const int32_t* __restrict__ ridx, const BlockTask* __restrict__ tasks,
+    int64_t* __restrict__ out_hist /* [T][n_slots][n_bins][2] */,
+    int n_bins, int n_slots,
+    const int32_t* __restrict__ feat_group_start,
+    const int32_t* __restrict__ bin_group_start,
+    const int32_t* __restrict__ cut_ptrs) {
+  const BlockTask task = tasks[blockIdx.x];
+  if (task.row_begin >= task.row_end) return;
+  const int group = blockIdx.y;
+  const int f_begin = feat_group_start[group];
+  const int f_end = feat_group_start[group + 1];
+  const int bin_begin = bin_group_start[group];
+  const int group_bins = bin_group_start[group + 1] - bin_begin;
+  const int gf = f_end - f_begin;
+
+  __shared__ int s_start[GBT_HIST_MAX_F];
+  __shared__ int s_width[GBT_HIST_MAX_F];
+  for (int f = threadIdx.x; f < gf; f += blockDim.x) {
+    const int c0 = cut_ptrs[f_begin + f];
+    s_start[f] = c0 - bin_begin;
+    s_width[f] = cut_ptrs[f_begin + f + 1] - c0;
+  }
+  extern __shared__ unsigned long long smem[];  // [group_bins][T][2]
+  const int total = group_bins * T * 2;
+  for (int i = threadIdx.x; i < total; i += blockDim.x) smem[i] = 0ULL;
+  __syncthreads();
+
+  for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
+       i += blockDim.x) {
+    const int row = ridx[i];
+    long long gv[GBT_HIST_MT_MAX_T], hv[GBT_HIST_MT_MAX_T];
+    const int32_t* qp = qg + (size_t)row * T * 2;
+    for (int t = 0; t < T; ++t) {
+      gv[t] = qp[2 * t];
+      hv[t] = qp[2 * t + 1];
+    }
+    const BinT* rowbins = gidx + (size_t)row * n_features + f_begin;
+    for (int f = 0; f < gf; ++f) {
+      const int local = (int)rowbins[f];
+      if (local >= s_width[f]) continue;
+      unsigned long long* cell =
+          smem + ((size_t)(s_start[f] + local) * T) * 2;
+      for (int t = 0; t < T; ++t) {
+        atomicAdd(&cell[2 * t], (unsigned long long)gv[t]);
+        atomicAdd(&cell[2 * t + 1], (unsigned long long)hv[t]);
+      }
+    }
+  }
+  __syncthreads();
+  // flush [bin][t][c] -> out[slot][t][bin_begin+bin][c]
+  for (int i = threadIdx.x; i < total; i += blockDim.x) {
+    const unsigned long long v = smem[i];
+    if (v == 0ULL) continue;
+    const int c = i & 1;
+    const int t = (i >> 1) % T;
+    const int bin = (i >> 1) / T;
+    atomicAdd((unsigned long long*)&out_hist[
+                  (((size_t)t * n_slots + task.out_slot) * n_bins
+                   + bin_begin + bin) * 2 + c], v);
+  }
+}
+
+extern "C" void gbt_hist_mt(const uint8_t* gidx8, const uint16_t* gidx16,
+                            int n_features, const int32_t* qg, int T,
+                            const int32_t* ridx, const BlockTask* tasks,
+                            int n_tasks, int64_t* out_hist, int n_bins,
+                            int n_slots,
+                            const int32_t* feat_group_start,
+                            const int32_t* bin_group_start, int n_groups,
+                            int max_group_bins, const int32_t* cut_ptrs,
+                            hipStream_t stream) {
+  dim3 grid(n_tasks, n_groups);
+  dim3 block(512);
+  size_t shmem = (size_t)max_group_bins * T * 2 * sizeof(int64_t);
+  if (gidx8 != nullptr) {
+    hipLaunchKernelGGL((HistMtKernel<uint8_t>), grid, block, shmem, stream,
+                       gidx8, n_features, qg, T, ridx, tasks, out_hist,
+                       n_bins, n_slots, feat_group_start, bin_group_start,
+                       cut_ptrs);
+  } else {
+    hipLaunchKernelGGL((HistMtKernel<uint16_t>), grid, block, shmem, stream,
+                       gidx16, n_features, qg, T, ridx, tasks, out_hist,
+                       n_bins, n_slots, feat_group_start, bin_group_start,
+                       cut_ptrs);
+  }
+}
