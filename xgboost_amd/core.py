@@ -614,21 +614,24 @@ class Booster:
             return self._predict_margin_extmem(dmat, out, lo, hi)
         if getattr(dmat, "_sparse_data", None) is not None:
             return self._predict_margin_sparse(dmat, out, lo, hi)
-        X = dmat.raw_data()
         train_cats = getattr(self, "cat_categories_", None)
         pred_cats = getattr(dmat, "categories_", None)
         aligned = False
+        X = None
         if train_cats and pred_cats and pred_cats != train_cats:
             # predict-frame categories re-coded to the training dictionary
             # (reference encoder/ordinal.h Recode)
             from .data import align_categories
-            X = align_categories(X, pred_cats, train_cats)
+            X = align_categories(dmat.raw_data(), pred_cats, train_cats)
             aligned = True
         has_mt = any(t.leaf_values is not None for t in self.trees[lo:hi])
         if self.device.type == "cuda" and (hi - lo) > 0 and not has_mt \
                 and not aligned:
+            # device-resident inputs never round-trip to host here
             from .backend.gpu import predict_margin_gpu
             return predict_margin_gpu(self, dmat, out, lo, hi)
+        if X is None:
+            X = dmat.raw_data()
         for t in range(lo, hi):
             tree = self.trees[t]
             pos = tree.predict_leaf_np(X, dmat.missing)
