@@ -252,3 +252,31 @@ def test_inplace_predict_zero_copy_proxy():
     r3 = bst.inplace_predict(Xm, missing=-999.0)
     r4 = bst.predict(xgb.DMatrix(Xn))
     assert np.allclose(r3, r4, atol=1e-6)
+
+
+def test_arrow_table_ingestion():
+    """pyarrow Table ingestion (reference _from_arrow_table) incl.
+    dictionary-encoded categoricals and nulls-as-missing."""
+    import pyarrow as pa
+    rng = np.random.RandomState(4)
+    a = rng.randn(300).astype(np.float32)
+    b = rng.randn(300).astype(np.float64)
+    b[5] = np.nan
+    cat = pa.array(rng.choice(["x", "y", "z"], 300)).dictionary_encode()
+    tbl = pa.table({"a": pa.array(a), "b": pa.array(b), "c": cat})
+    y = (a > 0).astype(np.float32)
+    d = xgb.DMatrix(tbl, label=y, enable_categorical=True)
+    assert d.num_col() == 3
+    assert d.feature_names == ["a", "b", "c"]
+    assert d.feature_types == ["float", "float", "c"]
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 3}, d, 5,
+                    verbose_eval=False)
+    acc = ((bst.predict(d) > 0.5) == y).mean()
+    assert acc > 0.9
+    # nulls became missing
+    X = d.raw_data()
+    assert np.isnan(X[5, 1])
+    # plain (non-categorical) tables work without the flag
+    tbl2 = pa.table({"a": pa.array(a), "b": pa.array(b)})
+    d2 = xgb.DMatrix(tbl2, label=y)
+    assert d2.num_col() == 2

@@ -569,6 +569,11 @@ def _ingest(data: Any, enable_categorical: bool):
         if X.ndim == 1:
             X = X.reshape(-1, 1)
         return np.ascontiguousarray(X, dtype=np.float32), names, types, cats
+    elif _is_arrow(data):
+        # Arrow table / record batch (reference _from_arrow_table,
+        # python-package/xgboost/data.py:836): columnar ingestion with
+        # dictionary columns as categoricals
+        return _from_arrow(data, enable_categorical)
     else:
         X = np.asarray(data)
     if X.ndim == 1:
@@ -581,6 +586,49 @@ def _ingest(data: Any, enable_categorical: bool):
 
 def _is_pandas(data: Any) -> bool:
     return type(data).__module__.startswith("pandas") and hasattr(data, "dtypes")
+
+
+def _is_arrow(data: Any) -> bool:
+    mod = type(data).__module__
+    return mod.startswith("pyarrow") and hasattr(data, "column_names")
+
+
+def _from_arrow(table, enable_categorical: bool):
+    """pyarrow Table/RecordBatch -> dense float32 + names/types/categories.
+
+    Dictionary-encoded columns become categoricals (codes as values,
+    like the pandas path); nulls become NaN (missing)."""
+    import pyarrow as pa
+    names = [str(c) for c in table.column_names]
+    types: List[str] = []
+    cols = []
+    categories: Dict[int, list] = {}
+    for j, name in enumerate(table.column_names):
+        col = table.column(name)
+        if isinstance(col, pa.ChunkedArray):
+            col = col.combine_chunks()
+        if pa.types.is_dictionary(col.type):
+            if not enable_categorical:
+                raise ValueError(
+                    f"categorical (dictionary) column {name!r} needs "
+                    f"enable_categorical=True")
+            categories[j] = col.dictionary.to_pylist()
+            codes = col.indices.to_numpy(zero_copy_only=False)
+            codes = codes.astype(np.float32)
+            null_mask = ~np.asarray(col.is_valid())
+            codes[null_mask] = np.nan
+            cols.append(codes)
+            types.append("c")
+        else:
+            arr = col.to_numpy(zero_copy_only=False).astype(np.float32)
+            null_mask = ~np.asarray(col.is_valid())
+            if null_mask.any():
+                arr = arr.copy()
+                arr[null_mask] = np.nan
+            cols.append(arr)
+            types.append("float")
+    X = np.ascontiguousarray(np.stack(cols, axis=1), dtype=np.float32)
+    return X, names, types, categories
 
 
 def _from_pandas(df, enable_categorical: bool):
