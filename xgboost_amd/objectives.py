@@ -817,8 +817,19 @@ class _LambdaRankBase(Objective):
             h_out.index_add_(0, idx_low, hess)
             if self.normalize:
                 if self.pair_method == "topk":
-                    s_g = torch.zeros(G, dtype=torch.float64, device=dev)
-                    s_g.index_add_(0, pgk, -2.0 * lam)
+                    # per-group sum WITHOUT atomics: pgk is sorted
+                    # (repeat_interleave order survives the keep mask),
+                    # and ~pairs-per-group-way contended fp64 atomics
+                    # serialize into a CAS loop on ROCm (measured 380 ms
+                    # of a 395 ms round); cumsum + searchsorted is
+                    # bandwidth-bound and deterministic
+                    cs = torch.zeros(pgk.numel() + 1, dtype=torch.float64,
+                                     device=dev)
+                    torch.cumsum(-2.0 * lam, 0, out=cs[1:])
+                    grange = torch.arange(G, device=dev)
+                    lo = torch.searchsorted(pgk, grange, side="left")
+                    hi = torch.searchsorted(pgk, grange, side="right")
+                    s_g = cs[hi] - cs[lo]
                     norm_g = torch.where(
                         s_g > 0,
                         torch.log2(1.0 + s_g) / s_g.clamp(min=1e-300),
