@@ -786,8 +786,10 @@ class _LambdaRankBase(Objective):
             ya, yb = y[da], y[db]
             keep = ya != yb
             da, db, ya, yb = da[keep], db[keep], ya[keep], yb[keep]
-            ra = (a_flat - gp[pg])[keep]
-            rb = (b_flat - gp[pg])[keep]
+            af = a_flat[keep]
+            bf = b_flat[keep]
+            ra = af - gp[pg][keep]
+            rb = bf - gp[pg][keep]
             pgk = pg[keep]
             swap = ya < yb
             idx_high = torch.where(swap, db, da)
@@ -811,10 +813,28 @@ class _LambdaRankBase(Objective):
                     ok, delta / ((s_high - s_low).abs() + 0.01), delta)
             lam = (sig - 1.0) * delta
             hess = torch.clamp(sig * (1.0 - sig), min=1e-16) * delta * 2.0
-            g_out.index_add_(0, idx_high, lam)
-            g_out.index_add_(0, idx_low, -lam)
-            h_out.index_add_(0, idx_high, hess)
-            h_out.index_add_(0, idx_low, hess)
+            # Scatter WITHOUT atomics (fp64 atomics CAS-loop on ROCm and
+            # the hot top-ranked docs make them ~100-way contended):
+            # accumulate by RANK-LIST POSITION with sort + cumsum +
+            # searchsorted segment sums, then write through `order` as a
+            # pure permutation.  Also makes gradients accumulation-order
+            # deterministic.
+            idxN = torch.arange(N + 1, device=dev)
+
+            def segsum(keys_sorted, vals):
+                cs = torch.zeros(vals.numel() + 1, dtype=torch.float64,
+                                 device=dev)
+                torch.cumsum(vals, 0, out=cs[1:])
+                bnd = torch.searchsorted(keys_sorted, idxN)
+                return cs[bnd[1:]] - cs[bnd[:-1]]
+
+            val_a = torch.where(swap, -lam, lam)  # pair grad at side a
+            ka, pa = torch.sort(af)
+            kb, pb = torch.sort(bf)
+            g_r = segsum(ka, val_a[pa]) + segsum(kb, -val_a[pb])
+            h_r = segsum(ka, hess[pa]) + segsum(kb, hess[pb])
+            g_out[order] = g_r
+            h_out[order] = h_r
             if self.normalize:
                 if self.pair_method == "topk":
                     # per-group sum WITHOUT atomics: pgk is sorted
