@@ -62,6 +62,14 @@ def _flat_finite_columns(X: torch.Tensor, weights: Optional[torch.Tensor],
     if total == 0:
         return (torch.zeros(0, dtype=torch.float32, device=dev),
                 torch.zeros(0, dtype=torch.float64, device=dev), cnt)
+    if total == n * f:
+        # no missing anywhere: the flat segmented layout is just the
+        # column-major transpose (skips two 2-D gather passes — the
+        # dominant cost of the per-tree approx regen on dense data)
+        vals = sorted_vals.t().reshape(-1).float()
+        w = (w_sorted.t().reshape(-1) if w_sorted is not None
+             else torch.ones(total, dtype=torch.float64, device=dev))
+        return vals, w, cnt
     col_ptr = torch.zeros(f + 1, dtype=torch.long, device=dev)
     torch.cumsum(cnt, 0, out=col_ptr[1:])
     f_ids = torch.repeat_interleave(torch.arange(f, device=dev), cnt)
