@@ -813,3 +813,28 @@ def test_gpu_multi_target_training_quality():
     pred = bst.predict(xgb.DMatrix(X))
     rmse = float(np.sqrt(((pred - Y) ** 2).mean()))
     assert rmse < 0.4, rmse
+
+
+def test_gpu_colsample_bytree_native_matches_python_driver():
+    """colsample_bytree now runs on the native whole-tree driver with a
+    broadcast feature mask; trees must match the Python GPU driver."""
+    from xgboost_amd.grower import TreeGrower
+    X, y = _data(30000, 12, seed=29)
+    pd = {"objective": "binary:logistic", "max_depth": 7, "max_bin": 128,
+          "colsample_bytree": 0.5, "seed": 11, "device": "cuda"}
+    bg = xgb.train(pd, xgb.DMatrix(X, label=y), 6)
+    orig = TreeGrower._try_native
+    TreeGrower._try_native = lambda self, qg, t: None
+    try:
+        bp = xgb.train(pd, xgb.DMatrix(X, label=y), 6)
+    finally:
+        TreeGrower._try_native = orig
+    used = set()
+    for tg, tc in zip(bg.trees, bp.trees):
+        assert tg.n_nodes == tc.n_nodes
+        assert np.array_equal(tg.split_index[:tg.n_nodes],
+                              tc.split_index[:tc.n_nodes])
+        assert np.array_equal(tg.left[:tg.n_nodes], tc.left[:tc.n_nodes])
+        used |= set(int(f) for i, f in enumerate(tg.split_index[:tg.n_nodes])
+                    if tg.left[i] != -1)
+    assert len(used) <= 12  # sanity; sampling varies per tree

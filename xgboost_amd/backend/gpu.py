@@ -522,7 +522,8 @@ class GpuOps(SegmentedOpsMixin):
     # -- native C++ level-loop driver (driver.hip) ----------------------
     def grow_tree_native(self, qgpair: torch.Tensor, tree, param,
                          quantizer, monotone: Optional[np.ndarray],
-                         root_sums: torch.Tensor):
+                         root_sums: torch.Tensor,
+                         feature_mask: Optional[np.ndarray] = None):
         """Run the whole per-tree loop in C++ (gbt_grow_tree).  Returns
         (tree, positions) or None when the config is unsupported (the
         Python driver handles those)."""
@@ -599,6 +600,11 @@ class GpuOps(SegmentedOpsMixin):
             ("split_index", np.int32), ("split_cond", np.float32),
             ("default_left", np.uint8), ("loss_chg", np.float64),
             ("sum_hess", np.float32), ("base_weight", np.float32)]}
+        fmask_ptr = None
+        if feature_mask is not None:
+            self._fmask_dev_t = torch.from_numpy(
+                np.ascontiguousarray(feature_mask, np.uint8)).to(dev)
+            fmask_ptr = self.hip.ptr(self._fmask_dev_t)
         mono_dev = mono_host = None
         if monotone is not None:
             m8 = np.ascontiguousarray(monotone, np.int8)
@@ -665,7 +671,7 @@ class GpuOps(SegmentedOpsMixin):
             gsc, hsc,
             param.reg_lambda, param.reg_alpha, param.max_delta_step,
             param.min_child_weight, param.gamma, param.eta, param.max_depth,
-            mono_dev, mono_host, cb,
+            mono_dev, mono_host, fmask_ptr, cb,
             *[host[k].ctypes.data_as(ctypes.c_void_p) for k in (
                 "left", "right", "parent", "split_index", "split_cond",
                 "default_left", "loss_chg", "sum_hess", "base_weight")],

@@ -165,9 +165,15 @@ class TreeGrower:
         if not hasattr(ops, "grow_tree_native"):
             return None
         if (self.cat_mask is not None or self.interaction is not None
-                or p.colsample_bytree < 1.0 or p.colsample_bylevel < 1.0
-                or p.colsample_bynode < 1.0):
+                or p.colsample_bylevel < 1.0 or p.colsample_bynode < 1.0):
             return None
+        fmask = None
+        if p.colsample_bytree < 1.0:
+            # per-TREE sampling is a constant feature set: the native
+            # evaluator masks it with one broadcast row (rank-identical:
+            # ColumnSampler seeding is deterministic per seed)
+            fmask = np.zeros(ops.qm.n_features, np.uint8)
+            fmask[self.col_sampler.tree_set] = 1
         needs_replay = p.grow_policy == "lossguide" or p.max_leaves > 0
         if needs_replay:
             if p.max_depth <= 0 or p.max_depth > 14:
@@ -184,7 +190,7 @@ class TreeGrower:
             if collective.is_distributed():
                 collective.allreduce_sum_(rs)
         out = ops.grow_tree_native(qgpair, tree, p, self.quantizer,
-                                   self.monotone, rs)
+                                   self.monotone, rs, feature_mask=fmask)
         if out is None or not needs_replay:
             return out
         dtree, pos = out

@@ -69,6 +69,7 @@ _SIGS = {
         _d, _d, _d, _d, _d, _d,  # lambda, alpha, mds, mcw, gamma, eta
         _i,                      # max_depth
         _p, _p,                  # monotone dev/host
+        _p,                      # colsample_bytree feature mask (dev)
         _p,                      # allreduce callback
         _p, _p, _p, _p, _p, _p, _p, _p, _p,  # tree out arrays
         _p,                      # stream

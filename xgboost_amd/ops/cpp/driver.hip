@@ -542,6 +542,8 @@ int gbt_grow_tree(
     double reg_lambda, double reg_alpha, double max_delta_step,
     double min_child_weight, double gamma, double eta, int max_depth,
     const int8_t* monotone_dev, const int8_t* monotone_host,
+    const uint8_t* fmask_dev,  // null, or [n_features] colsample_bytree
+                               // feature mask (broadcast to every node)
     AllreduceFn allreduce,
     // host tree outputs (caller-sized to 2^(max_depth+1))
     int32_t* out_left, int32_t* out_right, int32_t* out_parent,
@@ -650,21 +652,21 @@ int gbt_grow_tree(
       char* d = (char*)ctx->ring.dev[slot];
       ps_arg = (const int64_t*)(d + off_ps);
       if (has_mono) {
-        gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
+        gbt_evaluate_masked(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
                      maxabs_eval,
                      g_scale, h_scale, reg_lambda, reg_alpha, max_delta_step,
                      min_child_weight, monotone_dev,
-                     (const double*)(d + off_bd), nullptr, nullptr, eval_gain,
+                     (const double*)(d + off_bd), 0 /*mask_stride*/, fmask_dev, nullptr, eval_gain,
                      eval_bin, eval_dir, eval_lsum, nullptr, 0, stream);
         gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k,
                         n_features, eval_best, nullptr, stream);
         return 0;
       }
     }
-    gbt_evaluate(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
+    gbt_evaluate_masked(hists, k, n_bins, n_features, cut_ptrs_dev, ps_arg,
                  maxabs_eval, g_scale,
                  h_scale, reg_lambda, reg_alpha, max_delta_step,
-                 min_child_weight, monotone_dev, nullptr, nullptr, nullptr,
+                 min_child_weight, monotone_dev, nullptr, 0 /*mask_stride*/, fmask_dev, nullptr,
                  eval_gain, eval_bin, eval_dir, eval_lsum, nullptr, 0, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, k, n_features,
                     eval_best, nullptr, stream);
@@ -800,10 +802,10 @@ int gbt_grow_tree(
                                stream));
     }
     // root evaluation straight into best_rec[0]
-    gbt_evaluate(hist_pool_a, 1, n_bins, n_features, cut_ptrs_dev,
+    gbt_evaluate_masked(hist_pool_a, 1, n_bins, n_features, cut_ptrs_dev,
                  root_sums_dev, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
                  max_delta_step, min_child_weight,
-                 has_mono ? monotone_dev : nullptr, nullptr, nullptr,
+                 has_mono ? monotone_dev : nullptr, nullptr, 0 /*mask_stride*/, fmask_dev,
                  nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, nullptr,
                  0, stream);
     gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, 1, n_features,
@@ -868,12 +870,11 @@ int gbt_grow_tree(
                            stream, cur_pool, next_pool, next_pool, d_pslot,
                            (int)hist_row, 0, ps_next, d_pps, kp_arr + L + 1);
       }
-      gbt_evaluate(next_pool, cap_kids, n_bins, n_features, cut_ptrs_dev,
+      gbt_evaluate_masked(next_pool, cap_kids, n_bins, n_features, cut_ptrs_dev,
                    ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
                    max_delta_step, min_child_weight,
                    has_mono ? monotone_dev : nullptr,
-                   has_mono ? bnd_bufs[L & 1] : nullptr,
-                   nullptr, nullptr, eval_gain, eval_bin, eval_dir,
+                   has_mono ? bnd_bufs[L & 1] : nullptr, 0 /*mask_stride*/, fmask_dev, nullptr, eval_gain, eval_bin, eval_dir,
                    eval_lsum, kn_arr + L + 1, 0, stream);
       gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, cap_kids,
                       n_features, bo, kn_arr + L + 1, stream);

@@ -75,6 +75,8 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
     double reg_lambda, double reg_alpha, double max_delta_step,
     double min_child_weight, const int8_t* __restrict__ monotone,
     const double* __restrict__ node_bounds,
+    int mask_stride /* n_features = per-node mask rows; 0 = one
+                        per-tree row broadcast to every node */,
     const uint8_t* __restrict__ feature_mask,
     const uint8_t* __restrict__ cat_feature, double* __restrict__ out_gain,
     int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dir,
@@ -89,7 +91,8 @@ __global__ __launch_bounds__(64) void EvaluateKernel(
       (cat_feature == nullptr || !cat_feature[f])) {
     continue;  // the narrow kernel writes this slot
   }
-  if (feature_mask != nullptr && feature_mask[out_idx] == 0) {
+  if (feature_mask != nullptr &&
+      feature_mask[(size_t)node * mask_stride + f] == 0) {
     if (lane == 0) {
       out_gain[out_idx] = -INFINITY;
       out_bin[out_idx] = -1;
@@ -238,7 +241,7 @@ __global__ __launch_bounds__(256) void EvaluateNarrowKernel(
     double max_delta_step, double min_child_weight,
     const int8_t* __restrict__ monotone,
     const double* __restrict__ node_bounds,
-    const uint8_t* __restrict__ feature_mask,
+    int mask_stride, const uint8_t* __restrict__ feature_mask,
     const uint8_t* __restrict__ cat_feature, double* __restrict__ out_gain,
     int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dir,
     int64_t* __restrict__ out_lsum) {
@@ -259,7 +262,8 @@ __global__ __launch_bounds__(256) void EvaluateNarrowKernel(
     if (fb1 - fb0 > narrow_max) continue;           // wave kernel's
     if (cat_feature != nullptr && cat_feature[f]) continue;  // slot
     const size_t out_idx = (size_t)node * n_features + f;
-    if (feature_mask != nullptr && feature_mask[out_idx] == 0) {
+    if (feature_mask != nullptr &&
+        feature_mask[(size_t)node * mask_stride + f] == 0) {
       out_gain[out_idx] = -INFINITY;
       out_bin[out_idx] = -1;
       continue;
@@ -476,12 +480,34 @@ extern "C" void gbt_evaluate(
     const uint8_t* cat_feature, double* out_gain, int32_t* out_bin,
     uint8_t* out_dir, int64_t* out_lsum, const int32_t* k_dev,
     int narrow_max, hipStream_t stream) {
+  gbt_evaluate_masked(hist, n_nodes, n_bins, n_features, cut_ptrs,
+                      parent_sums, maxabs, g_scale, h_scale, reg_lambda,
+                      reg_alpha, max_delta_step, min_child_weight, monotone,
+                      node_bounds, n_features, feature_mask, cat_feature,
+                      out_gain, out_bin, out_dir, out_lsum, k_dev,
+                      narrow_max, stream);
+}
+
+// mask_stride: n_features = per-node mask rows; 0 = ONE per-tree row
+// broadcast to every node (the native driver's colsample_bytree path)
+extern "C" void gbt_evaluate_masked(
+    const int64_t* hist, int n_nodes, int n_bins, int n_features,
+    const int32_t* cut_ptrs, const int64_t* parent_sums,
+    const float* maxabs, double g_scale,
+    double h_scale, double reg_lambda, double reg_alpha,
+    double max_delta_step, double min_child_weight, const int8_t* monotone,
+    const double* node_bounds, int mask_stride,
+    const uint8_t* feature_mask,
+    const uint8_t* cat_feature, double* out_gain, int32_t* out_bin,
+    uint8_t* out_dir, int64_t* out_lsum, const int32_t* k_dev,
+    int narrow_max, hipStream_t stream) {
   dim3 grid(n_features > 65535 ? 65535 : n_features, n_nodes);
   hipLaunchKernelGGL(EvaluateKernel, grid, dim3(64), 0, stream, hist, n_nodes,
                      n_bins, n_features, cut_ptrs, parent_sums, maxabs, k_dev,
                      narrow_max, g_scale,
                      h_scale, reg_lambda, reg_alpha, max_delta_step,
-                     min_child_weight, monotone, node_bounds, feature_mask,
+                     min_child_weight, monotone, node_bounds, mask_stride,
+                     feature_mask,
                      cat_feature, out_gain, out_bin, out_dir, out_lsum);
   if (narrow_max > 0) {
     const long long total = (long long)n_nodes * n_features;
@@ -491,7 +517,8 @@ extern "C" void gbt_evaluate(
                        stream, hist, n_nodes, n_bins, n_features, cut_ptrs,
                        parent_sums, maxabs, k_dev, narrow_max, g_scale,
                        h_scale, reg_lambda, reg_alpha, max_delta_step,
-                       min_child_weight, monotone, node_bounds, feature_mask,
+                       min_child_weight, monotone, node_bounds, mask_stride,
+                       feature_mask,
                        cat_feature, out_gain, out_bin, out_dir, out_lsum);
   }
 }

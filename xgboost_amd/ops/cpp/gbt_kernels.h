@@ -66,6 +66,20 @@ void gbt_evaluate(const int64_t* hist /* [n_nodes, n_bins, 2] */,
                   int narrow_max,  // >0: scalar kernel for narrow features
                   hipStream_t stream);
 
+// mask_stride: n_features = per-node rows, 0 = one per-tree row
+void gbt_evaluate_masked(const int64_t* hist, int n_nodes, int n_bins,
+                         int n_features, const int32_t* cut_ptrs,
+                         const int64_t* parent_sums, const float* maxabs,
+                         double g_scale, double h_scale, double reg_lambda,
+                         double reg_alpha, double max_delta_step,
+                         double min_child_weight, const int8_t* monotone,
+                         const double* node_bounds, int mask_stride,
+                         const uint8_t* feature_mask,
+                         const uint8_t* cat_feature, double* out_gain,
+                         int32_t* out_bin, uint8_t* out_dir,
+                         int64_t* out_lsum, const int32_t* k_dev,
+                         int narrow_max, hipStream_t stream);
+
 void gbt_compress(const float* X, int64_t n_rows, int n_features,
                   const float* cut_values, const int32_t* cut_ptrs,
                   const uint8_t* cat_feature, float missing_value,

@@ -50,15 +50,24 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
   const int gf = f_end - f_begin;
 
   // stage per-feature (start bin - bin_begin, n_bins) in LDS so the hot
-  // loop does no global cut_ptrs loads
-  __shared__ int s_start[GBT_HIST_MAX_F];
+  // loop does no global cut_ptrs loads.  On the LDS path the start is
+  // group-relative (< 8192) so it PACKS with the width into one 32-bit
+  // word — one ds_read per (row, feature) instead of two (the hist
+  // kernel is LDS-issue bound; metadata was a third of the traffic)
+  __shared__ unsigned s_meta[GBT_HIST_MAX_F];   // (start << 16) | width
+  __shared__ int s_start[GBT_HIST_MAX_F];       // global path (wide start)
   __shared__ int s_width[GBT_HIST_MAX_F];
   const bool stage_meta = gf <= GBT_HIST_MAX_F;
   if (stage_meta) {
     for (int f = threadIdx.x; f < gf; f += blockDim.x) {
       const int c0 = cut_ptrs[f_begin + f];
-      s_start[f] = c0 - (kUseShared ? bin_begin : 0);
-      s_width[f] = cut_ptrs[f_begin + f + 1] - c0;
+      const int wdt = cut_ptrs[f_begin + f + 1] - c0;
+      if (kUseShared) {
+        s_meta[f] = ((unsigned)(c0 - bin_begin) << 16) | (unsigned)wdt;
+      } else {
+        s_start[f] = c0;
+        s_width[f] = wdt;
+      }
     }
   }
 
@@ -88,12 +97,15 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
     if (stage_meta) {
       for (int f = 0; f < gf; ++f) {
         const int local = (int)rowbins[f];
-        if (local >= s_width[f]) continue;  // missing sentinel
-        const int sbin = s_start[f] + local;
         if (kUseShared) {
+          const unsigned m = s_meta[f];
+          if (local >= (int)(m & 0xFFFFu)) continue;  // missing sentinel
+          const int sbin = (int)(m >> 16) + local;
           atomicAdd(&hist_s[2 * sbin], (unsigned long long)g);
           atomicAdd(&hist_s[2 * sbin + 1], (unsigned long long)h);
         } else {
+          if (local >= s_width[f]) continue;
+          const int sbin = s_start[f] + local;
           atomicAdd((unsigned long long*)&hist_g[2 * sbin],
                     (unsigned long long)g);
           atomicAdd((unsigned long long*)&hist_g[2 * sbin + 1],
