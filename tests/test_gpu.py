@@ -838,3 +838,56 @@ def test_gpu_colsample_bytree_native_matches_python_driver():
         used |= set(int(f) for i, f in enumerate(tg.split_index[:tg.n_nodes])
                     if tg.left[i] != -1)
     assert len(used) <= 12  # sanity; sampling varies per tree
+
+
+def test_distributed_whole_tree_chain_vs_duplicated_data():
+    """End-to-end check of the DISTRIBUTED whole-tree chain without
+    multiple devices: patch the collectives to emulate world=2 with
+    both ranks holding the SAME shard (allreduce-sum doubles buffers,
+    max/broadcast are identity).  With reg_lambda=0, min_child_weight=0
+    the doubled histograms must produce EXACTLY the tree that
+    single-rank training on the row-duplicated dataset produces
+    (int64 sums scale exactly by 2; gains scale uniformly; leaf values
+    -2G/2H == -G/H).  This exercises the padded fixed-count hist and
+    pair-sum reduces, the global-hessian sibling choice and the
+    built-is-left records — the exact code the 8-GPU run drives."""
+    from xgboost_amd import collective as coll
+    X, y = _data(20000, 8, seed=37)
+    pd = {"objective": "binary:logistic", "max_depth": 8, "max_bin": 128,
+          "reg_lambda": 0.0, "min_child_weight": 0.0, "seed": 2,
+          "device": "cuda"}
+    X2 = np.concatenate([X, X])
+    y2 = np.concatenate([y, y])
+    b_ref = xgb.train(pd, xgb.DMatrix(X2, label=y2), 6)
+
+    saved = {k: getattr(coll, k) for k in
+             ("is_distributed", "get_world_size", "get_rank",
+              "allreduce_sum_", "allreduce_max_", "broadcast_obj",
+              "allreduce_sum_scalars", "allreduce_max_scalars",
+              "barrier", "allgather_obj")}
+    try:
+        coll.is_distributed = lambda: True
+        coll.get_world_size = lambda: 2
+        coll.get_rank = lambda: 0
+        coll.allreduce_sum_ = lambda t: t.mul_(2)
+        coll.allreduce_max_ = lambda t: t
+        coll.broadcast_obj = lambda obj, src=0: obj
+        coll.allreduce_sum_scalars = lambda v: [2 * x for x in v]
+        coll.allreduce_max_scalars = lambda v: list(v)
+        coll.barrier = lambda: None
+        coll.allgather_obj = lambda obj: [obj, obj]
+        b_dist = xgb.train(pd, xgb.DMatrix(X, label=y), 6)
+    finally:
+        for k, v in saved.items():
+            setattr(coll, k, v)
+    assert len(b_dist.trees) == len(b_ref.trees)
+    for td, tr in zip(b_dist.trees, b_ref.trees):
+        assert td.n_nodes == tr.n_nodes
+        assert np.array_equal(td.split_index[:td.n_nodes],
+                              tr.split_index[:tr.n_nodes])
+        assert np.array_equal(td.left[:td.n_nodes], tr.left[:tr.n_nodes])
+        assert np.allclose(td.split_cond[:td.n_nodes],
+                           tr.split_cond[:tr.n_nodes], rtol=1e-6, atol=1e-7)
+        # doubled shard hessians == duplicated-data hessians directly
+        assert np.allclose(td.sum_hess[:td.n_nodes],
+                           tr.sum_hess[:tr.n_nodes], rtol=1e-5)
