@@ -851,6 +851,7 @@ def test_distributed_whole_tree_chain_vs_duplicated_data():
     -2G/2H == -G/H).  This exercises the padded fixed-count hist and
     pair-sum reduces, the global-hessian sibling choice and the
     built-is-left records — the exact code the 8-GPU run drives."""
+    import xgboost_amd.sketch as sketch_mod
     from xgboost_amd import collective as coll
     X, y = _data(20000, 8, seed=37)
     pd = {"objective": "binary:logistic", "max_depth": 8, "max_bin": 128,
@@ -858,7 +859,16 @@ def test_distributed_whole_tree_chain_vs_duplicated_data():
           "device": "cuda"}
     X2 = np.concatenate([X, X])
     y2 = np.concatenate([y, y])
-    b_ref = xgb.train(pd, xgb.DMatrix(X2, label=y2), 6)
+    # both runs must bin with IDENTICAL cuts (the emulated world=2 would
+    # otherwise take the summary-merge sketch while the reference run
+    # takes the exact sort — different cut points, different trees)
+    cuts_fixed = make_cuts(X, 128)
+    saved_sketch = sketch_mod.sketch_cuts
+    sketch_mod.sketch_cuts = lambda dmat, mb: cuts_fixed
+    try:
+        b_ref = xgb.train(pd, xgb.DMatrix(X2, label=y2), 6)
+    finally:
+        sketch_mod.sketch_cuts = saved_sketch
 
     saved = {k: getattr(coll, k) for k in
              ("is_distributed", "get_world_size", "get_rank",
@@ -876,8 +886,10 @@ def test_distributed_whole_tree_chain_vs_duplicated_data():
         coll.allreduce_max_scalars = lambda v: list(v)
         coll.barrier = lambda: None
         coll.allgather_obj = lambda obj: [obj, obj]
+        sketch_mod.sketch_cuts = lambda dmat, mb: cuts_fixed
         b_dist = xgb.train(pd, xgb.DMatrix(X, label=y), 6)
     finally:
+        sketch_mod.sketch_cuts = saved_sketch
         for k, v in saved.items():
             setattr(coll, k, v)
     assert len(b_dist.trees) == len(b_ref.trees)
