@@ -37,6 +37,12 @@ class GradQuantizer:
             m = gpair.abs().amax(dim=0)  # one reduce, one D2H sync
             if collective.is_distributed():
                 collective.allreduce_max_(m)
+            if gpair.is_cuda:
+                # keep the (allreduced) device max-abs: the native
+                # whole-tree driver derives the SAME scales on device
+                # (bit-identical formula), which unlocks the one-sync
+                # chain for every non-fused objective
+                self.maxabs_dev = m.contiguous()
             mh = m.cpu()
             max_g, max_h = float(mh[0]), float(mh[1])
         else:
