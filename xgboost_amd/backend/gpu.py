@@ -150,6 +150,16 @@ class GpuOps(SegmentedOpsMixin):
         self.n_groups = len(groups_f) - 1
         self.max_group_bins = int(np.max(np.diff(groups_b)))
         self.use_shared = 1 if self.max_group_bins <= LDS_MAX_GROUP_BINS else 0
+        # hist-kernel mode upgrade (see hist.hip launcher): 2 = the
+        # register-metadata kernel (every group <= 32 features, u8
+        # bins), 3 = + dword-packed bin loads (4-aligned stride/groups)
+        if (self.use_shared and qm.gidx.dtype == torch.uint8
+                and max(groups_f[i + 1] - groups_f[i]
+                        for i in range(self.n_groups)) <= 32):
+            aligned = (qm.n_features % 4 == 0
+                       and all(f % 4 == 0 for f in groups_f[:-1])
+                       and qm.gidx.data_ptr() % 4 == 0)
+            self.use_shared = 3 if aligned else 2
         self._ridx_out: Optional[torch.Tensor] = None
         if cuts.feature_types is not None:
             self.cat_feature = torch.tensor(

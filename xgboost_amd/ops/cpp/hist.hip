@@ -186,31 +186,39 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernelReg(
     const uint8_t* __restrict__ gidx, int n_features,
     const int32_t* __restrict__ qgpair, const int32_t* __restrict__ ridx,
     const BlockTask* __restrict__ tasks,
-    int64_t* __restrict__ out_hist, int n_bins, int gf,
+    int64_t* __restrict__ out_hist, int n_bins,
+    const int32_t* __restrict__ feat_group_start,
+    const int32_t* __restrict__ bin_group_start,
     const int32_t* __restrict__ cut_ptrs,
     int64_t* __restrict__ node_sums) {
   const BlockTask task = tasks[blockIdx.x];
   if (task.row_begin >= task.row_end) return;
+  const int group = blockIdx.y;
+  const int f_begin = feat_group_start[group];
+  const int gf = feat_group_start[group + 1] - f_begin;
+  const int bin_begin = bin_group_start[group];
+  const int group_bins = bin_group_start[group + 1] - bin_begin;
 
-  unsigned meta[kGF];  // (start << 16) | width, start/width < 65536
+  unsigned meta[kGF];  // (group-rel start << 16) | width, both < 65536
 #pragma unroll
   for (int f = 0; f < kGF; ++f) {
     if (f < gf) {
-      const int c0 = cut_ptrs[f];
-      meta[f] = ((unsigned)c0 << 16) | (unsigned)(cut_ptrs[f + 1] - c0);
+      const int c0 = cut_ptrs[f_begin + f];
+      meta[f] = ((unsigned)(c0 - bin_begin) << 16)
+                | (unsigned)(cut_ptrs[f_begin + f + 1] - c0);
     } else {
       meta[f] = 0;
     }
   }
 
   extern __shared__ unsigned long long smem[];
-  for (int i = threadIdx.x; i < n_bins * 2; i += blockDim.x) {
+  for (int i = threadIdx.x; i < group_bins * 2; i += blockDim.x) {
     smem[i] = 0ULL;
   }
   __syncthreads();
 
   int64_t* hist_g = out_hist + (size_t)task.out_slot * n_bins * 2;
-  const bool do_sums = node_sums != nullptr;
+  const bool do_sums = node_sums != nullptr && group == 0;
   long long sum_g = 0, sum_h = 0;
   for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
        i += blockDim.x) {
@@ -221,10 +229,10 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernelReg(
       sum_g += g;
       sum_h += h;
     }
-    const uint8_t* rowbins = gidx + (size_t)row * n_features;
+    const uint8_t* rowbins = gidx + (size_t)row * n_features + f_begin;
     if (kVec) {
       // dword-packed bin loads: 4 features per 32-bit load (launcher
-      // guarantees n_features % 4 == 0 so every row is 4-aligned)
+      // guarantees row stride and f_begin keep rows 4-aligned)
       const unsigned* rowu = (const unsigned*)rowbins;
 #pragma unroll
       for (int f4 = 0; f4 < kGF / 4; ++f4) {
@@ -256,10 +264,10 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernelReg(
   }
 
   __syncthreads();
-  for (int i = threadIdx.x; i < n_bins * 2; i += blockDim.x) {
+  for (int i = threadIdx.x; i < group_bins * 2; i += blockDim.x) {
     const unsigned long long v = smem[i];
     if (v != 0ULL) {
-      atomicAdd((unsigned long long*)&hist_g[i], v);
+      atomicAdd((unsigned long long*)&hist_g[2 * bin_begin + i], v);
     }
   }
   if (do_sums) {
@@ -313,27 +321,29 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
   dim3 grid(n_tasks, n_groups);
   dim3 block(block_size);
   size_t shmem = use_shared ? (size_t)max_group_bins * 2 * sizeof(int64_t) : 0;
-  // register-metadata fast path: one feature group, few features, u8
-  // bins (see HistKernelReg; env GBT_HIST_REG=0 to disable for A/B)
+  // register-metadata fast path (HistKernelReg): selected by the
+  // CALLER through use_shared (it owns the host-side group geometry):
+  //   0 = global-atomic fallback, 1 = generic LDS kernel,
+  //   2 = register-metadata (every group <= 32 features, u8),
+  //   3 = 2 + dword-packed bin loads (stride/group 4-aligned).
+  // Env GBT_HIST_REG=0 forces the generic kernel for A/B.
   static int use_reg = [] {
     const char* e = getenv("GBT_HIST_REG");
     return e ? atoi(e) : 1;
   }();
-  if (use_reg && use_shared && n_groups == 1 && gidx8 != nullptr &&
-      n_features <= 32) {
-    const bool vec = (n_features % 4 == 0) &&
-                     (((uintptr_t)gidx8 & 3u) == 0);
+  if (use_reg && use_shared >= 2 && gidx8 != nullptr) {
+    dim3 rgrid(n_tasks, n_groups);
     dim3 rblock(reg_block);
-    if (vec) {
-      hipLaunchKernelGGL((HistKernelReg<32, true>), dim3(n_tasks), rblock,
+    if (use_shared >= 3) {
+      hipLaunchKernelGGL((HistKernelReg<32, true>), rgrid, rblock,
                          shmem, stream, gidx8, n_features, qgpair, ridx,
-                         tasks, out_hist, n_bins, n_features, cut_ptrs,
-                         node_sums);
+                         tasks, out_hist, n_bins, feat_group_start,
+                         bin_group_start, cut_ptrs, node_sums);
     } else {
-      hipLaunchKernelGGL((HistKernelReg<32, false>), dim3(n_tasks), rblock,
+      hipLaunchKernelGGL((HistKernelReg<32, false>), rgrid, rblock,
                          shmem, stream, gidx8, n_features, qgpair, ridx,
-                         tasks, out_hist, n_bins, n_features, cut_ptrs,
-                         node_sums);
+                         tasks, out_hist, n_bins, feat_group_start,
+                         bin_group_start, cut_ptrs, node_sums);
     }
     return;
   }
