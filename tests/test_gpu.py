@@ -903,3 +903,20 @@ def test_distributed_whole_tree_chain_vs_duplicated_data():
         # doubled shard hessians == duplicated-data hessians directly
         assert np.allclose(td.sum_hess[:td.n_nodes],
                            tr.sum_hess[:tr.n_nodes], rtol=1e-5)
+
+
+def test_gpu_u16_bins_training_matches_cpu():
+    """max_bin > 256 stores u16 local bins — exercises the u16
+    register-metadata hist kernel variant against the CPU oracle."""
+    X, y = _data(30000, 6, seed=43)
+    pd = {"objective": "binary:logistic", "max_depth": 6, "max_bin": 700,
+          "eta": 0.3}
+    bc = xgb.train(pd, xgb.DMatrix(X, label=y), 5)
+    bg = xgb.train({**pd, "device": "cuda"}, xgb.DMatrix(X, label=y), 5)
+    for tg, tc in zip(bg.trees, bc.trees):
+        assert tg.n_nodes == tc.n_nodes
+        assert np.array_equal(tg.split_index[:tg.n_nodes],
+                              tc.split_index[:tc.n_nodes])
+        assert np.array_equal(tg.left[:tg.n_nodes], tc.left[:tc.n_nodes])
+        assert np.allclose(tg.split_cond[:tg.n_nodes],
+                           tc.split_cond[:tc.n_nodes], rtol=1e-6)

@@ -153,11 +153,13 @@ class GpuOps(SegmentedOpsMixin):
         # hist-kernel mode upgrade (see hist.hip launcher): 2 = the
         # register-metadata kernel (every group <= 32 features, u8
         # bins), 3 = + dword-packed bin loads (4-aligned stride/groups)
-        if (self.use_shared and qm.gidx.dtype == torch.uint8
+        esz = qm.gidx.element_size()
+        if (self.use_shared and esz <= 2
                 and max(groups_f[i + 1] - groups_f[i]
                         for i in range(self.n_groups)) <= 32):
-            aligned = (qm.n_features % 4 == 0
-                       and all(f % 4 == 0 for f in groups_f[:-1])
+            lanes = 4 // esz  # features per 32-bit load
+            aligned = (qm.n_features % lanes == 0
+                       and all(f % lanes == 0 for f in groups_f[:-1])
                        and qm.gidx.data_ptr() % 4 == 0)
             self.use_shared = 3 if aligned else 2
         self._ridx_out: Optional[torch.Tensor] = None

@@ -181,9 +181,9 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernel(
 // packed 32-bit loads (4 features per load) instead of 28 byte loads.
 // Atomic count is unchanged — this attacks the ISSUE-slot bound the
 // round-1 PMC ladder identified (bank conflicts were only ~6%).
-template <int kGF, bool kVec>
+template <typename BinT, int kGF, bool kVec>
 __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernelReg(
-    const uint8_t* __restrict__ gidx, int n_features,
+    const BinT* __restrict__ gidx, int n_features,
     const int32_t* __restrict__ qgpair, const int32_t* __restrict__ ridx,
     const BlockTask* __restrict__ tasks,
     int64_t* __restrict__ out_hist, int n_bins,
@@ -229,20 +229,22 @@ __global__ __launch_bounds__(GBT_HIST_BLOCK) void HistKernelReg(
       sum_g += g;
       sum_h += h;
     }
-    const uint8_t* rowbins = gidx + (size_t)row * n_features + f_begin;
+    const BinT* rowbins = gidx + (size_t)row * n_features + f_begin;
     if (kVec) {
-      // dword-packed bin loads: 4 features per 32-bit load (launcher
-      // guarantees row stride and f_begin keep rows 4-aligned)
+      // dword-packed bin loads: 4 (u8) / 2 (u16) features per 32-bit
+      // load (launcher guarantees rows stay 4-aligned)
+      constexpr int kLanes = 4 / (int)sizeof(BinT);
       const unsigned* rowu = (const unsigned*)rowbins;
 #pragma unroll
-      for (int f4 = 0; f4 < kGF / 4; ++f4) {
-        if (4 * f4 >= gf) break;
+      for (int f4 = 0; f4 < kGF / kLanes; ++f4) {
+        if (kLanes * f4 >= gf) break;
         const unsigned packed = rowu[f4];
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int f = 4 * f4 + j;
+        for (int j = 0; j < kLanes; ++j) {
+          const int f = kLanes * f4 + j;
           const unsigned m = meta[f];
-          const int local = (int)((packed >> (8 * j)) & 0xFFu);
+          const int local = (int)((packed >> (8 * (int)sizeof(BinT) * j))
+                                  & (sizeof(BinT) == 1 ? 0xFFu : 0xFFFFu));
           if (local >= (int)(m & 0xFFFFu)) continue;
           const int sbin = (int)(m >> 16) + local;
           atomicAdd(&smem[2 * sbin], (unsigned long long)g);
@@ -331,19 +333,35 @@ extern "C" void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
     const char* e = getenv("GBT_HIST_REG");
     return e ? atoi(e) : 1;
   }();
-  if (use_reg && use_shared >= 2 && gidx8 != nullptr) {
+  if (use_reg && use_shared >= 2) {
     dim3 rgrid(n_tasks, n_groups);
     dim3 rblock(reg_block);
-    if (use_shared >= 3) {
-      hipLaunchKernelGGL((HistKernelReg<32, true>), rgrid, rblock,
-                         shmem, stream, gidx8, n_features, qgpair, ridx,
-                         tasks, out_hist, n_bins, feat_group_start,
-                         bin_group_start, cut_ptrs, node_sums);
+    if (gidx8 != nullptr) {
+      if (use_shared >= 3) {
+        hipLaunchKernelGGL((HistKernelReg<uint8_t, 32, true>), rgrid,
+                           rblock, shmem, stream, gidx8, n_features, qgpair,
+                           ridx, tasks, out_hist, n_bins, feat_group_start,
+                           bin_group_start, cut_ptrs, node_sums);
+      } else {
+        hipLaunchKernelGGL((HistKernelReg<uint8_t, 32, false>), rgrid,
+                           rblock, shmem, stream, gidx8, n_features, qgpair,
+                           ridx, tasks, out_hist, n_bins, feat_group_start,
+                           bin_group_start, cut_ptrs, node_sums);
+      }
     } else {
-      hipLaunchKernelGGL((HistKernelReg<32, false>), rgrid, rblock,
-                         shmem, stream, gidx8, n_features, qgpair, ridx,
-                         tasks, out_hist, n_bins, feat_group_start,
-                         bin_group_start, cut_ptrs, node_sums);
+      if (use_shared >= 3) {
+        hipLaunchKernelGGL((HistKernelReg<uint16_t, 32, true>), rgrid,
+                           rblock, shmem, stream, gidx16, n_features,
+                           qgpair, ridx, tasks, out_hist, n_bins,
+                           feat_group_start, bin_group_start, cut_ptrs,
+                           node_sums);
+      } else {
+        hipLaunchKernelGGL((HistKernelReg<uint16_t, 32, false>), rgrid,
+                           rblock, shmem, stream, gidx16, n_features,
+                           qgpair, ridx, tasks, out_hist, n_bins,
+                           feat_group_start, bin_group_start, cut_ptrs,
+                           node_sums);
+      }
     }
     return;
   }
