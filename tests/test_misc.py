@@ -376,3 +376,26 @@ def test_early_stopping_last_dataset_last_metric():
     assert bst.best_iteration is not None
     assert len(res["valid"]["auc"]) < 60
     assert "logloss" in res["valid"] and "auc" in res["valid"]
+
+
+def test_training_observer(capfd, monkeypatch):
+    """TrainingObserver (reference src/common/observer.h): env-gated
+    per-iteration gradient/tree/prediction dumps."""
+    import numpy as np
+    import xgboost_amd as xgb
+    from xgboost_amd.monitor import TrainingObserver
+    monkeypatch.setenv("XGB_AMD_OBSERVER", "1")
+    TrainingObserver._enabled = None  # re-read env
+    try:
+        rng = np.random.RandomState(0)
+        X = rng.randn(300, 4).astype(np.float32)
+        y = (X[:, 0] > 0).astype(np.float32)
+        xgb.train({"objective": "binary:logistic", "max_depth": 3},
+                  xgb.DMatrix(X, label=y), 2, verbose_eval=False)
+    finally:
+        TrainingObserver._enabled = None
+        monkeypatch.delenv("XGB_AMD_OBSERVER")
+    out, _ = capfd.readouterr()
+    assert "[observer]" in out
+    assert "grad:" in out and "tree:" in out and "margin:" in out
+    TrainingObserver._enabled = None

@@ -242,7 +242,12 @@ class Booster:
             self._update_existing(dtrain, iteration)
             return
         margin = self._cached_margin(dtrain)
+        from .monitor import TrainingObserver
+        if TrainingObserver.enabled():
+            TrainingObserver.observe_predictions(iteration, margin)
         if fobj is None and self._boost_fused(dtrain, margin, iteration):
+            if TrainingObserver.enabled():
+                TrainingObserver.observe_tree(iteration, self.trees[-1])
             return
         if fobj is not None:
             preds = self.objective.pred_transform(margin).cpu().numpy()
@@ -253,6 +258,7 @@ class Booster:
                                    device=margin.device).view(margin.shape)
         else:
             grad, hess = self.objective.get_gradient(margin, dtrain.info, iteration)
+            TrainingObserver.observe_gradient(iteration, grad, hess)
         self.boost_gpair(dtrain, grad, hess, iteration)
 
     def boost(self, dtrain: DMatrix, iteration: int = 0, grad=None, hess=None) -> None:
@@ -323,6 +329,9 @@ class Booster:
                 margin[:, k] += leaf_vals[positions.to(margin.device).long()]
         self.iteration_indptr.append(self.iteration_indptr[-1] + new_trees)
         self._cache[id(dtrain)] = (margin, len(self.trees))
+        from .monitor import TrainingObserver
+        if TrainingObserver.enabled() and self.trees:
+            TrainingObserver.observe_tree(iteration, self.trees[-1])
         if self.tparam.debug_synchronize:
             collective.check_synchronized(
                 json.dumps(self.trees[-1].to_json()).encode(), "tree")

@@ -72,3 +72,52 @@ class Monitor:
 def _report_all():
     for m in Monitor._instances:
         m.maybe_print()
+
+
+class TrainingObserver:
+    """Debug observer (reference src/common/observer.h:38
+    TrainingObserver): dumps per-iteration gradient statistics and the
+    freshly committed tree when enabled.  The reference gates it at
+    compile time; here it is runtime-gated by the XGB_AMD_OBSERVER env
+    var or config_context(observer=True), so production pays a single
+    attribute check."""
+
+    _enabled = None
+
+    @classmethod
+    def enabled(cls) -> bool:
+        if cls._enabled is None:
+            import os
+            cls._enabled = os.environ.get("XGB_AMD_OBSERVER", "0") not in (
+                "0", "", "false")
+        return cls._enabled
+
+    @classmethod
+    def observe_gradient(cls, iteration: int, grad, hess) -> None:
+        if not cls.enabled():
+            return
+        g = grad.detach()
+        h = hess.detach()
+        print(f"[observer] it={iteration} grad: n={g.numel()} "
+              f"mean={float(g.mean()):+.6g} absmax={float(g.abs().max()):.6g}"
+              f" | hess: mean={float(h.mean()):+.6g} "
+              f"min={float(h.min()):.6g}", flush=True)
+
+    @classmethod
+    def observe_tree(cls, iteration: int, tree) -> None:
+        if not cls.enabled():
+            return
+        n = tree.n_nodes
+        leaves = int((tree.left[:n] == -1).sum())
+        print(f"[observer] it={iteration} tree: nodes={n} leaves={leaves} "
+              f"max_gain={float(tree.loss_chg[:n].max()):.6g} "
+              f"root_hess={float(tree.sum_hess[0]):.6g}", flush=True)
+
+    @classmethod
+    def observe_predictions(cls, iteration: int, margin) -> None:
+        if not cls.enabled():
+            return
+        m = margin.detach()
+        print(f"[observer] it={iteration} margin: mean="
+              f"{float(m.mean()):+.6g} std={float(m.std()):.6g}",
+              flush=True)
