@@ -172,6 +172,11 @@ class GpuOps(SegmentedOpsMixin):
         g8 = qm.gidx.dtype == torch.uint8
         self._gidx8 = qm.gidx if g8 else None
         self._gidx16 = None if g8 else qm.gidx
+        # feature-major copy for the partition/leaf-decide kernels:
+        # they read ONE feature per node, so the column layout keeps
+        # their gathers inside a single cache-resident column instead
+        # of touching a 64-byte line per row across the whole matrix
+        self._gidx_T = qm.gidx.t().contiguous()
         self.stager = _PinnedStager(dev)
 
     # ------------------------------------------------------------------
@@ -181,10 +186,22 @@ class GpuOps(SegmentedOpsMixin):
         g8 = gidx.dtype == torch.uint8
         self._gidx8 = gidx if g8 else None
         self._gidx16 = None if g8 else gidx
+        self._gidx_T = None  # streamed pages: transposing every sweep
+        # would cost more than the column gathers save
         self.qm = QuantizedMatrix(gidx, self.qm.cuts, self.qm.has_missing)
 
     def _gidx_ptrs(self):
         return self.hip.ptr(self._gidx8), self.hip.ptr(self._gidx16)
+
+    def _gidx_col_args(self):
+        """(col8_ptr, col16_ptr, leading dim) for the kernels that take
+        the optional feature-major copy."""
+        t = self._gidx_T
+        if t is None:
+            return None, None, 0
+        if t.dtype == torch.uint8:
+            return self.hip.ptr(t), None, t.shape[1]
+        return None, self.hip.ptr(t), t.shape[1]
 
     def make_ridx(self, n_rows: int) -> torch.Tensor:
         self._ridx_out = torch.empty(n_rows, dtype=torch.int32,
@@ -511,8 +528,9 @@ class GpuOps(SegmentedOpsMixin):
             self.stager.upload([tasks_np, feat, sbin, dleft, counters,
                                 cat_bits_np, cat_off_np])
         p8, p16 = self._gidx_ptrs()
+        c8, c16, cld = self._gidx_col_args()
         self.lib.gbt_partition(
-            p8, p16, self.qm.n_features, self.hip.ptr(ridx),
+            p8, p16, self.qm.n_features, c8, c16, cld, self.hip.ptr(ridx),
             self.hip.ptr(self._ridx_out), self.hip.ptr(tasks), len(tasks_np),
             self.hip.ptr(feat_t), self.hip.ptr(sbin_t), self.hip.ptr(dleft_t),
             self.hip.ptr(cat_bits_t), self.hip.ptr(cat_off_t),
@@ -652,12 +670,13 @@ class GpuOps(SegmentedOpsMixin):
 
             cb = self.hip.ALLREDUCE_FN(_allreduce)
         p8, p16 = self._gidx_ptrs()
+        c8, c16, cld = self._gidx_col_args()
         ma = getattr(quantizer, "maxabs_dev", None)
         out_scales = np.zeros(2, dtype=np.float64)
         gsc = quantizer.g_scale if ma is None else 0.0
         hsc = quantizer.h_scale if ma is None else 0.0
         rc = self.lib.gbt_grow_tree(
-            ws["driver"], p8, p16, self.qm.n_features, n_rows,
+            ws["driver"], p8, p16, self.qm.n_features, c8, c16, n_rows,
             self.hip.ptr(qgpair),
             self.hip.ptr(self.cut_ptrs),
             cut_values_host.ctypes.data_as(ctypes.c_void_p),

@@ -29,13 +29,16 @@ _d = _c.c_double
 
 _SIGS = {
     "gbt_hist": [_p, _p, _i, _p, _p, _p, _i, _p, _i, _p, _p, _i, _i, _p, _i, _p, _p],
-    "gbt_partition": [_p, _p, _i, _p, _p, _p, _i, _p, _p, _p, _p, _p, _p, _p, _p],
+    "gbt_partition": [_p, _p, _i, _p, _p, _i64, _p, _p, _p, _i, _p, _p, _p,
+                      _p, _p, _p, _p, _p],
     "gbt_evaluate": [_p, _i, _i, _i, _p, _p, _p, _d, _d, _d, _d, _d, _d,
                      _p, _p, _p, _p, _p, _p, _p, _p, _p, _i, _p],
     "gbt_compress": [_p, _i64, _i, _p, _p, _p, _f, _i, _p, _p, _p],
     "gbt_predict": [_p, _i64, _i, _f, _i, _p, _p, _p, _p, _p, _p, _p, _p,
                     _p, _p, _i, _i, _p, _p, _p],
     "gbt_leaf_partition": [_p, _p, _i, _p, _p, _p],
+    "gbt_leaf_decide": [_p, _p, _i, _p, _p, _i64, _p, _p, _i, _p, _p, _p,
+                        _p, _p, _p, _p],
     "gbt_copy_ranges": [_p, _p, _p, _i, _p],
     "gbt_select_best": [_p, _p, _p, _p, _i, _i, _p, _p, _p],
     "gbt_mt_evaluate": [_p, _i, _i, _i, _i, _p, _p, _p, _p,
@@ -55,7 +58,9 @@ _SIGS = {
     "gbt_driver_destroy": [_p],
     "gbt_grow_tree": [
         _p,                      # ctx
-        _p, _p, _i, _i64, _p,    # gidx8/16, n_features, n_rows, qgpair
+        _p, _p, _i,              # gidx8/16, n_features
+        _p, _p,                  # feature-major gidx copies (or null)
+        _i64, _p,                # n_rows, qgpair
         _p, _p, _p, _p,          # cut_ptrs_dev, cut_values_host, cut_ptrs_host, n_bins_feat_dev
         _p, _p, _i, _i, _i, _i,  # groups, n_groups, max_group_bins, use_shared, n_bins
         _p, _p, _p, _p,          # ridx, ridx_out, pool_a, pool_b

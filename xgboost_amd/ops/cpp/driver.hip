@@ -513,6 +513,7 @@ typedef void (*AllreduceFn)(long long* dev_ptr, long long n_elems);
 // Returns n_nodes (>0) on success, negative on failure.
 int gbt_grow_tree(
     void* vctx, const uint8_t* gidx8, const uint16_t* gidx16, int n_features,
+    const uint8_t* gidx8_col, const uint16_t* gidx16_col,  // feature-major
     long long n_rows, const int32_t* qgpair, const int32_t* cut_ptrs_dev,
     const float* cut_values_host, const int32_t* cut_ptrs_host,
     const int32_t* n_bins_feat_dev, const int32_t* feat_group_start_dev,
@@ -836,7 +837,8 @@ int gbt_grow_tree(
                          (L == 0 || !has_mono) ? nullptr : bnd_bufs[(L - 1) & 1],
                          has_mono ? bnd_bufs[L & 1] : nullptr,
                          L == 0 ? nullptr : d_mode + (size_t)(L - 1) * pool);
-      gbt_partition(gidx8, gidx16, n_features, cur_ridx, alt_ridx, d_pt,
+      gbt_partition(gidx8, gidx16, n_features, gidx8_col, gidx16_col,
+                    n_rows, cur_ridx, alt_ridx, d_pt,
                     wt_max_ptasks, d_feat, d_sbin, d_dl, nullptr, nullptr,
                     n_bins_feat_dev, d_cnt, stream);
       std::swap(cur_ridx, alt_ridx);
@@ -1033,7 +1035,8 @@ int gbt_grow_tree(
         HIP_CHECK(hipMemcpyAsync(ctx->ring.dev[slot], h, bytes,
                                  hipMemcpyHostToDevice, stream));
         char* d = (char*)ctx->ring.dev[slot];
-        gbt_leaf_decide(gidx8, gidx16, n_features, cur_ridx,
+        gbt_leaf_decide(gidx8, gidx16, n_features, gidx8_col, gidx16_col,
+                        n_rows, cur_ridx,
                         (const BlockTask*)d, (int)tasks.size(),
                         (const int32_t*)(d + offf),
                         (const int32_t*)(d + offs),
@@ -1318,7 +1321,8 @@ int gbt_grow_tree(
       // final level: the children are all leaves — write their
       // positions directly (one decide pass) instead of
       // partition + copy + counter sync + a later leaf sweep
-      gbt_leaf_decide(gidx8, gidx16, n_features, cur_ridx,
+      gbt_leaf_decide(gidx8, gidx16, n_features, gidx8_col, gidx16_col,
+                      n_rows, cur_ridx,
                       (const BlockTask*)d,
                       (int)ptasks.size(), (const int32_t*)(d + off_feat),
                       (const int32_t*)(d + off_sbin),
@@ -1328,7 +1332,8 @@ int gbt_grow_tree(
       level_nodes.clear();
       break;
     }
-    gbt_partition(gidx8, gidx16, n_features, cur_ridx, alt_ridx,
+    gbt_partition(gidx8, gidx16, n_features, gidx8_col, gidx16_col,
+                  n_rows, cur_ridx, alt_ridx,
                   (const BlockTask*)d, (int)ptasks.size(),
                   (const int32_t*)(d + off_feat),
                   (const int32_t*)(d + off_sbin),

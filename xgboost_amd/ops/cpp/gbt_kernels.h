@@ -35,7 +35,11 @@ void gbt_hist(const uint8_t* gidx8, const uint16_t* gidx16,
               hipStream_t stream);
 
 void gbt_partition(const uint8_t* gidx8, const uint16_t* gidx16,
-                   int n_features, const int32_t* ridx_in, int32_t* ridx_out,
+                   int n_features,
+                   const uint8_t* gidx8_col,   // optional feature-major
+                   const uint16_t* gidx16_col, // copy ([F][col_ld])
+                   int64_t col_ld,
+                   const int32_t* ridx_in, int32_t* ridx_out,
                    const BlockTask* tasks, int n_tasks,
                    const int32_t* split_feature,   // [n_slots]
                    const int32_t* split_bin_local, // [n_slots] (-1: cat)
@@ -102,7 +106,10 @@ void gbt_predict(const float* X, int64_t n_rows, int n_features,
                  hipStream_t stream);
 
 void gbt_leaf_decide(const uint8_t* gidx8, const uint16_t* gidx16,
-                     int n_features, const int32_t* ridx,
+                     int n_features,
+                     const uint8_t* gidx8_col, const uint16_t* gidx16_col,
+                     int64_t col_ld,
+                     const int32_t* ridx,
                      const BlockTask* tasks, int n_tasks,
                      const int32_t* split_feature,
                      const int32_t* split_bin_local,

@@ -35,9 +35,15 @@ __device__ __forceinline__ bool DecideLeft(int local_bin, int fbins,
   return local_bin <= split_bin_local;
 }
 
+// gidx_col: optional FEATURE-MAJOR copy of the bin matrix ([F][ld]).
+// The partition touches ONE feature per node, so the column layout
+// turns its per-row gather from a random 64-byte line across the whole
+// row-major matrix (LLC-bandwidth bound, ~4x the roofline) into
+// accesses confined to a single n_rows-byte column that sits in cache.
 template <typename BinT>
 __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
     const BinT* __restrict__ gidx, int n_features,
+    const BinT* __restrict__ gidx_col, long long col_ld,
     const int32_t* __restrict__ ridx_in, int32_t* __restrict__ ridx_out,
     const BlockTask* __restrict__ tasks,
     const int32_t* __restrict__ split_feature,
@@ -80,11 +86,14 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
     __syncthreads();
   }
 
+  const BinT* col = gidx_col ? gidx_col + (size_t)feature * col_ld
+                             : nullptr;
   // phase A: count left in my contiguous sub-range
   int my_left = 0;
   for (int i = my_begin; i < my_end; ++i) {
     const int row = ridx_in[i];
-    const int local = (int)gidx[(size_t)row * n_features + feature];
+    const int local = col ? (int)col[row]
+                          : (int)gidx[(size_t)row * n_features + feature];
     const bool left = DecideLeft(local, fbins, sbin, dleft, cats, cat_words);
     my_left += left ? 1 : 0;
     if (use_cache) {
@@ -142,7 +151,8 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
       const int k = i - task.row_begin;
       left = (decide_bits[k >> 5] >> (k & 31)) & 1u;
     } else {
-      const int local = (int)gidx[(size_t)row * n_features + feature];
+      const int local = col ? (int)col[row]
+                            : (int)gidx[(size_t)row * n_features + feature];
       left = DecideLeft(local, fbins, sbin, dleft, cats, cat_words);
     }
     if (left) {
@@ -155,6 +165,7 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void PartitionKernel(
 
 extern "C" void gbt_partition(
     const uint8_t* gidx8, const uint16_t* gidx16, int n_features,
+    const uint8_t* gidx8_col, const uint16_t* gidx16_col, int64_t col_ld,
     const int32_t* ridx_in, int32_t* ridx_out, const BlockTask* tasks,
     int n_tasks, const int32_t* split_feature, const int32_t* split_bin_local,
     const uint8_t* default_left, const uint32_t* cat_bits,
@@ -163,12 +174,14 @@ extern "C" void gbt_partition(
   if (gidx8 != nullptr) {
     hipLaunchKernelGGL((PartitionKernel<uint8_t>), dim3(n_tasks),
                        dim3(GBT_PART_BLOCK), 0, stream, gidx8, n_features,
+                       gidx8_col, (long long)col_ld,
                        ridx_in, ridx_out, tasks, split_feature,
                        split_bin_local, default_left, cat_bits,
                        cat_bits_offset, n_bins_feat, counters);
   } else {
     hipLaunchKernelGGL((PartitionKernel<uint16_t>), dim3(n_tasks),
                        dim3(GBT_PART_BLOCK), 0, stream, gidx16, n_features,
+                       gidx16_col, (long long)col_ld,
                        ridx_in, ridx_out, tasks, split_feature,
                        split_bin_local, default_left, cat_bits,
                        cat_bits_offset, n_bins_feat, counters);
@@ -208,6 +221,7 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void LeafPartitionKernel(
 template <typename BinT>
 __global__ __launch_bounds__(GBT_PART_BLOCK) void LeafDecideKernel(
     const BinT* __restrict__ gidx, int n_features,
+    const BinT* __restrict__ gidx_col, long long col_ld,
     const int32_t* __restrict__ ridx, const BlockTask* __restrict__ tasks,
     const int32_t* __restrict__ split_feature,
     const int32_t* __restrict__ split_bin_local,
@@ -222,17 +236,23 @@ __global__ __launch_bounds__(GBT_PART_BLOCK) void LeafDecideKernel(
   const bool dleft = default_left[slot] != 0;
   const int fbins = n_bins_feat[feature];
   const int lnid = kids[2 * slot], rnid = kids[2 * slot + 1];
+  const BinT* col = gidx_col ? gidx_col + (size_t)feature * col_ld
+                             : nullptr;
   for (int i = task.row_begin + (int)threadIdx.x; i < task.row_end;
        i += blockDim.x) {
     const int row = ridx[i];
-    const int local = (int)gidx[(size_t)row * n_features + feature];
+    const int local = col ? (int)col[row]
+                          : (int)gidx[(size_t)row * n_features + feature];
     const bool left = DecideLeft(local, fbins, sbin, dleft, nullptr, 0);
     out_pos[row] = left ? lnid : rnid;
   }
 }
 
 extern "C" void gbt_leaf_decide(const uint8_t* gidx8, const uint16_t* gidx16,
-                                int n_features, const int32_t* ridx,
+                                int n_features,
+                                const uint8_t* gidx8_col,
+                                const uint16_t* gidx16_col, int64_t col_ld,
+                                const int32_t* ridx,
                                 const BlockTask* tasks, int n_tasks,
                                 const int32_t* split_feature,
                                 const int32_t* split_bin_local,
@@ -243,11 +263,13 @@ extern "C" void gbt_leaf_decide(const uint8_t* gidx8, const uint16_t* gidx16,
   if (gidx8 != nullptr) {
     hipLaunchKernelGGL((LeafDecideKernel<uint8_t>), dim3(n_tasks),
                        dim3(GBT_PART_BLOCK), 0, stream, gidx8, n_features,
+                       gidx8_col, (long long)col_ld,
                        ridx, tasks, split_feature, split_bin_local,
                        default_left, kids, n_bins_feat, out_pos);
   } else {
     hipLaunchKernelGGL((LeafDecideKernel<uint16_t>), dim3(n_tasks),
                        dim3(GBT_PART_BLOCK), 0, stream, gidx16, n_features,
+                       gidx16_col, (long long)col_ld,
                        ridx, tasks, split_feature, split_bin_local,
                        default_left, kids, n_bins_feat, out_pos);
   }
