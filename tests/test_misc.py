@@ -399,3 +399,25 @@ def test_training_observer(capfd, monkeypatch):
     assert "[observer]" in out
     assert "grad:" in out and "tree:" in out and "margin:" in out
     TrainingObserver._enabled = None
+
+
+def test_bench_contract_cpu():
+    """bench.py is the driver's contract: default flags must finish
+    quickly and print one JSON line with the required fields."""
+    import json
+    import os
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "3", "--warmup", "1"],
+        capture_output=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr.decode()
+    line = out.stdout.decode().strip().splitlines()[-1]
+    d = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in d, k
+    assert d["metric"] == "boosting_rounds_per_sec"
+    assert d["config"]["parallelism"] == "dp1"
