@@ -49,33 +49,32 @@ def _flat_finite_columns(X: torch.Tensor, weights: Optional[torch.Tensor],
     n, f = X.shape
     if not np.isnan(missing):
         X = torch.where((X == missing), torch.full_like(X, float("nan")), X)
-    sorted_vals, order = torch.sort(X, dim=0)  # NaNs sort last
+    # sort along the CONTIGUOUS last dim of the transpose (the fast
+    # segmented-radix path; dim=0 sorting measured ~2.5x slower), and
+    # the flat per-feature layout falls straight out of the transpose
+    XT = X.t().contiguous()  # [F, n]
+    sorted_vals, order = torch.sort(XT, dim=-1)  # NaNs sort last
     # parity with the CPU oracle (quantile.make_cuts): only NaN/missing
     # is skipped; +-inf participates like any value
-    cnt = (~torch.isnan(sorted_vals)).sum(dim=0).long()  # [F]
-    if weights is not None:
-        w_sorted = weights.to(torch.float64)[order]
-    else:
-        w_sorted = None
+    cnt = (~torch.isnan(sorted_vals)).sum(dim=-1).long()  # [F]
+    w_sorted = (weights.to(torch.float64)[order]
+                if weights is not None else None)
     total = int(cnt.sum())
     dev = X.device
     if total == 0:
         return (torch.zeros(0, dtype=torch.float32, device=dev),
                 torch.zeros(0, dtype=torch.float64, device=dev), cnt)
     if total == n * f:
-        # no missing anywhere: the flat segmented layout is just the
-        # column-major transpose (skips two 2-D gather passes — the
-        # dominant cost of the per-tree approx regen on dense data)
-        vals = sorted_vals.t().reshape(-1).float()
-        w = (w_sorted.t().reshape(-1) if w_sorted is not None
+        vals = sorted_vals.reshape(-1).float()
+        w = (w_sorted.reshape(-1) if w_sorted is not None
              else torch.ones(total, dtype=torch.float64, device=dev))
         return vals, w, cnt
     col_ptr = torch.zeros(f + 1, dtype=torch.long, device=dev)
     torch.cumsum(cnt, 0, out=col_ptr[1:])
     f_ids = torch.repeat_interleave(torch.arange(f, device=dev), cnt)
     i_ids = torch.arange(total, device=dev) - col_ptr[f_ids]
-    vals = sorted_vals[i_ids, f_ids].float()
-    w = (w_sorted[i_ids, f_ids] if w_sorted is not None
+    vals = sorted_vals[f_ids, i_ids].float()
+    w = (w_sorted[f_ids, i_ids] if w_sorted is not None
          else torch.ones(total, dtype=torch.float64, device=dev))
     return vals, w, cnt
 
