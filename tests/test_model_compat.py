@@ -178,3 +178,70 @@ def test_dump_model_matches_reference_schema(tmp_path):
     assert len(dumps) == 3
     for d in dumps:
         _validate_dump_node(json.loads(d))
+
+
+def _ubj(value) -> bytes:
+    """Minimal independent UBJSON (draft-12) encoder for the fixture —
+    deliberately NOT xgboost_amd.ubjson, so this exercises our READER
+    against bytes produced by foreign code.  Uses plain (un-optimized)
+    containers plus one optimized typed array to cover both forms."""
+    import struct
+    out = bytearray()
+
+    def emit(v):
+        if isinstance(v, dict):
+            out.append(ord("{"))
+            for k, vv in v.items():
+                kb = k.encode()
+                out.append(ord("i"))
+                out.append(len(kb))
+                out.extend(kb)
+                emit(vv)
+            out.append(ord("}"))
+        elif isinstance(v, list):
+            if v and all(isinstance(x, float) for x in v):
+                # optimized float64 typed array: [$D#i<n> payload
+                out.extend(b"[$D#")
+                out.append(ord("i"))
+                out.append(len(v))
+                for x in v:
+                    out.extend(struct.pack(">d", x))
+            else:
+                out.append(ord("["))
+                for x in v:
+                    emit(x)
+                out.append(ord("]"))
+        elif isinstance(v, bool):
+            out.append(ord("T" if v else "F"))
+        elif isinstance(v, int):
+            out.append(ord("l"))
+            out.extend(struct.pack(">i", v))
+        elif isinstance(v, float):
+            out.append(ord("D"))
+            out.extend(struct.pack(">d", v))
+        elif isinstance(v, str):
+            b = v.encode()
+            out.append(ord("S"))
+            out.append(ord("i"))
+            out.append(len(b))
+            out.extend(b)
+        else:
+            raise TypeError(type(v))
+
+    emit(value)
+    return bytes(out)
+
+
+def test_load_foreign_ubjson_model(tmp_path):
+    """A UBJSON model encoded by an INDEPENDENT encoder (typed arrays +
+    plain containers, big-endian draft-12) must load and predict like
+    its JSON twin."""
+    path = str(tmp_path / "ref_model.ubj")
+    with open(path, "wb") as fh:
+        fh.write(_ubj(REFERENCE_STYLE_MODEL))
+    bst = xgb.Booster(model_file=path)
+    assert bst.num_boosted_rounds() == 2
+    X = np.array([[0.0, 9.9], [1.0, -1.0]], np.float32)
+    pred = bst.predict(xgb.DMatrix(X))
+    exp = [1 / (1 + math.exp(-(m + 0.05))) for m in (-0.2, 0.3)]
+    assert np.allclose(pred, exp, atol=1e-6), (pred, exp)
