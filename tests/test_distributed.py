@@ -579,3 +579,42 @@ def test_collective_failure_detection():
         assert procs[1].returncode == 17
         assert procs[0].returncode == 0, out0.decode()
         assert b"DETECTED" in out0, out0.decode()
+
+
+RANK_WORKER = r"""
+import os, sys
+import numpy as np
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+collective.init("gloo")
+rank = collective.get_rank()
+world = collective.get_world_size()
+rng = np.random.RandomState(3)
+n, f = 2000, 6
+X = rng.randn(n, f).astype(np.float32)
+y = np.clip((X[:, 0] * 2 + 2).astype(int), 0, 4).astype(np.float32)
+qid = np.repeat(np.arange(40), n // 40)
+# shard whole GROUPS per rank (ranking groups never straddle workers)
+gmask = (qid % world) == rank
+d = xgb.DMatrix(X[gmask], label=y[gmask], qid=qid[gmask])
+bst = xgb.train({"objective": "rank:ndcg", "max_depth": 4, "eta": 0.3,
+                 "debug_synchronize": True}, d, 5, verbose_eval=False)
+raw = bytes(bst.save_raw("json"))
+ref = collective.broadcast_obj(raw, 0)
+assert raw == ref, "ranking model differs across workers"
+if rank == 0:
+    import pickle
+    with open(os.environ["XGB_AMD_OUT"], "wb") as fh:
+        pickle.dump({"ok": True}, fh)
+import torch.distributed as dist
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_two_process_ranking_training():
+    """Group-sharded lambdarank: gradients are group-local, histogram
+    sync must still produce identical trees on every worker."""
+    _run_workers(2, RANK_WORKER)

@@ -113,7 +113,7 @@ from .cpu import SegmentedOpsMixin
 
 
 class GpuOps(SegmentedOpsMixin):
-    def __init__(self, qm: QuantizedMatrix):
+    def __init__(self, qm: QuantizedMatrix, col_copy: bool = True):
         from .. import ops as hip_ops
         if not torch.cuda.is_available():
             raise RuntimeError("GpuOps requires a GPU")
@@ -175,8 +175,10 @@ class GpuOps(SegmentedOpsMixin):
         # feature-major copy for the partition/leaf-decide kernels:
         # they read ONE feature per node, so the column layout keeps
         # their gathers inside a single cache-resident column instead
-        # of touching a 64-byte line per row across the whole matrix
-        self._gidx_T = qm.gidx.t().contiguous()
+        # of touching a 64-byte line per row across the whole matrix.
+        # (External-memory page caches skip it: doubling every cached
+        # page would blow the device budget for a secondary kernel.)
+        self._gidx_T = qm.gidx.t().contiguous() if col_copy else None
         self.stager = _PinnedStager(dev)
 
     # ------------------------------------------------------------------

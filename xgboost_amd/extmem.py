@@ -330,7 +330,8 @@ class ExtMemOps:
                 size = store.page_bytes(i)
                 if size <= budget and not disk:
                     budget -= size
-                    self.page_ops.append(GpuOps(qm.to(device)))
+                    self.page_ops.append(GpuOps(qm.to(device),
+                                                col_copy=False))
                 else:
                     if not hasattr(self, "_copy_stream"):
                         self._copy_stream = torch.cuda.Stream()
