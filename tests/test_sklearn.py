@@ -162,3 +162,21 @@ def test_sklearn_extended_surface():
     r.fit(X, yr, qid=q, verbose=False)
     s = r.score(X, yr, qid=q)
     assert 0.0 <= s <= 1.0
+
+
+def test_sklearn_callable_objective():
+    """reference _objective_decorator: a callable objective on the
+    sklearn wrapper takes (y_true, y_pred) -> (grad, hess)."""
+    rng = np.random.RandomState(5)
+    X = rng.randn(800, 5).astype(np.float32)
+    y = (X[:, 0] * 2 + rng.randn(800) * 0.1).astype(np.float32)
+
+    def sq_err(y_true, y_pred):
+        return (y_pred - y_true), np.ones_like(y_true)
+
+    m1 = xgb.XGBRegressor(n_estimators=8, max_depth=4,
+                          objective=sq_err).fit(X, y)
+    m2 = xgb.XGBRegressor(n_estimators=8, max_depth=4,
+                          objective="reg:squarederror").fit(X, y)
+    p1, p2 = m1.predict(X), m2.predict(X)
+    assert np.allclose(p1, p2, atol=1e-5), np.abs(p1 - p2).max()

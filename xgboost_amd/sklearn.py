@@ -153,11 +153,21 @@ class XGBModel:
                 evals.append((self._make_dmatrix(ex, ey, sw, bm),
                               f"validation_{i}"))
         params = self.get_xgb_params()
-        if self.objective is None:
+        obj = None
+        if callable(self.objective):
+            # reference _objective_decorator (sklearn.py): a callable
+            # objective takes (y_true, y_pred) -> (grad, hess); train's
+            # fobj convention is (preds, dtrain)
+            user_fn = self.objective
+            params["objective"] = self._default_objective()
+
+            def obj(preds, dmat):  # noqa: ANN001
+                return user_fn(dmat.get_label(), preds)
+        elif self.objective is None:
             params.setdefault("objective", self._default_objective())
         self.evals_result_ = {}
         self._Booster = _train(
-            params, dtrain, self.n_estimators_, evals=evals,
+            params, dtrain, self.n_estimators_, evals=evals, obj=obj,
             early_stopping_rounds=getattr(self, "early_stopping_rounds", None),
             evals_result=self.evals_result_, verbose_eval=verbose,
             xgb_model=xgb_model,
