@@ -652,15 +652,25 @@ class _LambdaRankBase(Objective):
             "lambdarank_score_normalization", True))
         self.exp_gain = bool(self.params.get("ndcg_exp_gain", True))
         self.unbiased = bool(self.params.get("lambdarank_unbiased", False))
-        if self.unbiased:
-            raise NotImplementedError(
-                "lambdarank_unbiased (position debiasing) is not implemented")
+        self.bias_norm = float(self.params.get("lambdarank_bias_norm", 1.0))
+        # Unbiased LambdaMART position-bias ratios ti+ / tj- (reference
+        # lambdarank_obj.cc:133-136), carried across boosting iterations.
+        self._ti = None
+        self._tj = None
 
     def _k(self) -> int:
         """reference LambdaRankParam::NumPair (ranking_utils.h:102)."""
         if self.num_pair > 0:
             return self.num_pair
         return 32 if self.pair_method == "topk" else 1
+
+    def _max_position(self, cnt: torch.Tensor) -> int:
+        """reference RankingCache::MaxPositionSize (ranking_utils.h:224):
+        truncation level when pair_method=topk, else min(max group, 32)."""
+        if self.pair_method == "topk":
+            return self._k()
+        mx = int(cnt.max()) if cnt.numel() else 0
+        return min(mx, 32)
 
     def _groups(self, info):
         if info.group_ptr is None:
@@ -813,6 +823,54 @@ class _LambdaRankBase(Objective):
                     ok, delta / ((s_high - s_low).abs() + 0.01), delta)
             lam = (sig - 1.0) * delta
             hess = torch.clamp(sig * (1.0 - sig), min=1e-16) * delta * 2.0
+            li_pos = lj_pos = None
+            if self.unbiased:
+                # Unbiased LambdaMART (reference lambdarank_obj.h:128-147
+                # and lambdarank_obj.cc:205-221): divide the pair gradient
+                # by ti+[idx_high] * tj-[idx_low] (positions on the ORIGINAL
+                # label-sorted list) and accumulate the pair cost
+                # log(1/(1-sigma)) * |delta| into per-position sums.
+                k_pos = max(self._max_position(cnt), 1)
+                if (self._ti is None or self._ti.numel() != k_pos
+                        or self._ti.device != dev):
+                    self._ti = torch.ones(k_pos, dtype=torch.float64,
+                                          device=dev)
+                    self._tj = torch.ones(k_pos, dtype=torch.float64,
+                                          device=dev)
+                eps64 = 1e-16
+                ih_pos = idx_high - gp[pgk]
+                il_pos = idx_low - gp[pgk]
+                inside = (ih_pos < k_pos) & (il_pos < k_pos)
+                ih_c = ih_pos.clamp(max=k_pos - 1)
+                il_c = il_pos.clamp(max=k_pos - 1)
+                t_hi = self._ti[ih_c]
+                t_lo = self._tj[il_c]
+                ok_t = inside & (t_hi >= eps64) & (t_lo >= eps64)
+                scale = torch.where(ok_t, 1.0 / (t_hi * t_lo),
+                                    torch.ones_like(t_hi))
+                lam = lam * scale
+                hess = hess * scale
+                # cost = log(1/(1-sigma)) * delta_metric (eq. 30/31 input)
+                cost = -torch.log1p(-sig) * delta
+                zero = torch.zeros((), dtype=torch.float64, device=dev)
+                li_add = torch.where(inside & (t_lo >= eps64),
+                                     cost / t_lo.clamp(min=eps64), zero)
+                lj_add = torch.where(inside & (t_hi >= eps64),
+                                     cost / t_hi.clamp(min=eps64), zero)
+
+                def possum(keys, vals):
+                    # deterministic per-position sum (k_pos bins):
+                    # sort + cumsum + searchsorted, no fp64 atomics
+                    ks, pi = torch.sort(keys)
+                    cs = torch.zeros(vals.numel() + 1, dtype=torch.float64,
+                                     device=dev)
+                    torch.cumsum(vals[pi], 0, out=cs[1:])
+                    bnd = torch.searchsorted(
+                        ks, torch.arange(k_pos + 1, device=dev))
+                    return cs[bnd[1:]] - cs[bnd[:-1]]
+
+                li_pos = possum(ih_c, li_add)
+                lj_pos = possum(il_c, lj_add)
             # Scatter WITHOUT atomics (fp64 atomics CAS-loop on ROCm and
             # the hot top-ranked docs make them ~100-way contended):
             # accumulate by RANK-LIST POSITION with sort + cumsum +
@@ -859,6 +917,15 @@ class _LambdaRankBase(Objective):
                 else:
                     g_out /= self._k()
                     h_out /= self._k()
+            if self.unbiased and li_pos is not None:
+                # UpdatePositionBias (reference lambdarank_obj.cc:40-86):
+                # ti+(i) = (li(i)/li(0))^(1/(1+bias_norm)), fresh each
+                # iteration (the accumulators reset; ti carries over).
+                reg = 1.0 / (1.0 + self.bias_norm)
+                if float(li_pos[0]) >= 1e-16:
+                    self._ti = (li_pos / li_pos[0]).pow(reg)
+                if float(lj_pos[0]) >= 1e-16:
+                    self._tj = (lj_pos / lj_pos[0]).pow(reg)
         # group weights * weight_norm (ranking_utils.cc:44)
         if info.weights is not None:
             w_np = np.asarray(info.weights, np.float64).reshape(-1)
@@ -886,6 +953,8 @@ class _LambdaRankBase(Objective):
             "lambdarank_normalization": str(int(self.normalize)),
             "lambdarank_score_normalization": str(int(self.score_norm)),
             "ndcg_exp_gain": str(int(self.exp_gain)),
+            "lambdarank_unbiased": str(int(self.unbiased)),
+            "lambdarank_bias_norm": str(self.bias_norm),
         }}
 
 
