@@ -920,3 +920,42 @@ def test_gpu_u16_bins_training_matches_cpu():
         assert np.array_equal(tg.left[:tg.n_nodes], tc.left[:tc.n_nodes])
         assert np.allclose(tg.split_cond[:tg.n_nodes],
                            tc.split_cond[:tc.n_nodes], rtol=1e-6)
+
+
+def test_wt_graph_replay_matches_no_graph():
+    """The hipGraph-captured whole-tree chain (driver.hip) must produce
+    bit-identical models to the direct-enqueue chain (GBT_WT_GRAPH=0),
+    including across rounds where the graph is REPLAYED."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import tempfile
+
+    script = r"""
+import json, sys
+import numpy as np
+import xgboost_amd as xgb
+rng = np.random.RandomState(7)
+X = rng.randn(60000, 12).astype(np.float32)
+y = ((X[:, 0] * X[:, 1] + X[:, 2]) > 0).astype(np.float32)
+d = xgb.DMatrix(X, label=y)
+bst = xgb.train({"objective": "binary:logistic", "max_depth": 7,
+                 "device": "cuda", "eta": 0.3}, d, 15)
+out = {"dump": bst.get_dump(with_stats=True),
+       "pred": bst.predict(d)[:512].tolist()}
+json.dump(out, open(sys.argv[1], "w"))
+"""
+    outs = []
+    with tempfile.TemporaryDirectory() as td:
+        sp = os.path.join(td, "run.py")
+        with open(sp, "w") as f:
+            f.write(script)
+        for flag in ("1", "0"):
+            env = dict(os.environ, GBT_WT_GRAPH=flag)
+            of = os.path.join(td, f"out{flag}.json")
+            subprocess.run([sys.executable, sp, of], check=True, env=env,
+                           timeout=300)
+            outs.append(json.load(open(of)))
+    assert outs[0]["dump"] == outs[1]["dump"]
+    assert outs[0]["pred"] == outs[1]["pred"]

@@ -606,6 +606,18 @@ class GpuOps(SegmentedOpsMixin):
                                           dtype=torch.int32, device=dev),
                 "tg_scratch": torch.empty(3 * max_build + 8,
                                           dtype=torch.int32, device=dev),
+                # pointer-stable staging for the per-round inputs: the
+                # driver's hipGraph replay bakes device addresses, so
+                # the fresh-per-round tensors (quantized gradients, root
+                # sums, max-abs, colsample mask) are copied into these
+                # persistent buffers instead of being passed directly
+                "qg_stage": torch.empty((n_rows, 2), dtype=torch.int32,
+                                        device=dev),
+                "rs_stage": torch.empty(2, dtype=torch.int64, device=dev),
+                "ma_stage": torch.empty(2, dtype=torch.float32,
+                                        device=dev),
+                "fm_stage": torch.empty(self.qm.n_features,
+                                        dtype=torch.uint8, device=dev),
                 "driver": self.lib.gbt_driver_create(),
             }
             # whole-tree mode arena (single sync per tree): per-level
@@ -634,16 +646,20 @@ class GpuOps(SegmentedOpsMixin):
             ("sum_hess", np.float32), ("base_weight", np.float32)]}
         fmask_ptr = None
         if feature_mask is not None:
-            self._fmask_dev_t = torch.from_numpy(
-                np.ascontiguousarray(feature_mask, np.uint8)).to(dev)
-            fmask_ptr = self.hip.ptr(self._fmask_dev_t)
+            fm = np.ascontiguousarray(feature_mask, np.uint8)
+            ws["fm_stage"].copy_(
+                torch.from_numpy(fm), non_blocking=True)
+            fmask_ptr = self.hip.ptr(ws["fm_stage"])
         mono_dev = mono_host = None
         if monotone is not None:
             m8 = np.ascontiguousarray(monotone, np.int8)
-            self._mono_dev_t = torch.from_numpy(m8).to(dev)
-            self._mono_host = m8
+            key = m8.tobytes()
+            if getattr(self, "_mono_key", None) != key:
+                self._mono_dev_t = torch.from_numpy(m8).to(dev)
+                self._mono_host = m8
+                self._mono_key = key
             mono_dev = self.hip.ptr(self._mono_dev_t)
-            mono_host = m8.ctypes.data_as(ctypes.c_void_p)
+            mono_host = self._mono_host.ctypes.data_as(ctypes.c_void_p)
         cuts = self.qm.cuts
         cut_ptrs_host = np.ascontiguousarray(cuts.ptrs, np.int32)
         cut_values_host = np.ascontiguousarray(cuts.values, np.float32)
@@ -677,9 +693,15 @@ class GpuOps(SegmentedOpsMixin):
         out_scales = np.zeros(2, dtype=np.float64)
         gsc = quantizer.g_scale if ma is None else 0.0
         hsc = quantizer.h_scale if ma is None else 0.0
+        # stage per-round tensors at stable addresses (hipGraph replay)
+        qg_st = ws["qg_stage"][:n_rows]
+        qg_st.copy_(qgpair.view(n_rows, 2))
+        ws["rs_stage"].copy_(root_sums.view(-1)[:2])
+        if ma is not None:
+            ws["ma_stage"].copy_(ma.view(-1)[:2])
         rc = self.lib.gbt_grow_tree(
             ws["driver"], p8, p16, self.qm.n_features, c8, c16, n_rows,
-            self.hip.ptr(qgpair),
+            self.hip.ptr(qg_st),
             self.hip.ptr(self.cut_ptrs),
             cut_values_host.ctypes.data_as(ctypes.c_void_p),
             cut_ptrs_host.ctypes.data_as(ctypes.c_void_p),
@@ -698,8 +720,8 @@ class GpuOps(SegmentedOpsMixin):
             self.hip.ptr(ws["wt_ws"]),
             0 if ws["wt_ws"] is None else ws["wt_ws"].numel(),
             ws["wt_max_ptasks"],
-            self.hip.ptr(root_sums),
-            self.hip.ptr(ma),
+            self.hip.ptr(ws["rs_stage"]),
+            self.hip.ptr(ws["ma_stage"]) if ma is not None else None,
             out_scales.ctypes.data_as(ctypes.c_void_p),
             gsc, hsc,
             param.reg_lambda, param.reg_alpha, param.max_delta_step,

@@ -26,6 +26,7 @@
 
 #include <algorithm>
 #include <cmath>
+#include <cstdio>
 #include <cstdlib>
 #include <cstring>
 #include <vector>
@@ -68,6 +69,34 @@ struct DriverCtx {
   void* readback_host = nullptr;
   size_t readback_cap = 0;
 
+  // hipGraph replay of the single-GPU whole-tree chain: the enqueue
+  // sequence is identical every round (no host decisions mid-chain), so
+  // after one capture the ~25 launches/tree replay as ONE graph launch.
+  // Valid only while every baked address is unchanged — wt_key records
+  // them all and any mismatch forces a recapture.  The chain's host
+  // staging lives in a DEDICATED pinned buffer (not the rotating ring)
+  // so the baked source addresses stay stable; its contents are
+  // rewritten before every launch (replays read them at execution).
+  hipGraphExec_t wt_graph = nullptr;
+  std::vector<unsigned long long> wt_key;
+  void* wtc_host = nullptr;
+  void* wtc_dev = nullptr;
+  size_t wtc_cap = 0;
+  // capture needs a REAL stream: the caller usually passes torch's
+  // default stream (0), and capturing a null-stream alias leaves the
+  // legacy stream wedged in capture state.  The chain runs on this
+  // dedicated stream, ordered behind the caller stream by gevent.
+  hipStream_t gstream = nullptr;
+  hipEvent_t gevent = nullptr;
+
+  int ensure_gstream() {
+    if (gstream == nullptr)
+      HIP_CHECK(hipStreamCreateWithFlags(&gstream, hipStreamNonBlocking));
+    if (gevent == nullptr)
+      HIP_CHECK(hipEventCreateWithFlags(&gevent, hipEventDisableTiming));
+    return 0;
+  }
+
   int ensure_readback(size_t bytes) {
     if (readback_cap >= bytes) return 0;
     size_t want = 4096;
@@ -77,14 +106,49 @@ struct DriverCtx {
     readback_cap = want;
     return 0;
   }
+  int ensure_const(size_t bytes) {
+    if (wtc_cap >= bytes) return 0;
+    size_t want = 4096;
+    while (want < bytes) want <<= 1;
+    if (wtc_host) hipHostFree(wtc_host);
+    if (wtc_dev) hipFree(wtc_dev);
+    HIP_CHECK(hipHostMalloc(&wtc_host, want));
+    HIP_CHECK(hipMalloc(&wtc_dev, want));
+    wtc_cap = want;
+    return 0;
+  }
   ~DriverCtx() {
     for (int i = 0; i < PinnedRing::kSlots; ++i) {
       if (ring.host[i]) hipHostFree(ring.host[i]);
       if (ring.dev[i]) hipFree(ring.dev[i]);
     }
     if (readback_host) hipHostFree(readback_host);
+    if (wt_graph) (void)hipGraphExecDestroy(wt_graph);
+    if (wtc_host) hipHostFree(wtc_host);
+    if (wtc_dev) hipFree(wtc_dev);
+    if (gstream) (void)hipStreamDestroy(gstream);
+    if (gevent) (void)hipEventDestroy(gevent);
   }
 };
+
+// hipGraph replay of the whole-tree chain (default on; GBT_WT_GRAPH=0
+// disables).  Single-GPU only: the distributed chain interleaves RCCL
+// enqueues through a host callback, which a capture cannot contain.
+bool WtGraphEnabled() {
+  static int v = [] {
+    const char* e = getenv("GBT_WT_GRAPH");
+    return e ? atoi(e) : 1;
+  }();
+  return v != 0;
+}
+
+bool WtGraphDebug() {
+  static int v = [] {
+    const char* e = getenv("GBT_WT_GRAPH_DEBUG");
+    return e ? atoi(e) : 0;
+  }();
+  return v != 0;
+}
 
 struct Node {
   int nid;
@@ -148,6 +212,12 @@ void ChunkTasks(const std::vector<Node*>& nodes, std::vector<BlockTask>* out,
     }
   }
   if (out->empty()) out->push_back(BlockTask{0, 0, 0, 0});
+}
+
+__global__ void ZeroI64Kernel(int64_t* p, long long n) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = 0;
 }
 
 __global__ void IotaKernel(int32_t* r, long long n) {
@@ -560,8 +630,12 @@ int gbt_grow_tree(
   const bool has_mono = monotone_host != nullptr;
   double inv_g = 1.0 / g_scale, inv_h = 1.0 / h_scale;
   const long long hist_row = (long long)n_bins * 2;
+  const bool whole_tree =
+      wt_ws != nullptr &&
+      maxabs_dev != nullptr && max_nodes_level <= 1024 && max_depth >= 2 &&
+      wt_max_ptasks > 0;
 
-  {
+  if (!whole_tree) {
     int blocks = (int)std::min<long long>((n_rows + 255) / 256, 4096);
     hipLaunchKernelGGL(IotaKernel, dim3(blocks), dim3(256), 0, stream, ridx,
                        n_rows);
@@ -601,7 +675,7 @@ int gbt_grow_tree(
   const long long hist_min_rows = 2048;
 
   // ---- root histogram: tasks host-generated (root segment is known)
-  {
+  if (!whole_tree) {
     std::vector<Node*> nodes{&root};
     std::vector<BlockTask> tasks;
     ChunkTasks(nodes, &tasks, hist_min_rows, hist_tasks);
@@ -736,10 +810,8 @@ int gbt_grow_tree(
   // sibling build/subtract choice comes from GLOBAL hessians so every
   // rank reduces the same slot layout.  Monotone: fp64 bound
   // propagation runs inside ApplyKernel (scales derived on device).
-  const bool whole_tree =
-      wt_ws != nullptr &&
-      maxabs_dev != nullptr && max_nodes_level <= 1024 && max_depth >= 2 &&
-      wt_max_ptasks > 0;
+  // Single-GPU: the fixed launch sequence is captured into a hipGraph
+  // once and replayed (one graph launch instead of ~25 enqueues/tree).
   if (whole_tree) {
     const int choice_global = (allreduce != nullptr || has_mono) ? 1 : 0;
     const int pool = 2 * max_nodes_level;
@@ -787,103 +859,34 @@ int gbt_grow_tree(
         has_mono ? (double*)(w + o_bndb) : nullptr};
     uint8_t* d_mode = (uint8_t*)(w + o_mode);  // [max_depth][pool]
 
-    {  // knodes[0]=1, kpairs[0]=0, seg_rec[0]={0, n_rows}
-      const int slot = ctx->ring.next();
-      if (int e = ctx->ring.ensure(slot, 16)) return e;
-      int32_t* hh = (int32_t*)ctx->ring.host[slot];
+    // root tasks: regenerated each call with IDENTICAL content (they
+    // depend only on n_rows and the task constants) and staged through
+    // the DEDICATED pinned const buffer, so a captured graph can bake
+    // the addresses and every replay re-reads the rewritten bytes.
+    // Layout: [0..15] kn/kp/seg init ints, then the root BlockTasks.
+    std::vector<BlockTask> rtasks;
+    {
+      std::vector<Node*> rnodes{&root};
+      ChunkTasks(rnodes, &rtasks, hist_min_rows, hist_tasks);
+    }
+    const size_t c_tasks_off = 16;
+    if (int e = ctx->ensure_const(c_tasks_off +
+                                  rtasks.size() * sizeof(BlockTask)))
+      return e;
+    {
+      int32_t* hh = (int32_t*)ctx->wtc_host;
       hh[0] = 1;
       hh[1] = 0;
       hh[2] = 0;
       hh[3] = (int)n_rows;
-      HIP_CHECK(hipMemcpyAsync(kn_arr, &hh[0], 4, hipMemcpyHostToDevice,
-                               stream));
-      HIP_CHECK(hipMemcpyAsync(kp_arr, &hh[1], 4, hipMemcpyHostToDevice,
-                               stream));
-      HIP_CHECK(hipMemcpyAsync(seg_rec, &hh[2], 8, hipMemcpyHostToDevice,
-                               stream));
+      memcpy((char*)ctx->wtc_host + c_tasks_off, rtasks.data(),
+             rtasks.size() * sizeof(BlockTask));
     }
-    // root evaluation straight into best_rec[0]
-    gbt_evaluate_masked(hist_pool_a, 1, n_bins, n_features, cut_ptrs_dev,
-                 root_sums_dev, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
-                 max_delta_step, min_child_weight,
-                 has_mono ? monotone_dev : nullptr, nullptr, 0 /*mask_stride*/, fmask_dev,
-                 nullptr, eval_gain, eval_bin, eval_dir, eval_lsum, nullptr,
-                 0, stream);
-    gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, 1, n_features,
-                    best_rec, nullptr, stream);
     const int wt_max_htasks = (int)std::min<long long>(
         std::min<long long>(n_rows / hist_min_rows, hist_tasks) +
             max_nodes_level + 1,
         (long long)hist_tasks_cap);
-    const int64_t* ps_prev = root_sums_dev;
-    for (int L = 0; L + 1 < max_depth; ++L) {
-      const int64_t* bl =
-          best_rec + (L == 0 ? 0 : (1 + (size_t)(L - 1) * pool) * 6);
-      const int32_t* sl =
-          seg_rec + (L == 0 ? 0 : (1 + (size_t)(L - 1) * pool) * 2);
-      int64_t* bo = best_rec + (1 + (size_t)L * pool) * 6;
-      int32_t* so = seg_rec + (1 + (size_t)L * pool) * 2;
-      int64_t* ps_next = ps_bufs[L & 1];
-      // level-L capacity: at most 2^L nodes expand, 2^(L+1) children
-      const int cap_kids = (int)std::min<long long>(2LL << L, (long long)pool);
-      hipLaunchKernelGGL(ApplyKernel, dim3(1), dim3(256), 0, stream, bl, sl,
-                         kn_arr + L, kp_arr + L, ps_prev, gamma, cut_ptrs_dev,
-                         (long long)1024, (long long)2048, wt_max_ptasks,
-                         kn_arr + L + 1, kp_arr + L + 1, d_feat, d_sbin,
-                         d_dl, d_cnt, d_pt, d_desc, d_pps, d_pslot,
-                         choice_global, maxabs_dev,
-                         has_mono ? monotone_dev : nullptr,
-                         reg_lambda, reg_alpha, max_delta_step,
-                         (L == 0 || !has_mono) ? nullptr : bnd_bufs[(L - 1) & 1],
-                         has_mono ? bnd_bufs[L & 1] : nullptr,
-                         L == 0 ? nullptr : d_mode + (size_t)(L - 1) * pool);
-      gbt_partition(gidx8, gidx16, n_features, gidx8_col, gidx16_col,
-                    n_rows, cur_ridx, alt_ridx, d_pt,
-                    wt_max_ptasks, d_feat, d_sbin, d_dl, nullptr, nullptr,
-                    n_bins_feat_dev, d_cnt, stream);
-      std::swap(cur_ridx, alt_ridx);
-      hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, stream,
-                         d_cnt, d_desc, 0, hist_min_rows, hist_tasks,
-                         wt_max_htasks, tg_scratch, hist_tasks_dev, ps_next,
-                         kp_arr + L + 1, so,
-                         allreduce != nullptr ? 2 * cap_kids : 0,
-                         d_mode + (size_t)L * pool);
-      HIP_CHECK(hipMemsetAsync(next_pool, 0,
-                               (size_t)cap_kids * hist_row * sizeof(int64_t),
-                               stream));
-      gbt_hist(gidx8, gidx16, n_features, qgpair, cur_ridx, hist_tasks_dev,
-               wt_max_htasks, next_pool, n_bins, feat_group_start_dev,
-               bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
-               use_shared, ps_next, stream);
-      if (allreduce) {
-        // fixed worst-case counts (host-known, rank-identical): built
-        // hist slots 0..cap/2-1 (unused slots are zeros — the memset
-        // covers the pool, HistTaskGenKernel zeroed the ps padding) and
-        // the built-child pair sums.  Sibling subtraction AFTER the
-        // reduce then yields global histograms/sums for every child.
-        allreduce((long long*)next_pool,
-                  (long long)(cap_kids / 2) * hist_row);
-        allreduce((long long*)ps_next, (long long)cap_kids);
-      }
-      {
-        const long long total = (long long)(cap_kids / 2 + 1) * hist_row;
-        int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
-        hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
-                           stream, cur_pool, next_pool, next_pool, d_pslot,
-                           (int)hist_row, 0, ps_next, d_pps, kp_arr + L + 1);
-      }
-      gbt_evaluate_masked(next_pool, cap_kids, n_bins, n_features, cut_ptrs_dev,
-                   ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
-                   max_delta_step, min_child_weight,
-                   has_mono ? monotone_dev : nullptr,
-                   has_mono ? bnd_bufs[L & 1] : nullptr, 0 /*mask_stride*/, fmask_dev, nullptr, eval_gain, eval_bin, eval_dir,
-                   eval_lsum, kn_arr + L + 1, 0, stream);
-      gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, cap_kids,
-                      n_features, bo, kn_arr + L + 1, stream);
-      ps_prev = ps_next;
-      std::swap(cur_pool, next_pool);
-    }
-    // ---- the ONE sync: all per-level records + root sums + scales ----
+    // pre-size the ONE readback so its (graph-baked) address is final
     size_t r_best = 0;
     size_t r_seg = ((size_t)rec_slots * 48 + 63) & ~63ULL;
     size_t r_kp = (r_seg + (size_t)rec_slots * 8 + 63) & ~63ULL;
@@ -893,20 +896,291 @@ int gbt_grow_tree(
     size_t r_total = r_mode + (size_t)max_depth * pool;
     if (int e = ctx->ensure_readback(r_total)) return e;
     char* rb = (char*)ctx->readback_host;
-    HIP_CHECK(hipMemcpyAsync(rb + r_best, best_rec, (size_t)rec_slots * 48,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipMemcpyAsync(rb + r_seg, seg_rec, (size_t)rec_slots * 8,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipMemcpyAsync(rb + r_kp, kp_arr,
-                             (size_t)(max_depth + 2) * 4,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipMemcpyAsync(rb + r_rs, root_sums_dev, 16,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipMemcpyAsync(rb + r_ma, maxabs_dev, 8,
-                             hipMemcpyDeviceToHost, stream));
-    HIP_CHECK(hipMemcpyAsync(rb + r_mode, d_mode, (size_t)max_depth * pool,
-                             hipMemcpyDeviceToHost, stream));
+
+    // ---- the chain: EVERY launch for the whole tree, no host sync.
+    // Ping-pong pointers are locals so a graph REPLAY (which skips this
+    // code entirely) leaves the outer state identical.
+    // H2D staging: tiny constant uploads, enqueued directly each call
+    // (kept OUT of the captured graph — graph-embedded memcpy nodes
+    // showed unreliable replay ordering on ROCm)
+    auto stage_h2d = [&](hipStream_t cs) -> int {
+      char* ch = (char*)ctx->wtc_host;
+      HIP_CHECK(hipMemcpyAsync(kn_arr, ch, 4, hipMemcpyHostToDevice,
+                               cs));
+      HIP_CHECK(hipMemcpyAsync(kp_arr, ch + 4, 4, hipMemcpyHostToDevice,
+                               cs));
+      HIP_CHECK(hipMemcpyAsync(seg_rec, ch + 8, 8, hipMemcpyHostToDevice,
+                               cs));
+      HIP_CHECK(hipMemcpyAsync((char*)ctx->wtc_dev + c_tasks_off,
+                               ch + c_tasks_off,
+                               rtasks.size() * sizeof(BlockTask),
+                               hipMemcpyHostToDevice, cs));
+      return 0;
+    };
+    auto enqueue_chain = [&](hipStream_t cs) -> int {
+      int32_t* cr = ridx;
+      int32_t* ar = ridx_out;
+      int64_t* cp = hist_pool_a;
+      int64_t* np2 = hist_pool_b;
+      {
+        int blocks = (int)std::min<long long>((n_rows + 255) / 256, 4096);
+        hipLaunchKernelGGL(IotaKernel, dim3(blocks), dim3(256), 0, cs,
+                           cr, n_rows);
+      }
+      {
+        // zero via a kernel, not hipMemsetAsync: memset NODES in a
+        // captured graph replay out of order on this ROCm
+        int zb = (int)std::min<long long>((hist_row + 255) / 256, 4096);
+        hipLaunchKernelGGL(ZeroI64Kernel, dim3(zb), dim3(256), 0, cs, cp,
+                           hist_row);
+      }
+      gbt_hist(gidx8, gidx16, n_features, qgpair, cr,
+               (const BlockTask*)((char*)ctx->wtc_dev + c_tasks_off),
+               (int)rtasks.size(), cp, n_bins, feat_group_start_dev,
+               bin_group_start_dev, n_groups, max_group_bins, cut_ptrs_dev,
+               use_shared, nullptr, cs);
+      if (allreduce) allreduce((long long*)cp, hist_row);
+      // root evaluation straight into best_rec[0]
+      gbt_evaluate_masked(cp, 1, n_bins, n_features, cut_ptrs_dev,
+                   root_sums_dev, maxabs_dev, 0.0, 0.0, reg_lambda,
+                   reg_alpha, max_delta_step, min_child_weight,
+                   has_mono ? monotone_dev : nullptr, nullptr,
+                   0 /*mask_stride*/, fmask_dev,
+                   nullptr, eval_gain, eval_bin, eval_dir, eval_lsum,
+                   nullptr, 0, cs);
+      gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, 1,
+                      n_features, best_rec, nullptr, cs);
+      const int64_t* ps_prev = root_sums_dev;
+      for (int L = 0; L + 1 < max_depth; ++L) {
+        const int64_t* bl =
+            best_rec + (L == 0 ? 0 : (1 + (size_t)(L - 1) * pool) * 6);
+        const int32_t* sl =
+            seg_rec + (L == 0 ? 0 : (1 + (size_t)(L - 1) * pool) * 2);
+        int64_t* bo = best_rec + (1 + (size_t)L * pool) * 6;
+        int32_t* so = seg_rec + (1 + (size_t)L * pool) * 2;
+        int64_t* ps_next = ps_bufs[L & 1];
+        // level-L capacity: at most 2^L nodes expand, 2^(L+1) children
+        const int cap_kids =
+            (int)std::min<long long>(2LL << L, (long long)pool);
+        hipLaunchKernelGGL(ApplyKernel, dim3(1), dim3(256), 0, cs, bl,
+                           sl, kn_arr + L, kp_arr + L, ps_prev, gamma,
+                           cut_ptrs_dev,
+                           (long long)1024, (long long)2048, wt_max_ptasks,
+                           kn_arr + L + 1, kp_arr + L + 1, d_feat, d_sbin,
+                           d_dl, d_cnt, d_pt, d_desc, d_pps, d_pslot,
+                           choice_global, maxabs_dev,
+                           has_mono ? monotone_dev : nullptr,
+                           reg_lambda, reg_alpha, max_delta_step,
+                           (L == 0 || !has_mono) ? nullptr
+                                                 : bnd_bufs[(L - 1) & 1],
+                           has_mono ? bnd_bufs[L & 1] : nullptr,
+                           L == 0 ? nullptr : d_mode + (size_t)(L - 1) * pool);
+        gbt_partition(gidx8, gidx16, n_features, gidx8_col, gidx16_col,
+                      n_rows, cr, ar, d_pt,
+                      wt_max_ptasks, d_feat, d_sbin, d_dl, nullptr, nullptr,
+                      n_bins_feat_dev, d_cnt, cs);
+        std::swap(cr, ar);
+        hipLaunchKernelGGL(HistTaskGenKernel, dim3(1), dim3(256), 0, cs,
+                           d_cnt, d_desc, 0, hist_min_rows, hist_tasks,
+                           wt_max_htasks, tg_scratch, hist_tasks_dev,
+                           ps_next, kp_arr + L + 1, so,
+                           allreduce != nullptr ? 2 * cap_kids : 0,
+                           d_mode + (size_t)L * pool);
+        {
+          const long long zn = (long long)cap_kids * hist_row;
+          int zb = (int)std::min<long long>((zn + 255) / 256, 4096);
+          hipLaunchKernelGGL(ZeroI64Kernel, dim3(zb), dim3(256), 0, cs,
+                             np2, zn);
+        }
+        gbt_hist(gidx8, gidx16, n_features, qgpair, cr, hist_tasks_dev,
+                 wt_max_htasks, np2, n_bins, feat_group_start_dev,
+                 bin_group_start_dev, n_groups, max_group_bins,
+                 cut_ptrs_dev, use_shared, ps_next, cs);
+        if (allreduce) {
+          // fixed worst-case counts (host-known, rank-identical): built
+          // hist slots 0..cap/2-1 (unused slots are zeros — the memset
+          // covers the pool, HistTaskGenKernel zeroed the ps padding)
+          // and the built-child pair sums.  Sibling subtraction AFTER
+          // the reduce yields global histograms/sums for every child.
+          allreduce((long long*)np2,
+                    (long long)(cap_kids / 2) * hist_row);
+          allreduce((long long*)ps_next, (long long)cap_kids);
+        }
+        {
+          const long long total = (long long)(cap_kids / 2 + 1) * hist_row;
+          int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
+          hipLaunchKernelGGL(SubtractHistKernel, dim3(blocks), dim3(256), 0,
+                             cs, cp, np2, np2, d_pslot,
+                             (int)hist_row, 0, ps_next, d_pps,
+                             kp_arr + L + 1);
+        }
+        gbt_evaluate_masked(np2, cap_kids, n_bins, n_features, cut_ptrs_dev,
+                     ps_next, maxabs_dev, 0.0, 0.0, reg_lambda, reg_alpha,
+                     max_delta_step, min_child_weight,
+                     has_mono ? monotone_dev : nullptr,
+                     has_mono ? bnd_bufs[L & 1] : nullptr,
+                     0 /*mask_stride*/, fmask_dev, nullptr, eval_gain,
+                     eval_bin, eval_dir, eval_lsum, kn_arr + L + 1, 0,
+                     cs);
+        gbt_select_best(eval_gain, eval_bin, eval_dir, eval_lsum, cap_kids,
+                        n_features, bo, kn_arr + L + 1, cs);
+        ps_prev = ps_next;
+        std::swap(cp, np2);
+      }
+      return 0;
+    };
+    // D2H readback burst: direct enqueue after the kernels (not a graph
+    // node), the ONE sync follows
+    auto readback = [&](hipStream_t cs) -> int {
+      HIP_CHECK(hipMemcpyAsync(rb + r_best, best_rec,
+                               (size_t)rec_slots * 48,
+                               hipMemcpyDeviceToHost, cs));
+      HIP_CHECK(hipMemcpyAsync(rb + r_seg, seg_rec, (size_t)rec_slots * 8,
+                               hipMemcpyDeviceToHost, cs));
+      HIP_CHECK(hipMemcpyAsync(rb + r_kp, kp_arr,
+                               (size_t)(max_depth + 2) * 4,
+                               hipMemcpyDeviceToHost, cs));
+      HIP_CHECK(hipMemcpyAsync(rb + r_rs, root_sums_dev, 16,
+                               hipMemcpyDeviceToHost, cs));
+      HIP_CHECK(hipMemcpyAsync(rb + r_ma, maxabs_dev, 8,
+                               hipMemcpyDeviceToHost, cs));
+      HIP_CHECK(hipMemcpyAsync(rb + r_mode, d_mode,
+                               (size_t)max_depth * pool,
+                               hipMemcpyDeviceToHost, cs));
+      return 0;
+    };
+
+    // hipGraph capture/replay (single-GPU only): the key records every
+    // address and scalar the chain bakes at enqueue time — any change
+    // forces a recapture, so replay is exactly equivalent.  Capture and
+    // launch run on the ctx-owned stream (never the caller's, which is
+    // usually the un-capturable default stream), ordered behind the
+    // caller's pending work by gevent and joined by the host sync.
+    bool enqueued = false;
+    if (allreduce == nullptr && WtGraphEnabled() &&
+        ctx->ensure_gstream() == 0) {
+      auto db = [](double d) {
+        unsigned long long u;
+        memcpy(&u, &d, 8);
+        return u;
+      };
+      std::vector<unsigned long long> key = {
+          (unsigned long long)gidx8, (unsigned long long)gidx16,
+          (unsigned long long)gidx8_col, (unsigned long long)gidx16_col,
+          (unsigned long long)qgpair, (unsigned long long)cut_ptrs_dev,
+          (unsigned long long)n_bins_feat_dev,
+          (unsigned long long)feat_group_start_dev,
+          (unsigned long long)bin_group_start_dev,
+          (unsigned long long)ridx, (unsigned long long)ridx_out,
+          (unsigned long long)hist_pool_a, (unsigned long long)hist_pool_b,
+          (unsigned long long)eval_gain, (unsigned long long)eval_bin,
+          (unsigned long long)eval_dir, (unsigned long long)eval_lsum,
+          (unsigned long long)hist_tasks_dev,
+          (unsigned long long)tg_scratch, (unsigned long long)wt_ws,
+          (unsigned long long)root_sums_dev,
+          (unsigned long long)maxabs_dev, (unsigned long long)monotone_dev,
+          (unsigned long long)fmask_dev,
+          (unsigned long long)ctx->readback_host,
+          (unsigned long long)ctx->wtc_host,
+          (unsigned long long)ctx->wtc_dev,
+          (unsigned long long)n_rows, (unsigned long long)n_features,
+          (unsigned long long)n_bins, (unsigned long long)use_shared,
+          (unsigned long long)n_groups,
+          (unsigned long long)max_group_bins,
+          (unsigned long long)max_depth,
+          (unsigned long long)max_nodes_level,
+          (unsigned long long)wt_max_ptasks,
+          (unsigned long long)hist_tasks_cap,
+          (unsigned long long)rtasks.size(),
+          (unsigned long long)has_mono,
+          db(reg_lambda), db(reg_alpha), db(max_delta_step),
+          db(min_child_weight), db(gamma)};
+      // order the graph stream behind the caller stream (staged
+      // gradient/root-sum copies enqueued by the wrapper).  HOST-side
+      // wait: a hipStreamWaitEvent dependency is not reliably honored
+      // by hipGraphLaunch on the target stream (observed: replays read
+      // half-staged gradients), and the whole chain is about to run for
+      // ~0.5 ms anyway, so the few-us host block is in the noise.
+      HIP_CHECK(hipEventRecord(ctx->gevent, stream));
+      HIP_CHECK(hipEventSynchronize(ctx->gevent));
+      if (int e = stage_h2d(ctx->gstream)) return e;
+      if (ctx->wt_graph != nullptr && key == ctx->wt_key) {
+        if (hipGraphLaunch(ctx->wt_graph, ctx->gstream) == hipSuccess) {
+          enqueued = true;
+          if (WtGraphDebug()) fprintf(stderr, "[wtgraph] replay\n");
+        } else {
+          (void)hipGetLastError();
+          (void)hipGraphExecDestroy(ctx->wt_graph);
+          ctx->wt_graph = nullptr;
+          ctx->wt_key.clear();
+        }
+      }
+      if (!enqueued &&
+          hipStreamBeginCapture(ctx->gstream,
+                                hipStreamCaptureModeRelaxed) ==
+              hipSuccess) {
+        int ec = enqueue_chain(ctx->gstream);
+        hipGraph_t g = nullptr;
+        hipError_t ce = hipStreamEndCapture(ctx->gstream, &g);
+        if (WtGraphDebug())
+          fprintf(stderr, "[wtgraph] capture ec=%d end=%d\n", ec, (int)ce);
+        if (ec != 0) {
+          if (g) (void)hipGraphDestroy(g);
+          return ec;
+        }
+        if (ce == hipSuccess && g != nullptr) {
+          if (ctx->wt_graph) {
+            (void)hipGraphExecDestroy(ctx->wt_graph);
+            ctx->wt_graph = nullptr;
+          }
+          ctx->wt_key.clear();
+          hipError_t ie =
+              hipGraphInstantiate(&ctx->wt_graph, g, nullptr, nullptr, 0);
+          hipError_t le =
+              ie == hipSuccess ? hipGraphLaunch(ctx->wt_graph, ctx->gstream)
+                               : hipErrorUnknown;
+          if (WtGraphDebug())
+            fprintf(stderr, "[wtgraph] inst=%d launch=%d\n", (int)ie,
+                    (int)le);
+          if (ie == hipSuccess && le == hipSuccess) {
+            ctx->wt_key = key;
+            enqueued = true;
+          } else if (ctx->wt_graph) {
+            (void)hipGraphExecDestroy(ctx->wt_graph);
+            ctx->wt_graph = nullptr;
+          }
+          (void)hipGraphDestroy(g);
+        } else {
+          (void)hipGetLastError();
+        }
+        // a capture enqueues nothing for execution — every failure path
+        // above leaves enqueued=false and falls through to the direct
+        // enqueue below
+      }
+      if (!enqueued) {
+        // safety: never leave the graph stream wedged mid-capture
+        hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+        if (hipStreamIsCapturing(ctx->gstream, &st) == hipSuccess &&
+            st != hipStreamCaptureStatusNone) {
+          hipGraph_t junk = nullptr;
+          (void)hipStreamEndCapture(ctx->gstream, &junk);
+          if (junk) (void)hipGraphDestroy(junk);
+        }
+      }
+    }
+    if (!enqueued) {
+      if (WtGraphDebug()) fprintf(stderr, "[wtgraph] direct enqueue\n");
+      if (int e = stage_h2d(stream)) return e;
+      if (int e = enqueue_chain(stream)) return e;
+      if (int e = readback(stream)) return e;
+    } else {
+      if (int e = readback(ctx->gstream)) return e;
+      HIP_CHECK(hipStreamSynchronize(ctx->gstream));
+    }
     HIP_CHECK(hipStreamSynchronize(stream));
+    // the chain used local ping-pong pointers; reproduce the final
+    // parity for the post-sync leaf decide ((max_depth-1) swaps)
+    if ((max_depth - 1) & 1) std::swap(cur_ridx, alt_ridx);
     const int64_t* h_best = (const int64_t*)(rb + r_best);
     const int32_t* h_seg = (const int32_t*)(rb + r_seg);
     const int32_t* h_kp = (const int32_t*)(rb + r_kp);
