@@ -951,11 +951,14 @@ json.dump(out, open(sys.argv[1], "w"))
         sp = os.path.join(td, "run.py")
         with open(sp, "w") as f:
             f.write(script)
+        pkg_root = os.path.dirname(os.path.dirname(
+            os.path.abspath(__import__("xgboost_amd").__file__)))
         for flag in ("1", "0"):
-            env = dict(os.environ, GBT_WT_GRAPH=flag)
+            env = dict(os.environ, GBT_WT_GRAPH=flag,
+                       PYTHONPATH=pkg_root)
             of = os.path.join(td, f"out{flag}.json")
             subprocess.run([sys.executable, sp, of], check=True, env=env,
-                           timeout=300)
+                           timeout=300, cwd=pkg_root)
             outs.append(json.load(open(of)))
     assert outs[0]["dump"] == outs[1]["dump"]
     assert outs[0]["pred"] == outs[1]["pred"]
