@@ -1,0 +1,150 @@
+"""DART dropout (reference gbm/gbtree.cc: DropTrees :474, NormalizeTrees
+:539, weighted PredictBatch :632).  The `dart` booster name is a
+deprecated alias; dropout is driven by rate_drop/one_drop/skip_drop on
+the tree booster."""
+import json
+
+import numpy as np
+import pytest
+import torch
+
+import xgboost_amd as xgb
+
+
+def _data(n=2000, seed=0):
+    rng = np.random.RandomState(seed)
+    X = rng.randn(n, 6).astype(np.float32)
+    y = (X[:, 0] * 1.5 + X[:, 1] ** 2 + 0.1 * rng.randn(n)).astype(
+        np.float32)
+    return X, y
+
+
+def test_skip_drop_one_is_plain_gbtree():
+    X, y = _data()
+    d = xgb.DMatrix(X, label=y)
+    p = {"max_depth": 4, "eta": 0.3, "seed": 7}
+    plain = xgb.train(dict(p), d, 10)
+    dart = xgb.train(dict(p, rate_drop=0.5, skip_drop=1.0), d, 10)
+    assert plain.get_dump(with_stats=True) == dart.get_dump(with_stats=True)
+    # weights tracked but all 1.0
+    assert dart.weight_drop == [1.0] * 10
+
+
+def test_margin_cache_matches_fresh_weighted_predict():
+    """The incremental margin cache (new-tree adds scaled by the DART
+    weight + (factor-1)*dropped fix-up) must equal a from-scratch
+    weighted prediction after every kind of round."""
+    X, y = _data(1500, seed=3)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 4, "eta": 0.3, "seed": 5,
+                     "rate_drop": 0.4, "one_drop": True}, d, 12)
+    assert len(bst.weight_drop) == len(bst.trees)
+    assert any(w != 1.0 for w in bst.weight_drop)
+    cached, _ = bst._cache[id(d)]
+    fresh = bst._predict_margin(d)
+    assert torch.allclose(cached, fresh, atol=1e-4), \
+        (cached - fresh).abs().max()
+
+
+def test_rate_drop_one_weights_recurrence():
+    """rate_drop=1.0 drops EVERY tree each round -> deterministic
+    NormalizeTrees recurrence (normalize_type=tree):
+    k=len(trees), dropped *= k/(k+lr), new tree weight 1/(k+lr)."""
+    X, y = _data(800, seed=1)
+    d = xgb.DMatrix(X, label=y)
+    lr = 0.5
+    rounds = 5
+    bst = xgb.train({"max_depth": 3, "eta": lr, "seed": 2,
+                     "rate_drop": 1.0}, d, rounds)
+    w = []
+    for _ in range(rounds):
+        k = len(w)
+        if k == 0:
+            w.append(1.0)
+        else:
+            factor = k / (k + lr)
+            w = [x * factor for x in w]
+            w.append(1.0 / (k + lr))
+    assert np.allclose(bst.weight_drop, w, rtol=1e-6), (bst.weight_drop, w)
+
+
+def test_normalize_type_forest():
+    X, y = _data(800, seed=4)
+    d = xgb.DMatrix(X, label=y)
+    lr = 0.3
+    bst = xgb.train({"max_depth": 3, "eta": lr, "seed": 2,
+                     "rate_drop": 1.0, "normalize_type": "forest"}, d, 3)
+    f = 1.0 / (1.0 + lr)
+    # round2: w=[f, f]; round3: w=[f*f, f*f, f]
+    assert np.allclose(bst.weight_drop, [f * f, f * f, f], rtol=1e-6)
+
+
+def test_dart_save_load_roundtrip(tmp_path):
+    X, y = _data(1000, seed=6)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 4, "eta": 0.3, "seed": 9,
+                     "rate_drop": 0.3}, d, 8)
+    p1 = bst.predict(d)
+    fn = str(tmp_path / "m.json")
+    bst.save_model(fn)
+    j = json.load(open(fn))
+    assert "weight_drop" in j["learner"]["gradient_booster"]
+    bst2 = xgb.Booster(model_file=fn)
+    assert bst2.weight_drop == pytest.approx(bst.weight_drop)
+    assert np.allclose(bst2.predict(d), p1, atol=1e-6)
+
+
+def test_legacy_dart_format_loads(tmp_path):
+    """Old models: gradient_booster = {name: dart, gbtree: {...},
+    weight_drop: [...]} (gbtree.cc:452-463 compat)."""
+    X, y = _data(500, seed=8)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 0.3,
+                     "seed": 1}, d, 4)
+    fn = str(tmp_path / "m.json")
+    bst.save_model(fn)
+    j = json.load(open(fn))
+    gb = j["learner"]["gradient_booster"]
+    wd = gb.pop("weight_drop")
+    j["learner"]["gradient_booster"] = {
+        "name": "dart", "gbtree": gb, "weight_drop": wd}
+    fn2 = str(tmp_path / "legacy.json")
+    json.dump(j, open(fn2, "w"))
+    bst2 = xgb.Booster(model_file=fn2)
+    assert bst2.weight_drop == pytest.approx(bst.weight_drop)
+    assert np.allclose(bst2.predict(d), bst.predict(d), atol=1e-6)
+
+
+def test_dart_booster_alias_and_quality():
+    X, y = _data(3000, seed=10)
+    ycls = (y > np.median(y)).astype(np.float32)
+    d = xgb.DMatrix(X, label=ycls)
+    res = {}
+    bst = xgb.train({"booster": "dart", "objective": "binary:logistic",
+                     "max_depth": 4, "eta": 0.3, "rate_drop": 0.2,
+                     "seed": 3, "eval_metric": "auc"}, d, 25,
+                    evals=[(d, "t")], evals_result=res, verbose_eval=False)
+    assert res["t"]["auc"][-1] > 0.9
+    assert len(bst.weight_drop) == len(bst.trees)
+
+
+def test_dart_sklearn_wrapper():
+    from xgboost_amd.sklearn import XGBRegressor
+    X, y = _data(1200, seed=11)
+    m = XGBRegressor(n_estimators=10, max_depth=4, booster="dart",
+                     rate_drop=0.3, learning_rate=0.3)
+    m.fit(X, y)
+    p = m.predict(X)
+    assert np.isfinite(p).all()
+    assert np.corrcoef(p, y)[0, 1] > 0.8
+
+
+def test_weighted_sample_type_runs():
+    X, y = _data(800, seed=12)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 0.4,
+                     "sample_type": "weighted", "seed": 4}, d, 10)
+    assert len(bst.weight_drop) == 10
+    cached, _ = bst._cache[id(d)]
+    fresh = bst._predict_margin(d)
+    assert torch.allclose(cached, fresh, atol=1e-4)

@@ -778,10 +778,15 @@ class GpuOps(SegmentedOpsMixin):
 
 
 class _ForestArrays:
-    """Device SoA for a tree range [lo, hi) of a booster."""
+    """Device SoA for a tree range [lo, hi) of a booster.  DART tree
+    weights are folded into the LEAF values (margin = sum w_i*tree_i),
+    so the predict kernel needs no weight array."""
 
-    def __init__(self, booster, lo: int, hi: int, device):
-        trees = booster.trees[lo:hi]
+    def __init__(self, booster, lo: int, hi: int, device, idxs=None):
+        if idxs is None:
+            idxs = range(lo, hi)
+        idxs = list(idxs)
+        trees = [booster.trees[i] for i in idxs]
         offs = np.zeros(len(trees) + 1, np.int32)
         for i, t in enumerate(trees):
             offs[i + 1] = offs[i] + t.n_nodes
@@ -803,6 +808,10 @@ class _ForestArrays:
             right[o:o + n] = t.right[:n]
             sidx[o:o + n] = t.split_index[:n]
             cond[o:o + n] = t.split_cond[:n]
+            w = booster._tw(idxs[i]) if hasattr(booster, "_tw") else 1.0
+            if w != 1.0:
+                leaf_mask = t.left[:n] == -1
+                cond[o:o + n][leaf_mask] *= np.float32(w)
             dft[o:o + n] = t.default_left[:n]
             stype[o:o + n] = t.split_type[:n]
             hess[o:o + n] = t.sum_hess[:n]
@@ -831,7 +840,7 @@ class _ForestArrays:
         else:
             self.cat_bits = torch.zeros(1, dtype=torch.int32, device=device)
         self.tree_group = torch.tensor(
-            [booster.tree_info[t] for t in range(lo, hi)],
+            [booster.tree_info[t] for t in idxs],
             dtype=torch.int32, device=device)
         self.n_trees = len(trees)
 
@@ -1001,7 +1010,9 @@ def shap_interactions_gpu(booster, dmat, lo: int, hi: int,
 def _cached_forest(booster, lo: int, hi: int, device) -> _ForestArrays:
     """Device forest SoA cached across predict calls (serving path:
     repeated inplace_predict uploads nothing but the rows)."""
-    key = (lo, hi, len(booster.trees), str(device))
+    wd = getattr(booster, "weight_drop", None)
+    key = (lo, hi, len(booster.trees), str(device),
+           None if not wd else tuple(wd))
     fc = booster.__dict__.setdefault("_forest_dev_cache", {})
     fa = fc.get(key)
     if fa is None:
@@ -1009,6 +1020,31 @@ def _cached_forest(booster, lo: int, hi: int, device) -> _ForestArrays:
         fa = _ForestArrays(booster, lo, hi, device)
         fc[key] = fa
     return fa
+
+
+def predict_subset_gpu(booster, dmat, idxs, out: torch.Tensor
+                       ) -> torch.Tensor:
+    """DART dropped-trees contribution sum(w_i * tree_i(X)) on device
+    (weights folded into the subset forest's leaf values)."""
+    from .. import ops as hip_ops
+    lib = hip_ops.load()
+    device = out.device
+    fa = _ForestArrays(booster, 0, 0, device, idxs=idxs)
+    dd = dmat.device_data() if hasattr(dmat, "device_data") else None
+    X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
+    n = dmat.num_row()
+    missing = dmat.missing
+    missing_is_nan = 1 if np.isnan(missing) else 0
+    lib.gbt_predict(
+        hip_ops.ptr(X), n, dmat.num_col(),
+        float(0.0 if missing_is_nan else missing), missing_is_nan,
+        hip_ops.ptr(fa.tree_offsets), hip_ops.ptr(fa.left),
+        hip_ops.ptr(fa.right), hip_ops.ptr(fa.split_index),
+        hip_ops.ptr(fa.split_cond), hip_ops.ptr(fa.default_left),
+        hip_ops.ptr(fa.split_type), hip_ops.ptr(fa.cat_offsets),
+        hip_ops.ptr(fa.cat_bits), hip_ops.ptr(fa.tree_group), fa.n_trees,
+        out.shape[1], hip_ops.ptr(out), None, hip_ops.stream())
+    return out
 
 
 def predict_margin_gpu(booster, dmat, out_margin: torch.Tensor,

@@ -60,6 +60,24 @@ class Booster:
         self.seed = int(self.raw_params.get("seed", 0))
         self.seed_per_iteration = bool(self.raw_params.get("seed_per_iteration", False))
 
+        # DART dropout lives in the tree booster (reference deprecates the
+        # `dart` booster name and folds rate_drop/skip_drop/one_drop into
+        # gbtree — gbm/gbtree.cc:474 DropTrees, :539 NormalizeTrees)
+        def _flag(v) -> bool:
+            return str(v).lower() in ("1", "true", "yes")
+        rp = self.raw_params
+        self.dart_params = {
+            "rate_drop": float(rp.get("rate_drop", 0.0)),
+            "one_drop": _flag(rp.get("one_drop", 0)),
+            "skip_drop": float(rp.get("skip_drop", 0.0)),
+            "sample_type": str(rp.get("sample_type", "uniform")),
+            "normalize_type": str(rp.get("normalize_type", "tree")),
+        }
+        self.weight_drop: List[float] = []   # per-tree weights (DART)
+        self._idx_drop: List[int] = []       # this iteration's drop set
+        self._dart_dc = None                 # dropped-trees margin term
+        self._dart_rng = np.random.RandomState((self.seed + 2027) % (1 << 31))
+
         self.trees: List[RegTree] = []
         self.tree_info: List[int] = []
         self.iteration_indptr: List[int] = [0]
@@ -233,6 +251,112 @@ class Booster:
             self._ops_cache[key] = ops
         return ops
 
+    # -- DART dropout (reference gbm/gbtree.cc) ------------------------
+    def _dart_configured(self) -> bool:
+        dp = self.dart_params
+        return dp["rate_drop"] != 0.0 or dp["one_drop"] or \
+            dp["skip_drop"] != 0.0
+
+    def _dart_track(self) -> bool:
+        """CommitModel tracks tree weights once dropout is configured or
+        a loaded model carries weights (gbtree.cc:347-356)."""
+        return bool(self.weight_drop) or self._dart_configured()
+
+    def _tw(self, t: int) -> float:
+        """Per-tree prediction weight (1.0 outside DART)."""
+        return self.weight_drop[t] if t < len(self.weight_drop) else 1.0
+
+    def _dart_drop_trees(self) -> List[int]:
+        """Select this iteration's drop set (GBTree::DropTrees,
+        gbtree.cc:474): uniform or weighted per-tree Bernoulli at
+        rate_drop, optional one_drop backstop, skip_drop gate."""
+        dp = self.dart_params
+        self._idx_drop = []
+        if not self._dart_configured() or not self.trees:
+            return []
+        if not self.weight_drop:
+            self.weight_drop = [1.0] * len(self.trees)
+        elif len(self.weight_drop) < len(self.trees):
+            self.weight_drop.extend(
+                [1.0] * (len(self.trees) - len(self.weight_drop)))
+        rng = self._dart_rng
+        if dp["skip_drop"] > 0.0 and rng.uniform() < dp["skip_drop"]:
+            return []
+        wd = self.weight_drop
+        if dp["sample_type"] == "weighted":
+            sw = float(sum(wd))
+            for i, w in enumerate(wd):
+                if rng.uniform() < dp["rate_drop"] * len(wd) * w / sw:
+                    self._idx_drop.append(i)
+            if dp["one_drop"] and not self._idx_drop and wd:
+                p = np.asarray(wd, np.float64)
+                self._idx_drop.append(int(rng.choice(len(wd), p=p / p.sum())))
+        else:
+            for i in range(len(wd)):
+                if rng.uniform() < dp["rate_drop"]:
+                    self._idx_drop.append(i)
+            if dp["one_drop"] and not self._idx_drop and wd:
+                self._idx_drop.append(int(rng.randint(len(wd))))
+        return list(self._idx_drop)
+
+    def _dart_new_weight(self) -> float:
+        """Weight the trees built this iteration will carry
+        (NormalizeTrees, gbtree.cc:539)."""
+        if not self._dart_track():
+            return 1.0
+        k = len(self._idx_drop)
+        if k == 0:
+            return 1.0
+        lr = float(self.tparam.eta)
+        if self.dart_params["normalize_type"] == "forest":
+            return 1.0 / (1.0 + lr)
+        return 1.0 / (k + lr)
+
+    def _dart_commit(self, margin: torch.Tensor, n_new: int) -> None:
+        """NormalizeTrees + incremental margin fix-up: the cached margin
+        holds sum(w_i * tree_i); scaling the dropped trees by `factor`
+        shifts it by (factor - 1) * dropped_contribution."""
+        if not self._dart_track():
+            return
+        n_old = len(self.trees) - n_new
+        if len(self.weight_drop) < n_old:
+            self.weight_drop.extend([1.0] * (n_old - len(self.weight_drop)))
+        k = len(self._idx_drop)
+        lr = float(self.tparam.eta)
+        if k == 0:
+            self.weight_drop.extend([1.0] * n_new)
+        else:
+            factor = (1.0 / (1.0 + lr)
+                      if self.dart_params["normalize_type"] == "forest"
+                      else k / (k + lr))
+            for i in self._idx_drop:
+                self.weight_drop[i] *= factor
+            if self._dart_dc is not None:
+                margin += (factor - 1.0) * self._dart_dc
+            self.weight_drop.extend([self._dart_new_weight()] * n_new)
+        self._idx_drop = []
+        self._dart_dc = None
+        self._forest_dev_cache = {}  # weights changed: drop stale SoA
+
+    def _predict_tree_subset(self, dmat: DMatrix,
+                             idxs: List[int]) -> torch.Tensor:
+        """sum(w_i * tree_i(X)) over the drop set, shaped like the
+        margin cache."""
+        n = dmat.num_row()
+        out = torch.zeros((n, self.n_outputs), dtype=torch.float32,
+                          device=self.device)
+        if self.device.type == "cuda":
+            from .backend.gpu import predict_subset_gpu
+            return predict_subset_gpu(self, dmat, idxs, out)
+        X = dmat.raw_data()
+        for t in idxs:
+            tree = self.trees[t]
+            pos = tree.predict_leaf_np(X, dmat.missing)
+            vals = tree.split_cond[:tree.n_nodes][pos] * self._tw(t)
+            out[:, self.tree_info[t]] += torch.as_tensor(
+                vals, device=out.device)
+        return out
+
     # -- training ------------------------------------------------------
     def update(self, dtrain: DMatrix, iteration: int,
                fobj=None) -> None:
@@ -245,19 +369,28 @@ class Booster:
         from .monitor import TrainingObserver
         if TrainingObserver.enabled():
             TrainingObserver.observe_predictions(iteration, margin)
-        if fobj is None and self._boost_fused(dtrain, margin, iteration):
+        use_margin = margin
+        if self.booster_kind != "gblinear" and self._dart_configured():
+            # DART: gradients come from the margin with this round's
+            # dropped trees removed (GBTree::PredictBatch is_training)
+            dropped = self._dart_drop_trees()
+            if dropped:
+                self._dart_dc = self._predict_tree_subset(dtrain, dropped)
+                use_margin = margin - self._dart_dc
+        elif fobj is None and self._boost_fused(dtrain, margin, iteration):
             if TrainingObserver.enabled():
                 TrainingObserver.observe_tree(iteration, self.trees[-1])
             return
         if fobj is not None:
-            preds = self.objective.pred_transform(margin).cpu().numpy()
+            preds = self.objective.pred_transform(use_margin).cpu().numpy()
             grad, hess = fobj(np.squeeze(preds), dtrain)
             grad = torch.as_tensor(np.asarray(grad, np.float32),
                                    device=margin.device).view(margin.shape)
             hess = torch.as_tensor(np.asarray(hess, np.float32),
                                    device=margin.device).view(margin.shape)
         else:
-            grad, hess = self.objective.get_gradient(margin, dtrain.info, iteration)
+            grad, hess = self.objective.get_gradient(use_margin, dtrain.info,
+                                                     iteration)
             TrainingObserver.observe_gradient(iteration, grad, hess)
         self.boost_gpair(dtrain, grad, hess, iteration)
 
@@ -299,6 +432,7 @@ class Booster:
             self._boost_multi_target(dtrain, ops, grad, hess, margin, seed)
             return
         is_approx = self.tparam.tree_method == "approx"
+        w_new = self._dart_new_weight()
         for k in range(n_out):
             if is_approx:
                 ops = self._ops_for(
@@ -326,7 +460,9 @@ class Booster:
                     (leaf_vals,) = ops.stager.upload([leaf_np])
                 else:
                     leaf_vals = torch.as_tensor(leaf_np, device=margin.device)
-                margin[:, k] += leaf_vals[positions.to(margin.device).long()]
+                add = leaf_vals[positions.to(margin.device).long()]
+                margin[:, k] += add if w_new == 1.0 else w_new * add
+        self._dart_commit(margin, new_trees)
         self.iteration_indptr.append(self.iteration_indptr[-1] + new_trees)
         self._cache[id(dtrain)] = (margin, len(self.trees))
         from .monitor import TrainingObserver
@@ -475,7 +611,10 @@ class Booster:
         self.iteration_indptr.append(self.iteration_indptr[-1] + 1)
         leaf_vals = torch.as_tensor(
             tree.leaf_values[:tree.n_nodes].copy(), device=margin.device)
-        margin += leaf_vals[positions.to(margin.device).long()]
+        w_new = self._dart_new_weight()
+        add = leaf_vals[positions.to(margin.device).long()]
+        margin += add if w_new == 1.0 else w_new * add
+        self._dart_commit(margin, 1)
         self._cache[id(dtrain)] = (margin, len(self.trees))
 
     def _update_existing(self, dtrain: DMatrix, iteration: int) -> None:
@@ -525,6 +664,7 @@ class Booster:
         cache = self.__dict__.setdefault("_exact_cache", {}).setdefault(
             id(dtrain), {})
         new_trees = 0
+        w_new = self._dart_new_weight()
         for k in range(n_out):
             for ptree in range(self.tparam.num_parallel_tree):
                 gpair = torch.stack([grad[:, k], hess[:, k]], dim=1)
@@ -542,7 +682,9 @@ class Booster:
                     tree.split_cond[:tree.n_nodes].copy(), device=margin.device)
                 pos_t = torch.as_tensor(positions.astype(np.int64),
                                         device=margin.device)
-                margin[:, k] += leaf_vals[pos_t]
+                add = leaf_vals[pos_t]
+                margin[:, k] += add if w_new == 1.0 else w_new * add
+        self._dart_commit(margin, new_trees)
         self.iteration_indptr.append(self.iteration_indptr[-1] + new_trees)
         self._cache[id(dtrain)] = (margin, len(self.trees))
 
@@ -644,12 +786,13 @@ class Booster:
         for t in range(lo, hi):
             tree = self.trees[t]
             pos = tree.predict_leaf_np(X, dmat.missing)
+            w = self._tw(t)
             if tree.leaf_values is not None:
-                out += torch.as_tensor(tree.leaf_values[:tree.n_nodes][pos],
-                                       device=out.device)
+                out += w * torch.as_tensor(
+                    tree.leaf_values[:tree.n_nodes][pos], device=out.device)
             else:
                 vals = tree.split_cond[:tree.n_nodes][pos]
-                out[:, self.tree_info[t]] += torch.as_tensor(
+                out[:, self.tree_info[t]] += w * torch.as_tensor(
                     vals, device=out.device)
         return out
 
@@ -685,7 +828,7 @@ class Booster:
             finally:
                 tree.split_index[:tree.n_nodes] = saved
             vals = tree.split_cond[:tree.n_nodes][pos]
-            out[:, self.tree_info[t]] += torch.as_tensor(
+            out[:, self.tree_info[t]] += self._tw(t) * torch.as_tensor(
                 vals, device=out.device)
         return out
 
@@ -700,7 +843,7 @@ class Booster:
                 tree = self.trees[t]
                 pos = tree.predict_leaf_bins(gg, dmat.cuts)
                 vals = tree.split_cond[:tree.n_nodes][pos]
-                out[s:e, self.tree_info[t]] += torch.as_tensor(
+                out[s:e, self.tree_info[t]] += self._tw(t) * torch.as_tensor(
                     vals, device=out.device)
         return out
 
@@ -950,6 +1093,12 @@ class Booster:
                 },
                 "name": "gbtree",
             }
+            if self.weight_drop:
+                # DART tree weights: the reference LOADS this key from the
+                # gbtree object (gbtree.cc:456 compat path) but its writer
+                # currently drops it; we write it so round-trips keep the
+                # weights and reference builds can still read the file
+                gb["weight_drop"] = [float(w) for w in self.weight_drop]
         learner = {
             "attributes": dict(self.attributes_),
             "feature_names": self.feature_names or [],
@@ -987,6 +1136,14 @@ class Booster:
         self.objective = create_objective(obj_name, obj_params)
         self.raw_params["objective"] = obj_name
         gb = learner["gradient_booster"]
+        if gb.get("name") == "dart":
+            # legacy DART format: {"name": "dart", "gbtree": {...},
+            # "weight_drop": [...]} (gbtree.cc:452-463)
+            wd = gb.get("weight_drop") or []
+            gb = dict(gb["gbtree"])
+            gb.setdefault("name", "gbtree")
+            if wd:
+                gb["weight_drop"] = wd
         if gb.get("name") == "gblinear":
             from .linear import GBLinearModel
             self.booster_kind = "gblinear"
@@ -1004,6 +1161,9 @@ class Booster:
         model = gb["model"]
         self.trees = [RegTree.from_json(t) for t in model["trees"]]
         self.tree_info = [int(x) for x in model["tree_info"]]
+        self.weight_drop = [float(w) for w in gb.get("weight_drop", [])]
+        if len(self.weight_drop) > len(self.trees):
+            raise ValueError("weight_drop longer than the tree list")
         indptr = model.get("iteration_indptr")
         if indptr:
             self.iteration_indptr = [int(x) for x in indptr]

@@ -139,6 +139,8 @@ LEARNER_PARAMS = {
     "lambdarank_normalization", "lambdarank_score_normalization",
     "lambdarank_unbiased", "lambdarank_bias_norm", "ndcg_exp_gain",
     "expectile_alpha",
+    # DART dropout (gbtree-level, reference gbm/gbtree.h DartTrainParam)
+    "rate_drop", "one_drop", "skip_drop", "sample_type", "normalize_type",
 }
 
 _KNOWN = {f.name for f in dataclasses.fields(TrainParam)} | {"lambda", "alpha"} | LEARNER_PARAMS | set(ALIASES)
