@@ -16,6 +16,7 @@ from .data import DMatrix, QuantileDMatrix  # noqa: F401
 from .plotting import plot_importance, plot_tree, to_graphviz  # noqa: F401
 from .training import cv, train  # noqa: F401
 from . import callback  # noqa: F401
+from . import objective  # noqa: F401  (experimental class objectives)
 from . import collective  # noqa: F401
 
 __version__ = "0.1.0"

@@ -382,8 +382,24 @@ class Booster:
                 TrainingObserver.observe_tree(iteration, self.trees[-1])
             return
         if fobj is not None:
-            preds = self.objective.pred_transform(use_margin).cpu().numpy()
-            grad, hess = fobj(np.squeeze(preds), dtrain)
+            # custom objectives receive the RAW margin (reference
+            # core.py:2315 predict(output_margin=True, training=True))
+            preds = use_margin.detach().cpu().numpy()
+            from .objective import Objective as _ClsObj, TreeObjective
+            if isinstance(fobj, TreeObjective):
+                # full gradient values the leaves; an optional reduced
+                # gradient finds the structure (reference core.py:2320)
+                vgrad, vhess = fobj(iteration, preds, dtrain)
+                sg = fobj.split_grad(iteration, vgrad, vhess)
+                if sg is not None:
+                    self._boost_split_grad(dtrain, sg[0], sg[1], vgrad,
+                                           vhess, iteration)
+                    return
+                grad, hess = vgrad, vhess
+            elif isinstance(fobj, _ClsObj):
+                grad, hess = fobj(iteration, preds, dtrain)
+            else:
+                grad, hess = fobj(np.squeeze(preds), dtrain)
             grad = torch.as_tensor(np.asarray(grad, np.float32),
                                    device=margin.device).view(margin.shape)
             hess = torch.as_tensor(np.asarray(hess, np.float32),
@@ -615,6 +631,96 @@ class Booster:
         add = leaf_vals[positions.to(margin.device).long()]
         margin += add if w_new == 1.0 else w_new * add
         self._dart_commit(margin, 1)
+        self._cache[id(dtrain)] = (margin, len(self.trees))
+
+    def _boost_split_grad(self, dtrain: DMatrix, sgrad, shess, vgrad,
+                          vhess, iteration: int) -> None:
+        """Reduced-gradient boosting (reference c_api.cc:1237
+        XGBoosterTrainOneIterWithSplitGrad + gbtree.cc:191): ONE
+        vector-leaf tree per iteration whose STRUCTURE comes from the
+        reduced (sgrad, shess) and whose leaf VALUES come from the
+        full-width (vgrad, vhess)."""
+        dev = self.device
+        n = dtrain.num_row()
+
+        def t2(a):
+            return torch.as_tensor(np.asarray(a, np.float32),
+                                   device=dev).view(n, -1)
+
+        sg, sh = t2(sgrad), t2(shess)
+        vg, vh = t2(vgrad), t2(vhess)
+        if vg.shape[1] < 2:
+            raise ValueError(
+                "split_grad requires vector-leaf trees: configure the "
+                "booster with num_target matching the full gradient "
+                "width (reference gbtree.cc:192)")
+        if vg.shape[1] != self.n_outputs:
+            raise ValueError(
+                f"value gradient width {vg.shape[1]} != model outputs "
+                f"{self.n_outputs}")
+        if self.tparam.monotone_constraints:
+            raise ValueError("monotone constraints are not supported "
+                             "with reduced gradients (gbtree.cc:194)")
+        margin, _ = self._cache[id(dtrain)]
+        ops = self._ops_for(dtrain)
+        seed = self.seed + iteration
+        if sg.shape[1] == 1:
+            gpair = torch.stack([sg[:, 0], sh[:, 0]],
+                                dim=1).contiguous().to(ops.device)
+            quant = GradQuantizer(gpair)
+            qg = quant.quantize(gpair)
+            tree = RegTree(self.n_features)
+            grower = TreeGrower(ops, self.tparam, quant, n, seed=seed,
+                                feature_weights=dtrain.info.feature_weights)
+            tree, positions = grower.grow(qg, tree)
+        else:
+            from .grower import MultiTargetGrower
+            qgpairs, quantizers = [], []
+            for k in range(sg.shape[1]):
+                gp = torch.stack([sg[:, k], sh[:, k]],
+                                 dim=1).contiguous().to(ops.device)
+                q = GradQuantizer(gp)
+                quantizers.append(q)
+                qgpairs.append(q.quantize(gp))
+            tree = RegTree(self.n_features, sg.shape[1])
+            grower = MultiTargetGrower(
+                ops, self.tparam, quantizers, n, seed,
+                feature_weights=dtrain.info.feature_weights)
+            tree, positions = grower.grow(qgpairs, tree)
+        # vector leaves from the FULL gradient at the found partition
+        # (deterministic sort+cumsum segment sums, no fp64 atomics)
+        C = vg.shape[1]
+        nn = tree.n_nodes
+        pos = positions.to(dev).long()
+        ps, pperm = torch.sort(pos)
+        bnd = torch.searchsorted(ps, torch.arange(nn + 1, device=dev))
+        Gv = torch.zeros((nn, C), dtype=torch.float64, device=dev)
+        Hv = torch.zeros((nn, C), dtype=torch.float64, device=dev)
+        cs = torch.zeros(n + 1, dtype=torch.float64, device=dev)
+        for c in range(C):
+            torch.cumsum(vg[:, c].double()[pperm], 0, out=cs[1:])
+            Gv[:, c] = cs[bnd[1:]] - cs[bnd[:-1]]
+            torch.cumsum(vh[:, c].double()[pperm], 0, out=cs[1:])
+            Hv[:, c] = cs[bnd[1:]] - cs[bnd[:-1]]
+        lam = self.tparam.reg_lambda
+        alpha = self.tparam.reg_alpha
+        w = -torch.sign(Gv) * torch.clamp(Gv.abs() - alpha, min=0.0) \
+            / (Hv + lam)
+        mds = self.tparam.max_delta_step
+        if mds > 0:
+            w = w.clamp(-mds, mds)
+        w = (w * self.tparam.eta).float()
+        # upgrade to a vector-leaf tree: values only at leaves
+        leaf_mask = torch.as_tensor(tree.left[:nn] == -1, device=dev)
+        w = torch.where(leaf_mask[:, None], w, torch.zeros_like(w))
+        lv = np.zeros((len(tree.split_cond), C), dtype=np.float32)
+        lv[:nn] = w.cpu().numpy()
+        tree.n_targets = C
+        tree.leaf_values = lv
+        self.trees.append(tree)
+        self.tree_info.append(0)
+        self.iteration_indptr.append(self.iteration_indptr[-1] + 1)
+        margin += torch.as_tensor(lv[:nn], device=margin.device)[pos]
         self._cache[id(dtrain)] = (margin, len(self.trees))
 
     def _update_existing(self, dtrain: DMatrix, iteration: int) -> None:
