@@ -148,3 +148,22 @@ def test_weighted_sample_type_runs():
     cached, _ = bst._cache[id(d)]
     fresh = bst._predict_margin(d)
     assert torch.allclose(cached, fresh, atol=1e-4)
+
+
+def test_dart_slice_carries_weights():
+    """Booster slicing keeps each tree's DART weight (reference
+    GBTree::Slice, gbtree.cc:625-631)."""
+    X, y = _data(800, seed=13)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 1.0,
+                     "seed": 4}, d, 6)
+    sl = bst[2:5]
+    assert sl.weight_drop == pytest.approx(bst.weight_drop[2:5])
+    # sliced predictions = weighted sum over just those trees
+    base = bst.base_score
+    exp = np.full(len(y), base, np.float32)
+    for i, t in enumerate(range(2, 5)):
+        pos = bst.trees[t].predict_leaf_np(X, float("nan"))
+        exp += bst.weight_drop[t] * \
+            bst.trees[t].split_cond[:bst.trees[t].n_nodes][pos]
+    assert np.allclose(sl.predict(d), exp, atol=1e-5)

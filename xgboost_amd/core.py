@@ -1324,6 +1324,10 @@ class Booster:
             for t in range(s, e):
                 b.trees.append(self.trees[t])
                 b.tree_info.append(self.tree_info[t])
+                # DART weights follow their trees (reference
+                # GBTree::Slice, gbtree.cc:625-631)
+                if self.weight_drop:
+                    b.weight_drop.append(self._tw(t))
             b.iteration_indptr.append(len(b.trees))
         return b
 
