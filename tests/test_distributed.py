@@ -618,3 +618,65 @@ def test_two_process_ranking_training():
     """Group-sharded lambdarank: gradients are group-local, histogram
     sync must still produce identical trees on every worker."""
     _run_workers(2, RANK_WORKER)
+
+
+DART_WORKER = r"""
+import os, pickle, sys
+import numpy as np
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+
+collective.init("gloo")
+rank = collective.get_rank()
+world = collective.get_world_size()
+rng = np.random.RandomState(3)
+n, f = 1600, 5
+X = rng.randn(n, f).astype(np.float32)
+y = (X[:, 0] - 0.5 * X[:, 1] + 0.1 * rng.randn(n)).astype(np.float32)
+sl = slice(rank * n // world, (rank + 1) * n // world)
+d = xgb.DMatrix(X[sl], label=y[sl])
+bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 0.4,
+                 "one_drop": True, "seed": 11}, d, 8)
+with open(os.environ["XGB_AMD_OUT"] + f".{rank}", "wb") as fh:
+    pickle.dump({"dump": bst.get_dump(with_stats=True),
+                 "wd": bst.weight_drop}, fh)
+collective.finalize()
+"""
+
+
+def test_two_process_dart_identical_models():
+    """DART under data-parallel training: the drop set comes from a
+    seed-deterministic RNG and tree counts, both rank-identical, so
+    every rank must produce the SAME model and weights."""
+    port = 29553
+    with tempfile.TemporaryDirectory() as td:
+        spath = os.path.join(td, "worker.py")
+        with open(spath, "w") as fh:
+            fh.write(DART_WORKER)
+        out = os.path.join(td, "out.pkl")
+        procs = []
+        for r in range(2):
+            env = dict(os.environ)
+            env.update({
+                "RANK": str(r), "WORLD_SIZE": "2",
+                "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                "XGB_AMD_REPO": REPO, "XGB_AMD_OUT": out,
+            })
+            procs.append(subprocess.Popen(
+                [sys.executable, spath], env=env,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        outputs = []
+        ok = True
+        for p in procs:
+            stdout, _ = p.communicate(timeout=300)
+            outputs.append(stdout.decode())
+            ok = ok and p.returncode == 0
+        assert ok, "worker failed:\n" + "\n---\n".join(outputs)
+        with open(out + ".0", "rb") as fh:
+            r0 = pickle.load(fh)
+        with open(out + ".1", "rb") as fh:
+            r1 = pickle.load(fh)
+    assert r0["dump"] == r1["dump"]
+    assert r0["wd"] == pytest.approx(r1["wd"])
+    assert any(w != 1.0 for w in r0["wd"])

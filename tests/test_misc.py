@@ -421,3 +421,38 @@ def test_bench_contract_cpu():
         assert k in d, k
     assert d["metric"] == "boosting_rounds_per_sec"
     assert d["config"]["parallelism"] == "dp1"
+
+
+def test_testing_module_generators():
+    """xgboost_amd.testing: offline analogs of the reference's public
+    testing data helpers, usable end-to-end."""
+    import xgboost_amd as xgb
+    from xgboost_amd import testing as tm
+
+    X, y = tm.make_regression(400, 6, seed=1)
+    bst = xgb.train({"max_depth": 3}, xgb.DMatrix(X, label=y), 5)
+    assert np.corrcoef(bst.predict(xgb.DMatrix(X)), y)[0, 1] > 0.7
+
+    Xc, yc = tm.make_classification(400, 6, n_classes=3, seed=2)
+    bst = xgb.train({"objective": "multi:softmax", "num_class": 3,
+                     "max_depth": 3}, xgb.DMatrix(Xc, label=yc), 5)
+    assert (bst.predict(xgb.DMatrix(Xc)) == yc).mean() > 0.6
+
+    Xb, yb, wb = tm.make_batches(100, 4, 3, seed=3)
+    assert len(Xb) == 3 and Xb[0].shape == (100, 4) and wb[0].min() >= 0
+
+    Xl, yl, qid = tm.make_ltr(600, 5, 20, max_rel=3, seed=4)
+    assert qid.shape == (600,) and (np.diff(qid) >= 0).all()
+    d = xgb.DMatrix(Xl, label=yl)
+    d.set_info(qid=qid)
+    bst = xgb.train({"objective": "rank:ndcg", "max_depth": 3,
+                     "eval_metric": "ndcg"}, d, 5)
+
+    csr, ys = tm.make_sparse_regression(300, 10, sparsity=0.8, seed=5)
+    assert csr.nnz < 300 * 10 * 0.4
+    bst = xgb.train({"max_depth": 3}, xgb.DMatrix(csr, label=ys), 3)
+
+    Xcat, ycat = tm.make_categorical(300, 4, 8, cat_ratio=0.5, seed=6)
+    d = xgb.DMatrix(Xcat, label=ycat, enable_categorical=True)
+    bst = xgb.train({"max_depth": 3}, d, 3)
+    assert np.isfinite(bst.predict(d)).all()
