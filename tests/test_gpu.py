@@ -962,3 +962,31 @@ json.dump(out, open(sys.argv[1], "w"))
             outs.append(json.load(open(of)))
     assert outs[0]["dump"] == outs[1]["dump"]
     assert outs[0]["pred"] == outs[1]["pred"]
+
+
+def test_dart_on_device():
+    """DART on cuda: dropped-tree contributions come from the subset
+    forest predict (predict_subset_gpu) and weighted prediction folds
+    weights into the cached forest's leaf values."""
+    rng = np.random.RandomState(21)
+    Xn = rng.randn(20000, 8).astype(np.float32)
+    yn = (Xn[:, 0] * 1.2 - Xn[:, 1] + 0.1 * rng.randn(20000)).astype(
+        np.float32)
+    d = DMatrix(Xn, label=yn)
+    bst = xgb.train({"max_depth": 4, "eta": 0.3, "device": "cuda",
+                     "rate_drop": 0.4, "one_drop": True, "seed": 9},
+                    d, 10)
+    assert len(bst.weight_drop) == 10
+    assert any(w != 1.0 for w in bst.weight_drop)
+    cached, _ = bst._cache[id(d)]
+    fresh = bst._predict_margin(d)
+    assert torch.allclose(cached, fresh, atol=1e-3), \
+        (cached - fresh).abs().max()
+    # CPU reference: same params/seed on cpu must give the same trees
+    bst_cpu = xgb.train({"max_depth": 4, "eta": 0.3, "device": "cpu",
+                         "rate_drop": 0.4, "one_drop": True, "seed": 9},
+                        d, 10)
+    assert bst.weight_drop == pytest.approx(bst_cpu.weight_drop)
+    pg = bst.predict(d)
+    pc = bst_cpu.predict(d)
+    assert np.allclose(pg, pc, atol=2e-3), np.abs(pg - pc).max()
