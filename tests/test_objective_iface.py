@@ -106,3 +106,21 @@ def test_split_grad_requires_multi_target():
 
     with pytest.raises(ValueError, match="vector-leaf"):
         xgb.train({"max_depth": 3, "base_score": 0.0}, d, 2, obj=Bad())
+
+
+def test_split_grad_with_dart():
+    """Reduced-gradient boosting composes with DART: the vector-leaf
+    tree's margin add is scaled by the DART weight and the dropped
+    trees are renormalized."""
+    import torch
+    X, Y = _mt_data(n=600, seed=9)
+    C = Y.shape[1]
+    d = xgb.DMatrix(X, label=Y)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "num_target": C,
+                     "base_score": 0.0, "rate_drop": 1.0, "seed": 3},
+                    d, 4, obj=Reduced())
+    assert len(bst.weight_drop) == 4
+    assert any(w != 1.0 for w in bst.weight_drop)
+    cached, _ = bst._cache[id(d)]
+    fresh = bst._predict_margin(d)
+    assert torch.allclose(cached, fresh, atol=1e-4)

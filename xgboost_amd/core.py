@@ -720,7 +720,10 @@ class Booster:
         self.trees.append(tree)
         self.tree_info.append(0)
         self.iteration_indptr.append(self.iteration_indptr[-1] + 1)
-        margin += torch.as_tensor(lv[:nn], device=margin.device)[pos]
+        w_new = self._dart_new_weight()
+        add = torch.as_tensor(lv[:nn], device=margin.device)[pos]
+        margin += add if w_new == 1.0 else w_new * add
+        self._dart_commit(margin, 1)
         self._cache[id(dtrain)] = (margin, len(self.trees))
 
     def _update_existing(self, dtrain: DMatrix, iteration: int) -> None:
