@@ -345,16 +345,22 @@ class Booster:
         n = dmat.num_row()
         out = torch.zeros((n, self.n_outputs), dtype=torch.float32,
                           device=self.device)
-        if self.device.type == "cuda":
+        if self.device.type == "cuda" and not any(
+                self.trees[t].leaf_values is not None for t in idxs):
             from .backend.gpu import predict_subset_gpu
             return predict_subset_gpu(self, dmat, idxs, out)
         X = dmat.raw_data()
         for t in idxs:
             tree = self.trees[t]
             pos = tree.predict_leaf_np(X, dmat.missing)
-            vals = tree.split_cond[:tree.n_nodes][pos] * self._tw(t)
-            out[:, self.tree_info[t]] += torch.as_tensor(
-                vals, device=out.device)
+            if tree.leaf_values is not None:  # vector leaves
+                out += self._tw(t) * torch.as_tensor(
+                    tree.leaf_values[:tree.n_nodes][pos],
+                    device=out.device)
+            else:
+                vals = tree.split_cond[:tree.n_nodes][pos] * self._tw(t)
+                out[:, self.tree_info[t]] += torch.as_tensor(
+                    vals, device=out.device)
         return out
 
     # -- training ------------------------------------------------------
