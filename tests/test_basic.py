@@ -280,3 +280,19 @@ def test_arrow_table_ingestion():
     tbl2 = pa.table({"a": pa.array(a), "b": pa.array(b)})
     d2 = xgb.DMatrix(tbl2, label=y)
     assert d2.num_col() == 2
+
+
+def test_deprecated_setters():
+    """set_label/set_weight/set_base_margin/set_group aliases
+    (reference core.py deprecated per-field setters)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(60, 3).astype(np.float32)
+    d = xgb.DMatrix(X)
+    d.set_label(np.arange(60, dtype=np.float32))
+    d.set_weight(np.ones(60, dtype=np.float32))
+    d.set_base_margin(np.zeros(60, dtype=np.float32))
+    assert d.get_label()[-1] == 59
+    assert d.get_weight().sum() == 60
+    d2 = xgb.DMatrix(X, label=np.zeros(60, np.float32))
+    d2.set_group([30, 30])
+    assert list(d2.get_group()) == [30, 30]  # group sizes

@@ -314,6 +314,20 @@ class DMatrix:
                      "label_lower_bound": "label_lower_bound",
                      "label_upper_bound": "label_upper_bound"}
 
+    # deprecated per-field setters kept for API compatibility
+    # (reference core.py set_label/set_weight/set_base_margin/set_group)
+    def set_label(self, label) -> None:
+        self.set_info(label=label)
+
+    def set_weight(self, weight) -> None:
+        self.set_info(weight=weight)
+
+    def set_base_margin(self, margin) -> None:
+        self.set_info(base_margin=margin)
+
+    def set_group(self, group) -> None:
+        self.set_info(group=group)
+
     def get_float_info(self, field: str) -> np.ndarray:
         if field in self._FLOAT_FIELDS:
             v = getattr(self.info, self._FLOAT_FIELDS[field])
