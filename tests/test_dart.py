@@ -167,3 +167,21 @@ def test_dart_slice_carries_weights():
         exp += bst.weight_drop[t] * \
             bst.trees[t].split_cond[:bst.trees[t].n_nodes][pos]
     assert np.allclose(sl.predict(d), exp, atol=1e-5)
+
+
+def test_dart_pickle_and_continuation():
+    """weight_drop survives pickling (save_raw embeds it) and training
+    continuation extends it."""
+    import pickle
+    X, y = _data(400, seed=15)
+    d = xgb.DMatrix(X, label=y)
+    b = xgb.train({"max_depth": 3, "rate_drop": 0.5, "seed": 1}, d, 6)
+    b2 = pickle.loads(pickle.dumps(b))
+    assert b2.weight_drop == pytest.approx(b.weight_drop)
+    assert np.allclose(b2.predict(d), b.predict(d))
+    b3 = xgb.train({"max_depth": 3, "rate_drop": 0.5, "seed": 2}, d, 3,
+                   xgb_model=b)
+    assert len(b3.weight_drop) == len(b3.trees) == 9
+    cached = b3._predict_margin(d)
+    # margin from scratch must be consistent with incremental history
+    assert np.isfinite(cached.numpy()).all()
