@@ -185,3 +185,19 @@ def test_dart_pickle_and_continuation():
     cached = b3._predict_margin(d)
     # margin from scratch must be consistent with incremental history
     assert np.isfinite(cached.numpy()).all()
+
+
+def test_shap_ignores_dart_weights():
+    """pred_contribs follow the reference: the predictor's contribution
+    path sees the raw tree model (no weight_drop), so phi sums to the
+    UNWEIGHTED margin."""
+    X, y = _data(500, seed=17)
+    d = xgb.DMatrix(X, label=y)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 1.0,
+                     "seed": 2}, d, 4)
+    phi = bst.predict(d, pred_contribs=True)
+    unweighted = np.full(len(y), bst.base_score, np.float64)
+    for t in bst.trees:
+        pos = t.predict_leaf_np(X, float("nan"))
+        unweighted += t.split_cond[:t.n_nodes][pos]
+    assert np.allclose(phi.sum(axis=1), unweighted, atol=1e-4)

@@ -782,7 +782,8 @@ class _ForestArrays:
     weights are folded into the LEAF values (margin = sum w_i*tree_i),
     so the predict kernel needs no weight array."""
 
-    def __init__(self, booster, lo: int, hi: int, device, idxs=None):
+    def __init__(self, booster, lo: int, hi: int, device, idxs=None,
+                 fold_weights: bool = False):
         if idxs is None:
             idxs = range(lo, hi)
         idxs = list(idxs)
@@ -808,7 +809,8 @@ class _ForestArrays:
             right[o:o + n] = t.right[:n]
             sidx[o:o + n] = t.split_index[:n]
             cond[o:o + n] = t.split_cond[:n]
-            w = booster._tw(idxs[i]) if hasattr(booster, "_tw") else 1.0
+            w = (booster._tw(idxs[i])
+                 if fold_weights and hasattr(booster, "_tw") else 1.0)
             if w != 1.0:
                 leaf_mask = t.left[:n] == -1
                 cond[o:o + n][leaf_mask] *= np.float32(w)
@@ -1017,7 +1019,7 @@ def _cached_forest(booster, lo: int, hi: int, device) -> _ForestArrays:
     fa = fc.get(key)
     if fa is None:
         fc.clear()  # model changed or different range: drop stale SoA
-        fa = _ForestArrays(booster, lo, hi, device)
+        fa = _ForestArrays(booster, lo, hi, device, fold_weights=True)
         fc[key] = fa
     return fa
 
@@ -1029,7 +1031,8 @@ def predict_subset_gpu(booster, dmat, idxs, out: torch.Tensor
     from .. import ops as hip_ops
     lib = hip_ops.load()
     device = out.device
-    fa = _ForestArrays(booster, 0, 0, device, idxs=idxs)
+    fa = _ForestArrays(booster, 0, 0, device, idxs=idxs,
+                       fold_weights=True)
     dd = dmat.device_data() if hasattr(dmat, "device_data") else None
     X = dd if dd is not None else torch.from_numpy(dmat.raw_data()).to(device)
     n = dmat.num_row()
