@@ -148,3 +148,38 @@ def test_extmem_disk_spill_requires_prefix(batched_data):
     d = ExtMemQuantileDMatrix(NumpyBatchIter(Xs, ys), max_bin=64,
                               max_host_cache_bytes=1)
     assert not any(d.store.is_disk(i) for i in range(4))
+
+
+def test_dart_with_external_memory():
+    """DART's dropped-tree contributions on quantized-only pages use
+    the bin-based traversal (no raw data exists to re-read)."""
+    import torch
+    rng = np.random.RandomState(31)
+    batches = [(rng.randn(300, 5).astype(np.float32),
+                rng.randn(300).astype(np.float32)) for _ in range(3)]
+
+    class It(DataIter):
+        def __init__(self):
+            super().__init__()
+            self.i = 0
+
+        def reset(self):
+            self.i = 0
+
+        def next(self, input_data):
+            if self.i >= len(batches):
+                return False
+            X, y = batches[self.i]
+            input_data(data=X, label=y)
+            self.i += 1
+            return self.i < len(batches)
+
+    d = ExtMemQuantileDMatrix(It(), max_bin=64)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 0.5,
+                     "seed": 3}, d, 6)
+    assert len(bst.weight_drop) == 6
+    assert any(w != 1.0 for w in bst.weight_drop)
+    cached, _ = bst._cache[id(d)]
+    fresh = bst._predict_margin(d)
+    assert torch.allclose(cached, fresh, atol=1e-4), \
+        (cached - fresh).abs().max()

@@ -345,6 +345,21 @@ class Booster:
         n = dmat.num_row()
         out = torch.zeros((n, self.n_outputs), dtype=torch.float32,
                           device=self.device)
+        from .extmem import ExtMemQuantileDMatrix
+        if isinstance(dmat, ExtMemQuantileDMatrix):
+            # quantized-only pages: bin-based traversal per page (the
+            # split conds are cut values, so this is exact)
+            for pi in range(len(dmat.pages)):
+                qm = dmat.page_qm(pi)
+                s, e = dmat.page_offsets[pi], dmat.page_offsets[pi + 1]
+                gg = qm.global_gidx().cpu().numpy()
+                for t in idxs:
+                    tree = self.trees[t]
+                    pos = tree.predict_leaf_bins(gg, dmat.cuts)
+                    vals = tree.split_cond[:tree.n_nodes][pos] * self._tw(t)
+                    out[s:e, self.tree_info[t]] += torch.as_tensor(
+                        vals, device=out.device)
+            return out
         if self.device.type == "cuda" and not any(
                 self.trees[t].leaf_values is not None for t in idxs):
             from .backend.gpu import predict_subset_gpu
