@@ -120,3 +120,23 @@ def test_sparse_gpu_matches_cpu():
                               t2.split_index[:t2.n_nodes])
         assert np.allclose(t1.split_cond[:t1.n_nodes],
                            t2.split_cond[:t2.n_nodes], rtol=1e-4, atol=1e-6)
+
+
+def test_dart_with_sparse_training():
+    """DART dropped-tree margins on scipy CSR training data route
+    through the used-features sparse traversal (no densification of
+    the full matrix)."""
+    import torch
+    from scipy import sparse
+    rng = np.random.RandomState(9)
+    csr = sparse.random(800, 50, density=0.1, format="csr",
+                        dtype=np.float32, random_state=rng)
+    y = np.asarray(csr @ rng.randn(50)).ravel().astype(np.float32)
+    d = xgb.DMatrix(csr, label=y)
+    bst = xgb.train({"max_depth": 3, "eta": 0.3, "rate_drop": 0.5,
+                     "seed": 5}, d, 6)
+    assert len(bst.weight_drop) == 6
+    cached, _ = bst._cache[id(d)]
+    fresh = bst._predict_margin(d)
+    assert torch.allclose(cached, fresh, atol=1e-4), \
+        (cached - fresh).abs().max()

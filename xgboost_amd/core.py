@@ -345,6 +345,8 @@ class Booster:
         n = dmat.num_row()
         out = torch.zeros((n, self.n_outputs), dtype=torch.float32,
                           device=self.device)
+        if getattr(dmat, "_sparse_data", None) is not None:
+            return self._sparse_margin_add(dmat, out, list(idxs))
         from .extmem import ExtMemQuantileDMatrix
         if isinstance(dmat, ExtMemQuantileDMatrix):
             # quantized-only pages: bin-based traversal per page (the
@@ -928,14 +930,18 @@ class Booster:
 
     def _predict_margin_sparse(self, dmat, out: torch.Tensor,
                                lo: int, hi: int) -> torch.Tensor:
+        return self._sparse_margin_add(dmat, out, list(range(lo, hi)))
+
+    def _sparse_margin_add(self, dmat, out: torch.Tensor,
+                           idxs: List[int]) -> torch.Tensor:
         """Sparse predict: densify ONLY the features used by the trees,
         absent entries become NaN (missing -> default direction)."""
-        if hi <= lo:
+        if not idxs:
             return out  # no trees yet (first margin of training)
         csr = dmat.sparse_data()
         n = csr.shape[0]
         used = sorted(set(
-            int(f) for t in range(lo, hi)
+            int(f) for t in idxs
             for nid in range(self.trees[t].n_nodes)
             if not self.trees[t].is_leaf(nid)
             for f in [self.trees[t].split_index[nid]]))
@@ -945,7 +951,7 @@ class Booster:
         sub = csr[:, used].tocoo()
         X_sub = np.full((n, len(used)), np.nan, dtype=np.float32)
         X_sub[sub.row, sub.col] = sub.data
-        for t in range(lo, hi):
+        for t in idxs:
             tree = self.trees[t]
             remap = tree.split_index[:tree.n_nodes].copy()
             saved = remap.copy()
