@@ -296,3 +296,17 @@ def test_deprecated_setters():
     d2 = xgb.DMatrix(X, label=np.zeros(60, np.float32))
     d2.set_group([30, 30])
     assert list(d2.get_group()) == [30, 30]  # group sizes
+
+
+def test_label_validation():
+    """reference data.cc:566: labels must be finite; AFT censoring
+    bounds (which may be +inf) go through label_lower/upper_bound."""
+    X = np.ones((3, 2), np.float32)
+    with pytest.raises(ValueError, match="NaN, infinity"):
+        xgb.DMatrix(X, label=np.array([1, np.nan, 2], np.float32))
+    d = xgb.DMatrix(X)
+    with pytest.raises(ValueError, match="NaN, infinity"):
+        d.set_info(label=np.array([np.inf, 0, 1], np.float32))
+    d.set_info(label=np.array([0, 1, 2], np.float32),
+               label_lower_bound=np.zeros(3, np.float32),
+               label_upper_bound=np.full(3, np.inf, np.float32))

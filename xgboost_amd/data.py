@@ -223,7 +223,7 @@ class DMatrix:
             n_row, n_col = self._device_data.shape
         self.info = MetaInfo(num_row=n_row, num_col=n_col)
         if label is not None:
-            self.info.labels = _as_float_array(label)
+            self.info.labels = _checked_labels(label)
         if weight is not None:
             self.info.weights = _as_float_array(weight).reshape(-1)
         if base_margin is not None:
@@ -277,7 +277,7 @@ class DMatrix:
                  label_lower_bound=None, label_upper_bound=None,
                  feature_weights=None) -> None:
         if label is not None:
-            self.info.labels = _as_float_array(label)
+            self.info.labels = _checked_labels(label)
         if weight is not None:
             self.info.weights = _as_float_array(weight).reshape(-1)
         if base_margin is not None:
@@ -566,6 +566,16 @@ def _as_float_array(v: Any) -> np.ndarray:
     if isinstance(v, torch.Tensor):
         return v.detach().cpu().numpy().astype(np.float32, copy=False)
     return np.asarray(v, dtype=np.float32)
+
+
+
+def _checked_labels(label) -> np.ndarray:
+    """reference data.cc:566 LabelsCheck: labels must be finite (AFT
+    censoring bounds live in label_lower/upper_bound, not here)."""
+    lab = _as_float_array(label)
+    if lab.size and not np.isfinite(lab).all():
+        raise ValueError("Label contains NaN, infinity or a value too large.")
+    return lab
 
 
 def _ingest(data: Any, enable_categorical: bool):
