@@ -118,3 +118,41 @@ def test_feature_weights_demo():
     w0 = score.get("f0", 0)
     others = [score.get(f"f{i}", 0) for i in range(1, 8)]
     assert w0 > np.mean(others)
+
+
+def test_custom_softmax_idiom():
+    """demo/guide-python/custom_softmax.py: multiclass custom objective
+    via raw-margin softmax gradients matches the builtin."""
+    rng = np.random.RandomState(4)
+    n, f, C = 900, 6, 3
+    X = rng.randn(n, f).astype(np.float32)
+    y = np.argmax(X[:, :C] + 0.3 * rng.randn(n, C), axis=1).astype(
+        np.float32)
+
+    def softprob_obj(preds, dmat):
+        labels = dmat.get_label().astype(int)
+        m = preds.reshape(n, C)
+        e = np.exp(m - m.max(axis=1, keepdims=True))
+        p = e / e.sum(axis=1, keepdims=True)
+        grad = p.copy()
+        grad[np.arange(n), labels] -= 1.0
+        hess = np.maximum(2.0 * p * (1.0 - p), 1e-6)
+        return grad, hess
+
+    params = {"max_depth": 4, "eta": 0.3, "num_class": C,
+              "base_score": 0.5, "seed": 0}
+    d1 = xgb.DMatrix(X, label=y)
+    custom = xgb.train(dict(params, objective="multi:softmax",
+                            disable_default_eval_metric=1),
+                       d1, 8, obj=softprob_obj)
+    d2 = xgb.DMatrix(X, label=y)
+    builtin = xgb.train(dict(params, objective="multi:softprob"), d2, 8)
+    pc = custom.predict(d1, output_margin=True).reshape(n, C).argmax(1)
+    pb = builtin.predict(d2).reshape(n, C).argmax(1)
+    assert (pc == pb).mean() > 0.95
+    assert (pc == y).mean() > 0.85
+
+
+def test_reduced_gradient_demo_runs():
+    import demo.reduced_gradient as rg
+    rg.main()
