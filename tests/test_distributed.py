@@ -55,9 +55,21 @@ dist.destroy_process_group()
 """
 
 
+
+
+def _free_port() -> int:
+    """OS-assigned free port: fixed ports collide when the suite runs
+    under pytest-xdist (parallel workers) or right after a crashed run
+    leaves TIME_WAIT sockets."""
+    import socket
+    with socket.socket() as sock:
+        sock.bind(("127.0.0.1", 0))
+        return sock.getsockname()[1]
+
+
 def _run_workers(world_size: int, script: str, env_extra=None) -> None:
     procs = []
-    port = 29517
+    port = _free_port()
     with tempfile.TemporaryDirectory() as td:
         spath = os.path.join(td, "worker.py")
         with open(spath, "w") as fh:
@@ -151,7 +163,7 @@ def test_hist_allreduce_exact():
 
 def _run_workers_simple(world_size: int, script: str) -> None:
     procs = []
-    port = 29531
+    port = _free_port()
     with tempfile.TemporaryDirectory() as td:
         spath = os.path.join(td, "worker.py")
         with open(spath, "w") as fh:
@@ -560,7 +572,7 @@ def test_collective_failure_detection():
     """A dead worker must surface as an error on the survivors within
     the watchdog budget instead of hanging the job."""
     procs = []
-    port = 29611
+    port = _free_port()
     with tempfile.TemporaryDirectory() as td:
         spath = os.path.join(td, "worker.py")
         with open(spath, "w") as fh:
@@ -649,7 +661,7 @@ def test_two_process_dart_identical_models():
     """DART under data-parallel training: the drop set comes from a
     seed-deterministic RNG and tree counts, both rank-identical, so
     every rank must produce the SAME model and weights."""
-    port = 29553
+    port = _free_port()
     with tempfile.TemporaryDirectory() as td:
         spath = os.path.join(td, "worker.py")
         with open(spath, "w") as fh:
