@@ -201,3 +201,21 @@ def test_shap_ignores_dart_weights():
         pos = t.predict_leaf_np(X, float("nan"))
         unweighted += t.split_cond[:t.n_nodes][pos]
     assert np.allclose(phi.sum(axis=1), unweighted, atol=1e-4)
+
+
+def test_dart_with_cv_and_early_stopping():
+    """xgb.cv and EarlyStopping work with dropout configured."""
+    X, y = _data(900, seed=19)
+    ycls = (y > np.median(y)).astype(np.float32)
+    d = xgb.DMatrix(X, label=ycls)
+    res = xgb.cv({"objective": "binary:logistic", "max_depth": 3,
+                  "rate_drop": 0.3, "seed": 1, "eval_metric": "logloss"},
+                 d, num_boost_round=8, nfold=3)
+    assert len(res["test-logloss-mean"]) == 8
+    es = xgb.callback.EarlyStopping(rounds=3, save_best=True)
+    bst = xgb.train({"objective": "binary:logistic", "max_depth": 3,
+                     "rate_drop": 0.3, "seed": 1,
+                     "eval_metric": "logloss"}, d, 30,
+                    evals=[(d, "t")], callbacks=[es], verbose_eval=False)
+    assert bst.num_boosted_rounds() <= 30
+    assert len(bst.weight_drop) == len(bst.trees)
