@@ -125,3 +125,22 @@ def test_multi_target_one_output_per_tree():
     assert len(bst.trees) == 20  # one tree per target per round
     p = bst.predict(d)
     assert p.shape == (1000, 2)
+
+
+def test_exact_rejects_unsupported():
+    """reference updater_colmaker.cc:104-113: exact rejects categorical
+    data, external memory and colsample_bynode."""
+    import pandas as pd
+    rng = np.random.RandomState(0)
+    y = rng.randn(60).astype(np.float32)
+    Xc = pd.DataFrame({
+        "c": pd.Series(rng.randint(0, 3, 60)).astype("category"),
+        "n": rng.randn(60).astype(np.float32)})
+    dc = xgb.DMatrix(Xc, label=y, enable_categorical=True)
+    with pytest.raises(ValueError, match="categorical"):
+        xgb.train({"tree_method": "exact", "max_depth": 2}, dc, 1)
+    X = rng.randn(60, 3).astype(np.float32)
+    d = xgb.DMatrix(X, label=y)
+    with pytest.raises(ValueError, match="colsample_bynode|column sample"):
+        xgb.train({"tree_method": "exact", "max_depth": 2,
+                   "colsample_bynode": 0.5}, d, 1)

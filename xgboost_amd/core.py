@@ -789,6 +789,22 @@ class Booster:
         from .exact import grow_exact
         if self.device.type == "cuda":
             raise ValueError("tree_method=exact is CPU-only; use hist on GPU")
+        # reference updater_colmaker.cc:104-113: the exact method
+        # rejects categorical data, external memory and colsample_bynode
+        ft = dtrain.feature_types or []
+        if any(t in ("c", "categorical") for t in ft) or \
+                getattr(dtrain, "categories_", None):
+            raise ValueError(
+                "Updater `grow_colmaker` or `exact` tree method does not"
+                " support categorical data")
+        from .extmem import ExtMemQuantileDMatrix
+        if isinstance(dtrain, ExtMemQuantileDMatrix):
+            raise ValueError(
+                "Updater `grow_colmaker` or `exact` tree method doesn't "
+                "support external memory training")
+        if self.tparam.colsample_bynode != 1.0:
+            raise ValueError("column sample by node is not yet supported "
+                             "by the exact tree method")
         n = dtrain.num_row()
         n_out = grad.shape[1]
         margin, _ = self._cache[id(dtrain)]
