@@ -96,3 +96,21 @@ def test_gblinear_multiclass():
     p = bst.predict(d)
     assert p.shape == (600, 3)
     assert np.allclose(p.sum(axis=1), 1.0, atol=1e-5)
+
+
+def test_gblinear_rejects_categorical_and_shotgun_selectors():
+    """reference gblinear.cc:129 (NoCategorical) and
+    updater_shotgun.cc:20 (cyclic/shuffle only)."""
+    import pandas as pd
+    rng = np.random.RandomState(0)
+    y = rng.randn(60).astype(np.float32)
+    Xc = pd.DataFrame({
+        "c": pd.Series(rng.randint(0, 3, 60)).astype("category"),
+        "n": rng.randn(60).astype(np.float32)})
+    dc = xgb.DMatrix(Xc, label=y, enable_categorical=True)
+    with pytest.raises(ValueError, match="categorical"):
+        xgb.train({"booster": "gblinear"}, dc, 1)
+    d = xgb.DMatrix(rng.randn(60, 3).astype(np.float32), label=y)
+    with pytest.raises(ValueError, match="shotgun"):
+        xgb.train({"booster": "gblinear", "updater": "shotgun",
+                   "feature_selector": "greedy"}, d, 1)

@@ -617,6 +617,12 @@ class Booster:
     def _boost_linear(self, dtrain: DMatrix, grad: torch.Tensor,
                       hess: torch.Tensor, iteration: int) -> None:
         from .linear import GBLinearModel
+        # reference gblinear.cc:129: linear boosting has no encoding for
+        # categorical inputs
+        ft = dtrain.feature_types or []
+        if any(t in ("c", "categorical") for t in ft) or \
+                getattr(dtrain, "categories_", None):
+            raise ValueError("`gblinear` does not support categorical data")
         if self._linear is None:
             self._linear = GBLinearModel(self.n_features, self.n_outputs,
                                          self.raw_params, self.device)

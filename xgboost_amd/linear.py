@@ -38,6 +38,13 @@ class GBLinearModel:
         self.top_k = int(p.get("top_k", 0))
         self.updater = str(p.get("updater", "coord_descent")
                            if p.get("updater") else "coord_descent")
+        if self.updater == "shotgun" and self.feature_selector not in (
+                "cyclic", "shuffle"):
+            # reference updater_shotgun.cc:20: parallel shotgun steps
+            # cannot honor gradient-ordered selectors
+            raise ValueError(
+                "Unsupported feature selector for shotgun updater. "
+                "Supported options are: {cyclic, shuffle}")
 
     def predict_margin(self, X: torch.Tensor) -> torch.Tensor:
         return X @ self.weights[:-1] + self.weights[-1]
