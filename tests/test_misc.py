@@ -456,3 +456,16 @@ def test_testing_module_generators():
     d = xgb.DMatrix(Xcat, label=ycat, enable_categorical=True)
     bst = xgb.train({"max_depth": 3}, d, 3)
     assert np.isfinite(bst.predict(d)).all()
+
+
+def test_vector_leaf_requires_hist():
+    """reference gbtree.cc:187: multi_output_tree rejects non-hist
+    tree methods."""
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(80, 4).astype(np.float32)
+    Y = rng.randn(80, 2).astype(np.float32)
+    d = xgb.DMatrix(X, label=Y)
+    with pytest.raises(ValueError, match="hist tree method"):
+        xgb.train({"multi_strategy": "multi_output_tree",
+                   "tree_method": "approx", "max_depth": 3}, d, 1)

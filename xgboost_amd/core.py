@@ -453,6 +453,11 @@ class Booster:
             self._boost_linear(dtrain, grad, hess, iteration)
             return
         if self.tparam.tree_method == "exact":
+            if str(self.raw_params.get(
+                    "multi_strategy", "one_output_per_tree")) ==                     "multi_output_tree" and                     (grad.dim() > 1 and grad.shape[1] > 1):
+                raise ValueError(
+                    "Only the hist tree method is supported for building "
+                    "multi-target trees with vector leaf.")
             self._boost_exact(dtrain, grad, hess, iteration)
             return
         ops = self._ops_for(dtrain)
@@ -468,6 +473,11 @@ class Booster:
         multi_strategy = str(self.raw_params.get("multi_strategy",
                                                  "one_output_per_tree"))
         if multi_strategy == "multi_output_tree" and n_out > 1:
+            # reference gbtree.cc:187: vector-leaf trees are hist-only
+            if self.tparam.tree_method not in ("auto", "hist"):
+                raise ValueError(
+                    "Only the hist tree method is supported for building "
+                    "multi-target trees with vector leaf.")
             self._boost_multi_target(dtrain, ops, grad, hess, margin, seed)
             return
         is_approx = self.tparam.tree_method == "approx"
