@@ -310,3 +310,17 @@ def test_label_validation():
     d.set_info(label=np.array([0, 1, 2], np.float32),
                label_lower_bound=np.zeros(3, np.float32),
                label_upper_bound=np.full(3, np.inf, np.float32))
+
+
+def test_quantile_dmatrix_hist_only():
+    """reference IterativeDMatrix: no SparsePage for exact
+    (iterative_dmatrix.cc:159), no cut regeneration for approx
+    (:129) — QuantileDMatrix trains with hist only."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(100, 4).astype(np.float32)
+    y = rng.randn(100).astype(np.float32)
+    qd = xgb.QuantileDMatrix(X, label=y, max_bin=32)
+    for tm in ("exact", "approx"):
+        with pytest.raises(ValueError, match="QuantileDMatrix"):
+            xgb.train({"tree_method": tm, "max_depth": 2}, qd, 1)
+    xgb.train({"tree_method": "hist", "max_depth": 2}, qd, 1)

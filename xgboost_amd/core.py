@@ -452,6 +452,16 @@ class Booster:
         if self.booster_kind == "gblinear":
             self._boost_linear(dtrain, grad, hess, iteration)
             return
+        from .data import QuantileDMatrix as _QDM
+        if self.tparam.tree_method in ("exact", "approx") and \
+                type(dtrain) is _QDM:
+            # reference: QuantileDMatrix carries only quantized pages —
+            # exact has no SparsePage to walk (iterative_dmatrix.cc:159)
+            # and approx would need hessian-weighted cut regeneration
+            # (iterative_dmatrix.cc:129 RegenGHist CHECK)
+            raise ValueError(
+                f"tree_method={self.tparam.tree_method} does not support "
+                "QuantileDMatrix; use hist or a plain DMatrix")
         if self.tparam.tree_method == "exact":
             if str(self.raw_params.get(
                     "multi_strategy", "one_output_per_tree")) ==                     "multi_output_tree" and                     (grad.dim() > 1 and grad.shape[1] > 1):
