@@ -469,3 +469,14 @@ def test_vector_leaf_requires_hist():
     with pytest.raises(ValueError, match="hist tree method"):
         xgb.train({"multi_strategy": "multi_output_tree",
                    "tree_method": "approx", "max_depth": 3}, d, 1)
+
+
+def test_param_lower_bounds():
+    """reference param.h field bounds: eta >= 0, max_depth >= 0."""
+    import xgboost_amd as xgb
+    X = np.ones((20, 2), np.float32)
+    d = xgb.DMatrix(X, label=np.zeros(20, np.float32))
+    with pytest.raises(ValueError, match="eta"):
+        xgb.train({"eta": -0.1, "max_depth": 2}, d, 1)
+    with pytest.raises(ValueError, match="max_depth"):
+        xgb.train({"max_depth": -2}, d, 1)

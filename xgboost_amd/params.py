@@ -114,6 +114,12 @@ def _validate(tp: TrainParam) -> None:
         raise ValueError(f"unknown tree_method: {tp.tree_method}")
     if tp.max_bin < 2:
         raise ValueError("max_bin must be >= 2")
+    if tp.eta < 0:
+        raise ValueError("eta (learning_rate) must be >= 0 "
+                         "(reference param.h:83 lower bound)")
+    if tp.max_depth < 0:
+        raise ValueError("max_depth must be >= 0 "
+                         "(0 = unbounded; reference param.h:90)")
     if not (0.0 < tp.subsample <= 1.0):
         raise ValueError("subsample must be in (0, 1]")
     for name in ("colsample_bytree", "colsample_bylevel", "colsample_bynode"):
