@@ -324,3 +324,22 @@ def test_quantile_dmatrix_hist_only():
         with pytest.raises(ValueError, match="QuantileDMatrix"):
             xgb.train({"tree_method": tm, "max_depth": 2}, qd, 1)
     xgb.train({"tree_method": "hist", "max_depth": 2}, qd, 1)
+
+
+def test_feature_names_validation():
+    """reference _validate_features: predict-frame names (incl. order)
+    must match training names."""
+    import pandas as pd
+    rng = np.random.RandomState(0)
+    X = pd.DataFrame({"a": rng.randn(50).astype(np.float32),
+                      "b": rng.randn(50).astype(np.float32)})
+    y = rng.randn(50).astype(np.float32)
+    bst = xgb.train({"max_depth": 2}, xgb.DMatrix(X, label=y), 2)
+    with pytest.raises(ValueError, match="feature_names mismatch"):
+        bst.predict(xgb.DMatrix(X.rename(columns={"b": "c"})))
+    with pytest.raises(ValueError, match="feature_names mismatch"):
+        bst.predict(xgb.DMatrix(X[["b", "a"]]))
+    p = bst.predict(xgb.DMatrix(X))  # matching names fine
+    # validate_features=False skips the check
+    p2 = bst.predict(xgb.DMatrix(X[["b", "a"]]), validate_features=False)
+    assert p.shape == p2.shape

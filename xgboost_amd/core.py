@@ -1046,6 +1046,19 @@ class Booster:
                 and data.num_col() != self.n_features:
             raise ValueError(
                 f"feature mismatch: {data.num_col()} vs {self.n_features}")
+        if validate_features and self.feature_names and data.feature_names \
+                and list(self.feature_names) != list(data.feature_names):
+            # reference core.py _validate_features: names (and order)
+            # must match the training frame
+            trained = set(self.feature_names)
+            given = set(data.feature_names)
+            raise ValueError(
+                "feature_names mismatch: "
+                f"{list(self.feature_names)} vs {list(data.feature_names)}"
+                + (f"; missing from data: {sorted(trained - given)}"
+                   if trained - given else "")
+                + (f"; unexpected in data: {sorted(given - trained)}"
+                   if given - trained else ""))
         if pred_leaf:
             lo, hi = self._tree_range(iteration_range)
             X = data.raw_data()
