@@ -308,3 +308,14 @@ def test_unbiased_trains_end_to_end():
     hits = sum(rel[g * gsz + np.argmax(p[g * gsz:(g + 1) * gsz])]
                for g in range(n_g))
     assert hits > n_g * 0.7
+
+
+def test_qid_must_be_sorted():
+    """reference data.cc:621."""
+    import xgboost_amd as xgb
+    X = np.random.RandomState(0).randn(10, 2).astype(np.float32)
+    d = xgb.DMatrix(X, label=np.zeros(10, np.float32))
+    with pytest.raises(ValueError, match="non-decreasing"):
+        d.set_info(qid=np.array([1, 0] * 5))
+    d.set_info(qid=np.array([0] * 5 + [1] * 5))
+    assert list(d.info.group_ptr) == [0, 5, 10]

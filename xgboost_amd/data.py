@@ -291,6 +291,10 @@ class DMatrix:
             self.info.group_ptr = np.concatenate([[0], np.cumsum(g)]).astype(np.int64)
         if qid is not None:
             q = np.asarray(qid).reshape(-1)
+            if q.size > 1 and not (q[1:] >= q[:-1]).all():
+                # reference data.cc:621
+                raise ValueError("`qid` must be sorted in non-decreasing "
+                                 "order along with data.")
             boundaries = np.nonzero(np.diff(q))[0] + 1
             self.info.group_ptr = np.concatenate([[0], boundaries, [q.size]]).astype(np.int64)
         if feature_names is not None:
