@@ -319,3 +319,15 @@ def test_qid_must_be_sorted():
         d.set_info(qid=np.array([1, 0] * 5))
     d.set_info(qid=np.array([0] * 5 + [1] * 5))
     assert list(d.info.group_ptr) == [0, 5, 10]
+
+
+def test_group_sizes_must_cover_rows():
+    """reference ValidateQueryGroup: group sizes must sum to num_row."""
+    import xgboost_amd as xgb
+    X = np.random.RandomState(0).randn(30, 2).astype(np.float32)
+    d = xgb.DMatrix(X, label=np.zeros(30, np.float32))
+    with pytest.raises(ValueError, match="Invalid group structure"):
+        d.set_info(group=[10, 10])
+    with pytest.raises(ValueError, match="Invalid group structure"):
+        xgb.DMatrix(X, label=np.zeros(30, np.float32), group=[40])
+    d.set_info(group=[15, 15])

@@ -234,7 +234,14 @@ class DMatrix:
             self.info.label_upper_bound = _as_float_array(label_upper_bound).reshape(-1)
         if group is not None:
             g = np.asarray(group, dtype=np.int64).reshape(-1)
-            self.info.group_ptr = np.concatenate([[0], np.cumsum(g)]).astype(np.int64)
+            self.info.group_ptr = np.concatenate(
+                [[0], np.cumsum(g)]).astype(np.int64)
+            if int(self.info.group_ptr[-1]) != self.info.num_row:
+                # reference metainfo.h ValidateQueryGroup
+                raise ValueError(
+                    "Invalid group structure: group sizes sum to "
+                    f"{int(self.info.group_ptr[-1])} but the data has "
+                    f"{self.info.num_row} rows")
         elif qid is not None:
             q = np.asarray(qid).reshape(-1)
             if np.any(q[1:] < q[:-1]):
@@ -288,7 +295,14 @@ class DMatrix:
             self.info.label_upper_bound = _as_float_array(label_upper_bound).reshape(-1)
         if group is not None:
             g = np.asarray(group, dtype=np.int64).reshape(-1)
-            self.info.group_ptr = np.concatenate([[0], np.cumsum(g)]).astype(np.int64)
+            self.info.group_ptr = np.concatenate(
+                [[0], np.cumsum(g)]).astype(np.int64)
+            if int(self.info.group_ptr[-1]) != self.info.num_row:
+                # reference metainfo.h ValidateQueryGroup
+                raise ValueError(
+                    "Invalid group structure: group sizes sum to "
+                    f"{int(self.info.group_ptr[-1])} but the data has "
+                    f"{self.info.num_row} rows")
         if qid is not None:
             q = np.asarray(qid).reshape(-1)
             if q.size > 1 and not (q[1:] >= q[:-1]).all():
