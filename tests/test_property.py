@@ -106,3 +106,31 @@ def test_quantile_sketch_rank_bound(seed):
     # binning never loses rows and respects cut boundaries
     binned = np.searchsorted(c, X[:, 0], side="left")
     assert binned.max() < len(c)
+
+
+_json_val = st.recursive(
+    st.one_of(st.none(), st.booleans(),
+              st.integers(-2 ** 62, 2 ** 62),
+              st.floats(allow_nan=False, width=64),
+              st.text(max_size=20)),
+    lambda children: st.one_of(
+        st.lists(children, max_size=5),
+        st.dictionaries(st.text(max_size=8), children, max_size=5)),
+    max_leaves=25)
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(_json_val)
+def test_ubjson_roundtrip_fuzz(v):
+    """UBJSON draft-12 writer/reader round-trips arbitrary JSON values
+    (the model format's binary carrier)."""
+    from xgboost_amd.ubjson import dumps_ubjson, loads_ubjson
+    assert loads_ubjson(dumps_ubjson(v)) == v
+
+
+def test_ubjson_special_floats():
+    import math
+    from xgboost_amd.ubjson import dumps_ubjson, loads_ubjson
+    assert loads_ubjson(dumps_ubjson([float("inf"), float("-inf")])) == \
+        [float("inf"), float("-inf")]
+    assert math.isnan(loads_ubjson(dumps_ubjson([float("nan")]))[0])
