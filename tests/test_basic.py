@@ -343,3 +343,24 @@ def test_feature_names_validation():
     # validate_features=False skips the check
     p2 = bst.predict(xgb.DMatrix(X[["b", "a"]]), validate_features=False)
     assert p.shape == p2.shape
+
+
+def test_trees_to_dataframe_columns():
+    """reference core.py:3259 column set, incl. Category for
+    categorical splits."""
+    import pandas as pd
+    rng = np.random.RandomState(0)
+    Xc = pd.DataFrame({
+        "c": pd.Series(rng.randint(0, 6, 300)).astype("category"),
+        "n": rng.randn(300).astype(np.float32)})
+    y = (Xc["c"].cat.codes.to_numpy() % 2 +
+         rng.randn(300) * 0.1).astype(np.float32)
+    bst = xgb.train({"max_depth": 3, "max_cat_to_onehot": 1}, 
+                    xgb.DMatrix(Xc, label=y, enable_categorical=True), 3)
+    df = bst.trees_to_dataframe()
+    assert list(df.columns) == [
+        "Tree", "Target", "Node", "ID", "Feature", "Split", "Yes", "No",
+        "Missing", "Gain", "Cover", "Category"]
+    cat_rows = df[df["Category"].notna()]
+    assert len(cat_rows) > 0  # categorical splits present
+    assert all(isinstance(c, list) for c in cat_rows["Category"])

@@ -1508,17 +1508,26 @@ class Booster:
         return self.get_score(fmap, "weight")
 
     def trees_to_dataframe(self, fmap: str = ""):
+        """reference core.py trees_to_dataframe (:3259): columns
+        Tree/Target/Node/ID/Feature/Split/Yes/No/Missing/Gain/Cover/
+        Category (Category holds the go-right category list for
+        categorical splits)."""
         import pandas as pd
         rows = []
         for ti, tree in enumerate(self.trees):
             for nid in range(tree.n_nodes):
                 leaf = tree.is_leaf(nid)
+                is_cat = (not leaf and tree.split_type[nid] == 1)
+                cats = (list(map(int, tree.cat_segments.get(nid, [])))
+                        if is_cat else None)
                 rows.append({
-                    "Tree": ti, "Node": nid, "ID": f"{ti}-{nid}",
+                    "Tree": ti, "Target": 0, "Node": nid,
+                    "ID": f"{ti}-{nid}",
                     "Feature": "Leaf" if leaf else
                     (self.feature_names[tree.split_index[nid]]
                      if self.feature_names else f"f{tree.split_index[nid]}"),
-                    "Split": None if leaf else float(tree.split_cond[nid]),
+                    "Split": None if (leaf or is_cat)
+                    else float(tree.split_cond[nid]),
                     "Yes": None if leaf else f"{ti}-{tree.left[nid]}",
                     "No": None if leaf else f"{ti}-{tree.right[nid]}",
                     "Missing": None if leaf else
@@ -1526,6 +1535,7 @@ class Booster:
                     "Gain": float(tree.split_cond[nid]) if leaf
                     else float(tree.loss_chg[nid]),
                     "Cover": float(tree.sum_hess[nid]),
+                    "Category": cats,
                 })
         return pd.DataFrame(rows)
 
