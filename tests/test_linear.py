@@ -114,3 +114,18 @@ def test_gblinear_rejects_categorical_and_shotgun_selectors():
     with pytest.raises(ValueError, match="shotgun"):
         xgb.train({"booster": "gblinear", "updater": "shotgun",
                    "feature_selector": "greedy"}, d, 1)
+
+
+def test_gblinear_feature_importance_is_coefficients():
+    """reference gblinear.cc:210 FeatureScore: weight importance = the
+    coefficients (bias excluded); other types are undefined."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(300, 4).astype(np.float32)
+    y = (X[:, 0] * 2 + X[:, 1]).astype(np.float32)
+    bst = xgb.train({"booster": "gblinear", "eta": 0.5}, 
+                    xgb.DMatrix(X, label=y), 20)
+    s = bst.get_score(importance_type="weight")
+    assert set(s) == {"f0", "f1", "f2", "f3"}
+    assert abs(s["f0"]) > abs(s["f2"])  # real coefficient magnitudes
+    with pytest.raises(ValueError, match="weight"):
+        bst.get_score(importance_type="gain")

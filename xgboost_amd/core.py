@@ -1456,6 +1456,24 @@ class Booster:
         if importance_type not in ("weight", "gain", "cover",
                                    "total_gain", "total_cover"):
             raise ValueError(f"unknown importance_type {importance_type}")
+        if self.booster_kind == "gblinear":
+            # reference gblinear.cc:210 FeatureScore: the per-feature
+            # coefficients (bias excluded); only `weight` is defined
+            if importance_type != "weight":
+                raise ValueError("gblinear only has `weight` defined "
+                                 "for feature importance.")
+            if self._linear is None:
+                return {}
+            names = self._fmap_names(fmap)
+            w = self._linear.weights[:-1].detach().cpu().numpy()
+
+            def lname(f):
+                return names[f] if names and f < len(names) else f"f{f}"
+
+            if w.shape[1] == 1:
+                return {lname(f): float(w[f, 0]) for f in range(w.shape[0])}
+            return {lname(f): [float(x) for x in w[f]]
+                    for f in range(w.shape[0])}
         counts: Dict[int, float] = {}
         sums: Dict[int, float] = {}
         for tree in self.trees:
