@@ -180,3 +180,25 @@ def test_sklearn_callable_objective():
                           objective="reg:squarederror").fit(X, y)
     p1, p2 = m1.predict(X), m2.predict(X)
     assert np.allclose(p1, p2, atol=1e-5), np.abs(p1 - p2).max()
+
+
+def test_sklearn_meta_estimator_compat():
+    """sklearn >= 1.6 tags protocol: GridSearchCV / cross_val_score /
+    Pipeline work (estimators inherit BaseEstimator + mixins like the
+    reference's XGBModelBase)."""
+    from sklearn.model_selection import GridSearchCV, cross_val_score
+    from sklearn.pipeline import Pipeline
+    from sklearn.preprocessing import StandardScaler
+    rng = np.random.RandomState(0)
+    X = rng.randn(150, 4).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    gs = GridSearchCV(xgb.XGBClassifier(n_estimators=4),
+                      {"max_depth": [2, 3]}, cv=2)
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.8
+    sc = cross_val_score(xgb.XGBRegressor(n_estimators=4), X, X[:, 0], cv=2)
+    assert (sc > 0.5).all()
+    pipe = Pipeline([("s", StandardScaler()),
+                     ("m", xgb.XGBRegressor(n_estimators=4))])
+    pipe.fit(X, X[:, 0])
+    assert pipe.score(X, X[:, 0]) > 0.5

@@ -14,6 +14,24 @@ from .core import Booster
 from .data import DMatrix
 from .training import train as _train
 
+# Inherit scikit-learn's estimator protocol when it is installed
+# (reference sklearn.py XGBModelBase extends BaseEstimator through the
+# compat shim): modern sklearn (>=1.6) meta-estimators call
+# `__sklearn_tags__`, which only BaseEstimator + the mixins provide.
+try:
+    from sklearn.base import (BaseEstimator as _SklBaseEstimator,
+                              ClassifierMixin as _SklClassifierMixin,
+                              RegressorMixin as _SklRegressorMixin)
+except ImportError:  # sklearn not installed: plain classes
+    class _SklBaseEstimator:  # type: ignore[no-redef]
+        pass
+
+    class _SklClassifierMixin:  # type: ignore[no-redef]
+        pass
+
+    class _SklRegressorMixin:  # type: ignore[no-redef]
+        pass
+
 _PARAM_NAMES = [
     "max_depth", "max_leaves", "max_bin", "grow_policy", "learning_rate",
     "n_estimators", "verbosity", "objective", "booster", "tree_method",
@@ -29,7 +47,7 @@ _PARAM_NAMES = [
 ]
 
 
-class XGBModel:
+class XGBModel(_SklBaseEstimator):
     """Base sklearn-style estimator (reference sklearn.py:868)."""
 
     _estimator_type = "regressor"
@@ -275,14 +293,11 @@ class XGBModel:
         self._Booster.load_model(fname)
         self.n_features_in_ = self._Booster.num_features()
 
-    def __sklearn_tags__(self):
-        # minimal sklearn >=1.6 tags protocol
-        class T:
-            estimator_type = self._estimator_type
-        return T()
+    # __sklearn_tags__ comes from sklearn's BaseEstimator + the
+    # Classifier/Regressor mixins (sklearn >= 1.6 tags protocol)
 
 
-class XGBRegressor(XGBModel):
+class XGBRegressor(_SklRegressorMixin, XGBModel):
     _estimator_type = "regressor"
 
     def _default_objective(self) -> str:
@@ -293,7 +308,7 @@ class XGBRegressor(XGBModel):
         return r2_score(y, self.predict(X), sample_weight=sample_weight)
 
 
-class XGBClassifier(XGBModel):
+class XGBClassifier(_SklClassifierMixin, XGBModel):
     _estimator_type = "classifier"
 
     def _default_objective(self) -> str:
