@@ -364,3 +364,24 @@ def test_trees_to_dataframe_columns():
     cat_rows = df[df["Category"].notna()]
     assert len(cat_rows) > 0  # categorical splits present
     assert all(isinstance(c, list) for c in cat_rows["Category"])
+
+
+def test_pandas_nullable_and_bad_dtypes():
+    """pandas nullable extension dtypes (Int64/Float64/boolean) convert
+    with NA -> NaN; non-numeric dtypes are rejected (reference pandas
+    adapter semantics)."""
+    import pandas as pd
+    df = pd.DataFrame({
+        "i": pd.array([1, None, 3, 4] * 10, dtype="Int64"),
+        "f": pd.array([0.5, 1.5, None, 2.5] * 10, dtype="Float64"),
+        "b": pd.array([True, False, None, True] * 10, dtype="boolean"),
+    })
+    y = np.arange(40, dtype=np.float32)
+    d = xgb.DMatrix(df, label=y)
+    X0 = d.raw_data()
+    assert np.isnan(X0[1, 0]) and np.isnan(X0[2, 1]) and np.isnan(X0[2, 2])
+    assert X0[0, 0] == 1.0 and X0[0, 2] == 1.0
+    xgb.train({"max_depth": 2}, d, 2)
+    with pytest.raises(ValueError, match="int, float, bool or category"):
+        xgb.DMatrix(pd.DataFrame({"t": pd.date_range("2020", periods=5)}),
+                    label=y[:5])

@@ -691,9 +691,23 @@ def _from_pandas(df, enable_categorical: bool):
             categories[j] = list(s.cat.categories)
             types.append("c")
         else:
-            v = s.to_numpy(np.float32, na_value=np.nan)
+            import pandas.api.types as pdt
+            # accepts numpy AND pandas nullable extension dtypes
+            # (Int64/Float64/boolean, NA -> NaN); anything non-numeric
+            # is rejected like the reference data adapter
+            if pdt.is_bool_dtype(s.dtype):
+                kind = "int"
+            elif pdt.is_integer_dtype(s.dtype):
+                kind = "int"
+            elif pdt.is_float_dtype(s.dtype):
+                kind = "float"
+            else:
+                raise ValueError(
+                    "DataFrame.dtypes for data must be int, float, bool "
+                    f"or category; column {c!r} is {s.dtype}")
+            v = s.to_numpy(dtype=np.float32, na_value=np.nan)
             cols.append(v)
-            types.append("int" if np.issubdtype(s.dtype, np.integer) else "float")
+            types.append(kind)
     X = np.stack(cols, axis=1)
     if "c" not in types:
         types_out = None
