@@ -172,3 +172,18 @@ def test_torch_metrics_match_numpy():
     pt = np.round(rng.rand(n), 2)
     m = create_metric("auc")
     assert abs(m(pt, info) - m(torch.from_numpy(pt), info)) < 1e-10
+
+
+def test_degenerate_auc_is_nan():
+    """reference auc.cc:351: single-class AUC -> NaN + warning."""
+    import warnings
+    import xgboost_amd as xgb
+    X = np.random.RandomState(0).randn(30, 2).astype(np.float32)
+    d = xgb.DMatrix(X, label=np.ones(30, np.float32))
+    res = {}
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        xgb.train({"objective": "binary:logistic", "eval_metric": "auc",
+                   "max_depth": 2}, d, 1, evals=[(d, "t")],
+                  evals_result=res, verbose_eval=False)
+    assert np.isnan(res["t"]["auc"][0])

@@ -287,7 +287,9 @@ def auc(preds, info, param=None):
                                   device=dev))
             a, valid = _binary_auc_t(p.view(-1), y, w)
             s_, v_ = collective.allreduce_sum_scalars([a * valid, valid])
-            return s_ / max(v_, 1e-16)
+            if v_ == 0:
+                return _degenerate_auc()
+            return s_ / v_
         preds = p.cpu().numpy()
     p = np.asarray(preds, dtype=np.float64)
     if p.ndim == 2 and p.shape[1] > 1:
@@ -300,7 +302,17 @@ def auc(preds, info, param=None):
     a, valid = _binary_auc(p.reshape(-1), y, w)
     # distributed: weighted mean of per-worker AUC (reference auc.cc:125)
     s, v = collective.allreduce_sum_scalars([a * valid, valid])
-    return s / max(v, 1e-16)
+    if v == 0:
+        return _degenerate_auc()
+    return s / v
+
+
+def _degenerate_auc() -> float:
+    """reference auc.cc:351: AUC over a single-class dataset is NaN."""
+    import warnings
+    warnings.warn("Dataset is empty, or contains only positive or "
+                  "negative samples.")
+    return float("nan")
 
 
 def _binary_auc_t(p, y, w) -> Tuple[float, float]:
