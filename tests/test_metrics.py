@@ -187,3 +187,22 @@ def test_degenerate_auc_is_nan():
                    "max_depth": 2}, d, 1, evals=[(d, "t")],
                   evals_result=res, verbose_eval=False)
     assert np.isnan(res["t"]["auc"][0])
+
+
+def test_rank_metric_minus_suffix():
+    """reference rank_metric.cc:385/:446: a group with no relevant docs
+    scores 1 by default and 0 with the `-` suffix, in both cases
+    remaining in the denominator."""
+    from xgboost_amd.data import MetaInfo
+    from xgboost_amd.metrics import create_metric
+    info = MetaInfo()
+    info.labels = np.array([1, 0, 0, 0, 0, 0], np.float32)
+    info.num_row = 6
+    info.group_ptr = np.array([0, 3, 6], np.int64)  # group 2: no positives
+    preds = np.array([0.9, 0.1, 0.2, 0.5, 0.4, 0.3], np.float64)
+    for name in ("ndcg", "map"):
+        full = create_metric(f"{name}@3")(preds, info)
+        minus = create_metric(f"{name}@3-")(preds, info)
+        # group1 scores 1.0 (perfect); group2: 1 vs 0
+        assert full == pytest.approx((1.0 + 1.0) / 2)
+        assert minus == pytest.approx((1.0 + 0.0) / 2)

@@ -434,9 +434,10 @@ def ndcg(preds, info, param=None):
         yy, pp = y[s:e], p[s:e]
         k = min(topn, e - s)
         if yy.sum() == 0:
-            if not minus:
-                scores.append(1.0)
-                nvalid += 1
+            # reference rank_metric.cc:385: no-relevance groups score
+            # minus ? 0 : 1 and STAY in the denominator
+            scores.append(0.0 if minus else 1.0)
+            nvalid += 1
             continue
         order = np.argsort(-pp, kind="stable")
         gains = (2.0 ** yy - 1.0)
@@ -452,6 +453,7 @@ def ndcg(preds, info, param=None):
 @register("map")
 def map_metric(preds, info, param=None):
     topn = int(param.rstrip("-")) if param else 2 ** 31 - 1
+    minus = bool(param and param.endswith("-"))
     p = np.asarray(preds, dtype=np.float64).reshape(-1)
     y = np.asarray(info.labels, dtype=np.float64).reshape(-1)
     gp = _groups(info, p.size)
@@ -464,7 +466,10 @@ def map_metric(preds, info, param=None):
         hits = np.cumsum(rel)
         prec_at = rel[:k] * (hits[:k] / np.arange(1, k + 1))
         npos = rel.sum()
-        scores.append(float(prec_at.sum() / npos) if npos > 0 else 1.0)
+        # reference rank_metric.cc:446: no-relevance groups score
+        # minus ? 0 : 1 and stay in the denominator
+        scores.append(float(prec_at.sum() / npos) if npos > 0
+                      else (0.0 if minus else 1.0))
         nvalid += 1
     sa, sv = collective.allreduce_sum_scalars([float(np.sum(scores)), nvalid])
     return sa / max(sv, 1e-16)
