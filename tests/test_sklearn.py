@@ -202,3 +202,23 @@ def test_sklearn_meta_estimator_compat():
                      ("m", xgb.XGBRegressor(n_estimators=4))])
     pipe.fit(X, X[:, 0])
     assert pipe.score(X, X[:, 0]) > 0.5
+
+
+def test_sklearn_callable_eval_metric():
+    """reference _metric_decorator: eval_metric may be an sklearn-style
+    callable (y_true, y_pred) used as the custom metric."""
+    from sklearn.metrics import mean_absolute_error
+    rng = np.random.RandomState(3)
+    X = rng.randn(120, 4).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    m = xgb.XGBRegressor(n_estimators=5,
+                         eval_metric=mean_absolute_error)
+    m.fit(X, y, eval_set=[(X, y)], verbose=False)
+    vals = m.evals_result_["validation_0"]["mean_absolute_error"]
+    assert len(vals) == 5 and vals[-1] < vals[0]
+    # early stopping on the callable metric
+    m2 = xgb.XGBRegressor(n_estimators=50,
+                          eval_metric=mean_absolute_error,
+                          early_stopping_rounds=3)
+    m2.fit(X, y, eval_set=[(X, y)], verbose=False)
+    assert m2.get_booster().best_iteration is not None

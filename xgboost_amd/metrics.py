@@ -403,7 +403,7 @@ def aucpr(preds, info, param=None):
     fp = np.cumsum(w * (1 - y))
     total_pos = tp[-1]
     if total_pos == 0:
-        return 0.0
+        return _degenerate_auc()
     prec = tp / np.clip(tp + fp, 1e-16, None)
     rec = tp / total_pos
     rec0 = np.concatenate([[0.0], rec[:-1]])

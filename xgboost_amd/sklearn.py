@@ -183,9 +183,21 @@ class XGBModel(_SklBaseEstimator):
                 return user_fn(dmat.get_label(), preds)
         elif self.objective is None:
             params.setdefault("objective", self._default_objective())
+        custom_metric = None
+        if callable(getattr(self, "eval_metric", None)):
+            # reference _metric_decorator: sklearn-style metrics take
+            # (y_true, y_pred); predictions arrive transformed for
+            # builtin objectives
+            metric_fn = self.eval_metric
+            metric_name = getattr(metric_fn, "__name__", "custom_metric")
+            params.pop("eval_metric", None)
+
+            def custom_metric(preds, dmat):  # noqa: ANN001
+                return metric_name, float(metric_fn(dmat.get_label(), preds))
         self.evals_result_ = {}
         self._Booster = _train(
             params, dtrain, self.n_estimators_, evals=evals, obj=obj,
+            custom_metric=custom_metric,
             early_stopping_rounds=getattr(self, "early_stopping_rounds", None),
             evals_result=self.evals_result_, verbose_eval=verbose,
             xgb_model=xgb_model,
