@@ -385,3 +385,19 @@ def test_pandas_nullable_and_bad_dtypes():
     with pytest.raises(ValueError, match="int, float, bool or category"):
         xgb.DMatrix(pd.DataFrame({"t": pd.date_range("2020", periods=5)}),
                     label=y[:5])
+
+
+def test_pred_leaf_strict_shape():
+    """reference strict_shape leaf predictions:
+    (n, n_iterations, n_groups, n_parallel_tree)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(40, 3).astype(np.float32)
+    y3 = rng.randint(0, 3, 40).astype(np.float32)
+    d = xgb.DMatrix(X, label=y3)
+    bst = xgb.train({"objective": "multi:softprob", "num_class": 3,
+                     "num_parallel_tree": 2, "max_depth": 2}, d, 4)
+    leaves = bst.predict(d, pred_leaf=True)
+    assert leaves.shape == (40, 24)  # 4 iters * 3 classes * 2 trees
+    strict = bst.predict(d, pred_leaf=True, strict_shape=True)
+    assert strict.shape == (40, 4, 3, 2)
+    assert np.array_equal(strict.reshape(40, -1), leaves)

@@ -1064,7 +1064,16 @@ class Booster:
             X = data.raw_data()
             out = np.stack([self.trees[t].predict_leaf_np(X, data.missing)
                             for t in range(lo, hi)], axis=1)
-            return out.astype(np.float32)
+            out = out.astype(np.float32)
+            if strict_shape:
+                # reference: (n, n_iterations, n_groups, n_parallel_tree)
+                npt = max(1, self.tparam.num_parallel_tree)
+                per_iter = max(1, (self.iteration_indptr[1]
+                                   - self.iteration_indptr[0]))
+                n_iter = (hi - lo) // per_iter
+                out = out.reshape(out.shape[0], n_iter,
+                                  per_iter // npt, npt)
+            return out
         if pred_interactions:
             from .shap import shap_interactions
             return shap_interactions(self, data, iteration_range)
