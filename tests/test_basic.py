@@ -401,3 +401,24 @@ def test_pred_leaf_strict_shape():
     strict = bst.predict(d, pred_leaf=True, strict_shape=True)
     assert strict.shape == (40, 4, 3, 2)
     assert np.array_equal(strict.reshape(40, -1), leaves)
+
+
+def test_booster_iter_and_index_bounds():
+    """Booster.__iter__ yields per-iteration slices; out-of-range int
+    indexing raises IndexError (the legacy iteration protocol otherwise
+    loops forever)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(40, 3).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    bst = xgb.train({"max_depth": 2}, xgb.DMatrix(X, label=y), 3)
+    parts = list(bst)
+    assert len(parts) == 3
+    assert all(p.num_boosted_rounds() == 1 for p in parts)
+    with pytest.raises(IndexError):
+        bst[3]
+    assert bst[-1].num_boosted_rounds() == 1  # negative indexing
+    d = xgb.DMatrix(X)
+    total = sum(p.predict(d, output_margin=True) - bst.base_score
+                for p in parts) + bst.base_score
+    assert np.allclose(total, bst.predict(d, output_margin=True),
+                       atol=1e-5)

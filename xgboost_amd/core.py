@@ -1401,8 +1401,19 @@ class Booster:
         b.load_model(self.save_raw("json"))
         return b
 
+    def __iter__(self):
+        """Per-iteration slices (reference core.py Booster.__iter__)."""
+        for i in range(self.num_boosted_rounds()):
+            yield self[i]
+
     def __getitem__(self, val) -> "Booster":
         if isinstance(val, int):
+            n = self.num_boosted_rounds()
+            if val < -n or val >= n:
+                raise IndexError(
+                    f"index {val} out of range for {n} boosted rounds")
+            if val < 0:
+                val += n
             val = slice(val, val + 1)
         lo, hi, step = val.indices(self.num_boosted_rounds())
         b = Booster(self.raw_params)
