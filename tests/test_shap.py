@@ -178,3 +178,18 @@ def test_path_pair_decomposition_matches_exact_interactions():
         out[:, i_, i_] = base[:, i_] - (out[:, i_, :].sum(axis=-1)
                                         - out[:, i_, i_])
     assert np.allclose(out, ref, atol=1e-4), np.abs(out - ref).max()
+
+
+def test_contribs_strict_shape():
+    """reference strict_shape: contribs (n, groups, ncol+1),
+    interactions (n, groups, ncol+1, ncol+1)."""
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(30, 3).astype(np.float32)
+    bst = xgb.train({"max_depth": 2},
+                    xgb.DMatrix(X, label=X[:, 0]), 2)
+    d = xgb.DMatrix(X)
+    assert bst.predict(d, pred_contribs=True,
+                       strict_shape=True).shape == (30, 1, 4)
+    assert bst.predict(d, pred_interactions=True,
+                       strict_shape=True).shape == (30, 1, 4, 4)

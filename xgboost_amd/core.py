@@ -1076,10 +1076,17 @@ class Booster:
             return out
         if pred_interactions:
             from .shap import shap_interactions
-            return shap_interactions(self, data, iteration_range)
+            out = shap_interactions(self, data, iteration_range)
+            if strict_shape and out.ndim == 3:
+                out = out[:, None]  # (n, groups=1, ncol+1, ncol+1)
+            return out
         if pred_contribs:
             from .shap import shap_values
-            return shap_values(self, data, iteration_range, approx=approx_contribs)
+            out = shap_values(self, data, iteration_range,
+                              approx=approx_contribs)
+            if strict_shape and out.ndim == 2:
+                out = out[:, None]  # (n, groups=1, ncol+1)
+            return out
         margin = self._predict_margin(data, iteration_range)
         if output_margin:
             res = margin
