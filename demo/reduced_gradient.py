@@ -7,6 +7,12 @@ iteration instead of n_targets trees.  Reference analog:
 demo/guide-python/multioutput_reduced_gradient.py +
 XGBoosterTrainOneIterWithSplitGrad.
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
 import numpy as np
 
 import xgboost_amd as xgb
