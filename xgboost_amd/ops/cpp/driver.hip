@@ -79,6 +79,11 @@ struct DriverCtx {
   // rewritten before every launch (replays read them at execution).
   hipGraphExec_t wt_graph = nullptr;
   std::vector<unsigned long long> wt_key;
+  // capture-on-second-sight: a key must repeat once before we pay the
+  // capture+instantiate cost (fresh per-round contexts — e.g. the gpu
+  // approx path rebuilds its ops every iteration — would otherwise
+  // capture every round and never replay)
+  std::vector<unsigned long long> wt_seen_key;
   void* wtc_host = nullptr;
   void* wtc_dev = nullptr;
   size_t wtc_cap = 0;
@@ -1115,7 +1120,10 @@ int gbt_grow_tree(
           ctx->wt_key.clear();
         }
       }
-      if (!enqueued &&
+      if (!enqueued && key != ctx->wt_seen_key) {
+        // first sighting of this configuration: run direct, remember it
+        ctx->wt_seen_key = key;
+      } else if (!enqueued &&
           hipStreamBeginCapture(ctx->gstream,
                                 hipStreamCaptureModeRelaxed) ==
               hipSuccess) {
