@@ -14,7 +14,6 @@ pred_transform applied first except for output_margin-style metrics).
 from __future__ import annotations
 
 import math
-import re
 from typing import Callable, Dict, Tuple
 
 import numpy as np

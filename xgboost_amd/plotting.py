@@ -7,7 +7,6 @@ from typing import Optional
 
 import numpy as np
 
-from .core import Booster
 
 
 def plot_importance(booster, ax=None, height: float = 0.2,

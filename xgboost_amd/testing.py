@@ -5,7 +5,7 @@ Offline analog of the reference's public ``xgboost.testing`` helpers
 regression data, learning-to-rank groups, sparse regression matrices,
 categorical frames — generated locally instead of downloaded.
 """
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 

@@ -5,7 +5,6 @@ Reference behavior: python-package/xgboost/training.py:53 (train),
 """
 from __future__ import annotations
 
-import copy
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
