@@ -1108,8 +1108,8 @@ int gbt_grow_tree(
       // ~0.5 ms anyway, so the few-us host block is in the noise.
       HIP_CHECK(hipEventRecord(ctx->gevent, stream));
       HIP_CHECK(hipEventSynchronize(ctx->gevent));
-      if (int e = stage_h2d(ctx->gstream)) return e;
       if (ctx->wt_graph != nullptr && key == ctx->wt_key) {
+        if (int e = stage_h2d(ctx->gstream)) return e;
         if (hipGraphLaunch(ctx->wt_graph, ctx->gstream) == hipSuccess) {
           enqueued = true;
           if (WtGraphDebug()) fprintf(stderr, "[wtgraph] replay\n");
@@ -1123,10 +1123,14 @@ int gbt_grow_tree(
       if (!enqueued && key != ctx->wt_seen_key) {
         // first sighting of this configuration: run direct, remember it
         ctx->wt_seen_key = key;
-      } else if (!enqueued &&
-          hipStreamBeginCapture(ctx->gstream,
-                                hipStreamCaptureModeRelaxed) ==
-              hipSuccess) {
+      } else if (!enqueued) {
+        if (int e = stage_h2d(ctx->gstream)) return e;
+        if (hipStreamBeginCapture(ctx->gstream,
+                                  hipStreamCaptureModeRelaxed) !=
+            hipSuccess) {
+          (void)hipGetLastError();
+          goto graph_done;
+        }
         int ec = enqueue_chain(ctx->gstream);
         hipGraph_t g = nullptr;
         hipError_t ce = hipStreamEndCapture(ctx->gstream, &g);
@@ -1165,6 +1169,7 @@ int gbt_grow_tree(
         // above leaves enqueued=false and falls through to the direct
         // enqueue below
       }
+    graph_done:
       if (!enqueued) {
         // safety: never leave the graph stream wedged mid-capture
         hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
