@@ -422,3 +422,20 @@ def test_booster_iter_and_index_bounds():
                 for p in parts) + bst.base_score
     assert np.allclose(total, bst.predict(d, output_margin=True),
                        atol=1e-5)
+
+
+def test_inplace_predict_options():
+    """iteration_range slices and base_margin adds (reference
+    inplace_predict surface)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(50, 3).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    bst = xgb.train({"max_depth": 2, "base_score": 0.0, "seed": 1},
+                    xgb.DMatrix(X, label=y), 4)
+    p2 = bst.inplace_predict(X, iteration_range=(0, 2))
+    bst2 = xgb.train({"max_depth": 2, "base_score": 0.0, "seed": 1},
+                     xgb.DMatrix(X, label=y), 2)
+    assert np.allclose(p2, bst2.predict(xgb.DMatrix(X)), atol=1e-6)
+    bm = np.full(50, 1.5, np.float32)
+    assert np.allclose(bst.inplace_predict(X, base_margin=bm),
+                       bst.inplace_predict(X) + 1.5, atol=1e-5)

@@ -222,3 +222,19 @@ def test_sklearn_callable_eval_metric():
                           early_stopping_rounds=3)
     m2.fit(X, y, eval_set=[(X, y)], verbose=False)
     assert m2.get_booster().best_iteration is not None
+
+
+def test_rf_wrapper_defaults():
+    """reference XGBRF defaults: one boosting round of n_estimators
+    parallel trees, eta=1, subsample=0.8, colsample_bynode=0.8,
+    reg_lambda=1e-5 (sklearn.py:2053-2056)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(80, 3).astype(np.float32)
+    m = xgb.XGBRFRegressor(n_estimators=6).fit(X, X[:, 0])
+    b = m.get_booster()
+    assert b.num_boosted_rounds() == 1
+    assert len(b.trees) == 6
+    tp = b.tparam
+    assert (tp.num_parallel_tree, tp.eta, tp.subsample,
+            tp.colsample_bynode) == (6, 1.0, 0.8, 0.8)
+    assert tp.reg_lambda == pytest.approx(1e-5)
