@@ -331,3 +331,25 @@ def test_group_sizes_must_cover_rows():
     with pytest.raises(ValueError, match="Invalid group structure"):
         xgb.DMatrix(X, label=np.zeros(30, np.float32), group=[40])
     d.set_info(group=[15, 15])
+
+
+def test_lambdarank_bool_params_survive_model_io():
+    """Model JSON stores lambdarank booleans as "0"/"1" strings —
+    loading must not read "0" as truthy."""
+    import xgboost_amd as xgb
+    X = np.random.RandomState(0).randn(60, 3).astype(np.float32)
+    y = np.random.RandomState(1).randint(0, 3, 60).astype(np.float32)
+    d = xgb.DMatrix(X, label=y)
+    d.set_info(group=[20, 20, 20])
+    b = xgb.train({"objective": "rank:ndcg", "max_depth": 2,
+                   "lambdarank_normalization": 0, "ndcg_exp_gain": 0,
+                   "lambdarank_score_normalization": 0}, d, 2)
+    b2 = xgb.Booster()
+    b2.load_model(bytearray(b.save_raw("json")))
+    o = b2.objective
+    assert (o.normalize, o.exp_gain, o.score_norm) == (False, False, False)
+    b3 = xgb.Booster()
+    b3.load_model(bytearray(b.save_raw("ubj")))
+    o3 = b3.objective
+    assert (o3.normalize, o3.exp_gain, o3.score_norm) == (
+        False, False, False)

@@ -620,6 +620,17 @@ def _weighted_quantile(v: np.ndarray, w: Optional[np.ndarray], q: float) -> floa
 # pair generation lambdarank_obj.cuh:76, NDCG deltas ranking_utils.h)
 
 
+def _pbool(v, default: bool) -> bool:
+    """Parse a boolean objective parameter: model JSON stores them as
+    "0"/"1" strings, so bool("0") must not read as True."""
+    if v is None:
+        return default
+    if isinstance(v, str):
+        return v.strip().lower() in ("1", "true", "yes")
+    return bool(v)
+
+
+
 class _LambdaRankBase(Objective):
     """LambdaMART gradients, fully vectorized torch ops (device-resident
     when predictions live on the GPU).
@@ -646,12 +657,13 @@ class _LambdaRankBase(Objective):
                                             0) or 0)
         self.pair_method = str(self.params.get("lambdarank_pair_method",
                                                "topk"))
-        self.normalize = bool(self.params.get("lambdarank_normalization",
-                                              True))
-        self.score_norm = bool(self.params.get(
-            "lambdarank_score_normalization", True))
-        self.exp_gain = bool(self.params.get("ndcg_exp_gain", True))
-        self.unbiased = bool(self.params.get("lambdarank_unbiased", False))
+        self.normalize = _pbool(
+            self.params.get("lambdarank_normalization"), True)
+        self.score_norm = _pbool(
+            self.params.get("lambdarank_score_normalization"), True)
+        self.exp_gain = _pbool(self.params.get("ndcg_exp_gain"), True)
+        self.unbiased = _pbool(self.params.get("lambdarank_unbiased"),
+                               False)
         self.bias_norm = float(self.params.get("lambdarank_bias_norm", 1.0))
         # Unbiased LambdaMART position-bias ratios ti+ / tj- (reference
         # lambdarank_obj.cc:133-136), carried across boosting iterations.
