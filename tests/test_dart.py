@@ -219,3 +219,27 @@ def test_dart_with_cv_and_early_stopping():
                     evals=[(d, "t")], callbacks=[es], verbose_eval=False)
     assert bst.num_boosted_rounds() <= 30
     assert len(bst.weight_drop) == len(bst.trees)
+
+
+def test_dart_heavy_feature_combination():
+    """DART composed with multiclass, parallel trees, monotone,
+    colsample, subsample and weights: margin cache stays consistent
+    with a fresh weighted predict and the model round-trips."""
+    import torch
+    rng = np.random.RandomState(0)
+    X = rng.randn(1000, 6).astype(np.float32)
+    y = rng.randint(0, 3, 1000).astype(np.float32)
+    w = rng.rand(1000).astype(np.float32) + 0.5
+    d = xgb.DMatrix(X, label=y, weight=w)
+    b = xgb.train({
+        "objective": "multi:softprob", "num_class": 3, "max_depth": 4,
+        "rate_drop": 0.3, "one_drop": True,
+        "monotone_constraints": "(1,0,0,0,0,-1)",
+        "colsample_bytree": 0.8, "subsample": 0.9,
+        "num_parallel_tree": 2, "seed": 5}, d, 6)
+    assert len(b.trees) == len(b.weight_drop) == 36
+    cached, _ = b._cache[id(d)]
+    assert torch.allclose(cached, b._predict_margin(d), atol=1e-3)
+    b2 = xgb.Booster()
+    b2.load_model(bytearray(b.save_raw("ubj")))
+    assert np.allclose(b2.predict(d), b.predict(d), atol=1e-5)
