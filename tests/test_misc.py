@@ -480,3 +480,13 @@ def test_param_lower_bounds():
         xgb.train({"eta": -0.1, "max_depth": 2}, d, 1)
     with pytest.raises(ValueError, match="max_depth"):
         xgb.train({"max_depth": -2}, d, 1)
+
+
+def test_gpu_id_deprecated_spelling():
+    """reference maps gpu_id=N to device=cuda:N; the name must pass
+    parameter validation."""
+    import xgboost_amd as xgb
+    b = xgb.Booster({"gpu_id": 1, "validate_parameters": 1})
+    assert str(b.device) == "cuda:1"
+    b2 = xgb.Booster({"device": "cpu", "gpu_id": 0})
+    assert b2.device.type == "cpu"  # explicit device wins

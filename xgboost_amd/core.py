@@ -32,6 +32,9 @@ VERSION = (3, 5, 0)
 
 def _resolve_device(params: Dict[str, Any]) -> torch.device:
     dev = str(params.get("device", "cpu"))
+    if "device" not in params and params.get("gpu_id") is not None:
+        # deprecated reference spelling: gpu_id=N means cuda:N
+        dev = f"cuda:{int(params['gpu_id'])}"
     if dev in ("cuda", "gpu"):
         return torch.device("cuda", torch.cuda.current_device()
                             if torch.cuda.is_available() else 0)

@@ -147,6 +147,8 @@ LEARNER_PARAMS = {
     "expectile_alpha",
     # DART dropout (gbtree-level, reference gbm/gbtree.h DartTrainParam)
     "rate_drop", "one_drop", "skip_drop", "sample_type", "normalize_type",
+    # deprecated device spelling (reference maps gpu_id -> device)
+    "gpu_id",
 }
 
 _KNOWN = {f.name for f in dataclasses.fields(TrainParam)} | {"lambda", "alpha"} | LEARNER_PARAMS | set(ALIASES)
