@@ -144,3 +144,24 @@ def test_exact_rejects_unsupported():
     with pytest.raises(ValueError, match="colsample_bynode|column sample"):
         xgb.train({"tree_method": "exact", "max_depth": 2,
                    "colsample_bynode": 0.5}, d, 1)
+
+
+def test_hist_equals_exact_on_discrete_data():
+    """Differential oracle: with few distinct feature values (every
+    value its own bin), the hist updater must find the SAME trees as
+    the exact enumeration (reference relationship between
+    grow_quantile_histmaker and grow_colmaker)."""
+    rng = np.random.RandomState(0)
+    X = rng.randint(0, 20, (500, 4)).astype(np.float32)
+    y = (X[:, 0] * 0.5 - X[:, 1] * 0.2 + rng.randn(500)).astype(
+        np.float32)
+    p = {"max_depth": 4, "base_score": 0.0, "reg_lambda": 1.0}
+    bh = xgb.train(dict(p, tree_method="hist", max_bin=256),
+                   xgb.DMatrix(X, label=y), 3)
+    be = xgb.train(dict(p, tree_method="exact"),
+                   xgb.DMatrix(X, label=y), 3)
+    d = xgb.DMatrix(X)
+    assert np.allclose(bh.predict(d), be.predict(d), atol=1e-6)
+    for a, b in zip(bh.trees, be.trees):
+        assert np.array_equal(a.split_index[:a.n_nodes],
+                              b.split_index[:b.n_nodes])
