@@ -165,3 +165,19 @@ def test_hist_equals_exact_on_discrete_data():
     for a, b in zip(bh.trees, be.trees):
         assert np.array_equal(a.split_index[:a.n_nodes],
                               b.split_index[:b.n_nodes])
+
+
+def test_approx_equals_hist_with_uniform_hessian():
+    """With reg:squarederror the hessian is 1 everywhere, so the
+    hessian-weighted re-sketch reduces to plain quantiles and approx
+    must reproduce hist exactly (same max_bin)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(800, 4).astype(np.float32)
+    y = (X[:, 0] - X[:, 1]).astype(np.float32)
+    p = {"max_depth": 3, "base_score": 0.0, "max_bin": 64}
+    bh = xgb.train(dict(p, tree_method="hist"),
+                   xgb.DMatrix(X, label=y), 3)
+    ba = xgb.train(dict(p, tree_method="approx"),
+                   xgb.DMatrix(X, label=y), 3)
+    d = xgb.DMatrix(X)
+    assert np.allclose(bh.predict(d), ba.predict(d), atol=1e-7)
