@@ -129,3 +129,23 @@ def test_gblinear_feature_importance_is_coefficients():
     assert abs(s["f0"]) > abs(s["f2"])  # real coefficient magnitudes
     with pytest.raises(ValueError, match="weight"):
         bst.get_score(importance_type="gain")
+
+
+def test_gblinear_converges_to_ridge_solution():
+    """Differential oracle: coordinate descent must converge to the
+    closed-form ridge solution with the reference's DENORMALIZED
+    penalty (lambda * sum_instance_weight, gblinear.cc
+    DenormalizePenalties)."""
+    rng = np.random.RandomState(0)
+    n, f = 500, 4
+    X = rng.randn(n, f).astype(np.float32)
+    w_true = np.array([1.0, -2.0, 0.5, 0.0], np.float32)
+    y = (X @ w_true).astype(np.float32)
+    lam = 1.0
+    bst = xgb.train({"booster": "gblinear", "eta": 0.8, "lambda": lam,
+                     "alpha": 0.0, "base_score": 0.0},
+                    xgb.DMatrix(X, label=y), 300)
+    w_fit = bst._linear.weights[:-1, 0].numpy()
+    A = X.T @ X + lam * n * np.eye(f)
+    w_ridge = np.linalg.solve(A, X.T @ y)
+    assert np.abs(w_fit - w_ridge).max() < 5e-3
