@@ -168,3 +168,31 @@ def test_aft_uncensored():
 def test_unknown_objective_raises():
     with pytest.raises(ValueError):
         create_objective("not:a:loss")
+
+
+def test_cox_gradient_matches_autograd():
+    """survival:cox gradient vs torch autograd of the Breslow partial
+    likelihood (negative labels = censored, reference convention)."""
+    rng = np.random.RandomState(0)
+    n = 60
+    t = rng.exponential(2, n).astype(np.float32)
+    event = rng.rand(n) > 0.3
+    y = np.where(event, t, -t).astype(np.float32)
+    info = MetaInfo()
+    info.labels = y
+    info.num_row = n
+    obj = create_objective("survival:cox")
+    margin = torch.tensor(rng.randn(n, 1).astype(np.float32))
+    g, _ = obj.get_gradient(margin, info, 0)
+    m = margin.double().clone().requires_grad_(True)
+    times = torch.tensor(np.abs(y).astype(np.float64))
+    exp_m = torch.exp(m.view(-1))
+    nll = torch.zeros((), dtype=torch.float64)
+    for i in range(n):
+        if not event[i]:
+            continue
+        risk = times >= times[i]
+        nll = nll - (m.view(-1)[i] - torch.log(exp_m[risk].sum()))
+    nll.backward()
+    assert np.abs(g.view(-1).numpy()
+                  - m.grad.view(-1).numpy()).max() < 1e-5
