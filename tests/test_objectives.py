@@ -196,3 +196,50 @@ def test_cox_gradient_matches_autograd():
     nll.backward()
     assert np.abs(g.view(-1).numpy()
                   - m.grad.view(-1).numpy()).max() < 1e-5
+
+
+@pytest.mark.parametrize("dist", ["normal", "logistic", "extreme"])
+def test_aft_interval_gradient_matches_autograd(dist):
+    """survival:aft censored-interval gradients vs torch autograd of
+    the interval likelihood CDF(zu) - CDF(zl) (probability_distribution
+    .h formulas).  Rows whose interval likelihood underflows float32
+    are excluded (cancellation noise, same as the reference's fp32
+    evaluation)."""
+    rng = np.random.RandomState(0)
+    n = 50
+    lb = rng.exponential(2, n).astype(np.float32) + 0.1
+    ub = lb + rng.exponential(1, n).astype(np.float32)
+    ub[::5] = np.inf
+    lb2 = lb.copy()
+    lb2[::7] = 0.0
+    info = MetaInfo()
+    info.num_row = n
+    info.labels = lb2
+    info.label_lower_bound = lb2
+    info.label_upper_bound = ub
+    obj = create_objective("survival:aft", {
+        "aft_loss_distribution": dist,
+        "aft_loss_distribution_scale": 1.1})
+    margin = torch.tensor(rng.randn(n, 1).astype(np.float32))
+    g, _ = obj.get_gradient(margin, info, 0)
+    m = margin.double().clone().requires_grad_(True)
+    s = 1.1
+    yl = torch.tensor(lb2.astype(np.float64)).clamp(min=1e-12)
+    yu = torch.tensor(ub.astype(np.float64))
+    zl = (torch.log(yl) - m.view(-1)) / s
+    zu = (torch.log(yu.clamp(max=1e12)) - m.view(-1)) / s
+    if dist == "normal":
+        def cdf(z):
+            return 0.5 * (1 + torch.erf(z / np.sqrt(2)))
+    elif dist == "logistic":
+        cdf = torch.sigmoid
+    else:
+        def cdf(z):
+            return 1 - torch.exp(-torch.exp(z))
+    like = torch.where(torch.tensor(np.isfinite(ub)),
+                       cdf(zu) - cdf(zl), 1 - cdf(zl))
+    nll = -torch.log(like.clamp(min=1e-300)).sum()
+    nll.backward()
+    ok = like.detach().numpy() > 1e-5
+    diff = np.abs(g.view(-1).numpy() - m.grad.view(-1).numpy())[ok]
+    assert diff.max() < 5e-3, diff.max()
