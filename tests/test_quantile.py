@@ -84,3 +84,23 @@ def test_sentinel_covers_max():
         local = g[:, f] - cuts.ptrs[f]
         assert local.max() < cuts.n_bins(f)
         assert local.min() >= 0
+
+
+def test_quantile_dmatrix_trains_identically_to_dmatrix():
+    """In-core QuantileDMatrix carries the same cuts/bins as quantizing
+    a plain DMatrix, so hist training must be bit-identical; a ref'd
+    validation QDM bins with the training cuts."""
+    import xgboost_amd as xgb
+    rng = np.random.RandomState(0)
+    X = rng.randn(600, 5).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    p = {"max_depth": 4, "max_bin": 64, "seed": 3}
+    b1 = xgb.train(p, xgb.DMatrix(X, label=y), 5)
+    b2 = xgb.train(p, xgb.QuantileDMatrix(X, label=y, max_bin=64), 5)
+    assert b1.get_dump(with_stats=True) == b2.get_dump(with_stats=True)
+    dtr = xgb.QuantileDMatrix(X, label=y, max_bin=64)
+    dva = xgb.QuantileDMatrix(X[:100], label=y[:100], max_bin=64, ref=dtr)
+    res = {}
+    xgb.train(p, dtr, 3, evals=[(dva, "v")], evals_result=res,
+              verbose_eval=False)
+    assert len(res["v"]["rmse"]) == 3
