@@ -140,3 +140,26 @@ def test_dart_with_sparse_training():
     fresh = bst._predict_margin(d)
     assert torch.allclose(cached, fresh, atol=1e-4), \
         (cached - fresh).abs().max()
+
+
+def test_sparse_equals_dense_nan_training():
+    """CSR missing entries behave exactly like NaN in a dense matrix:
+    first-round trees match exactly; later rounds may break exact gain
+    TIES differently between the two code paths, but predictions
+    coincide on the training data."""
+    from scipy import sparse
+    rng = np.random.RandomState(0)
+    csr = sparse.random(500, 20, density=0.2, format="csr",
+                        dtype=np.float32, random_state=rng)
+    y = np.asarray(csr @ rng.randn(20)).ravel().astype(np.float32)
+    p = {"max_depth": 4, "max_bin": 64, "seed": 1}
+    Xd = np.full((500, 20), np.nan, np.float32)
+    coo = csr.tocoo()
+    Xd[coo.row, coo.col] = coo.data
+    bs = xgb.train(p, xgb.DMatrix(csr, label=y), 1)
+    bd = xgb.train(p, xgb.DMatrix(Xd, label=y), 1)
+    assert bs.get_dump(with_stats=True) == bd.get_dump(with_stats=True)
+    bs5 = xgb.train(p, xgb.DMatrix(csr, label=y), 5)
+    bd5 = xgb.train(p, xgb.DMatrix(Xd, label=y), 5)
+    assert np.allclose(bs5.predict(xgb.DMatrix(Xd)),
+                       bd5.predict(xgb.DMatrix(Xd)), atol=1e-6)
