@@ -692,3 +692,37 @@ def test_two_process_dart_identical_models():
     assert r0["dump"] == r1["dump"]
     assert r0["wd"] == pytest.approx(r1["wd"])
     assert any(w != 1.0 for w in r0["wd"])
+
+
+def test_tracker_done_and_error_paths():
+    """RabitTracker bookkeeping: wait_for returns once every worker
+    posts done; a posted error surfaces via .error and makes wait_for
+    raise (reference collective tracker error port semantics)."""
+    from xgboost_amd.tracker import (RabitTracker, connect_tracker,
+                                     post_done, post_error)
+    t = RabitTracker(n_workers=2)
+    t.start()
+    args = t.worker_args()
+    stores = [connect_tracker(args["DMLC_TRACKER_URI"],
+                              args["DMLC_TRACKER_PORT"], 2, rank=None,
+                              timeout_s=30)
+              for _ in range(2)]
+    ranks = sorted(r for _, r in stores)
+    assert ranks == [0, 1]  # tracker-assigned ranks
+    for st, _ in stores:
+        post_done(st)
+    t.wait_for(timeout=30)  # returns, no raise
+    assert t.error() is None
+    t.free()
+
+    t2 = RabitTracker(n_workers=2)
+    t2.start()
+    a2 = t2.worker_args()
+    s0, _ = connect_tracker(a2["DMLC_TRACKER_URI"],
+                            a2["DMLC_TRACKER_PORT"], 2, rank=None,
+                            timeout_s=30)
+    post_error(s0, "worker exploded")
+    with pytest.raises(RuntimeError, match="worker exploded"):
+        t2.wait_for(timeout=10)
+    assert "exploded" in (t2.error() or "")
+    t2.free()
