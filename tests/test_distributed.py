@@ -726,3 +726,33 @@ def test_tracker_done_and_error_paths():
         t2.wait_for(timeout=10)
     assert "exploded" in (t2.error() or "")
     t2.free()
+
+
+WORLD4_WORKER = r"""
+import os, pickle, sys
+import numpy as np
+sys.path.insert(0, os.environ["XGB_AMD_REPO"])
+import xgboost_amd as xgb
+from xgboost_amd import collective
+collective.init("gloo")
+rank, world = collective.get_rank(), collective.get_world_size()
+rng = np.random.RandomState(0)
+n, f = 2000, 6
+X = rng.randn(n, f).astype(np.float32)
+y = (X[:, 0] - X[:, 1] > 0).astype(np.float32)
+sl = slice(rank * n // world, (rank + 1) * n // world)
+bst = xgb.train({"objective": "binary:logistic", "max_depth": 4,
+                 "seed": 3, "debug_synchronize": True},
+                xgb.DMatrix(X[sl], label=y[sl]), 5)
+if rank == 0:
+    with open(os.environ["XGB_AMD_OUT"], "wb") as fh:
+        pickle.dump(bst.get_dump(), fh)
+collective.finalize()
+"""
+
+
+def test_four_process_training():
+    """World size 4 (beyond the usual 2): rank layout generality +
+    debug_synchronize's per-iteration tree-equality allreduce check."""
+    dump = _run_workers(4, WORLD4_WORKER)
+    assert len(dump) == 5
