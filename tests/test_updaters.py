@@ -181,3 +181,20 @@ def test_approx_equals_hist_with_uniform_hessian():
                    xgb.DMatrix(X, label=y), 3)
     d = xgb.DMatrix(X)
     assert np.allclose(bh.predict(d), ba.predict(d), atol=1e-7)
+
+
+def test_explicit_updater_selects_growth_algorithm():
+    """reference gbtree specified_updater_: updater="grow_colmaker,
+    prune" runs the exact enumeration; grow_quantile_histmaker -> hist."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(150, 3).astype(np.float32)
+    d = xgb.DMatrix(X, label=X[:, 0])
+    b = xgb.train({"updater": "grow_colmaker,prune", "max_depth": 3},
+                  d, 2)
+    assert b.tparam.tree_method == "exact"
+    b2 = xgb.train({"updater": "grow_colmaker,prune", "max_depth": 3,
+                    "tree_method": "exact"}, d, 2)
+    assert b.get_dump() == b2.get_dump()
+    b3 = xgb.train({"updater": "grow_quantile_histmaker",
+                    "max_depth": 3}, d, 2)
+    assert b3.tparam.tree_method == "hist"

@@ -97,6 +97,22 @@ def make_train_param(params: Dict[str, Any]) -> TrainParam:
             setattr(tp, f.name, v)
     if tp.tree_method in _DEPRECATED_TREE_METHODS:
         tp.tree_method = _DEPRECATED_TREE_METHODS[tp.tree_method]
+    if tp.updater and tp.process_type == "default" \
+            and tp.tree_method == "auto":
+        # explicit updater sequences select the growth algorithm
+        # (reference gbtree.cc specified_updater_): prune/refresh/sync
+        # suffixes need no separate pass here — gamma gating during
+        # growth subsumes the prune updater's loss_chg test
+        _grow_updaters = {"grow_colmaker": "exact",
+                          "grow_histmaker": "approx",
+                          "grow_quantile_histmaker": "hist",
+                          "grow_gpu_hist": "hist",
+                          "grow_gpu_approx": "approx"}
+        for u in str(tp.updater).split(","):
+            u = u.strip()
+            if u in _grow_updaters:
+                tp.tree_method = _grow_updaters[u]
+                break
     _validate(tp)
     return tp
 
