@@ -439,3 +439,19 @@ def test_inplace_predict_options():
     bm = np.full(50, 1.5, np.float32)
     assert np.allclose(bst.inplace_predict(X, base_margin=bm),
                        bst.inplace_predict(X) + 1.5, atol=1e-5)
+
+
+def test_dmatrix_slice_propagates_meta():
+    """DMatrix.slice(rindex) carries labels/weights for the selected
+    rows (reference DMatrix::SliceCol/Slice semantics)."""
+    rng = np.random.RandomState(0)
+    X = rng.randn(100, 3).astype(np.float32)
+    y = X[:, 0].astype(np.float32)
+    d = xgb.DMatrix(X, label=y, weight=np.arange(100, dtype=np.float32))
+    sub = d.slice([5, 10, 20])
+    assert sub.num_row() == 3
+    assert np.allclose(sub.get_label(), y[[5, 10, 20]])
+    assert np.allclose(sub.get_weight(), [5, 10, 20])
+    bst = xgb.train({"max_depth": 2}, d, 2)
+    assert np.allclose(bst.predict(sub),
+                       bst.predict(d)[[5, 10, 20]], atol=1e-6)
